@@ -1,0 +1,196 @@
+"""Worker for the 2-process collective correctness battery.
+
+Launched by tests/test_collective_2proc_gpu.py with RANK/WORLD_SIZE env.
+Both ranks may share one physical GPU (gpurun boxes have a single MI355X):
+HIP IPC handles open fine across processes on the same device, so the full
+multi-rank protocol (flags, LL slots, parity scratch) is exercised even
+when xGMI itself is not.
+
+Every collective result is checked against a torch fp32/exact reference
+computed from deterministic per-rank inputs.
+"""
+
+from __future__ import annotations
+
+import os
+import signal
+import sys
+
+signal.alarm(int(os.environ.get("UCCL_TEST_ALARM", "240")))
+
+import torch
+import torch.distributed as dist
+
+
+def make_input(rank: int, count: int, dtype, seed_tag: int) -> torch.Tensor:
+    g = torch.Generator().manual_seed(1234 + 97 * rank + seed_tag)
+    if dtype == torch.int32:
+        return torch.randint(-1000, 1000, (count,), generator=g,
+                             dtype=dtype).cuda()
+    return torch.randn(count, generator=g, dtype=torch.float32).to(dtype).cuda()
+
+
+def expected_sum(world: int, count: int, dtype, seed_tag: int) -> torch.Tensor:
+    # fp32 (or int64) reference accumulation on CPU, then cast once — this is
+    # exactly what the kernels' AccumV16 fp32 accumulators implement.
+    if dtype == torch.int32:
+        acc = torch.zeros(count, dtype=torch.int64)
+        for r in range(world):
+            g = torch.Generator().manual_seed(1234 + 97 * r + seed_tag)
+            acc += torch.randint(-1000, 1000, (count,), generator=g,
+                                 dtype=torch.int32).to(torch.int64)
+        return acc.to(torch.int32)
+    acc = torch.zeros(count, dtype=torch.float32)
+    for r in range(world):
+        g = torch.Generator().manual_seed(1234 + 97 * r + seed_tag)
+        acc += torch.randn(count, generator=g, dtype=torch.float32).to(
+            dtype).to(torch.float32)
+    return acc.to(dtype)
+
+
+def check(name: str, got: torch.Tensor, want: torch.Tensor, tol: float):
+    got = got.float().cpu()
+    want = want.float().cpu()
+    if tol == 0:
+        ok = torch.equal(got, want)
+    else:
+        ok = torch.allclose(got, want, rtol=tol, atol=tol)
+    if not ok:
+        diff = (got - want).abs().max().item()
+        raise AssertionError(f"{name}: max diff {diff}")
+    print(f"[rank {dist.get_rank()}] {name} OK", flush=True)
+
+
+def main():
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.cuda.set_device(int(os.environ.get("UCCL_TEST_DEVICE", "0")))
+
+    import uccl_amd.collective as ucol
+
+    comm = ucol.init(device=torch.cuda.current_device())
+
+    tol = {torch.float32: 0.0, torch.int32: 0.0,
+           torch.bfloat16: 1e-2, torch.float16: 1e-3}
+
+    seed = 0
+    # --- allreduce across the three algorithm paths -------------------------
+    # LL (<=32KB), one-shot (<=2MB), two-shot (>2MB), chunked two-shot
+    for count, label in [(1000, "ll"), (100000, "oneshot"),
+                         (3 << 20, "twoshot"), (1000003, "oneshot-odd")]:
+        for dtype in (torch.float32, torch.bfloat16, torch.float16,
+                      torch.int32):
+            seed += 1
+            x = make_input(rank, count, dtype, seed)
+            comm.all_reduce(x)
+            torch.cuda.synchronize()
+            check(f"allreduce[{label},{dtype}]", x,
+                  expected_sum(world, count, dtype, seed), tol[dtype])
+
+    # chunked two-shot: shrink scratch via a dedicated small-heap comm is
+    # heavy; instead pick a size above the default 2MB oneshot threshold and
+    # large enough to need >1 chunk only if scratch < size. Default scratch
+    # is tens of MB, so force chunking with a big-ish message.
+    seed += 1
+    count = 48 << 20  # 192MB fp32 > default per-parity scratch
+    x = make_input(rank, count, torch.float32, seed)
+    comm.all_reduce(x)
+    torch.cuda.synchronize()
+    check("allreduce[chunked]", x,
+          expected_sum(world, count, torch.float32, seed), 0.0)
+
+    # --- allgather ----------------------------------------------------------
+    seed += 1
+    n = 12345
+    mine = make_input(rank, n, torch.float32, seed)
+    out = torch.empty(world * n, dtype=torch.float32, device="cuda")
+    comm.all_gather(out, mine)
+    torch.cuda.synchronize()
+    want = torch.cat([
+        make_input(r, n, torch.float32, seed).cpu() for r in range(world)])
+    check("allgather", out, want, 0.0)
+
+    # --- reduce_scatter -----------------------------------------------------
+    seed += 1
+    per = 4096
+    inp = make_input(rank, per * world, torch.float32, seed)
+    out = torch.empty(per, dtype=torch.float32, device="cuda")
+    comm.reduce_scatter(out, inp)
+    torch.cuda.synchronize()
+    full = expected_sum(world, per * world, torch.float32, seed)
+    check("reduce_scatter", out, full[rank * per:(rank + 1) * per], 0.0)
+
+    # --- broadcast ----------------------------------------------------------
+    seed += 1
+    x = make_input(rank, 9999, torch.bfloat16, seed)
+    comm.broadcast(x, root=0)
+    torch.cuda.synchronize()
+    check("broadcast", x, make_input(0, 9999, torch.bfloat16, seed).cpu(),
+          0.0)
+
+    # --- all_to_all ---------------------------------------------------------
+    seed += 1
+    per = 2048
+    inp = make_input(rank, per * world, torch.float32, seed)
+    out = torch.empty_like(inp)
+    comm.all_to_all(out, inp)
+    torch.cuda.synchronize()
+    want = torch.cat([
+        make_input(r, per * world, torch.float32,
+                   seed)[rank * per:(rank + 1) * per].cpu()
+        for r in range(world)
+    ])
+    check("all_to_all", out, want, 0.0)
+
+    # --- send/recv (pairwise ring) ------------------------------------------
+    seed += 1
+    n = 300000  # spans >1 p2p slot chunk at 2MB slots? 1.2MB -> single chunk
+    if world >= 2:
+        peer_to = (rank + 1) % world
+        peer_from = (rank - 1 + world) % world
+        payload = make_input(rank, n, torch.float32, seed)
+        got = torch.empty(n, dtype=torch.float32, device="cuda")
+        if rank % 2 == 0:
+            comm.send(payload, peer_to)
+            comm.recv(got, peer_from)
+        else:
+            comm.recv(got, peer_from)
+            comm.send(payload, peer_to)
+        torch.cuda.synchronize()
+        check("send/recv", got, make_input(peer_from, n, torch.float32,
+                                           seed).cpu(), 0.0)
+
+        # multi-chunk send (> 2MB slot)
+        seed += 1
+        n2 = 1500000  # 6MB fp32 -> 3 chunks
+        payload = make_input(rank, n2, torch.float32, seed)
+        got = torch.empty(n2, dtype=torch.float32, device="cuda")
+        if rank % 2 == 0:
+            comm.send(payload, peer_to)
+            comm.recv(got, peer_from)
+        else:
+            comm.recv(got, peer_from)
+            comm.send(payload, peer_to)
+        torch.cuda.synchronize()
+        check("send/recv-chunked", got,
+              make_input(peer_from, n2, torch.float32, seed).cpu(), 0.0)
+
+    # --- barrier + interleave stress ---------------------------------------
+    comm.barrier()
+    for i in range(20):  # rapid-fire small collectives: parity/seq stress
+        seed += 1
+        x = make_input(rank, 257, torch.float32, seed)
+        comm.all_reduce(x)
+        torch.cuda.synchronize()
+        check(f"stress[{i}]", x, expected_sum(world, 257, torch.float32,
+                                              seed), 0.0)
+
+    dist.barrier()
+    if rank == 0:
+        print("ALL COLLECTIVE TESTS PASSED", flush=True)
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

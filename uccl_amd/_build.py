@@ -1,0 +1,126 @@
+"""In-tree build driver for the uccl_amd native extension.
+
+Compiles the HIP/C++ sources directly with hipcc for gfx950 (no hipify, no
+CUDA compatibility layer) and links against the torch-ROCm libraries. The
+resulting .so lives inside the package tree so it travels with repo
+snapshots (gpurun) and is importable without installation.
+"""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import sysconfig
+from pathlib import Path
+
+PKG_DIR = Path(__file__).resolve().parent
+CSRC = PKG_DIR / "csrc"
+ROCM = Path(os.environ.get("ROCM_PATH", "/opt/rocm"))
+HIPCC = str(ROCM / "bin" / "hipcc")
+GPU_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+
+SOURCES = [
+    CSRC / "collective" / "kernels.hip",
+    CSRC / "collective" / "communicator.cpp",
+    CSRC / "ep" / "ep_kernels.hip",
+    CSRC / "ep" / "ep_buffer.cpp",
+    CSRC / "bindings" / "module.cpp",
+]
+
+EXT_NAME = "_C" + (sysconfig.get_config_var("EXT_SUFFIX") or ".so")
+
+
+def _torch_paths():
+    import torch
+
+    troot = Path(torch.__file__).parent
+    return [troot / "include", troot / "include/torch/csrc/api/include"], troot / "lib"
+
+
+def _stale(target: Path, deps) -> bool:
+    if not target.exists():
+        return True
+    t = target.stat().st_mtime
+    return any(d.stat().st_mtime > t for d in deps)
+
+
+def build(verbose: bool = False, force: bool = False) -> Path:
+    """Compile the extension in-tree. Returns the path to the built .so."""
+    target = PKG_DIR / EXT_NAME
+    incs, libdir = _torch_paths()
+    headers = list(CSRC.rglob("*.h"))
+    sources = [s for s in SOURCES if s.exists()]
+    if not force and not _stale(target, sources + headers):
+        return target
+
+    objdir = PKG_DIR / ".build"
+    objdir.mkdir(exist_ok=True)
+
+    py_inc = sysconfig.get_paths()["include"]
+    common = [
+        f"--offload-arch={GPU_ARCH}",
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        "-D__HIP_PLATFORM_AMD__=1",
+        "-DUSE_ROCM=1",
+        "-DTORCH_EXTENSION_NAME=_C",
+        "-D_GLIBCXX_USE_CXX11_ABI=1",
+        "-fno-gpu-rdc",
+        "-Wno-unused-result",
+        f"-I{py_inc}",
+    ] + [f"-I{i}" for i in incs]
+
+    objs = []
+    procs = []
+    for src in sources:
+        obj = objdir / (src.stem + ".o")
+        objs.append(obj)
+        if not force and not _stale(obj, [src] + headers):
+            continue
+        cmd = [HIPCC, "-c", str(src), "-o", str(obj)] + common
+        if verbose:
+            print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)
+        procs.append((src, subprocess.Popen(cmd, stdout=subprocess.PIPE,
+                                            stderr=subprocess.STDOUT)))
+    failed = False
+    for src, p in procs:
+        out, _ = p.communicate()
+        if p.returncode != 0:
+            failed = True
+            print(f"[uccl_amd build] FAILED {src}:\n{out.decode()}",
+                  file=sys.stderr)
+        elif verbose and out:
+            print(out.decode(), file=sys.stderr)
+    if failed:
+        raise RuntimeError("uccl_amd native build failed")
+
+    link = (
+        [HIPCC, "-shared", "-o", str(target)]
+        + [str(o) for o in objs]
+        + [
+            f"-L{libdir}",
+            "-ltorch",
+            "-ltorch_cpu",
+            "-ltorch_hip",
+            "-ltorch_python",
+            "-lc10",
+            "-lc10_hip",
+            "-lamdhip64",
+            f"-Wl,-rpath,{libdir}",
+            f"-L{ROCM}/lib",
+            f"-Wl,-rpath,{ROCM}/lib",
+        ]
+    )
+    if verbose:
+        print("[uccl_amd build]", " ".join(link), file=sys.stderr)
+    r = subprocess.run(link, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    if r.returncode != 0:
+        raise RuntimeError(f"uccl_amd link failed:\n{r.stdout.decode()}")
+    return target
+
+
+if __name__ == "__main__":
+    build(verbose=True, force="--force" in sys.argv)
+    print("built:", PKG_DIR / EXT_NAME)

@@ -1,0 +1,128 @@
+// pybind11 module `uccl_amd._C` — Python surface over the native engine.
+// Parity role: the reference's nanobind modules uccl.p2p (p2p/engine_api.cc)
+// and uccl.ep (ep/src/uccl_ep.cc:1783+); here a single extension hosts the
+// collective Communicator (and, as they land, the p2p Endpoint and EP
+// Buffer), built HIP-native for gfx950.
+
+#include <torch/extension.h>
+
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <hip/hip_runtime.h>
+
+#include "../collective/communicator.h"
+#include "../core/log.h"
+
+namespace py = pybind11;
+using uccl::Communicator;
+using uccl::Dtype;
+
+namespace {
+
+hipStream_t current_stream(int device) {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA(device).stream();
+}
+
+Dtype dtype_of(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return Dtype::kF32;
+    case at::kHalf: return Dtype::kF16;
+    case at::kBFloat16: return Dtype::kBF16;
+    case at::kInt: return Dtype::kI32;
+    default:
+      TORCH_CHECK(false, "uccl_amd: unsupported dtype ", t.scalar_type());
+  }
+}
+
+void check_tensor(const at::Tensor& t) {
+  TORCH_CHECK(t.is_cuda(), "uccl_amd: tensor must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), "uccl_amd: tensor must be contiguous");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_C, m) {
+  m.doc() = "uccl_amd native engine (MI355X / gfx950)";
+
+  m.def("device_count", [] {
+    int n = 0;
+    (void)hipGetDeviceCount(&n);
+    return n;
+  });
+
+  py::class_<Communicator>(m, "Communicator")
+      .def(py::init([](int rank, int world, int device, size_t heap_bytes) {
+             return new Communicator(rank, world, device, heap_bytes);
+           }),
+           py::arg("rank"), py::arg("world"), py::arg("device"),
+           py::arg("heap_bytes") = 0)
+      .def_property_readonly("rank", &Communicator::rank)
+      .def_property_readonly("world", &Communicator::world)
+      .def_property_readonly("device", &Communicator::device)
+      .def_property_readonly("scratch_capacity",
+                             &Communicator::scratch_capacity_bytes)
+      .def("handle_bytes",
+           [](Communicator& c) { return py::bytes(c.handle_bytes()); })
+      .def("connect",
+           [](Communicator& c, const std::vector<std::string>& handles) {
+             c.connect(handles);
+           })
+      .def("all_reduce",
+           [](Communicator& c, at::Tensor t) {
+             check_tensor(t);
+             c.all_reduce(t.data_ptr(), t.numel(), dtype_of(t),
+                          current_stream(c.device()));
+           })
+      .def("all_gather",
+           [](Communicator& c, at::Tensor out, at::Tensor in) {
+             check_tensor(out);
+             check_tensor(in);
+             TORCH_CHECK(out.numel() == in.numel() * c.world(),
+                         "all_gather: out must have world*numel(in) elems");
+             TORCH_CHECK(out.scalar_type() == in.scalar_type());
+             c.all_gather(out.data_ptr(), in.data_ptr(), in.numel(),
+                          dtype_of(in), current_stream(c.device()));
+           })
+      .def("reduce_scatter",
+           [](Communicator& c, at::Tensor out, at::Tensor in) {
+             check_tensor(out);
+             check_tensor(in);
+             TORCH_CHECK(in.numel() == out.numel() * c.world(),
+                         "reduce_scatter: in must have world*numel(out)");
+             TORCH_CHECK(out.scalar_type() == in.scalar_type());
+             c.reduce_scatter(out.data_ptr(), in.data_ptr(), out.numel(),
+                              dtype_of(in), current_stream(c.device()));
+           })
+      .def("broadcast",
+           [](Communicator& c, at::Tensor t, int root) {
+             check_tensor(t);
+             c.broadcast(t.data_ptr(), t.numel(), dtype_of(t), root,
+                         current_stream(c.device()));
+           })
+      .def("all_to_all",
+           [](Communicator& c, at::Tensor out, at::Tensor in) {
+             check_tensor(out);
+             check_tensor(in);
+             TORCH_CHECK(out.numel() == in.numel());
+             TORCH_CHECK(in.numel() % c.world() == 0,
+                         "all_to_all: numel must divide world");
+             TORCH_CHECK(out.scalar_type() == in.scalar_type());
+             c.all_to_all(out.data_ptr(), in.data_ptr(),
+                          in.numel() / c.world(), dtype_of(in),
+                          current_stream(c.device()));
+           })
+      .def("send",
+           [](Communicator& c, at::Tensor t, int dst) {
+             check_tensor(t);
+             c.send(t.data_ptr(), t.numel() * t.element_size(), dst,
+                    current_stream(c.device()));
+           })
+      .def("recv",
+           [](Communicator& c, at::Tensor t, int src) {
+             check_tensor(t);
+             c.recv(t.data_ptr(), t.numel() * t.element_size(), src,
+                    current_stream(c.device()));
+           })
+      .def("barrier", [](Communicator& c) {
+        c.barrier(current_stream(c.device()));
+      });
+}

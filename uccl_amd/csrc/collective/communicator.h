@@ -1,0 +1,95 @@
+// Host-side Communicator for the xGMI collective engine.
+//
+// Role parity: the comm-lifecycle half of the reference's lite-collective
+// NCCL drop-in (experimental/lite/lite-collective/nccl/nccl.cu:1455
+// ncclCommInitRank + collective entry points), re-designed for MI355X:
+// symmetric HIP-IPC heaps per rank, algorithm selection by message size
+// (LL packet / one-shot fullmesh / two-shot RS+AG push), host-sequenced
+// send/recv with per-destination staging slots and ack credits.
+//
+// Bootstrap (exchanging the 64-byte IPC handles + device ids) is left to
+// the caller — in Python, torch.distributed (gloo or any backend) or any
+// out-of-band channel; this class is transport-agnostic about rendezvous,
+// the same split as the reference's TCP-OOB bootstrap.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <array>
+#include <cstdint>
+#include <string>
+#include <vector>
+
+#include "kernels.h"
+#include "layout.h"
+
+namespace uccl {
+
+struct IpcInfo {
+  hipIpcMemHandle_t handle;
+  int device;
+  int pid;
+};
+
+class Communicator {
+ public:
+  Communicator(int rank, int world, int device, size_t heap_bytes);
+  ~Communicator();
+
+  Communicator(const Communicator&) = delete;
+  Communicator& operator=(const Communicator&) = delete;
+
+  // Serialized IpcInfo for this rank, to be exchanged out-of-band.
+  std::string handle_bytes() const;
+
+  // Install all ranks' serialized IpcInfo blobs (ordered by rank).
+  void connect(const std::vector<std::string>& all_handles);
+
+  int rank() const { return rank_; }
+  int world() const { return world_; }
+  int device() const { return device_; }
+  size_t scratch_capacity_bytes() const { return scratch_cap_; }
+
+  // All ops are asynchronous on `stream` and in-place where natural.
+  void all_reduce(void* data, size_t count, Dtype dt, hipStream_t stream);
+  void all_gather(void* out, void const* in, size_t count_per_rank, Dtype dt,
+                  hipStream_t stream);
+  void reduce_scatter(void* out, void const* in, size_t count_per_rank,
+                      Dtype dt, hipStream_t stream);
+  void broadcast(void* data, size_t count, Dtype dt, int root,
+                 hipStream_t stream);
+  void all_to_all(void* out, void const* in, size_t count_per_rank, Dtype dt,
+                  hipStream_t stream);
+  void send(void const* data, size_t bytes, int dst, hipStream_t stream);
+  void recv(void* data, size_t bytes, int src, hipStream_t stream);
+  void barrier(hipStream_t stream);
+
+ private:
+  CommView view(uint64_t seq) const;
+  uint64_t next_seq() {
+    uint64_t s = seq_;
+    seq_ += 2;
+    return s;
+  }
+
+  int rank_, world_, device_;
+  size_t heap_bytes_, scratch_cap_;
+  void* heap_ = nullptr;
+  std::array<void*, kMaxRanks> peers_{};
+  std::array<bool, kMaxRanks> ipc_opened_{};
+  uint64_t seq_ = 2;  // flag regions start zeroed; first live seq must be >0
+  // per-destination send / per-source recv sequence counters (p2p channel)
+  std::array<uint64_t, kMaxRanks> send_seq_{};
+  std::array<uint64_t, kMaxRanks> recv_seq_{};
+  bool connected_ = false;
+
+  // thresholds (env-tunable)
+  size_t ll_threshold_;
+  size_t oneshot_threshold_;
+
+  static constexpr int kCollCh = 0;
+  static constexpr int kP2PDataCh = 2;
+  static constexpr int kP2PAckCh = 3;
+};
+
+}  // namespace uccl

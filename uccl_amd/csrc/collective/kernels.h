@@ -1,0 +1,45 @@
+// Host-side launcher declarations for the CDNA4 collective kernels.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include "layout.h"
+
+namespace uccl {
+
+enum class Dtype : int { kF32 = 0, kF16 = 1, kBF16 = 2, kI32 = 3 };
+
+inline size_t dtype_size(Dtype d) {
+  switch (d) {
+    case Dtype::kF16:
+    case Dtype::kBF16: return 2;
+    default: return 4;
+  }
+}
+
+void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s);
+void launch_oneshot_allreduce(const CommView& cv, void* out, size_t count,
+                              Dtype dt, hipStream_t s);
+void launch_twoshot_rs_push(const CommView& cv, size_t count, Dtype dt,
+                            hipStream_t s);
+void launch_twoshot_copyout(const CommView& cv, void* out, size_t bytes,
+                            hipStream_t s);
+void launch_ll_allreduce(const CommView& cv, void const* in, void* out,
+                         size_t count, Dtype dt, hipStream_t s);
+void launch_allgather_pull(const CommView& cv, void* out, size_t chunk_bytes,
+                           hipStream_t s);
+void launch_reducescatter_pull(const CommView& cv, void* out, size_t count,
+                               Dtype dt, hipStream_t s);
+void launch_broadcast_pull(const CommView& cv, int root, void* out,
+                           size_t bytes, hipStream_t s);
+void launch_alltoall_pull(const CommView& cv, void* out, size_t chunk_bytes,
+                          hipStream_t s);
+void launch_barrier(const CommView& cv, hipStream_t s);
+void launch_signal_peer(const CommView& cv, int dst, int ch, uint64_t val,
+                        hipStream_t s);
+void launch_wait_peer(const CommView& cv, int src, int ch, uint64_t val,
+                      hipStream_t s);
+void launch_copy_from_peer(const CommView& cv, int src, size_t src_off,
+                           void* dst, size_t bytes, hipStream_t s);
+
+}  // namespace uccl

@@ -1,0 +1,241 @@
+// CDNA4 (gfx950) device-side primitives for the uccl_amd collective engine:
+// system-scope flag signal/wait, LL (low-latency) 8-byte flagged packets,
+// vectorized copy/reduce helpers. This is the uccl_amd analog of the
+// reference's device channel substrate (experimental/lite/lite-collective
+// core/*_device.hpp and ep/include/ring_buffer.cuh) re-designed for the
+// CDNA4 memory model: wave64, per-XCD non-coherent L2, xGMI peer-HBM
+// access. All cross-GPU ordering goes through __hip_atomic_* at
+// __HIP_MEMORY_SCOPE_SYSTEM (release stores emit the L2 writeback, acquire
+// loads emit the invalidate) — the verified-fence discipline the reference
+// learned the hard way (ep/README.md:139 "aggressive atomics").
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#include <cstdint>
+
+namespace uccl {
+namespace device {
+
+constexpr int kWave = 64;  // CDNA wavefront width (not 32!)
+
+// ---------------------------------------------------------------------------
+// System-scope atomics: the only legal way to order data across GPUs on xGMI.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void st_release_sys(uint64_t* p, uint64_t v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+__device__ __forceinline__ void st_relaxed_sys(uint64_t* p, uint64_t v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+__device__ __forceinline__ uint64_t ld_acquire_sys(uint64_t const* p) {
+  return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+__device__ __forceinline__ uint64_t ld_relaxed_sys(uint64_t const* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+// Scope "" means system scope in the amdgcn fence builtin.
+__device__ __forceinline__ void fence_release_sys() {
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+}
+
+__device__ __forceinline__ void fence_acquire_sys() {
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+}
+
+// Nanosleep-style backoff for spin loops (CDNA has s_sleep, no __nanosleep).
+__device__ __forceinline__ void backoff() { __builtin_amdgcn_s_sleep(2); }
+
+// Bounded spin-wait: waits until *flag >= target.  Traps (aborting the
+// kernel, and with it the process) instead of hanging the GPU forever —
+// a deadlocked collective should kill the job, not wedge the node.
+__device__ __forceinline__ void wait_flag_ge(uint64_t const* flag,
+                                             uint64_t target) {
+  // ~2.7e8 polls with s_sleep backoff ≈ tens of seconds: far beyond any
+  // sane collective wait, but short enough that a deadlock aborts the
+  // kernel instead of wedging the GPU (gpurun strike avoidance).
+  for (uint64_t i = 0; i < (1ull << 28); ++i) {
+    if (ld_acquire_sys(flag) >= target) return;
+    backoff();
+  }
+  __builtin_trap();
+}
+
+// ---------------------------------------------------------------------------
+// LL packet: 8 bytes = {u32 data, u32 flag}, written/read as one system-scope
+// 64-bit atomic so data+flag arrive together over xGMI (same design point as
+// NCCL LL / lite-collective allreduce_packet.cu, sized for wave64 here).
+// ---------------------------------------------------------------------------
+
+union LLPacket {
+  uint64_t u64;
+  struct {
+    uint32_t data;
+    uint32_t flag;
+  };
+};
+
+__device__ __forceinline__ void ll_write(uint64_t* slot, uint32_t data,
+                                         uint32_t flag) {
+  LLPacket p;
+  p.data = data;
+  p.flag = flag;
+  st_relaxed_sys(slot, p.u64);
+}
+
+__device__ __forceinline__ uint32_t ll_read(uint64_t const* slot,
+                                            uint32_t flag) {
+  LLPacket p;
+  for (uint64_t i = 0; i < (1ull << 28); ++i) {
+    p.u64 = ld_relaxed_sys(slot);
+    if (p.flag == flag) return p.data;
+    backoff();
+  }
+  __builtin_trap();
+}
+
+// ---------------------------------------------------------------------------
+// 16-byte vector type for coalesced HBM/xGMI traffic (16 B/lane × wave64 =
+// 1 KiB per instruction).
+// ---------------------------------------------------------------------------
+
+union alignas(16) V16 {
+  uint4 u;
+  float f32[4];
+  __hip_bfloat16 bf16[8];
+  __half f16[8];
+  unsigned short u16[8];
+};
+
+// Elementwise fp32-accumulate add of two 16B vectors of T.
+template <typename T>
+__device__ __forceinline__ V16 v16_add(V16 a, V16 b);
+
+template <>
+__device__ __forceinline__ V16 v16_add<float>(V16 a, V16 b) {
+  V16 r;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) r.f32[i] = a.f32[i] + b.f32[i];
+  return r;
+}
+
+template <>
+__device__ __forceinline__ V16 v16_add<__hip_bfloat16>(V16 a, V16 b) {
+  V16 r;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    r.bf16[i] = __float2bfloat16(__bfloat162float(a.bf16[i]) +
+                                 __bfloat162float(b.bf16[i]));
+  return r;
+}
+
+template <>
+__device__ __forceinline__ V16 v16_add<__half>(V16 a, V16 b) {
+  V16 r;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    r.f16[i] = __float2half(__half2float(a.f16[i]) + __half2float(b.f16[i]));
+  return r;
+}
+
+template <>
+__device__ __forceinline__ V16 v16_add<int>(V16 a, V16 b) {
+  V16 r;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    reinterpret_cast<int*>(r.f32)[i] = reinterpret_cast<int*>(a.f32)[i] +
+                                       reinterpret_cast<int*>(b.f32)[i];
+  return r;
+}
+
+// fp32-accumulator variant: unpack T into float lanes, accumulate exactly,
+// pack once at the end (used by the reduce kernels so bf16 sums don't lose
+// bits per-step with world_size up to 8).
+template <typename T, int N>
+struct AccumV16;
+
+template <int N>
+struct AccumV16<float, N> {
+  float v[4];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) v[i] = a.f32[i];
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) v[i] += a.f32[i];
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) r.f32[i] = v[i];
+    return r;
+  }
+};
+
+template <int N>
+struct AccumV16<__hip_bfloat16, N> {
+  float v[8];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v[i] = __bfloat162float(a.bf16[i]);
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v[i] += __bfloat162float(a.bf16[i]);
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) r.bf16[i] = __float2bfloat16(v[i]);
+    return r;
+  }
+};
+
+template <int N>
+struct AccumV16<__half, N> {
+  float v[8];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v[i] = __half2float(a.f16[i]);
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v[i] += __half2float(a.f16[i]);
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) r.f16[i] = __float2half(v[i]);
+    return r;
+  }
+};
+
+template <int N>
+struct AccumV16<int, N> {
+  int v[4];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) v[i] = reinterpret_cast<int const*>(a.f32)[i];
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) v[i] += reinterpret_cast<int const*>(a.f32)[i];
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) reinterpret_cast<int*>(r.f32)[i] = v[i];
+    return r;
+  }
+};
+
+}  // namespace device
+}  // namespace uccl
