@@ -10,6 +10,10 @@
 
 namespace uccl {
 
+static inline unsigned all_mask(int world) {
+  return world >= 32 ? 0xffffffffu : ((1u << world) - 1u);
+}
+
 Communicator::Communicator(int rank, int world, int device, size_t heap_bytes)
     : rank_(rank), world_(world), device_(device) {
   UCCL_CHECK(world >= 1 && world <= kMaxRanks)
@@ -120,6 +124,7 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
       char* p = static_cast<char*>(data) + off * es;
       CommView const cv = view(next_seq());
       launch_copy(static_cast<char*>(heap_) + cv.sa_off, p, n * es, stream);
+      launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
       launch_oneshot_allreduce(cv, p, n, dt, stream);
     }
     return;
@@ -139,10 +144,12 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
     char* p = static_cast<char*>(data) + off * es;
     CommView const cv = view(next_seq());
     launch_copy(static_cast<char*>(heap_) + cv.sa_off, p, n * es, stream);
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     if (oneshot) {
       launch_oneshot_allreduce(cv, p, n, dt, stream);
     } else {
       launch_twoshot_rs_push(cv, n, dt, stream);
+      launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
       launch_twoshot_copyout(cv, p, n * es, stream);
     }
   }
@@ -162,11 +169,11 @@ void Communicator::all_gather(void* out, void const* in, size_t count_per_rank,
     CommView const cv = view(next_seq());
     launch_copy(static_cast<char*>(heap_) + cv.sa_off,
                 static_cast<char const*>(in) + off * es, n * es, stream);
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     if (n == count_per_rank) {
       launch_allgather_pull(cv, out, n * es, stream);
     } else {
-      // chunked: barrier then strided per-peer pulls into out[p][off..]
-      launch_barrier(cv, stream);
+      // chunked: strided per-peer pulls into out[p][off..]
       for (int p = 0; p < world_; ++p) {
         launch_copy_from_peer(
             cv, p, cv.sa_off,
@@ -199,6 +206,7 @@ void Communicator::reduce_scatter(void* out, void const* in,
                       (static_cast<size_t>(p) * count_per_rank + off) * es,
                   n * es, stream);
     }
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     launch_reducescatter_pull(cv, static_cast<char*>(out) + off * es, n, dt,
                               stream);
   }
@@ -217,6 +225,9 @@ void Communicator::broadcast(void* data, size_t count, Dtype dt, int root,
       launch_copy(static_cast<char*>(heap_) + cv.sa_off,
                   static_cast<char*>(data) + off * es, n * es, stream);
     }
+    // wait ALL (not just root): a rank's next-call signal is what proves
+    // its reads of this parity finished (scratch-reuse safety)
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     launch_broadcast_pull(cv, root, static_cast<char*>(data) + off * es,
                           n * es, stream);
   }
@@ -242,11 +253,11 @@ void Communicator::all_to_all(void* out, void const* in, size_t count_per_rank,
                       (static_cast<size_t>(p) * count_per_rank + off) * es,
                   n * es, stream);
     }
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     if (n == count_per_rank) {
       launch_alltoall_pull(cv, out, n * es, stream);
     } else {
-      // chunked: barrier then strided per-peer pulls
-      launch_barrier(cv, stream);
+      // chunked: strided per-peer pulls
       for (int p = 0; p < world_; ++p) {
         launch_copy_from_peer(
             cv, p, cv.sa_off + static_cast<size_t>(rank_) * n * es,

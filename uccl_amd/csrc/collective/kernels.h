@@ -34,6 +34,8 @@ void launch_broadcast_pull(const CommView& cv, int root, void* out,
                            size_t bytes, hipStream_t s);
 void launch_alltoall_pull(const CommView& cv, void* out, size_t chunk_bytes,
                           hipStream_t s);
+void launch_signal_wait(const CommView& cv, uint64_t val, unsigned wait_mask,
+                        hipStream_t s);
 void launch_barrier(const CommView& cv, hipStream_t s);
 void launch_signal_peer(const CommView& cv, int dst, int ch, uint64_t val,
                         hipStream_t s);

@@ -50,21 +50,22 @@ __device__ inline char* scratch_b(void* base, const CommView& cv) {
   return reinterpret_cast<char*>(base) + cv.sb_off;
 }
 
-// Signal `val` on our writer slot in every rank's heap (including our own),
-// then wait until every rank has signalled `val` into our heap. Every block
-// performs the wait (the acquire load invalidates that block's XCD caches —
-// required before reading peer data; see primitives.h).
-__device__ inline void signal_all(const CommView& cv, uint64_t val) {
-  if (blockIdx.x == 0 && threadIdx.x < static_cast<unsigned>(cv.world)) {
-    st_release_sys(flag_ptr(cv.peers[threadIdx.x], cv.rank, cv.channel), val);
-  }
-}
-
-__device__ inline void wait_all(const CommView& cv, uint64_t val) {
+// Flag round = one tiny single-block kernel: signal `val` on our writer
+// slot in every rank's heap, then spin until every rank in `wait_mask` has
+// signalled `val` into ours. Runs as its own dispatch so that (a) worker
+// kernels never spin — no co-residency/occupancy deadlock risk at any
+// grid size, and (b) cross-GPU visibility is inherited from the HSA
+// dispatch boundaries: the dispatch-completion system release publishes
+// the preceding stream-ordered kernel's stores, and the next dispatch's
+// system acquire invalidates stale caches before worker kernels read
+// peer HBM.
+__global__ void k_signal_wait(CommView cv, uint64_t val, unsigned wait_mask) {
   if (threadIdx.x < static_cast<unsigned>(cv.world)) {
-    wait_flag_ge(flag_ptr(cv.peers[cv.rank], threadIdx.x, cv.channel), val);
+    st_release_sys(flag_ptr(cv.peers[threadIdx.x], cv.rank, cv.channel), val);
+    if ((wait_mask >> threadIdx.x) & 1u) {
+      wait_flag_ge(flag_ptr(cv.peers[cv.rank], threadIdx.x, cv.channel), val);
+    }
   }
-  __syncthreads();
 }
 
 // ---------------------------------------------------------------------------
@@ -95,9 +96,6 @@ __global__ void k_copy(void* __restrict__ dst, void const* __restrict__ src,
 template <typename T>
 __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
                                     size_t count) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
-
   size_t const vper = 16 / sizeof(T);
   size_t const nvec = count / vper;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -129,8 +127,6 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
 template <>
 __global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
                                          size_t count) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
   size_t const vper = 4;
   size_t const nvec = count / vper;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -166,9 +162,6 @@ __global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
 
 template <typename T>
 __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
-
   size_t const vper = 16 / sizeof(T);
   size_t const nvec = count / vper;
   size_t const shard = (nvec + cv.world - 1) / cv.world;
@@ -213,9 +206,6 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
 // wait for everyone's pushes, copy assembled scratchB to the output.
 __global__ void k_twoshot_copyout(CommView cv, void* __restrict__ out,
                                   size_t bytes) {
-  signal_all(cv, cv.seq + 1);
-  wait_all(cv, cv.seq + 1);
-
   char const* src = scratch_b(cv.peers[cv.rank], cv);
   size_t const nvec = bytes / 16;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -322,8 +312,6 @@ __global__ void k_ll_allreduce(CommView cv, void const* __restrict__ in,
 // out[r*chunk .. ] = rank r's scratchA chunk, pulled from each peer.
 __global__ void k_allgather_pull(CommView cv, void* __restrict__ out,
                                  size_t chunk_bytes) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
   size_t const nvec = chunk_bytes / 16;
   size_t const tail = chunk_bytes & 15;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -344,8 +332,6 @@ __global__ void k_allgather_pull(CommView cv, void* __restrict__ out,
 template <typename T>
 __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
                                      size_t count) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
   size_t const vper = 16 / sizeof(T);
   size_t const nvec = count / vper;
   size_t const elem_off = static_cast<size_t>(cv.rank) * count;
@@ -378,10 +364,6 @@ __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
 // have staged; only root's flag is awaited.
 __global__ void k_broadcast_pull(CommView cv, int root, void* __restrict__ out,
                                  size_t bytes) {
-  signal_all(cv, cv.seq);
-  if (threadIdx.x == 0)
-    wait_flag_ge(flag_ptr(cv.peers[cv.rank], root, cv.channel), cv.seq);
-  __syncthreads();
   auto const* s = reinterpret_cast<V16 const*>(scratch_a(cv.peers[root], cv));
   auto* d = reinterpret_cast<V16*>(out);
   size_t const nvec = bytes / 16;
@@ -398,8 +380,6 @@ __global__ void k_broadcast_pull(CommView cv, int root, void* __restrict__ out,
 // chunk offset r.
 __global__ void k_alltoall_pull(CommView cv, void* __restrict__ out,
                                 size_t chunk_bytes) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
   size_t const nvec = chunk_bytes / 16;
   size_t const tail = chunk_bytes & 15;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -420,11 +400,6 @@ __global__ void k_alltoall_pull(CommView cv, void* __restrict__ out,
 // ---------------------------------------------------------------------------
 // Barrier and point-to-point signal/wait (host-sequenced send/recv staging)
 // ---------------------------------------------------------------------------
-
-__global__ void k_barrier(CommView cv) {
-  signal_all(cv, cv.seq);
-  wait_all(cv, cv.seq);
-}
 
 // Signal `val` into rank dst's flags[rank][ch]
 __global__ void k_signal_peer(CommView cv, int dst, int ch, uint64_t val) {
@@ -548,8 +523,15 @@ void launch_alltoall_pull(const CommView& cv, void* out, size_t chunk_bytes,
       cv, out, chunk_bytes);
 }
 
+void launch_signal_wait(const CommView& cv, uint64_t val, unsigned wait_mask,
+                        hipStream_t s) {
+  k_signal_wait<<<1, 64, 0, s>>>(cv, val, wait_mask);
+}
+
 void launch_barrier(const CommView& cv, hipStream_t s) {
-  k_barrier<<<1, 64, 0, s>>>(cv);
+  unsigned const mask = (cv.world >= 32) ? 0xffffffffu
+                                         : ((1u << cv.world) - 1u);
+  k_signal_wait<<<1, 64, 0, s>>>(cv, cv.seq, mask);
 }
 
 void launch_signal_peer(const CommView& cv, int dst, int ch, uint64_t val,
