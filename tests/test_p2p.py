@@ -1,0 +1,59 @@
+"""P2P engine tests. The TCP data plane runs everywhere (CPU tier); the
+HIP-IPC one-copy path needs a GPU (gpu tier, 2 processes on one device)."""
+
+import os
+import subprocess
+import sys
+import tempfile
+import uuid
+from pathlib import Path
+
+import pytest
+import torch
+
+REPO = Path(__file__).resolve().parent.parent
+WORKER = REPO / "tests" / "workers" / "p2p_worker.py"
+
+
+def _run_pair(extra_env):
+    meta = os.path.join(tempfile.gettempdir(),
+                        f"uccl_p2p_{uuid.uuid4().hex}.meta")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.update(extra_env)
+    ps = [subprocess.Popen([sys.executable, str(WORKER), role, meta],
+                           env=env, stdout=subprocess.PIPE,
+                           stderr=subprocess.STDOUT)
+          for role in ("server", "client")]
+    outs, ok = [], True
+    for p in ps:
+        try:
+            out, _ = p.communicate(timeout=200)
+        except subprocess.TimeoutExpired:
+            p.kill()
+            out, _ = p.communicate()
+            ok = False
+        outs.append(out.decode())
+        ok = ok and p.returncode == 0
+    joined = "\n=====\n".join(outs)
+    assert ok, joined
+    assert "SERVER OK" in joined and "CLIENT OK" in joined, joined
+
+
+def test_p2p_tcp_cpu():
+    _run_pair({"UCCL_P2P_TEST_GPU": "0"})
+
+
+@pytest.mark.gpu
+def test_p2p_gpu_ipc():
+    if not torch.cuda.is_available():
+        pytest.skip("requires GPU")
+    _run_pair({"UCCL_P2P_TEST_GPU": "1"})
+
+
+@pytest.mark.gpu
+def test_p2p_gpu_tcp_staging():
+    # force the TCP staging path for GPU tensors (IPC disabled)
+    if not torch.cuda.is_available():
+        pytest.skip("requires GPU")
+    _run_pair({"UCCL_P2P_TEST_GPU": "1", "UCCL_P2P_ENABLE_IPC": "0"})

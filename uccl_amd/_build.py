@@ -23,6 +23,7 @@ GPU_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 SOURCES = [
     CSRC / "collective" / "kernels.hip",
     CSRC / "collective" / "communicator.cpp",
+    CSRC / "p2p" / "endpoint.cpp",
     CSRC / "ep" / "ep_kernels.hip",
     CSRC / "ep" / "ep_buffer.cpp",
     CSRC / "bindings" / "module.cpp",

@@ -11,10 +11,12 @@
 
 #include "../collective/communicator.h"
 #include "../core/log.h"
+#include "../p2p/endpoint.h"
 
 namespace py = pybind11;
 using uccl::Communicator;
 using uccl::Dtype;
+using uccl::p2p::Endpoint;
 
 namespace {
 
@@ -125,4 +127,100 @@ PYBIND11_MODULE(_C, m) {
       .def("barrier", [](Communicator& c) {
         c.barrier(current_stream(c.device()));
       });
+
+  // --- P2P engine --------------------------------------------------------
+  auto dev_of = [](const at::Tensor& t) {
+    return t.is_cuda() ? static_cast<int>(t.get_device()) : -1;
+  };
+
+  py::class_<Endpoint>(m, "Endpoint")
+      .def(py::init<int, int>(), py::arg("gpu") = -1,
+           py::arg("num_workers") = 2)
+      .def("metadata",
+           [](Endpoint& e) { return py::bytes(e.metadata()); })
+      .def("connect", &Endpoint::connect,
+           py::call_guard<py::gil_scoped_release>())
+      .def("accept", &Endpoint::accept,
+           py::call_guard<py::gil_scoped_release>())
+      .def("num_conns", &Endpoint::num_conns)
+      .def("reg",
+           [](Endpoint& e, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous());
+             return e.reg(t.data_ptr(), t.numel() * t.element_size(),
+                          t.is_cuda() ? t.get_device() : -1);
+           })
+      .def("dereg", &Endpoint::dereg)
+      .def("advertise",
+           [](Endpoint& e, uint64_t mr, uint64_t off, uint64_t bytes) {
+             return py::bytes(e.advertise(mr, off, bytes));
+           },
+           py::arg("mr_id"), py::arg("offset") = 0, py::arg("bytes") = 0)
+      .def("send",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous());
+             int dev = dev_of(t);
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.send(cid, p, n, dev);
+           })
+      .def("recv",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous());
+             int dev = dev_of(t);
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.recv(cid, p, n, dev);
+           })
+      .def("write",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t,
+                    const std::string& ad) {
+             TORCH_CHECK(t.is_contiguous());
+             int dev = dev_of(t);
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.write(cid, p, n, dev, ad);
+           })
+      .def("read",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t,
+                    const std::string& ad) {
+             TORCH_CHECK(t.is_contiguous());
+             int dev = dev_of(t);
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.read(cid, p, n, dev, ad);
+           })
+      .def("send_async",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous());
+             return e.send_async(cid, t.data_ptr(),
+                                 t.numel() * t.element_size(), dev_of(t));
+           })
+      .def("recv_async",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous());
+             return e.recv_async(cid, t.data_ptr(),
+                                 t.numel() * t.element_size(), dev_of(t));
+           })
+      .def("write_async",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t,
+                    const std::string& ad) {
+             TORCH_CHECK(t.is_contiguous());
+             return e.write_async(cid, t.data_ptr(),
+                                  t.numel() * t.element_size(), dev_of(t),
+                                  ad);
+           })
+      .def("read_async",
+           [dev_of](Endpoint& e, uint64_t cid, at::Tensor t,
+                    const std::string& ad) {
+             TORCH_CHECK(t.is_contiguous());
+             return e.read_async(cid, t.data_ptr(),
+                                 t.numel() * t.element_size(), dev_of(t),
+                                 ad);
+           })
+      .def("poll_async", &Endpoint::poll_async,
+           py::call_guard<py::gil_scoped_release>());
 }

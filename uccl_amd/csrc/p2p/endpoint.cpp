@@ -1,0 +1,626 @@
+#include "endpoint.h"
+
+#include <hip/hip_runtime.h>
+#include <unistd.h>
+
+#include <cstring>
+
+#include "../core/env.h"
+#include "../core/log.h"
+#include "../core/net.h"
+
+namespace uccl {
+namespace p2p {
+
+namespace {
+
+constexpr uint64_t kAdvertMagic = 0x75636361647665ULL;  // "uccladve"
+constexpr size_t kStagingBytes = 8ull << 20;
+
+enum Op : uint64_t {
+  kHello = 1,
+  kSendData = 2,     // a=bytes                               + payload
+  kSendIpc = 3,      // a=bytes, d=token                      + {handle,off}
+  kWriteData = 4,    // a=mr, b=off, c=bytes, d=token         + payload
+  kWriteIpc = 5,     // a=mr, b=off, c=bytes, d=token         + {handle,off}
+  kWriteAck = 6,     // d=token
+  kReadReq = 7,      // a=mr, b=off, c=bytes, d=token
+  kReadResp = 8,     // a=bytes, d=token                      + payload
+  kIpcDone = 9,      // d=token
+};
+
+struct MsgHdr {
+  uint64_t op;
+  uint64_t a, b, c, d;
+};
+
+struct IpcBlob {
+  hipIpcMemHandle_t handle;
+  uint64_t offset;
+  int device;
+  int src_pid;
+};
+
+struct Meta {
+  char ip[48];
+  uint16_t port;
+  int gpu;
+  int pid;
+  char host[64];
+};
+
+bool is_gpu(int device) { return device >= 0; }
+
+}  // namespace
+
+struct Endpoint::RxItem {
+  // Either inline host data, or an IPC descriptor to copy from.
+  std::vector<char> data;
+  bool ipc = false;
+  IpcBlob blob{};
+  uint64_t token = 0;
+  size_t bytes = 0;
+};
+
+struct Endpoint::Conn {
+  uint64_t id = 0;
+  int fd = -1;
+  std::string peer_ip;
+  int peer_gpu = -1;
+  int peer_pid = -1;
+  bool same_host = false;
+  std::thread rx;
+  std::mutex tx_mu;
+  std::atomic<bool> alive{true};
+
+  // in-order two-sided queue
+  std::deque<std::shared_ptr<RxItem>> rxq;
+  std::mutex rx_mu;
+  std::condition_variable rx_cv;
+
+  // token -> completion latch (acks, read responses)
+  std::mutex tok_mu;
+  std::condition_variable tok_cv;
+  std::unordered_map<uint64_t, std::shared_ptr<RxItem>> completed;
+  std::atomic<uint64_t> next_token{1};
+
+  // IPC handle cache: src (pid, base-handle bytes) -> mapped ptr
+  std::unordered_map<std::string, void*> ipc_cache;
+  std::mutex ipc_mu;
+
+  void send_msg(MsgHdr const& h, void const* payload = nullptr,
+                size_t payload_bytes = 0) {
+    std::lock_guard<std::mutex> g(tx_mu);
+    net::send_all(fd, &h, sizeof(h));
+    if (payload_bytes) net::send_all(fd, payload, payload_bytes);
+  }
+};
+
+Endpoint::Endpoint(int gpu, int num_workers) : gpu_(gpu) {
+  char host[256] = {0};
+  gethostname(host, sizeof(host) - 1);
+  host_id_ = std::string(host) + ":" + net::local_ip();
+  listen_fd_ = net::listen_on(&port_);
+  listener_ = std::thread([this] {
+    while (!stop_) {
+      int fd = ::accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) {
+        if (stop_) break;
+        continue;
+      }
+      int one = 1;
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      auto c = std::make_shared<Conn>();
+      c->fd = fd;
+      c->id = next_conn_++;
+      // passive side: expect hello, reply hello
+      MsgHdr h{};
+      if (!net::recv_all(fd, &h, sizeof(h)) || h.op != kHello) {
+        ::close(fd);
+        continue;
+      }
+      std::vector<char> hostbuf(h.b);
+      net::recv_all(fd, hostbuf.data(), hostbuf.size());
+      c->peer_gpu = static_cast<int>(h.a);
+      c->peer_pid = static_cast<int>(h.c);
+      std::string peer_host(hostbuf.begin(), hostbuf.end());
+      c->same_host = (peer_host == host_id_);
+      MsgHdr r{kHello, static_cast<uint64_t>(gpu_), host_id_.size(),
+               static_cast<uint64_t>(getpid()), 0};
+      net::send_all(fd, &r, sizeof(r));
+      net::send_all(fd, host_id_.data(), host_id_.size());
+      {
+        std::lock_guard<std::mutex> g(conn_mu_);
+        conns_[c->id] = c;
+        accepted_.push_back(c->id);
+      }
+      c->rx = std::thread([this, c] { rx_loop(c); });
+      accept_cv_.notify_all();
+    }
+  });
+  for (int i = 0; i < std::max(1, num_workers); ++i)
+    workers_.emplace_back([this] { worker_loop(); });
+}
+
+Endpoint::~Endpoint() {
+  stop_ = true;
+  ::shutdown(listen_fd_, SHUT_RDWR);
+  ::close(listen_fd_);
+  task_cv_.notify_all();
+  for (auto& w : workers_)
+    if (w.joinable()) w.join();
+  {
+    std::lock_guard<std::mutex> g(conn_mu_);
+    for (auto& [id, c] : conns_) {
+      c->alive = false;
+      ::shutdown(c->fd, SHUT_RDWR);
+    }
+  }
+  if (listener_.joinable()) listener_.join();
+  {
+    std::lock_guard<std::mutex> g(conn_mu_);
+    for (auto& [id, c] : conns_) {
+      if (c->rx.joinable()) c->rx.join();
+      ::close(c->fd);
+      for (auto& [k, p] : c->ipc_cache) (void)hipIpcCloseMemHandle(p);
+    }
+    conns_.clear();
+  }
+  if (staging_) (void)hipHostFree(staging_);
+}
+
+std::string Endpoint::metadata() const {
+  Meta m{};
+  std::string ip = net::local_ip();
+  strncpy(m.ip, ip.c_str(), sizeof(m.ip) - 1);
+  m.port = port_;
+  m.gpu = gpu_;
+  m.pid = static_cast<int>(getpid());
+  strncpy(m.host, host_id_.c_str(), sizeof(m.host) - 1);
+  return std::string(reinterpret_cast<char*>(&m), sizeof(m));
+}
+
+uint64_t Endpoint::connect(const std::string& remote_metadata) {
+  UCCL_CHECK(remote_metadata.size() == sizeof(Meta)) << "bad metadata blob";
+  Meta m{};
+  memcpy(&m, remote_metadata.data(), sizeof(m));
+  std::string ip = m.ip;
+  if (std::string(m.host) == host_id_) ip = "127.0.0.1";
+  int fd = net::connect_to(ip, m.port);
+  auto c = std::make_shared<Conn>();
+  c->fd = fd;
+  c->id = next_conn_++;
+  c->peer_ip = ip;
+  MsgHdr h{kHello, static_cast<uint64_t>(gpu_), host_id_.size(),
+           static_cast<uint64_t>(getpid()), 0};
+  net::send_all(fd, &h, sizeof(h));
+  net::send_all(fd, host_id_.data(), host_id_.size());
+  MsgHdr r{};
+  UCCL_CHECK(net::recv_all(fd, &r, sizeof(r)) && r.op == kHello)
+      << "hello handshake failed";
+  std::vector<char> hostbuf(r.b);
+  net::recv_all(fd, hostbuf.data(), hostbuf.size());
+  c->peer_gpu = static_cast<int>(r.a);
+  c->peer_pid = static_cast<int>(r.c);
+  c->same_host = (std::string(hostbuf.begin(), hostbuf.end()) == host_id_);
+  {
+    std::lock_guard<std::mutex> g(conn_mu_);
+    conns_[c->id] = c;
+  }
+  c->rx = std::thread([this, c] { rx_loop(c); });
+  return c->id;
+}
+
+uint64_t Endpoint::accept() {
+  std::unique_lock<std::mutex> lk(conn_mu_);
+  accept_cv_.wait(lk, [this] { return !accepted_.empty() || stop_; });
+  UCCL_CHECK(!accepted_.empty()) << "endpoint shut down";
+  uint64_t id = accepted_.front();
+  accepted_.pop_front();
+  return id;
+}
+
+std::shared_ptr<Endpoint::Conn> Endpoint::conn(uint64_t id) {
+  std::lock_guard<std::mutex> g(conn_mu_);
+  auto it = conns_.find(id);
+  UCCL_CHECK(it != conns_.end()) << "unknown conn " << id;
+  return it->second;
+}
+
+uint64_t Endpoint::reg(void* ptr, size_t bytes, int device) {
+  MR mr{next_mr_++, ptr, bytes, device};
+  std::lock_guard<std::mutex> g(mr_mu_);
+  mrs_[mr.id] = mr;
+  return mr.id;
+}
+
+void Endpoint::dereg(uint64_t mr_id) {
+  std::lock_guard<std::mutex> g(mr_mu_);
+  mrs_.erase(mr_id);
+}
+
+// ---------------------------------------------------------------------------
+// rx loop: demultiplex inbound messages
+// ---------------------------------------------------------------------------
+
+void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
+  if (gpu_ >= 0) (void)hipSetDevice(gpu_);
+  while (c->alive && !stop_) {
+    MsgHdr h{};
+    if (!net::recv_all(c->fd, &h, sizeof(h))) break;
+    switch (h.op) {
+      case kSendData: {
+        auto item = std::make_shared<RxItem>();
+        item->bytes = h.a;
+        item->data.resize(h.a);
+        net::recv_all(c->fd, item->data.data(), h.a);
+        {
+          std::lock_guard<std::mutex> g(c->rx_mu);
+          c->rxq.push_back(item);
+        }
+        c->rx_cv.notify_all();
+        break;
+      }
+      case kSendIpc: {
+        auto item = std::make_shared<RxItem>();
+        item->bytes = h.a;
+        item->ipc = true;
+        item->token = h.d;
+        net::recv_all(c->fd, &item->blob, sizeof(IpcBlob));
+        {
+          std::lock_guard<std::mutex> g(c->rx_mu);
+          c->rxq.push_back(item);
+        }
+        c->rx_cv.notify_all();
+        break;
+      }
+      case kWriteData: {
+        MR mr;
+        {
+          std::lock_guard<std::mutex> g(mr_mu_);
+          auto it = mrs_.find(h.a);
+          UCCL_CHECK(it != mrs_.end()) << "write to unknown mr " << h.a;
+          mr = it->second;
+        }
+        UCCL_CHECK(h.b + h.c <= mr.bytes) << "write overflows mr";
+        char* dst = static_cast<char*>(mr.ptr) + h.b;
+        if (!is_gpu(mr.device)) {
+          net::recv_all(c->fd, dst, h.c);
+        } else {
+          std::vector<char> tmp(h.c);
+          net::recv_all(c->fd, tmp.data(), h.c);
+          UCCL_CHECK_HIP(hipMemcpy(dst, tmp.data(), h.c,
+                                   hipMemcpyHostToDevice));
+        }
+        if (h.d) c->send_msg(MsgHdr{kWriteAck, 0, 0, 0, h.d});
+        break;
+      }
+      case kWriteIpc: {
+        IpcBlob blob{};
+        net::recv_all(c->fd, &blob, sizeof(blob));
+        MR mr;
+        {
+          std::lock_guard<std::mutex> g(mr_mu_);
+          auto it = mrs_.find(h.a);
+          UCCL_CHECK(it != mrs_.end()) << "ipc write to unknown mr " << h.a;
+          mr = it->second;
+        }
+        UCCL_CHECK(h.b + h.c <= mr.bytes) << "ipc write overflows mr";
+        void* src_base = open_ipc(*c, &blob, blob.device);
+        UCCL_CHECK_HIP(hipMemcpy(static_cast<char*>(mr.ptr) + h.b,
+                                 static_cast<char*>(src_base) + blob.offset,
+                                 h.c, hipMemcpyDeviceToDevice));
+        if (h.d) c->send_msg(MsgHdr{kIpcDone, 0, 0, 0, h.d});
+        break;
+      }
+      case kReadReq: {
+        MR mr;
+        {
+          std::lock_guard<std::mutex> g(mr_mu_);
+          auto it = mrs_.find(h.a);
+          UCCL_CHECK(it != mrs_.end()) << "read of unknown mr " << h.a;
+          mr = it->second;
+        }
+        UCCL_CHECK(h.b + h.c <= mr.bytes) << "read overflows mr";
+        char const* src = static_cast<char const*>(mr.ptr) + h.b;
+        MsgHdr resp{kReadResp, h.c, 0, 0, h.d};
+        if (!is_gpu(mr.device)) {
+          c->send_msg(resp, src, h.c);
+        } else {
+          std::vector<char> tmp(h.c);
+          UCCL_CHECK_HIP(hipMemcpy(tmp.data(), src, h.c,
+                                   hipMemcpyDeviceToHost));
+          c->send_msg(resp, tmp.data(), h.c);
+        }
+        break;
+      }
+      case kReadResp: {
+        auto item = std::make_shared<RxItem>();
+        item->bytes = h.a;
+        item->data.resize(h.a);
+        net::recv_all(c->fd, item->data.data(), h.a);
+        {
+          std::lock_guard<std::mutex> g(c->tok_mu);
+          c->completed[h.d] = item;
+        }
+        c->tok_cv.notify_all();
+        break;
+      }
+      case kWriteAck:
+      case kIpcDone: {
+        {
+          std::lock_guard<std::mutex> g(c->tok_mu);
+          c->completed[h.d] = std::make_shared<RxItem>();
+        }
+        c->tok_cv.notify_all();
+        break;
+      }
+      default:
+        UCCL_LOG_ERROR << "unknown p2p op " << h.op;
+        c->alive = false;
+        return;
+    }
+  }
+}
+
+void* Endpoint::open_ipc(Conn& c, const void* blob_bytes, int src_device) {
+  auto const* blob = static_cast<IpcBlob const*>(blob_bytes);
+  std::string key(reinterpret_cast<char const*>(&blob->handle),
+                  sizeof(blob->handle));
+  key += std::to_string(blob->src_pid);
+  std::lock_guard<std::mutex> g(c.ipc_mu);
+  auto it = c.ipc_cache.find(key);
+  if (it != c.ipc_cache.end()) return it->second;
+  void* p = nullptr;
+  UCCL_CHECK_HIP(hipIpcOpenMemHandle(&p, blob->handle,
+                                     hipIpcMemLazyEnablePeerAccess));
+  c.ipc_cache[key] = p;
+  return p;
+}
+
+// ---------------------------------------------------------------------------
+// data-plane ops
+// ---------------------------------------------------------------------------
+
+static bool ipc_enabled() {
+  static bool v = env_bool("UCCL_P2P_ENABLE_IPC", true);
+  return v;
+}
+
+void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
+  if (is_gpu(device) && c.same_host && ipc_enabled()) {
+    // one-copy IPC path: ship {handle, offset}; receiver DtoD-copies
+    IpcBlob blob{};
+    void* base = nullptr;
+    size_t base_sz = 0;
+    UCCL_CHECK_HIP(hipMemGetAddressRange(
+        reinterpret_cast<hipDeviceptr_t*>(&base), &base_sz,
+        reinterpret_cast<hipDeviceptr_t>(const_cast<void*>(ptr))));
+    UCCL_CHECK_HIP(hipIpcGetMemHandle(&blob.handle, base));
+    blob.offset = static_cast<char const*>(ptr) - static_cast<char*>(base);
+    blob.device = device;
+    blob.src_pid = static_cast<int>(getpid());
+    uint64_t token = c.next_token++;
+    c.send_msg(MsgHdr{kSendIpc, bytes, 0, 0, token}, &blob, sizeof(blob));
+    // wait for receiver's copy (src must stay valid until then)
+    std::unique_lock<std::mutex> lk(c.tok_mu);
+    c.tok_cv.wait(lk, [&] { return c.completed.count(token) || !c.alive; });
+    c.completed.erase(token);
+    return;
+  }
+  if (!is_gpu(device)) {
+    c.send_msg(MsgHdr{kSendData, bytes, 0, 0, 0}, ptr, bytes);
+    return;
+  }
+  // GPU over TCP: pinned staging chunks
+  std::lock_guard<std::mutex> sg(staging_mu_);
+  if (!staging_) {
+    UCCL_CHECK_HIP(hipSetDevice(gpu_ >= 0 ? gpu_ : device));
+    UCCL_CHECK_HIP(hipHostMalloc(&staging_, kStagingBytes));
+    staging_bytes_ = kStagingBytes;
+  }
+  std::lock_guard<std::mutex> g(c.tx_mu);
+  MsgHdr h{kSendData, bytes, 0, 0, 0};
+  net::send_all(c.fd, &h, sizeof(h));
+  for (size_t off = 0; off < bytes; off += staging_bytes_) {
+    size_t n = std::min(staging_bytes_, bytes - off);
+    UCCL_CHECK_HIP(hipMemcpy(staging_,
+                             static_cast<char const*>(ptr) + off, n,
+                             hipMemcpyDeviceToHost));
+    net::send_all(c.fd, staging_, n);
+  }
+}
+
+void Endpoint::copy_to_user(RxItem& item, void* dst, size_t bytes,
+                            int device) {
+  UCCL_CHECK(item.bytes <= bytes)
+      << "recv buffer too small: " << bytes << " < " << item.bytes;
+  if (!is_gpu(device)) {
+    memcpy(dst, item.data.data(), item.bytes);
+  } else {
+    UCCL_CHECK_HIP(hipMemcpy(dst, item.data.data(), item.bytes,
+                             hipMemcpyHostToDevice));
+  }
+}
+
+void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
+  std::shared_ptr<RxItem> item;
+  {
+    std::unique_lock<std::mutex> lk(c.rx_mu);
+    c.rx_cv.wait(lk, [&] { return !c.rxq.empty() || !c.alive || stop_; });
+    UCCL_CHECK(!c.rxq.empty()) << "connection closed during recv";
+    item = c.rxq.front();
+    c.rxq.pop_front();
+  }
+  if (item->ipc) {
+    UCCL_CHECK(is_gpu(device)) << "IPC send into host recv buffer";
+    UCCL_CHECK(item->bytes <= bytes) << "recv buffer too small";
+    void* src_base = open_ipc(c, &item->blob, item->blob.device);
+    UCCL_CHECK_HIP(
+        hipMemcpy(ptr, static_cast<char*>(src_base) + item->blob.offset,
+                  item->bytes, hipMemcpyDeviceToDevice));
+    c.send_msg(MsgHdr{kIpcDone, 0, 0, 0, item->token});
+  } else {
+    copy_to_user(*item, ptr, bytes, device);
+  }
+}
+
+void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
+                        Advert ad) {
+  UCCL_CHECK(bytes <= ad.bytes) << "write larger than advertised window";
+  uint64_t token = c.next_token++;
+  if (is_gpu(device) && c.same_host && ipc_enabled()) {
+    IpcBlob blob{};
+    void* base = nullptr;
+    size_t base_sz = 0;
+    UCCL_CHECK_HIP(hipMemGetAddressRange(
+        reinterpret_cast<hipDeviceptr_t*>(&base), &base_sz,
+        reinterpret_cast<hipDeviceptr_t>(const_cast<void*>(ptr))));
+    UCCL_CHECK_HIP(hipIpcGetMemHandle(&blob.handle, base));
+    blob.offset = static_cast<char const*>(ptr) - static_cast<char*>(base);
+    blob.device = device;
+    blob.src_pid = static_cast<int>(getpid());
+    c.send_msg(MsgHdr{kWriteIpc, ad.mr_id, ad.offset, bytes, token}, &blob,
+               sizeof(blob));
+  } else if (!is_gpu(device)) {
+    c.send_msg(MsgHdr{kWriteData, ad.mr_id, ad.offset, bytes, token}, ptr,
+               bytes);
+  } else {
+    std::vector<char> tmp(bytes);
+    UCCL_CHECK_HIP(hipMemcpy(tmp.data(), ptr, bytes, hipMemcpyDeviceToHost));
+    c.send_msg(MsgHdr{kWriteData, ad.mr_id, ad.offset, bytes, token},
+               tmp.data(), bytes);
+  }
+  std::unique_lock<std::mutex> lk(c.tok_mu);
+  c.tok_cv.wait(lk, [&] { return c.completed.count(token) || !c.alive; });
+  c.completed.erase(token);
+}
+
+void Endpoint::do_read(Conn& c, void* ptr, size_t bytes, int device,
+                       Advert ad) {
+  UCCL_CHECK(bytes <= ad.bytes) << "read larger than advertised window";
+  uint64_t token = c.next_token++;
+  c.send_msg(MsgHdr{kReadReq, ad.mr_id, ad.offset, bytes, token});
+  std::shared_ptr<RxItem> item;
+  {
+    std::unique_lock<std::mutex> lk(c.tok_mu);
+    c.tok_cv.wait(lk, [&] { return c.completed.count(token) || !c.alive; });
+    item = c.completed[token];
+    c.completed.erase(token);
+  }
+  UCCL_CHECK(item) << "read failed (connection closed)";
+  copy_to_user(*item, ptr, bytes, device);
+}
+
+// ---------------------------------------------------------------------------
+// public (sync wrappers + async submission)
+// ---------------------------------------------------------------------------
+
+static Advert parse_advert(const std::string& s) {
+  Advert ad{};
+  UCCL_CHECK(s.size() == sizeof(Advert)) << "bad advert blob";
+  memcpy(&ad, s.data(), sizeof(ad));
+  UCCL_CHECK(ad.magic == kAdvertMagic) << "bad advert magic";
+  return ad;
+}
+
+std::string Endpoint::advertise(uint64_t mr_id, uint64_t offset,
+                                uint64_t bytes) {
+  {
+    std::lock_guard<std::mutex> g(mr_mu_);
+    auto it = mrs_.find(mr_id);
+    UCCL_CHECK(it != mrs_.end()) << "advertise of unknown mr";
+    UCCL_CHECK(offset + bytes <= it->second.bytes) << "advert out of range";
+  }
+  Advert ad{kAdvertMagic, mr_id, offset, bytes};
+  return std::string(reinterpret_cast<char*>(&ad), sizeof(ad));
+}
+
+void Endpoint::send(uint64_t cid, void const* p, size_t n, int dev) {
+  do_send(*conn(cid), p, n, dev);
+}
+void Endpoint::recv(uint64_t cid, void* p, size_t n, int dev) {
+  do_recv(*conn(cid), p, n, dev);
+}
+void Endpoint::write(uint64_t cid, void const* p, size_t n, int dev,
+                     const std::string& ad) {
+  do_write(*conn(cid), p, n, dev, parse_advert(ad));
+}
+void Endpoint::read(uint64_t cid, void* p, size_t n, int dev,
+                    const std::string& ad) {
+  do_read(*conn(cid), p, n, dev, parse_advert(ad));
+}
+
+uint64_t Endpoint::submit(std::function<void()> fn) {
+  auto status = std::make_shared<std::atomic<int>>(0);
+  uint64_t id = next_xfer_++;
+  {
+    std::lock_guard<std::mutex> g(xfer_mu_);
+    xfers_[id] = status;
+  }
+  {
+    std::lock_guard<std::mutex> g(task_mu_);
+    tasks_.push_back([fn = std::move(fn), status] {
+      fn();
+      status->store(1);
+    });
+  }
+  task_cv_.notify_one();
+  return id;
+}
+
+void Endpoint::worker_loop() {
+  if (gpu_ >= 0) (void)hipSetDevice(gpu_);
+  while (true) {
+    std::function<void()> task;
+    {
+      std::unique_lock<std::mutex> lk(task_mu_);
+      task_cv_.wait(lk, [this] { return !tasks_.empty() || stop_; });
+      if (stop_ && tasks_.empty()) return;
+      task = std::move(tasks_.front());
+      tasks_.pop_front();
+    }
+    task();
+  }
+}
+
+uint64_t Endpoint::send_async(uint64_t cid, void const* p, size_t n,
+                              int dev) {
+  auto c = conn(cid);
+  return submit([this, c, p, n, dev] { do_send(*c, p, n, dev); });
+}
+uint64_t Endpoint::recv_async(uint64_t cid, void* p, size_t n, int dev) {
+  auto c = conn(cid);
+  return submit([this, c, p, n, dev] { do_recv(*c, p, n, dev); });
+}
+uint64_t Endpoint::write_async(uint64_t cid, void const* p, size_t n, int dev,
+                               const std::string& ad) {
+  auto c = conn(cid);
+  Advert a = parse_advert(ad);
+  return submit([this, c, p, n, dev, a] { do_write(*c, p, n, dev, a); });
+}
+uint64_t Endpoint::read_async(uint64_t cid, void* p, size_t n, int dev,
+                              const std::string& ad) {
+  auto c = conn(cid);
+  Advert a = parse_advert(ad);
+  return submit([this, c, p, n, dev, a] { do_read(*c, p, n, dev, a); });
+}
+
+bool Endpoint::poll_async(uint64_t xfer_id) {
+  std::lock_guard<std::mutex> g(xfer_mu_);
+  auto it = xfers_.find(xfer_id);
+  UCCL_CHECK(it != xfers_.end()) << "unknown transfer " << xfer_id;
+  if (it->second->load() == 1) {
+    xfers_.erase(it);
+    return true;
+  }
+  return false;
+}
+
+int Endpoint::num_conns() {
+  std::lock_guard<std::mutex> g(conn_mu_);
+  return static_cast<int>(conns_.size());
+}
+
+}  // namespace p2p
+}  // namespace uccl
