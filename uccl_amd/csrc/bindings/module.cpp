@@ -11,6 +11,7 @@
 
 #include "../collective/communicator.h"
 #include "../core/log.h"
+#include "../ep/ep_buffer.h"
 #include "../p2p/endpoint.h"
 
 namespace py = pybind11;
@@ -223,4 +224,67 @@ PYBIND11_MODULE(_C, m) {
            })
       .def("poll_async", &Endpoint::poll_async,
            py::call_guard<py::gil_scoped_release>());
+
+  // --- EP (DeepEP-compatible expert parallel) -----------------------------
+  py::class_<uccl::ep::EpBuffer>(m, "EpBuffer")
+      .def(py::init([](int rank, int world, int device, int num_experts,
+                       int topk, int hidden, int max_tokens,
+                       int elem_size) {
+             return new uccl::ep::EpBuffer(rank, world, device, num_experts,
+                                           topk, hidden, max_tokens,
+                                           elem_size);
+           }),
+           py::arg("rank"), py::arg("world"), py::arg("device"),
+           py::arg("num_experts"), py::arg("topk"), py::arg("hidden"),
+           py::arg("max_tokens"), py::arg("elem_size"))
+      .def("handle_bytes",
+           [](uccl::ep::EpBuffer& b) { return py::bytes(b.handle_bytes()); })
+      .def("connect",
+           [](uccl::ep::EpBuffer& b, const std::vector<std::string>& h) {
+             b.connect(h);
+           })
+      .def("dispatch",
+           [](uccl::ep::EpBuffer& b, at::Tensor x, at::Tensor topk_idx) {
+             TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+             TORCH_CHECK(topk_idx.scalar_type() == at::kLong &&
+                         topk_idx.is_cuda() && topk_idx.is_contiguous());
+             auto const& v = b.view();
+             TORCH_CHECK(x.dim() == 2 && x.size(1) == v.hidden);
+             TORCH_CHECK(x.element_size() == v.elem_size);
+             TORCH_CHECK(topk_idx.dim() == 2 && topk_idx.size(1) == v.topk &&
+                         topk_idx.size(0) == x.size(0));
+             auto counts = at::empty(
+                 {v.local_experts, v.world},
+                 at::TensorOptions().dtype(at::kInt).device(x.device()));
+             b.dispatch(x.data_ptr(), topk_idx.data_ptr<int64_t>(),
+                        static_cast<int>(x.size(0)), counts.data_ptr<int>(),
+                        current_stream(b.device()));
+             auto recv_x = at::from_blob(
+                 b.recv_x_ptr(),
+                 {v.local_experts,
+                  static_cast<int64_t>(v.world) * v.max_tokens, v.hidden},
+                 at::TensorOptions().dtype(x.scalar_type())
+                     .device(x.device()));
+             return py::make_tuple(recv_x, counts);
+           })
+      .def("combine",
+           [](uccl::ep::EpBuffer& b, at::Tensor expert_out,
+              at::Tensor topk_idx, at::Tensor topk_w) {
+             auto const& v = b.view();
+             TORCH_CHECK(expert_out.is_cuda() && expert_out.is_contiguous());
+             TORCH_CHECK(expert_out.element_size() == v.elem_size);
+             TORCH_CHECK(topk_idx.scalar_type() == at::kLong &&
+                         topk_idx.is_contiguous());
+             TORCH_CHECK(topk_w.scalar_type() == at::kFloat &&
+                         topk_w.is_contiguous());
+             auto out = at::empty(
+                 {topk_idx.size(0), v.hidden},
+                 at::TensorOptions().dtype(expert_out.scalar_type())
+                     .device(expert_out.device()));
+             b.combine(expert_out.data_ptr(), out.data_ptr(),
+                       topk_idx.data_ptr<int64_t>(),
+                       topk_w.data_ptr<float>(),
+                       current_stream(b.device()));
+             return out;
+           });
 }

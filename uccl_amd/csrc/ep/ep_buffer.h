@@ -1,0 +1,55 @@
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <array>
+#include <string>
+#include <vector>
+
+#include "ep_layout.h"
+
+namespace uccl {
+namespace ep {
+
+class EpBuffer {
+ public:
+  EpBuffer(int rank, int world, int device, int num_experts, int topk,
+           int hidden, int max_tokens, int elem_size);
+  ~EpBuffer();
+  EpBuffer(const EpBuffer&) = delete;
+
+  std::string handle_bytes() const;
+  void connect(const std::vector<std::string>& handles);
+
+  // x: [num_tokens, hidden] (elem_size); topk_idx: [num_tokens, topk] i64.
+  // out_counts: device int32 [local_experts, world].
+  void dispatch(void const* x, int64_t const* topk_idx, int num_tokens,
+                int* out_counts, hipStream_t stream);
+
+  // expert_out: [local_experts, world*max_tokens, hidden];
+  // out: [num_tokens, hidden]; topk_w: [num_tokens, topk] f32.
+  void combine(void const* expert_out, void* out, int64_t const* topk_idx,
+               float const* topk_w, hipStream_t stream);
+
+  const EpView& view() const { return v_; }
+  void* recv_x_ptr() const {
+    return static_cast<char*>(heap_) + v_.off_disp_x;
+  }
+  void* recv_meta_ptr() const {
+    return static_cast<char*>(heap_) + v_.off_disp_meta;
+  }
+  int rank() const { return rank_; }
+  int world() const { return world_; }
+  int device() const { return device_; }
+
+ private:
+  int rank_, world_, device_;
+  void* heap_ = nullptr;
+  EpView v_{};
+  std::array<bool, kMaxRanks> ipc_opened_{};
+  bool connected_ = false;
+  int last_num_tokens_ = -1;
+};
+
+}  // namespace ep
+}  // namespace uccl

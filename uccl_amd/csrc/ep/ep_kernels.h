@@ -1,0 +1,18 @@
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include "ep_layout.h"
+
+namespace uccl {
+namespace ep {
+
+void launch_ep_dispatch(const EpView& v, void const* x,
+                        int64_t const* topk_idx, int num_tokens,
+                        int* out_counts, hipStream_t s);
+void launch_ep_combine(const EpView& v, void const* expert_out, void* out,
+                       int64_t const* topk_idx, float const* topk_w,
+                       int num_tokens, hipStream_t s);
+
+}  // namespace ep
+}  // namespace uccl

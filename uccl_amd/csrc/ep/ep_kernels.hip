@@ -1,0 +1,254 @@
+// CDNA4 expert-parallel dispatch/combine kernels (intranode xGMI).
+//
+// Protocol (see ep_layout.h): per destination expert e, one block scans the
+// local top-k table, compacts matching (token, k) pairs in token order
+// (stable two-pass prefix compaction in LDS), copies each token's hidden
+// vector into the destination rank's slot array over xGMI, then publishes a
+// seq-tagged count with a system-scope release store. Combine returns each
+// processed slot to its source (token, k) cell and the source reduces with
+// top-k weights in fp32. Waits are isolated in single-block kernels (same
+// no-spinning-workers rule as the collective engine).
+//
+// Functional parity: reference ep/src/internode_ll.cu dispatch/combine
+// (phases, per-expert slots, src-idx meta) minus the RDMA/proxy path, which
+// a single 8-OAM MI355X node does not need.
+
+#include <hip/hip_runtime.h>
+
+#include "../device/primitives.h"
+#include "ep_kernels.h"
+
+namespace uccl {
+namespace ep {
+
+using namespace uccl::device;
+
+namespace {
+
+__device__ inline uint64_t tag_count(uint64_t seq, uint32_t count) {
+  return (seq << 32) | count;
+}
+
+// Copy `bytes` from src to dst cooperatively with the whole block.
+__device__ inline void block_copy(char* __restrict__ dst,
+                                  char const* __restrict__ src,
+                                  size_t bytes) {
+  size_t const nvec = bytes / 16;
+  auto* d = reinterpret_cast<V16*>(dst);
+  auto const* s = reinterpret_cast<V16 const*>(src);
+  for (size_t i = threadIdx.x; i < nvec; i += blockDim.x) d[i] = s[i];
+  size_t const tail = bytes & 15;
+  if (tail && threadIdx.x < tail)
+    dst[bytes - tail + threadIdx.x] = src[bytes - tail + threadIdx.x];
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// dispatch send: grid = num_experts blocks (one per destination expert)
+// ---------------------------------------------------------------------------
+
+__global__ void k_ep_dispatch_send(EpView v, void const* __restrict__ x,
+                                   int64_t const* __restrict__ topk_idx,
+                                   int num_tokens) {
+  int const e = blockIdx.x;              // global destination expert
+  int const dst = e / v.local_experts;   // destination rank
+  int const le = e % v.local_experts;    // local expert index on dst
+  void* dbase = v.peers[dst];
+
+  // --- stable compaction of matching (token, k) pairs ---------------------
+  // segment the token range contiguously across threads so slot order is
+  // token order (deterministic; tests rely on it)
+  extern __shared__ uint32_t smem[];  // [blockDim.x + 1] prefix, then list
+  uint32_t* prefix = smem;
+  uint32_t* list = smem + blockDim.x + 1;  // max_tokens entries (t | k<<24)
+
+  int const seg = (num_tokens + blockDim.x - 1) / blockDim.x;
+  int const t0 = threadIdx.x * seg;
+  int const t1 = min(t0 + seg, num_tokens);
+  uint32_t mine = 0;
+  for (int t = t0; t < t1; ++t)
+    for (int k = 0; k < v.topk; ++k)
+      if (topk_idx[static_cast<size_t>(t) * v.topk + k] == e) ++mine;
+  prefix[threadIdx.x + 1] = mine;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    prefix[0] = 0;
+    for (unsigned i = 1; i <= blockDim.x; ++i) prefix[i] += prefix[i - 1];
+  }
+  __syncthreads();
+  uint32_t pos = prefix[threadIdx.x];
+  for (int t = t0; t < t1; ++t)
+    for (int k = 0; k < v.topk; ++k)
+      if (topk_idx[static_cast<size_t>(t) * v.topk + k] == e)
+        list[pos++] = static_cast<uint32_t>(t) |
+                      (static_cast<uint32_t>(k) << 24);
+  __syncthreads();
+  uint32_t const count = prefix[blockDim.x];
+
+  // --- copy matched tokens into dst's slots over xGMI ---------------------
+  size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
+  for (uint32_t i = 0; i < count; ++i) {
+    uint32_t const tk = list[i];
+    uint32_t const t = tk & kMetaTokMask;
+    size_t const slot = static_cast<size_t>(v.rank) * v.max_tokens + i;
+    block_copy(disp_x_ptr(dbase, v, le, slot),
+               static_cast<char const*>(x) + static_cast<size_t>(t) *
+                                                  row_bytes,
+               row_bytes);
+    if (threadIdx.x == 0) *disp_meta_ptr(dbase, v, le, slot) = tk;
+  }
+  __syncthreads();
+  fence_release_sys();
+  if (threadIdx.x == 0) {
+    st_release_sys(disp_count_ptr(dbase, v, le, v.rank),
+                   tag_count(v.seq, count));
+  }
+}
+
+// single-block wait: spin until every (local_expert, src) count carries this
+// seq tag, then write the plain counts into out_counts [local_experts][world]
+__global__ void k_ep_dispatch_wait(EpView v,
+                                   int* __restrict__ out_counts) {
+  int const n = v.local_experts * v.world;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+    int const le = i / v.world;
+    int const src = i % v.world;
+    uint64_t const* p = disp_count_ptr(v.peers[v.rank], v, le, src);
+    uint64_t got = 0;
+    for (uint64_t it = 0;; ++it) {
+      got = ld_acquire_sys(p);
+      if ((got >> 32) == v.seq) break;
+      if (it > (1ull << 28)) __builtin_trap();
+      backoff();
+    }
+    out_counts[i] = static_cast<int>(got & 0xffffffffu);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// combine send: grid = local_experts * world blocks; block (le, src) walks
+// its received slots and writes each expert output row back into the source
+// rank's (token, k) cell.
+// ---------------------------------------------------------------------------
+
+__global__ void k_ep_combine_send(EpView v,
+                                  void const* __restrict__ expert_out) {
+  int const le = blockIdx.x / v.world;
+  int const src = blockIdx.x % v.world;
+  void* me = v.peers[v.rank];
+  uint64_t const tagged = *disp_count_ptr(me, v, le, src);
+  // count published at dispatch time with this seq (combine reuses it)
+  uint32_t const count = static_cast<uint32_t>(tagged & 0xffffffffu);
+  void* sbase = v.peers[src];
+  size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
+  for (uint32_t i = 0; i < count; ++i) {
+    size_t const slot = static_cast<size_t>(src) * v.max_tokens + i;
+    uint32_t const meta = *disp_meta_ptr(me, v, le, slot);
+    uint32_t const t = meta & kMetaTokMask;
+    uint32_t const k = meta >> 24;
+    char const* srcrow =
+        static_cast<char const*>(expert_out) +
+        ((static_cast<size_t>(le) * v.world * v.max_tokens + slot) *
+         v.hidden) *
+            v.elem_size;
+    block_copy(comb_x_ptr(sbase, v, t, k), srcrow, row_bytes);
+  }
+}
+
+// one block: publish "my returns to you are complete" to every rank
+// (launched after k_ep_combine_send; dispatch boundary flushed the writes)
+__global__ void k_ep_combine_signal(EpView v) {
+  if (threadIdx.x < static_cast<unsigned>(v.world))
+    st_release_sys(comb_flag_ptr(v.peers[threadIdx.x], v, v.rank), v.seq);
+}
+
+__global__ void k_ep_combine_wait(EpView v) {
+  if (threadIdx.x < static_cast<unsigned>(v.world)) {
+    uint64_t const* p = comb_flag_ptr(v.peers[v.rank], v, threadIdx.x);
+    for (uint64_t it = 0;; ++it) {
+      if (ld_acquire_sys(p) >= v.seq) break;
+      if (it > (1ull << 28)) __builtin_trap();
+      backoff();
+    }
+  }
+}
+
+// reduce: grid = num_tokens blocks; out[t] = sum_k w[t][k] * comb_x[t][k]
+// in fp32, skipping masked (idx<0) entries.
+template <typename T>
+__global__ void k_ep_combine_reduce(EpView v, void* __restrict__ out,
+                                    int64_t const* __restrict__ topk_idx,
+                                    float const* __restrict__ topk_w,
+                                    int num_tokens) {
+  int const t = blockIdx.x;
+  if (t >= num_tokens) return;
+  void* me = v.peers[v.rank];
+  size_t const vper = 16 / sizeof(T);
+  size_t const nvec = static_cast<size_t>(v.hidden) / vper;
+  auto* orow = reinterpret_cast<V16*>(static_cast<char*>(out) +
+                                      static_cast<size_t>(t) * v.hidden *
+                                          sizeof(T));
+  for (size_t i = threadIdx.x; i < nvec; i += blockDim.x) {
+    float acc[16 / sizeof(T)] = {};
+    for (int k = 0; k < v.topk; ++k) {
+      if (topk_idx[static_cast<size_t>(t) * v.topk + k] < 0) continue;
+      float const w = topk_w[static_cast<size_t>(t) * v.topk + k];
+      V16 const val =
+          reinterpret_cast<V16 const*>(comb_x_ptr(me, v, t, k))[i];
+#pragma unroll
+      for (size_t j = 0; j < vper; ++j)
+        acc[j] += w * static_cast<float>(reinterpret_cast<T const*>(
+                          &val)[j]);
+    }
+    V16 r;
+#pragma unroll
+    for (size_t j = 0; j < vper; ++j)
+      reinterpret_cast<T*>(&r)[j] = static_cast<T>(acc[j]);
+    orow[i] = r;
+  }
+  // hidden not divisible by vec width: scalar tail
+  size_t const tail = v.hidden - nvec * vper;
+  if (tail && threadIdx.x < tail) {
+    size_t const j = nvec * vper + threadIdx.x;
+    float acc = 0.f;
+    for (int k = 0; k < v.topk; ++k) {
+      if (topk_idx[static_cast<size_t>(t) * v.topk + k] < 0) continue;
+      acc += topk_w[static_cast<size_t>(t) * v.topk + k] *
+             static_cast<float>(reinterpret_cast<T const*>(
+                 comb_x_ptr(me, v, t, 0))[k * v.hidden + j]);
+    }
+    reinterpret_cast<T*>(orow)[j] = static_cast<T>(acc);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+void launch_ep_dispatch(const EpView& v, void const* x,
+                        int64_t const* topk_idx, int num_tokens,
+                        int* out_counts, hipStream_t s) {
+  size_t const smem = (257 + v.max_tokens) * sizeof(uint32_t);
+  k_ep_dispatch_send<<<v.num_experts, 256, smem, s>>>(v, x, topk_idx,
+                                                      num_tokens);
+  k_ep_dispatch_wait<<<1, 256, 0, s>>>(v, out_counts);
+}
+
+void launch_ep_combine(const EpView& v, void const* expert_out, void* out,
+                       int64_t const* topk_idx, float const* topk_w,
+                       int num_tokens, hipStream_t s) {
+  k_ep_combine_send<<<v.local_experts * v.world, 256, 0, s>>>(v, expert_out);
+  k_ep_combine_signal<<<1, 64, 0, s>>>(v);
+  k_ep_combine_wait<<<1, 64, 0, s>>>(v);
+  if (v.elem_size == 2) {
+    k_ep_combine_reduce<__hip_bfloat16>
+        <<<num_tokens, 256, 0, s>>>(v, out, topk_idx, topk_w, num_tokens);
+  } else {
+    k_ep_combine_reduce<float>
+        <<<num_tokens, 256, 0, s>>>(v, out, topk_idx, topk_w, num_tokens);
+  }
+}
+
+}  // namespace ep
+}  // namespace uccl

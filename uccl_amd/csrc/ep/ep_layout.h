@@ -1,0 +1,85 @@
+// Expert-parallel (DeepEP-compatible) symmetric-heap layout.
+//
+// Parity role: the reference's ep/ Buffer + internode_ll kernels
+// (ep/src/uccl_ep.cc:348, ep/src/internode_ll.cu:62/:747), re-designed for
+// a single MI355X node: no NIC, no proxy threads — dispatch tokens are
+// written straight into the destination rank's HBM over xGMI, per-expert
+// per-source slot arrays carry seq-tagged counts, and combine returns ride
+// the same path back. 288 GB HBM3E per GPU makes the fully-partitioned
+// slot layout ([local_expert][src_rank*max_tokens]) affordable, which
+// removes all cross-source contention.
+#pragma once
+
+#include <cstddef>
+#include <cstdint>
+
+#include "../collective/layout.h"  // kMaxRanks
+
+namespace uccl {
+namespace ep {
+
+// Runtime-sized layout; offsets are computed host-side once.
+struct EpView {
+  int rank;
+  int world;
+  int num_experts;     // global
+  int local_experts;   // num_experts / world
+  int topk;
+  int hidden;          // elements per token
+  int max_tokens;      // per source rank (DeepEP num_max_dispatch_tokens_per_rank)
+  int elem_size;       // bytes per element of x
+  uint64_t seq;
+  void* peers[kMaxRanks];
+
+  // byte offsets into every rank's heap:
+  size_t off_disp_count;  // u64 [local_experts][world]    (seq<<32 | count)
+  size_t off_disp_x;      // [local_experts][world*max_tokens][hidden] elems
+  size_t off_disp_meta;   // u32 [local_experts][world*max_tokens]
+                          //   meta = src_token_idx | (k << 24)
+  size_t off_comb_flag;   // u64 [world]                   (seq)
+  size_t off_comb_x;      // [max_tokens][topk][hidden] elems
+  size_t heap_bytes;
+};
+
+constexpr uint32_t kMetaTokMask = 0x00ffffffu;
+
+__host__ __device__ inline uint64_t* disp_count_ptr(void* base,
+                                                    const EpView& v, int le,
+                                                    int src) {
+  return reinterpret_cast<uint64_t*>(static_cast<char*>(base) +
+                                     v.off_disp_count) +
+         static_cast<size_t>(le) * v.world + src;
+}
+
+__host__ __device__ inline char* disp_x_ptr(void* base, const EpView& v,
+                                            int le, size_t slot) {
+  return static_cast<char*>(base) + v.off_disp_x +
+         ((static_cast<size_t>(le) * v.world * v.max_tokens + slot) *
+          v.hidden) *
+             v.elem_size;
+}
+
+__host__ __device__ inline uint32_t* disp_meta_ptr(void* base,
+                                                   const EpView& v, int le,
+                                                   size_t slot) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
+                                     v.off_disp_meta) +
+         static_cast<size_t>(le) * v.world * v.max_tokens + slot;
+}
+
+__host__ __device__ inline uint64_t* comb_flag_ptr(void* base,
+                                                   const EpView& v,
+                                                   int src) {
+  return reinterpret_cast<uint64_t*>(static_cast<char*>(base) +
+                                     v.off_comb_flag) +
+         src;
+}
+
+__host__ __device__ inline char* comb_x_ptr(void* base, const EpView& v,
+                                            size_t tok, int k) {
+  return static_cast<char*>(base) + v.off_comb_x +
+         ((tok * v.topk + k) * static_cast<size_t>(v.hidden)) * v.elem_size;
+}
+
+}  // namespace ep
+}  // namespace uccl

@@ -1,0 +1,100 @@
+"""DeepEP-compatible expert-parallel communication for MI355X.
+
+API parity target: the reference's `deep_ep.Buffer`-replicating Python layer
+(ep/bench/buffer.py) — intranode xGMI path. One node, up to 8 ranks:
+
+    buf = uccl_amd.ep.Buffer(group, num_experts=64, topk=8,
+                             hidden=7168, max_tokens=4096,
+                             dtype=torch.bfloat16)
+    recv_x, recv_count, handle = buf.dispatch(x, topk_idx)
+    ...run local experts over recv_x[e, :recv_count_total(e)]...
+    combined = buf.combine(expert_out, topk_idx, topk_weights)
+
+Dispatch writes each token's hidden vector straight into every destination
+rank's per-(expert, source) slot array over xGMI (no proxy, no staging);
+recv_x is a zero-copy view of that slot memory. Combine returns expert
+outputs to the (token, k) cells of their source ranks and reduces with
+top-k weights in fp32.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+
+def get_dispatch_layout(topk_idx: torch.Tensor, num_experts: int,
+                        num_ranks: int):
+    """Token routing layout, mirroring the reference's get_dispatch_layout
+    (ep/src/layout.cu:10): returns (num_tokens_per_rank [R],
+    num_tokens_per_expert [E], is_token_in_rank [T, R] bool)."""
+    T, K = topk_idx.shape
+    valid = topk_idx >= 0
+    flat = topk_idx.clamp(min=0)
+    num_tokens_per_expert = torch.zeros(num_experts, dtype=torch.int32,
+                                        device=topk_idx.device)
+    num_tokens_per_expert.scatter_add_(
+        0, flat.reshape(-1),
+        valid.reshape(-1).to(torch.int32))
+    experts_per_rank = num_experts // num_ranks
+    token_rank = flat // experts_per_rank  # [T, K]
+    is_token_in_rank = torch.zeros(T, num_ranks, dtype=torch.bool,
+                                   device=topk_idx.device)
+    is_token_in_rank.scatter_(1, token_rank,
+                              valid)
+    num_tokens_per_rank = is_token_in_rank.sum(0).to(torch.int32)
+    return num_tokens_per_rank, num_tokens_per_expert, is_token_in_rank
+
+
+class Buffer:
+    def __init__(self, group=None, num_experts: int = 8, topk: int = 2,
+                 hidden: int = 7168, max_tokens: int = 4096,
+                 dtype: torch.dtype = torch.bfloat16,
+                 device: Optional[int] = None):
+        from uccl_amd import _load_native
+
+        C = _load_native(required=True)
+        import torch.distributed as dist
+
+        if dist.is_available() and dist.is_initialized():
+            self.rank = dist.get_rank(group)
+            self.world = dist.get_world_size(group)
+        else:
+            self.rank, self.world = 0, 1
+        if device is None:
+            device = torch.cuda.current_device()
+        self.device = device
+        self.dtype = dtype
+        self.num_experts = num_experts
+        self.topk = topk
+        self.hidden = hidden
+        self.max_tokens = max_tokens
+        elem = torch.tensor([], dtype=dtype).element_size()
+        self._b = C.EpBuffer(self.rank, self.world, device, num_experts,
+                             topk, hidden, max_tokens, elem)
+        if self.world > 1:
+            handles = [None] * self.world
+            dist.all_gather_object(handles, self._b.handle_bytes(),
+                                   group=group)
+            self._b.connect(handles)
+
+    @property
+    def local_experts(self) -> int:
+        return self.num_experts // self.world
+
+    def dispatch(self, x: torch.Tensor, topk_idx: torch.Tensor
+                 ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Returns (packed_recv_x [local_E, world*max_tokens, hidden] view,
+        recv_count [local_E, world] int32). Slots for source rank r live at
+        [e, r*max_tokens : r*max_tokens + recv_count[e, r]]."""
+        return self._b.dispatch(x, topk_idx)
+
+    def combine(self, expert_out: torch.Tensor, topk_idx: torch.Tensor,
+                topk_weights: torch.Tensor) -> torch.Tensor:
+        return self._b.combine(expert_out, topk_idx,
+                               topk_weights.float().contiguous())
+
+    # DeepEP-compatible aliases (low-latency intranode semantics)
+    low_latency_dispatch = dispatch
+    low_latency_combine = combine
