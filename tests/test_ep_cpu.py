@@ -12,7 +12,7 @@ def test_dispatch_layout_basic():
     assert npe.tolist() == [2, 1, 1, 3]
     assert in_rank.tolist() == [[True, True], [True, True], [False, True],
                                 [True, False]]
-    assert npr.tolist() == [3, 2]
+    assert npr.tolist() == [3, 3]
 
 
 def test_dispatch_layout_random_consistency():

@@ -39,10 +39,13 @@ def get_dispatch_layout(topk_idx: torch.Tensor, num_experts: int,
         valid.reshape(-1).to(torch.int32))
     experts_per_rank = num_experts // num_ranks
     token_rank = flat // experts_per_rank  # [T, K]
-    is_token_in_rank = torch.zeros(T, num_ranks, dtype=torch.bool,
+    # masked (idx<0) entries are routed to a throwaway bucket so they can
+    # never set a real rank bit (scatter order is undefined on duplicates)
+    tr = token_rank.masked_fill(~valid, num_ranks)
+    is_token_in_rank = torch.zeros(T, num_ranks + 1, dtype=torch.bool,
                                    device=topk_idx.device)
-    is_token_in_rank.scatter_(1, token_rank,
-                              valid)
+    is_token_in_rank.scatter_(1, tr, torch.ones_like(tr, dtype=torch.bool))
+    is_token_in_rank = is_token_in_rank[:, :num_ranks].contiguous()
     num_tokens_per_rank = is_token_in_rank.sum(0).to(torch.int32)
     return num_tokens_per_rank, num_tokens_per_expert, is_token_in_rank
 
