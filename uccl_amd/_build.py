@@ -46,6 +46,43 @@ def _stale(target: Path, deps) -> bool:
     return any(d.stat().st_mtime > t for d in deps)
 
 
+def build_plugin(verbose: bool = False) -> Path:
+    """Build librccl-net-uccl.so (RCCL net plugin, pure sockets, no HIP)
+    and the dlopen test harness."""
+    plugdir = PKG_DIR / "lib"
+    plugdir.mkdir(exist_ok=True)
+    target = plugdir / "librccl-net-uccl.so"
+    src = CSRC / "plugin" / "tcp_plugin.cpp"
+    harness_src = CSRC / "plugin" / "plugin_test_main.cpp"
+    harness = plugdir / "plugin_test"
+    deps = [src, harness_src] + list((CSRC / "plugin").glob("*.h")) + [
+        CSRC / "core" / "net.h", CSRC / "core" / "log.h",
+        CSRC / "core" / "env.h"]
+    if not _stale(target, deps) and not _stale(harness, deps):
+        return target
+    import subprocess as sp
+
+    for cmd in (
+        ["g++", "-O2", "-std=c++17", "-fPIC", "-shared", str(src), "-o",
+         str(target), "-pthread"],
+        ["g++", "-O2", "-std=c++17", str(harness_src), "-o", str(harness),
+         "-ldl", "-pthread"],
+    ):
+        if verbose:
+            print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)
+        r = sp.run(cmd, stdout=sp.PIPE, stderr=sp.STDOUT)
+        if r.returncode != 0:
+            raise RuntimeError(f"plugin build failed:\n{r.stdout.decode()}")
+    # NCCL-compatible alias
+    alias = plugdir / "libnccl-net-uccl.so"
+    if not alias.exists():
+        try:
+            alias.symlink_to(target.name)
+        except FileExistsError:
+            pass
+    return target
+
+
 def build(verbose: bool = False, force: bool = False) -> Path:
     """Compile the extension in-tree. Returns the path to the built .so."""
     target = PKG_DIR / EXT_NAME
@@ -119,6 +156,7 @@ def build(verbose: bool = False, force: bool = False) -> Path:
     r = subprocess.run(link, stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
     if r.returncode != 0:
         raise RuntimeError(f"uccl_amd link failed:\n{r.stdout.decode()}")
+    build_plugin(verbose=verbose)
     return target
 
 
