@@ -24,6 +24,7 @@ SOURCES = [
     CSRC / "collective" / "kernels.hip",
     CSRC / "collective" / "communicator.cpp",
     CSRC / "p2p" / "endpoint.cpp",
+    CSRC / "transport" / "reliable.cpp",
     CSRC / "ep" / "ep_kernels.hip",
     CSRC / "ep" / "ep_buffer.cpp",
     CSRC / "bindings" / "module.cpp",

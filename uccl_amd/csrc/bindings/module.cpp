@@ -13,6 +13,7 @@
 #include "../core/log.h"
 #include "../ep/ep_buffer.h"
 #include "../p2p/endpoint.h"
+#include "../transport/reliable.h"
 
 namespace py = pybind11;
 using uccl::Communicator;
@@ -287,4 +288,49 @@ PYBIND11_MODULE(_C, m) {
                        current_stream(b.device()));
              return out;
            });
+
+  // --- software multipath reliable transport ------------------------------
+  using uccl::transport::TransportEndpoint;
+  py::class_<uccl::transport::Stats>(m, "TransportStats")
+      .def_readonly("data_sent", &uccl::transport::Stats::data_sent)
+      .def_readonly("data_recv", &uccl::transport::Stats::data_recv)
+      .def_readonly("acks_sent", &uccl::transport::Stats::acks_sent)
+      .def_readonly("acks_recv", &uccl::transport::Stats::acks_recv)
+      .def_readonly("retransmits", &uccl::transport::Stats::retransmits)
+      .def_readonly("rto_retransmits",
+                    &uccl::transport::Stats::rto_retransmits)
+      .def_readonly("injected_drops",
+                    &uccl::transport::Stats::injected_drops)
+      .def_readonly("msgs_sent", &uccl::transport::Stats::msgs_sent)
+      .def_readonly("msgs_recv", &uccl::transport::Stats::msgs_recv)
+      .def_readonly("srtt_us", &uccl::transport::Stats::srtt_us)
+      .def_readonly("cwnd", &uccl::transport::Stats::cwnd);
+
+  py::class_<TransportEndpoint>(m, "TransportEndpoint")
+      .def(py::init<int, size_t>(), py::arg("num_paths") = 8,
+           py::arg("chunk_bytes") = 8192)
+      .def("metadata", &TransportEndpoint::metadata)
+      .def("connect", &TransportEndpoint::connect,
+           py::call_guard<py::gil_scoped_release>())
+      .def("accept", &TransportEndpoint::accept,
+           py::call_guard<py::gil_scoped_release>())
+      .def("send",
+           [](TransportEndpoint& e, uint64_t flow, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),
+                         "transport send: host tensors");
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.send_msg(flow, p, n);
+           })
+      .def("recv",
+           [](TransportEndpoint& e, uint64_t flow, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),
+                         "transport recv: host tensors");
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.recv_msg(flow, p, n);
+           })
+      .def("stats", &TransportEndpoint::stats);
 }

@@ -1,0 +1,558 @@
+#include "reliable.h"
+
+#include <poll.h>
+#include <sys/socket.h>
+
+#include <algorithm>
+#include <cstdlib>
+#include <cstring>
+
+#include "../core/env.h"
+#include "../core/log.h"
+#include "../core/net.h"
+
+namespace uccl {
+namespace transport {
+
+namespace {
+
+constexpr uint32_t kMagic = 0x55434354;  // "UCCT"
+enum Kind : uint32_t { kData = 1, kAck = 2, kWake = 3 };
+
+struct DataHdr {
+  uint32_t magic;
+  uint32_t kind;
+  uint64_t flow;
+  uint64_t msg_id;
+  uint64_t msg_bytes;
+  uint64_t off;
+  uint32_t len;
+  uint32_t csn;
+  uint64_t ts_ns;
+};
+
+struct AckHdr {
+  uint32_t magic;
+  uint32_t kind;
+  uint64_t flow;
+  uint32_t cum;   // all csn < cum received
+  uint32_t pad;
+  uint64_t sack0;  // bits for csn in [cum, cum+64)
+  uint64_t sack1;  // [cum+64, cum+128)
+  uint64_t ts_echo;
+};
+
+uint64_t now_ns() {
+  timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return static_cast<uint64_t>(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
+}
+
+// deterministic per-packet drop decision (reproducible loss injection,
+// analogous to the reference's compile-time kTestLoss knobs)
+bool inject_drop(uint32_t csn, uint32_t attempt, int pct) {
+  if (pct <= 0) return false;
+  uint64_t h = (static_cast<uint64_t>(csn) << 20) ^ (attempt * 0x9e3779b9u);
+  h ^= h >> 33;
+  h *= 0xff51afd7ed558ccdULL;
+  h ^= h >> 33;
+  return static_cast<int>(h % 100) < pct;
+}
+
+}  // namespace
+
+struct MsgTx {
+  uint64_t id;
+  char const* ptr;
+  size_t bytes;
+  size_t next_off = 0;       // next byte to chunk out
+  size_t acked_bytes = 0;
+  bool done = false;
+};
+
+struct ChunkTx {
+  std::shared_ptr<MsgTx> msg;
+  uint64_t off;
+  uint32_t len;
+  uint64_t send_ts = 0;
+  uint32_t attempts = 0;
+  int dupacks = 0;
+};
+
+struct MsgRx {
+  char* user_ptr = nullptr;
+  std::vector<char> staging;
+  size_t bytes = 0;
+  size_t recv_bytes = 0;
+  bool known = false;  // first chunk seen
+  char* dest() { return user_ptr ? user_ptr : staging.data(); }
+};
+
+struct TransportEndpoint::Flow {
+  uint64_t id;
+  sockaddr_in peer_paths[64];
+  int num_paths;
+
+  // --- TX direction ---
+  uint32_t next_csn = 0;
+  std::map<uint32_t, ChunkTx> inflight;   // csn -> chunk
+  std::deque<std::shared_ptr<MsgTx>> txq;  // messages not fully chunked
+  uint64_t next_tx_msg = 0;
+  double cwnd = 16.0;
+  double srtt_us = 0.0;
+  double prev_rtt_us = 0.0;
+  uint32_t tx_cum = 0;  // lowest unacked csn
+
+  // --- RX direction ---
+  uint32_t rx_cum = 0;  // all csn < rx_cum received
+  std::map<uint32_t, bool> rx_ooo;  // received csn >= rx_cum
+  std::unordered_map<uint64_t, MsgRx> rxmsgs;
+  uint64_t next_post_msg = 0;   // msg_id the next recv_msg call will take
+  uint64_t next_done_msg = 0;   // completion watermark for in-order delivery
+  uint64_t last_data_ts = 0;    // ts to echo in acks
+};
+
+struct TransportEndpoint::Impl {
+  int num_paths;
+  size_t chunk_bytes;
+  int loss_pct;
+  std::vector<int> socks;          // UDP path sockets
+  std::vector<uint16_t> ports;
+  int wake_fd = -1;                // self-addressed UDP for wakeups
+  uint16_t wake_port = 0;
+  int ctrl_listen = -1;
+  uint16_t ctrl_port = 0;
+  std::thread ctrl_thread;
+  std::thread progress;
+  std::atomic<bool> stop{false};
+
+  std::mutex mu;
+  std::condition_variable cv;
+  std::unordered_map<uint64_t, std::unique_ptr<Flow>> flows;
+  std::deque<uint64_t> accepted;
+  std::atomic<uint64_t> next_flow{1};
+
+  Stats st;
+
+  // ---- helpers ----
+  void wake() {
+    sockaddr_in a{};
+    a.sin_family = AF_INET;
+    a.sin_port = htons(wake_port);
+    inet_pton(AF_INET, "127.0.0.1", &a.sin_addr);
+    uint32_t w[2] = {kMagic, kWake};
+    (void)sendto(wake_fd, w, sizeof(w), 0, reinterpret_cast<sockaddr*>(&a),
+                 sizeof(a));
+  }
+
+  void send_chunk(Flow& f, uint32_t csn, ChunkTx& c) {
+    DataHdr h{kMagic, kData, f.id, c.msg->id, c.msg->bytes, c.off, c.len,
+              csn, now_ns()};
+    char buf[sizeof(DataHdr) + 65536];
+    memcpy(buf, &h, sizeof(h));
+    memcpy(buf + sizeof(h), c.msg->ptr + c.off, c.len);
+    int const path = csn % f.num_paths;  // spray round-robin across paths
+    c.send_ts = now_ns();
+    ++c.attempts;
+    if (inject_drop(csn, c.attempts, loss_pct)) {
+      ++st.injected_drops;
+      return;  // "sent" into the void
+    }
+    (void)sendto(socks[path], buf, sizeof(h) + c.len, 0,
+                 reinterpret_cast<sockaddr*>(&f.peer_paths[path]),
+                 sizeof(sockaddr_in));
+    ++st.data_sent;
+  }
+
+  void pump_tx(Flow& f) {
+    while (static_cast<double>(f.inflight.size()) < f.cwnd && !f.txq.empty()) {
+      auto m = f.txq.front();
+      uint32_t const csn = f.next_csn++;
+      ChunkTx c;
+      c.msg = m;
+      c.off = m->next_off;
+      c.len = static_cast<uint32_t>(
+          std::min(chunk_bytes, m->bytes - m->next_off));
+      m->next_off += c.len;
+      if (m->next_off >= m->bytes) f.txq.pop_front();
+      send_chunk(f, csn, c);
+      f.inflight.emplace(csn, std::move(c));
+      if (m->bytes == 0) {  // zero-byte message: len-0 chunk carries it
+        break;
+      }
+    }
+  }
+
+  void send_ack(Flow& f, int sock_idx, sockaddr_in const& to) {
+    AckHdr a{kMagic, kAck, f.id, f.rx_cum, 0, 0, 0, f.last_data_ts};
+    for (auto const& [csn, _] : f.rx_ooo) {
+      uint32_t const d = csn - f.rx_cum;
+      if (d < 64)
+        a.sack0 |= 1ull << d;
+      else if (d < 128)
+        a.sack1 |= 1ull << (d - 64);
+      else
+        break;
+    }
+    (void)sendto(socks[sock_idx], &a, sizeof(a), 0,
+                 reinterpret_cast<sockaddr const*>(&to), sizeof(to));
+    ++st.acks_sent;
+  }
+
+  void timely_update(Flow& f, double rtt_us) {
+    // Timely-style RTT-gradient window adaptation (SIGCOMM'15 adapted to
+    // window form; pacing is bypassed, matching the reference default
+    // BYPASS_PACING=1, transport_config.h:35)
+    static double const t_low = env_int("UCCL_TP_TLOW_US", 50);
+    static double const t_high = env_int("UCCL_TP_THIGH_US", 1000);
+    static double const add = 1.0, beta = 0.8;
+    if (f.srtt_us == 0) f.srtt_us = rtt_us;
+    double const grad = (rtt_us - f.prev_rtt_us) / std::max(f.srtt_us, 1.0);
+    f.prev_rtt_us = rtt_us;
+    f.srtt_us = 0.875 * f.srtt_us + 0.125 * rtt_us;
+    if (rtt_us < t_low) {
+      f.cwnd += add;
+    } else if (rtt_us > t_high) {
+      f.cwnd *= 1.0 - beta * (1.0 - t_high / rtt_us);
+    } else if (grad <= 0) {
+      f.cwnd += add;
+    } else {
+      f.cwnd *= 1.0 - beta * std::min(grad, 0.25);
+    }
+    f.cwnd = std::min(std::max(f.cwnd, 2.0), 4096.0);
+    st.srtt_us = f.srtt_us;
+    st.cwnd = f.cwnd;
+  }
+
+  void ack_chunk(Flow& f, uint32_t csn) {
+    auto it = f.inflight.find(csn);
+    if (it == f.inflight.end()) return;
+    it->second.msg->acked_bytes += it->second.len;
+    auto& m = *it->second.msg;
+    if (m.acked_bytes >= m.bytes && !m.done) {
+      m.done = true;
+      ++st.msgs_sent;
+      cv.notify_all();
+    }
+    f.inflight.erase(it);
+  }
+
+  void handle_ack(Flow& f, AckHdr const& a) {
+    ++st.acks_recv;
+    if (a.ts_echo) {
+      double const rtt_us = (now_ns() - a.ts_echo) / 1000.0;
+      timely_update(f, rtt_us);
+    }
+    // cumulative
+    while (!f.inflight.empty() && f.inflight.begin()->first < a.cum)
+      ack_chunk(f, f.inflight.begin()->first);
+    // SACK bits
+    uint32_t highest_sacked = a.cum;
+    for (int i = 0; i < 128; ++i) {
+      bool const set = i < 64 ? (a.sack0 >> i) & 1 : (a.sack1 >> (i - 64)) & 1;
+      if (set) {
+        ack_chunk(f, a.cum + i);
+        highest_sacked = a.cum + i;
+      }
+    }
+    // dup-ack style fast retransmit: unacked csns below the highest SACKed
+    // one accumulate "holes seen" counts. The threshold is large (default
+    // 32) because packet spraying across paths reorders heavily — the same
+    // design point as the reference's ROCE_DUP_ACK_THRES=32
+    // (collective/rdma/transport_config.h:145).
+    static int const dup_thres =
+        static_cast<int>(env_int("UCCL_TP_DUPACK_THRES", 32));
+    for (auto& [csn, c] : f.inflight) {
+      if (csn >= highest_sacked) break;
+      if (++c.dupacks >= dup_thres) {
+        c.dupacks = 0;
+        ++st.retransmits;
+        send_chunk(f, csn, c);
+      }
+    }
+    pump_tx(f);
+  }
+
+  void handle_data(Flow& f, DataHdr const& h, char const* payload,
+                   int sock_idx, sockaddr_in const& from) {
+    ++st.data_recv;
+    f.last_data_ts = h.ts_ns;
+    bool const fresh =
+        (h.csn >= f.rx_cum) && !f.rx_ooo.count(h.csn);
+    if (fresh) {
+      auto& m = f.rxmsgs[h.msg_id];
+      if (!m.known) {
+        m.known = true;
+        m.bytes = h.msg_bytes;
+        if (!m.user_ptr && m.staging.empty()) m.staging.resize(h.msg_bytes);
+      }
+      if (h.len) memcpy(m.dest() + h.off, payload, h.len);
+      m.recv_bytes += h.len;
+      f.rx_ooo[h.csn] = true;
+      while (f.rx_ooo.count(f.rx_cum)) {
+        f.rx_ooo.erase(f.rx_cum);
+        ++f.rx_cum;
+      }
+      if (m.recv_bytes >= m.bytes) {
+        ++st.msgs_recv;
+        cv.notify_all();
+      }
+    }
+    send_ack(f, sock_idx, from);
+  }
+
+  void rto_scan() {
+    static uint64_t const rto_ns =
+        static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 20000)) * 1000;
+    uint64_t const now = now_ns();
+    for (auto& [fid, fp] : flows) {
+      Flow& f = *fp;
+      for (auto& [csn, c] : f.inflight) {
+        uint64_t const rto =
+            std::max<uint64_t>(rto_ns, 4ull * 1000 *
+                                           static_cast<uint64_t>(f.srtt_us));
+        if (c.send_ts && now - c.send_ts > rto) {
+          ++st.rto_retransmits;
+          f.cwnd = std::max(2.0, f.cwnd / 2);
+          send_chunk(f, csn, c);
+        }
+      }
+      pump_tx(f);
+    }
+  }
+
+  void progress_loop() {
+    std::vector<pollfd> pfds;
+    for (int s : socks) pfds.push_back({s, POLLIN, 0});
+    pfds.push_back({wake_fd, POLLIN, 0});
+    std::vector<char> buf(sizeof(DataHdr) + 65536 + 64);
+    while (!stop) {
+      (void)poll(pfds.data(), pfds.size(), 5);
+      std::lock_guard<std::mutex> g(mu);
+      for (size_t i = 0; i < pfds.size(); ++i) {
+        while (true) {
+          sockaddr_in from{};
+          socklen_t fl = sizeof(from);
+          ssize_t n = recvfrom(pfds[i].fd, buf.data(), buf.size(),
+                               MSG_DONTWAIT,
+                               reinterpret_cast<sockaddr*>(&from), &fl);
+          if (n <= 0) break;
+          if (n < static_cast<ssize_t>(8)) continue;
+          auto kind = reinterpret_cast<uint32_t const*>(buf.data())[1];
+          if (kind == kWake) continue;
+          if (kind == kData && n >= static_cast<ssize_t>(sizeof(DataHdr))) {
+            auto const* h = reinterpret_cast<DataHdr const*>(buf.data());
+            auto it = flows.find(h->flow);
+            if (it != flows.end())
+              handle_data(*it->second, *h, buf.data() + sizeof(DataHdr),
+                          static_cast<int>(i), from);
+          } else if (kind == kAck &&
+                     n >= static_cast<ssize_t>(sizeof(AckHdr))) {
+            auto const* a = reinterpret_cast<AckHdr const*>(buf.data());
+            auto it = flows.find(a->flow);
+            if (it != flows.end()) handle_ack(*it->second, *a);
+          }
+        }
+      }
+      rto_scan();
+    }
+  }
+
+  // ---- flow setup over TCP ctrl ----
+  struct CtrlMsg {
+    uint64_t flow;
+    int num_paths;
+    uint16_t ports[64];
+    char ip[48];
+  };
+
+  void ctrl_loop() {
+    while (!stop) {
+      int fd = ::accept(ctrl_listen, nullptr, nullptr);
+      if (fd < 0) {
+        if (stop) return;
+        continue;
+      }
+      CtrlMsg peer{};
+      if (!net::recv_all(fd, &peer, sizeof(peer))) {
+        ::close(fd);
+        continue;
+      }
+      uint64_t const flow = next_flow.fetch_add(1);
+      CtrlMsg mine = self_ctrl(flow);
+      net::send_all(fd, &mine, sizeof(mine));
+      ::close(fd);
+      install_flow(flow, peer);
+      {
+        std::lock_guard<std::mutex> g(mu);
+        accepted.push_back(flow);
+      }
+      cv.notify_all();
+    }
+  }
+
+  CtrlMsg self_ctrl(uint64_t flow) {
+    CtrlMsg m{};
+    m.flow = flow;
+    m.num_paths = num_paths;
+    for (int i = 0; i < num_paths; ++i) m.ports[i] = ports[i];
+    std::string ip = net::local_ip();
+    strncpy(m.ip, ip.c_str(), sizeof(m.ip) - 1);
+    return m;
+  }
+
+  void install_flow(uint64_t flow, CtrlMsg const& peer) {
+    auto f = std::make_unique<Flow>();
+    f->id = flow;
+    f->num_paths = std::min(num_paths, peer.num_paths);
+    for (int i = 0; i < f->num_paths; ++i) {
+      sockaddr_in a{};
+      a.sin_family = AF_INET;
+      a.sin_port = htons(peer.ports[i]);
+      inet_pton(AF_INET, peer.ip, &a.sin_addr);
+      f->peer_paths[i] = a;
+    }
+    std::lock_guard<std::mutex> g(mu);
+    flows[flow] = std::move(f);
+  }
+};
+
+TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
+    : impl_(new Impl()) {
+  UCCL_CHECK(num_paths >= 1 && num_paths <= 64) << "1..64 paths";
+  UCCL_CHECK(chunk_bytes >= 512 && chunk_bytes <= 60000)
+      << "chunk must fit a UDP datagram";
+  impl_->num_paths = num_paths;
+  impl_->chunk_bytes = chunk_bytes;
+  impl_->loss_pct = static_cast<int>(env_int("UCCL_TP_LOSS_PCT", 0));
+  for (int i = 0; i < num_paths; ++i) {
+    int s = ::socket(AF_INET, SOCK_DGRAM, 0);
+    UCCL_CHECK(s >= 0) << "udp socket";
+    int sz = 4 << 20;
+    setsockopt(s, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+    setsockopt(s, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    sockaddr_in a{};
+    a.sin_family = AF_INET;
+    a.sin_addr.s_addr = htonl(INADDR_ANY);
+    UCCL_CHECK(::bind(s, reinterpret_cast<sockaddr*>(&a), sizeof(a)) == 0);
+    socklen_t al = sizeof(a);
+    getsockname(s, reinterpret_cast<sockaddr*>(&a), &al);
+    impl_->socks.push_back(s);
+    impl_->ports.push_back(ntohs(a.sin_port));
+  }
+  {
+    impl_->wake_fd = ::socket(AF_INET, SOCK_DGRAM, 0);
+    sockaddr_in a{};
+    a.sin_family = AF_INET;
+    a.sin_addr.s_addr = htonl(INADDR_ANY);
+    UCCL_CHECK(::bind(impl_->wake_fd, reinterpret_cast<sockaddr*>(&a),
+                      sizeof(a)) == 0);
+    socklen_t al = sizeof(a);
+    getsockname(impl_->wake_fd, reinterpret_cast<sockaddr*>(&a), &al);
+    impl_->wake_port = ntohs(a.sin_port);
+  }
+  impl_->ctrl_listen = net::listen_on(&impl_->ctrl_port);
+  impl_->ctrl_thread = std::thread([this] { impl_->ctrl_loop(); });
+  impl_->progress = std::thread([this] { impl_->progress_loop(); });
+}
+
+TransportEndpoint::~TransportEndpoint() {
+  impl_->stop = true;
+  ::shutdown(impl_->ctrl_listen, SHUT_RDWR);
+  ::close(impl_->ctrl_listen);
+  impl_->wake();
+  if (impl_->ctrl_thread.joinable()) impl_->ctrl_thread.join();
+  if (impl_->progress.joinable()) impl_->progress.join();
+  for (int s : impl_->socks) ::close(s);
+  ::close(impl_->wake_fd);
+}
+
+std::string TransportEndpoint::metadata() const {
+  char buf[64];
+  snprintf(buf, sizeof(buf), "%s:%u", net::local_ip().c_str(),
+           impl_->ctrl_port);
+  return buf;
+}
+
+uint64_t TransportEndpoint::connect(const std::string& md) {
+  auto const pos = md.rfind(':');
+  UCCL_CHECK(pos != std::string::npos) << "bad transport metadata";
+  std::string ip = md.substr(0, pos);
+  uint16_t port = static_cast<uint16_t>(atoi(md.c_str() + pos + 1));
+  int fd = net::connect_to(ip, port);
+  Impl::CtrlMsg mine = impl_->self_ctrl(0);
+  net::send_all(fd, &mine, sizeof(mine));
+  Impl::CtrlMsg peer{};
+  UCCL_CHECK(net::recv_all(fd, &peer, sizeof(peer))) << "ctrl handshake";
+  ::close(fd);
+  impl_->install_flow(peer.flow, peer);
+  return peer.flow;
+}
+
+uint64_t TransportEndpoint::accept() {
+  std::unique_lock<std::mutex> lk(impl_->mu);
+  impl_->cv.wait(lk, [this] {
+    return !impl_->accepted.empty() || impl_->stop;
+  });
+  UCCL_CHECK(!impl_->accepted.empty()) << "endpoint closed";
+  uint64_t f = impl_->accepted.front();
+  impl_->accepted.pop_front();
+  return f;
+}
+
+void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
+                                 size_t bytes) {
+  std::shared_ptr<MsgTx> m;
+  {
+    std::lock_guard<std::mutex> g(impl_->mu);
+    auto it = impl_->flows.find(flow);
+    UCCL_CHECK(it != impl_->flows.end()) << "unknown flow " << flow;
+    Flow& f = *it->second;
+    m = std::make_shared<MsgTx>();
+    m->id = f.next_tx_msg++;
+    m->ptr = static_cast<char const*>(ptr);
+    m->bytes = bytes;
+    f.txq.push_back(m);
+    impl_->pump_tx(f);
+  }
+  impl_->wake();
+  std::unique_lock<std::mutex> lk(impl_->mu);
+  impl_->cv.wait(lk, [&] { return m->done || impl_->stop; });
+  UCCL_CHECK(m->done) << "endpoint closed during send";
+}
+
+void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
+  uint64_t msg_id;
+  {
+    std::lock_guard<std::mutex> g(impl_->mu);
+    auto it = impl_->flows.find(flow);
+    UCCL_CHECK(it != impl_->flows.end()) << "unknown flow " << flow;
+    Flow& f = *it->second;
+    msg_id = f.next_post_msg++;
+    auto& m = f.rxmsgs[msg_id];
+    if (!m.staging.empty()) {
+      memcpy(ptr, m.staging.data(), std::min(bytes, m.staging.size()));
+      m.staging.clear();
+      m.staging.shrink_to_fit();
+    }
+    m.user_ptr = static_cast<char*>(ptr);
+  }
+  std::unique_lock<std::mutex> lk(impl_->mu);
+  Flow& f = *impl_->flows[flow];
+  impl_->cv.wait(lk, [&] {
+    auto it = f.rxmsgs.find(msg_id);
+    return (it != f.rxmsgs.end() && it->second.known &&
+            it->second.recv_bytes >= it->second.bytes) ||
+           impl_->stop;
+  });
+  UCCL_CHECK(!impl_->stop) << "endpoint closed during recv";
+  f.rxmsgs.erase(msg_id);
+}
+
+Stats TransportEndpoint::stats() const {
+  std::lock_guard<std::mutex> g(impl_->mu);
+  return impl_->st;
+}
+
+}  // namespace transport
+}  // namespace uccl
