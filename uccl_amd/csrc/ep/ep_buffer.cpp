@@ -52,6 +52,9 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   v_.off_comb_x = off;
   off = align256(off + static_cast<size_t>(max_tokens) * topk * hidden *
                            elem_size);
+  v_.off_plan = off;
+  off = align256(off + sizeof(uint32_t) * num_experts *
+                           (1 + static_cast<size_t>(max_tokens)));
   v_.heap_bytes = off;
 
   UCCL_CHECK_HIP(hipSetDevice(device_));

@@ -38,6 +38,8 @@ struct EpView {
                           //   meta = src_token_idx | (k << 24)
   size_t off_comb_flag;   // u64 [world]                   (seq)
   size_t off_comb_x;      // [max_tokens][topk][hidden] elems
+  size_t off_plan;        // u32 [num_experts][1 + max_tokens]  (private
+                          //   per-rank scratch: count then (t|k<<24) list)
   size_t heap_bytes;
 };
 
@@ -65,6 +67,12 @@ __host__ __device__ inline uint32_t* disp_meta_ptr(void* base,
   return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
                                      v.off_disp_meta) +
          static_cast<size_t>(le) * v.world * v.max_tokens + slot;
+}
+
+__host__ __device__ inline uint32_t* plan_ptr(void* base, const EpView& v,
+                                              int e) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) + v.off_plan) +
+         static_cast<size_t>(e) * (1 + v.max_tokens);
 }
 
 __host__ __device__ inline uint64_t* comb_flag_ptr(void* base,
