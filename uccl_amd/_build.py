@@ -23,6 +23,7 @@ GPU_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 SOURCES = [
     CSRC / "collective" / "kernels.hip",
     CSRC / "collective" / "communicator.cpp",
+    CSRC / "collective" / "pg_backend.cpp",
     CSRC / "p2p" / "endpoint.cpp",
     CSRC / "transport" / "reliable.cpp",
     CSRC / "ep" / "ep_kernels.hip",

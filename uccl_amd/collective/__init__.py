@@ -90,3 +90,26 @@ def init(group=None, device: Optional[int] = None,
         dist.all_gather_object(handles, comm.handle_bytes(), group=group)
         comm.connect(handles)
     return comm
+
+
+_backend_registered = False
+
+
+def register_torch_backend() -> None:
+    """Register the native "uccl" c10d backend so that
+    torch.distributed.init_process_group(backend="uccl") routes collectives
+    through the xGMI engine directly (drop-in, no code changes)."""
+    global _backend_registered
+    if _backend_registered:
+        return
+    import torch.distributed as dist
+
+    from uccl_amd import _load_native
+
+    C = _load_native(required=True)
+
+    def _creator(store, rank, size, timeout):
+        return C._create_uccl_backend(store, rank, size, timeout)
+
+    dist.Backend.register_backend("uccl", _creator, devices=["cuda"])
+    _backend_registered = True

@@ -16,6 +16,11 @@
 #include "../transport/reliable.h"
 
 namespace py = pybind11;
+
+namespace uccl {
+void register_pg_backend(pybind11::module_& m);
+}
+
 using uccl::Communicator;
 using uccl::Dtype;
 using uccl::p2p::Endpoint;
@@ -46,6 +51,8 @@ void check_tensor(const at::Tensor& t) {
 
 PYBIND11_MODULE(_C, m) {
   m.doc() = "uccl_amd native engine (MI355X / gfx950)";
+
+  uccl::register_pg_backend(m);
 
   m.def("device_count", [] {
     int n = 0;
