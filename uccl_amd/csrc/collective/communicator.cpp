@@ -130,6 +130,7 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
     return;
   }
   UCCL_CHECK(connected_) << "connect() not called";
+  UCCL_CHECK(dt != Dtype::kU8) << "allreduce needs a typed dtype";
   size_t const es = dtype_size(dt);
   size_t const bytes = count * es;
 
@@ -194,6 +195,7 @@ void Communicator::reduce_scatter(void* out, void const* in,
     return;
   }
   UCCL_CHECK(connected_) << "connect() not called";
+  UCCL_CHECK(dt != Dtype::kU8) << "reduce_scatter needs a typed dtype";
   size_t const chunk_elems = (scratch_cap_ / es / world_) & ~size_t(63);
   for (size_t off = 0; off < count_per_rank; off += chunk_elems) {
     size_t const n = std::min(chunk_elems, count_per_rank - off);

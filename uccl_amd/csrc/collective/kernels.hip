@@ -450,6 +450,7 @@ void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s) {
     case Dtype::kF16: fn<__half>(__VA_ARGS__); break;   \
     case Dtype::kBF16: fn<__hip_bfloat16>(__VA_ARGS__); break; \
     case Dtype::kI32: fn<int>(__VA_ARGS__); break;      \
+    case Dtype::kU8: /* reducing collectives reject kU8 host-side */ break; \
   }
 
 template <typename T>

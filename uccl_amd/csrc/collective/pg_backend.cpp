@@ -11,6 +11,8 @@
 
 #include <torch/python.h>
 
+#include <pybind11/chrono.h>
+
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 #include <torch/csrc/distributed/c10d/Backend.hpp>
@@ -123,8 +125,8 @@ class UcclBackend : public c10d::Backend {
     auto& t = tensors[0];
     check(t);
     hipStream_t s = cur_stream(comm_->device());
-    comm_->broadcast(t.data_ptr(), t.numel(), to_dtype(t),
-                     static_cast<int>(opts.rootRank), s);
+    comm_->broadcast(t.data_ptr(), t.numel() * t.element_size(),
+                     Dtype::kU8, static_cast<int>(opts.rootRank), s);
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::BROADCAST,
                                          s, tensors);
   }
@@ -135,9 +137,10 @@ class UcclBackend : public c10d::Backend {
     check(input);
     check(output);
     TORCH_CHECK(output.numel() == input.numel() * getSize());
+    TORCH_CHECK(output.scalar_type() == input.scalar_type());
     hipStream_t s = cur_stream(comm_->device());
-    comm_->all_gather(output.data_ptr(), input.data_ptr(), input.numel(),
-                      to_dtype(input), s);
+    comm_->all_gather(output.data_ptr(), input.data_ptr(),
+                      input.numel() * input.element_size(), Dtype::kU8, s);
     return c10::make_intrusive<UcclWork>(
         getRank(), c10d::OpType::_ALLGATHER_BASE, s,
         std::vector<at::Tensor>{output});
@@ -153,8 +156,8 @@ class UcclBackend : public c10d::Backend {
     TORCH_CHECK(static_cast<int>(outputs[0].size()) == getSize());
     hipStream_t s = cur_stream(comm_->device());
     at::Tensor flat = at::empty({getSize() * in.numel()}, in.options());
-    comm_->all_gather(flat.data_ptr(), in.data_ptr(), in.numel(),
-                      to_dtype(in), s);
+    comm_->all_gather(flat.data_ptr(), in.data_ptr(),
+                      in.numel() * in.element_size(), Dtype::kU8, s);
     for (int r = 0; r < getSize(); ++r)
       outputs[0][r].view(-1).copy_(
           flat.narrow(0, r * in.numel(), in.numel()), /*non_blocking=*/true);
@@ -190,7 +193,8 @@ class UcclBackend : public c10d::Backend {
                 input.numel() % getSize() == 0);
     hipStream_t s = cur_stream(comm_->device());
     comm_->all_to_all(output.data_ptr(), input.data_ptr(),
-                      input.numel() / getSize(), to_dtype(input), s);
+                      input.numel() / getSize() * input.element_size(),
+                      Dtype::kU8, s);
     return c10::make_intrusive<UcclWork>(
         getRank(), c10d::OpType::ALLTOALL_BASE, s,
         std::vector<at::Tensor>{output});
