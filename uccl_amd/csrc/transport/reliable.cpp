@@ -219,7 +219,9 @@ struct TransportEndpoint::Impl {
     } else {
       f.cwnd *= 1.0 - beta * std::min(grad, 0.25);
     }
-    f.cwnd = std::min(std::max(f.cwnd, 2.0), 4096.0);
+    static double const cwnd_max =
+        static_cast<double>(env_int("UCCL_TP_CWND_MAX", 1024));
+    f.cwnd = std::min(std::max(f.cwnd, 2.0), cwnd_max);
     st.srtt_us = f.srtt_us;
     st.cwnd = f.cwnd;
   }
@@ -428,9 +430,13 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   for (int i = 0; i < num_paths; ++i) {
     int s = ::socket(AF_INET, SOCK_DGRAM, 0);
     UCCL_CHECK(s >= 0) << "udp socket";
-    int sz = 4 << 20;
-    setsockopt(s, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
-    setsockopt(s, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    int sz = 16 << 20;
+    // FORCE variants bypass net.core.{r,w}mem_max when running as root —
+    // without them loopback drops under bursts and masquerades as loss
+    if (setsockopt(s, SOL_SOCKET, SO_RCVBUFFORCE, &sz, sizeof(sz)) != 0)
+      setsockopt(s, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+    if (setsockopt(s, SOL_SOCKET, SO_SNDBUFFORCE, &sz, sizeof(sz)) != 0)
+      setsockopt(s, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
     sockaddr_in a{};
     a.sin_family = AF_INET;
     a.sin_addr.s_addr = htonl(INADDR_ANY);
