@@ -101,7 +101,9 @@ struct TransportEndpoint::Flow {
   double cwnd = 16.0;
   double srtt_us = 0.0;
   double prev_rtt_us = 0.0;
-  uint32_t tx_cum = 0;  // lowest unacked csn
+  uint32_t tx_cum = 0;       // lowest unacked csn
+  uint32_t last_cum = 0;     // cum of the previous ack (hole detection)
+  int hole_dupacks = 0;      // acks with stalled cum + new SACKs above
 
   // --- RX direction ---
   uint32_t rx_cum = 0;  // all csn < rx_cum received
@@ -257,19 +259,26 @@ struct TransportEndpoint::Impl {
         highest_sacked = a.cum + i;
       }
     }
-    // dup-ack style fast retransmit: unacked csns below the highest SACKed
-    // one accumulate "holes seen" counts. The threshold is large (default
-    // 32) because packet spraying across paths reorders heavily — the same
-    // design point as the reference's ROCE_DUP_ACK_THRES=32
-    // (collective/rdma/transport_config.h:145).
+    // SACK-hole fast retransmit: spraying reorders heavily (and the
+    // progress loop drains path sockets in batches), so per-chunk dup-ack
+    // counting misfires. Instead, count acks whose cumulative edge is
+    // STALLED while SACKs keep arriving above it — the classic SACK loss
+    // signal — and retransmit only the first hole. The threshold stays
+    // large for the same reason the reference uses ROCE_DUP_ACK_THRES=32
+    // (collective/rdma/transport_config.h:145); RTO backstops the rest.
     static int const dup_thres =
         static_cast<int>(env_int("UCCL_TP_DUPACK_THRES", 32));
-    for (auto& [csn, c] : f.inflight) {
-      if (csn >= highest_sacked) break;
-      if (++c.dupacks >= dup_thres) {
-        c.dupacks = 0;
-        ++st.retransmits;
-        send_chunk(f, csn, c);
+    if (a.cum != f.last_cum) {
+      f.last_cum = a.cum;
+      f.hole_dupacks = 0;
+    } else if (highest_sacked > a.cum) {
+      if (++f.hole_dupacks >= dup_thres) {
+        f.hole_dupacks = 0;
+        auto hole = f.inflight.find(a.cum);
+        if (hole != f.inflight.end()) {
+          ++st.retransmits;
+          send_chunk(f, hole->first, hole->second);
+        }
       }
     }
     pump_tx(f);
