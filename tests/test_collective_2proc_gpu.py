@@ -22,12 +22,15 @@ REPO = Path(__file__).resolve().parent.parent
 WORKER = REPO / "tests" / "workers" / "collective_worker.py"
 
 
-def test_two_process_collectives():
-    world = 2
+import pytest
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_two_process_collectives(world):
     env_base = dict(os.environ)
     env_base.update({
         "MASTER_ADDR": "127.0.0.1",
-        "MASTER_PORT": "29471",
+        "MASTER_PORT": str(29471 + world),
         "WORLD_SIZE": str(world),
         "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         "PYTHONPATH": str(REPO),

@@ -186,6 +186,24 @@ def main():
         check(f"stress[{i}]", x, expected_sum(world, 257, torch.float32,
                                               seed), 0.0)
 
+    # --- fp8 (OCP e4m3) allreduce across LL / oneshot / twoshot paths ------
+    if hasattr(torch, "float8_e4m3fn"):
+        for count in (1000, 300000, 5 << 20):
+            seed += 1
+            gens = [torch.Generator().manual_seed(4321 + 13 * r + seed)
+                    for r in range(world)]
+            vals = [torch.randn(count, generator=g) * 0.5 for g in gens]
+            x8 = vals[rank].to(torch.float8_e4m3fn).cuda()
+            comm.all_reduce(x8)
+            torch.cuda.synchronize()
+            ref = sum(v.to(torch.float8_e4m3fn).float() for v in vals)
+            ref8 = ref.to(torch.float8_e4m3fn).float()
+            got = x8.float().cpu()
+            # fp8 quantization of the fp32-accumulated sum: one-ulp slack
+            assert torch.allclose(got, ref8, rtol=0.15, atol=0.1), \
+                (got - ref8).abs().max()
+            print(f"[rank {rank}] fp8 allreduce[{count}] OK", flush=True)
+
     dist.barrier()
     if rank == 0:
         print("ALL COLLECTIVE TESTS PASSED", flush=True)

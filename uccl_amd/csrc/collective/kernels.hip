@@ -258,9 +258,8 @@ __global__ void k_ll_allreduce(CommView cv, void const* __restrict__ in,
   }
 
   for (size_t w = i; w < nw; w += stride) {
-    float accf[2] = {0.f, 0.f};
+    float accf[4] = {0.f, 0.f, 0.f, 0.f};
     int acci[1] = {0};
-    uint32_t raw0 = 0;
     for (int p = 0; p < cv.world; ++p) {
       uint32_t const d = ll_read(
           reinterpret_cast<uint64_t const*>(
@@ -272,14 +271,21 @@ __global__ void k_ll_allreduce(CommView cv, void const* __restrict__ in,
         } else {
           accf[0] += __uint_as_float(d);
         }
-      } else {  // 2-byte types: two lanes per word
+      } else if constexpr (sizeof(T) == 2) {  // two lanes per word
         T lo, hi;
         reinterpret_cast<uint16_t&>(lo) = d & 0xffff;
         reinterpret_cast<uint16_t&>(hi) = d >> 16;
         accf[0] += static_cast<float>(lo);
         accf[1] += static_cast<float>(hi);
+      } else {  // 1-byte types (fp8): four lanes per word
+#pragma unroll
+        for (int b = 0; b < 4; ++b) {
+          T e;
+          reinterpret_cast<uint8_t&>(e) =
+              static_cast<uint8_t>((d >> (8 * b)) & 0xff);
+          accf[b] += static_cast<float>(e);
+        }
       }
-      raw0 = d;
     }
     uint32_t r;
     if constexpr (sizeof(T) == 4) {
@@ -288,12 +294,19 @@ __global__ void k_ll_allreduce(CommView cv, void const* __restrict__ in,
       } else {
         r = __float_as_uint(accf[0]);
       }
-    } else {
+    } else if constexpr (sizeof(T) == 2) {
       T lo = static_cast<T>(accf[0]), hi = static_cast<T>(accf[1]);
       r = reinterpret_cast<uint16_t&>(lo) |
           (static_cast<uint32_t>(reinterpret_cast<uint16_t&>(hi)) << 16);
+    } else {
+      r = 0;
+#pragma unroll
+      for (int b = 0; b < 4; ++b) {
+        T e = static_cast<T>(accf[b]);
+        r |= static_cast<uint32_t>(reinterpret_cast<uint8_t&>(e))
+             << (8 * b);
+      }
     }
-    (void)raw0;
     if ((w + 1) * 4 <= bytes) {
       reinterpret_cast<uint32_t*>(out)[w] = r;
     } else {
@@ -451,6 +464,7 @@ void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s) {
     case Dtype::kBF16: fn<__hip_bfloat16>(__VA_ARGS__); break; \
     case Dtype::kI32: fn<int>(__VA_ARGS__); break;      \
     case Dtype::kU8: /* reducing collectives reject kU8 host-side */ break; \
+    case Dtype::kF8E4M3: fn<__hip_fp8_e4m3>(__VA_ARGS__); break; \
   }
 
 template <typename T>

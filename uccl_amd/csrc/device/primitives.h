@@ -13,6 +13,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
+#include <hip/hip_fp8.h>
 
 #include <cstdint>
 
@@ -214,6 +215,32 @@ struct AccumV16<__half, N> {
     V16 r;
 #pragma unroll
     for (int i = 0; i < 8; ++i) r.f16[i] = __float2half(v[i]);
+    return r;
+  }
+};
+
+// OCP fp8 e4m3 (gfx950-native, NOT the MI300X fnuz variant): 16 elems per
+// 16B vector, fp32 accumulation, saturating pack.
+template <int N>
+struct AccumV16<__hip_fp8_e4m3, N> {
+  float v[16];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 16; ++i)
+      v[i] = static_cast<float>(
+          reinterpret_cast<__hip_fp8_e4m3 const*>(&a)[i]);
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 16; ++i)
+      v[i] += static_cast<float>(
+          reinterpret_cast<__hip_fp8_e4m3 const*>(&a)[i]);
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 16; ++i)
+      reinterpret_cast<__hip_fp8_e4m3*>(&r)[i] = __hip_fp8_e4m3(v[i]);
     return r;
   }
 };
