@@ -69,7 +69,9 @@ def main():
 
     import uccl_amd.collective as ucol
 
-    comm = ucol.init(device=torch.cuda.current_device())
+    # small heap so the 192MB message below genuinely exercises chunking
+    comm = ucol.init(device=torch.cuda.current_device(),
+                     heap_bytes=192 * (1 << 20))
 
     tol = {torch.float32: 0.0, torch.int32: 0.0,
            torch.bfloat16: 1e-2, torch.float16: 1e-3}
