@@ -34,6 +34,8 @@ def test_two_process_collectives(world):
         "WORLD_SIZE": str(world),
         "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         "PYTHONPATH": str(REPO),
+        "UCCL_TEST_LIGHT": "1" if world > 2 else "0",
+        "UCCL_TEST_ALARM": "420",
     })
     procs = []
     for r in range(world):
@@ -46,7 +48,7 @@ def test_two_process_collectives(world):
     ok = True
     for p in procs:
         try:
-            out, _ = p.communicate(timeout=300)
+            out, _ = p.communicate(timeout=440)
         except subprocess.TimeoutExpired:
             p.kill()
             out, _ = p.communicate()

@@ -69,20 +69,26 @@ def main():
 
     import uccl_amd.collective as ucol
 
-    # small heap so the 192MB message below genuinely exercises chunking
+    # small heap so the large message below genuinely exercises chunking
+    light_heap = 64 if os.environ.get("UCCL_TEST_LIGHT", "0") == "1" else 192
     comm = ucol.init(device=torch.cuda.current_device(),
-                     heap_bytes=192 * (1 << 20))
+                     heap_bytes=light_heap * (1 << 20))
 
+    light = os.environ.get("UCCL_TEST_LIGHT", "0") == "1"
     tol = {torch.float32: 0.0, torch.int32: 0.0,
            torch.bfloat16: 1e-2, torch.float16: 1e-3}
 
     seed = 0
     # --- allreduce across the three algorithm paths -------------------------
     # LL (<=32KB), one-shot (<=2MB), two-shot (>2MB), chunked two-shot
-    for count, label in [(1000, "ll"), (100000, "oneshot"),
-                         (3 << 20, "twoshot"), (1000003, "oneshot-odd")]:
-        for dtype in (torch.float32, torch.bfloat16, torch.float16,
-                      torch.int32):
+    cases = [(1000, "ll"), (100000, "oneshot"), (3 << 20, "twoshot"),
+             (1000003, "oneshot-odd")]
+    dtypes = (torch.float32, torch.bfloat16, torch.float16, torch.int32)
+    if light:
+        cases = [(1000, "ll"), (100000, "oneshot"), (3 << 20, "twoshot")]
+        dtypes = (torch.float32, torch.bfloat16)
+    for count, label in cases:
+        for dtype in dtypes:
             seed += 1
             x = make_input(rank, count, dtype, seed)
             comm.all_reduce(x)
@@ -95,7 +101,7 @@ def main():
     # large enough to need >1 chunk only if scratch < size. Default scratch
     # is tens of MB, so force chunking with a big-ish message.
     seed += 1
-    count = 48 << 20  # 192MB fp32 > default per-parity scratch
+    count = (8 if light else 48) << 20  # > per-parity scratch -> chunked
     x = make_input(rank, count, torch.float32, seed)
     comm.all_reduce(x)
     torch.cuda.synchronize()
@@ -180,7 +186,7 @@ def main():
 
     # --- barrier + interleave stress ---------------------------------------
     comm.barrier()
-    for i in range(20):  # rapid-fire small collectives: parity/seq stress
+    for i in range(5 if light else 20):  # rapid-fire parity/seq stress
         seed += 1
         x = make_input(rank, 257, torch.float32, seed)
         comm.all_reduce(x)
@@ -190,7 +196,7 @@ def main():
 
     # --- fp8 (OCP e4m3) allreduce across LL / oneshot / twoshot paths ------
     if hasattr(torch, "float8_e4m3fn"):
-        for count in (1000, 300000, 5 << 20):
+        for count in ((1000, 300000) if light else (1000, 300000, 5 << 20)):
             seed += 1
             gens = [torch.Generator().manual_seed(4321 + 13 * r + seed)
                     for r in range(world)]
