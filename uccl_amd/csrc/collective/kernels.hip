@@ -102,10 +102,10 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
     AccumV16<T, 0> acc;
-    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[cv.rank], cv))[i]);
+    // fixed rank order => bitwise-identical results on every rank
+    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv))[i]);
 #pragma unroll 7
-    for (int k = 1; k < cv.world; ++k) {
-      int const p = (cv.rank + k) % cv.world;
+    for (int p = 1; p < cv.world; ++p) {
       acc.add(reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv))[i]);
     }
     reinterpret_cast<V16*>(out)[i] = acc.pack();
@@ -133,9 +133,8 @@ __global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
     AccumV16<int, 0> acc;
-    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[cv.rank], cv))[i]);
-    for (int k = 1; k < cv.world; ++k) {
-      int const p = (cv.rank + k) % cv.world;
+    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv))[i]);
+    for (int p = 1; p < cv.world; ++p) {
       acc.add(reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv))[i]);
     }
     reinterpret_cast<V16*>(out)[i] = acc.pack();
@@ -172,10 +171,10 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < end; i += stride) {
     AccumV16<T, 0> acc;
-    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[cv.rank], cv))[i]);
+    // fixed rank order => bitwise-identical results on every rank
+    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv))[i]);
 #pragma unroll 7
-    for (int k = 1; k < cv.world; ++k) {
-      int const p = (cv.rank + k) % cv.world;
+    for (int p = 1; p < cv.world; ++p) {
       acc.add(reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv))[i]);
     }
     V16 const r = acc.pack();
@@ -353,10 +352,9 @@ __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
   for (; i < nvec; i += stride) {
     AccumV16<T, 0> acc;
     acc.init(reinterpret_cast<V16 const*>(
-        reinterpret_cast<T const*>(scratch_a(cv.peers[cv.rank], cv)) +
+        reinterpret_cast<T const*>(scratch_a(cv.peers[0], cv)) +
         elem_off)[i]);
-    for (int k = 1; k < cv.world; ++k) {
-      int const p = (cv.rank + k) % cv.world;
+    for (int p = 1; p < cv.world; ++p) {
       acc.add(reinterpret_cast<V16 const*>(
           reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv)) + elem_off)[i]);
     }
