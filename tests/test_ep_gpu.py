@@ -30,12 +30,13 @@ def test_ep_world1():
     assert "EP ALL OK" in out
 
 
-def test_ep_world2():
+def _run_world2(extra_env, port):
     env_base = dict(os.environ)
     env_base.update({
-        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29473",
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
         "WORLD_SIZE": "2", "PYTHONPATH": str(REPO),
     })
+    env_base.update(extra_env)
     ps = []
     for r in range(2):
         env = dict(env_base)
@@ -56,3 +57,14 @@ def test_ep_world2():
     joined = "\n=====\n".join(outs)
     assert ok, joined
     assert joined.count("EP ALL OK") == 2, joined
+
+
+def test_ep_world2():
+    _run_world2({}, 29473)
+
+
+def test_ep_world2_forced_proxy():
+    """Internode-emulation: no IPC mapping between the two ranks; all EP
+    traffic flows GPU -> D2H command ring -> CPU proxy -> multipath
+    reliable transport -> peer proxy -> peer GPU memory."""
+    _run_world2({"UCCL_EP_FORCE_PROXY": "1"}, 29477)

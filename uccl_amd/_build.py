@@ -28,6 +28,7 @@ SOURCES = [
     CSRC / "transport" / "reliable.cpp",
     CSRC / "ep" / "ep_kernels.hip",
     CSRC / "ep" / "ep_buffer.cpp",
+    CSRC / "ep" / "ep_proxy.cpp",
     CSRC / "bindings" / "module.cpp",
 ]
 

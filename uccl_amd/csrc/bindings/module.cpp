@@ -318,10 +318,17 @@ PYBIND11_MODULE(_C, m) {
       .def(py::init<int, size_t>(), py::arg("num_paths") = 8,
            py::arg("chunk_bytes") = 8192)
       .def("metadata", &TransportEndpoint::metadata)
-      .def("connect", &TransportEndpoint::connect,
-           py::call_guard<py::gil_scoped_release>())
-      .def("accept", &TransportEndpoint::accept,
-           py::call_guard<py::gil_scoped_release>())
+      .def("connect",
+           [](TransportEndpoint& e, const std::string& md, uint64_t tag) {
+             py::gil_scoped_release rel;
+             return e.connect(md, tag);
+           },
+           py::arg("metadata"), py::arg("tag") = 0)
+      .def("accept",
+           [](TransportEndpoint& e) {
+             py::gil_scoped_release rel;
+             return e.accept(nullptr);
+           })
       .def("send",
            [](TransportEndpoint& e, uint64_t flow, at::Tensor t) {
              TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),

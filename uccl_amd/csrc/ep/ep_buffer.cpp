@@ -11,7 +11,9 @@
 
 #include "../core/env.h"
 #include "../core/log.h"
+#include "d2h_ring.h"
 #include "ep_kernels.h"
+#include "ep_proxy.h"
 
 namespace uccl {
 namespace ep {
@@ -54,7 +56,20 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
                            elem_size);
   v_.off_plan = off;
   off = align256(off + sizeof(uint32_t) * num_experts *
-                           (1 + static_cast<size_t>(max_tokens)));
+                           (2 + static_cast<size_t>(max_tokens)));
+  size_t const egress_rows = static_cast<size_t>(max_tokens) * topk;
+  v_.off_egress = off;
+  off = align256(off + egress_rows * hidden * elem_size);
+  v_.off_egress_meta = off;
+  off = align256(off + egress_rows * sizeof(uint32_t));
+  // ingress: per-source-peer staging slices of max_tokens rows (each
+  // proxy rx message covers one (expert, src) pair <= max_tokens rows)
+  size_t const ingress_rows =
+      static_cast<size_t>(world) * max_tokens;
+  v_.off_ingress = off;
+  off = align256(off + ingress_rows * hidden * elem_size);
+  v_.off_ingress_meta = off;
+  off = align256(off + ingress_rows * sizeof(uint32_t));
   v_.heap_bytes = off;
 
   UCCL_CHECK_HIP(hipSetDevice(device_));
@@ -63,15 +78,31 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   UCCL_CHECK_HIP(hipDeviceSynchronize());
   for (int r = 0; r < kMaxRanks; ++r) v_.peers[r] = nullptr;
   v_.peers[rank_] = heap_;
+  v_.proxy_mask = 0;
+  v_.ring = nullptr;
+  if (env_bool("UCCL_EP_FORCE_PROXY", false) && world_ > 1) {
+    UCCL_CHECK_HIP(hipHostMalloc(&ring_host_, sizeof(D2HRing),
+                                 hipHostMallocMapped));
+    memset(ring_host_, 0, sizeof(D2HRing));
+    void* dev = nullptr;
+    UCCL_CHECK_HIP(hipHostGetDevicePointer(&dev, ring_host_, 0));
+    v_.ring = static_cast<D2HRing*>(dev);
+  }
+  UCCL_CHECK_HIP(hipHostMalloc(
+      reinterpret_cast<void**>(&host_counts_),
+      sizeof(int) * v_.local_experts * world_));
   UCCL_LOG_INFO << "EpBuffer rank " << rank << "/" << world << " experts="
                 << num_experts << " hidden=" << hidden << " max_tokens="
                 << max_tokens << " heap=" << (v_.heap_bytes >> 20) << "MB";
 }
 
 EpBuffer::~EpBuffer() {
+  proxy_.reset();  // joins proxy threads before the heap goes away
   for (int r = 0; r < world_; ++r)
     if (ipc_opened_[r] && v_.peers[r]) (void)hipIpcCloseMemHandle(v_.peers[r]);
   if (heap_) (void)hipFree(heap_);
+  if (ring_host_) (void)hipHostFree(ring_host_);
+  if (host_counts_) (void)hipHostFree(host_counts_);
 }
 
 std::string EpBuffer::handle_bytes() const {
@@ -83,12 +114,24 @@ std::string EpBuffer::handle_bytes() const {
   UCCL_CHECK_HIP(hipIpcGetMemHandle(&b.h, heap_));
   b.device = device_;
   b.pid = static_cast<int>(getpid());
-  return std::string(reinterpret_cast<char*>(&b), sizeof(b));
+  std::string out(reinterpret_cast<char*>(&b), sizeof(b));
+  if (v_.ring) {
+    // proxy mode: append the transport rendezvous metadata (the proxy is
+    // created lazily here so its endpoint exists before exchange)
+    auto* self = const_cast<EpBuffer*>(this);
+    if (!self->proxy_)
+      self->proxy_ = std::make_unique<EpProxy>(v_, heap_, ring_host_,
+                                               device_);
+    out += proxy_->transport_metadata();
+  }
+  return out;
 }
 
 void EpBuffer::connect(const std::vector<std::string>& handles) {
   UCCL_CHECK(static_cast<int>(handles.size()) == world_);
   UCCL_CHECK_HIP(hipSetDevice(device_));
+  bool const force_proxy = v_.ring != nullptr;
+  std::vector<std::string> tp_md(world_);
   for (int r = 0; r < world_; ++r) {
     if (r == rank_) continue;
     struct Blob {
@@ -96,8 +139,14 @@ void EpBuffer::connect(const std::vector<std::string>& handles) {
       int device;
       int pid;
     } b{};
-    UCCL_CHECK(handles[r].size() == sizeof(b)) << "bad ep handle blob";
+    UCCL_CHECK(handles[r].size() >= sizeof(b)) << "bad ep handle blob";
     memcpy(&b, handles[r].data(), sizeof(b));
+    tp_md[r] = handles[r].substr(sizeof(b));
+    if (force_proxy) {
+      // emulate internode: no IPC mapping, all traffic via proxy
+      v_.proxy_mask |= 1u << r;
+      continue;
+    }
     if (b.device != device_) {
       hipError_t e = hipDeviceEnablePeerAccess(b.device, 0);
       UCCL_CHECK(e == hipSuccess || e == hipErrorPeerAccessAlreadyEnabled)
@@ -108,6 +157,12 @@ void EpBuffer::connect(const std::vector<std::string>& handles) {
                                        hipIpcMemLazyEnablePeerAccess));
     v_.peers[r] = p;
     ipc_opened_[r] = true;
+  }
+  if (force_proxy) {
+    UCCL_CHECK(proxy_) << "handle_bytes() must run before connect()";
+    proxy_->set_view(v_);
+    proxy_->establish_flows(tp_md, v_.proxy_mask);
+    proxy_->start();
   }
   connected_ = true;
 }
@@ -120,6 +175,10 @@ void EpBuffer::dispatch(void const* x, int64_t const* topk_idx,
       << num_tokens << " tokens > max_tokens " << v_.max_tokens;
   ++v_.seq;
   launch_ep_dispatch(v_, x, topk_idx, num_tokens, out_counts, stream);
+  // pinned host copy of counts (combine proxy shipping needs them)
+  UCCL_CHECK_HIP(hipMemcpyAsync(host_counts_, out_counts,
+                                sizeof(int) * v_.local_experts * world_,
+                                hipMemcpyDeviceToHost, stream));
   last_num_tokens_ = num_tokens;
 }
 
@@ -127,8 +186,17 @@ void EpBuffer::combine(void const* expert_out, void* out,
                        int64_t const* topk_idx, float const* topk_w,
                        hipStream_t stream) {
   UCCL_CHECK(last_num_tokens_ >= 0) << "combine without a prior dispatch";
-  launch_ep_combine(v_, expert_out, out, topk_idx, topk_w, last_num_tokens_,
-                    stream);
+  launch_ep_combine_send(v_, expert_out, stream);
+  if (v_.proxy_mask) {
+    // proxy peers: ship expert_out rows host-side once the stream reaches
+    // this point (expert_out fully produced + host_counts_ landed)
+    hipEvent_t ev = nullptr;
+    UCCL_CHECK_HIP(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    UCCL_CHECK_HIP(hipEventRecord(ev, stream));
+    proxy_->enqueue_combine(expert_out, v_.seq, ev, host_counts_);
+  }
+  launch_ep_combine_finish(v_, out, topk_idx, topk_w, last_num_tokens_,
+                           stream);
 }
 
 }  // namespace ep

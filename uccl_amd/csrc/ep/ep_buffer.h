@@ -6,10 +6,15 @@
 #include <string>
 #include <vector>
 
+#include <memory>
+
 #include "ep_layout.h"
 
 namespace uccl {
 namespace ep {
+
+class EpProxy;
+struct D2HRing;
 
 class EpBuffer {
  public:
@@ -42,6 +47,8 @@ class EpBuffer {
   int world() const { return world_; }
   int device() const { return device_; }
 
+  bool proxy_enabled() const { return proxy_ != nullptr; }
+
  private:
   int rank_, world_, device_;
   void* heap_ = nullptr;
@@ -49,6 +56,12 @@ class EpBuffer {
   std::array<bool, kMaxRanks> ipc_opened_{};
   bool connected_ = false;
   int last_num_tokens_ = -1;
+
+  // proxy path (internode / forced): D2H command ring + CPU proxy over
+  // the reliable transport
+  std::unique_ptr<EpProxy> proxy_;
+  D2HRing* ring_host_ = nullptr;
+  int* host_counts_ = nullptr;  // pinned copy of dispatch counts
 };
 
 }  // namespace ep

@@ -10,9 +10,13 @@ namespace ep {
 void launch_ep_dispatch(const EpView& v, void const* x,
                         int64_t const* topk_idx, int num_tokens,
                         int* out_counts, hipStream_t s);
-void launch_ep_combine(const EpView& v, void const* expert_out, void* out,
-                       int64_t const* topk_idx, float const* topk_w,
-                       int num_tokens, hipStream_t s);
+void launch_ep_combine_send(const EpView& v, void const* expert_out,
+                            hipStream_t s);
+void launch_ep_combine_finish(const EpView& v, void* out,
+                              int64_t const* topk_idx, float const* topk_w,
+                              int num_tokens, hipStream_t s);
+void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
+                            hipStream_t s);
 
 }  // namespace ep
 }  // namespace uccl

@@ -15,7 +15,10 @@
 
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
+
 #include "../device/primitives.h"
+#include "d2h_ring.h"
 #include "ep_kernels.h"
 
 namespace uccl {
@@ -83,8 +86,20 @@ __global__ void k_ep_dispatch_plan(EpView v,
   for (int t = t0; t < t1; ++t)
     for (int k = 0; k < v.topk; ++k)
       if (topk_idx[static_cast<size_t>(t) * v.topk + k] == e)
-        plan[1 + pos++] = static_cast<uint32_t>(t) |
+        plan[2 + pos++] = static_cast<uint32_t>(t) |
                           (static_cast<uint32_t>(k) << 24);
+}
+
+// single block: exclusive prefix of per-expert counts -> plan[e][1]
+// (egress row offsets for the proxy path)
+__global__ void k_ep_plan_prefix(EpView v) {
+  if (threadIdx.x != 0) return;
+  uint32_t acc = 0;
+  for (int e = 0; e < v.num_experts; ++e) {
+    uint32_t* plan = plan_ptr(v.peers[v.rank], v, e);
+    plan[1] = acc;
+    acc += plan[0];
+  }
 }
 
 __global__ void k_ep_dispatch_copy(EpView v, void const* __restrict__ x) {
@@ -92,30 +107,54 @@ __global__ void k_ep_dispatch_copy(EpView v, void const* __restrict__ x) {
   int const b = blockIdx.x % kDispatchFanout;
   int const dst = e / v.local_experts;
   int const le = e % v.local_experts;
+  bool const via_proxy = (v.proxy_mask >> dst) & 1u;
+  void* me = v.peers[v.rank];
   void* dbase = v.peers[dst];
-  uint32_t const* plan = plan_ptr(v.peers[v.rank], v, e);
+  uint32_t const* plan = plan_ptr(me, v, e);
   uint32_t const count = plan[0];
+  uint32_t const pfx = plan[1];
   size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
   for (uint32_t i = b; i < count; i += kDispatchFanout) {
-    uint32_t const tk = plan[1 + i];
+    uint32_t const tk = plan[2 + i];
     uint32_t const t = tk & kMetaTokMask;
-    size_t const slot = static_cast<size_t>(v.rank) * v.max_tokens + i;
-    block_copy(disp_x_ptr(dbase, v, le, slot),
-               static_cast<char const*>(x) +
-                   static_cast<size_t>(t) * row_bytes,
-               row_bytes);
-    if (threadIdx.x == 0) *disp_meta_ptr(dbase, v, le, slot) = tk;
+    char const* src_row =
+        static_cast<char const*>(x) + static_cast<size_t>(t) * row_bytes;
+    if (via_proxy) {
+      // stage into packed local egress; the CPU proxy ships it
+      block_copy(egress_row(me, v, pfx + i), src_row, row_bytes);
+      if (threadIdx.x == 0) *egress_meta(me, v, pfx + i) = tk;
+    } else {
+      size_t const slot = static_cast<size_t>(v.rank) * v.max_tokens + i;
+      block_copy(disp_x_ptr(dbase, v, le, slot), src_row, row_bytes);
+      if (threadIdx.x == 0) *disp_meta_ptr(dbase, v, le, slot) = tk;
+    }
   }
 }
 
-// publish counts after the copy kernel's dispatch-boundary flush
+// publish counts after the copy kernel's dispatch-boundary flush.
+// Direct peers get the seq-tagged count release-store; proxy peers get a
+// 32-byte TransferCmd pushed into the D2H ring (GPU-initiated transfer,
+// executed by the CPU proxy — the IBGDA-replacement design).
 __global__ void k_ep_dispatch_publish(EpView v) {
   for (int e = threadIdx.x; e < v.num_experts; e += blockDim.x) {
     int const dst = e / v.local_experts;
+    if ((v.proxy_mask >> dst) & 1u) continue;
     int const le = e % v.local_experts;
     uint32_t const count = plan_ptr(v.peers[v.rank], v, e)[0];
     st_release_sys(disp_count_ptr(v.peers[dst], v, le, v.rank),
                    tag_count(v.seq, count));
+  }
+  __syncthreads();
+  if (threadIdx.x == 0 && v.proxy_mask) {
+    for (int e = 0; e < v.num_experts; ++e) {
+      int const dst = e / v.local_experts;
+      if (!((v.proxy_mask >> dst) & 1u)) continue;
+      uint32_t const* plan = plan_ptr(v.peers[v.rank], v, e);
+      TransferCmd c{static_cast<uint32_t>(CmdOp::kDispatchWrite),
+                    static_cast<uint32_t>(v.seq), static_cast<uint64_t>(e),
+                    plan[1], plan[0]};
+      ring_push(v.ring, c);
+    }
   }
 }
 
@@ -151,6 +190,7 @@ __global__ void k_ep_combine_send(EpView v,
   int const b = blockIdx.x % kDispatchFanout;
   int const le = pair / v.world;
   int const src = pair % v.world;
+  if ((v.proxy_mask >> src) & 1u) return;  // proxy ships these host-side
   void* me = v.peers[v.rank];
   uint64_t const tagged = *disp_count_ptr(me, v, le, src);
   // count published at dispatch time with this seq (combine reuses it)
@@ -174,7 +214,8 @@ __global__ void k_ep_combine_send(EpView v,
 // one block: publish "my returns to you are complete" to every rank
 // (launched after k_ep_combine_send; dispatch boundary flushed the writes)
 __global__ void k_ep_combine_signal(EpView v) {
-  if (threadIdx.x < static_cast<unsigned>(v.world))
+  if (threadIdx.x < static_cast<unsigned>(v.world) &&
+      !((v.proxy_mask >> threadIdx.x) & 1u))
     st_release_sys(comb_flag_ptr(v.peers[threadIdx.x], v, v.rank), v.seq);
 }
 
@@ -186,6 +227,20 @@ __global__ void k_ep_combine_wait(EpView v) {
       if (it > (1ull << 28)) __builtin_trap();
       backoff();
     }
+  }
+}
+
+// proxy-ingress scatter: rows staged at ingress[row0..row0+count) are
+// moved into their (token, k) comb_x cells (proxy RX path).
+__global__ void k_ep_comb_scatter(EpView v, size_t row0, size_t count) {
+  void* me = v.peers[v.rank];
+  size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
+  for (size_t i = blockIdx.x; i < count; i += gridDim.x) {
+    uint32_t const meta = *ingress_meta(me, v, row0 + i);
+    uint32_t const t = meta & kMetaTokMask;
+    uint32_t const k = meta >> 24;
+    block_copy(comb_x_ptr(me, v, t, k), ingress_row(me, v, row0 + i),
+               row_bytes);
   }
 }
 
@@ -247,17 +302,28 @@ void launch_ep_dispatch(const EpView& v, void const* x,
   size_t const smem = 257 * sizeof(uint32_t);
   k_ep_dispatch_plan<<<v.num_experts, 256, smem, s>>>(v, topk_idx,
                                                       num_tokens);
+  k_ep_plan_prefix<<<1, 64, 0, s>>>(v);
   k_ep_dispatch_copy<<<v.num_experts * kDispatchFanout, 256, 0, s>>>(v, x);
   k_ep_dispatch_publish<<<1, 256, 0, s>>>(v);
   k_ep_dispatch_wait<<<1, 256, 0, s>>>(v, out_counts);
 }
 
-void launch_ep_combine(const EpView& v, void const* expert_out, void* out,
-                       int64_t const* topk_idx, float const* topk_w,
-                       int num_tokens, hipStream_t s) {
+void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
+                            hipStream_t s) {
+  int const grid = count ? static_cast<int>(std::min<size_t>(count, 512)) : 1;
+  k_ep_comb_scatter<<<grid, 256, 0, s>>>(v, row0, count);
+}
+
+void launch_ep_combine_send(const EpView& v, void const* expert_out,
+                            hipStream_t s) {
   k_ep_combine_send<<<v.local_experts * v.world * kDispatchFanout, 256, 0,
                       s>>>(v, expert_out);
   k_ep_combine_signal<<<1, 64, 0, s>>>(v);
+}
+
+void launch_ep_combine_finish(const EpView& v, void* out,
+                              int64_t const* topk_idx, float const* topk_w,
+                              int num_tokens, hipStream_t s) {
   k_ep_combine_wait<<<1, 64, 0, s>>>(v);
   if (v.elem_size == 2) {
     k_ep_combine_reduce<__hip_bfloat16>

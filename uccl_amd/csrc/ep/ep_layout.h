@@ -19,9 +19,13 @@ namespace uccl {
 namespace ep {
 
 // Runtime-sized layout; offsets are computed host-side once.
+struct D2HRing;
+
 struct EpView {
   int rank;
   int world;
+  uint32_t proxy_mask;   // bit r set => traffic to rank r goes via proxy
+  D2HRing* ring;         // device-visible pointer (proxy mode), else null
   int num_experts;     // global
   int local_experts;   // num_experts / world
   int topk;
@@ -38,8 +42,13 @@ struct EpView {
                           //   meta = src_token_idx | (k << 24)
   size_t off_comb_flag;   // u64 [world]                   (seq)
   size_t off_comb_x;      // [max_tokens][topk][hidden] elems
-  size_t off_plan;        // u32 [num_experts][1 + max_tokens]  (private
-                          //   per-rank scratch: count then (t|k<<24) list)
+  size_t off_plan;        // u32 [num_experts][2 + max_tokens]  (private
+                          //   per-rank scratch: count, egress prefix, list)
+  size_t off_egress;      // packed egress rows [max_tokens*topk][hidden]
+                          //   (proxy mode: rows destined to remote ranks)
+  size_t off_egress_meta; // u32 [max_tokens*topk]
+  size_t off_ingress;     // proxy RX staging rows (combine returns)
+  size_t off_ingress_meta;
   size_t heap_bytes;
 };
 
@@ -72,7 +81,34 @@ __host__ __device__ inline uint32_t* disp_meta_ptr(void* base,
 __host__ __device__ inline uint32_t* plan_ptr(void* base, const EpView& v,
                                               int e) {
   return reinterpret_cast<uint32_t*>(static_cast<char*>(base) + v.off_plan) +
-         static_cast<size_t>(e) * (1 + v.max_tokens);
+         static_cast<size_t>(e) * (2 + v.max_tokens);
+}
+
+__host__ __device__ inline char* egress_row(void* base, const EpView& v,
+                                            size_t row) {
+  return static_cast<char*>(base) + v.off_egress +
+         row * static_cast<size_t>(v.hidden) * v.elem_size;
+}
+
+__host__ __device__ inline uint32_t* egress_meta(void* base, const EpView& v,
+                                                 size_t row) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
+                                     v.off_egress_meta) +
+         row;
+}
+
+__host__ __device__ inline char* ingress_row(void* base, const EpView& v,
+                                             size_t row) {
+  return static_cast<char*>(base) + v.off_ingress +
+         row * static_cast<size_t>(v.hidden) * v.elem_size;
+}
+
+__host__ __device__ inline uint32_t* ingress_meta(void* base,
+                                                  const EpView& v,
+                                                  size_t row) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
+                                     v.off_ingress_meta) +
+         row;
 }
 
 __host__ __device__ inline uint64_t* comb_flag_ptr(void* base,

@@ -1,0 +1,268 @@
+#include "ep_proxy.h"
+
+#include <unistd.h>
+
+#include <cstring>
+
+#include "../core/env.h"
+#include "../core/log.h"
+
+namespace uccl {
+namespace ep {
+
+namespace {
+constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3;
+constexpr size_t kStageBytes = 16ull << 20;  // pinned staging chunk
+}  // namespace
+
+EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
+                 int device)
+    : v_(view), heap_(heap), ring_(ring_host), device_(device) {
+  tp_ = std::make_unique<transport::TransportEndpoint>(
+      static_cast<int>(env_int("UCCL_EP_PROXY_PATHS", 8)),
+      static_cast<size_t>(env_int("UCCL_EP_PROXY_CHUNK", 16384)));
+  flows_.resize(v_.world, 0);
+  UCCL_CHECK_HIP(hipSetDevice(device_));
+  UCCL_CHECK_HIP(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+  UCCL_CHECK_HIP(
+      hipStreamCreateWithFlags(&rx_stream_, hipStreamNonBlocking));
+  stage_bytes_ = kStageBytes;
+  UCCL_CHECK_HIP(hipHostMalloc(&host_buf_, stage_bytes_));
+  UCCL_CHECK_HIP(hipHostMalloc(&rx_buf_, stage_bytes_));
+}
+
+EpProxy::~EpProxy() {
+  stop_ = true;
+  cv_.notify_all();
+  if (tp_) tp_->shutdown();
+  if (ring_thread_.joinable()) ring_thread_.join();
+  if (comb_thread_.joinable()) comb_thread_.join();
+  for (auto& t : rx_threads_)
+    if (t.joinable()) t.join();
+  tp_.reset();
+  if (host_buf_) (void)hipHostFree(host_buf_);
+  if (rx_buf_) (void)hipHostFree(rx_buf_);
+  if (stream_) (void)hipStreamDestroy(stream_);
+  if (rx_stream_) (void)hipStreamDestroy(rx_stream_);
+}
+
+void EpProxy::establish_flows(const std::vector<std::string>& tp_md,
+                              uint32_t proxy_mask) {
+  proxy_mask_ = proxy_mask;
+  // higher rank connects to lower rank; lower rank accepts and matches by
+  // the connector's tag (= its rank)
+  int expected_accepts = 0;
+  for (int r = 0; r < v_.world; ++r) {
+    if (!((proxy_mask_ >> r) & 1u)) continue;
+    if (r < v_.rank) {
+      flows_[r] = tp_->connect(tp_md[r], static_cast<uint64_t>(v_.rank));
+    } else {
+      ++expected_accepts;
+    }
+  }
+  for (int i = 0; i < expected_accepts; ++i) {
+    uint64_t tag = 0;
+    uint64_t flow = tp_->accept(&tag);
+    UCCL_CHECK(tag < static_cast<uint64_t>(v_.world)) << "bad flow tag";
+    flows_[tag] = flow;
+  }
+}
+
+void EpProxy::start() {
+  ring_thread_ = std::thread([this] { ring_loop(); });
+  comb_thread_ = std::thread([this] { comb_tx_loop(); });
+  for (int r = 0; r < v_.world; ++r)
+    if ((proxy_mask_ >> r) & 1u)
+      rx_threads_.emplace_back([this, r] { rx_loop(r); });
+}
+
+// D2H-copy `count` rows from dev_rows and stream them over `flow` in
+// staging-sized chunks, then the metas.
+void EpProxy::ship_rows(uint64_t flow, WireHdr const& h,
+                        void const* dev_rows,
+                        uint32_t const* dev_metas_or_null,
+                        std::vector<uint32_t> const* host_metas) {
+  size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
+  size_t const total = h.count * row_bytes;
+  tp_->send_msg(flow, &h, sizeof(h));
+  for (size_t off = 0; off < total; off += stage_bytes_) {
+    size_t const n = std::min(stage_bytes_, total - off);
+    UCCL_CHECK_HIP(hipMemcpyAsync(host_buf_,
+                                  static_cast<char const*>(dev_rows) + off,
+                                  n, hipMemcpyDeviceToHost, stream_));
+    UCCL_CHECK_HIP(hipStreamSynchronize(stream_));
+    tp_->send_msg(flow, host_buf_, n);
+  }
+  std::vector<uint32_t> metas;
+  if (host_metas) {
+    metas = *host_metas;
+  } else {
+    metas.resize(h.count);
+    UCCL_CHECK_HIP(hipMemcpyAsync(metas.data(), dev_metas_or_null,
+                                  h.count * sizeof(uint32_t),
+                                  hipMemcpyDeviceToHost, stream_));
+    UCCL_CHECK_HIP(hipStreamSynchronize(stream_));
+  }
+  if (h.count) tp_->send_msg(flow, metas.data(), h.count * sizeof(uint32_t));
+}
+
+void EpProxy::ring_loop() {
+  (void)hipSetDevice(device_);
+  uint64_t head = 0;
+  while (!stop_) {
+    uint64_t tail =
+        __atomic_load_n(const_cast<uint64_t*>(&ring_->tail),
+                        __ATOMIC_ACQUIRE);
+    if (head == tail) {
+      usleep(20);
+      continue;
+    }
+    while (head != tail) {
+      TransferCmd c;
+      memcpy(&c, const_cast<TransferCmd*>(&ring_->cmds[head % kRingSlots]),
+             sizeof(c));
+      ++head;
+      __atomic_store_n(const_cast<uint64_t*>(&ring_->head), head,
+                       __ATOMIC_RELEASE);
+      if (c.op != static_cast<uint32_t>(CmdOp::kDispatchWrite)) continue;
+      int const e = static_cast<int>(c.a);
+      int const dst = e / v_.local_experts;
+      WireHdr h{kDisp, c.seq32, static_cast<uint32_t>(e % v_.local_experts),
+                static_cast<uint32_t>(v_.rank), c.c};
+      try {
+        ship_rows(flows_[dst], h, egress_row(heap_, v_, c.b),
+                  egress_meta(heap_, v_, c.b), nullptr);
+      } catch (std::exception const&) {
+        return;  // transport closed
+      }
+    }
+  }
+}
+
+void EpProxy::comb_tx_loop() {
+  (void)hipSetDevice(device_);
+  while (true) {
+    CombTask task;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      cv_.wait(lk, [this] { return !comb_q_.empty() || stop_; });
+      if (stop_ && comb_q_.empty()) return;
+      task = std::move(comb_q_.front());
+      comb_q_.pop_front();
+    }
+    (void)hipEventSynchronize(task.ready);
+    size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
+    try {
+      for (int src = 0; src < v_.world; ++src) {
+        if (!((proxy_mask_ >> src) & 1u)) continue;
+        for (int le = 0; le < v_.local_experts; ++le) {
+          uint64_t const count =
+              static_cast<uint64_t>(task.counts[le * v_.world + src]);
+          WireHdr h{kComb, static_cast<uint32_t>(task.seq),
+                    static_cast<uint32_t>(le),
+                    static_cast<uint32_t>(v_.rank), count};
+          size_t const slot0 =
+              static_cast<size_t>(src) * v_.max_tokens;
+          char const* rows =
+              static_cast<char const*>(task.expert_out) +
+              ((static_cast<size_t>(le) * v_.world * v_.max_tokens +
+                slot0)) *
+                  row_bytes;
+          ship_rows(flows_[src], h, rows,
+                    disp_meta_ptr(heap_, v_, le, slot0), nullptr);
+        }
+        WireHdr done{kCombDone, static_cast<uint32_t>(task.seq), 0,
+                     static_cast<uint32_t>(v_.rank), 0};
+        tp_->send_msg(flows_[src], &done, sizeof(done));
+      }
+    } catch (std::exception const&) {
+      return;
+    }
+    (void)hipEventDestroy(task.ready);
+  }
+}
+
+void EpProxy::enqueue_combine(void const* expert_out, uint64_t seq,
+                              hipEvent_t ready, int const* counts) {
+  CombTask t;
+  t.expert_out = expert_out;
+  t.seq = seq;
+  t.ready = ready;
+  t.counts.assign(counts, counts + v_.local_experts * v_.world);
+  {
+    std::lock_guard<std::mutex> g(mu_);
+    comb_q_.push_back(std::move(t));
+  }
+  cv_.notify_one();
+}
+
+// scatter kernel for combine ingress (defined in ep_kernels.hip)
+void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
+                            hipStream_t s);
+
+void EpProxy::rx_loop(int peer) {
+  (void)hipSetDevice(device_);
+  uint64_t const flow = flows_[peer];
+  size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
+  std::vector<uint32_t> metas;
+  try {
+    while (!stop_) {
+      WireHdr h{};
+      tp_->recv_msg(flow, &h, sizeof(h));
+      if (h.kind == kCombDone) {
+        // all combine payloads from `peer` for this seq have been
+        // scattered (rx_stream_ is in-order); publish the flag
+        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+        uint64_t const seq = h.seq32;
+        UCCL_CHECK_HIP(hipMemcpy(comb_flag_ptr(heap_, v_, peer), &seq,
+                                 sizeof(seq), hipMemcpyHostToDevice));
+        continue;
+      }
+      size_t const total = h.count * row_bytes;
+      // destination in device memory
+      size_t const ing0 = static_cast<size_t>(peer) * v_.max_tokens;
+      char* dev_dst = (h.kind == kDisp)
+                          ? disp_x_ptr(heap_, v_, h.le,
+                                       static_cast<size_t>(h.src) *
+                                           v_.max_tokens)
+                          : ingress_row(heap_, v_, ing0);
+      for (size_t off = 0; off < total; off += stage_bytes_) {
+        size_t const n = std::min(stage_bytes_, total - off);
+        tp_->recv_msg(flow, rx_buf_, n);
+        UCCL_CHECK_HIP(hipMemcpyAsync(dev_dst + off, rx_buf_, n,
+                                      hipMemcpyHostToDevice, rx_stream_));
+        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+      }
+      metas.resize(h.count);
+      if (h.count)
+        tp_->recv_msg(flow, metas.data(), h.count * sizeof(uint32_t));
+      if (h.kind == kDisp) {
+        UCCL_CHECK_HIP(hipMemcpyAsync(
+            disp_meta_ptr(heap_, v_, h.le,
+                          static_cast<size_t>(h.src) * v_.max_tokens),
+            metas.data(), h.count * sizeof(uint32_t),
+            hipMemcpyHostToDevice, rx_stream_));
+        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+        uint64_t const tagged =
+            (static_cast<uint64_t>(h.seq32) << 32) | h.count;
+        UCCL_CHECK_HIP(
+            hipMemcpy(disp_count_ptr(heap_, v_, h.le, h.src), &tagged,
+                      sizeof(tagged), hipMemcpyHostToDevice));
+      } else {  // kComb: scatter ingress rows into comb_x cells by meta
+        if (h.count) {
+          UCCL_CHECK_HIP(hipMemcpyAsync(ingress_meta(heap_, v_, ing0),
+                                        metas.data(),
+                                        h.count * sizeof(uint32_t),
+                                        hipMemcpyHostToDevice, rx_stream_));
+          launch_ep_comb_scatter(v_, ing0, h.count, rx_stream_);
+          UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+        }
+      }
+    }
+  } catch (std::exception const&) {
+    // transport closed
+  }
+}
+
+}  // namespace ep
+}  // namespace uccl

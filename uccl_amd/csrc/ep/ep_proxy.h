@@ -1,0 +1,98 @@
+// CPU proxy for GPU-initiated EP transfers.
+//
+// Parity role: the reference's ep/src/proxy.cpp run_dual loop — drain the
+// device→host command rings, execute the network writes, complete back
+// into peer GPU memory. MI355X re-design: the wire is
+// the uccl_amd multipath reliable transport (csrc/transport) instead of
+// raw ibverbs, the completion write-back is a host→device copy of the
+// seq-tagged count (system-visible to the spinning wait kernel), and
+// combine returns are shipped host-side off a HIP event (the dispatch
+// direction is the GPU-initiated one, via the D2H ring).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <atomic>
+#include <condition_variable>
+#include <deque>
+#include <memory>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "../transport/reliable.h"
+#include "d2h_ring.h"
+#include "ep_layout.h"
+
+namespace uccl {
+namespace ep {
+
+class EpProxy {
+ public:
+  EpProxy(const EpView& view, void* heap, D2HRing* ring_host, int device);
+  ~EpProxy();
+
+  std::string transport_metadata() const { return tp_->metadata(); }
+  void set_view(const EpView& v) { v_ = v; }
+  // flows[r] for every proxy peer r; establish with connect/accept by rank
+  // order convention (lower rank accepts, higher rank connects).
+  void establish_flows(const std::vector<std::string>& tp_md,
+                       uint32_t proxy_mask);
+
+  // combine returns: ship expert_out rows for every proxy peer once
+  // `ready` has been recorded on the compute stream; counts[le*world+src]
+  // is the pinned host copy of the dispatch counts.
+  void enqueue_combine(void const* expert_out, uint64_t seq,
+                       hipEvent_t ready, int const* counts);
+
+  void start();
+
+ private:
+  struct WireHdr {
+    uint32_t kind;   // 1=dispatch, 2=combine, 3=combine-done
+    uint32_t seq32;
+    uint32_t le;
+    uint32_t src;
+    uint64_t count;
+  };
+
+  void ring_loop();
+  void rx_loop(int peer);
+  void comb_tx_loop();
+  void ship_rows(uint64_t flow, WireHdr const& h, void const* dev_rows,
+                 uint32_t const* dev_metas_or_null,
+                 std::vector<uint32_t> const* host_metas);
+
+  EpView v_;
+  void* heap_;
+  D2HRing* ring_;
+  int device_;
+  uint32_t proxy_mask_ = 0;
+
+  std::unique_ptr<transport::TransportEndpoint> tp_;
+  std::vector<uint64_t> flows_;
+
+  hipStream_t stream_ = nullptr;     // non-blocking proxy stream (tx)
+  hipStream_t rx_stream_ = nullptr;  // non-blocking proxy stream (rx)
+  void* host_buf_ = nullptr;         // pinned staging (tx)
+  void* rx_buf_ = nullptr;           // pinned staging (rx)
+  size_t stage_bytes_ = 0;
+
+  struct CombTask {
+    void const* expert_out;
+    uint64_t seq;
+    hipEvent_t ready;
+    std::vector<int> counts;
+  };
+  std::deque<CombTask> comb_q_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::atomic<bool> stop_{false};
+
+  std::thread ring_thread_;
+  std::thread comb_thread_;
+  std::vector<std::thread> rx_threads_;
+};
+
+}  // namespace ep
+}  // namespace uccl

@@ -4,6 +4,7 @@
 #include <sys/socket.h>
 
 #include <algorithm>
+#include <stdexcept>
 #include <cstdlib>
 #include <cstring>
 
@@ -132,6 +133,7 @@ struct TransportEndpoint::Impl {
   std::condition_variable cv;
   std::unordered_map<uint64_t, std::unique_ptr<Flow>> flows;
   std::deque<uint64_t> accepted;
+  std::deque<uint64_t> accepted_tags;
   std::atomic<uint64_t> next_flow{1};
 
   Stats st;
@@ -372,6 +374,7 @@ struct TransportEndpoint::Impl {
   // ---- flow setup over TCP ctrl ----
   struct CtrlMsg {
     uint64_t flow;
+    uint64_t tag;  // connector-supplied peer identity
     int num_paths;
     uint16_t ports[64];
     char ip[48];
@@ -397,6 +400,7 @@ struct TransportEndpoint::Impl {
       {
         std::lock_guard<std::mutex> g(mu);
         accepted.push_back(flow);
+        accepted_tags.push_back(peer.tag);
       }
       cv.notify_all();
     }
@@ -471,11 +475,16 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   impl_->progress = std::thread([this] { impl_->progress_loop(); });
 }
 
-TransportEndpoint::~TransportEndpoint() {
-  impl_->stop = true;
+void TransportEndpoint::shutdown() {
+  if (impl_->stop.exchange(true)) return;
   ::shutdown(impl_->ctrl_listen, SHUT_RDWR);
-  ::close(impl_->ctrl_listen);
   impl_->wake();
+  impl_->cv.notify_all();
+}
+
+TransportEndpoint::~TransportEndpoint() {
+  shutdown();
+  ::close(impl_->ctrl_listen);
   if (impl_->ctrl_thread.joinable()) impl_->ctrl_thread.join();
   if (impl_->progress.joinable()) impl_->progress.join();
   for (int s : impl_->socks) ::close(s);
@@ -489,13 +498,14 @@ std::string TransportEndpoint::metadata() const {
   return buf;
 }
 
-uint64_t TransportEndpoint::connect(const std::string& md) {
+uint64_t TransportEndpoint::connect(const std::string& md, uint64_t tag) {
   auto const pos = md.rfind(':');
   UCCL_CHECK(pos != std::string::npos) << "bad transport metadata";
   std::string ip = md.substr(0, pos);
   uint16_t port = static_cast<uint16_t>(atoi(md.c_str() + pos + 1));
   int fd = net::connect_to(ip, port);
   Impl::CtrlMsg mine = impl_->self_ctrl(0);
+  mine.tag = tag;
   net::send_all(fd, &mine, sizeof(mine));
   Impl::CtrlMsg peer{};
   UCCL_CHECK(net::recv_all(fd, &peer, sizeof(peer))) << "ctrl handshake";
@@ -504,7 +514,7 @@ uint64_t TransportEndpoint::connect(const std::string& md) {
   return peer.flow;
 }
 
-uint64_t TransportEndpoint::accept() {
+uint64_t TransportEndpoint::accept(uint64_t* peer_tag) {
   std::unique_lock<std::mutex> lk(impl_->mu);
   impl_->cv.wait(lk, [this] {
     return !impl_->accepted.empty() || impl_->stop;
@@ -512,6 +522,8 @@ uint64_t TransportEndpoint::accept() {
   UCCL_CHECK(!impl_->accepted.empty()) << "endpoint closed";
   uint64_t f = impl_->accepted.front();
   impl_->accepted.pop_front();
+  if (peer_tag) *peer_tag = impl_->accepted_tags.front();
+  impl_->accepted_tags.pop_front();
   return f;
 }
 
@@ -533,7 +545,7 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
   impl_->wake();
   std::unique_lock<std::mutex> lk(impl_->mu);
   impl_->cv.wait(lk, [&] { return m->done || impl_->stop; });
-  UCCL_CHECK(m->done) << "endpoint closed during send";
+  if (!m->done) throw std::runtime_error("transport closed during send");
 }
 
 void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
@@ -560,7 +572,7 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
             it->second.recv_bytes >= it->second.bytes) ||
            impl_->stop;
   });
-  UCCL_CHECK(!impl_->stop) << "endpoint closed during recv";
+  if (impl_->stop) throw std::runtime_error("transport closed during recv");
   f.rxmsgs.erase(msg_id);
 }
 

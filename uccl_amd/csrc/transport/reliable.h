@@ -57,14 +57,20 @@ class TransportEndpoint {
   ~TransportEndpoint();
 
   std::string metadata() const;  // {ip, ctrl_port}
-  uint64_t connect(const std::string& remote_metadata);
-  uint64_t accept();
+  // `tag` identifies the connecting peer to the acceptor (e.g. its rank).
+  uint64_t connect(const std::string& remote_metadata, uint64_t tag = 0);
+  uint64_t accept(uint64_t* peer_tag = nullptr);
 
   // Blocking reliable message ops (in-order per flow).
   void send_msg(uint64_t flow, void const* ptr, size_t bytes);
   void recv_msg(uint64_t flow, void* ptr, size_t bytes);
 
   Stats stats() const;
+
+  // Unblock all pending send/recv (they throw std::runtime_error) and stop
+  // the progress machinery; safe to call before destruction while other
+  // threads are still blocked in recv_msg.
+  void shutdown();
 
  private:
   struct Flow;
