@@ -15,6 +15,12 @@ constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3;
 constexpr size_t kStageBytes = 16ull << 20;  // pinned staging chunk
 }  // namespace
 
+int EpProxy::flow_peer(uint64_t flow) const {
+  for (int r = 0; r < v_.world; ++r)
+    if (flows_[r] == flow) return r;
+  return 0;
+}
+
 EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
                  int device)
     : v_(view), heap_(heap), ring_(ring_host), device_(device) {
@@ -22,6 +28,8 @@ EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
       static_cast<int>(env_int("UCCL_EP_PROXY_PATHS", 8)),
       static_cast<size_t>(env_int("UCCL_EP_PROXY_CHUNK", 16384)));
   flows_.resize(v_.world, 0);
+  for (int r = 0; r < v_.world; ++r)
+    flow_mu_.emplace_back(new std::mutex());
   UCCL_CHECK_HIP(hipSetDevice(device_));
   UCCL_CHECK_HIP(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
   UCCL_CHECK_HIP(
@@ -82,6 +90,7 @@ void EpProxy::ship_rows(uint64_t flow, WireHdr const& h,
                         void const* dev_rows,
                         uint32_t const* dev_metas_or_null,
                         std::vector<uint32_t> const* host_metas) {
+  std::lock_guard<std::mutex> guard(*flow_mu_[flow_peer(flow)]);
   size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
   size_t const total = h.count * row_bytes;
   tp_->send_msg(flow, &h, sizeof(h));
@@ -158,6 +167,7 @@ void EpProxy::comb_tx_loop() {
         for (int le = 0; le < v_.local_experts; ++le) {
           uint64_t const count =
               static_cast<uint64_t>(task.counts[le * v_.world + src]);
+          (void)row_bytes;
           WireHdr h{kComb, static_cast<uint32_t>(task.seq),
                     static_cast<uint32_t>(le),
                     static_cast<uint32_t>(v_.rank), count};
@@ -173,7 +183,10 @@ void EpProxy::comb_tx_loop() {
         }
         WireHdr done{kCombDone, static_cast<uint32_t>(task.seq), 0,
                      static_cast<uint32_t>(v_.rank), 0};
-        tp_->send_msg(flows_[src], &done, sizeof(done));
+        {
+          std::lock_guard<std::mutex> guard(*flow_mu_[src]);
+          tp_->send_msg(flows_[src], &done, sizeof(done));
+        }
       }
     } catch (std::exception const&) {
       return;
@@ -188,7 +201,7 @@ void EpProxy::enqueue_combine(void const* expert_out, uint64_t seq,
   t.expert_out = expert_out;
   t.seq = seq;
   t.ready = ready;
-  t.counts.assign(counts, counts + v_.local_experts * v_.world);
+  t.counts = counts;
   {
     std::lock_guard<std::mutex> g(mu_);
     comb_q_.push_back(std::move(t));
@@ -213,9 +226,14 @@ void EpProxy::rx_loop(int peer) {
         // all combine payloads from `peer` for this seq have been
         // scattered (rx_stream_ is in-order); publish the flag
         UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+        // NB: must NOT use the legacy default stream here — it would wait
+        // for the peer's spinning wait-kernel (deadlock). rx_stream_ is
+        // non-blocking.
         uint64_t const seq = h.seq32;
-        UCCL_CHECK_HIP(hipMemcpy(comb_flag_ptr(heap_, v_, peer), &seq,
-                                 sizeof(seq), hipMemcpyHostToDevice));
+        UCCL_CHECK_HIP(hipMemcpyAsync(comb_flag_ptr(heap_, v_, peer), &seq,
+                                      sizeof(seq), hipMemcpyHostToDevice,
+                                      rx_stream_));
+        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
         continue;
       }
       size_t const total = h.count * row_bytes;
@@ -245,9 +263,10 @@ void EpProxy::rx_loop(int peer) {
         UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
         uint64_t const tagged =
             (static_cast<uint64_t>(h.seq32) << 32) | h.count;
-        UCCL_CHECK_HIP(
-            hipMemcpy(disp_count_ptr(heap_, v_, h.le, h.src), &tagged,
-                      sizeof(tagged), hipMemcpyHostToDevice));
+        UCCL_CHECK_HIP(hipMemcpyAsync(disp_count_ptr(heap_, v_, h.le, h.src),
+                                      &tagged, sizeof(tagged),
+                                      hipMemcpyHostToDevice, rx_stream_));
+        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
       } else {  // kComb: scatter ingress rows into comb_x cells by meta
         if (h.count) {
           UCCL_CHECK_HIP(hipMemcpyAsync(ingress_meta(heap_, v_, ing0),

@@ -59,6 +59,7 @@ class EpProxy {
   void ring_loop();
   void rx_loop(int peer);
   void comb_tx_loop();
+  int flow_peer(uint64_t flow) const;
   void ship_rows(uint64_t flow, WireHdr const& h, void const* dev_rows,
                  uint32_t const* dev_metas_or_null,
                  std::vector<uint32_t> const* host_metas);
@@ -71,6 +72,9 @@ class EpProxy {
 
   std::unique_ptr<transport::TransportEndpoint> tp_;
   std::vector<uint64_t> flows_;
+  // serializes multi-message sequences (hdr, payload..., metas) per flow:
+  // the ring thread and the combine-tx thread share peer flows
+  std::vector<std::unique_ptr<std::mutex>> flow_mu_;
 
   hipStream_t stream_ = nullptr;     // non-blocking proxy stream (tx)
   hipStream_t rx_stream_ = nullptr;  // non-blocking proxy stream (rx)
@@ -82,7 +86,8 @@ class EpProxy {
     void const* expert_out;
     uint64_t seq;
     hipEvent_t ready;
-    std::vector<int> counts;
+    int const* counts;  // pinned host buffer; valid to read once `ready`
+                        // has completed (the D2H precedes it in-stream)
   };
   std::deque<CombTask> comb_q_;
   std::mutex mu_;
