@@ -142,6 +142,16 @@ Endpoint::Endpoint(int gpu, int num_workers) : gpu_(gpu) {
     workers_.emplace_back([this] { worker_loop(); });
 }
 
+hipStream_t Endpoint::copy_stream() {
+  std::lock_guard<std::mutex> g(copy_mu_);
+  if (!copy_stream_) {
+    hipStream_t s = nullptr;
+    UCCL_CHECK_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    copy_stream_ = s;
+  }
+  return static_cast<hipStream_t>(copy_stream_);
+}
+
 Endpoint::~Endpoint() {
   stop_ = true;
   ::shutdown(listen_fd_, SHUT_RDWR);
@@ -167,6 +177,8 @@ Endpoint::~Endpoint() {
     conns_.clear();
   }
   if (staging_) (void)hipHostFree(staging_);
+  if (copy_stream_)
+    (void)hipStreamDestroy(static_cast<hipStream_t>(copy_stream_));
 }
 
 std::string Endpoint::metadata() const {
@@ -307,9 +319,12 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
         }
         UCCL_CHECK(h.b + h.c <= mr.bytes) << "ipc write overflows mr";
         void* src_base = open_ipc(*c, &blob, blob.device);
-        UCCL_CHECK_HIP(hipMemcpy(static_cast<char*>(mr.ptr) + h.b,
-                                 static_cast<char*>(src_base) + blob.offset,
-                                 h.c, hipMemcpyDeviceToDevice));
+        hipStream_t cs = copy_stream();
+        UCCL_CHECK_HIP(hipMemcpyAsync(
+            static_cast<char*>(mr.ptr) + h.b,
+            static_cast<char*>(src_base) + blob.offset, h.c,
+            hipMemcpyDeviceToDevice, cs));
+        UCCL_CHECK_HIP(hipStreamSynchronize(cs));
         if (h.d) c->send_msg(MsgHdr{kIpcDone, 0, 0, 0, h.d});
         break;
       }
@@ -456,9 +471,11 @@ void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
     UCCL_CHECK(is_gpu(device)) << "IPC send into host recv buffer";
     UCCL_CHECK(item->bytes <= bytes) << "recv buffer too small";
     void* src_base = open_ipc(c, &item->blob, item->blob.device);
-    UCCL_CHECK_HIP(
-        hipMemcpy(ptr, static_cast<char*>(src_base) + item->blob.offset,
-                  item->bytes, hipMemcpyDeviceToDevice));
+    hipStream_t cs = copy_stream();
+    UCCL_CHECK_HIP(hipMemcpyAsync(
+        ptr, static_cast<char*>(src_base) + item->blob.offset, item->bytes,
+        hipMemcpyDeviceToDevice, cs));
+    UCCL_CHECK_HIP(hipStreamSynchronize(cs));
     c.send_msg(MsgHdr{kIpcDone, 0, 0, 0, item->token});
   } else {
     copy_to_user(*item, ptr, bytes, device);

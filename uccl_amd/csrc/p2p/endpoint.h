@@ -14,6 +14,8 @@
 //     (p2p/engine.cc:2267)
 #pragma once
 
+#include <hip/hip_runtime.h>
+
 #include <atomic>
 #include <condition_variable>
 #include <cstdint>
@@ -127,6 +129,13 @@ class Endpoint {
   void* staging_ = nullptr;
   size_t staging_bytes_ = 0;
   std::mutex staging_mu_;
+
+  // dedicated non-blocking stream for IPC DtoD copies: hipMemcpy DtoD is
+  // async w.r.t. the host, so completion must be an explicit stream sync
+  // before acking the sender (and must not ride the legacy null stream)
+  hipStream_t copy_stream();
+  void* copy_stream_ = nullptr;
+  std::mutex copy_mu_;
 };
 
 }  // namespace p2p
