@@ -22,7 +22,7 @@ Communicator::Communicator(int rank, int world, int device, size_t heap_bytes)
   UCCL_CHECK(rank >= 0 && rank < world) << "bad rank " << rank;
   heap_bytes_ = heap_bytes ? heap_bytes
                            : static_cast<size_t>(
-                                 env_int("UCCL_SYM_HEAP_MB", 1056)) *
+                                 env_int("UCCL_SYM_HEAP_MB", 1184)) *
                                  (1 << 20);
   UCCL_CHECK(heap_bytes_ > kScratchAOffset + (4 << 20))
       << "heap too small: " << heap_bytes_;
