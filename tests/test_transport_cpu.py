@@ -126,3 +126,10 @@ def test_transport_stats_cc():
     st = a.stats()
     assert st.srtt_us > 0
     assert 2.0 <= st.cwnd <= 4096.0
+
+
+def test_transport_swift_cc():
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_CC="swift")
+    xfer(a, b, fa, fb, 2 << 20, 31)
+    st = a.stats()
+    assert st.msgs_sent == 1 and st.srtt_us > 0

@@ -203,10 +203,13 @@ struct TransportEndpoint::Impl {
     ++st.acks_sent;
   }
 
+  // Congestion control, selectable like the reference's cc_state mux
+  // (include/cc/cc_state.h:24 {kNone,kTimely,kSwift}):
+  //   UCCL_TP_CC=timely  RTT-gradient (SIGCOMM'15), the default
+  //   UCCL_TP_CC=swift   delay-target window (SIGCOMM'20)
+  //   UCCL_TP_CC=none    fixed window (UCCL_TP_CWND_MAX)
   void timely_update(Flow& f, double rtt_us) {
-    // Timely-style RTT-gradient window adaptation (SIGCOMM'15 adapted to
-    // window form; pacing is bypassed, matching the reference default
-    // BYPASS_PACING=1, transport_config.h:35)
+    static std::string const cc = env_str("UCCL_TP_CC", "timely");
     static double const t_low = env_int("UCCL_TP_TLOW_US", 50);
     static double const t_high = env_int("UCCL_TP_THIGH_US", 1000);
     static double const add = 1.0, beta = 0.8;
@@ -214,7 +217,21 @@ struct TransportEndpoint::Impl {
     double const grad = (rtt_us - f.prev_rtt_us) / std::max(f.srtt_us, 1.0);
     f.prev_rtt_us = rtt_us;
     f.srtt_us = 0.875 * f.srtt_us + 0.125 * rtt_us;
-    if (rtt_us < t_low) {
+    if (cc == "none") {
+      f.cwnd = 1e9;  // clamped to cwnd_max below
+    } else if (cc == "swift") {
+      // Swift: additive increase below the delay target, multiplicative
+      // decrease proportional to the overshoot (capped)
+      static double const target = env_int("UCCL_TP_SWIFT_TARGET_US", 300);
+      static double const max_mdf = 0.5;
+      if (rtt_us < target) {
+        f.cwnd += add / std::max(f.cwnd, 1.0) * 8.0;
+      } else {
+        double const mdf =
+            std::min(beta * (rtt_us - target) / rtt_us, max_mdf);
+        f.cwnd *= 1.0 - mdf;
+      }
+    } else if (rtt_us < t_low) {
       f.cwnd += add;
     } else if (rtt_us > t_high) {
       f.cwnd *= 1.0 - beta * (1.0 - t_high / rtt_us);
