@@ -237,15 +237,16 @@ PYBIND11_MODULE(_C, m) {
   // --- EP (DeepEP-compatible expert parallel) -----------------------------
   py::class_<uccl::ep::EpBuffer>(m, "EpBuffer")
       .def(py::init([](int rank, int world, int device, int num_experts,
-                       int topk, int hidden, int max_tokens,
-                       int elem_size) {
+                       int topk, int hidden, int max_tokens, int elem_size,
+                       bool use_fp8) {
              return new uccl::ep::EpBuffer(rank, world, device, num_experts,
                                            topk, hidden, max_tokens,
-                                           elem_size);
+                                           elem_size, use_fp8);
            }),
            py::arg("rank"), py::arg("world"), py::arg("device"),
            py::arg("num_experts"), py::arg("topk"), py::arg("hidden"),
-           py::arg("max_tokens"), py::arg("elem_size"))
+           py::arg("max_tokens"), py::arg("elem_size"),
+           py::arg("use_fp8") = false)
       .def("handle_bytes",
            [](uccl::ep::EpBuffer& b) { return py::bytes(b.handle_bytes()); })
       .def("connect",
@@ -268,6 +269,21 @@ PYBIND11_MODULE(_C, m) {
              b.dispatch(x.data_ptr(), topk_idx.data_ptr<int64_t>(),
                         static_cast<int>(x.size(0)), counts.data_ptr<int>(),
                         current_stream(b.device()));
+             if (v.disp_fp8) {
+               auto recv_x = at::from_blob(
+                   b.recv_x_ptr(),
+                   {v.local_experts,
+                    static_cast<int64_t>(v.world) * v.max_tokens, v.hidden},
+                   at::TensorOptions().dtype(at::kFloat8_e4m3fn)
+                       .device(x.device()));
+               auto recv_scale = at::from_blob(
+                   b.recv_scale_ptr(),
+                   {v.local_experts,
+                    static_cast<int64_t>(v.world) * v.max_tokens,
+                    v.hidden / 128},
+                   at::TensorOptions().dtype(at::kFloat).device(x.device()));
+               return py::make_tuple(recv_x, counts, recv_scale);
+             }
              auto recv_x = at::from_blob(
                  b.recv_x_ptr(),
                  {v.local_experts,

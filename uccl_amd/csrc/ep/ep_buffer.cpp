@@ -21,8 +21,13 @@ namespace ep {
 static size_t align256(size_t x) { return (x + 255) & ~size_t(255); }
 
 EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
-                   int topk, int hidden, int max_tokens, int elem_size)
+                   int topk, int hidden, int max_tokens, int elem_size,
+                   bool use_fp8)
     : rank_(rank), world_(world), device_(device) {
+  if (use_fp8) {
+    UCCL_CHECK(hidden % 128 == 0) << "fp8 dispatch needs hidden % 128 == 0";
+    UCCL_CHECK(elem_size == 2) << "fp8 dispatch quantizes bf16/fp16 input";
+  }
   UCCL_CHECK(world >= 1 && world <= kMaxRanks) << "world=" << world;
   UCCL_CHECK(num_experts % world == 0)
       << "num_experts " << num_experts << " must divide world " << world;
@@ -38,6 +43,8 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   v_.hidden = hidden;
   v_.max_tokens = max_tokens;
   v_.elem_size = elem_size;
+  v_.disp_fp8 = use_fp8 ? 1 : 0;
+  v_.disp_elem = use_fp8 ? 1 : elem_size;
   v_.seq = 0;
 
   size_t off = 0;
@@ -50,7 +57,11 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
                            static_cast<size_t>(max_tokens));
   v_.off_disp_x = off;
   off = align256(off + static_cast<size_t>(v_.local_experts) * world *
-                           max_tokens * hidden * elem_size);
+                           max_tokens * hidden * v_.disp_elem);
+  v_.off_disp_scale = off;
+  if (use_fp8)
+    off = align256(off + static_cast<size_t>(v_.local_experts) * world *
+                             max_tokens * (hidden / 128) * sizeof(float));
   v_.off_comb_x = off;
   off = align256(off + static_cast<size_t>(max_tokens) * topk * hidden *
                            elem_size);
@@ -80,6 +91,8 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   v_.peers[rank_] = heap_;
   v_.proxy_mask = 0;
   v_.ring = nullptr;
+  UCCL_CHECK(!(use_fp8 && env_bool("UCCL_EP_FORCE_PROXY", false)))
+      << "fp8 dispatch not yet supported on the proxy path";
   if (env_bool("UCCL_EP_FORCE_PROXY", false) && world_ > 1) {
     UCCL_CHECK_HIP(hipHostMalloc(&ring_host_, sizeof(D2HRing),
                                  hipHostMallocMapped));

@@ -19,7 +19,7 @@ struct D2HRing;
 class EpBuffer {
  public:
   EpBuffer(int rank, int world, int device, int num_experts, int topk,
-           int hidden, int max_tokens, int elem_size);
+           int hidden, int max_tokens, int elem_size, bool use_fp8 = false);
   ~EpBuffer();
   EpBuffer(const EpBuffer&) = delete;
 
@@ -39,6 +39,9 @@ class EpBuffer {
   const EpView& view() const { return v_; }
   void* recv_x_ptr() const {
     return static_cast<char*>(heap_) + v_.off_disp_x;
+  }
+  void* recv_scale_ptr() const {
+    return static_cast<char*>(heap_) + v_.off_disp_scale;
   }
   void* recv_meta_ptr() const {
     return static_cast<char*>(heap_) + v_.off_disp_meta;

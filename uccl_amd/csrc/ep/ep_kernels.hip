@@ -32,6 +32,41 @@ __device__ inline uint64_t tag_count(uint64_t seq, uint32_t count) {
   return (seq << 32) | count;
 }
 
+// Quantize one hidden row (bf16) to fp8-e4m3 with per-128-element scales,
+// cooperatively with a 256-thread block (two 128-element groups in
+// flight; cross-wave amax via LDS). DeepEP LL fp8 semantics
+// (reference ep/src/internode_ll.cu:96-200): scale = amax/448.
+__device__ inline void block_quant_row_fp8(
+    char* __restrict__ dst, float* __restrict__ dscale,
+    __hip_bfloat16 const* __restrict__ src, int hidden, float* red) {
+  int const ngroups = hidden / 128;
+  for (int g0 = 0; g0 < ngroups; g0 += 2) {
+    int const g = g0 + (threadIdx.x >> 7);
+    int const lane = threadIdx.x & 127;
+    bool const active = g < ngroups;
+    float v = 0.f;
+    if (active) v = __bfloat162float(src[g * 128 + lane]);
+    red[threadIdx.x] = fabsf(v);
+    __syncthreads();
+    // tree-reduce amax within each 128-thread half
+#pragma unroll
+    for (int off = 64; off > 0; off >>= 1) {
+      if (lane < off)
+        red[threadIdx.x] =
+            fmaxf(red[threadIdx.x], red[threadIdx.x + off]);
+      __syncthreads();
+    }
+    float const amax = red[(threadIdx.x >> 7) << 7];
+    float const scale = amax > 0.f ? amax / 448.f : 1.f;
+    if (active) {
+      reinterpret_cast<__hip_fp8_e4m3*>(dst)[g * 128 + lane] =
+          __hip_fp8_e4m3(v / scale);
+      if (lane == 0) dscale[g] = scale;
+    }
+    __syncthreads();
+  }
+}
+
 // Copy `bytes` from src to dst cooperatively with the whole block.
 __device__ inline void block_copy(char* __restrict__ dst,
                                   char const* __restrict__ src,
@@ -125,7 +160,16 @@ __global__ void k_ep_dispatch_copy(EpView v, void const* __restrict__ x) {
       if (threadIdx.x == 0) *egress_meta(me, v, pfx + i) = tk;
     } else {
       size_t const slot = static_cast<size_t>(v.rank) * v.max_tokens + i;
-      block_copy(disp_x_ptr(dbase, v, le, slot), src_row, row_bytes);
+      if (v.disp_fp8) {
+        __shared__ float red[256];
+        block_quant_row_fp8(
+            disp_x_ptr(dbase, v, le, slot),
+            disp_scale_ptr(dbase, v, le, slot),
+            reinterpret_cast<__hip_bfloat16 const*>(src_row), v.hidden,
+            red);
+      } else {
+        block_copy(disp_x_ptr(dbase, v, le, slot), src_row, row_bytes);
+      }
       if (threadIdx.x == 0) *disp_meta_ptr(dbase, v, le, slot) = tk;
     }
   }

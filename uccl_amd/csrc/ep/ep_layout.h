@@ -31,13 +31,18 @@ struct EpView {
   int topk;
   int hidden;          // elements per token
   int max_tokens;      // per source rank (DeepEP num_max_dispatch_tokens_per_rank)
-  int elem_size;       // bytes per element of x
+  int elem_size;       // bytes per element of x (combine payload dtype)
+  int disp_fp8;        // 1 => dispatch payload is fp8-e4m3 with per-128
+                       //   scales (DeepEP LL fp8 mode); else raw elem_size
+  int disp_elem;       // bytes per dispatched element (1 if disp_fp8)
   uint64_t seq;
   void* peers[kMaxRanks];
 
   // byte offsets into every rank's heap:
   size_t off_disp_count;  // u64 [local_experts][world]    (seq<<32 | count)
-  size_t off_disp_x;      // [local_experts][world*max_tokens][hidden] elems
+  size_t off_disp_x;      // [local_experts][world*max_tokens][hidden] disp_elem
+  size_t off_disp_scale;  // f32 [local_experts][world*max_tokens][hidden/128]
+                          //   (fp8 mode only)
   size_t off_disp_meta;   // u32 [local_experts][world*max_tokens]
                           //   meta = src_token_idx | (k << 24)
   size_t off_comb_flag;   // u64 [world]                   (seq)
@@ -67,7 +72,15 @@ __host__ __device__ inline char* disp_x_ptr(void* base, const EpView& v,
   return static_cast<char*>(base) + v.off_disp_x +
          ((static_cast<size_t>(le) * v.world * v.max_tokens + slot) *
           v.hidden) *
-             v.elem_size;
+             v.disp_elem;
+}
+
+__host__ __device__ inline float* disp_scale_ptr(void* base, const EpView& v,
+                                                 int le, size_t slot) {
+  return reinterpret_cast<float*>(static_cast<char*>(base) +
+                                  v.off_disp_scale) +
+         (static_cast<size_t>(le) * v.world * v.max_tokens + slot) *
+             (v.hidden / 128);
 }
 
 __host__ __device__ inline uint32_t* disp_meta_ptr(void* base,

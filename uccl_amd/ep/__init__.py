@@ -54,7 +54,7 @@ class Buffer:
     def __init__(self, group=None, num_experts: int = 8, topk: int = 2,
                  hidden: int = 7168, max_tokens: int = 4096,
                  dtype: torch.dtype = torch.bfloat16,
-                 device: Optional[int] = None):
+                 device: Optional[int] = None, use_fp8: bool = False):
         from uccl_amd import _load_native
 
         C = _load_native(required=True)
@@ -73,9 +73,10 @@ class Buffer:
         self.topk = topk
         self.hidden = hidden
         self.max_tokens = max_tokens
+        self.use_fp8 = use_fp8
         elem = torch.tensor([], dtype=dtype).element_size()
         self._b = C.EpBuffer(self.rank, self.world, device, num_experts,
-                             topk, hidden, max_tokens, elem)
+                             topk, hidden, max_tokens, elem, use_fp8)
         if self.world > 1:
             handles = [None] * self.world
             dist.all_gather_object(handles, self._b.handle_bytes(),
@@ -86,11 +87,12 @@ class Buffer:
     def local_experts(self) -> int:
         return self.num_experts // self.world
 
-    def dispatch(self, x: torch.Tensor, topk_idx: torch.Tensor
-                 ) -> Tuple[torch.Tensor, torch.Tensor]:
+    def dispatch(self, x: torch.Tensor, topk_idx: torch.Tensor):
         """Returns (packed_recv_x [local_E, world*max_tokens, hidden] view,
-        recv_count [local_E, world] int32). Slots for source rank r live at
-        [e, r*max_tokens : r*max_tokens + recv_count[e, r]]."""
+        recv_count [local_E, world] int32) — plus recv_scales
+        [local_E, world*max_tokens, hidden/128] f32 when use_fp8 (payloads
+        are e4m3 with per-128 scales, DeepEP LL fp8 semantics). Slots for
+        source rank r live at [e, r*max_tokens : + recv_count[e, r]]."""
         return self._b.dispatch(x, topk_idx)
 
     def combine(self, expert_out: torch.Tensor, topk_idx: torch.Tensor,
