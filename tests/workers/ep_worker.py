@@ -26,6 +26,41 @@ def rank_inputs(r: int, T: int, H: int, K: int, E: int, dtype, seed: int):
     return x, topk, w
 
 
+def fp8_check():
+    """fp8 dispatch: quantized payloads + per-128 scales must dequantize
+    back to the input within e4m3 relative error."""
+    import uccl_amd.ep as uep
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    T, H, K = 128, 512, 2
+    E = 4 * world
+    buf = uep.Buffer(num_experts=E, topk=K, hidden=H, max_tokens=256,
+                     dtype=torch.bfloat16, use_fp8=True)
+    x, topk, w = rank_inputs(rank, T, H, K, E, torch.bfloat16, 77)
+    recv_x, counts, recv_scale = buf.dispatch(x.cuda(), topk.cuda())
+    torch.cuda.synchronize()
+    local_E = E // world
+    all_inputs = [rank_inputs(r, T, H, K, E, torch.bfloat16, 77)
+                  for r in range(world)]
+    counts_cpu = counts.cpu()
+    for le in range(local_E):
+        e = rank * local_E + le
+        for src in range(world):
+            sx, stopk, _ = all_inputs[src]
+            sel = [t for t in range(T) if (stopk[t] == e).any()]
+            n = counts_cpu[le, src].item()
+            assert n == len(sel), (n, len(sel))
+            q = recv_x[le, src * 256: src * 256 + n].float().cpu()
+            sc = recv_scale[le, src * 256: src * 256 + n].cpu()
+            deq = q.view(n, H // 128, 128) * sc.unsqueeze(-1)
+            want = sx[sel].float().view(n, H // 128, 128)
+            err = (deq - want).abs()
+            ref = want.abs().amax(dim=-1, keepdim=True).clamp(min=1e-6)
+            assert (err / ref).max() <= 0.08, float((err / ref).max())
+    print(f"[rank {rank}] EP FP8 OK", flush=True)
+
+
 def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -114,38 +149,3 @@ def main():
 
 if __name__ == "__main__":
     main()
-
-
-def fp8_check():
-    """fp8 dispatch: quantized payloads + per-128 scales must dequantize
-    back to the input within e4m3 relative error."""
-    import uccl_amd.ep as uep
-
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    T, H, K = 128, 512, 2
-    E = 4 * world
-    buf = uep.Buffer(num_experts=E, topk=K, hidden=H, max_tokens=256,
-                     dtype=torch.bfloat16, use_fp8=True)
-    x, topk, w = rank_inputs(rank, T, H, K, E, torch.bfloat16, 77)
-    recv_x, counts, recv_scale = buf.dispatch(x.cuda(), topk.cuda())
-    torch.cuda.synchronize()
-    local_E = E // world
-    all_inputs = [rank_inputs(r, T, H, K, E, torch.bfloat16, 77)
-                  for r in range(world)]
-    counts_cpu = counts.cpu()
-    for le in range(local_E):
-        e = rank * local_E + le
-        for src in range(world):
-            sx, stopk, _ = all_inputs[src]
-            sel = [t for t in range(T) if (stopk[t] == e).any()]
-            n = counts_cpu[le, src].item()
-            assert n == len(sel), (n, len(sel))
-            q = recv_x[le, src * 256: src * 256 + n].float().cpu()
-            sc = recv_scale[le, src * 256: src * 256 + n].cpu()
-            deq = q.view(n, H // 128, 128) * sc.unsqueeze(-1)
-            want = sx[sel].float().view(n, H // 128, 128)
-            err = (deq - want).abs()
-            ref = want.abs().amax(dim=-1, keepdim=True).clamp(min=1e-6)
-            assert (err / ref).max() <= 0.08, float((err / ref).max())
-    print(f"[rank {rank}] EP FP8 OK", flush=True)
