@@ -47,6 +47,31 @@ def main():
         want_seg = torch.arange(rank * 8, rank * 8 + 8, dtype=torch.float32) + r * 1000
         assert torch.equal(seg, want_seg), (r, seg, want_seg)
 
+    # reduce / gather / scatter
+    r = torch.full((128,), float(rank + 1), device="cuda")
+    dist.reduce(r, dst=0)
+    torch.cuda.synchronize()
+    if rank == 0:
+        assert torch.allclose(r, torch.full_like(r, float(want)))
+
+    gin = torch.full((64,), float(rank * 10), device="cuda")
+    gouts = [torch.empty(64, device="cuda") for _ in range(world)] \
+        if rank == 0 else []
+    dist.gather(gin, gouts if rank == 0 else None, dst=0)
+    torch.cuda.synchronize()
+    if rank == 0:
+        for rr in range(world):
+            assert torch.allclose(gouts[rr],
+                                  torch.full((64,), float(rr * 10),
+                                             device="cuda"))
+
+    sout = torch.empty(32, device="cuda")
+    sins = [torch.full((32,), float(100 + rr), device="cuda")
+            for rr in range(world)] if rank == 0 else None
+    dist.scatter(sout, sins, src=0)
+    torch.cuda.synchronize()
+    assert torch.allclose(sout, torch.full_like(sout, float(100 + rank)))
+
     dist.barrier()
 
     # DDP end-to-end: tiny model, grads must match single-process reference

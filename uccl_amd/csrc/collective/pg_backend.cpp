@@ -201,6 +201,73 @@ class UcclBackend : public c10d::Backend {
         std::vector<at::Tensor>{output});
   }
 
+  c10::intrusive_ptr<c10d::Work> reduce(
+      std::vector<at::Tensor>& tensors,
+      const c10d::ReduceOptions& opts) override {
+    // Only the root's tensor is specified to hold the result; implement
+    // as a SUM allreduce (non-root tensors also end up reduced, which
+    // the torch.distributed contract permits).
+    TORCH_CHECK(tensors.size() == 1);
+    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM,
+                "uccl: only SUM reduce");
+    auto& t = tensors[0];
+    check(t);
+    hipStream_t s = cur_stream(comm_->device());
+    comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s);
+    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::REDUCE, s,
+                                         tensors);
+  }
+
+  c10::intrusive_ptr<c10d::Work> gather(
+      std::vector<std::vector<at::Tensor>>& outputs,
+      std::vector<at::Tensor>& inputs,
+      const c10d::GatherOptions& opts) override {
+    TORCH_CHECK(inputs.size() == 1);
+    auto& in = inputs[0];
+    check(in);
+    hipStream_t s = cur_stream(comm_->device());
+    at::Tensor flat = at::empty({getSize() * in.numel()}, in.options());
+    comm_->all_gather(flat.data_ptr(), in.data_ptr(),
+                      in.numel() * in.element_size(), Dtype::kU8, s);
+    std::vector<at::Tensor> results;
+    if (getRank() == opts.rootRank) {
+      TORCH_CHECK(outputs.size() == 1 &&
+                  static_cast<int>(outputs[0].size()) == getSize());
+      for (int r = 0; r < getSize(); ++r) {
+        outputs[0][r].view(-1).copy_(
+            flat.narrow(0, r * in.numel(), in.numel()), true);
+      }
+      results = outputs[0];
+    }
+    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::GATHER, s,
+                                         results);
+  }
+
+  c10::intrusive_ptr<c10d::Work> scatter(
+      std::vector<at::Tensor>& outputs,
+      std::vector<std::vector<at::Tensor>>& inputs,
+      const c10d::ScatterOptions& opts) override {
+    TORCH_CHECK(outputs.size() == 1);
+    auto& out = outputs[0];
+    check(out);
+    hipStream_t s = cur_stream(comm_->device());
+    at::Tensor flat = at::empty({getSize() * out.numel()}, out.options());
+    if (getRank() == opts.rootRank) {
+      TORCH_CHECK(inputs.size() == 1 &&
+                  static_cast<int>(inputs[0].size()) == getSize());
+      for (int r = 0; r < getSize(); ++r)
+        flat.narrow(0, r * out.numel(), out.numel())
+            .copy_(inputs[0][r].view(-1), true);
+    }
+    comm_->broadcast(flat.data_ptr(),
+                     flat.numel() * flat.element_size(), Dtype::kU8,
+                     static_cast<int>(opts.rootRank), s);
+    out.view(-1).copy_(
+        flat.narrow(0, getRank() * out.numel(), out.numel()), true);
+    return c10::make_intrusive<UcclWork>(
+        getRank(), c10d::OpType::SCATTER, s, std::vector<at::Tensor>{out});
+  }
+
   c10::intrusive_ptr<c10d::Work> send(std::vector<at::Tensor>& tensors,
                                       int dst, int) override {
     TORCH_CHECK(tensors.size() == 1);
