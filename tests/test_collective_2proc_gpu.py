@@ -25,7 +25,7 @@ WORKER = REPO / "tests" / "workers" / "collective_worker.py"
 import pytest
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_two_process_collectives(world):
     env_base = dict(os.environ)
     env_base.update({
@@ -35,7 +35,7 @@ def test_two_process_collectives(world):
         "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         "PYTHONPATH": str(REPO),
         "UCCL_TEST_LIGHT": "1" if world > 2 else "0",
-        "UCCL_TEST_ALARM": "420",
+        "UCCL_TEST_ALARM": "600",
     })
     procs = []
     for r in range(world):
@@ -48,7 +48,7 @@ def test_two_process_collectives(world):
     ok = True
     for p in procs:
         try:
-            out, _ = p.communicate(timeout=440)
+            out, _ = p.communicate(timeout=620)
         except subprocess.TimeoutExpired:
             p.kill()
             out, _ = p.communicate()

@@ -136,6 +136,21 @@ PYBIND11_MODULE(_C, m) {
            })
       .def("barrier", [](Communicator& c) {
         c.barrier(current_stream(c.device()));
+      })
+      .def("stats", [](Communicator& c) {
+        static const char* names[8] = {"all_reduce", "all_gather",
+                                       "reduce_scatter", "broadcast",
+                                       "all_to_all", "send", "recv",
+                                       "barrier"};
+        py::dict d;
+        auto const& st = c.stats();
+        for (int i = 0; i < 8; ++i) {
+          py::dict e;
+          e["calls"] = st[i].calls;
+          e["bytes"] = st[i].bytes;
+          d[names[i]] = e;
+        }
+        return d;
       });
 
   // --- P2P engine --------------------------------------------------------

@@ -110,6 +110,7 @@ CommView Communicator::view(uint64_t seq) const {
 
 void Communicator::all_reduce(void* data, size_t count, Dtype dt,
                               hipStream_t stream) {
+  tally(0, count * dtype_size(dt));
   if (world_ == 1) {
     // Sum over one rank is the identity. By default this is a no-op; with
     // UCCL_WORLD1_STAGED=1 we still run the full staged kernel path
@@ -158,6 +159,7 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
 
 void Communicator::all_gather(void* out, void const* in, size_t count_per_rank,
                               Dtype dt, hipStream_t stream) {
+  tally(1, count_per_rank * dtype_size(dt));
   size_t const es = dtype_size(dt);
   if (world_ == 1) {
     if (out != in) launch_copy(out, in, count_per_rank * es, stream);
@@ -189,6 +191,7 @@ void Communicator::all_gather(void* out, void const* in, size_t count_per_rank,
 void Communicator::reduce_scatter(void* out, void const* in,
                                   size_t count_per_rank, Dtype dt,
                                   hipStream_t stream) {
+  tally(2, count_per_rank * dtype_size(dt) * world_);
   size_t const es = dtype_size(dt);
   if (world_ == 1) {
     if (out != in) launch_copy(out, in, count_per_rank * es, stream);
@@ -216,6 +219,7 @@ void Communicator::reduce_scatter(void* out, void const* in,
 
 void Communicator::broadcast(void* data, size_t count, Dtype dt, int root,
                              hipStream_t stream) {
+  tally(3, count * dtype_size(dt));
   if (world_ == 1) return;
   UCCL_CHECK(connected_) << "connect() not called";
   size_t const es = dtype_size(dt);
@@ -237,6 +241,7 @@ void Communicator::broadcast(void* data, size_t count, Dtype dt, int root,
 
 void Communicator::all_to_all(void* out, void const* in, size_t count_per_rank,
                               Dtype dt, hipStream_t stream) {
+  tally(4, count_per_rank * dtype_size(dt) * world_);
   size_t const es = dtype_size(dt);
   if (world_ == 1) {
     if (out != in) launch_copy(out, in, count_per_rank * es, stream);
@@ -273,6 +278,7 @@ void Communicator::all_to_all(void* out, void const* in, size_t count_per_rank,
 
 void Communicator::send(void const* data, size_t bytes, int dst,
                         hipStream_t stream) {
+  tally(5, bytes);
   UCCL_CHECK(connected_ && dst != rank_ && dst >= 0 && dst < world_)
       << "bad send dst " << dst;
   size_t const slot_off = kP2POffset + static_cast<size_t>(dst) * kP2PSlotBytes;
@@ -294,6 +300,7 @@ void Communicator::send(void const* data, size_t bytes, int dst,
 
 void Communicator::recv(void* data, size_t bytes, int src,
                         hipStream_t stream) {
+  tally(6, bytes);
   UCCL_CHECK(connected_ && src != rank_ && src >= 0 && src < world_)
       << "bad recv src " << src;
   size_t const slot_off =
@@ -312,6 +319,7 @@ void Communicator::recv(void* data, size_t bytes, int src,
 }
 
 void Communicator::barrier(hipStream_t stream) {
+  tally(7, 0);
   if (world_ == 1) return;
   UCCL_CHECK(connected_) << "connect() not called";
   launch_barrier(view(next_seq()), stream);

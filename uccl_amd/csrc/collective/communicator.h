@@ -64,6 +64,14 @@ class Communicator {
   void recv(void* data, size_t bytes, int src, hipStream_t stream);
   void barrier(hipStream_t stream);
 
+  struct OpStats {
+    uint64_t calls = 0;
+    uint64_t bytes = 0;
+  };
+  // per-op host-side tallies (parity: the reference's #ifdef STATS
+  // per-engine counters, collective/rdma/transport.cc:1797)
+  std::array<OpStats, 8> const& stats() const { return stats_; }
+
  private:
   CommView view(uint64_t seq) const;
   uint64_t next_seq() {
@@ -82,6 +90,12 @@ class Communicator {
   std::array<uint64_t, kMaxRanks> send_seq_{};
   std::array<uint64_t, kMaxRanks> recv_seq_{};
   bool connected_ = false;
+
+  std::array<OpStats, 8> stats_{};
+  void tally(int op, size_t bytes) {
+    ++stats_[op].calls;
+    stats_[op].bytes += bytes;
+  }
 
   // thresholds (env-tunable)
   size_t ll_threshold_;
