@@ -4,6 +4,7 @@
 //   hipcc --offload-arch=gfx950 -O3 tools/probe_copy.hip -o gpurun_out/probe_copy
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 
 #include <cstdio>
 #include <vector>
@@ -49,9 +50,10 @@ __global__ void k_copy_nt(void* __restrict__ dst, void const* __restrict__ src,
   auto const* s = reinterpret_cast<V16 const*>(src);
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
+  using VU = unsigned __attribute__((ext_vector_type(4)));
   for (; i < n; i += stride) {
-    V16 v = __builtin_nontemporal_load(&s[i]);
-    __builtin_nontemporal_store(v, &d[i]);
+    VU v = __builtin_nontemporal_load(reinterpret_cast<VU const*>(s) + i);
+    __builtin_nontemporal_store(v, reinterpret_cast<VU*>(d) + i);
   }
 }
 
