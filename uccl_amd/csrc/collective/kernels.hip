@@ -79,7 +79,7 @@ __global__ void k_copy(void* __restrict__ dst, void const* __restrict__ src,
   auto const* s = reinterpret_cast<V16 const*>(src);
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
-  for (; i < nvec; i += stride) d[i] = s[i];
+  for (; i < nvec; i += stride) nt_store(&d[i], nt_load(&s[i]));
   // byte tail
   size_t const tail = bytes & 15;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
@@ -103,12 +103,14 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
   for (; i < nvec; i += stride) {
     AccumV16<T, 0> acc;
     // fixed rank order => bitwise-identical results on every rank
-    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv))[i]);
+    acc.init(nt_load(
+        reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv)) + i));
 #pragma unroll 7
     for (int p = 1; p < cv.world; ++p) {
-      acc.add(reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv))[i]);
+      acc.add(nt_load(
+          reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv)) + i));
     }
-    reinterpret_cast<V16*>(out)[i] = acc.pack();
+    nt_store(reinterpret_cast<V16*>(out) + i, acc.pack());
   }
   // scalar tail
   size_t const tail = count - nvec * vper;
@@ -133,11 +135,13 @@ __global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
     AccumV16<int, 0> acc;
-    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv))[i]);
+    acc.init(nt_load(
+        reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv)) + i));
     for (int p = 1; p < cv.world; ++p) {
-      acc.add(reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv))[i]);
+      acc.add(nt_load(
+          reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv)) + i));
     }
-    reinterpret_cast<V16*>(out)[i] = acc.pack();
+    nt_store(reinterpret_cast<V16*>(out) + i, acc.pack());
   }
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
@@ -172,17 +176,19 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   for (; i < end; i += stride) {
     AccumV16<T, 0> acc;
     // fixed rank order => bitwise-identical results on every rank
-    acc.init(reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv))[i]);
+    acc.init(nt_load(
+        reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv)) + i));
 #pragma unroll 7
     for (int p = 1; p < cv.world; ++p) {
-      acc.add(reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv))[i]);
+      acc.add(nt_load(
+          reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv)) + i));
     }
     V16 const r = acc.pack();
     // push to every rank's scratchB (spread across links; self included)
 #pragma unroll 8
     for (int k = 0; k < cv.world; ++k) {
       int const p = (cv.rank + k) % cv.world;
-      reinterpret_cast<V16*>(scratch_b(cv.peers[p], cv))[i] = r;
+      nt_store(reinterpret_cast<V16*>(scratch_b(cv.peers[p], cv)) + i, r);
     }
   }
 
@@ -211,7 +217,7 @@ __global__ void k_twoshot_copyout(CommView cv, void* __restrict__ out,
   size_t const stride = gridDim.x * blockDim.x;
   auto* d = reinterpret_cast<V16*>(out);
   auto const* s = reinterpret_cast<V16 const*>(src);
-  for (; i < nvec; i += stride) d[i] = s[i];
+  for (; i < nvec; i += stride) nt_store(&d[i], nt_load(&s[i]));
   size_t const tail = bytes & 15;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail)
     reinterpret_cast<char*>(out)[bytes - tail + threadIdx.x] =
@@ -333,7 +339,7 @@ __global__ void k_allgather_pull(CommView cv, void* __restrict__ out,
     auto const* s = reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv));
     auto* d = reinterpret_cast<V16*>(reinterpret_cast<char*>(out) +
                                      static_cast<size_t>(p) * chunk_bytes);
-    for (size_t j = i; j < nvec; j += stride) d[j] = s[j];
+    for (size_t j = i; j < nvec; j += stride) nt_store(&d[j], nt_load(&s[j]));
     if (tail && blockIdx.x == 0 && threadIdx.x < tail)
       reinterpret_cast<char*>(d)[chunk_bytes - tail + threadIdx.x] =
           reinterpret_cast<char const*>(s)[chunk_bytes - tail + threadIdx.x];
@@ -351,14 +357,15 @@ __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
     AccumV16<T, 0> acc;
-    acc.init(reinterpret_cast<V16 const*>(
+    acc.init(nt_load(reinterpret_cast<V16 const*>(
         reinterpret_cast<T const*>(scratch_a(cv.peers[0], cv)) +
-        elem_off)[i]);
+        elem_off) + i));
     for (int p = 1; p < cv.world; ++p) {
-      acc.add(reinterpret_cast<V16 const*>(
-          reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv)) + elem_off)[i]);
+      acc.add(nt_load(reinterpret_cast<V16 const*>(
+          reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv)) +
+          elem_off) + i));
     }
-    reinterpret_cast<V16*>(out)[i] = acc.pack();
+    nt_store(reinterpret_cast<V16*>(out) + i, acc.pack());
   }
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
@@ -380,7 +387,7 @@ __global__ void k_broadcast_pull(CommView cv, int root, void* __restrict__ out,
   size_t const nvec = bytes / 16;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
-  for (; i < nvec; i += stride) d[i] = s[i];
+  for (; i < nvec; i += stride) nt_store(&d[i], nt_load(&s[i]));
   size_t const tail = bytes & 15;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail)
     reinterpret_cast<char*>(out)[bytes - tail + threadIdx.x] =
@@ -401,7 +408,7 @@ __global__ void k_alltoall_pull(CommView cv, void* __restrict__ out,
         scratch_a(cv.peers[p], cv) + static_cast<size_t>(cv.rank) * chunk_bytes);
     auto* d = reinterpret_cast<V16*>(reinterpret_cast<char*>(out) +
                                      static_cast<size_t>(p) * chunk_bytes);
-    for (size_t j = i; j < nvec; j += stride) d[j] = s[j];
+    for (size_t j = i; j < nvec; j += stride) nt_store(&d[j], nt_load(&s[j]));
     if (tail && blockIdx.x == 0 && threadIdx.x < tail)
       reinterpret_cast<char*>(d)[chunk_bytes - tail + threadIdx.x] =
           reinterpret_cast<char const*>(s)[chunk_bytes - tail + threadIdx.x];
@@ -433,7 +440,7 @@ __global__ void k_copy_from_peer(CommView cv, int src, size_t src_off,
   size_t const nvec = bytes / 16;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
-  for (; i < nvec; i += stride) d[i] = s[i];
+  for (; i < nvec; i += stride) nt_store(&d[i], nt_load(&s[i]));
   size_t const tail = bytes & 15;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail)
     reinterpret_cast<char*>(dst)[bytes - tail + threadIdx.x] =
@@ -445,9 +452,9 @@ __global__ void k_copy_from_peer(CommView cv, int src, size_t src_off,
 // ---------------------------------------------------------------------------
 
 static inline int grid_for(size_t bytes) {
-  // memory-bound: cap at ~2048 workgroups (256 CUs × 8), grid-stride rest
+  // memory-bound: probe_copy shows peak HBM bw at ~8192 workgroups
   size_t const want = (bytes / 16 + 255) / 256;
-  size_t const g = want < 8 ? 8 : (want > 2048 ? 2048 : want);
+  size_t const g = want < 8 ? 8 : (want > 8192 ? 8192 : want);
   return static_cast<int>(g);
 }
 

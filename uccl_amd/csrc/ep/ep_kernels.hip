@@ -74,7 +74,8 @@ __device__ inline void block_copy(char* __restrict__ dst,
   size_t const nvec = bytes / 16;
   auto* d = reinterpret_cast<V16*>(dst);
   auto const* s = reinterpret_cast<V16 const*>(src);
-  for (size_t i = threadIdx.x; i < nvec; i += blockDim.x) d[i] = s[i];
+  for (size_t i = threadIdx.x; i < nvec; i += blockDim.x)
+    nt_store(&d[i], nt_load(&s[i]));
   size_t const tail = bytes & 15;
   if (tail && threadIdx.x < tail)
     dst[bytes - tail + threadIdx.x] = src[bytes - tail + threadIdx.x];
