@@ -61,7 +61,11 @@ def build_plugin(verbose: bool = False) -> Path:
     deps = [src, harness_src] + list((CSRC / "plugin").glob("*.h")) + [
         CSRC / "core" / "net.h", CSRC / "core" / "log.h",
         CSRC / "core" / "env.h"]
-    if not _stale(target, deps) and not _stale(harness, deps):
+    capi = plugdir / "libuccl_p2p.so"
+    capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
+                 CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h"]
+    if (not _stale(target, deps) and not _stale(harness, deps)
+            and not _stale(capi, capi_srcs + deps)):
         return target
     import subprocess as sp
 
@@ -83,6 +87,18 @@ def build_plugin(verbose: bool = False) -> Path:
             alias.symlink_to(target.name)
         except FileExistsError:
             pass
+    # flat C API lib for NIXL-style integrators (reference: p2p/uccl_engine.h)
+    if _stale(capi, capi_srcs + deps):
+        cc_srcs = [x for x in capi_srcs if x.suffix == ".cpp"]
+        cmd = [HIPCC, "-O2", "-std=c++17", "-fPIC", "-shared",
+               f"--offload-arch={GPU_ARCH}"] + [str(x) for x in cc_srcs] + [
+               "-o", str(capi), "-pthread", f"-L{ROCM}/lib", "-lamdhip64",
+               f"-Wl,-rpath,{ROCM}/lib"]
+        if verbose:
+            print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)
+        r = sp.run(cmd, stdout=sp.PIPE, stderr=sp.STDOUT)
+        if r.returncode != 0:
+            raise RuntimeError(f"c_api build failed:\n{r.stdout.decode()}")
     return target
 
 
