@@ -26,8 +26,18 @@ def test_c_api_loopback():
     lib.uccl_engine_advertise.argtypes = [ctypes.c_void_p, ctypes.c_uint64,
                                           ctypes.c_uint64, ctypes.c_uint64,
                                           ctypes.c_void_p, ctypes.c_size_t]
-    for fn in ("send", "recv", "write", "read"):
-        getattr(lib, f"uccl_engine_{fn}").restype = ctypes.c_int
+    for fn in ("send", "recv"):
+        f = getattr(lib, f"uccl_engine_{fn}")
+        f.restype = ctypes.c_int
+        f.argtypes = [ctypes.c_void_p, ctypes.c_uint64, ctypes.c_void_p,
+                      ctypes.c_size_t, ctypes.c_int]
+    for fn in ("write", "read"):
+        f = getattr(lib, f"uccl_engine_{fn}")
+        f.restype = ctypes.c_int
+        f.argtypes = [ctypes.c_void_p, ctypes.c_uint64, ctypes.c_void_p,
+                      ctypes.c_size_t, ctypes.c_int, ctypes.c_void_p,
+                      ctypes.c_size_t]
+    lib.uccl_engine_destroy.argtypes = [ctypes.c_void_p]
 
     a = lib.uccl_engine_create(-1, 1)
     b = lib.uccl_engine_create(-1, 1)
