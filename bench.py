@@ -37,6 +37,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--mbytes", type=int, default=256,
                    help="per-GPU buffer size in MiB (bf16)")
+    p.add_argument("--symmetric", action="store_true",
+                   help="use the zero-copy symmetric-tensor path")
     return p.parse_args()
 
 
@@ -64,13 +66,20 @@ def main():
 
     torch.cuda.set_device(local_rank % torch.cuda.device_count())
 
+    if args.symmetric:  # must be set before the Communicator allocates
+        os.environ.setdefault("UCCL_SYM_USER_MB", str(args.mbytes + 64))
+
     import uccl_amd.collective as ucol
 
     comm = ucol.init()
 
     nbytes = args.mbytes * (1 << 20)
     count = nbytes // 2  # bf16
-    t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
+    if args.symmetric:
+        t = comm.symmetric_tensor([count], torch.bfloat16)
+        t.copy_(torch.randn(count, dtype=torch.bfloat16))
+    else:
+        t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
 
     def barrier():
         torch.cuda.synchronize()
@@ -133,7 +142,10 @@ def main():
                 "global_batch": None,
                 "seq_len": None,
                 "parallelism": f"xgmi-fullmesh-{world}gpu",
-                "engine": "uccl_amd twoshot RS+AG push kernels",
+                "engine": ("uccl_amd symmetric zero-copy twoshot"
+                           if args.symmetric else
+                           "uccl_amd twoshot RS+AG push kernels"),
+                "symmetric": args.symmetric,
                 "n1_value_is_local_algbw": world == 1,
             },
         }))

@@ -194,6 +194,27 @@ def main():
         check(f"stress[{i}]", x, expected_sum(world, 257, torch.float32,
                                               seed), 0.0)
 
+    # --- symmetric (zero-copy) allreduce -----------------------------------
+    for count in ([100000, 3 << 20] if not light else [100000]):
+        for dtype in (torch.float32, torch.bfloat16):
+            seed += 1
+            st = comm.symmetric_tensor([count], dtype)
+            st.copy_(make_input(rank, count, dtype, seed))
+            assert comm.is_symmetric(st)
+            comm.all_reduce(st)
+            torch.cuda.synchronize()
+            check(f"allreduce[sym,{dtype},{count}]", st,
+                  expected_sum(world, count, dtype, seed), tol[dtype])
+    # repeated in-place reuse of the same symmetric tensor
+    seed += 1
+    st = comm.symmetric_tensor([4096], torch.float32)
+    st.copy_(make_input(rank, 4096, torch.float32, seed))
+    comm.all_reduce(st)
+    comm.all_reduce(st)  # sum of sums
+    torch.cuda.synchronize()
+    once = expected_sum(world, 4096, torch.float32, seed).float()
+    check("allreduce[sym,repeat]", st, (once * world).to(torch.float32), 1e-4)
+
     # --- fp8 (OCP e4m3) allreduce across LL / oneshot / twoshot paths ------
     if hasattr(torch, "float8_e4m3fn"):
         for count in ((1000, 300000) if light else (1000, 300000, 5 << 20)):

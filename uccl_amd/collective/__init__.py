@@ -68,6 +68,19 @@ class Communicator:
     def barrier(self) -> None:
         self._c.barrier()
 
+    # -- symmetric (zero-copy) tensors --------------------------------------
+    def symmetric_tensor(self, sizes, dtype=torch.bfloat16) -> torch.Tensor:
+        """Allocate a tensor in the registered symmetric region. Every rank
+        must make the same symmetric allocations in the same order.
+        Collectives on these tensors skip all staging copies (the kernels
+        read/write every rank's buffer directly over xGMI)."""
+        if isinstance(sizes, int):
+            sizes = [sizes]
+        return self._c.symmetric_tensor(list(sizes), dtype)
+
+    def is_symmetric(self, t: torch.Tensor) -> bool:
+        return self._c.is_symmetric(t)
+
 
 def init(group=None, device: Optional[int] = None,
          heap_bytes: int = 0) -> Communicator:

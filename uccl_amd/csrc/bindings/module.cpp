@@ -137,6 +137,27 @@ PYBIND11_MODULE(_C, m) {
       .def("barrier", [](Communicator& c) {
         c.barrier(current_stream(c.device()));
       })
+      .def("symmetric_tensor",
+           [](Communicator& c, std::vector<int64_t> sizes,
+              py::object dtype_obj) {
+             auto dtype = torch::python::detail::py_object_to_dtype(
+                 dtype_obj);
+             int64_t numel = 1;
+             for (auto s : sizes) numel *= s;
+             size_t const bytes =
+                 static_cast<size_t>(numel) *
+                 c10::elementSize(dtype);
+             size_t const off = c.sym_alloc(bytes);
+             return at::from_blob(
+                 static_cast<char*>(c.heap_base()) + off, sizes,
+                 at::TensorOptions().dtype(dtype).device(
+                     at::Device(at::kCUDA, c.device())));
+           },
+           py::arg("sizes"), py::arg("dtype"))
+      .def("is_symmetric",
+           [](Communicator& c, at::Tensor t) {
+             return c.is_symmetric_ptr(t.data_ptr());
+           })
       .def("stats", [](Communicator& c) {
         static const char* names[8] = {"all_reduce", "all_gather",
                                        "reduce_scatter", "broadcast",

@@ -20,15 +20,20 @@ Communicator::Communicator(int rank, int world, int device, size_t heap_bytes)
       << "world=" << world << " (intranode engine supports up to "
       << kMaxRanks << ")";
   UCCL_CHECK(rank >= 0 && rank < world) << "bad rank " << rank;
-  heap_bytes_ = heap_bytes ? heap_bytes
-                           : static_cast<size_t>(
-                                 env_int("UCCL_SYM_HEAP_MB", 1184)) *
-                                 (1 << 20);
-  UCCL_CHECK(heap_bytes_ > kScratchAOffset + (4 << 20))
-      << "heap too small: " << heap_bytes_;
+  size_t const base_bytes =
+      heap_bytes ? heap_bytes
+                 : static_cast<size_t>(env_int("UCCL_SYM_HEAP_MB", 1184)) *
+                       (1 << 20);
+  UCCL_CHECK(base_bytes > kScratchAOffset + (4 << 20))
+      << "heap too small: " << base_bytes;
   // per-parity capacity: the A/B scratch regions are each split into two
   // parity halves (see layout.h CommView comment)
-  scratch_cap_ = (scratch_capacity(heap_bytes_) / 2) & ~size_t(255);
+  scratch_cap_ = (scratch_capacity(base_bytes) / 2) & ~size_t(255);
+  // symmetric user region (zero-copy collectives) appended after scratch
+  user_cap_ = static_cast<size_t>(env_int("UCCL_SYM_USER_MB", 512))
+              << 20;
+  user_off_ = kScratchAOffset + 4 * scratch_cap_;
+  heap_bytes_ = user_off_ + user_cap_;
   ll_threshold_ = env_int("UCCL_LL_THRESHOLD", 32 * 1024);
   if (ll_threshold_ > kLLMaxMsgBytes) ll_threshold_ = kLLMaxMsgBytes;
   oneshot_threshold_ = env_int("UCCL_ONESHOT_THRESHOLD", 2 * 1024 * 1024);
@@ -94,6 +99,15 @@ void Communicator::connect(const std::vector<std::string>& all_handles) {
   connected_ = true;
 }
 
+size_t Communicator::sym_alloc(size_t bytes) {
+  size_t const off = (user_bump_ + 255) & ~size_t(255);
+  UCCL_CHECK(off + bytes <= user_cap_)
+      << "symmetric region exhausted (" << (user_cap_ >> 20)
+      << "MB; raise UCCL_SYM_USER_MB)";
+  user_bump_ = off + bytes;
+  return user_off_ + off;
+}
+
 CommView Communicator::view(uint64_t seq) const {
   CommView cv{};
   cv.rank = rank_;
@@ -137,6 +151,27 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
 
   if (bytes <= ll_threshold_) {
     launch_ll_allreduce(view(next_seq()), data, data, count, dt, stream);
+    return;
+  }
+  if (is_symmetric_ptr(data)) {
+    // zero-copy path: reduce straight out of every rank's symmetric user
+    // region, push results back in place (no staging copies at all)
+    size_t const shard_bytes = (count * es + world_ - 1) / world_ + 256;
+    UCCL_CHECK(shard_bytes <= scratch_cap_)
+        << "symmetric allreduce shard exceeds scratch";
+    size_t const uoff =
+        static_cast<char const*>(data) - static_cast<char*>(heap_);
+    CommView const cv = view(next_seq());
+    CommView const cv2 = view(next_seq());
+    // entry barrier: every rank's input is produced + published
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
+    launch_twoshot_sym_rs(cv, uoff, count, dt, stream);
+    // mid barrier: nobody reads user inputs any more -> pushes may land
+    launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
+    launch_twoshot_sym_push(cv, uoff, count, dt, stream);
+    // exit barrier: all pushes into my buffer are visible before my
+    // stream continues (the caller reads the result right after)
+    launch_signal_wait(cv2, cv2.seq, all_mask(world_), stream);
     return;
   }
   bool const oneshot = bytes <= oneshot_threshold_;

@@ -50,6 +50,21 @@ class Communicator {
   int device() const { return device_; }
   size_t scratch_capacity_bytes() const { return scratch_cap_; }
 
+  // --- symmetric (zero-copy) user region ---------------------------------
+  // Bump-allocates `bytes` in the registered symmetric region; every rank
+  // must perform the same allocations in the same order (standard
+  // symmetric-heap contract). Collectives on tensors inside this region
+  // skip all staging copies.
+  size_t sym_alloc(size_t bytes);
+  void* heap_base() const { return heap_; }
+  size_t user_region_offset() const { return user_off_; }
+  size_t user_region_capacity() const { return user_cap_; }
+  bool is_symmetric_ptr(void const* p) const {
+    auto const u = reinterpret_cast<uintptr_t>(p);
+    auto const b = reinterpret_cast<uintptr_t>(heap_);
+    return u >= b + user_off_ && u < b + heap_bytes_;
+  }
+
   // All ops are asynchronous on `stream` and in-place where natural.
   void all_reduce(void* data, size_t count, Dtype dt, hipStream_t stream);
   void all_gather(void* out, void const* in, size_t count_per_rank, Dtype dt,
@@ -86,6 +101,9 @@ class Communicator {
   std::array<void*, kMaxRanks> peers_{};
   std::array<bool, kMaxRanks> ipc_opened_{};
   uint64_t seq_ = 2;  // flag regions start zeroed; first live seq must be >0
+  size_t user_off_ = 0;   // symmetric user region offset in the heap
+  size_t user_cap_ = 0;
+  size_t user_bump_ = 0;
   // per-destination send / per-source recv sequence counters (p2p channel)
   std::array<uint64_t, kMaxRanks> send_seq_{};
   std::array<uint64_t, kMaxRanks> recv_seq_{};

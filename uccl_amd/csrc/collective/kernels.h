@@ -29,6 +29,10 @@ void launch_twoshot_rs_push(const CommView& cv, size_t count, Dtype dt,
                             hipStream_t s);
 void launch_twoshot_copyout(const CommView& cv, void* out, size_t bytes,
                             hipStream_t s);
+void launch_twoshot_sym_rs(const CommView& cv, size_t uoff, size_t count,
+                           Dtype dt, hipStream_t s);
+void launch_twoshot_sym_push(const CommView& cv, size_t uoff, size_t count,
+                             Dtype dt, hipStream_t s);
 void launch_ll_allreduce(const CommView& cv, void const* in, void* out,
                          size_t count, Dtype dt, hipStream_t s);
 void launch_allgather_pull(const CommView& cv, void* out, size_t chunk_bytes,

@@ -207,6 +207,82 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   }
 }
 
+// ---------------------------------------------------------------------------
+// Symmetric (zero-copy) two-shot: the tensor lives in every rank's
+// registered symmetric user region at the same offset, so there is no
+// copy-in/copy-out at all. Phase 1 reduces my 1/N shard straight from all
+// peers' user buffers into my parity scratchB; after the flag round
+// (nobody reads user inputs any more), phase 2 pushes the reduced shard
+// into every rank's user buffer in place.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void k_twoshot_sym_rs(CommView cv, size_t uoff, size_t count) {
+  size_t const vper = 16 / sizeof(T);
+  size_t const nvec = count / vper;
+  size_t const shard = (nvec + cv.world - 1) / cv.world;
+  size_t const beg = cv.rank * shard;
+  size_t const end = min(beg + shard, nvec);
+  size_t i = beg + blockIdx.x * blockDim.x + threadIdx.x;
+  size_t const stride = gridDim.x * blockDim.x;
+  auto* sb = reinterpret_cast<V16*>(scratch_b(cv.peers[cv.rank], cv));
+  for (; i < end; i += stride) {
+    AccumV16<T, 0> acc;
+    acc.init(nt_load(reinterpret_cast<V16 const*>(
+                 static_cast<char*>(cv.peers[0]) + uoff) + i));
+#pragma unroll 7
+    for (int p = 1; p < cv.world; ++p) {
+      acc.add(nt_load(reinterpret_cast<V16 const*>(
+                  static_cast<char*>(cv.peers[p]) + uoff) + i));
+    }
+    nt_store(&sb[i - beg], acc.pack());
+  }
+  size_t const tail = count - nvec * vper;
+  if (tail && cv.rank == cv.world - 1 && blockIdx.x == 0 &&
+      threadIdx.x < tail) {
+    size_t const j = nvec * vper + threadIdx.x;
+    float a = 0.f;
+    for (int p = 0; p < cv.world; ++p)
+      a += static_cast<float>(reinterpret_cast<T const*>(
+          static_cast<char*>(cv.peers[p]) + uoff)[j]);
+    // stash tail results after the vector shard in scratchB
+    reinterpret_cast<T*>(sb)[(end - beg) * vper + threadIdx.x] =
+        static_cast<T>(a);
+  }
+}
+
+template <typename T>
+__global__ void k_twoshot_sym_push(CommView cv, size_t uoff, size_t count) {
+  size_t const vper = 16 / sizeof(T);
+  size_t const nvec = count / vper;
+  size_t const shard = (nvec + cv.world - 1) / cv.world;
+  size_t const beg = cv.rank * shard;
+  size_t const end = min(beg + shard, nvec);
+  size_t i = beg + blockIdx.x * blockDim.x + threadIdx.x;
+  size_t const stride = gridDim.x * blockDim.x;
+  auto const* sb =
+      reinterpret_cast<V16 const*>(scratch_b(cv.peers[cv.rank], cv));
+  for (; i < end; i += stride) {
+    V16 const r = sb[i - beg];
+#pragma unroll 8
+    for (int k = 0; k < cv.world; ++k) {
+      int const p = (cv.rank + k) % cv.world;
+      nt_store(reinterpret_cast<V16*>(
+                   static_cast<char*>(cv.peers[p]) + uoff) + i,
+               r);
+    }
+  }
+  size_t const tail = count - nvec * vper;
+  if (tail && cv.rank == cv.world - 1 && blockIdx.x == 0 &&
+      threadIdx.x < tail) {
+    size_t const j = nvec * vper + threadIdx.x;
+    T const r = reinterpret_cast<T const*>(sb)[(end - beg) * vper +
+                                               threadIdx.x];
+    for (int p = 0; p < cv.world; ++p)
+      reinterpret_cast<T*>(static_cast<char*>(cv.peers[p]) + uoff)[j] = r;
+  }
+}
+
 // Phase 2 kernel — signal seq+1 (my pushes are visible: dispatch boundary),
 // wait for everyone's pushes, copy assembled scratchB to the output.
 __global__ void k_twoshot_copyout(CommView cv, void* __restrict__ out,
@@ -509,6 +585,32 @@ void launch_oneshot_allreduce(const CommView& cv, void* out, size_t count,
 void launch_twoshot_rs_push(const CommView& cv, size_t count, Dtype dt,
                             hipStream_t s) {
   DT_DISPATCH(dt, l_twoshot_rs, cv, count, s);
+}
+
+template <typename T>
+static void l_sym_rs(const CommView& cv, size_t uoff, size_t count,
+                     hipStream_t s) {
+  k_twoshot_sym_rs<T>
+      <<<grid_for(count * sizeof(T) / (cv.world ? cv.world : 1)), 256, 0,
+         s>>>(cv, uoff, count);
+}
+
+template <typename T>
+static void l_sym_push(const CommView& cv, size_t uoff, size_t count,
+                       hipStream_t s) {
+  k_twoshot_sym_push<T>
+      <<<grid_for(count * sizeof(T) / (cv.world ? cv.world : 1)), 256, 0,
+         s>>>(cv, uoff, count);
+}
+
+void launch_twoshot_sym_rs(const CommView& cv, size_t uoff, size_t count,
+                           Dtype dt, hipStream_t s) {
+  DT_DISPATCH(dt, l_sym_rs, cv, uoff, count, s);
+}
+
+void launch_twoshot_sym_push(const CommView& cv, size_t uoff, size_t count,
+                             Dtype dt, hipStream_t s) {
+  DT_DISPATCH(dt, l_sym_push, cv, uoff, count, s);
 }
 
 void launch_twoshot_copyout(const CommView& cv, void* out, size_t bytes,
