@@ -189,6 +189,8 @@ PYBIND11_MODULE(_C, m) {
       .def("accept", &Endpoint::accept,
            py::call_guard<py::gil_scoped_release>())
       .def("num_conns", &Endpoint::num_conns)
+      .def("close_conn", &Endpoint::close_conn,
+           py::call_guard<py::gil_scoped_release>())
       .def("reg",
            [](Endpoint& e, at::Tensor t) {
              TORCH_CHECK(t.is_contiguous());

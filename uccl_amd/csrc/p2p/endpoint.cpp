@@ -634,6 +634,26 @@ bool Endpoint::poll_async(uint64_t xfer_id) {
   return false;
 }
 
+void Endpoint::close_conn(uint64_t conn_id) {
+  std::shared_ptr<Conn> c;
+  {
+    std::lock_guard<std::mutex> g(conn_mu_);
+    auto it = conns_.find(conn_id);
+    if (it == conns_.end()) return;
+    c = it->second;
+    conns_.erase(it);
+  }
+  c->alive = false;
+  ::shutdown(c->fd, SHUT_RDWR);
+  c->rx_cv.notify_all();
+  c->tok_cv.notify_all();
+  if (c->rx.joinable()) c->rx.join();
+  ::close(c->fd);
+  std::lock_guard<std::mutex> g2(c->ipc_mu);
+  for (auto& [k, p] : c->ipc_cache) (void)hipIpcCloseMemHandle(p);
+  c->ipc_cache.clear();
+}
+
 int Endpoint::num_conns() {
   std::lock_guard<std::mutex> g(conn_mu_);
   return static_cast<int>(conns_.size());

@@ -82,6 +82,9 @@ class Endpoint {
   bool poll_async(uint64_t xfer_id);  // true once complete (then forgets it)
 
   int num_conns();
+  // tear down one connection (reference parity: p2p remove_remote_endpoint,
+  // p2p/engine.cc:2208); pending ops on it fail
+  void close_conn(uint64_t conn_id);
 
  private:
   struct Conn;

@@ -78,6 +78,7 @@ struct ChunkTx {
   uint64_t send_ts = 0;
   uint32_t attempts = 0;
   int dupacks = 0;
+  int rto_count = 0;  // consecutive RTO hits (abort threshold)
 };
 
 struct MsgRx {
@@ -105,6 +106,8 @@ struct TransportEndpoint::Flow {
   uint32_t tx_cum = 0;       // lowest unacked csn
   uint32_t last_cum = 0;     // cum of the previous ack (hole detection)
   int hole_dupacks = 0;      // acks with stalled cum + new SACKs above
+
+  bool failed = false;  // flow marked dead after RTO abort threshold
 
   // --- RX direction ---
   uint32_t rx_cum = 0;  // all csn < rx_cum received
@@ -250,6 +253,7 @@ struct TransportEndpoint::Impl {
   void ack_chunk(Flow& f, uint32_t csn) {
     auto it = f.inflight.find(csn);
     if (it == f.inflight.end()) return;
+    it->second.rto_count = 0;
     it->second.msg->acked_bytes += it->second.len;
     auto& m = *it->second.msg;
     if (m.acked_bytes >= m.bytes && !m.done) {
@@ -334,14 +338,28 @@ struct TransportEndpoint::Impl {
   void rto_scan() {
     static uint64_t const rto_ns =
         static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 20000)) * 1000;
+    // flow-failure detection, parity with the reference's RTO abort
+    // threshold (kRTOAbortThreshold=50, transport_config.h:202 +
+    // mark_flow_timeout): a chunk that hits RTO this many times in a row
+    // marks the flow failed and fails its blocked senders/receivers.
+    static int const abort_thres =
+        static_cast<int>(env_int("UCCL_TP_RTO_ABORT", 50));
     uint64_t const now = now_ns();
     for (auto& [fid, fp] : flows) {
       Flow& f = *fp;
+      if (f.failed) continue;
       for (auto& [csn, c] : f.inflight) {
         uint64_t const rto =
             std::max<uint64_t>(rto_ns, 4ull * 1000 *
                                            static_cast<uint64_t>(f.srtt_us));
         if (c.send_ts && now - c.send_ts > rto) {
+          if (++c.rto_count >= abort_thres) {
+            UCCL_LOG_ERROR << "flow " << fid << " csn " << csn
+                           << " exceeded RTO abort threshold; marking dead";
+            f.failed = true;
+            cv.notify_all();
+            break;
+          }
           ++st.rto_retransmits;
           f.cwnd = std::max(2.0, f.cwnd / 2);
           send_chunk(f, csn, c);
@@ -561,7 +579,9 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
   }
   impl_->wake();
   std::unique_lock<std::mutex> lk(impl_->mu);
-  impl_->cv.wait(lk, [&] { return m->done || impl_->stop; });
+  Flow& fl = *impl_->flows[flow];
+  impl_->cv.wait(lk, [&] { return m->done || impl_->stop || fl.failed; });
+  if (fl.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (!m->done) throw std::runtime_error("transport closed during send");
 }
 
@@ -587,8 +607,9 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
     auto it = f.rxmsgs.find(msg_id);
     return (it != f.rxmsgs.end() && it->second.known &&
             it->second.recv_bytes >= it->second.bytes) ||
-           impl_->stop;
+           impl_->stop || f.failed;
   });
+  if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (impl_->stop) throw std::runtime_error("transport closed during recv");
   f.rxmsgs.erase(msg_id);
 }
