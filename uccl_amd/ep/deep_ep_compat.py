@@ -1,0 +1,110 @@
+"""DeepEP-shaped API over uccl_amd.ep — signature-level compatibility with
+`deep_ep.Buffer`'s low-latency path (the reference replicates this surface
+in ep/bench/buffer.py; here it is a thin shim over the native engine so
+existing DeepEP callers can switch imports).
+
+    from uccl_amd.ep.deep_ep_compat import Buffer
+    buf = Buffer(group, num_rdma_bytes=0, low_latency_mode=True, ...)
+    recv_x, recv_count, handle, event, hook = buf.low_latency_dispatch(
+        x, topk_idx, num_max_dispatch_tokens_per_rank, num_experts)
+    combined, event, hook = buf.low_latency_combine(
+        recv_x, topk_idx, topk_weights, handle)
+
+Differences from NVIDIA DeepEP (documented, not hidden):
+  - events are torch.cuda.Event objects recorded on the current stream
+    (ops are stream-ordered; there is no proxy hook to defer, so `hook`
+    is a no-op callable)
+  - fp8 dispatch returns (payload, scales) like DeepEP's
+    use_fp8=True path
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Optional, Tuple
+
+import torch
+
+from uccl_amd.ep import Buffer as _NativeBuffer
+from uccl_amd.ep import get_dispatch_layout
+
+
+class _Handle:
+    def __init__(self, topk_idx, num_tokens):
+        self.topk_idx = topk_idx
+        self.num_tokens = num_tokens
+
+
+def _event():
+    ev = torch.cuda.Event()
+    ev.record()
+    return ev
+
+
+def _noop_hook():
+    return None
+
+
+class Buffer:
+    def __init__(self, group=None, num_nvl_bytes: int = 0,
+                 num_rdma_bytes: int = 0, low_latency_mode: bool = True,
+                 num_qps_per_rank: int = 1, **kwargs):
+        # geometry is taken lazily from the first dispatch
+        self._group = group
+        self._native: Optional[_NativeBuffer] = None
+        self._cfg = None
+
+    # -- DeepEP static helpers ----------------------------------------------
+    @staticmethod
+    def get_dispatch_layout(topk_idx: torch.Tensor, num_experts: int,
+                            previous_event=None, async_finish=False,
+                            allocate_on_comm_stream=False):
+        import torch.distributed as dist
+
+        num_ranks = (dist.get_world_size()
+                     if dist.is_available() and dist.is_initialized() else 1)
+        npr, npe, in_rank = get_dispatch_layout(topk_idx, num_experts,
+                                                num_ranks)
+        return npr, None, npe, in_rank, _event()
+
+    def _ensure(self, hidden: int, max_tokens: int, num_experts: int,
+                topk: int, dtype, use_fp8: bool):
+        cfg = (hidden, max_tokens, num_experts, topk, dtype, use_fp8)
+        if self._cfg != cfg:
+            self._native = _NativeBuffer(
+                group=self._group, num_experts=num_experts, topk=topk,
+                hidden=hidden, max_tokens=max_tokens, dtype=dtype,
+                use_fp8=use_fp8)
+            self._cfg = cfg
+        return self._native
+
+    # -- low latency path ----------------------------------------------------
+    def low_latency_dispatch(self, x: torch.Tensor, topk_idx: torch.Tensor,
+                             num_max_dispatch_tokens_per_rank: int,
+                             num_experts: int, use_fp8: bool = False,
+                             async_finish: bool = False,
+                             return_recv_hook: bool = False):
+        nb = self._ensure(x.shape[1], num_max_dispatch_tokens_per_rank,
+                          num_experts, topk_idx.shape[1], x.dtype, use_fp8)
+        out = nb.dispatch(x, topk_idx)
+        handle = _Handle(topk_idx, x.shape[0])
+        if use_fp8:
+            recv_x, counts, scales = out
+            packed = (recv_x, scales)
+        else:
+            recv_x, counts = out
+            packed = recv_x
+        # DeepEP's packed_recv_count is per-local-expert
+        recv_count = counts.sum(dim=1)
+        return packed, recv_count, handle, _event(), _noop_hook
+
+    def low_latency_combine(self, x: torch.Tensor, topk_idx: torch.Tensor,
+                            topk_weights: torch.Tensor, handle: _Handle,
+                            async_finish: bool = False,
+                            return_recv_hook: bool = False):
+        assert self._native is not None, "combine before dispatch"
+        combined = self._native.combine(x, topk_idx, topk_weights)
+        return combined, _event(), _noop_hook
+
+    # convenience aliases matching the high-throughput entry points
+    dispatch = low_latency_dispatch
+    combine = low_latency_combine
