@@ -119,6 +119,38 @@ def main():
         make_input(r, n, torch.float32, seed).cpu() for r in range(world)])
     check("allgather", out, want, 0.0)
 
+    # --- chunked allgather (count > per-parity scratch => strided pulls) ---
+    if not light:
+        seed += 1
+        n = 3 << 20  # 12MB fp32 > 192MB-heap parity scratch? cap=(192-48)/4=36MB; force via more
+        n = 12 << 20  # 48MB fp32 > 36MB parity scratch -> chunked path
+        mine = make_input(rank, n, torch.float32, seed)
+        out = torch.empty(world * n, dtype=torch.float32, device="cuda")
+        comm.all_gather(out, mine)
+        torch.cuda.synchronize()
+        for r in range(world):
+            want = make_input(r, n, torch.float32, seed).cpu()
+            got = out[r * n:(r + 1) * n].cpu()
+            assert torch.equal(got, want), f"chunked allgather rank {r}"
+        print(f"[rank {rank}] allgather[chunked] OK", flush=True)
+        del out, mine
+
+    # --- chunked alltoall ---------------------------------------------------
+    if not light:
+        seed += 1
+        per = (10 << 20) // world  # total 40MB fp32 > cap/world chunks
+        inp = make_input(rank, per * world, torch.float32, seed)
+        out = torch.empty_like(inp)
+        comm.all_to_all(out, inp)
+        torch.cuda.synchronize()
+        want = torch.cat([
+            make_input(r, per * world, torch.float32,
+                       seed)[rank * per:(rank + 1) * per].cpu()
+            for r in range(world)
+        ])
+        check("alltoall[chunked]", out, want, 0.0)
+        del out, inp
+
     # --- reduce_scatter -----------------------------------------------------
     seed += 1
     per = 4096

@@ -72,6 +72,14 @@ def main():
     torch.cuda.synchronize()
     assert torch.allclose(sout, torch.full_like(sout, float(100 + rank)))
 
+    # subgroup: each rank in its own single-member group (creator is
+    # invoked per group with a prefixed store)
+    solo = dist.new_group([rank], backend="uccl")
+    t1 = torch.full((16,), float(rank), device="cuda")
+    dist.all_reduce(t1, group=solo)
+    torch.cuda.synchronize()
+    assert torch.allclose(t1, torch.full_like(t1, float(rank)))
+
     dist.barrier()
 
     # DDP end-to-end: tiny model, grads must match single-process reference
