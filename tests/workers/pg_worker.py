@@ -73,8 +73,13 @@ def main():
     assert torch.allclose(sout, torch.full_like(sout, float(100 + rank)))
 
     # subgroup: each rank in its own single-member group (creator is
-    # invoked per group with a prefixed store)
-    solo = dist.new_group([rank], backend="uccl")
+    # invoked per group with a prefixed store). new_group must be called
+    # by ALL ranks with the SAME list, once per group:
+    solo = None
+    for r in range(world):
+        g = dist.new_group([r], backend="uccl")
+        if r == rank:
+            solo = g
     t1 = torch.full((16,), float(rank), device="cuda")
     dist.all_reduce(t1, group=solo)
     torch.cuda.synchronize()
