@@ -67,6 +67,20 @@ __device__ inline void block_quant_row_fp8(
   }
 }
 
+// Copy for read-once sources (combine returns): NT on both sides.
+__device__ inline void block_copy_nt(char* __restrict__ dst,
+                                     char const* __restrict__ src,
+                                     size_t bytes) {
+  size_t const nvec = bytes / 16;
+  auto* d = reinterpret_cast<V16*>(dst);
+  auto const* s = reinterpret_cast<V16 const*>(src);
+  for (size_t i = threadIdx.x; i < nvec; i += blockDim.x)
+    nt_store(&d[i], nt_load(&s[i]));
+  size_t const tail = bytes & 15;
+  if (tail && threadIdx.x < tail)
+    dst[bytes - tail + threadIdx.x] = src[bytes - tail + threadIdx.x];
+}
+
 // Copy `bytes` from src to dst cooperatively with the whole block.
 __device__ inline void block_copy(char* __restrict__ dst,
                                   char const* __restrict__ src,
@@ -96,7 +110,12 @@ __device__ inline void block_copy(char* __restrict__ dst,
 //   publish (grid=1):        seq-tagged release-store of counts
 // ---------------------------------------------------------------------------
 
-constexpr int kDispatchFanout = 32;  // copy blocks per destination expert
+// copy blocks per (expert|pair): launchers size the grid for ~2048 blocks
+// total; kernels derive the fanout from gridDim
+__host__ __device__ inline int fanout_for(int pairs) {
+  int f = 2048 / (pairs > 0 ? pairs : 1);
+  return f < 1 ? 1 : (f > 256 ? 256 : f);
+}
 
 __global__ void k_ep_dispatch_plan(EpView v,
                                    int64_t const* __restrict__ topk_idx,
@@ -142,8 +161,10 @@ __global__ void k_ep_plan_prefix(EpView v) {
 }
 
 __global__ void k_ep_dispatch_copy(EpView v, void const* __restrict__ x) {
-  int const e = blockIdx.x / kDispatchFanout;
-  int const b = blockIdx.x % kDispatchFanout;
+  int const fanout = gridDim.x / v.num_experts;
+  int const e = blockIdx.x / fanout;
+  int const b = blockIdx.x % fanout;
+  if (e >= v.num_experts) return;
   int const dst = e / v.local_experts;
   int const le = e % v.local_experts;
   bool const via_proxy = (v.proxy_mask >> dst) & 1u;
@@ -153,7 +174,7 @@ __global__ void k_ep_dispatch_copy(EpView v, void const* __restrict__ x) {
   uint32_t const count = plan[0];
   uint32_t const pfx = plan[1];
   size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
-  for (uint32_t i = b; i < count; i += kDispatchFanout) {
+  for (uint32_t i = b; i < count; i += fanout) {
     uint32_t const tk = plan[2 + i];
     uint32_t const t = tk & kMetaTokMask;
     char const* src_row =
@@ -234,8 +255,11 @@ __global__ void k_ep_dispatch_wait(EpView v,
 
 __global__ void k_ep_combine_send(EpView v,
                                   void const* __restrict__ expert_out) {
-  int const pair = blockIdx.x / kDispatchFanout;
-  int const b = blockIdx.x % kDispatchFanout;
+  int const npairs = v.local_experts * v.world;
+  int const fanout = gridDim.x / npairs;
+  int const pair = blockIdx.x / fanout;
+  int const b = blockIdx.x % fanout;
+  if (pair >= npairs) return;
   int const le = pair / v.world;
   int const src = pair % v.world;
   if ((v.proxy_mask >> src) & 1u) return;  // proxy ships these host-side
@@ -245,7 +269,7 @@ __global__ void k_ep_combine_send(EpView v,
   uint32_t const count = static_cast<uint32_t>(tagged & 0xffffffffu);
   void* sbase = v.peers[src];
   size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
-  for (uint32_t i = b; i < count; i += kDispatchFanout) {
+  for (uint32_t i = b; i < count; i += fanout) {
     size_t const slot = static_cast<size_t>(src) * v.max_tokens + i;
     uint32_t const meta = *disp_meta_ptr(me, v, le, slot);
     uint32_t const t = meta & kMetaTokMask;
@@ -255,7 +279,8 @@ __global__ void k_ep_combine_send(EpView v,
         ((static_cast<size_t>(le) * v.world * v.max_tokens + slot) *
          v.hidden) *
             v.elem_size;
-    block_copy(comb_x_ptr(sbase, v, t, k), srcrow, row_bytes);
+    // expert outputs are read exactly once -> NT both sides
+    block_copy_nt(comb_x_ptr(sbase, v, t, k), srcrow, row_bytes);
   }
 }
 
@@ -351,7 +376,8 @@ void launch_ep_dispatch(const EpView& v, void const* x,
   k_ep_dispatch_plan<<<v.num_experts, 256, smem, s>>>(v, topk_idx,
                                                       num_tokens);
   k_ep_plan_prefix<<<1, 64, 0, s>>>(v);
-  k_ep_dispatch_copy<<<v.num_experts * kDispatchFanout, 256, 0, s>>>(v, x);
+  k_ep_dispatch_copy<<<v.num_experts * fanout_for(v.num_experts), 256, 0,
+                       s>>>(v, x);
   k_ep_dispatch_publish<<<1, 256, 0, s>>>(v);
   k_ep_dispatch_wait<<<1, 256, 0, s>>>(v, out_counts);
 }
@@ -364,8 +390,9 @@ void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
 
 void launch_ep_combine_send(const EpView& v, void const* expert_out,
                             hipStream_t s) {
-  k_ep_combine_send<<<v.local_experts * v.world * kDispatchFanout, 256, 0,
-                      s>>>(v, expert_out);
+  int const npairs = v.local_experts * v.world;
+  k_ep_combine_send<<<npairs * fanout_for(npairs), 256, 0, s>>>(v,
+                                                                expert_out);
   k_ep_combine_signal<<<1, 64, 0, s>>>(v);
 }
 
