@@ -1,0 +1,116 @@
+"""Worker driving libuccl_nccl.so (the NCCL C-ABI drop-in) via ctypes:
+GetUniqueId / CommInitRank bootstrap, then allreduce / broadcast /
+allgather / reducescatter on torch GPU tensors, checked against torch."""
+
+from __future__ import annotations
+
+import ctypes
+import os
+import signal
+import sys
+import time
+
+signal.alarm(int(os.environ.get("UCCL_TEST_ALARM", "200")))
+
+import torch
+
+
+def main():
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    id_path = sys.argv[1]
+    torch.cuda.set_device(0)
+    torch.cuda.init()
+
+    from uccl_amd._build import PKG_DIR
+
+    lib = ctypes.CDLL(str(PKG_DIR / "lib" / "libuccl_nccl.so"))
+    NCCL_ID = ctypes.c_byte * 128
+    lib.ncclGetUniqueId.argtypes = [ctypes.POINTER(NCCL_ID)]
+    lib.ncclCommInitRank.argtypes = [ctypes.POINTER(ctypes.c_void_p),
+                                     ctypes.c_int, NCCL_ID, ctypes.c_int]
+    for f, extra in (("ncclAllReduce", [ctypes.c_int]),
+                     ("ncclReduceScatter", [ctypes.c_int])):
+        fn = getattr(lib, f)
+        fn.restype = ctypes.c_int
+        fn.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t,
+                       ctypes.c_int] + extra + [ctypes.c_void_p,
+                                                ctypes.c_void_p]
+    lib.ncclAllGather.restype = ctypes.c_int
+    lib.ncclAllGather.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_size_t, ctypes.c_int,
+                                  ctypes.c_void_p, ctypes.c_void_p]
+    lib.ncclBroadcast.restype = ctypes.c_int
+    lib.ncclBroadcast.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_size_t, ctypes.c_int,
+                                  ctypes.c_int, ctypes.c_void_p,
+                                  ctypes.c_void_p]
+    lib.ncclCommDestroy.argtypes = [ctypes.c_void_p]
+
+    ncclFloat32, ncclSum = 7, 0
+
+    uid = NCCL_ID()
+    if rank == 0:
+        lib.ncclGetUniqueId(ctypes.byref(uid))
+        with open(id_path + ".tmp", "wb") as f:
+            f.write(bytes(uid))
+        os.rename(id_path + ".tmp", id_path)
+    else:
+        while not os.path.exists(id_path):
+            time.sleep(0.05)
+        uid = NCCL_ID(*open(id_path, "rb").read())
+
+    comm = ctypes.c_void_p()
+    rc = lib.ncclCommInitRank(ctypes.byref(comm), world, uid, rank)
+    assert rc == 0, f"CommInitRank rc={rc}"
+
+    stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+    want = sum(r + 1.0 for r in range(world))
+
+    # in-place allreduce
+    t = torch.full((65536,), float(rank + 1), device="cuda")
+    rc = lib.ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                           ncclFloat32, ncclSum, comm, stream)
+    torch.cuda.synchronize()
+    assert rc == 0 and torch.allclose(t, torch.full_like(t, want)), t[:3]
+
+    # out-of-place allreduce
+    src = torch.full((4096,), float(rank + 1), device="cuda")
+    dst = torch.zeros(4096, device="cuda")
+    rc = lib.ncclAllReduce(src.data_ptr(), dst.data_ptr(), 4096,
+                           ncclFloat32, ncclSum, comm, stream)
+    torch.cuda.synchronize()
+    assert rc == 0 and torch.allclose(dst, torch.full_like(dst, want))
+
+    # broadcast
+    b = torch.full((1024,), float(rank * 5), device="cuda")
+    rc = lib.ncclBroadcast(b.data_ptr(), b.data_ptr(), 1024, ncclFloat32,
+                           0, comm, stream)
+    torch.cuda.synchronize()
+    assert rc == 0 and torch.allclose(b, torch.zeros_like(b))
+
+    # allgather
+    ag_in = torch.full((256,), float(rank), device="cuda")
+    ag_out = torch.empty(256 * world, device="cuda")
+    rc = lib.ncclAllGather(ag_in.data_ptr(), ag_out.data_ptr(), 256,
+                           ncclFloat32, comm, stream)
+    torch.cuda.synchronize()
+    assert rc == 0
+    for r in range(world):
+        assert torch.allclose(ag_out[r * 256:(r + 1) * 256],
+                              torch.full((256,), float(r), device="cuda"))
+
+    # reduce_scatter
+    rs_in = torch.full((512 * world,), float(rank + 1), device="cuda")
+    rs_out = torch.empty(512, device="cuda")
+    rc = lib.ncclReduceScatter(rs_in.data_ptr(), rs_out.data_ptr(), 512,
+                               ncclFloat32, ncclSum, comm, stream)
+    torch.cuda.synchronize()
+    assert rc == 0 and torch.allclose(rs_out, torch.full_like(rs_out, want))
+
+    lib.ncclCommDestroy(comm)
+    print(f"[rank {rank}] NCCL SHIM OK", flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -64,8 +64,10 @@ def build_plugin(verbose: bool = False) -> Path:
     capi = plugdir / "libuccl_p2p.so"
     capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
                  CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h"]
+    shim_check = plugdir / "libuccl_nccl.so"
     if (not _stale(target, deps) and not _stale(harness, deps)
-            and not _stale(capi, capi_srcs + deps)):
+            and not _stale(capi, capi_srcs + deps)
+            and shim_check.exists()):
         return target
     import subprocess as sp
 
@@ -87,6 +89,24 @@ def build_plugin(verbose: bool = False) -> Path:
             alias.symlink_to(target.name)
         except FileExistsError:
             pass
+    # NCCL C-ABI drop-in over the collective engine (lite-collective role)
+    shim = plugdir / "libuccl_nccl.so"
+    shim_srcs = [CSRC / "nccl_shim" / "nccl_shim.cpp",
+                 CSRC / "collective" / "kernels.hip",
+                 CSRC / "collective" / "communicator.cpp"]
+    shim_hdrs = list((CSRC / "collective").glob("*.h")) + [
+        CSRC / "device" / "primitives.h"]
+    if _stale(shim, shim_srcs + shim_hdrs + deps):
+        cmd = [HIPCC, "-O3", "-std=c++17", "-fPIC", "-shared",
+               f"--offload-arch={GPU_ARCH}"] + [str(x) for x in shim_srcs] + [
+               "-o", str(shim), "-pthread", f"-L{ROCM}/lib", "-lamdhip64",
+               f"-Wl,-rpath,{ROCM}/lib"]
+        if verbose:
+            print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)
+        r = sp.run(cmd, stdout=sp.PIPE, stderr=sp.STDOUT)
+        if r.returncode != 0:
+            raise RuntimeError(f"nccl shim build failed:\n{r.stdout.decode()}")
+
     # flat C API lib for NIXL-style integrators (reference: p2p/uccl_engine.h)
     if _stale(capi, capi_srcs + deps):
         cc_srcs = [x for x in capi_srcs if x.suffix == ".cpp"]
