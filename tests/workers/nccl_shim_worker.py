@@ -25,10 +25,13 @@ def main():
     from uccl_amd._build import PKG_DIR
 
     lib = ctypes.CDLL(str(PKG_DIR / "lib" / "libuccl_nccl.so"))
-    NCCL_ID = ctypes.c_byte * 128
-    lib.ncclGetUniqueId.argtypes = [ctypes.POINTER(NCCL_ID)]
+    class NcclId(ctypes.Structure):
+        _fields_ = [("b", ctypes.c_byte * 128)]  # passed BY VALUE
+
+    NCCL_ID = NcclId
+    lib.ncclGetUniqueId.argtypes = [ctypes.POINTER(NcclId)]
     lib.ncclCommInitRank.argtypes = [ctypes.POINTER(ctypes.c_void_p),
-                                     ctypes.c_int, NCCL_ID, ctypes.c_int]
+                                     ctypes.c_int, NcclId, ctypes.c_int]
     for f, extra in (("ncclAllReduce", [ctypes.c_int]),
                      ("ncclReduceScatter", [ctypes.c_int])):
         fn = getattr(lib, f)
@@ -53,12 +56,14 @@ def main():
     if rank == 0:
         lib.ncclGetUniqueId(ctypes.byref(uid))
         with open(id_path + ".tmp", "wb") as f:
-            f.write(bytes(uid))
+            f.write(bytes(uid.b))
         os.rename(id_path + ".tmp", id_path)
     else:
         while not os.path.exists(id_path):
             time.sleep(0.05)
-        uid = NCCL_ID(*open(id_path, "rb").read())
+        raw = open(id_path, "rb").read()
+        uid = NCCL_ID()
+        ctypes.memmove(uid.b, raw, 128)
 
     comm = ctypes.c_void_p()
     rc = lib.ncclCommInitRank(ctypes.byref(comm), world, uid, rank)
