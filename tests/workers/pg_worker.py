@@ -118,6 +118,25 @@ def main():
         assert torch.allclose(p.grad, ref, atol=1e-5), \
             (p.grad - ref).abs().max()
 
+    # FSDP end-to-end (exercises _allgather_base/_reduce_scatter_base)
+    try:
+        from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
+
+        fm = torch.nn.Sequential(
+            torch.nn.Linear(64, 128), torch.nn.ReLU(),
+            torch.nn.Linear(128, 16)).cuda()
+        for p_ in fm.parameters():
+            dist.broadcast(p_.data, src=0)
+        fsdp = FSDP(fm, device_id=torch.cuda.current_device())
+        gx = torch.Generator().manual_seed(900 + rank)
+        loss = fsdp(torch.randn(8, 64, generator=gx).cuda()).square().mean()
+        loss.backward()
+        torch.cuda.synchronize()
+        if rank == 0:
+            print("FSDP OK", flush=True)
+    except ImportError:
+        pass
+
     dist.barrier()
     if rank == 0:
         print("PG BACKEND ALL OK", flush=True)
