@@ -183,6 +183,28 @@ def main():
     ])
     check("all_to_all", out, want, 0.0)
 
+    # --- chunked broadcast / reduce_scatter --------------------------------
+    if not light:
+        seed += 1
+        n = 24 << 20  # 96MB fp32 > 36MB parity scratch -> multi-chunk bcast
+        x = make_input(rank, n, torch.float32, seed)
+        comm.broadcast(x, root=1 if world > 1 else 0)
+        torch.cuda.synchronize()
+        check("broadcast[chunked]", x,
+              make_input(1 if world > 1 else 0, n, torch.float32,
+                         seed).cpu(), 0.0)
+        del x
+        seed += 1
+        per = 10 << 20  # per-rank 40MB fp32; full input world*40MB chunked
+        inp = make_input(rank, per * world, torch.float32, seed)
+        out = torch.empty(per, dtype=torch.float32, device="cuda")
+        comm.reduce_scatter(out, inp)
+        torch.cuda.synchronize()
+        full = expected_sum(world, per * world, torch.float32, seed)
+        check("reduce_scatter[chunked]", out,
+              full[rank * per:(rank + 1) * per], 0.0)
+        del inp, out
+
     # --- send/recv (pairwise ring) ------------------------------------------
     seed += 1
     n = 300000  # spans >1 p2p slot chunk at 2MB slots? 1.2MB -> single chunk
