@@ -259,6 +259,28 @@ def main():
             torch.cuda.synchronize()
             check(f"allreduce[sym,{dtype},{count}]", st,
                   expected_sum(world, count, dtype, seed), tol[dtype])
+    # symmetric all_gather (out symmetric) + reduce_scatter (in symmetric)
+    seed += 1
+    n = 65536
+    ag_out = comm.symmetric_tensor([world * n], torch.float32)
+    mine = make_input(rank, n, torch.float32, seed)
+    comm.all_gather(ag_out, mine)
+    torch.cuda.synchronize()
+    want = torch.cat([make_input(r, n, torch.float32, seed).cpu()
+                      for r in range(world)])
+    check("allgather[sym]", ag_out, want, 0.0)
+
+    seed += 1
+    per = 32768
+    rs_in = comm.symmetric_tensor([world * per], torch.float32)
+    rs_in.copy_(make_input(rank, world * per, torch.float32, seed))
+    rs_out = torch.empty(per, dtype=torch.float32, device="cuda")
+    comm.reduce_scatter(rs_out, rs_in)
+    torch.cuda.synchronize()
+    full = expected_sum(world, world * per, torch.float32, seed)
+    check("reduce_scatter[sym]", rs_out,
+          full[rank * per:(rank + 1) * per], 0.0)
+
     # repeated in-place reuse of the same symmetric tensor
     seed += 1
     st = comm.symmetric_tensor([4096], torch.float32)

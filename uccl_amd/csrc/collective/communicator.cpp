@@ -201,6 +201,16 @@ void Communicator::all_gather(void* out, void const* in, size_t count_per_rank,
     return;
   }
   UCCL_CHECK(connected_) << "connect() not called";
+  if (is_symmetric_ptr(out)) {
+    // zero-copy: push my slice straight into every rank's `out` slot,
+    // then one completion flag round
+    size_t const uoff =
+        static_cast<char const*>(out) - static_cast<char*>(heap_);
+    CommView const cv = view(next_seq());
+    launch_allgather_sym_push(cv, in, uoff, count_per_rank * es, stream);
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
+    return;
+  }
   size_t const chunk_elems = (scratch_cap_ / es) & ~size_t(63);
   for (size_t off = 0; off < count_per_rank; off += chunk_elems) {
     size_t const n = std::min(chunk_elems, count_per_rank - off);
@@ -234,6 +244,17 @@ void Communicator::reduce_scatter(void* out, void const* in,
   }
   UCCL_CHECK(connected_) << "connect() not called";
   UCCL_CHECK(dt != Dtype::kU8) << "reduce_scatter needs a typed dtype";
+  if (is_symmetric_ptr(in)) {
+    // zero-copy: reduce my shard straight out of every rank's symmetric
+    // input; entry barrier publishes inputs, exit barrier frees them
+    size_t const uoff =
+        static_cast<char const*>(in) - static_cast<char*>(heap_);
+    CommView const cv = view(next_seq());
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
+    launch_reducescatter_sym(cv, uoff, out, count_per_rank, dt, stream);
+    launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
+    return;
+  }
   size_t const chunk_elems = (scratch_cap_ / es / world_) & ~size_t(63);
   for (size_t off = 0; off < count_per_rank; off += chunk_elems) {
     size_t const n = std::min(chunk_elems, count_per_rank - off);

@@ -283,6 +283,63 @@ __global__ void k_twoshot_sym_push(CommView cv, size_t uoff, size_t count) {
   }
 }
 
+// Symmetric all_gather: `out` (symmetric, world*n elems) receives rank r's
+// local `in` at slot r on EVERY rank — pure pushes, no read dependencies,
+// so the only flag round is the completion wait after the push.
+__global__ void k_allgather_sym_push(CommView cv, void const* __restrict__ in,
+                                     size_t uoff, size_t slot_bytes) {
+  size_t const nvec = slot_bytes / 16;
+  size_t const tail = slot_bytes & 15;
+  auto const* s = reinterpret_cast<V16 const*>(in);
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t const stride = gridDim.x * blockDim.x;
+  for (int k = 0; k < cv.world; ++k) {
+    int const p = (cv.rank + k) % cv.world;
+    auto* d = reinterpret_cast<V16*>(static_cast<char*>(cv.peers[p]) + uoff +
+                                     static_cast<size_t>(cv.rank) *
+                                         slot_bytes);
+    for (size_t j = i; j < nvec; j += stride) nt_store(&d[j], nt_load(&s[j]));
+    if (tail && blockIdx.x == 0 && threadIdx.x < tail)
+      reinterpret_cast<char*>(d)[slot_bytes - tail + threadIdx.x] =
+          reinterpret_cast<char const*>(s)[slot_bytes - tail + threadIdx.x];
+  }
+}
+
+// Symmetric reduce_scatter: `in` symmetric [world*count]; out[i] =
+// sum_p in_p[rank*count + i]. Entry barrier (inputs published) before the
+// remote reads; exit barrier so callers may overwrite `in` afterwards.
+template <typename T>
+__global__ void k_reducescatter_sym(CommView cv, size_t uoff,
+                                    void* __restrict__ out, size_t count) {
+  size_t const vper = 16 / sizeof(T);
+  size_t const nvec = count / vper;
+  size_t const elem_off = static_cast<size_t>(cv.rank) * count;
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t const stride = gridDim.x * blockDim.x;
+  for (; i < nvec; i += stride) {
+    AccumV16<T, 0> acc;
+    acc.init(nt_load(reinterpret_cast<V16 const*>(
+        reinterpret_cast<T const*>(static_cast<char*>(cv.peers[0]) + uoff) +
+        elem_off) + i));
+    for (int p = 1; p < cv.world; ++p) {
+      acc.add(nt_load(reinterpret_cast<V16 const*>(
+          reinterpret_cast<T const*>(static_cast<char*>(cv.peers[p]) +
+                                     uoff) +
+          elem_off) + i));
+    }
+    nt_store(reinterpret_cast<V16*>(out) + i, acc.pack());
+  }
+  size_t const tail = count - nvec * vper;
+  if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
+    size_t const j = nvec * vper + threadIdx.x;
+    float a = 0.f;
+    for (int p = 0; p < cv.world; ++p)
+      a += static_cast<float>(reinterpret_cast<T const*>(
+          static_cast<char*>(cv.peers[p]) + uoff)[elem_off + j]);
+    reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
+  }
+}
+
 // Phase 2 kernel — signal seq+1 (my pushes are visible: dispatch boundary),
 // wait for everyone's pushes, copy assembled scratchB to the output.
 __global__ void k_twoshot_copyout(CommView cv, void* __restrict__ out,
@@ -606,6 +663,25 @@ static void l_sym_push(const CommView& cv, size_t uoff, size_t count,
 void launch_twoshot_sym_rs(const CommView& cv, size_t uoff, size_t count,
                            Dtype dt, hipStream_t s) {
   DT_DISPATCH(dt, l_sym_rs, cv, uoff, count, s);
+}
+
+void launch_allgather_sym_push(const CommView& cv, void const* in,
+                               size_t uoff, size_t slot_bytes,
+                               hipStream_t s) {
+  k_allgather_sym_push<<<grid_for(slot_bytes * cv.world), 256, 0, s>>>(
+      cv, in, uoff, slot_bytes);
+}
+
+template <typename T>
+static void l_rs_sym2(const CommView& cv, size_t uoff, void* out,
+                      size_t count, hipStream_t s) {
+  k_reducescatter_sym<T>
+      <<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, uoff, out, count);
+}
+
+void launch_reducescatter_sym(const CommView& cv, size_t uoff, void* out,
+                              size_t count, Dtype dt, hipStream_t s) {
+  DT_DISPATCH(dt, l_rs_sym2, cv, uoff, out, count, s);
 }
 
 void launch_twoshot_sym_push(const CommView& cv, size_t uoff, size_t count,
