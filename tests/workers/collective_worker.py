@@ -14,7 +14,6 @@ from __future__ import annotations
 
 import os
 import signal
-import sys
 
 signal.alarm(int(os.environ.get("UCCL_TEST_ALARM", "240")))
 

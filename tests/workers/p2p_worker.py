@@ -6,7 +6,6 @@ from __future__ import annotations
 import os
 import signal
 import sys
-import tempfile
 import time
 
 signal.alarm(int(os.environ.get("UCCL_TEST_ALARM", "180")))

@@ -12,7 +12,6 @@ CDNA4 + xGMI):
 
 from __future__ import annotations
 
-import os
 
 __version__ = "0.1.0"
 
