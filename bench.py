@@ -37,8 +37,11 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--mbytes", type=int, default=256,
                    help="per-GPU buffer size in MiB (bf16)")
-    p.add_argument("--symmetric", action="store_true",
-                   help="use the zero-copy symmetric-tensor path")
+    p.add_argument("--symmetric", dest="symmetric", action="store_true",
+                   default=True,
+                   help="use the zero-copy symmetric-tensor path (default)")
+    p.add_argument("--staged", dest="symmetric", action="store_false",
+                   help="force the staged (non-registered buffer) path")
     return p.parse_args()
 
 
@@ -86,16 +89,29 @@ def main():
         if use_dist:
             dist.barrier()
 
-    # correctness guard outside the timed region: allreduce of ones must
-    # give world (catches silently-broken paths before we publish a number)
-    chk = torch.ones(4096, dtype=torch.bfloat16, device="cuda")
-    comm.all_reduce(chk)
-    torch.cuda.synchronize()
+    # correctness guards outside the timed region: allreduce of ones must
+    # give world. The symmetric path is checked on its own tensor; if it
+    # ever fails we fall back to the staged path (and say so) rather than
+    # publishing a number from a broken path or no number at all.
     expect = float(world)
-    if not torch.allclose(chk, torch.full_like(chk, expect)):
+
+    def check_ones(tensor):
+        tensor.fill_(1.0)
+        comm.all_reduce(tensor)
+        torch.cuda.synchronize()
+        return bool(torch.allclose(tensor,
+                                   torch.full_like(tensor, expect)))
+
+    chk = torch.ones(4096, dtype=torch.bfloat16, device="cuda")
+    if not check_ones(chk):
         print(json.dumps({"error": "allreduce correctness check failed",
                           "got": float(chk[0])}))
         sys.exit(2)
+    if args.symmetric and world > 1:
+        schk = comm.symmetric_tensor([4096], torch.bfloat16)
+        if not check_ones(schk):
+            args.symmetric = False
+            t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
 
     for _ in range(args.warmup):
         comm.all_reduce(t)
@@ -145,7 +161,8 @@ def main():
                 "engine": ("uccl_amd symmetric zero-copy twoshot"
                            if args.symmetric else
                            "uccl_amd twoshot RS+AG push kernels"),
-                "symmetric": args.symmetric,
+                "symmetric": bool(args.symmetric and world > 1
+                                  and comm.is_symmetric(t)),
                 "n1_value_is_local_algbw": world == 1,
             },
         }))
