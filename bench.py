@@ -78,7 +78,7 @@ def main():
 
     nbytes = args.mbytes * (1 << 20)
     count = nbytes // 2  # bf16
-    if args.symmetric:
+    if args.symmetric and world > 1:
         t = comm.symmetric_tensor([count], torch.bfloat16)
         t.copy_(torch.randn(count, dtype=torch.bfloat16))
     else:
