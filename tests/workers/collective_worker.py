@@ -280,6 +280,20 @@ def main():
     check("reduce_scatter[sym]", rs_out,
           full[rank * per:(rank + 1) * per], 0.0)
 
+    # symmetric all_to_all (out symmetric)
+    seed += 1
+    per = 16384
+    a2a_out = comm.symmetric_tensor([world * per], torch.float32)
+    a2a_in = make_input(rank, per * world, torch.float32, seed)
+    comm.all_to_all(a2a_out, a2a_in)
+    torch.cuda.synchronize()
+    want = torch.cat([
+        make_input(r, per * world, torch.float32,
+                   seed)[rank * per:(rank + 1) * per].cpu()
+        for r in range(world)
+    ])
+    check("alltoall[sym]", a2a_out, want, 0.0)
+
     # repeated in-place reuse of the same symmetric tensor
     seed += 1
     st = comm.symmetric_tensor([4096], torch.float32)

@@ -304,6 +304,14 @@ void Communicator::all_to_all(void* out, void const* in, size_t count_per_rank,
     return;
   }
   UCCL_CHECK(connected_) << "connect() not called";
+  if (is_symmetric_ptr(out)) {
+    size_t const uoff =
+        static_cast<char const*>(out) - static_cast<char*>(heap_);
+    CommView const cv = view(next_seq());
+    launch_alltoall_sym_push(cv, in, uoff, count_per_rank * es, stream);
+    launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
+    return;
+  }
   size_t const chunk_elems = (scratch_cap_ / es / world_) & ~size_t(63);
   UCCL_CHECK(chunk_elems > 0) << "scratch too small for alltoall";
   for (size_t off = 0; off < count_per_rank; off += chunk_elems) {

@@ -38,6 +38,9 @@ void launch_allgather_sym_push(const CommView& cv, void const* in,
                                hipStream_t s);
 void launch_reducescatter_sym(const CommView& cv, size_t uoff, void* out,
                               size_t count, Dtype dt, hipStream_t s);
+void launch_alltoall_sym_push(const CommView& cv, void const* in,
+                              size_t uoff, size_t chunk_bytes,
+                              hipStream_t s);
 void launch_ll_allreduce(const CommView& cv, void const* in, void* out,
                          size_t count, Dtype dt, hipStream_t s);
 void launch_allgather_pull(const CommView& cv, void* out, size_t chunk_bytes,

@@ -305,6 +305,29 @@ __global__ void k_allgather_sym_push(CommView cv, void const* __restrict__ in,
   }
 }
 
+// Symmetric all_to_all: `out` symmetric [world*n]; rank r pushes its
+// input chunk p into peer p's out at slot r. Pure pushes + completion
+// round, like the symmetric all_gather.
+__global__ void k_alltoall_sym_push(CommView cv, void const* __restrict__ in,
+                                    size_t uoff, size_t chunk_bytes) {
+  size_t const nvec = chunk_bytes / 16;
+  size_t const tail = chunk_bytes & 15;
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t const stride = gridDim.x * blockDim.x;
+  for (int k = 0; k < cv.world; ++k) {
+    int const p = (cv.rank + k) % cv.world;
+    auto const* s = reinterpret_cast<V16 const*>(
+        static_cast<char const*>(in) + static_cast<size_t>(p) * chunk_bytes);
+    auto* d = reinterpret_cast<V16*>(static_cast<char*>(cv.peers[p]) + uoff +
+                                     static_cast<size_t>(cv.rank) *
+                                         chunk_bytes);
+    for (size_t j = i; j < nvec; j += stride) nt_store(&d[j], nt_load(&s[j]));
+    if (tail && blockIdx.x == 0 && threadIdx.x < tail)
+      reinterpret_cast<char*>(d)[chunk_bytes - tail + threadIdx.x] =
+          reinterpret_cast<char const*>(s)[chunk_bytes - tail + threadIdx.x];
+  }
+}
+
 // Symmetric reduce_scatter: `in` symmetric [world*count]; out[i] =
 // sum_p in_p[rank*count + i]. Entry barrier (inputs published) before the
 // remote reads; exit barrier so callers may overwrite `in` afterwards.
@@ -670,6 +693,13 @@ void launch_allgather_sym_push(const CommView& cv, void const* in,
                                hipStream_t s) {
   k_allgather_sym_push<<<grid_for(slot_bytes * cv.world), 256, 0, s>>>(
       cv, in, uoff, slot_bytes);
+}
+
+void launch_alltoall_sym_push(const CommView& cv, void const* in,
+                              size_t uoff, size_t chunk_bytes,
+                              hipStream_t s) {
+  k_alltoall_sym_push<<<grid_for(chunk_bytes * cv.world), 256, 0, s>>>(
+      cv, in, uoff, chunk_bytes);
 }
 
 template <typename T>
