@@ -202,13 +202,16 @@ void Communicator::all_gather(void* out, void const* in, size_t count_per_rank,
   }
   UCCL_CHECK(connected_) << "connect() not called";
   if (is_symmetric_ptr(out)) {
-    // zero-copy: push my slice straight into every rank's `out` slot,
-    // then one completion flag round
+    // zero-copy: entry flag round first — remote writers must not touch a
+    // peer's `out` before that peer has entered the call (its preceding
+    // stream work may still be reading the buffer) — then push my slice
+    // into every rank's slot, then the completion round.
     size_t const uoff =
         static_cast<char const*>(out) - static_cast<char*>(heap_);
     CommView const cv = view(next_seq());
-    launch_allgather_sym_push(cv, in, uoff, count_per_rank * es, stream);
     launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
+    launch_allgather_sym_push(cv, in, uoff, count_per_rank * es, stream);
+    launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
     return;
   }
   size_t const chunk_elems = (scratch_cap_ / es) & ~size_t(63);
@@ -305,11 +308,13 @@ void Communicator::all_to_all(void* out, void const* in, size_t count_per_rank,
   }
   UCCL_CHECK(connected_) << "connect() not called";
   if (is_symmetric_ptr(out)) {
+    // entry round before remote writes (see all_gather comment)
     size_t const uoff =
         static_cast<char const*>(out) - static_cast<char*>(heap_);
     CommView const cv = view(next_seq());
-    launch_alltoall_sym_push(cv, in, uoff, count_per_rank * es, stream);
     launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
+    launch_alltoall_sym_push(cv, in, uoff, count_per_rank * es, stream);
+    launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
     return;
   }
   size_t const chunk_elems = (scratch_cap_ / es / world_) & ~size_t(63);
