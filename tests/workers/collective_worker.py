@@ -23,22 +23,30 @@ import torch.distributed as dist
 
 def make_input(rank: int, count: int, dtype, seed_tag: int) -> torch.Tensor:
     g = torch.Generator().manual_seed(1234 + 97 * rank + seed_tag)
-    if dtype == torch.int32:
-        return torch.randint(-1000, 1000, (count,), generator=g,
+    if dtype in (torch.int32, torch.int64):
+        return torch.randint(-1000000, 1000000, (count,), generator=g,
                              dtype=dtype).cuda()
+    if dtype == torch.float64:
+        return torch.randn(count, generator=g, dtype=torch.float64).cuda()
     return torch.randn(count, generator=g, dtype=torch.float32).to(dtype).cuda()
 
 
 def expected_sum(world: int, count: int, dtype, seed_tag: int) -> torch.Tensor:
     # fp32 (or int64) reference accumulation on CPU, then cast once — this is
     # exactly what the kernels' AccumV16 fp32 accumulators implement.
-    if dtype == torch.int32:
+    if dtype in (torch.int32, torch.int64):
         acc = torch.zeros(count, dtype=torch.int64)
         for r in range(world):
             g = torch.Generator().manual_seed(1234 + 97 * r + seed_tag)
-            acc += torch.randint(-1000, 1000, (count,), generator=g,
-                                 dtype=torch.int32).to(torch.int64)
-        return acc.to(torch.int32)
+            acc += torch.randint(-1000000, 1000000, (count,), generator=g,
+                                 dtype=dtype).to(torch.int64)
+        return acc.to(dtype)
+    if dtype == torch.float64:
+        acc = torch.zeros(count, dtype=torch.float64)
+        for r in range(world):
+            g = torch.Generator().manual_seed(1234 + 97 * r + seed_tag)
+            acc += torch.randn(count, generator=g, dtype=torch.float64)
+        return acc
     acc = torch.zeros(count, dtype=torch.float32)
     for r in range(world):
         g = torch.Generator().manual_seed(1234 + 97 * r + seed_tag)
@@ -48,12 +56,12 @@ def expected_sum(world: int, count: int, dtype, seed_tag: int) -> torch.Tensor:
 
 
 def check(name: str, got: torch.Tensor, want: torch.Tensor, tol: float):
-    got = got.float().cpu()
-    want = want.float().cpu()
+    got = got.cpu()
+    want = want.cpu()
     if tol == 0:
-        ok = torch.equal(got, want)
+        ok = torch.equal(got, want)  # raw-dtype exact compare
     else:
-        ok = torch.allclose(got, want, rtol=tol, atol=tol)
+        ok = torch.allclose(got.float(), want.float(), rtol=tol, atol=tol)
     if not ok:
         diff = (got - want).abs().max().item()
         raise AssertionError(f"{name}: max diff {diff}")
@@ -74,15 +82,17 @@ def main():
                      heap_bytes=light_heap * (1 << 20))
 
     light = os.environ.get("UCCL_TEST_LIGHT", "0") == "1"
-    tol = {torch.float32: 0.0, torch.int32: 0.0,
-           torch.bfloat16: 1e-2, torch.float16: 1e-3}
+    # fixed-order accumulation makes fp32/fp64/int results bit-exact
+    tol = {torch.float32: 0.0, torch.int32: 0.0, torch.int64: 0.0,
+           torch.float64: 0.0, torch.bfloat16: 1e-2, torch.float16: 1e-3}
 
     seed = 0
     # --- allreduce across the three algorithm paths -------------------------
     # LL (<=32KB), one-shot (<=2MB), two-shot (>2MB), chunked two-shot
     cases = [(1000, "ll"), (100000, "oneshot"), (3 << 20, "twoshot"),
              (1000003, "oneshot-odd")]
-    dtypes = (torch.float32, torch.bfloat16, torch.float16, torch.int32)
+    dtypes = (torch.float32, torch.bfloat16, torch.float16, torch.int32,
+              torch.int64, torch.float64)
     if light:
         cases = [(1000, "ll"), (100000, "oneshot"), (3 << 20, "twoshot")]
         dtypes = (torch.float32, torch.bfloat16)

@@ -37,6 +37,8 @@ Dtype dtype_of(const at::Tensor& t) {
     case at::kHalf: return Dtype::kF16;
     case at::kBFloat16: return Dtype::kBF16;
     case at::kInt: return Dtype::kI32;
+    case at::kLong: return Dtype::kI64;
+    case at::kDouble: return Dtype::kF64;
     case at::kFloat8_e4m3fn: return Dtype::kF8E4M3;
     default:
       TORCH_CHECK(false, "uccl_amd: unsupported dtype ", t.scalar_type());

@@ -149,7 +149,8 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
   size_t const es = dtype_size(dt);
   size_t const bytes = count * es;
 
-  if (bytes <= ll_threshold_) {
+  if (bytes <= ll_threshold_ && es < 8) {
+    // (8-byte dtypes skip LL: its 4B packet lanes can't carry i64/f64)
     launch_ll_allreduce(view(next_seq()), data, data, count, dt, stream);
     return;
   }

@@ -10,7 +10,7 @@ namespace uccl {
 // kU8 is the raw-byte dtype used by copy-shaped collectives (broadcast,
 // allgather, alltoall, send/recv) so any tensor dtype can ride them; the
 // reducing collectives reject it.
-enum class Dtype : int { kF32 = 0, kF16 = 1, kBF16 = 2, kI32 = 3, kU8 = 4, kF8E4M3 = 5 };
+enum class Dtype : int { kF32 = 0, kF16 = 1, kBF16 = 2, kI32 = 3, kU8 = 4, kF8E4M3 = 5, kI64 = 6, kF64 = 7 };
 
 inline size_t dtype_size(Dtype d) {
   switch (d) {
@@ -18,6 +18,8 @@ inline size_t dtype_size(Dtype d) {
     case Dtype::kBF16: return 2;
     case Dtype::kU8:
     case Dtype::kF8E4M3: return 1;
+    case Dtype::kI64:
+    case Dtype::kF64: return 8;
     default: return 4;
   }
 }

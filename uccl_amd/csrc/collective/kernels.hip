@@ -116,9 +116,9 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    float a = 0.f;
+    typename TailAcc<T>::type a = 0;
     for (int p = 0; p < cv.world; ++p)
-      a += static_cast<float>(
+      a += static_cast<typename TailAcc<T>::type>(
           reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv))[j]);
     reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
   }
@@ -197,9 +197,9 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   if (tail && cv.rank == cv.world - 1 && blockIdx.x == 0 &&
       threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    float a = 0.f;
+    typename TailAcc<T>::type a = 0;
     for (int p = 0; p < cv.world; ++p)
-      a += static_cast<float>(
+      a += static_cast<typename TailAcc<T>::type>(
           reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv))[j]);
     T const r = static_cast<T>(a);
     for (int p = 0; p < cv.world; ++p)
@@ -241,9 +241,9 @@ __global__ void k_twoshot_sym_rs(CommView cv, size_t uoff, size_t count) {
   if (tail && cv.rank == cv.world - 1 && blockIdx.x == 0 &&
       threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    float a = 0.f;
+    typename TailAcc<T>::type a = 0;
     for (int p = 0; p < cv.world; ++p)
-      a += static_cast<float>(reinterpret_cast<T const*>(
+      a += static_cast<typename TailAcc<T>::type>(reinterpret_cast<T const*>(
           static_cast<char*>(cv.peers[p]) + uoff)[j]);
     // stash tail results after the vector shard in scratchB
     reinterpret_cast<T*>(sb)[(end - beg) * vper + threadIdx.x] =
@@ -355,9 +355,9 @@ __global__ void k_reducescatter_sym(CommView cv, size_t uoff,
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    float a = 0.f;
+    typename TailAcc<T>::type a = 0;
     for (int p = 0; p < cv.world; ++p)
-      a += static_cast<float>(reinterpret_cast<T const*>(
+      a += static_cast<typename TailAcc<T>::type>(reinterpret_cast<T const*>(
           static_cast<char*>(cv.peers[p]) + uoff)[elem_off + j]);
     reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
   }
@@ -526,9 +526,9 @@ __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    float a = 0.f;
+    typename TailAcc<T>::type a = 0;
     for (int p = 0; p < cv.world; ++p)
-      a += static_cast<float>(reinterpret_cast<T const*>(
+      a += static_cast<typename TailAcc<T>::type>(reinterpret_cast<T const*>(
           scratch_a(cv.peers[p], cv))[elem_off + j]);
     reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
   }
@@ -626,6 +626,8 @@ void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s) {
     case Dtype::kI32: fn<int>(__VA_ARGS__); break;      \
     case Dtype::kU8: /* reducing collectives reject kU8 host-side */ break; \
     case Dtype::kF8E4M3: fn<__hip_fp8_e4m3>(__VA_ARGS__); break; \
+    case Dtype::kI64: fn<long long>(__VA_ARGS__); break; \
+    case Dtype::kF64: fn<double>(__VA_ARGS__); break; \
   }
 
 template <typename T>

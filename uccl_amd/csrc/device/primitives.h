@@ -264,6 +264,66 @@ struct AccumV16<__hip_fp8_e4m3, N> {
 };
 
 template <int N>
+struct AccumV16<long long, N> {
+  long long v[2];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      v[i] = reinterpret_cast<long long const*>(&a)[i];
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      v[i] += reinterpret_cast<long long const*>(&a)[i];
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) reinterpret_cast<long long*>(&r)[i] = v[i];
+    return r;
+  }
+};
+
+template <int N>
+struct AccumV16<double, N> {
+  double v[2];
+  __device__ __forceinline__ void init(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      v[i] = reinterpret_cast<double const*>(&a)[i];
+  }
+  __device__ __forceinline__ void add(V16 a) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      v[i] += reinterpret_cast<double const*>(&a)[i];
+  }
+  __device__ __forceinline__ V16 pack() const {
+    V16 r;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) reinterpret_cast<double*>(&r)[i] = v[i];
+    return r;
+  }
+};
+
+// scalar accumulator type for ragged tails (exact for integers/doubles)
+template <typename T>
+struct TailAcc {
+  using type = float;
+};
+template <>
+struct TailAcc<int> {
+  using type = long long;
+};
+template <>
+struct TailAcc<long long> {
+  using type = long long;
+};
+template <>
+struct TailAcc<double> {
+  using type = double;
+};
+
+template <int N>
 struct AccumV16<int, N> {
   int v[4];
   __device__ __forceinline__ void init(V16 a) {
