@@ -30,15 +30,15 @@ def test_ep_world1():
     assert "EP ALL OK" in out
 
 
-def _run_world2(extra_env, port):
+def _run_world2(extra_env, port, world=2):
     env_base = dict(os.environ)
     env_base.update({
         "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
-        "WORLD_SIZE": "2", "PYTHONPATH": str(REPO),
+        "WORLD_SIZE": str(world), "PYTHONPATH": str(REPO),
     })
     env_base.update(extra_env)
     ps = []
-    for r in range(2):
+    for r in range(world):
         env = dict(env_base)
         env["RANK"] = str(r)
         ps.append(subprocess.Popen([sys.executable, str(WORKER)], env=env,
@@ -56,7 +56,7 @@ def _run_world2(extra_env, port):
         ok = ok and p.returncode == 0
     joined = "\n=====\n".join(outs)
     assert ok, joined
-    assert joined.count("EP ALL OK") == 2, joined
+    assert joined.count("EP ALL OK") == world, joined
 
 
 def test_ep_world2():
@@ -68,3 +68,7 @@ def test_ep_world2_forced_proxy():
     traffic flows GPU -> D2H command ring -> CPU proxy -> multipath
     reliable transport -> peer proxy -> peer GPU memory."""
     _run_world2({"UCCL_EP_FORCE_PROXY": "1"}, 29477)
+
+
+def test_ep_world4_forced_proxy():
+    _run_world2({"UCCL_EP_FORCE_PROXY": "1"}, 29479, world=4)
