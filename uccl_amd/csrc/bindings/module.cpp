@@ -63,6 +63,25 @@ PYBIND11_MODULE(_C, m) {
     return n;
   });
 
+  // peer-access matrix + link attributes (xGMI topology introspection;
+  // parity role: the reference's NIC/GPU topology discovery, util.h)
+  m.def("peer_matrix", [] {
+    int n = 0;
+    (void)hipGetDeviceCount(&n);
+    std::vector<std::vector<int>> can(n, std::vector<int>(n, 0));
+    for (int a = 0; a < n; ++a)
+      for (int b = 0; b < n; ++b) {
+        if (a == b) {
+          can[a][b] = 1;
+          continue;
+        }
+        int ok = 0;
+        (void)hipDeviceCanAccessPeer(&ok, a, b);
+        can[a][b] = ok;
+      }
+    return can;
+  });
+
   py::class_<Communicator>(m, "Communicator")
       .def(py::init([](int rank, int world, int device, size_t heap_bytes) {
              return new Communicator(rank, world, device, heap_bytes);
