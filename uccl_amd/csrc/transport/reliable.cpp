@@ -4,6 +4,7 @@
 #include <sys/socket.h>
 
 #include <algorithm>
+#include <random>
 #include <stdexcept>
 #include <cstdlib>
 #include <cstring>
@@ -505,6 +506,13 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
     getsockname(impl_->wake_fd, reinterpret_cast<sockaddr*>(&a), &al);
     impl_->wake_port = ntohs(a.sin_port);
   }
+  // Flow ids are assigned by the ACCEPTOR and used verbatim by both
+  // sides (wire id == map key). Endpoints can hold flows accepted
+  // locally AND flows assigned by remote acceptors, so ids must be
+  // globally unique: embed a per-endpoint random tag in the high bits.
+  impl_->next_flow =
+      (static_cast<uint64_t>(std::random_device{}() & 0x7fffffffu) << 32) |
+      1;
   impl_->ctrl_listen = net::listen_on(&impl_->ctrl_port);
   impl_->ctrl_thread = std::thread([this] { impl_->ctrl_loop(); });
   impl_->progress = std::thread([this] { impl_->progress_loop(); });
