@@ -21,6 +21,17 @@ int EpProxy::flow_peer(uint64_t flow) const {
   return 0;
 }
 
+EpProxy::Lane::Lane(int device, size_t bytes) {
+  UCCL_CHECK_HIP(hipSetDevice(device));
+  UCCL_CHECK_HIP(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+  UCCL_CHECK_HIP(hipHostMalloc(&buf, bytes));
+}
+
+EpProxy::Lane::~Lane() {
+  if (buf) (void)hipHostFree(buf);
+  if (stream) (void)hipStreamDestroy(stream);
+}
+
 EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
                  int device)
     : v_(view), heap_(heap), ring_(ring_host), device_(device) {
@@ -30,13 +41,7 @@ EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
   flows_.resize(v_.world, 0);
   for (int r = 0; r < v_.world; ++r)
     flow_mu_.emplace_back(new std::mutex());
-  UCCL_CHECK_HIP(hipSetDevice(device_));
-  UCCL_CHECK_HIP(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
-  UCCL_CHECK_HIP(
-      hipStreamCreateWithFlags(&rx_stream_, hipStreamNonBlocking));
   stage_bytes_ = kStageBytes;
-  UCCL_CHECK_HIP(hipHostMalloc(&host_buf_, stage_bytes_));
-  UCCL_CHECK_HIP(hipHostMalloc(&rx_buf_, stage_bytes_));
 }
 
 EpProxy::~EpProxy() {
@@ -48,10 +53,6 @@ EpProxy::~EpProxy() {
   for (auto& t : rx_threads_)
     if (t.joinable()) t.join();
   tp_.reset();
-  if (host_buf_) (void)hipHostFree(host_buf_);
-  if (rx_buf_) (void)hipHostFree(rx_buf_);
-  if (stream_) (void)hipStreamDestroy(stream_);
-  if (rx_stream_) (void)hipStreamDestroy(rx_stream_);
 }
 
 void EpProxy::establish_flows(const std::vector<std::string>& tp_md,
@@ -86,7 +87,7 @@ void EpProxy::start() {
 
 // D2H-copy `count` rows from dev_rows and stream them over `flow` in
 // staging-sized chunks, then the metas.
-void EpProxy::ship_rows(uint64_t flow, WireHdr const& h,
+void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                         void const* dev_rows,
                         uint32_t const* dev_metas_or_null,
                         std::vector<uint32_t> const* host_metas) {
@@ -96,11 +97,11 @@ void EpProxy::ship_rows(uint64_t flow, WireHdr const& h,
   tp_->send_msg(flow, &h, sizeof(h));
   for (size_t off = 0; off < total; off += stage_bytes_) {
     size_t const n = std::min(stage_bytes_, total - off);
-    UCCL_CHECK_HIP(hipMemcpyAsync(host_buf_,
+    UCCL_CHECK_HIP(hipMemcpyAsync(lane.buf,
                                   static_cast<char const*>(dev_rows) + off,
-                                  n, hipMemcpyDeviceToHost, stream_));
-    UCCL_CHECK_HIP(hipStreamSynchronize(stream_));
-    tp_->send_msg(flow, host_buf_, n);
+                                  n, hipMemcpyDeviceToHost, lane.stream));
+    UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+    tp_->send_msg(flow, lane.buf, n);
   }
   std::vector<uint32_t> metas;
   if (host_metas) {
@@ -109,14 +110,15 @@ void EpProxy::ship_rows(uint64_t flow, WireHdr const& h,
     metas.resize(h.count);
     UCCL_CHECK_HIP(hipMemcpyAsync(metas.data(), dev_metas_or_null,
                                   h.count * sizeof(uint32_t),
-                                  hipMemcpyDeviceToHost, stream_));
-    UCCL_CHECK_HIP(hipStreamSynchronize(stream_));
+                                  hipMemcpyDeviceToHost, lane.stream));
+    UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
   }
   if (h.count) tp_->send_msg(flow, metas.data(), h.count * sizeof(uint32_t));
 }
 
 void EpProxy::ring_loop() {
   (void)hipSetDevice(device_);
+  Lane lane(device_, stage_bytes_);
   uint64_t head = 0;
   while (!stop_) {
     uint64_t tail =
@@ -139,7 +141,7 @@ void EpProxy::ring_loop() {
       WireHdr h{kDisp, c.seq32, static_cast<uint32_t>(e % v_.local_experts),
                 static_cast<uint32_t>(v_.rank), c.c};
       try {
-        ship_rows(flows_[dst], h, egress_row(heap_, v_, c.b),
+        ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
                   egress_meta(heap_, v_, c.b), nullptr);
       } catch (std::exception const&) {
         return;  // transport closed
@@ -150,6 +152,7 @@ void EpProxy::ring_loop() {
 
 void EpProxy::comb_tx_loop() {
   (void)hipSetDevice(device_);
+  Lane lane(device_, stage_bytes_);
   while (true) {
     CombTask task;
     {
@@ -178,7 +181,7 @@ void EpProxy::comb_tx_loop() {
               ((static_cast<size_t>(le) * v_.world * v_.max_tokens +
                 slot0)) *
                   row_bytes;
-          ship_rows(flows_[src], h, rows,
+          ship_rows(lane, flows_[src], h, rows,
                     disp_meta_ptr(heap_, v_, le, slot0), nullptr);
         }
         WireHdr done{kCombDone, static_cast<uint32_t>(task.seq), 0,
@@ -215,6 +218,7 @@ void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
 
 void EpProxy::rx_loop(int peer) {
   (void)hipSetDevice(device_);
+  Lane lane(device_, stage_bytes_);
   uint64_t const flow = flows_[peer];
   size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
   std::vector<uint32_t> metas;
@@ -224,16 +228,16 @@ void EpProxy::rx_loop(int peer) {
       tp_->recv_msg(flow, &h, sizeof(h));
       if (h.kind == kCombDone) {
         // all combine payloads from `peer` for this seq have been
-        // scattered (rx_stream_ is in-order); publish the flag
-        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+        // scattered (lane.stream is in-order); publish the flag
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
         // NB: must NOT use the legacy default stream here — it would wait
-        // for the peer's spinning wait-kernel (deadlock). rx_stream_ is
+        // for the peer's spinning wait-kernel (deadlock). lane.stream is
         // non-blocking.
         uint64_t const seq = h.seq32;
         UCCL_CHECK_HIP(hipMemcpyAsync(comb_flag_ptr(heap_, v_, peer), &seq,
                                       sizeof(seq), hipMemcpyHostToDevice,
-                                      rx_stream_));
-        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+                                      lane.stream));
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
         continue;
       }
       size_t const total = h.count * row_bytes;
@@ -246,10 +250,10 @@ void EpProxy::rx_loop(int peer) {
                           : ingress_row(heap_, v_, ing0);
       for (size_t off = 0; off < total; off += stage_bytes_) {
         size_t const n = std::min(stage_bytes_, total - off);
-        tp_->recv_msg(flow, rx_buf_, n);
-        UCCL_CHECK_HIP(hipMemcpyAsync(dev_dst + off, rx_buf_, n,
-                                      hipMemcpyHostToDevice, rx_stream_));
-        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+        tp_->recv_msg(flow, lane.buf, n);
+        UCCL_CHECK_HIP(hipMemcpyAsync(dev_dst + off, lane.buf, n,
+                                      hipMemcpyHostToDevice, lane.stream));
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
       }
       metas.resize(h.count);
       if (h.count)
@@ -259,22 +263,22 @@ void EpProxy::rx_loop(int peer) {
             disp_meta_ptr(heap_, v_, h.le,
                           static_cast<size_t>(h.src) * v_.max_tokens),
             metas.data(), h.count * sizeof(uint32_t),
-            hipMemcpyHostToDevice, rx_stream_));
-        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+            hipMemcpyHostToDevice, lane.stream));
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
         uint64_t const tagged =
             (static_cast<uint64_t>(h.seq32) << 32) | h.count;
         UCCL_CHECK_HIP(hipMemcpyAsync(disp_count_ptr(heap_, v_, h.le, h.src),
                                       &tagged, sizeof(tagged),
-                                      hipMemcpyHostToDevice, rx_stream_));
-        UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+                                      hipMemcpyHostToDevice, lane.stream));
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
       } else {  // kComb: scatter ingress rows into comb_x cells by meta
         if (h.count) {
           UCCL_CHECK_HIP(hipMemcpyAsync(ingress_meta(heap_, v_, ing0),
                                         metas.data(),
                                         h.count * sizeof(uint32_t),
-                                        hipMemcpyHostToDevice, rx_stream_));
-          launch_ep_comb_scatter(v_, ing0, h.count, rx_stream_);
-          UCCL_CHECK_HIP(hipStreamSynchronize(rx_stream_));
+                                        hipMemcpyHostToDevice, lane.stream));
+          launch_ep_comb_scatter(v_, ing0, h.count, lane.stream);
+          UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
         }
       }
     }

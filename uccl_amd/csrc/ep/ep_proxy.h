@@ -56,12 +56,22 @@ class EpProxy {
     uint64_t count;
   };
 
+  // Per-thread pinned staging + non-blocking stream: the ring thread,
+  // the combine-tx thread, and every per-peer rx thread run concurrently
+  // and must never share host staging (corruption observed at world=4).
+  struct Lane {
+    void* buf = nullptr;
+    hipStream_t stream = nullptr;
+    Lane(int device, size_t bytes);
+    ~Lane();
+  };
+
   void ring_loop();
   void rx_loop(int peer);
   void comb_tx_loop();
   int flow_peer(uint64_t flow) const;
-  void ship_rows(uint64_t flow, WireHdr const& h, void const* dev_rows,
-                 uint32_t const* dev_metas_or_null,
+  void ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
+                 void const* dev_rows, uint32_t const* dev_metas_or_null,
                  std::vector<uint32_t> const* host_metas);
 
   EpView v_;
@@ -76,10 +86,6 @@ class EpProxy {
   // the ring thread and the combine-tx thread share peer flows
   std::vector<std::unique_ptr<std::mutex>> flow_mu_;
 
-  hipStream_t stream_ = nullptr;     // non-blocking proxy stream (tx)
-  hipStream_t rx_stream_ = nullptr;  // non-blocking proxy stream (rx)
-  void* host_buf_ = nullptr;         // pinned staging (tx)
-  void* rx_buf_ = nullptr;           // pinned staging (rx)
   size_t stage_bytes_ = 0;
 
   struct CombTask {
