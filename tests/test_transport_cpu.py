@@ -120,6 +120,16 @@ def test_transport_loss_recovery(loss):
     assert st.msgs_sent == 2
 
 
+def test_transport_ack_loss_recovery():
+    # dropped ACKs leave data unacknowledged; RTO must retransmit and the
+    # receiver must dedupe (payload integrity preserved)
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_ACK_LOSS_PCT=30)
+    xfer(a, b, fa, fb, 500000, 41)
+    st = a.stats()
+    assert st.msgs_sent == 1
+    assert st.retransmits + st.rto_retransmits > 0
+
+
 def test_transport_stats_cc():
     C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0)
     xfer(a, b, fa, fb, 8 << 20, 21)

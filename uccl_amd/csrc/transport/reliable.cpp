@@ -192,6 +192,15 @@ struct TransportEndpoint::Impl {
   }
 
   void send_ack(Flow& f, int sock_idx, sockaddr_in const& to) {
+    // independent ACK-loss injection (tests RTO-driven recovery when the
+    // reverse path drops)
+    static int const ack_loss =
+        static_cast<int>(env_int("UCCL_TP_ACK_LOSS_PCT", 0));
+    if (ack_loss > 0 &&
+        inject_drop(static_cast<uint32_t>(st.acks_sent), 7, ack_loss)) {
+      ++st.injected_drops;
+      return;
+    }
     AckHdr a{kMagic, kAck, f.id, f.rx_cum, 0, 0, 0, f.last_data_ts};
     for (auto const& [csn, _] : f.rx_ooo) {
       uint32_t const d = csn - f.rx_cum;
