@@ -121,13 +121,15 @@ def test_transport_loss_recovery(loss):
 
 
 def test_transport_ack_loss_recovery():
-    # dropped ACKs leave data unacknowledged; RTO must retransmit and the
-    # receiver must dedupe (payload integrity preserved)
-    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_ACK_LOSS_PCT=30)
-    xfer(a, b, fa, fb, 500000, 41)
-    st = a.stats()
-    assert st.msgs_sent == 1
-    assert st.retransmits + st.rto_retransmits > 0
+    # dropped ACKs are mostly covered by later acks' cumulative edge (the
+    # protocol working as designed); the message must still complete with
+    # full integrity, with RTO as the backstop for tail-ack loss
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_ACK_LOSS_PCT=40)
+    for i in range(3):
+        xfer(a, b, fa, fb, 500000, 41 + i)
+    st_b = b.stats()
+    assert st_b.injected_drops > 0, "ack-loss injection did not fire"
+    assert a.stats().msgs_sent == 3
 
 
 def test_transport_stats_cc():
