@@ -56,7 +56,7 @@ def xfer(a, b, fa, fb, nbytes, seed):
         b.recv(fb, dst)
         done["rx"] = True
 
-    t = threading.Thread(target=rx)
+    t = threading.Thread(target=rx, daemon=True)
     t.start()
     a.send(fa, src)
     t.join(timeout=60)
@@ -88,7 +88,7 @@ def test_transport_ordering_and_bidirectional():
         for o in outs:
             b.recv(fb, o)
 
-    t = threading.Thread(target=rx)
+    t = threading.Thread(target=rx, daemon=True)
     t.start()
     for m in msgs:
         a.send(fa, m)
@@ -102,7 +102,7 @@ def test_transport_ordering_and_bidirectional():
     def rx2():
         a.recv(fa, got)
 
-    t = threading.Thread(target=rx2)
+    t = threading.Thread(target=rx2, daemon=True)
     t.start()
     b.send(fb, back)
     t.join(timeout=60)
@@ -162,7 +162,7 @@ def test_transport_star_multi_flow():
         for _ in range(3):
             hub_flows.append(hub.accept())
 
-    t = threading.Thread(target=acceptor)
+    t = threading.Thread(target=acceptor, daemon=True)
     t.start()
     spoke_flows = [sp.connect(hub.metadata()) for sp in spokes]
     t.join(timeout=30)
@@ -183,12 +183,12 @@ def test_transport_star_multi_flow():
         def hub_io(i=idx, fl=f):
             hub.recv(fl, got_hub[i])
             hub.send(fl, from_hub[i])
-        threads.append(threading.Thread(target=hub_io))
+        threads.append(threading.Thread(target=hub_io, daemon=True))
     for i, (sp, f) in enumerate(zip(spokes, spoke_flows)):
         def spoke_io(i=i, sp=sp, fl=f):
             sp.send(fl, to_hub[i])
             sp.recv(fl, got_spoke[i])
-        threads.append(threading.Thread(target=spoke_io))
+        threads.append(threading.Thread(target=spoke_io, daemon=True))
     for th in threads:
         th.start()
     for th in threads:

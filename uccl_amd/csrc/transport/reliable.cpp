@@ -123,6 +123,7 @@ struct TransportEndpoint::Impl {
   int num_paths;
   size_t chunk_bytes;
   int loss_pct;
+  int ack_loss_pct;
   std::vector<int> socks;          // UDP path sockets
   std::vector<uint16_t> ports;
   int wake_fd = -1;                // self-addressed UDP for wakeups
@@ -192,12 +193,12 @@ struct TransportEndpoint::Impl {
   }
 
   void send_ack(Flow& f, int sock_idx, sockaddr_in const& to) {
-    // independent ACK-loss injection (tests RTO-driven recovery when the
-    // reverse path drops)
-    static int const ack_loss =
-        static_cast<int>(env_int("UCCL_TP_ACK_LOSS_PCT", 0));
-    if (ack_loss > 0 &&
-        inject_drop(static_cast<uint32_t>(st.acks_sent), 7, ack_loss)) {
+    // independent ACK-loss injection (exercises cumulative-ack coverage
+    // and the RTO backstop on reverse-path drops)
+    // key the decision on data_recv (which always advances) — keying on
+    // acks_sent would freeze the hash after a drop and drop forever
+    if (ack_loss_pct > 0 &&
+        inject_drop(static_cast<uint32_t>(st.data_recv), 7, ack_loss_pct)) {
       ++st.injected_drops;
       return;
     }
@@ -485,6 +486,8 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   impl_->num_paths = num_paths;
   impl_->chunk_bytes = chunk_bytes;
   impl_->loss_pct = static_cast<int>(env_int("UCCL_TP_LOSS_PCT", 0));
+  impl_->ack_loss_pct =
+      static_cast<int>(env_int("UCCL_TP_ACK_LOSS_PCT", 0));
   for (int i = 0; i < num_paths; ++i) {
     int s = ::socket(AF_INET, SOCK_DGRAM, 0);
     UCCL_CHECK(s >= 0) << "udp socket";
