@@ -20,6 +20,7 @@
 #include <torch/csrc/distributed/c10d/Types.hpp>
 #include <torch/csrc/distributed/c10d/Work.hpp>
 
+#include "../core/env.h"
 #include "../core/log.h"
 #include "communicator.h"
 
@@ -46,40 +47,60 @@ Dtype to_dtype(const at::Tensor& t) {
   }
 }
 
+// Work with ProcessGroupNCCL-style semantics: ops are enqueued on the
+// caller's stream; wait()/synchronize() make the *current stream* wait on
+// the op's completion event (no host block), so DDP keeps compute/comm
+// overlap; getFuture() completes asynchronously off a watcher thread.
 class UcclWork : public c10d::Work {
  public:
-  UcclWork(int rank, c10d::OpType op, hipStream_t stream,
+  UcclWork(int rank, c10d::OpType op, hipStream_t stream, int device,
            std::vector<at::Tensor> results)
-      : c10d::Work(rank, op), results_(std::move(results)) {
-    (void)hipEventCreateWithFlags(&event_, hipEventDisableTiming);
-    (void)hipEventRecord(event_, stream);
-  }
-  ~UcclWork() override {
-    if (event_) (void)hipEventDestroy(event_);
+      : c10d::Work(rank, op),
+        device_(device),
+        event_(new EventHolder),
+        results_(std::move(results)) {
+    (void)hipEventCreateWithFlags(&event_->ev, hipEventDisableTiming);
+    (void)hipEventRecord(event_->ev, stream);
   }
 
   bool isCompleted() override {
-    return hipEventQuery(event_) == hipSuccess;
+    return hipEventQuery(event_->ev) == hipSuccess;
   }
   bool isSuccess() const override { return true; }
   std::vector<at::Tensor> result() override { return results_; }
-  void synchronize() override { (void)hipEventSynchronize(event_); }
+  void synchronize() override {
+    hipStream_t cur =
+        at::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_).stream();
+    (void)hipStreamWaitEvent(cur, event_->ev, 0);
+  }
   bool wait(std::chrono::milliseconds) override {
     synchronize();
+    static bool const blocking =
+        uccl::env_bool("UCCL_BLOCKING_WAIT", false);
+    if (blocking) (void)hipEventSynchronize(event_->ev);
     return true;
   }
   c10::intrusive_ptr<c10::ivalue::Future> getFuture() override {
-    // Stream-ordered semantics: consumers on the enqueue stream are
-    // already ordered; complete the future with the result tensors.
     auto fut = c10::make_intrusive<c10::ivalue::Future>(
         c10::ListType::create(c10::TensorType::get()));
-    (void)hipEventSynchronize(event_);
-    fut->markCompleted(c10::IValue(results_));
+    auto holder = event_;  // keeps the event alive past Work destruction
+    auto results = results_;
+    std::thread([fut, holder, results]() {
+      (void)hipEventSynchronize(holder->ev);
+      fut->markCompleted(c10::IValue(results));
+    }).detach();
     return fut;
   }
 
  private:
-  hipEvent_t event_ = nullptr;
+  struct EventHolder {
+    hipEvent_t ev = nullptr;
+    ~EventHolder() {
+      if (ev) (void)hipEventDestroy(ev);
+    }
+  };
+  int device_;
+  std::shared_ptr<EventHolder> event_;
   std::vector<at::Tensor> results_;
 };
 
@@ -117,8 +138,8 @@ class UcclBackend : public c10d::Backend {
     check(t);
     hipStream_t s = cur_stream(comm_->device());
     comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s);
-    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::ALLREDUCE,
-                                         s, tensors);
+    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::ALLREDUCE, s,
+                                         comm_->device(), tensors);
   }
 
   c10::intrusive_ptr<c10d::Work> broadcast(
@@ -130,8 +151,8 @@ class UcclBackend : public c10d::Backend {
     hipStream_t s = cur_stream(comm_->device());
     comm_->broadcast(t.data_ptr(), t.numel() * t.element_size(),
                      Dtype::kU8, static_cast<int>(opts.rootRank), s);
-    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::BROADCAST,
-                                         s, tensors);
+    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::BROADCAST, s,
+                                         comm_->device(), tensors);
   }
 
   c10::intrusive_ptr<c10d::Work> _allgather_base(
@@ -145,7 +166,7 @@ class UcclBackend : public c10d::Backend {
     comm_->all_gather(output.data_ptr(), input.data_ptr(),
                       input.numel() * input.element_size(), Dtype::kU8, s);
     return c10::make_intrusive<UcclWork>(
-        getRank(), c10d::OpType::_ALLGATHER_BASE, s,
+        getRank(), c10d::OpType::_ALLGATHER_BASE, s, comm_->device(),
         std::vector<at::Tensor>{output});
   }
 
@@ -164,8 +185,8 @@ class UcclBackend : public c10d::Backend {
     for (int r = 0; r < getSize(); ++r)
       outputs[0][r].view(-1).copy_(
           flat.narrow(0, r * in.numel(), in.numel()), /*non_blocking=*/true);
-    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::ALLGATHER,
-                                         s, outputs[0]);
+    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::ALLGATHER, s,
+                                         comm_->device(), outputs[0]);
   }
 
   c10::intrusive_ptr<c10d::Work> _reduce_scatter_base(
@@ -180,7 +201,7 @@ class UcclBackend : public c10d::Backend {
     comm_->reduce_scatter(output.data_ptr(), input.data_ptr(),
                           output.numel(), to_dtype(input), s);
     return c10::make_intrusive<UcclWork>(
-        getRank(), c10d::OpType::_REDUCE_SCATTER_BASE, s,
+        getRank(), c10d::OpType::_REDUCE_SCATTER_BASE, s, comm_->device(),
         std::vector<at::Tensor>{output});
   }
 
@@ -199,7 +220,7 @@ class UcclBackend : public c10d::Backend {
                       input.numel() / getSize() * input.element_size(),
                       Dtype::kU8, s);
     return c10::make_intrusive<UcclWork>(
-        getRank(), c10d::OpType::ALLTOALL_BASE, s,
+        getRank(), c10d::OpType::ALLTOALL_BASE, s, comm_->device(),
         std::vector<at::Tensor>{output});
   }
 
@@ -217,6 +238,7 @@ class UcclBackend : public c10d::Backend {
     hipStream_t s = cur_stream(comm_->device());
     comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s);
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::REDUCE, s,
+                                         comm_->device(),
                                          tensors);
   }
 
@@ -242,6 +264,7 @@ class UcclBackend : public c10d::Backend {
       results = outputs[0];
     }
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::GATHER, s,
+                                         comm_->device(),
                                          results);
   }
 
@@ -267,7 +290,8 @@ class UcclBackend : public c10d::Backend {
     out.view(-1).copy_(
         flat.narrow(0, getRank() * out.numel(), out.numel()), true);
     return c10::make_intrusive<UcclWork>(
-        getRank(), c10d::OpType::SCATTER, s, std::vector<at::Tensor>{out});
+        getRank(), c10d::OpType::SCATTER, s, comm_->device(),
+        std::vector<at::Tensor>{out});
   }
 
   c10::intrusive_ptr<c10d::Work> send(std::vector<at::Tensor>& tensors,
@@ -278,6 +302,7 @@ class UcclBackend : public c10d::Backend {
     hipStream_t s = cur_stream(comm_->device());
     comm_->send(t.data_ptr(), t.numel() * t.element_size(), dst, s);
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::SEND, s,
+                                         comm_->device(),
                                          tensors);
   }
 
@@ -289,6 +314,7 @@ class UcclBackend : public c10d::Backend {
     hipStream_t s = cur_stream(comm_->device());
     comm_->recv(t.data_ptr(), t.numel() * t.element_size(), src, s);
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::RECV, s,
+                                         comm_->device(),
                                          tensors);
   }
 
@@ -297,8 +323,8 @@ class UcclBackend : public c10d::Backend {
     hipStream_t s = cur_stream(comm_->device());
     comm_->barrier(s);
     (void)hipStreamSynchronize(s);  // NCCL-like host-blocking barrier
-    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::BARRIER,
-                                         s, std::vector<at::Tensor>{});
+    return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::BARRIER, s,
+                                         comm_->device(), std::vector<at::Tensor>{});
   }
 
  private:
