@@ -301,8 +301,10 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
         } else {
           std::vector<char> tmp(h.c);
           net::recv_all(c->fd, tmp.data(), h.c);
-          UCCL_CHECK_HIP(hipMemcpy(dst, tmp.data(), h.c,
-                                   hipMemcpyHostToDevice));
+          hipStream_t cs = copy_stream();
+          UCCL_CHECK_HIP(hipMemcpyAsync(dst, tmp.data(), h.c,
+                                        hipMemcpyHostToDevice, cs));
+          UCCL_CHECK_HIP(hipStreamSynchronize(cs));
         }
         if (h.d) c->send_msg(MsgHdr{kWriteAck, 0, 0, 0, h.d});
         break;
@@ -343,8 +345,10 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
           c->send_msg(resp, src, h.c);
         } else {
           std::vector<char> tmp(h.c);
-          UCCL_CHECK_HIP(hipMemcpy(tmp.data(), src, h.c,
-                                   hipMemcpyDeviceToHost));
+          hipStream_t cs = copy_stream();
+          UCCL_CHECK_HIP(hipMemcpyAsync(tmp.data(), src, h.c,
+                                        hipMemcpyDeviceToHost, cs));
+          UCCL_CHECK_HIP(hipStreamSynchronize(cs));
           c->send_msg(resp, tmp.data(), h.c);
         }
         break;
@@ -437,11 +441,13 @@ void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
   std::lock_guard<std::mutex> g(c.tx_mu);
   MsgHdr h{kSendData, bytes, 0, 0, 0};
   net::send_all(c.fd, &h, sizeof(h));
+  hipStream_t cs = copy_stream();
   for (size_t off = 0; off < bytes; off += staging_bytes_) {
     size_t n = std::min(staging_bytes_, bytes - off);
-    UCCL_CHECK_HIP(hipMemcpy(staging_,
-                             static_cast<char const*>(ptr) + off, n,
-                             hipMemcpyDeviceToHost));
+    UCCL_CHECK_HIP(hipMemcpyAsync(staging_,
+                                  static_cast<char const*>(ptr) + off, n,
+                                  hipMemcpyDeviceToHost, cs));
+    UCCL_CHECK_HIP(hipStreamSynchronize(cs));
     net::send_all(c.fd, staging_, n);
   }
 }
@@ -453,8 +459,10 @@ void Endpoint::copy_to_user(RxItem& item, void* dst, size_t bytes,
   if (!is_gpu(device)) {
     memcpy(dst, item.data.data(), item.bytes);
   } else {
-    UCCL_CHECK_HIP(hipMemcpy(dst, item.data.data(), item.bytes,
-                             hipMemcpyHostToDevice));
+    hipStream_t cs = copy_stream();
+    UCCL_CHECK_HIP(hipMemcpyAsync(dst, item.data.data(), item.bytes,
+                                  hipMemcpyHostToDevice, cs));
+    UCCL_CHECK_HIP(hipStreamSynchronize(cs));
   }
 }
 
@@ -504,7 +512,10 @@ void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
                bytes);
   } else {
     std::vector<char> tmp(bytes);
-    UCCL_CHECK_HIP(hipMemcpy(tmp.data(), ptr, bytes, hipMemcpyDeviceToHost));
+    hipStream_t cs = copy_stream();
+    UCCL_CHECK_HIP(hipMemcpyAsync(tmp.data(), ptr, bytes,
+                                  hipMemcpyDeviceToHost, cs));
+    UCCL_CHECK_HIP(hipStreamSynchronize(cs));
     c.send_msg(MsgHdr{kWriteData, ad.mr_id, ad.offset, bytes, token},
                tmp.data(), bytes);
   }
