@@ -124,8 +124,8 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
   }
 }
 
-// int specialization of the scalar tail accumulate needs integer math; to
-// keep one template, route int through double-precision-free path:
+// int specialization predates TailAcc and is now equivalent to the generic
+// template's int instantiation; kept as the validated int entry point.
 template <>
 __global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
                                          size_t count) {
