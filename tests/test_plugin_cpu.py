@@ -8,13 +8,21 @@ from pathlib import Path
 REPO = Path(__file__).resolve().parent.parent
 
 
-def test_plugin_harness():
+import os
+
+import pytest
+
+
+@pytest.mark.parametrize("plane", ["multipath", "tcp"])
+def test_plugin_harness(plane):
     from uccl_amd._build import build_plugin
 
     so = build_plugin()
     harness = so.parent / "plugin_test"
+    env = dict(os.environ)
+    env["UCCL_NET_TRANSPORT"] = plane
     r = subprocess.run([str(harness), str(so)], capture_output=True,
-                       timeout=120)
+                       timeout=120, env=env)
     out = r.stdout.decode() + r.stderr.decode()
     assert r.returncode == 0, out
     assert "PLUGIN HARNESS OK" in out

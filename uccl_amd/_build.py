@@ -60,7 +60,8 @@ def build_plugin(verbose: bool = False) -> Path:
     harness = plugdir / "plugin_test"
     deps = [src, harness_src] + list((CSRC / "plugin").glob("*.h")) + [
         CSRC / "core" / "net.h", CSRC / "core" / "log.h",
-        CSRC / "core" / "env.h"]
+        CSRC / "core" / "env.h", CSRC / "transport" / "reliable.h",
+        CSRC / "transport" / "reliable.cpp"]
     capi = plugdir / "libuccl_p2p.so"
     capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
                  CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h"]
@@ -71,9 +72,10 @@ def build_plugin(verbose: bool = False) -> Path:
         return target
     import subprocess as sp
 
+    reliable = CSRC / "transport" / "reliable.cpp"
     for cmd in (
-        ["g++", "-O2", "-std=c++17", "-fPIC", "-shared", str(src), "-o",
-         str(target), "-pthread"],
+        ["g++", "-O2", "-std=c++17", "-fPIC", "-shared", str(src),
+         str(reliable), "-o", str(target), "-pthread"],
         ["g++", "-O2", "-std=c++17", str(harness_src), "-o", str(harness),
          "-ldl", "-pthread"],
     ):

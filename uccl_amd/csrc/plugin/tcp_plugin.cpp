@@ -1,21 +1,24 @@
 // librccl-net-uccl.so — RCCL network plugin (net ABI v6).
 //
 // Parity role: the reference's collective/rdma/nccl_plugin.cc vtable
-// (:85-633) — the drop-in transport under stock RCCL. This first transport
-// is a clean TCP implementation (host pointers; RCCL stages GPU data
-// through its own pinned buffers, exactly like NCCL's built-in socket
-// transport) with per-comm TX/RX threads, tag-matched frames, and
-// non-blocking connect/accept state machines per the plugin contract.
-// The multipath reliable transport (csrc/transport/) slots in behind the
-// same vtable for RDMA-capable fabrics.
+// (:85-633) — the drop-in transport under stock RCCL, carrying NCCL's
+// traffic over the software multipath reliable transport
+// (csrc/transport/: chunking, path spraying, SACK selective repeat,
+// Timely/Swift CC) exactly like the reference's plugin rides its
+// UcclRDMAEngine. Two data planes, selected by UCCL_NET_TRANSPORT:
+//   multipath (default) — TransportEndpoint flows (message-oriented)
+//   tcp               — plain stream sockets (NCCL-socket-transport-like)
+// Host pointers only (RCCL stages GPU data through its own pinned
+// buffers, like NCCL's built-in socket transport).
 //
-// Pure sockets + pthreads: no HIP dependency, so the plugin also serves
-// as a CPU-testable artifact (tests drive the vtable via dlopen).
+// No HIP dependency, so the plugin is fully CPU-testable (the dlopen
+// harness drives the vtable over both data planes).
 
 #include <fcntl.h>
 #include <poll.h>
 
 #include <atomic>
+#include <map>
 #include <condition_variable>
 #include <cstring>
 #include <deque>
@@ -27,6 +30,7 @@
 #include "../core/env.h"
 #include "../core/log.h"
 #include "../core/net.h"
+#include "../transport/reliable.h"
 #include "net_plugin_abi.h"
 
 namespace {
@@ -70,12 +74,60 @@ struct SendOp {
   Request* req;
 };
 
+bool use_multipath() {
+  static bool v = uccl::env_str("UCCL_NET_TRANSPORT", "multipath") !=
+                  std::string("tcp");
+  return v;
+}
+
+// --- multipath fabric singleton -------------------------------------------
+// One TransportEndpoint per process; RCCL's many listenComms are
+// distinguished by a nonce carried in the connect tag, so a global
+// acceptor thread can route inbound flows to the right listener queue.
+struct MpFabric {
+  uccl::transport::TransportEndpoint tp{
+      static_cast<int>(uccl::env_int("UCCL_NET_PATHS", 8)),
+      static_cast<size_t>(uccl::env_int("UCCL_NET_CHUNK", 16384))};
+  std::mutex mu;
+  std::condition_variable cv;
+  std::map<uint64_t, std::deque<uint64_t>> queues;  // nonce -> flows
+  std::atomic<uint64_t> next_nonce{1};
+  std::thread acceptor;
+  MpFabric() {
+    acceptor = std::thread([this] {
+      try {
+        while (true) {
+          uint64_t tag = 0;
+          uint64_t flow = tp.accept(&tag);
+          std::lock_guard<std::mutex> g(mu);
+          queues[tag].push_back(flow);
+          cv.notify_all();
+        }
+      } catch (std::exception const&) {
+      }
+    });
+    acceptor.detach();  // process-lifetime singleton
+  }
+  static MpFabric& get() {
+    static MpFabric f;
+    return f;
+  }
+};
+
+struct MpHandle {
+  uint64_t nonce;
+  uint16_t md_len;
+  char md[96];
+};
+
 struct ListenComm {
-  int fd = -1;
+  int fd = -1;        // tcp mode
+  uint64_t nonce = 0;  // multipath mode
 };
 
 struct Comm {
-  int fd = -1;
+  int fd = -1;        // tcp mode
+  uint64_t flow = 0;  // multipath mode (0 = tcp)
   bool sender = false;
   std::atomic<bool> alive{true};
   std::thread worker;
@@ -89,13 +141,7 @@ struct Comm {
   std::mutex mu;
   std::condition_variable cv;
 
-  ~Comm() {
-    alive = false;
-    cv.notify_all();
-    if (fd >= 0) ::shutdown(fd, SHUT_RDWR);
-    if (worker.joinable()) worker.join();
-    if (fd >= 0) ::close(fd);
-  }
+  ~Comm();
 };
 
 struct WireHdr {
@@ -103,6 +149,38 @@ struct WireHdr {
   int32_t tag;
   int32_t pad;
 };
+
+Comm::~Comm() {
+  alive = false;
+  cv.notify_all();
+  if (fd >= 0) ::shutdown(fd, SHUT_RDWR);
+  if (flow) MpFabric::get().tp.close_flow(flow);  // unblock recv_msg
+  if (worker.joinable()) worker.join();
+  if (fd >= 0) ::close(fd);
+}
+
+// channel ops spanning both data planes (message framing == byte framing:
+// the tcp path streams the same {hdr}{payload} sequence the multipath
+// path sends as two discrete reliable messages)
+void chan_send(Comm* c, void const* buf, size_t n) {
+  if (c->flow) {
+    MpFabric::get().tp.send_msg(c->flow, buf, n);
+  } else {
+    send_all(c->fd, buf, n);
+  }
+}
+
+bool chan_recv(Comm* c, void* buf, size_t n) {
+  if (c->flow) {
+    try {
+      MpFabric::get().tp.recv_msg(c->flow, buf, n);
+      return true;
+    } catch (std::exception const&) {
+      return false;
+    }
+  }
+  return recv_all(c->fd, buf, n);
+}
 
 void tx_loop(Comm* c) {
   while (c->alive) {
@@ -115,8 +193,8 @@ void tx_loop(Comm* c) {
       c->sendq.pop_front();
     }
     WireHdr h{static_cast<uint64_t>(op.size), op.tag, 0};
-    send_all(c->fd, &h, sizeof(h));
-    if (op.size) send_all(c->fd, op.data, op.size);
+    chan_send(c, &h, sizeof(h));
+    if (op.size) chan_send(c, op.data, op.size);
     op.req->size = op.size;
     op.req->done.store(1, std::memory_order_release);
   }
@@ -125,7 +203,7 @@ void tx_loop(Comm* c) {
 void rx_loop(Comm* c) {
   while (c->alive) {
     WireHdr h{};
-    if (!recv_all(c->fd, &h, sizeof(h))) return;
+    if (!chan_recv(c, &h, sizeof(h))) return;
     // try to match a posted recv by tag
     PostedRecv pr{};
     bool matched = false;
@@ -146,7 +224,7 @@ void rx_loop(Comm* c) {
              h.tag, (unsigned long)h.bytes, pr.cap);
         return;
       }
-      if (h.bytes) recv_all(c->fd, pr.data, h.bytes);
+      if (h.bytes) chan_recv(c, pr.data, h.bytes);
       pr.req->size = static_cast<int>(h.bytes);
       pr.req->done.store(1, std::memory_order_release);
     } else {
@@ -154,7 +232,7 @@ void rx_loop(Comm* c) {
       f.bytes = h.bytes;
       f.tag = h.tag;
       f.data.resize(h.bytes);
-      if (h.bytes) recv_all(c->fd, f.data.data(), h.bytes);
+      if (h.bytes) chan_recv(c, f.data.data(), h.bytes);
       std::lock_guard<std::mutex> g(c->mu);
       c->unmatched.push_back(std::move(f));
     }
@@ -167,7 +245,8 @@ void rx_loop(Comm* c) {
 
 ncclResult_t p_init(ncclDebugLogger_t logfn) {
   g_log = logfn;
-  PLOG(NCCL_LOG_INFO, "uccl-net tcp plugin init");
+  PLOG(NCCL_LOG_INFO, "uccl-net plugin init (%s data plane)",
+       use_multipath() ? "multipath" : "tcp");
   return ncclSuccess;
 }
 
@@ -193,6 +272,24 @@ ncclResult_t p_getProperties(int dev, ncclNetProperties_v6_t* props) {
 }
 
 ncclResult_t p_listen(int dev, void* opaque, void** listenComm) {
+  if (use_multipath()) {
+    auto& f = MpFabric::get();
+    auto* lc = new ListenComm();
+    lc->nonce = f.next_nonce.fetch_add(1);
+    {
+      std::lock_guard<std::mutex> g(f.mu);
+      f.queues[lc->nonce];  // create the routing queue
+    }
+    MpHandle h{};
+    h.nonce = lc->nonce;
+    std::string md = f.tp.metadata();
+    h.md_len = static_cast<uint16_t>(md.size());
+    memcpy(h.md, md.data(), std::min(md.size(), sizeof(h.md)));
+    static_assert(sizeof(MpHandle) <= NCCL_NET_HANDLE_MAXSIZE, "handle");
+    memcpy(opaque, &h, sizeof(h));
+    *listenComm = lc;
+    return ncclSuccess;
+  }
   auto* lc = new ListenComm();
   uint16_t port = 0;
   lc->fd = uccl::net::listen_on(&port);
@@ -209,6 +306,22 @@ ncclResult_t p_listen(int dev, void* opaque, void** listenComm) {
 }
 
 ncclResult_t p_connect(int dev, void* opaque, void** sendComm) {
+  if (use_multipath()) {
+    MpHandle h{};
+    memcpy(&h, opaque, sizeof(h));
+    try {
+      uint64_t flow = MpFabric::get().tp.connect(
+          std::string(h.md, h.md_len), h.nonce);
+      auto* c = new Comm();
+      c->flow = flow;
+      c->sender = true;
+      c->worker = std::thread(tx_loop, c);
+      *sendComm = c;
+      return ncclSuccess;
+    } catch (std::exception const&) {
+      return ncclSystemError;
+    }
+  }
   Handle h{};
   memcpy(&h, opaque, sizeof(h));
   // non-blocking contract: attempt one quick connect; if not ready yet,
@@ -236,6 +349,25 @@ ncclResult_t p_connect(int dev, void* opaque, void** sendComm) {
 
 ncclResult_t p_accept(void* listenComm, void** recvComm) {
   auto* lc = static_cast<ListenComm*>(listenComm);
+  if (use_multipath()) {
+    auto& f = MpFabric::get();
+    uint64_t flow = 0;
+    {
+      std::lock_guard<std::mutex> g(f.mu);
+      auto& q = f.queues[lc->nonce];
+      if (q.empty()) {
+        *recvComm = nullptr;  // not ready; RCCL retries
+        return ncclSuccess;
+      }
+      flow = q.front();
+      q.pop_front();
+    }
+    auto* c = new Comm();
+    c->flow = flow;
+    c->worker = std::thread(rx_loop, c);
+    *recvComm = c;
+    return ncclSuccess;
+  }
   int fd = ::accept(lc->fd, nullptr, nullptr);
   if (fd < 0) {
     *recvComm = nullptr;  // not ready; RCCL retries
@@ -330,7 +462,7 @@ ncclResult_t p_closeRecv(void* comm) {
 }
 ncclResult_t p_closeListen(void* comm) {
   auto* lc = static_cast<ListenComm*>(comm);
-  ::close(lc->fd);
+  if (lc->fd >= 0) ::close(lc->fd);
   delete lc;
   return ncclSuccess;
 }

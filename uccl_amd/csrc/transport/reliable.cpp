@@ -530,6 +530,15 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   impl_->progress = std::thread([this] { impl_->progress_loop(); });
 }
 
+void TransportEndpoint::close_flow(uint64_t flow) {
+  std::lock_guard<std::mutex> g(impl_->mu);
+  auto it = impl_->flows.find(flow);
+  if (it != impl_->flows.end()) {
+    it->second->failed = true;
+    impl_->cv.notify_all();
+  }
+}
+
 void TransportEndpoint::shutdown() {
   if (impl_->stop.exchange(true)) return;
   ::shutdown(impl_->ctrl_listen, SHUT_RDWR);

@@ -72,6 +72,10 @@ class TransportEndpoint {
   // threads are still blocked in recv_msg.
   void shutdown();
 
+  // Fail one flow: pending and future send/recv on it throw. (The analog
+  // of closing one connection while the endpoint stays up.)
+  void close_flow(uint64_t flow);
+
  private:
   struct Flow;
   struct Impl;
