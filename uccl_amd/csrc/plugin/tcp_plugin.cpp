@@ -85,9 +85,12 @@ bool use_multipath() {
 // distinguished by a nonce carried in the connect tag, so a global
 // acceptor thread can route inbound flows to the right listener queue.
 struct MpFabric {
-  uccl::transport::TransportEndpoint tp{
-      static_cast<int>(uccl::env_int("UCCL_NET_PATHS", 8)),
-      static_cast<size_t>(uccl::env_int("UCCL_NET_CHUNK", 16384))};
+  // process-lifetime singleton: the endpoint is intentionally leaked so
+  // static-destruction order can never race the detached acceptor thread
+  uccl::transport::TransportEndpoint& tp =
+      *new uccl::transport::TransportEndpoint(
+          static_cast<int>(uccl::env_int("UCCL_NET_PATHS", 8)),
+          static_cast<size_t>(uccl::env_int("UCCL_NET_CHUNK", 16384)));
   std::mutex mu;
   std::condition_variable cv;
   std::map<uint64_t, std::deque<uint64_t>> queues;  // nonce -> flows

@@ -583,7 +583,8 @@ uint64_t TransportEndpoint::accept(uint64_t* peer_tag) {
   impl_->cv.wait(lk, [this] {
     return !impl_->accepted.empty() || impl_->stop;
   });
-  UCCL_CHECK(!impl_->accepted.empty()) << "endpoint closed";
+  if (impl_->accepted.empty())
+    throw std::runtime_error("transport endpoint closed during accept");
   uint64_t f = impl_->accepted.front();
   impl_->accepted.pop_front();
   if (peer_tag) *peer_tag = impl_->accepted_tags.front();
