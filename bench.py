@@ -89,10 +89,11 @@ def main():
         if use_dist:
             dist.barrier()
 
-    # correctness guards outside the timed region: allreduce of ones must
-    # give world. The symmetric path is checked on its own tensor; if it
-    # ever fails we fall back to the staged path (and say so) rather than
-    # publishing a number from a broken path or no number at all.
+    # correctness guards outside the timed region, each exercising exactly
+    # the algorithm the timed loop would use (sizes are above the LL
+    # threshold so the gate matches the measured path):
+    #   staged two-shot (4 MiB > oneshot threshold) must pass, else no
+    #   number is published; the symmetric gate decides sym vs staged.
     expect = float(world)
 
     def check_ones(tensor):
@@ -102,13 +103,13 @@ def main():
         return bool(torch.allclose(tensor,
                                    torch.full_like(tensor, expect)))
 
-    chk = torch.ones(4096, dtype=torch.bfloat16, device="cuda")
+    chk = torch.ones(2 << 20, dtype=torch.bfloat16, device="cuda")
     if not check_ones(chk):
         print(json.dumps({"error": "allreduce correctness check failed",
                           "got": float(chk[0])}))
         sys.exit(2)
     if args.symmetric and world > 1:
-        schk = comm.symmetric_tensor([4096], torch.bfloat16)
+        schk = comm.symmetric_tensor([1 << 20], torch.bfloat16)
         if not check_ones(schk):
             args.symmetric = False
             t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
