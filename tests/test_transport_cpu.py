@@ -140,6 +140,17 @@ def test_transport_stats_cc():
     assert 2.0 <= st.cwnd <= 4096.0
 
 
+def test_transport_eqds_credit():
+    # receiver-driven credit mode: sender must stall when the granted
+    # window is exhausted and resume on ack grants; payload intact
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_CC="eqds",
+                                UCCL_TP_RWND_KB=64)
+    for n, sd in [(1 << 20, 51), (4 << 20, 52)]:
+        xfer(a, b, fa, fb, n, sd)
+    st = a.stats()
+    assert st.msgs_sent == 2
+
+
 def test_transport_swift_cc():
     C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_CC="swift")
     xfer(a, b, fa, fb, 2 << 20, 31)

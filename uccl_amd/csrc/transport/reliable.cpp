@@ -42,6 +42,8 @@ struct AckHdr {
   uint64_t sack0;  // bits for csn in [cum, cum+64)
   uint64_t sack1;  // [cum+64, cum+128)
   uint64_t ts_echo;
+  uint64_t credit_cum;  // receiver-driven grant (eqds mode): cumulative
+                        // bytes the sender may have chunked out
 };
 
 uint64_t now_ns() {
@@ -107,6 +109,9 @@ struct TransportEndpoint::Flow {
   uint32_t tx_cum = 0;       // lowest unacked csn
   uint32_t last_cum = 0;     // cum of the previous ack (hole detection)
   int hole_dupacks = 0;      // acks with stalled cum + new SACKs above
+  uint64_t bytes_chunked = 0;   // cumulative bytes handed to the wire
+  uint64_t credit_limit = 0;    // eqds: granted chunk-byte budget
+  uint64_t bytes_received = 0;  // rx side: cumulative fresh payload
 
   bool failed = false;  // flow marked dead after RTO abort threshold
 
@@ -173,8 +178,23 @@ struct TransportEndpoint::Impl {
     ++st.data_sent;
   }
 
+  static bool eqds_mode() {
+    static bool v = env_str("UCCL_TP_CC", "timely") == std::string("eqds");
+    return v;
+  }
+  static uint64_t rwnd_bytes() {
+    static uint64_t v =
+        static_cast<uint64_t>(env_int("UCCL_TP_RWND_KB", 4096)) << 10;
+    return v;
+  }
+
   void pump_tx(Flow& f) {
     while (static_cast<double>(f.inflight.size()) < f.cwnd && !f.txq.empty()) {
+      // EQDS-style receiver-driven credit: stop when the granted budget
+      // is exhausted; later acks raise credit_limit and re-pump.
+      if (eqds_mode() &&
+          f.bytes_chunked + chunk_bytes > f.credit_limit)
+        break;
       auto m = f.txq.front();
       uint32_t const csn = f.next_csn++;
       ChunkTx c;
@@ -183,6 +203,7 @@ struct TransportEndpoint::Impl {
       c.len = static_cast<uint32_t>(
           std::min(chunk_bytes, m->bytes - m->next_off));
       m->next_off += c.len;
+      f.bytes_chunked += c.len;
       if (m->next_off >= m->bytes) f.txq.pop_front();
       send_chunk(f, csn, c);
       f.inflight.emplace(csn, std::move(c));
@@ -193,6 +214,8 @@ struct TransportEndpoint::Impl {
   }
 
   void send_ack(Flow& f, int sock_idx, sockaddr_in const& to) {
+    // grant: allow the sender to stay rwnd bytes ahead of what we've seen
+    uint64_t const grant = f.bytes_received + rwnd_bytes();
     // independent ACK-loss injection (exercises cumulative-ack coverage
     // and the RTO backstop on reverse-path drops)
     // key the decision on data_recv (which always advances) — keying on
@@ -202,7 +225,7 @@ struct TransportEndpoint::Impl {
       ++st.injected_drops;
       return;
     }
-    AckHdr a{kMagic, kAck, f.id, f.rx_cum, 0, 0, 0, f.last_data_ts};
+    AckHdr a{kMagic, kAck, f.id, f.rx_cum, 0, 0, 0, f.last_data_ts, grant};
     for (auto const& [csn, _] : f.rx_ooo) {
       uint32_t const d = csn - f.rx_cum;
       if (d < 64)
@@ -277,6 +300,7 @@ struct TransportEndpoint::Impl {
 
   void handle_ack(Flow& f, AckHdr const& a) {
     ++st.acks_recv;
+    if (a.credit_cum > f.credit_limit) f.credit_limit = a.credit_cum;
     if (a.ts_echo) {
       double const rtt_us = (now_ns() - a.ts_echo) / 1000.0;
       timely_update(f, rtt_us);
@@ -333,6 +357,7 @@ struct TransportEndpoint::Impl {
       }
       if (h.len) memcpy(m.dest() + h.off, payload, h.len);
       m.recv_bytes += h.len;
+      f.bytes_received += h.len;
       f.rx_ooo[h.csn] = true;
       while (f.rx_ooo.count(f.rx_cum)) {
         f.rx_ooo.erase(f.rx_cum);
@@ -465,6 +490,7 @@ struct TransportEndpoint::Impl {
   void install_flow(uint64_t flow, CtrlMsg const& peer) {
     auto f = std::make_unique<Flow>();
     f->id = flow;
+    f->credit_limit = rwnd_bytes();  // initial grant (pre-first-ack)
     f->num_paths = std::min(num_paths, peer.num_paths);
     for (int i = 0; i < f->num_paths; ++i) {
       sockaddr_in a{};
