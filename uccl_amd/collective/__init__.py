@@ -81,6 +81,11 @@ class Communicator:
     def is_symmetric(self, t: torch.Tensor) -> bool:
         return self._c.is_symmetric(t)
 
+    def stats(self) -> dict:
+        """Per-op host-side call/byte tallies (observability parity with
+        the reference's stats counters)."""
+        return self._c.stats()
+
 
 def init(group=None, device: Optional[int] = None,
          heap_bytes: int = 0) -> Communicator:
