@@ -44,6 +44,17 @@ def test_p2p_tcp_cpu():
     _run_pair({"UCCL_P2P_TEST_GPU": "0"})
 
 
+def test_p2p_multipath_cpu():
+    # cross-host-style data plane: p2p bytes ride the reliable multipath
+    # transport (reference architecture: p2p engine over its own engine)
+    _run_pair({"UCCL_P2P_TEST_GPU": "0", "UCCL_P2P_TRANSPORT": "multipath"})
+
+
+def test_p2p_multipath_lossy_cpu():
+    _run_pair({"UCCL_P2P_TEST_GPU": "0", "UCCL_P2P_TRANSPORT": "multipath",
+               "UCCL_TP_LOSS_PCT": "5"})
+
+
 @pytest.mark.gpu
 def test_p2p_gpu_ipc():
     if not torch.cuda.is_available():

@@ -64,7 +64,9 @@ def build_plugin(verbose: bool = False) -> Path:
         CSRC / "transport" / "reliable.cpp"]
     capi = plugdir / "libuccl_p2p.so"
     capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
-                 CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h"]
+                 CSRC / "transport" / "reliable.cpp",
+                 CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h",
+                 CSRC / "transport" / "reliable.h"]
     shim_check = plugdir / "libuccl_nccl.so"
     if (not _stale(target, deps) and not _stale(harness, deps)
             and not _stale(capi, capi_srcs + deps)

@@ -3,11 +3,13 @@
 #include <hip/hip_runtime.h>
 #include <unistd.h>
 
+#include <chrono>
 #include <cstring>
 
 #include "../core/env.h"
 #include "../core/log.h"
 #include "../core/net.h"
+#include "../transport/reliable.h"
 
 namespace uccl {
 namespace p2p {
@@ -47,9 +49,19 @@ struct Meta {
   int gpu;
   int pid;
   char host[64];
+  uint16_t tp_len;   // multipath rendezvous metadata (0 = tcp-only peer)
+  char tp_md[64];
 };
 
 bool is_gpu(int device) { return device >= 0; }
+
+bool p2p_multipath() {
+  static bool v = uccl::env_str("UCCL_P2P_TRANSPORT", "tcp") ==
+                  std::string("multipath");
+  return v;
+}
+
+constexpr size_t kMsgChunk = 8ull << 20;  // payload message granule
 
 }  // namespace
 
@@ -65,6 +77,8 @@ struct Endpoint::RxItem {
 struct Endpoint::Conn {
   uint64_t id = 0;
   int fd = -1;
+  uccl::transport::TransportEndpoint* tp = nullptr;  // multipath mode
+  uint64_t flow = 0;
   std::string peer_ip;
   int peer_gpu = -1;
   int peer_pid = -1;
@@ -88,11 +102,48 @@ struct Endpoint::Conn {
   std::unordered_map<std::string, void*> ipc_cache;
   std::mutex ipc_mu;
 
+  // one logical message = header + chunked payload; over the multipath
+  // plane each piece is a discrete reliable message in the same order
   void send_msg(MsgHdr const& h, void const* payload = nullptr,
                 size_t payload_bytes = 0) {
     std::lock_guard<std::mutex> g(tx_mu);
-    net::send_all(fd, &h, sizeof(h));
-    if (payload_bytes) net::send_all(fd, payload, payload_bytes);
+    if (flow) {
+      tp->send_msg(flow, &h, sizeof(h));
+      for (size_t off = 0; off < payload_bytes; off += kMsgChunk) {
+        size_t const n = std::min(kMsgChunk, payload_bytes - off);
+        tp->send_msg(flow, static_cast<char const*>(payload) + off, n);
+      }
+    } else {
+      net::send_all(fd, &h, sizeof(h));
+      if (payload_bytes) net::send_all(fd, payload, payload_bytes);
+    }
+  }
+
+  bool recv_hdr(MsgHdr* h) {
+    if (flow) {
+      try {
+        tp->recv_msg(flow, h, sizeof(*h));
+        return true;
+      } catch (std::exception const&) {
+        return false;
+      }
+    }
+    return net::recv_all(fd, h, sizeof(*h));
+  }
+
+  bool recv_payload(void* buf, size_t bytes) {
+    if (flow) {
+      try {
+        for (size_t off = 0; off < bytes; off += kMsgChunk) {
+          size_t const n = std::min(kMsgChunk, bytes - off);
+          tp->recv_msg(flow, static_cast<char*>(buf) + off, n);
+        }
+        return true;
+      } catch (std::exception const&) {
+        return false;
+      }
+    }
+    return net::recv_all(fd, buf, bytes);
   }
 };
 
@@ -129,6 +180,20 @@ Endpoint::Endpoint(int gpu, int num_workers) : gpu_(gpu) {
                static_cast<uint64_t>(getpid()), 0};
       net::send_all(fd, &r, sizeof(r));
       net::send_all(fd, host_id_.data(), host_id_.size());
+      if (tp_ && h.d) {
+        // connector opens a flow tagged with its hello nonce
+        std::unique_lock<std::mutex> lk(tp_mu_);
+        bool ok = tp_cv_.wait_for(lk, std::chrono::seconds(30), [&] {
+          return tp_flows_.count(h.d) || stop_.load();
+        });
+        if (!ok || stop_) {
+          ::close(fd);
+          continue;
+        }
+        c->tp = tp_.get();
+        c->flow = tp_flows_[h.d];
+        tp_flows_.erase(h.d);
+      }
       {
         std::lock_guard<std::mutex> g(conn_mu_);
         conns_[c->id] = c;
@@ -138,6 +203,25 @@ Endpoint::Endpoint(int gpu, int num_workers) : gpu_(gpu) {
       accept_cv_.notify_all();
     }
   });
+  if (p2p_multipath()) {
+    tp_ = std::make_unique<uccl::transport::TransportEndpoint>(
+        static_cast<int>(env_int("UCCL_P2P_PATHS", 8)),
+        static_cast<size_t>(env_int("UCCL_P2P_TP_CHUNK", 16384)));
+    tp_acceptor_ = std::thread([this] {
+      try {
+        while (!stop_) {
+          uint64_t tag = 0;
+          uint64_t flow = tp_->accept(&tag);
+          {
+            std::lock_guard<std::mutex> g(tp_mu_);
+            tp_flows_[tag] = flow;
+          }
+          tp_cv_.notify_all();
+        }
+      } catch (std::exception const&) {
+      }
+    });
+  }
   for (int i = 0; i < std::max(1, num_workers); ++i)
     workers_.emplace_back([this] { worker_loop(); });
 }
@@ -154,6 +238,7 @@ hipStream_t Endpoint::copy_stream() {
 
 Endpoint::~Endpoint() {
   stop_ = true;
+  if (tp_) tp_->shutdown();
   ::shutdown(listen_fd_, SHUT_RDWR);
   ::close(listen_fd_);
   task_cv_.notify_all();
@@ -164,9 +249,11 @@ Endpoint::~Endpoint() {
     for (auto& [id, c] : conns_) {
       c->alive = false;
       ::shutdown(c->fd, SHUT_RDWR);
+      if (c->flow && tp_) tp_->close_flow(c->flow);
     }
   }
   if (listener_.joinable()) listener_.join();
+  if (tp_acceptor_.joinable()) tp_acceptor_.join();
   {
     std::lock_guard<std::mutex> g(conn_mu_);
     for (auto& [id, c] : conns_) {
@@ -189,6 +276,11 @@ std::string Endpoint::metadata() const {
   m.gpu = gpu_;
   m.pid = static_cast<int>(getpid());
   strncpy(m.host, host_id_.c_str(), sizeof(m.host) - 1);
+  if (tp_) {
+    std::string md = tp_->metadata();
+    m.tp_len = static_cast<uint16_t>(md.size());
+    memcpy(m.tp_md, md.data(), std::min(md.size(), sizeof(m.tp_md)));
+  }
   return std::string(reinterpret_cast<char*>(&m), sizeof(m));
 }
 
@@ -203,8 +295,11 @@ uint64_t Endpoint::connect(const std::string& remote_metadata) {
   c->fd = fd;
   c->id = next_conn_++;
   c->peer_ip = ip;
+  // nonce identifies this connection's multipath flow on the peer side
+  uint64_t const nonce =
+      tp_ ? ((static_cast<uint64_t>(getpid()) << 32) | c->id) : 0;
   MsgHdr h{kHello, static_cast<uint64_t>(gpu_), host_id_.size(),
-           static_cast<uint64_t>(getpid()), 0};
+           static_cast<uint64_t>(getpid()), nonce};
   net::send_all(fd, &h, sizeof(h));
   net::send_all(fd, host_id_.data(), host_id_.size());
   MsgHdr r{};
@@ -215,6 +310,10 @@ uint64_t Endpoint::connect(const std::string& remote_metadata) {
   c->peer_gpu = static_cast<int>(r.a);
   c->peer_pid = static_cast<int>(r.c);
   c->same_host = (std::string(hostbuf.begin(), hostbuf.end()) == host_id_);
+  if (tp_ && m.tp_len) {
+    c->tp = tp_.get();
+    c->flow = tp_->connect(std::string(m.tp_md, m.tp_len), nonce);
+  }
   {
     std::lock_guard<std::mutex> g(conn_mu_);
     conns_[c->id] = c;
@@ -259,13 +358,13 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
   if (gpu_ >= 0) (void)hipSetDevice(gpu_);
   while (c->alive && !stop_) {
     MsgHdr h{};
-    if (!net::recv_all(c->fd, &h, sizeof(h))) break;
+    if (!c->recv_hdr(&h)) break;
     switch (h.op) {
       case kSendData: {
         auto item = std::make_shared<RxItem>();
         item->bytes = h.a;
         item->data.resize(h.a);
-        net::recv_all(c->fd, item->data.data(), h.a);
+        c->recv_payload(item->data.data(), h.a);
         {
           std::lock_guard<std::mutex> g(c->rx_mu);
           c->rxq.push_back(item);
@@ -278,7 +377,7 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
         item->bytes = h.a;
         item->ipc = true;
         item->token = h.d;
-        net::recv_all(c->fd, &item->blob, sizeof(IpcBlob));
+        c->recv_payload(&item->blob, sizeof(IpcBlob));
         {
           std::lock_guard<std::mutex> g(c->rx_mu);
           c->rxq.push_back(item);
@@ -297,10 +396,10 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
         UCCL_CHECK(h.b + h.c <= mr.bytes) << "write overflows mr";
         char* dst = static_cast<char*>(mr.ptr) + h.b;
         if (!is_gpu(mr.device)) {
-          net::recv_all(c->fd, dst, h.c);
+          c->recv_payload(dst, h.c);
         } else {
           std::vector<char> tmp(h.c);
-          net::recv_all(c->fd, tmp.data(), h.c);
+          c->recv_payload(tmp.data(), h.c);
           hipStream_t cs = copy_stream();
           UCCL_CHECK_HIP(hipMemcpyAsync(dst, tmp.data(), h.c,
                                         hipMemcpyHostToDevice, cs));
@@ -311,7 +410,7 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
       }
       case kWriteIpc: {
         IpcBlob blob{};
-        net::recv_all(c->fd, &blob, sizeof(blob));
+        c->recv_payload(&blob, sizeof(blob));
         MR mr;
         {
           std::lock_guard<std::mutex> g(mr_mu_);
@@ -357,7 +456,7 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
         auto item = std::make_shared<RxItem>();
         item->bytes = h.a;
         item->data.resize(h.a);
-        net::recv_all(c->fd, item->data.data(), h.a);
+        c->recv_payload(item->data.data(), h.a);
         {
           std::lock_guard<std::mutex> g(c->tok_mu);
           c->completed[h.d] = item;
@@ -431,16 +530,21 @@ void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
     c.send_msg(MsgHdr{kSendData, bytes, 0, 0, 0}, ptr, bytes);
     return;
   }
-  // GPU over TCP: pinned staging chunks
+  // GPU over the wire: pinned staging chunks (kMsgChunk-sized so the
+  // multipath plane's message framing mirrors the receiver's reads)
   std::lock_guard<std::mutex> sg(staging_mu_);
   if (!staging_) {
     UCCL_CHECK_HIP(hipSetDevice(gpu_ >= 0 ? gpu_ : device));
-    UCCL_CHECK_HIP(hipHostMalloc(&staging_, kStagingBytes));
-    staging_bytes_ = kStagingBytes;
+    UCCL_CHECK_HIP(hipHostMalloc(&staging_, kMsgChunk));
+    staging_bytes_ = kMsgChunk;
   }
   std::lock_guard<std::mutex> g(c.tx_mu);
   MsgHdr h{kSendData, bytes, 0, 0, 0};
-  net::send_all(c.fd, &h, sizeof(h));
+  if (c.flow) {
+    c.tp->send_msg(c.flow, &h, sizeof(h));
+  } else {
+    net::send_all(c.fd, &h, sizeof(h));
+  }
   hipStream_t cs = copy_stream();
   for (size_t off = 0; off < bytes; off += staging_bytes_) {
     size_t n = std::min(staging_bytes_, bytes - off);
@@ -448,7 +552,11 @@ void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
                                   static_cast<char const*>(ptr) + off, n,
                                   hipMemcpyDeviceToHost, cs));
     UCCL_CHECK_HIP(hipStreamSynchronize(cs));
-    net::send_all(c.fd, staging_, n);
+    if (c.flow) {
+      c.tp->send_msg(c.flow, staging_, n);
+    } else {
+      net::send_all(c.fd, staging_, n);
+    }
   }
 }
 
@@ -656,6 +764,7 @@ void Endpoint::close_conn(uint64_t conn_id) {
   }
   c->alive = false;
   ::shutdown(c->fd, SHUT_RDWR);
+  if (c->flow && tp_) tp_->close_flow(c->flow);
   c->rx_cv.notify_all();
   c->tok_cv.notify_all();
   if (c->rx.joinable()) c->rx.join();

@@ -17,6 +17,7 @@
 #include <hip/hip_runtime.h>
 
 #include <atomic>
+#include <map>
 #include <condition_variable>
 #include <cstdint>
 #include <deque>
@@ -29,6 +30,9 @@
 #include <vector>
 
 namespace uccl {
+namespace transport {
+class TransportEndpoint;
+}
 namespace p2p {
 
 struct MR {
@@ -139,6 +143,15 @@ class Endpoint {
   hipStream_t copy_stream();
   void* copy_stream_ = nullptr;
   std::mutex copy_mu_;
+
+  // optional multipath data plane (UCCL_P2P_TRANSPORT=multipath): conns
+  // carry their bytes as reliable-transport messages instead of raw TCP
+  // (the TCP socket stays as the control/handshake channel)
+  std::unique_ptr<transport::TransportEndpoint> tp_;
+  std::thread tp_acceptor_;
+  std::mutex tp_mu_;
+  std::condition_variable tp_cv_;
+  std::map<uint64_t, uint64_t> tp_flows_;  // hello nonce -> flow
 };
 
 }  // namespace p2p
