@@ -35,6 +35,8 @@ class _Handle:
 
 
 def _event():
+    if not torch.cuda.is_available():
+        return None
     ev = torch.cuda.Event()
     ev.record()
     return ev
