@@ -4,6 +4,7 @@
 #include <unistd.h>
 
 #include <cstring>
+#include <sstream>
 
 #include "../core/env.h"
 #include "../core/log.h"
@@ -48,9 +49,28 @@ Communicator::Communicator(int rank, int world, int device, size_t heap_bytes)
   UCCL_LOG_INFO << "Communicator rank " << rank_ << "/" << world_
                 << " heap=" << (heap_bytes_ >> 20) << "MB scratch_cap="
                 << (scratch_cap_ >> 20) << "MB";
+  if (env_bool("UCCL_ENGINE_STATS", false)) {
+    stats_thread_ = std::thread([this] {
+      static const char* names[8] = {"ar", "ag", "rs", "bc",
+                                     "a2a", "snd", "rcv", "bar"};
+      while (!stop_stats_) {
+        for (int i = 0; i < 20 && !stop_stats_; ++i)
+          usleep(100 * 1000);
+        std::ostringstream os;
+        for (int i = 0; i < 8; ++i)
+          if (stats_[i].calls)
+            os << names[i] << "=" << stats_[i].calls << "/"
+               << (stats_[i].bytes >> 20) << "MB ";
+        if (!os.str().empty())
+          UCCL_LOG_INFO << "[stats rank " << rank_ << "] " << os.str();
+      }
+    });
+  }
 }
 
 Communicator::~Communicator() {
+  stop_stats_ = true;
+  if (stats_thread_.joinable()) stats_thread_.join();
   for (int r = 0; r < world_; ++r) {
     if (ipc_opened_[r] && peers_[r]) (void)hipIpcCloseMemHandle(peers_[r]);
   }

@@ -16,8 +16,10 @@
 #include <hip/hip_runtime.h>
 
 #include <array>
+#include <atomic>
 #include <cstdint>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "kernels.h"
@@ -108,6 +110,11 @@ class Communicator {
   std::array<uint64_t, kMaxRanks> send_seq_{};
   std::array<uint64_t, kMaxRanks> recv_seq_{};
   bool connected_ = false;
+
+  // optional periodic stats dump (UCCL_ENGINE_STATS=1), the analog of the
+  // reference's per-engine stats thread (transport.cc:1797)
+  std::thread stats_thread_;
+  std::atomic<bool> stop_stats_{false};
 
   std::array<OpStats, 8> stats_{};
   void tally(int op, size_t bytes) {
