@@ -8,6 +8,7 @@
 #include <cassert>
 #include <cstdio>
 #include <cstring>
+#include <thread>
 #include <vector>
 
 #include "net_plugin_abi.h"
@@ -107,6 +108,51 @@ int main(int argc, char** argv) {
   assert(net->closeSend(sc) == ncclSuccess);
   assert(net->closeRecv(rc) == ncclSuccess);
   assert(net->closeListen(lc) == ncclSuccess);
+
+  // --- concurrent multi-pair: 3 listeners, interleaved traffic ------------
+  // (exercises listener demultiplexing — on the multipath plane, the
+  // nonce-routed flow queues of the shared fabric endpoint)
+  {
+    std::vector<std::thread> ths;
+    for (int pair = 0; pair < 3; ++pair) {
+      ths.emplace_back([net, pair] {
+        char hdl[NCCL_NET_HANDLE_MAXSIZE] = {};
+        void *plc = nullptr, *psc = nullptr, *prc = nullptr;
+        assert(net->listen(0, hdl, &plc) == ncclSuccess);
+        std::thread acc([&] {
+          while (!prc) assert(net->accept(plc, &prc) == ncclSuccess);
+        });
+        while (!psc) assert(net->connect(0, hdl, &psc) == ncclSuccess);
+        acc.join();
+        for (int it = 0; it < 5; ++it) {
+          std::vector<char> sbuf(100000 + 1000 * pair,
+                                 static_cast<char>(pair * 31 + it));
+          std::vector<char> rbuf(sbuf.size());
+          void *sr = nullptr, *rr = nullptr;
+          int const tag = pair * 100 + it;
+          assert(net->isend(psc, sbuf.data(),
+                            static_cast<int>(sbuf.size()), tag, nullptr,
+                            &sr) == ncclSuccess);
+          void* d[1] = {rbuf.data()};
+          int z[1] = {static_cast<int>(rbuf.size())};
+          int t[1] = {tag};
+          void* mh[1] = {nullptr};
+          assert(net->irecv(prc, 1, d, z, t, mh, &rr) == ncclSuccess);
+          wait_req(net, sr, nullptr);
+          int got = 0;
+          wait_req(net, rr, &got);
+          assert(got == static_cast<int>(sbuf.size()));
+          assert(memcmp(sbuf.data(), rbuf.data(), sbuf.size()) == 0);
+        }
+        assert(net->closeSend(psc) == ncclSuccess);
+        assert(net->closeRecv(prc) == ncclSuccess);
+        assert(net->closeListen(plc) == ncclSuccess);
+      });
+    }
+    for (auto& t : ths) t.join();
+    printf("multi-pair OK\n");
+  }
+
   printf("PLUGIN HARNESS OK\n");
   return 0;
 }
