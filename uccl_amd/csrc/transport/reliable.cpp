@@ -500,7 +500,12 @@ struct TransportEndpoint::Impl {
       f->peer_paths[i] = a;
     }
     std::lock_guard<std::mutex> g(mu);
-    flows[flow] = std::move(f);
+    // self-connections (one endpoint dialing its own metadata, e.g. the
+    // in-process plugin fabric) install the same id from both the ctrl
+    // acceptor and the connector: the second install must NOT replace the
+    // live Flow (threads may already hold references into it) — both
+    // directions share the one object.
+    if (!flows.count(flow)) flows[flow] = std::move(f);
   }
 };
 
@@ -658,7 +663,10 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
     m.user_ptr = static_cast<char*>(ptr);
   }
   std::unique_lock<std::mutex> lk(impl_->mu);
-  Flow& f = *impl_->flows[flow];
+  auto fit = impl_->flows.find(flow);
+  if (fit == impl_->flows.end() || !fit->second)
+    throw std::runtime_error("transport flow vanished");
+  Flow& f = *fit->second;
   impl_->cv.wait(lk, [&] {
     auto it = f.rxmsgs.find(msg_id);
     return (it != f.rxmsgs.end() && it->second.known &&
