@@ -22,7 +22,7 @@ def test_plugin_harness(plane):
     env = dict(os.environ)
     env["UCCL_NET_TRANSPORT"] = plane
     r = subprocess.run([str(harness), str(so)], capture_output=True,
-                       timeout=120, env=env)
+                       timeout=300, env=env)
     out = r.stdout.decode() + r.stderr.decode()
     assert r.returncode == 0, out
     assert "PLUGIN HARNESS OK" in out
