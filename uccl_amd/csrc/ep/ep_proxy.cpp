@@ -170,7 +170,6 @@ void EpProxy::comb_tx_loop() {
         for (int le = 0; le < v_.local_experts; ++le) {
           uint64_t const count =
               static_cast<uint64_t>(task.counts[le * v_.world + src]);
-          (void)row_bytes;
           WireHdr h{kComb, static_cast<uint32_t>(task.seq),
                     static_cast<uint32_t>(le),
                     static_cast<uint32_t>(v_.rank), count};
