@@ -208,3 +208,15 @@ def test_transport_star_multi_flow():
     assert sorted(g.long().sum().item() for g in got_hub) ==         sorted(p.long().sum().item() for p in to_hub)
     for g in got_spoke:
         assert any(torch.equal(g, fh) for fh in from_hub)
+
+
+def test_transport_pacing():
+    """UCCL_TP_PACE_MBPS throttles chunk emission: a 1MB message at a
+    10 MB/s pace must take >= ~80ms (unpaced loopback: ~1ms)."""
+    import time
+
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_PACE_MBPS=10)
+    t0 = time.perf_counter()
+    xfer(a, b, fa, fb, 1 << 20, 61)
+    dt = time.perf_counter() - t0
+    assert dt >= 0.08, f"pacing had no effect ({dt*1e3:.1f} ms)"

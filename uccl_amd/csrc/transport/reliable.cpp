@@ -112,6 +112,7 @@ struct TransportEndpoint::Flow {
   uint64_t bytes_chunked = 0;   // cumulative bytes handed to the wire
   uint64_t credit_limit = 0;    // eqds: granted chunk-byte budget
   uint64_t bytes_received = 0;  // rx side: cumulative fresh payload
+  uint64_t next_send_ns = 0;    // pacing release time (UCCL_TP_PACE_MBPS)
 
   bool failed = false;  // flow marked dead after RTO abort threshold
 
@@ -129,6 +130,7 @@ struct TransportEndpoint::Impl {
   size_t chunk_bytes;
   int loss_pct;
   int ack_loss_pct;
+  uint64_t pace_q32 = 0;  // ns-per-byte in Q32 (0 = pacing bypassed)
   std::vector<int> socks;          // UDP path sockets
   std::vector<uint16_t> ports;
   int wake_fd = -1;                // self-addressed UDP for wakeups
@@ -195,6 +197,10 @@ struct TransportEndpoint::Impl {
       if (eqds_mode() &&
           f.bytes_chunked + chunk_bytes > f.credit_limit)
         break;
+      if (pace_q32) {
+        uint64_t const now = now_ns();
+        if (f.next_send_ns > now) break;  // paced: retry on a later pump
+      }
       auto m = f.txq.front();
       uint32_t const csn = f.next_csn++;
       ChunkTx c;
@@ -204,6 +210,11 @@ struct TransportEndpoint::Impl {
           std::min(chunk_bytes, m->bytes - m->next_off));
       m->next_off += c.len;
       f.bytes_chunked += c.len;
+      if (pace_q32) {
+        uint64_t const now = now_ns();
+        uint64_t const base = std::max(f.next_send_ns, now);
+        f.next_send_ns = base + ((pace_q32 * c.len) >> 32);
+      }
       if (m->next_off >= m->bytes) f.txq.pop_front();
       send_chunk(f, csn, c);
       f.inflight.emplace(csn, std::move(c));
@@ -519,6 +530,11 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   impl_->loss_pct = static_cast<int>(env_int("UCCL_TP_LOSS_PCT", 0));
   impl_->ack_loss_pct =
       static_cast<int>(env_int("UCCL_TP_ACK_LOSS_PCT", 0));
+  // optional sender pacing (the reference's Carousel timing-wheel role;
+  // bypassed by default there and here — BYPASS_PACING=1)
+  if (int64_t mbps = env_int("UCCL_TP_PACE_MBPS", 0); mbps > 0)
+    impl_->pace_q32 =
+        static_cast<uint64_t>((1e9 * 4294967296.0) / (mbps * 1e6));
   for (int i = 0; i < num_paths; ++i) {
     int s = ::socket(AF_INET, SOCK_DGRAM, 0);
     UCCL_CHECK(s >= 0) << "udp socket";
