@@ -42,8 +42,9 @@ def main():
         win = torch.zeros(4096, dtype=torch.float32, device=dev)
         mr = ep.reg(win)
         ad = ep.advertise(mr, 0, win.numel() * 4)
-        with open(meta_path + ".ad", "wb") as f:
+        with open(meta_path + ".ad.tmp", "wb") as f:
             f.write(ad)
+        os.rename(meta_path + ".ad.tmp", meta_path + ".ad")
         # wait for client's done marker
         while not os.path.exists(meta_path + ".done"):
             time.sleep(0.05)
@@ -82,8 +83,9 @@ def main():
         back = torch.zeros(4096, dtype=torch.float32, device=dev)
         ep.read(cid, back, ad)
         assert torch.allclose(back.cpu(), src.cpu())
-        with open(meta_path + ".done", "w") as f:
+        with open(meta_path + ".done.tmp", "w") as f:
             f.write("x")
+        os.rename(meta_path + ".done.tmp", meta_path + ".done")
 
         # async large transfer
         g = torch.Generator().manual_seed(99)
