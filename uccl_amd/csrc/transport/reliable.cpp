@@ -131,6 +131,15 @@ struct TransportEndpoint::Impl {
   int loss_pct;
   int ack_loss_pct;
   uint64_t pace_q32 = 0;  // ns-per-byte in Q32 (0 = pacing bypassed)
+  // CC configuration, resolved per endpoint at construction (statics
+  // would freeze the first process-wide value and break test isolation)
+  std::string cc_mode = "timely";
+  double t_low_us = 50, t_high_us = 1000, swift_target_us = 300;
+  double cwnd_max = 1024;
+  uint64_t rwnd = 4u << 20;
+  int dup_thres = 32;
+  uint64_t rto_base_ns = 20000000;
+  int rto_abort_thres = 50;
   std::vector<int> socks;          // UDP path sockets
   std::vector<uint16_t> ports;
   int wake_fd = -1;                // self-addressed UDP for wakeups
@@ -180,15 +189,8 @@ struct TransportEndpoint::Impl {
     ++st.data_sent;
   }
 
-  static bool eqds_mode() {
-    static bool v = env_str("UCCL_TP_CC", "timely") == std::string("eqds");
-    return v;
-  }
-  static uint64_t rwnd_bytes() {
-    static uint64_t v =
-        static_cast<uint64_t>(env_int("UCCL_TP_RWND_KB", 4096)) << 10;
-    return v;
-  }
+  bool eqds_mode() const { return cc_mode == "eqds"; }
+  uint64_t rwnd_bytes() const { return rwnd; }
 
   void pump_tx(Flow& f) {
     while (static_cast<double>(f.inflight.size()) < f.cwnd && !f.txq.empty()) {
@@ -257,9 +259,9 @@ struct TransportEndpoint::Impl {
   //   UCCL_TP_CC=swift   delay-target window (SIGCOMM'20)
   //   UCCL_TP_CC=none    fixed window (UCCL_TP_CWND_MAX)
   void timely_update(Flow& f, double rtt_us) {
-    static std::string const cc = env_str("UCCL_TP_CC", "timely");
-    static double const t_low = env_int("UCCL_TP_TLOW_US", 50);
-    static double const t_high = env_int("UCCL_TP_THIGH_US", 1000);
+    std::string const& cc = cc_mode;
+    double const t_low = t_low_us;
+    double const t_high = t_high_us;
     static double const add = 1.0, beta = 0.8;
     if (f.srtt_us == 0) f.srtt_us = rtt_us;
     double const grad = (rtt_us - f.prev_rtt_us) / std::max(f.srtt_us, 1.0);
@@ -270,7 +272,7 @@ struct TransportEndpoint::Impl {
     } else if (cc == "swift") {
       // Swift: additive increase below the delay target, multiplicative
       // decrease proportional to the overshoot (capped)
-      static double const target = env_int("UCCL_TP_SWIFT_TARGET_US", 300);
+      double const target = swift_target_us;
       static double const max_mdf = 0.5;
       if (rtt_us < target) {
         f.cwnd += add / std::max(f.cwnd, 1.0) * 8.0;
@@ -288,8 +290,6 @@ struct TransportEndpoint::Impl {
     } else {
       f.cwnd *= 1.0 - beta * std::min(grad, 0.25);
     }
-    static double const cwnd_max =
-        static_cast<double>(env_int("UCCL_TP_CWND_MAX", 1024));
     f.cwnd = std::min(std::max(f.cwnd, 2.0), cwnd_max);
     st.srtt_us = f.srtt_us;
     st.cwnd = f.cwnd;
@@ -335,8 +335,6 @@ struct TransportEndpoint::Impl {
     // signal — and retransmit only the first hole. The threshold stays
     // large for the same reason the reference uses ROCE_DUP_ACK_THRES=32
     // (collective/rdma/transport_config.h:145); RTO backstops the rest.
-    static int const dup_thres =
-        static_cast<int>(env_int("UCCL_TP_DUPACK_THRES", 32));
     if (a.cum != f.last_cum) {
       f.last_cum = a.cum;
       f.hole_dupacks = 0;
@@ -383,14 +381,12 @@ struct TransportEndpoint::Impl {
   }
 
   void rto_scan() {
-    static uint64_t const rto_ns =
-        static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 20000)) * 1000;
+    uint64_t const rto_ns = rto_base_ns;
     // flow-failure detection, parity with the reference's RTO abort
     // threshold (kRTOAbortThreshold=50, transport_config.h:202 +
     // mark_flow_timeout): a chunk that hits RTO this many times in a row
     // marks the flow failed and fails its blocked senders/receivers.
-    static int const abort_thres =
-        static_cast<int>(env_int("UCCL_TP_RTO_ABORT", 50));
+    int const abort_thres = rto_abort_thres;
     uint64_t const now = now_ns();
     for (auto& [fid, fp] : flows) {
       Flow& f = *fp;
@@ -530,6 +526,21 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   impl_->loss_pct = static_cast<int>(env_int("UCCL_TP_LOSS_PCT", 0));
   impl_->ack_loss_pct =
       static_cast<int>(env_int("UCCL_TP_ACK_LOSS_PCT", 0));
+  impl_->cc_mode = env_str("UCCL_TP_CC", "timely");
+  impl_->t_low_us = static_cast<double>(env_int("UCCL_TP_TLOW_US", 50));
+  impl_->t_high_us = static_cast<double>(env_int("UCCL_TP_THIGH_US", 1000));
+  impl_->swift_target_us =
+      static_cast<double>(env_int("UCCL_TP_SWIFT_TARGET_US", 300));
+  impl_->cwnd_max =
+      static_cast<double>(env_int("UCCL_TP_CWND_MAX", 1024));
+  impl_->rwnd = static_cast<uint64_t>(env_int("UCCL_TP_RWND_KB", 4096))
+                << 10;
+  impl_->dup_thres =
+      static_cast<int>(env_int("UCCL_TP_DUPACK_THRES", 32));
+  impl_->rto_base_ns =
+      static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 20000)) * 1000;
+  impl_->rto_abort_thres =
+      static_cast<int>(env_int("UCCL_TP_RTO_ABORT", 50));
   // optional sender pacing (the reference's Carousel timing-wheel role;
   // bypassed by default there and here — BYPASS_PACING=1)
   if (int64_t mbps = env_int("UCCL_TP_PACE_MBPS", 0); mbps > 0)
