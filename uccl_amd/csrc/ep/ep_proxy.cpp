@@ -45,7 +45,10 @@ EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
 }
 
 EpProxy::~EpProxy() {
-  stop_ = true;
+  {
+    std::lock_guard<std::mutex> g(mu_);  // lost-wakeup guard
+    stop_ = true;
+  }
   cv_.notify_all();
   if (tp_) tp_->shutdown();
   if (ring_thread_.joinable()) ring_thread_.join();

@@ -237,7 +237,10 @@ hipStream_t Endpoint::copy_stream() {
 }
 
 Endpoint::~Endpoint() {
-  stop_ = true;
+  {
+    std::lock_guard<std::mutex> g(task_mu_);  // lost-wakeup guard
+    stop_ = true;
+  }
   if (tp_) tp_->shutdown();
   ::shutdown(listen_fd_, SHUT_RDWR);
   ::close(listen_fd_);
@@ -247,9 +250,15 @@ Endpoint::~Endpoint() {
   {
     std::lock_guard<std::mutex> g(conn_mu_);
     for (auto& [id, c] : conns_) {
-      c->alive = false;
+      {
+        std::lock_guard<std::mutex> g1(c->rx_mu);
+        std::lock_guard<std::mutex> g2(c->tok_mu);
+        c->alive = false;
+      }
       ::shutdown(c->fd, SHUT_RDWR);
       if (c->flow && tp_) tp_->close_flow(c->flow);
+      c->rx_cv.notify_all();
+      c->tok_cv.notify_all();
     }
   }
   if (listener_.joinable()) listener_.join();
@@ -762,7 +771,11 @@ void Endpoint::close_conn(uint64_t conn_id) {
     c = it->second;
     conns_.erase(it);
   }
-  c->alive = false;
+  {
+    std::lock_guard<std::mutex> g1(c->rx_mu);
+    std::lock_guard<std::mutex> g2(c->tok_mu);
+    c->alive = false;
+  }
   ::shutdown(c->fd, SHUT_RDWR);
   if (c->flow && tp_) tp_->close_flow(c->flow);
   c->rx_cv.notify_all();

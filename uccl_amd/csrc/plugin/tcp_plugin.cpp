@@ -48,6 +48,10 @@ struct Handle {
   uint16_t port;
 };
 
+// first bytes on every tcp-plane connection: rejects strays (e.g. another
+// process's connect-retry landing on a recycled ephemeral port)
+constexpr uint64_t kPluginCookie = 0x7563636c2d6e6574ULL;  // "uccl-net"
+
 struct Request {
   std::atomic<int> done{0};
   int size = 0;
@@ -154,7 +158,12 @@ struct WireHdr {
 };
 
 Comm::~Comm() {
-  alive = false;
+  {
+    // flag must flip under the mutex or a worker between its predicate
+    // check and cv sleep misses the wake forever (lost wakeup)
+    std::lock_guard<std::mutex> g(mu);
+    alive = false;
+  }
   cv.notify_all();
   if (fd >= 0) ::shutdown(fd, SHUT_RDWR);
   if (flow) MpFabric::get().tp.close_flow(flow);  // unblock recv_msg
@@ -342,6 +351,8 @@ ncclResult_t p_connect(int dev, void* opaque, void** sendComm) {
   }
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+  uint64_t cookie = kPluginCookie;
+  send_all(fd, &cookie, sizeof(cookie));
   auto* c = new Comm();
   c->fd = fd;
   c->sender = true;
@@ -378,6 +389,17 @@ ncclResult_t p_accept(void* listenComm, void** recvComm) {
   }
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+  // validate the cookie (bounded wait) — drop stray connections
+  timeval tv{2, 0};
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  uint64_t cookie = 0;
+  if (!recv_all(fd, &cookie, sizeof(cookie)) || cookie != kPluginCookie) {
+    ::close(fd);
+    *recvComm = nullptr;
+    return ncclSuccess;
+  }
+  timeval tv0{0, 0};
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv0, sizeof(tv0));
   auto* c = new Comm();
   c->fd = fd;
   c->worker = std::thread(rx_loop, c);

@@ -451,12 +451,14 @@ struct TransportEndpoint::Impl {
 
   // ---- flow setup over TCP ctrl ----
   struct CtrlMsg {
+    uint64_t magic;  // rejects stray connections on recycled ports
     uint64_t flow;
     uint64_t tag;  // connector-supplied peer identity
     int num_paths;
     uint16_t ports[64];
     char ip[48];
   };
+  static constexpr uint64_t kCtrlMagic = 0x756363746e737074ULL;
 
   void ctrl_loop() {
     while (!stop) {
@@ -465,8 +467,11 @@ struct TransportEndpoint::Impl {
         if (stop) return;
         continue;
       }
+      timeval tv{2, 0};  // bounded read; strays must not stall accepts
+      setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
       CtrlMsg peer{};
-      if (!net::recv_all(fd, &peer, sizeof(peer))) {
+      if (!net::recv_all(fd, &peer, sizeof(peer)) ||
+          peer.magic != kCtrlMagic) {
         ::close(fd);
         continue;
       }
@@ -486,6 +491,7 @@ struct TransportEndpoint::Impl {
 
   CtrlMsg self_ctrl(uint64_t flow) {
     CtrlMsg m{};
+    m.magic = kCtrlMagic;
     m.flow = flow;
     m.num_paths = num_paths;
     for (int i = 0; i < num_paths; ++i) m.ports[i] = ports[i];
@@ -598,7 +604,10 @@ void TransportEndpoint::close_flow(uint64_t flow) {
 }
 
 void TransportEndpoint::shutdown() {
-  if (impl_->stop.exchange(true)) return;
+  {
+    std::lock_guard<std::mutex> g(impl_->mu);  // lost-wakeup guard
+    if (impl_->stop.exchange(true)) return;
+  }
   ::shutdown(impl_->ctrl_listen, SHUT_RDWR);
   impl_->wake();
   impl_->cv.notify_all();
@@ -630,7 +639,9 @@ uint64_t TransportEndpoint::connect(const std::string& md, uint64_t tag) {
   mine.tag = tag;
   net::send_all(fd, &mine, sizeof(mine));
   Impl::CtrlMsg peer{};
-  UCCL_CHECK(net::recv_all(fd, &peer, sizeof(peer))) << "ctrl handshake";
+  UCCL_CHECK(net::recv_all(fd, &peer, sizeof(peer)) &&
+             peer.magic == Impl::kCtrlMagic)
+      << "ctrl handshake";
   ::close(fd);
   impl_->install_flow(peer.flow, peer);
   return peer.flow;
