@@ -29,6 +29,7 @@ SOURCES = [
     CSRC / "ep" / "ep_kernels.hip",
     CSRC / "ep" / "ep_buffer.cpp",
     CSRC / "ep" / "ep_proxy.cpp",
+    CSRC / "ukernel" / "ukernel.cpp",
     CSRC / "bindings" / "module.cpp",
 ]
 

@@ -14,6 +14,7 @@
 #include "../ep/ep_buffer.h"
 #include "../p2p/endpoint.h"
 #include "../transport/reliable.h"
+#include "../ukernel/ukernel.h"
 
 namespace py = pybind11;
 
@@ -423,4 +424,68 @@ PYBIND11_MODULE(_C, m) {
              e.recv_msg(flow, p, n);
            })
       .def("stats", &TransportEndpoint::stats);
+
+  // --- chunk-graph planner / spray executor (csrc/ukernel) -----------------
+  // reference analog: experimental/ukernel planner->lower->SprayExecutor,
+  // unit-tested the same way (host mock backend, single process)
+  namespace uk = uccl::uk;
+  py::class_<uk::Topology>(m, "UkTopology")
+      .def(py::init<int>(), py::arg("world"))
+      .def_readonly("world", &uk::Topology::world)
+      .def("set_link_weight",
+           [](uk::Topology& t, int src, int dst, double w) {
+             TORCH_CHECK(src >= 0 && src < t.world && dst >= 0 &&
+                         dst < t.world && src != dst, "bad link");
+             t.link_weight[size_t(src) * t.world + dst] = w;
+           });
+  py::class_<uk::ChunkGraph>(m, "UkGraph")
+      .def_property_readonly(
+          "num_tasks",
+          [](uk::ChunkGraph const& g) { return g.tasks.size(); })
+      .def_readonly("world", &uk::ChunkGraph::world)
+      .def_readonly("scratch_bytes", &uk::ChunkGraph::scratch_bytes)
+      .def("dump", &uk::ChunkGraph::dump);
+  m.def("uk_plan_sendrecv", &uk::plan_sendrecv_spray, py::arg("topo"),
+        py::arg("src"), py::arg("dst"), py::arg("nbytes"),
+        py::arg("chunk_bytes"));
+  m.def("uk_plan_allreduce_rsag", &uk::plan_allreduce_rsag, py::arg("topo"),
+        py::arg("nbytes"), py::arg("elem_bytes") = 4,
+        py::arg("chunk_bytes") = 1 << 20);
+  m.def("uk_plan_allreduce_oneshot", &uk::plan_allreduce_oneshot,
+        py::arg("topo"), py::arg("nbytes"), py::arg("elem_bytes") = 4);
+  m.def("uk_plan_broadcast", &uk::plan_broadcast, py::arg("topo"),
+        py::arg("root"), py::arg("nbytes"), py::arg("chunk_bytes"));
+  m.def("uk_lower", &uk::lower);
+  m.def("uk_execute_host",
+        [](uk::ChunkGraph const& g, std::vector<at::Tensor> inputs,
+           int64_t out_bytes) {
+          int const world = g.world;
+          TORCH_CHECK(static_cast<int>(inputs.size()) == world,
+                      "one input tensor per rank");
+          for (auto const& t : inputs)
+            TORCH_CHECK(t.is_contiguous() && !t.is_cuda() &&
+                            t.scalar_type() == at::kFloat,
+                        "host float32 contiguous inputs");
+          uint64_t const in_bytes = inputs[0].numel() * 4;
+          uk::HostBackend hb(world, in_bytes, out_bytes, g.scratch_bytes);
+          for (int r = 0; r < world; ++r)
+            std::memcpy(hb.input(r), inputs[r].data_ptr<float>(), in_bytes);
+          uk::ExecStats st;
+          {
+            py::gil_scoped_release rel;
+            st = uk::execute(g, hb);
+          }
+          std::vector<at::Tensor> outs;
+          for (int r = 0; r < world; ++r) {
+            auto t = at::empty({out_bytes / 4}, at::kFloat);
+            std::memcpy(t.data_ptr<float>(), hb.output(r), out_bytes);
+            outs.push_back(t);
+          }
+          py::dict d;
+          d["tasks_run"] = st.tasks_run;
+          d["wait_requeues"] = st.wait_requeues;
+          d["link_bytes"] = st.link_bytes;
+          return py::make_tuple(outs, d);
+        },
+        py::arg("graph"), py::arg("inputs"), py::arg("out_bytes"));
 }
