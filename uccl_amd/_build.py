@@ -30,6 +30,7 @@ SOURCES = [
     CSRC / "ep" / "ep_buffer.cpp",
     CSRC / "ep" / "ep_proxy.cpp",
     CSRC / "ukernel" / "ukernel.cpp",
+    CSRC / "p2p" / "compress.cpp",
     CSRC / "bindings" / "module.cpp",
 ]
 
@@ -190,6 +191,7 @@ def build(verbose: bool = False, force: bool = False) -> Path:
             "-lc10",
             "-lc10_hip",
             "-lamdhip64",
+            "-lz",
             f"-Wl,-rpath,{libdir}",
             f"-L{ROCM}/lib",
             f"-Wl,-rpath,{ROCM}/lib",

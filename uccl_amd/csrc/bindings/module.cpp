@@ -14,6 +14,7 @@
 #include "../ep/ep_buffer.h"
 #include "../p2p/endpoint.h"
 #include "../transport/reliable.h"
+#include "../p2p/compress.h"
 #include "../ukernel/ukernel.h"
 
 namespace py = pybind11;
@@ -424,6 +425,52 @@ PYBIND11_MODULE(_C, m) {
              e.recv_msg(flow, p, n);
            })
       .def("stats", &TransportEndpoint::stats);
+
+  // --- lossless float codec (csrc/p2p/compress) ----------------------------
+  // reference analog: DietGPU compression layer, p2p/rdma/compression.cc
+  namespace comp = uccl::p2p::comp;
+  m.def("comp_compress",
+        [](at::Tensor t, int strategy) {
+          TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),
+                      "compress: host contiguous tensors");
+          int elem = static_cast<int>(t.element_size());
+          int code;
+          switch (t.scalar_type()) {
+            case at::kFloat: code = 0; break;
+            case at::kHalf: code = 1; break;
+            case at::kBFloat16: code = 2; break;
+            default: code = 3; elem = 1;
+          }
+          size_t const nbytes = t.numel() * t.element_size();
+          std::string s;
+          {
+            py::gil_scoped_release rel;
+            s = comp::compress(t.data_ptr(), nbytes, elem, code, strategy);
+          }
+          auto out = at::empty({static_cast<int64_t>(s.size())}, at::kByte);
+          std::memcpy(out.data_ptr(), s.data(), s.size());
+          return out;
+        },
+        py::arg("tensor"), py::arg("strategy") = static_cast<int>(comp::kSplitDeflate));
+  m.def("comp_decompress", [](at::Tensor frame) {
+    TORCH_CHECK(frame.is_contiguous() && !frame.is_cuda() &&
+                    frame.scalar_type() == at::kByte,
+                "decompress: host uint8 frame");
+    size_t const fb = frame.numel();
+    size_t const n = comp::orig_bytes(frame.data_ptr(), fb);
+    int const code = comp::dtype_code(frame.data_ptr(), fb);
+    at::ScalarType const st = code == 0   ? at::kFloat
+                              : code == 1 ? at::kHalf
+                              : code == 2 ? at::kBFloat16
+                                          : at::kByte;
+    int const es = code == 3 ? 1 : (code == 0 ? 4 : 2);
+    auto out = at::empty({static_cast<int64_t>(n / es)}, st);
+    {
+      py::gil_scoped_release rel;
+      comp::decompress(frame.data_ptr(), fb, out.data_ptr(), n);
+    }
+    return out;
+  });
 
   // --- chunk-graph planner / spray executor (csrc/ukernel) -----------------
   // reference analog: experimental/ukernel planner->lower->SprayExecutor,
