@@ -31,6 +31,7 @@ SOURCES = [
     CSRC / "ep" / "ep_proxy.cpp",
     CSRC / "ukernel" / "ukernel.cpp",
     CSRC / "p2p" / "compress.cpp",
+    CSRC / "core" / "trace.cpp",
     CSRC / "bindings" / "module.cpp",
 ]
 
@@ -67,6 +68,7 @@ def build_plugin(verbose: bool = False) -> Path:
     capi = plugdir / "libuccl_p2p.so"
     capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
                  CSRC / "transport" / "reliable.cpp",
+                 CSRC / "core" / "trace.cpp",
                  CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h",
                  CSRC / "transport" / "reliable.h"]
     shim_check = plugdir / "libuccl_nccl.so"
@@ -77,9 +79,10 @@ def build_plugin(verbose: bool = False) -> Path:
     import subprocess as sp
 
     reliable = CSRC / "transport" / "reliable.cpp"
+    tracecc = CSRC / "core" / "trace.cpp"
     for cmd in (
         ["g++", "-O2", "-std=c++17", "-fPIC", "-shared", str(src),
-         str(reliable), "-o", str(target), "-pthread"],
+         str(reliable), str(tracecc), "-o", str(target), "-pthread"],
         ["g++", "-O2", "-std=c++17", str(harness_src), "-o", str(harness),
          "-ldl", "-pthread"],
     ):

@@ -15,6 +15,8 @@
 #include "../p2p/endpoint.h"
 #include "../transport/reliable.h"
 #include "../p2p/compress.h"
+#include "../core/latency.h"
+#include "../core/trace.h"
 #include "../ukernel/ukernel.h"
 
 namespace py = pybind11;
@@ -389,7 +391,9 @@ PYBIND11_MODULE(_C, m) {
       .def_readonly("msgs_sent", &uccl::transport::Stats::msgs_sent)
       .def_readonly("msgs_recv", &uccl::transport::Stats::msgs_recv)
       .def_readonly("srtt_us", &uccl::transport::Stats::srtt_us)
-      .def_readonly("cwnd", &uccl::transport::Stats::cwnd);
+      .def_readonly("cwnd", &uccl::transport::Stats::cwnd)
+      .def_readonly("rtt_p50_us", &uccl::transport::Stats::rtt_p50_us)
+      .def_readonly("rtt_p99_us", &uccl::transport::Stats::rtt_p99_us);
 
   py::class_<TransportEndpoint>(m, "TransportEndpoint")
       .def(py::init<int, size_t>(), py::arg("num_paths") = 8,
@@ -425,6 +429,22 @@ PYBIND11_MODULE(_C, m) {
              e.recv_msg(flow, p, n);
            })
       .def("stats", &TransportEndpoint::stats);
+
+  // --- tracing / latency observability (csrc/core) -------------------------
+  // reference analogs: NPKit event tracing (lite-collective core/npkit.cc)
+  // and the latency percentile recorder (include/util/latency.h)
+  m.def("trace_enabled", &uccl::trace::enabled);
+  m.def("trace_set_enabled", &uccl::trace::set_enabled);
+  m.def("trace_dump_json", &uccl::trace::dump_json);
+  m.def("trace_clear", &uccl::trace::clear);
+  m.def("trace_num_events", &uccl::trace::num_events);
+  py::class_<uccl::LatencyHist>(m, "LatencyHist")
+      .def(py::init<>())
+      .def("record_us", &uccl::LatencyHist::record_us)
+      .def("record_ns", &uccl::LatencyHist::record_ns)
+      .def("count", &uccl::LatencyHist::count)
+      .def("percentile_us", &uccl::LatencyHist::percentile_us)
+      .def("reset", &uccl::LatencyHist::reset);
 
   // --- lossless float codec (csrc/p2p/compress) ----------------------------
   // reference analog: DietGPU compression layer, p2p/rdma/compression.cc

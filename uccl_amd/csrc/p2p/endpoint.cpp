@@ -8,6 +8,7 @@
 
 #include "../core/env.h"
 #include "../core/log.h"
+#include "../core/trace.h"
 #include "../core/net.h"
 #include "../transport/reliable.h"
 
@@ -515,6 +516,7 @@ static bool ipc_enabled() {
 }
 
 void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
+  trace::Span span("p2p", "send");
   if (is_gpu(device) && c.same_host && ipc_enabled()) {
     // one-copy IPC path: ship {handle, offset}; receiver DtoD-copies
     IpcBlob blob{};
@@ -584,6 +586,7 @@ void Endpoint::copy_to_user(RxItem& item, void* dst, size_t bytes,
 }
 
 void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
+  trace::Span span("p2p", "recv");
   std::shared_ptr<RxItem> item;
   {
     std::unique_lock<std::mutex> lk(c.rx_mu);
@@ -609,6 +612,7 @@ void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
 
 void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
                         Advert ad) {
+  trace::Span span("p2p", "write");
   UCCL_CHECK(bytes <= ad.bytes) << "write larger than advertised window";
   uint64_t token = c.next_token++;
   if (is_gpu(device) && c.same_host && ipc_enabled()) {
@@ -643,6 +647,7 @@ void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
 
 void Endpoint::do_read(Conn& c, void* ptr, size_t bytes, int device,
                        Advert ad) {
+  trace::Span span("p2p", "read");
   UCCL_CHECK(bytes <= ad.bytes) << "read larger than advertised window";
   uint64_t token = c.next_token++;
   c.send_msg(MsgHdr{kReadReq, ad.mr_id, ad.offset, bytes, token});

@@ -10,6 +10,8 @@
 #include <cstring>
 
 #include "../core/env.h"
+#include "../core/latency.h"
+#include "../core/trace.h"
 #include "../core/log.h"
 #include "../core/net.h"
 
@@ -158,6 +160,7 @@ struct TransportEndpoint::Impl {
   std::atomic<uint64_t> next_flow{1};
 
   Stats st;
+  LatencyHist rtt_hist;
 
   // ---- helpers ----
   void wake() {
@@ -314,6 +317,7 @@ struct TransportEndpoint::Impl {
     if (a.credit_cum > f.credit_limit) f.credit_limit = a.credit_cum;
     if (a.ts_echo) {
       double const rtt_us = (now_ns() - a.ts_echo) / 1000.0;
+      rtt_hist.record_us(rtt_us);
       timely_update(f, rtt_us);
     }
     // cumulative
@@ -663,6 +667,7 @@ uint64_t TransportEndpoint::accept(uint64_t* peer_tag) {
 
 void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
                                  size_t bytes) {
+  trace::Span span("transport", "send_msg");
   std::shared_ptr<MsgTx> m;
   {
     std::lock_guard<std::mutex> g(impl_->mu);
@@ -685,6 +690,7 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
 }
 
 void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
+  trace::Span span("transport", "recv_msg");
   uint64_t msg_id;
   {
     std::lock_guard<std::mutex> g(impl_->mu);
@@ -718,7 +724,10 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
 
 Stats TransportEndpoint::stats() const {
   std::lock_guard<std::mutex> g(impl_->mu);
-  return impl_->st;
+  Stats st = impl_->st;
+  st.rtt_p50_us = impl_->rtt_hist.percentile_us(50);
+  st.rtt_p99_us = impl_->rtt_hist.percentile_us(99);
+  return st;
 }
 
 }  // namespace transport

@@ -47,6 +47,8 @@ struct Stats {
   uint64_t msgs_recv = 0;
   double srtt_us = 0;
   double cwnd = 0;
+  double rtt_p50_us = 0;  // from the log-bucket latency recorder
+  double rtt_p99_us = 0;
 };
 
 class TransportEndpoint {

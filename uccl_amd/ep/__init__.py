@@ -103,3 +103,11 @@ class Buffer:
     # DeepEP-compatible aliases (low-latency intranode semantics)
     low_latency_dispatch = dispatch
     low_latency_combine = combine
+
+
+def __getattr__(name):
+    if name in ("ElasticBuffer", "MembershipChanged", "expert_rank_table"):
+        from uccl_amd.ep import elastic
+
+        return getattr(elastic, name)
+    raise AttributeError(name)
