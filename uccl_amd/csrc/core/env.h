@@ -27,6 +27,26 @@ inline bool env_bool(const char* name, bool dflt) {
            e[0] == 'F');
 }
 
+// UCCL_<X> with NCCL_<X> / RCCL_<X> fallback (the reference's param.h
+// honors NCCL aliases, param.h:31-44): pass the UCCL-prefixed name.
+inline const char* env_aliased(const char* uccl_name) {
+  const char* e = std::getenv(uccl_name);
+  if (e && *e) return e;
+  std::string tail(uccl_name);
+  if (tail.rfind("UCCL_", 0) == 0) tail = tail.substr(5);
+  std::string n = "NCCL_" + tail;
+  if ((e = std::getenv(n.c_str())) && *e) return e;
+  n = "RCCL_" + tail;
+  if ((e = std::getenv(n.c_str())) && *e) return e;
+  return nullptr;
+}
+
+inline std::string env_str_aliased(const char* uccl_name,
+                                   const std::string& dflt) {
+  const char* e = env_aliased(uccl_name);
+  return e ? std::string(e) : dflt;
+}
+
 // Declares a lazily-cached env parameter accessor:  uccl_param_Foo()
 #define UCCL_PARAM(Name, Env, Default)              \
   inline int64_t uccl_param_##Name() {              \

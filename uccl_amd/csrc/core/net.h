@@ -5,6 +5,8 @@
 
 #include <arpa/inet.h>
 #include <ifaddrs.h>
+
+#include "env.h"
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
@@ -92,17 +94,23 @@ inline bool recv_all(int fd, void* buf, size_t n) {
 }
 
 // Best-effort non-loopback IPv4 of this host (for metadata blobs).
+// UCCL_SOCKET_IFNAME (NCCL_/RCCL_SOCKET_IFNAME honored as aliases, like
+// the reference's param system) pins the interface by name prefix.
 inline std::string local_ip() {
   std::string result = "127.0.0.1";
   ifaddrs* ifs = nullptr;
   if (getifaddrs(&ifs) != 0) return result;
+  std::string want = env_str_aliased("UCCL_SOCKET_IFNAME", "");
   for (ifaddrs* it = ifs; it; it = it->ifa_next) {
     if (!it->ifa_addr || it->ifa_addr->sa_family != AF_INET) continue;
+    if (!want.empty() &&
+        std::string(it->ifa_name).rfind(want, 0) != 0)
+      continue;
     auto* sin = reinterpret_cast<sockaddr_in*>(it->ifa_addr);
     char buf[INET_ADDRSTRLEN];
     inet_ntop(AF_INET, &sin->sin_addr, buf, sizeof(buf));
     std::string ip(buf);
-    if (ip != "127.0.0.1") {
+    if (ip != "127.0.0.1" || !want.empty()) {
       result = ip;
       break;
     }
