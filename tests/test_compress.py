@@ -107,3 +107,36 @@ def test_endpoint_compressed_transfer():
     rth.join()
     assert torch.equal(got["t"], src)
     assert wire_bytes < src.numel() * 2  # actually compressed on the wire
+
+
+def test_object_transfer():
+    # Ray-API-analog object send: nested state dict with tensors
+    a = p2p.Endpoint(gpu=0, num_workers=1)
+    b = p2p.Endpoint(gpu=0, num_workers=1)
+    ids = {}
+    th = threading.Thread(target=lambda: ids.setdefault("b", b.accept()))
+    th.start()
+    cid_a = a.connect(b.metadata())
+    th.join()
+
+    obj = {
+        "step": 1234,
+        "lr": 3e-4,
+        "weights": {"w1": torch.randn(64, 32).bfloat16(),
+                    "ids": torch.arange(100, dtype=torch.int64)},
+        "shapes": [(1, 2), [3, torch.zeros(5)]],
+    }
+    got = {}
+    rx = threading.Thread(
+        target=lambda: got.setdefault("o", p2p.recv_object(b, ids["b"])))
+    rx.start()
+    p2p.send_object(a, cid_a, obj)
+    rx.join()
+
+    o = got["o"]
+    assert o["step"] == 1234 and o["lr"] == 3e-4
+    assert torch.equal(o["weights"]["w1"], obj["weights"]["w1"])
+    assert o["weights"]["ids"].dtype == torch.int64
+    assert torch.equal(o["weights"]["ids"], obj["weights"]["ids"])
+    assert o["shapes"][0] == (1, 2)
+    assert torch.equal(o["shapes"][1][1], torch.zeros(5))

@@ -523,6 +523,12 @@ PYBIND11_MODULE(_C, m) {
   m.def("uk_plan_broadcast", &uk::plan_broadcast, py::arg("topo"),
         py::arg("root"), py::arg("nbytes"), py::arg("chunk_bytes"));
   m.def("uk_lower", &uk::lower);
+  m.def("uk_estimate_us", &uk::estimate_us, py::arg("graph"),
+        py::arg("topo"), py::arg("link_gbps") = 150.0,
+        py::arg("local_gbps") = 1500.0, py::arg("overhead_us") = 4.0);
+  m.def("uk_plan_allreduce_auto", &uk::plan_allreduce_auto, py::arg("topo"),
+        py::arg("nbytes"), py::arg("elem_bytes") = 4,
+        py::arg("chunk_bytes") = 1 << 20);
   m.def("uk_execute_host",
         [](uk::ChunkGraph const& g, std::vector<at::Tensor> inputs,
            int64_t out_bytes) {

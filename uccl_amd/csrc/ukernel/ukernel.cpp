@@ -265,6 +265,71 @@ ChunkGraph plan_broadcast(Topology const& topo, int root, uint64_t nbytes,
 }
 
 // ---------------------------------------------------------------------------
+// Cost model
+// ---------------------------------------------------------------------------
+
+double estimate_us(ChunkGraph const& g, Topology const& topo,
+                   double link_gbps, double local_gbps, double overhead_us) {
+  size_t const n = g.tasks.size();
+  int const world = g.world;
+  // topological order (deps are acyclic; lower() or planners guarantee it)
+  std::vector<int> indeg(n, 0);
+  std::vector<std::vector<int>> out(n);
+  for (size_t i = 0; i < n; ++i)
+    for (int d : g.tasks[i].deps) {
+      indeg[i]++;
+      out[d].push_back(static_cast<int>(i));
+    }
+  std::deque<int> q;
+  for (size_t i = 0; i < n; ++i)
+    if (!indeg[i]) q.push_back(static_cast<int>(i));
+
+  std::vector<double> finish(n, 0.0);
+  std::vector<double> link_free(size_t(world) * world, 0.0);
+  std::vector<double> rank_free(world, 0.0);
+  double total = 0.0;
+  size_t seen = 0;
+  while (!q.empty()) {
+    int const i = q.front();
+    q.pop_front();
+    ++seen;
+    Task const& t = g.tasks[i];
+    double ready = 0.0;
+    for (int d : t.deps) ready = std::max(ready, finish[d]);
+    double dur = overhead_us;
+    double* res = nullptr;
+    if (t.op == Op::kPut) {
+      double const w = topo.weight(t.src.rank, t.dst.rank);
+      double const rate = link_gbps * (w > 0 ? w : 1e-3);  // GB/s
+      dur = overhead_us + t.bytes / rate * 1e-3;           // B/(GB/s)=ns
+      res = &link_free[size_t(t.src.rank) * world + t.dst.rank];
+    } else if (t.op == Op::kCopy || t.op == Op::kReduce) {
+      double const mult = t.op == Op::kReduce ? 2.0 : 1.0;  // rd+rd+wr vs rd+wr
+      dur = overhead_us + mult * t.bytes / local_gbps * 1e-3;
+      res = &rank_free[t.rank];
+    }
+    double const start = res ? std::max(ready, *res) : ready;
+    finish[i] = start + dur;
+    if (res) *res = finish[i];
+    total = std::max(total, finish[i]);
+    for (int s2 : out[i])
+      if (--indeg[s2] == 0) q.push_back(s2);
+  }
+  if (seen != n) throw std::invalid_argument("estimate: graph has a cycle");
+  return total;
+}
+
+ChunkGraph plan_allreduce_auto(Topology const& topo, uint64_t nbytes,
+                               uint64_t elem_bytes, uint64_t chunk_bytes) {
+  ChunkGraph one = plan_allreduce_oneshot(topo, nbytes, elem_bytes);
+  ChunkGraph rsag = plan_allreduce_rsag(topo, nbytes, elem_bytes,
+                                        chunk_bytes);
+  double const t1 = estimate_us(one, topo);
+  double const t2 = estimate_us(rsag, topo);
+  return lower(t1 <= t2 ? one : rsag);
+}
+
+// ---------------------------------------------------------------------------
 // Lowering
 // ---------------------------------------------------------------------------
 

@@ -124,6 +124,24 @@ ChunkGraph plan_broadcast(Topology const& topo, int root, uint64_t nbytes,
                           uint64_t chunk_bytes);
 
 // ---------------------------------------------------------------------------
+// Cost model: deterministic list-scheduling estimate of a plan's
+// completion time. Resources: every directed link serializes its kPuts at
+// link_gbps x topo.weight; every rank serializes local kCopy/kReduce at
+// local_gbps; signal/wait cost overhead_us each. This is what lets the
+// planner CHOOSE an algorithm per (topology, size) instead of hard
+// thresholds (the reference planner's topology-aware role).
+// ---------------------------------------------------------------------------
+double estimate_us(ChunkGraph const& g, Topology const& topo,
+                   double link_gbps = 150.0, double local_gbps = 1500.0,
+                   double overhead_us = 4.0);
+
+// Build + lower the better allreduce plan for this size on this topology
+// (one-shot vs RS+AG by estimated time).
+ChunkGraph plan_allreduce_auto(Topology const& topo, uint64_t nbytes,
+                               uint64_t elem_bytes = 4,
+                               uint64_t chunk_bytes = 1 << 20);
+
+// ---------------------------------------------------------------------------
 // Lowering: validate the DAG, convert every cross-rank dependency edge
 // into an explicit kSignal/kWait pair, and assign unique flags.
 // Returns the lowered graph (topologically executable by rank workers).

@@ -106,7 +106,14 @@ def recv_compressed(ep, conn_id):
     frame = torch.empty(fbytes, dtype=torch.uint8)
     ep.recv(conn_id, frame)
     out = decompress(frame)
-    return out.view(*dims[:ndim]) if ndim else out
+    shape = dims[:ndim]
+    import math
+
+    want = math.prod(shape) if ndim else 1
+    if out.numel() == want:
+        return out.view(*shape) if ndim else out.view(())
+    return out  # non-float payload rode as raw bytes; caller restores dtype
+
 
 
 def __getattr__(name):
@@ -117,3 +124,78 @@ def __getattr__(name):
     if name in ("writev", "readv"):
         return getattr(_EndpointExtras, name)
     raise AttributeError(name)
+
+# --- object transfer (reference parity: the Ray-style object API on the
+# p2p engine, p2p/tests/test_ray_api.py — fast checkpoint/object movement;
+# tensors ride the codec, metadata rides pickle) ---------------------------
+
+def send_object(ep, conn_id, obj, compress_tensors: bool = True):
+    """Ship an arbitrary picklable object; torch tensors inside are
+    extracted and sent as (optionally compressed) binary payloads rather
+    than pickled bytes."""
+    import pickle
+    import struct
+
+    import torch
+
+    tensors = []
+
+    def strip(o):
+        if isinstance(o, torch.Tensor):
+            tensors.append(o.detach().contiguous().cpu())
+            return ("__uccl_tensor__", len(tensors) - 1,
+                    tuple(o.shape), str(o.dtype))
+        if isinstance(o, dict):
+            return {k: strip(v) for k, v in o.items()}
+        if isinstance(o, (list, tuple)):
+            t = [strip(v) for v in o]
+            return t if isinstance(o, list) else ("__uccl_tuple__", t)
+        return o
+
+    skeleton = pickle.dumps(strip(obj))
+    hdr = struct.pack("<IQI", 0x554F424A, len(skeleton), len(tensors))
+    ep.send(conn_id, torch.frombuffer(bytearray(hdr), dtype=torch.uint8))
+    ep.send(conn_id, torch.frombuffer(bytearray(skeleton),
+                                      dtype=torch.uint8))
+    for t in tensors:
+        if compress_tensors:
+            send_compressed(ep, conn_id, t)
+        else:
+            # uncompressed: reuse the framing with strategy none
+            send_compressed(ep, conn_id, t, strategy=STRATEGY_NONE)
+
+
+def recv_object(ep, conn_id):
+    import pickle
+    import struct
+
+    import torch
+
+    hdr = torch.empty(16, dtype=torch.uint8)
+    ep.recv(conn_id, hdr)
+    magic, skel_len, ntensors = struct.unpack("<IQI", bytes(hdr.tolist()))
+    assert magic == 0x554F424A, "recv_object: bad header"
+    skel = torch.empty(skel_len, dtype=torch.uint8)
+    ep.recv(conn_id, skel)
+    skeleton = pickle.loads(bytes(skel.tolist()))
+    tensors = [recv_compressed(ep, conn_id) for _ in range(ntensors)]
+
+    def rebuild(o):
+        if isinstance(o, tuple) and len(o) == 4 and \
+                o[0] == "__uccl_tensor__":
+            _, idx, shape, dtype = o
+            t = tensors[idx]
+            want = getattr(torch, dtype.replace("torch.", ""))
+            if t.dtype != want:  # non-float dtypes ride as raw bytes
+                t = t.view(want)
+            return t.view(*shape) if shape else t.view(())
+        if isinstance(o, tuple) and len(o) == 2 and \
+                o[0] == "__uccl_tuple__":
+            return tuple(rebuild(v) for v in o[1])
+        if isinstance(o, dict):
+            return {k: rebuild(v) for k, v in o.items()}
+        if isinstance(o, list):
+            return [rebuild(v) for v in o]
+        return o
+
+    return rebuild(skeleton)

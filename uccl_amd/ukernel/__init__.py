@@ -57,6 +57,21 @@ def lower(graph: Graph) -> Graph:
     return _C.uk_lower(graph)
 
 
+def estimate_us(graph: Graph, topo, link_gbps: float = 150.0,
+                local_gbps: float = 1500.0, overhead_us: float = 4.0):
+    """List-scheduling completion-time estimate: per-link serialization
+    at link_gbps x weight, per-rank local ops at local_gbps."""
+    return _C.uk_estimate_us(graph, topo, link_gbps, local_gbps,
+                             overhead_us)
+
+
+def plan_allreduce_auto(topo, nbytes: int, elem_bytes: int = 4,
+                        chunk_bytes: int = 1 << 20) -> Graph:
+    """Pick one-shot vs RS+AG by estimated time; returns a LOWERED
+    graph ready for execute_host."""
+    return _C.uk_plan_allreduce_auto(topo, nbytes, elem_bytes, chunk_bytes)
+
+
 def execute_host(graph: Graph, inputs: List[torch.Tensor],
                  out_bytes: int | None = None) -> Tuple[list, dict]:
     """Run a LOWERED graph on the host mock backend (one worker thread per
