@@ -220,3 +220,84 @@ def test_transport_pacing():
     xfer(a, b, fa, fb, 1 << 20, 61)
     dt = time.perf_counter() - t0
     assert dt >= 0.08, f"pacing had no effect ({dt*1e3:.1f} ms)"
+
+
+def test_transport_eqds_paced_pull():
+    """Paced pull quanta (UCCL_TP_EQDS_MBPS): the receiver doles credit
+    at the configured rate, so a 2MB transfer at 20 MB/s (with a 64KB
+    initial window) must take >= ~70ms."""
+    import time
+
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_CC="eqds",
+                                UCCL_TP_EQDS_MBPS=20, UCCL_TP_RWND_KB=64)
+    t0 = time.perf_counter()
+    xfer(a, b, fa, fb, 2 << 20, 71)
+    dt = time.perf_counter() - t0
+    assert dt >= 0.07, f"pull pacing had no effect ({dt*1e3:.1f} ms)"
+    st = a.stats()
+    assert st.msgs_sent == 1
+
+
+def test_transport_eqds_incast_sharing():
+    """Two senders incast into one receiver with a paced aggregate pull
+    rate: both must complete, and the total must take at least as long
+    as the aggregate rate allows (rate is split across active flows)."""
+    import time
+
+    import os
+
+    old = {}
+    env = {"UCCL_TP_CC": "eqds", "UCCL_TP_EQDS_MBPS": "40",
+           "UCCL_TP_RWND_KB": "64", "UCCL_TP_CWND_MAX": "256"}
+    for k, v in env.items():
+        old[k] = os.environ.get(k)
+        os.environ[k] = v
+    try:
+        from uccl_amd import _load_native
+
+        C = _load_native(required=False)
+        sink = C.TransportEndpoint(num_paths=2, chunk_bytes=4096)
+        srcs = [C.TransportEndpoint(num_paths=2, chunk_bytes=4096)
+                for _ in range(2)]
+        sink_flows = []
+
+        def acceptor():
+            for _ in range(2):
+                sink_flows.append(sink.accept())
+
+        t = threading.Thread(target=acceptor, daemon=True)
+        t.start()
+        src_flows = [sp.connect(sink.metadata()) for sp in srcs]
+        t.join(timeout=30)
+
+        n = 1 << 20
+        payloads = [torch.randint(0, 255, (n,), dtype=torch.uint8)
+                    for _ in range(2)]
+        outs = [torch.zeros(n, dtype=torch.uint8) for _ in range(2)]
+        t0 = time.perf_counter()
+        rx = []
+        for i, f in enumerate(sink_flows):
+            th = threading.Thread(target=lambda i=i, f=f:
+                                  sink.recv(f, outs[i]), daemon=True)
+            th.start()
+            rx.append(th)
+        tx = []
+        for i, (sp, f) in enumerate(zip(srcs, src_flows)):
+            th = threading.Thread(target=lambda i=i, sp=sp, f=f:
+                                  sp.send(f, payloads[i]), daemon=True)
+            th.start()
+            tx.append(th)
+        for th in tx + rx:
+            th.join(timeout=60)
+        dt = time.perf_counter() - t0
+        # 2MB total at 40MB/s aggregate minus 2x64KB initial windows
+        assert dt >= 0.035, f"incast pull pacing absent ({dt*1e3:.1f} ms)"
+        got = sorted(o.long().sum().item() for o in outs)
+        want = sorted(p.long().sum().item() for p in payloads)
+        assert got == want
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v

@@ -125,6 +125,10 @@ struct TransportEndpoint::Flow {
   uint64_t next_post_msg = 0;   // msg_id the next recv_msg call will take
   uint64_t next_done_msg = 0;   // completion watermark for in-order delivery
   uint64_t last_data_ts = 0;    // ts to echo in acks
+  // eqds pull pacer (receiver side): paced cumulative grant
+  uint64_t granted = 0;
+  uint64_t last_grant_ns = 0;
+  uint64_t last_rx_ns = 0;
 };
 
 struct TransportEndpoint::Impl {
@@ -139,6 +143,7 @@ struct TransportEndpoint::Impl {
   double t_low_us = 50, t_high_us = 1000, swift_target_us = 300;
   double cwnd_max = 1024;
   uint64_t rwnd = 4u << 20;
+  double eqds_bytes_per_ns = 0;  // UCCL_TP_EQDS_MBPS: paced pull quanta
   int dup_thres = 32;
   uint64_t rto_base_ns = 20000000;
   int rto_abort_thres = 50;
@@ -229,19 +234,47 @@ struct TransportEndpoint::Impl {
     }
   }
 
-  void send_ack(Flow& f, int sock_idx, sockaddr_in const& to) {
-    // grant: allow the sender to stay rwnd bytes ahead of what we've seen
-    uint64_t const grant = f.bytes_received + rwnd_bytes();
+  // flows with inbound activity in the last 100 ms share the pull rate
+  int active_inbound() {
+    uint64_t const now = now_ns();
+    int n = 0;
+    for (auto const& [id, fp] : flows)
+      if (fp && now - fp->last_rx_ns < 100'000'000ull) ++n;
+    return n > 0 ? n : 1;
+  }
+
+  void send_ack(Flow& f, int sock_idx, sockaddr_in const& to,
+                bool allow_drop = true) {
+    // grant: allow the sender to stay rwnd bytes ahead of what we've
+    // seen; in paced-EQDS mode (UCCL_TP_EQDS_MBPS) the receiver doles
+    // that window out as pull quanta at the configured aggregate rate,
+    // split evenly across flows with inbound demand — the EQDS incast
+    // discipline (reference include/cc/eqds.h pull-quanta pacer)
+    uint64_t grant = f.bytes_received + rwnd_bytes();
+    if (eqds_bytes_per_ns > 0) {
+      uint64_t const now = now_ns();
+      if (!f.last_grant_ns) {
+        f.last_grant_ns = now;
+        f.granted = rwnd_bytes();  // match the sender's initial credit
+      }
+      double const share = eqds_bytes_per_ns / active_inbound();
+      auto const quanta =
+          static_cast<uint64_t>((now - f.last_grant_ns) * share);
+      f.granted = std::min(f.granted + quanta, grant);
+      f.last_grant_ns = now;
+      grant = f.granted;
+    }
     // independent ACK-loss injection (exercises cumulative-ack coverage
     // and the RTO backstop on reverse-path drops)
     // key the decision on data_recv (which always advances) — keying on
     // acks_sent would freeze the hash after a drop and drop forever
-    if (ack_loss_pct > 0 &&
+    if (allow_drop && ack_loss_pct > 0 &&
         inject_drop(static_cast<uint32_t>(st.data_recv), 7, ack_loss_pct)) {
       ++st.injected_drops;
       return;
     }
     AckHdr a{kMagic, kAck, f.id, f.rx_cum, 0, 0, 0, f.last_data_ts, grant};
+
     for (auto const& [csn, _] : f.rx_ooo) {
       uint32_t const d = csn - f.rx_cum;
       if (d < 64)
@@ -381,6 +414,7 @@ struct TransportEndpoint::Impl {
         cv.notify_all();
       }
     }
+    f.last_rx_ns = now_ns();
     send_ack(f, sock_idx, from);
   }
 
@@ -450,6 +484,21 @@ struct TransportEndpoint::Impl {
         }
       }
       rto_scan();
+      // paced-EQDS credit refresh: a credit-stalled sender emits no
+      // data, so acks (which carry grants) would never flow again —
+      // the receiver must top up pulls from the progress loop
+      if (eqds_bytes_per_ns > 0) {
+        uint64_t const now = now_ns();
+        for (auto& [id, fp] : flows) {
+          if (!fp || !fp->last_grant_ns) continue;
+          if (now - fp->last_rx_ns > 500'000'000ull) continue;  // idle
+          if (fp->granted >= fp->bytes_received + rwnd_bytes()) continue;
+          // refresh acks bypass loss injection: the drop hash is keyed
+          // on data_recv, which is frozen while the sender is stalled,
+          // so an injected drop here would repeat forever (livelock)
+          send_ack(*fp, 0, fp->peer_paths[0], /*allow_drop=*/false);
+        }
+      }
     }
   }
 
@@ -543,6 +592,8 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
       static_cast<double>(env_int("UCCL_TP_SWIFT_TARGET_US", 300));
   impl_->cwnd_max =
       static_cast<double>(env_int("UCCL_TP_CWND_MAX", 1024));
+  impl_->eqds_bytes_per_ns =
+      static_cast<double>(env_int("UCCL_TP_EQDS_MBPS", 0)) * 1e-3;
   impl_->rwnd = static_cast<uint64_t>(env_int("UCCL_TP_RWND_KB", 4096))
                 << 10;
   impl_->dup_thres =
