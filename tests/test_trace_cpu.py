@@ -103,3 +103,58 @@ def test_transport_stats_rtt_percentiles():
     st = a.stats()
     assert st.rtt_p50_us > 0
     assert st.rtt_p99_us >= st.rtt_p50_us
+
+
+# --- lockless rings (reference jring analog, core/ring.h) -----------------
+
+def test_spsc_ring_order_and_capacity():
+    r = C.SpscRingU64(8)
+    for i in range(8):
+        assert r.push(i)
+    assert not r.push(99)  # full
+    assert r.size() == 8
+    for i in range(8):
+        assert r.pop() == i  # FIFO
+    assert r.pop() is None
+
+
+def test_mpmc_ring_threads():
+    r = C.MpmcRingU64(1 << 12)
+    nprod, nitems = 4, 20000
+    popped = []
+    lock = threading.Lock()
+    stop = threading.Event()
+
+    def producer(base):
+        for i in range(nitems):
+            v = base * nitems + i
+            while not r.push(v):
+                pass
+
+    def consumer():
+        local = []
+        while not stop.is_set() or r.size_approx():
+            v = r.pop()
+            if v is not None:
+                local.append(v)
+        with lock:
+            popped.extend(local)
+
+    cons = [threading.Thread(target=consumer) for _ in range(3)]
+    prods = [threading.Thread(target=producer, args=(b,))
+             for b in range(nprod)]
+    for t in cons + prods:
+        t.start()
+    for t in prods:
+        t.join(timeout=60)
+    stop.set()
+    for t in cons:
+        t.join(timeout=60)
+    assert sorted(popped) == list(range(nprod * nitems))
+
+
+def test_ring_capacity_validation():
+    import pytest
+
+    with pytest.raises(Exception):
+        C.MpmcRingU64(100)  # not a power of two

@@ -16,6 +16,7 @@
 #include "../transport/reliable.h"
 #include "../p2p/compress.h"
 #include "../core/latency.h"
+#include "../core/ring.h"
 #include "../core/trace.h"
 #include "../ukernel/ukernel.h"
 
@@ -438,6 +439,32 @@ PYBIND11_MODULE(_C, m) {
   m.def("trace_dump_json", &uccl::trace::dump_json);
   m.def("trace_clear", &uccl::trace::clear);
   m.def("trace_num_events", &uccl::trace::num_events);
+  py::class_<uccl::SpscRing<uint64_t>>(m, "SpscRingU64")
+      .def(py::init<size_t>(), py::arg("capacity_pow2"))
+      .def("push", &uccl::SpscRing<uint64_t>::push)
+      .def("pop",
+           [](uccl::SpscRing<uint64_t>& r) -> py::object {
+             uint64_t v;
+             if (r.pop(&v)) return py::cast(v);
+             return py::none();
+           })
+      .def("size", &uccl::SpscRing<uint64_t>::size);
+  py::class_<uccl::MpmcRing<uint64_t>>(m, "MpmcRingU64")
+      .def(py::init<size_t>(), py::arg("capacity_pow2"))
+      .def("push", &uccl::MpmcRing<uint64_t>::push,
+           py::call_guard<py::gil_scoped_release>())
+      .def("pop",
+           [](uccl::MpmcRing<uint64_t>& r) -> py::object {
+             uint64_t v;
+             bool ok;
+             {
+               py::gil_scoped_release rel;
+               ok = r.pop(&v);
+             }
+             if (ok) return py::cast(v);
+             return py::none();
+           })
+      .def("size_approx", &uccl::MpmcRing<uint64_t>::size_approx);
   py::class_<uccl::LatencyHist>(m, "LatencyHist")
       .def(py::init<>())
       .def("record_us", &uccl::LatencyHist::record_us)

@@ -8,6 +8,8 @@
 #include <thread>
 #include <vector>
 
+#include "ring.h"
+
 namespace uccl {
 namespace trace {
 
@@ -21,15 +23,30 @@ struct Ev {
   uint64_t tid;
 };
 
-constexpr size_t kMaxEvents = 1u << 20;
+// bounded lock-free recording (core/ring.h MPMC): engines on hot paths
+// never take a lock to emit an event; the ring is allocated lazily on
+// first enable so a disabled process pays one pointer load
+constexpr size_t kRingCap = 1u << 18;
 
 std::atomic<bool> g_on{[] {
   char const* e = std::getenv("UCCL_TRACE");
   return e && e[0] == '1';
 }()};
-std::mutex g_mu;
-std::vector<Ev> g_events;
+std::atomic<MpmcRing<Ev>*> g_ring{nullptr};
+std::mutex g_mu;  // guards dump/clear (drain side), not recording
 std::atomic<uint64_t> g_dropped{0};
+
+MpmcRing<Ev>* ring() {
+  MpmcRing<Ev>* r = g_ring.load(std::memory_order_acquire);
+  if (r) return r;
+  std::lock_guard<std::mutex> lk(g_mu);
+  r = g_ring.load(std::memory_order_acquire);
+  if (!r) {
+    r = new MpmcRing<Ev>(kRingCap);
+    g_ring.store(r, std::memory_order_release);
+  }
+  return r;
+}
 
 uint64_t now_us() {
   return std::chrono::duration_cast<std::chrono::microseconds>(
@@ -47,23 +64,20 @@ void set_enabled(bool on) { g_on.store(on, std::memory_order_relaxed); }
 
 void event(char const* cat, char const* name, char ph, int64_t arg) {
   if (!enabled()) return;
-  uint64_t const ts = now_us();
-  uint64_t const t = tid();
-  std::lock_guard<std::mutex> lk(g_mu);
-  if (g_events.size() >= kMaxEvents) {
-    g_dropped.fetch_add(1, std::memory_order_relaxed);
-    return;
-  }
-  g_events.push_back({cat, name, ph, arg, ts, t});
+  Ev e{cat, name, ph, arg, now_us(), tid()};
+  if (!ring()->push(e)) g_dropped.fetch_add(1, std::memory_order_relaxed);
 }
 
 std::string dump_json() {
   std::lock_guard<std::mutex> lk(g_mu);
   std::ostringstream os;
   os << "{\"traceEvents\":[";
-  for (size_t i = 0; i < g_events.size(); ++i) {
-    auto const& e = g_events[i];
-    if (i) os << ",";
+  MpmcRing<Ev>* r = g_ring.load(std::memory_order_acquire);
+  bool first = true;
+  Ev e;
+  while (r && r->pop(&e)) {
+    if (!first) os << ",";
+    first = false;
     os << "{\"cat\":\"" << e.cat << "\",\"name\":\"" << e.name
        << "\",\"ph\":\"" << e.ph << "\",\"ts\":" << e.ts_us
        << ",\"pid\":1,\"tid\":" << e.tid;
@@ -79,13 +93,16 @@ std::string dump_json() {
 
 void clear() {
   std::lock_guard<std::mutex> lk(g_mu);
-  g_events.clear();
+  MpmcRing<Ev>* r = g_ring.load(std::memory_order_acquire);
+  Ev e;
+  while (r && r->pop(&e)) {
+  }
   g_dropped.store(0, std::memory_order_relaxed);
 }
 
 size_t num_events() {
-  std::lock_guard<std::mutex> lk(g_mu);
-  return g_events.size();
+  MpmcRing<Ev>* r = g_ring.load(std::memory_order_acquire);
+  return r ? r->size_approx() : 0;
 }
 
 }  // namespace trace

@@ -39,6 +39,7 @@ class Span {
   bool on_ = false;
 };
 
+// drains the recorded events into Trace Event Format JSON
 std::string dump_json();
 void clear();
 size_t num_events();
