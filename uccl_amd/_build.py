@@ -72,9 +72,12 @@ def build_plugin(verbose: bool = False) -> Path:
                  CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h",
                  CSRC / "transport" / "reliable.h"]
     shim_check = plugdir / "libuccl_nccl.so"
+    shim_deps = [CSRC / "nccl_shim" / "nccl_shim.cpp",
+                 CSRC / "collective" / "kernels.hip",
+                 CSRC / "collective" / "communicator.cpp"]
     if (not _stale(target, deps) and not _stale(harness, deps)
             and not _stale(capi, capi_srcs + deps)
-            and shim_check.exists()):
+            and not _stale(shim_check, shim_deps)):
         return target
     import subprocess as sp
 

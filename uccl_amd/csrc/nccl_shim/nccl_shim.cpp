@@ -326,4 +326,38 @@ ncclResult_t ncclRecv(void* recvbuff, size_t count,
   return ncclSuccess;
 }
 
+// RCCL extension, also provided by the reference's drop-in
+// (lite-collective nccl.cu:2069): symmetric all-to-all, count elements
+// per peer.
+ncclResult_t ncclAllToAll(const void* sendbuff, void* recvbuff,
+                          size_t count, ncclDataType_t datatype,
+                          ncclComm_t comm, hipStream_t stream) {
+  Dtype dt;
+  if (to_dtype(datatype, &dt) != ncclSuccess) return ncclInvalidArgument;
+  COMM(comm)->comm->all_to_all(recvbuff, sendbuff,
+                               count * uccl::dtype_size(dt), Dtype::kU8,
+                               stream);
+  return ncclSuccess;
+}
+
+ncclResult_t ncclCommFinalize(ncclComm_t comm) {
+  // flush outstanding engine work; destruction stays with CommDestroy
+  if (comm) COMM(comm)->comm->barrier(nullptr);
+  return ncclSuccess;
+}
+
+const char* ncclGetLastError(ncclComm_t) {
+  return "";  // entry points fail fast with result codes; no deferred log
+}
+
+ncclResult_t ncclMemAlloc(void** ptr, size_t size) {
+  if (!ptr) return ncclInvalidArgument;
+  return hipMalloc(ptr, size) == hipSuccess ? ncclSuccess
+                                            : ncclUnhandledCudaError;
+}
+
+ncclResult_t ncclMemFree(void* ptr) {
+  return hipFree(ptr) == hipSuccess ? ncclSuccess : ncclUnhandledCudaError;
+}
+
 }  // extern "C"
