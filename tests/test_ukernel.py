@@ -170,3 +170,44 @@ def test_auto_planner_executes():
     outs, _ = uk.execute_host(g, ins)
     for o in outs:
         assert torch.equal(o, expect)
+
+
+def test_planner_fuzz():
+    # randomized geometry sweep across every planner family; fp32 sums on
+    # small-int data are exact, so verification is strict equality
+    import random
+
+    rng = random.Random(20260912)
+    for trial in range(12):
+        world = rng.choice([2, 3, 4, 5, 8])
+        elems = rng.randint(1, 5000)
+        chunk = rng.choice([64, 512, 4096]) * 4
+        topo = uk.Topology(world)
+        if rng.random() < 0.5 and world > 2:
+            a, b = rng.sample(range(world), 2)
+            topo.set_link_weight(a, b, rng.choice([0.1, 0.5, 2.0]))
+        ins = _inputs(world, elems, seed=trial)
+        expect = torch.stack(ins).sum(0)
+        fam = rng.choice(["rsag", "oneshot", "auto", "bcast", "send"])
+        if fam == "rsag":
+            g = uk.lower(uk.plan_allreduce_rsag(topo, elems * 4, 4, chunk))
+        elif fam == "oneshot":
+            g = uk.lower(uk.plan_allreduce_oneshot(topo, elems * 4))
+        elif fam == "auto":
+            g = uk.plan_allreduce_auto(topo, elems * 4, 4, chunk)
+        elif fam == "bcast":
+            g = uk.lower(uk.plan_broadcast(topo, world - 1, elems * 4,
+                                           chunk))
+        else:
+            g = uk.lower(uk.plan_sendrecv(topo, 0, world - 1, elems * 4,
+                                          chunk))
+        outs, stats = uk.execute_host(g, ins)
+        assert stats["tasks_run"] == g.num_tasks, (fam, world, elems)
+        if fam in ("rsag", "oneshot", "auto"):
+            for o in outs:
+                assert torch.equal(o, expect), (fam, world, elems, chunk)
+        elif fam == "bcast":
+            for o in outs:
+                assert torch.equal(o, ins[world - 1]), (world, elems)
+        else:
+            assert torch.equal(outs[world - 1], ins[0]), (world, elems)

@@ -28,6 +28,16 @@ from uccl_amd.ep import Buffer as _NativeBuffer
 from uccl_amd.ep import get_dispatch_layout
 
 
+class Config:
+    """DeepEP Config stand-in (buffer.py get_dispatch_config): the native
+    kernels pick their own launch shapes (adaptive fanout, ~2048 blocks),
+    so the knobs are accepted and recorded but not required."""
+
+    def __init__(self, num_sms: int = 24, **kwargs):
+        self.num_sms = num_sms
+        self.extra = kwargs
+
+
 class _Handle:
     def __init__(self, topk_idx, num_tokens):
         self.topk_idx = topk_idx
@@ -107,6 +117,44 @@ class Buffer:
         combined = self._native.combine(x, topk_idx, topk_weights)
         return combined, _event(), _noop_hook
 
-    # convenience aliases matching the high-throughput entry points
+    # -- maintenance / config surface ---------------------------------------
+    def clean_low_latency_buffer(self, num_max_dispatch_tokens_per_rank=None,
+                                 hidden=None, num_experts=None):
+        """No-op by design: the native engine seq-tags every per-(expert,
+        src) count word (ep_kernels.hip k_ep_dispatch_publish), so stale
+        state from a previous iteration can never be confused with the
+        current one and nothing needs zeroing between calls (DeepEP zeroes
+        count/flag regions here, internode_ll.cu:23)."""
+        return None
+
+    @staticmethod
+    def get_low_latency_rdma_size_hint(num_max_dispatch_tokens_per_rank,
+                                       hidden, num_ranks, num_experts):
+        # heap sizing is handled natively; returned for API compatibility
+        per_token = hidden * 2 + 16
+        return int(num_max_dispatch_tokens_per_rank * num_ranks * per_token)
+
+    @staticmethod
+    def get_dispatch_config(num_ranks: int):
+        return Config(num_sms=24)
+
+    @staticmethod
+    def get_combine_config(num_ranks: int):
+        return Config(num_sms=24)
+
+    @property
+    def group_size(self):
+        import torch.distributed as dist
+
+        if self._group is not None and dist.is_available() \
+                and dist.is_initialized():
+            return dist.get_world_size(group=self._group)
+        return 1
+
+    # convenience aliases matching the high-throughput and internode entry
+    # points (one xGMI engine serves all three DeepEP modes; internode
+    # peers ride the proxy path automatically)
     dispatch = low_latency_dispatch
     combine = low_latency_combine
+    internode_dispatch = low_latency_dispatch
+    internode_combine = low_latency_combine
