@@ -211,3 +211,25 @@ def test_planner_fuzz():
                 assert torch.equal(o, ins[world - 1]), (world, elems)
         else:
             assert torch.equal(outs[world - 1], ins[0]), (world, elems)
+
+
+def test_allgather_plan():
+    world, elems = 4, 1000  # per-rank contribution
+    topo = uk.Topology(world)
+    ins = _inputs(world, elems, seed=21)
+    g = uk.lower(uk.plan_allgather(topo, elems * 4, chunk_bytes=1024))
+    outs, _ = uk.execute_host(g, ins, out_bytes=world * elems * 4)
+    expect = torch.cat(ins)
+    for o in outs:
+        assert torch.equal(o, expect)
+
+
+def test_reducescatter_plan():
+    world, shard = 4, 600  # elems per shard; input = world*shard per rank
+    topo = uk.Topology(world)
+    ins = _inputs(world, world * shard, seed=22)
+    g = uk.lower(uk.plan_reducescatter(topo, shard * 4, 4, chunk_bytes=512))
+    outs, _ = uk.execute_host(g, ins, out_bytes=shard * 4)
+    total = torch.stack(ins).sum(0)
+    for r, o in enumerate(outs):
+        assert torch.equal(o, total[r * shard:(r + 1) * shard]), r

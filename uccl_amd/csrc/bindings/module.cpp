@@ -549,6 +549,11 @@ PYBIND11_MODULE(_C, m) {
         py::arg("topo"), py::arg("nbytes"), py::arg("elem_bytes") = 4);
   m.def("uk_plan_broadcast", &uk::plan_broadcast, py::arg("topo"),
         py::arg("root"), py::arg("nbytes"), py::arg("chunk_bytes"));
+  m.def("uk_plan_allgather", &uk::plan_allgather, py::arg("topo"),
+        py::arg("nbytes"), py::arg("chunk_bytes") = 1 << 20);
+  m.def("uk_plan_reducescatter", &uk::plan_reducescatter, py::arg("topo"),
+        py::arg("shard_bytes"), py::arg("elem_bytes") = 4,
+        py::arg("chunk_bytes") = 1 << 20);
   m.def("uk_lower", &uk::lower);
   m.def("uk_estimate_us", &uk::estimate_us, py::arg("graph"),
         py::arg("topo"), py::arg("link_gbps") = 150.0,

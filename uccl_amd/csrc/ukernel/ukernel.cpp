@@ -264,6 +264,79 @@ ChunkGraph plan_broadcast(Topology const& topo, int root, uint64_t nbytes,
   return g;
 }
 
+ChunkGraph plan_allgather(Topology const& topo, uint64_t nbytes,
+                          uint64_t chunk_bytes) {
+  int const world = topo.world;
+  ChunkGraph g;
+  g.world = world;
+  g.scratch_bytes = 0;
+  for (int src = 0; src < world; ++src) {
+    uint64_t const base = uint64_t(src) * nbytes;
+    for (uint64_t c = 0; c < nbytes; c += chunk_bytes) {
+      uint64_t const len = std::min(chunk_bytes, nbytes - c);
+      Task self;
+      self.op = Op::kCopy;
+      self.rank = src;
+      self.src = {src, Space::kInput, c};
+      self.dst = {src, Space::kOutput, base + c};
+      self.bytes = len;
+      g.add(std::move(self));
+      for (int dst = 0; dst < world; ++dst) {
+        if (dst == src) continue;
+        Task put;
+        put.op = Op::kPut;
+        put.rank = src;
+        put.src = {src, Space::kInput, c};
+        put.dst = {dst, Space::kOutput, base + c};
+        put.bytes = len;
+        g.add(std::move(put));
+      }
+    }
+  }
+  return g;
+}
+
+ChunkGraph plan_reducescatter(Topology const& topo, uint64_t shard_bytes,
+                              uint64_t elem_bytes, uint64_t chunk_bytes) {
+  (void)elem_bytes;
+  int const world = topo.world;
+  ChunkGraph g;
+  g.world = world;
+  g.scratch_bytes = shard_bytes * world;
+  for (int owner = 0; owner < world; ++owner) {
+    uint64_t const base = uint64_t(owner) * shard_bytes;
+    for (uint64_t c = 0; c < shard_bytes; c += chunk_bytes) {
+      uint64_t const len = std::min(chunk_bytes, shard_bytes - c);
+      Task seed;
+      seed.op = Op::kCopy;
+      seed.rank = owner;
+      seed.src = {owner, Space::kInput, base + c};
+      seed.dst = {owner, Space::kOutput, c};
+      seed.bytes = len;
+      int last = g.add(std::move(seed));
+      for (int r = 0; r < world; ++r) {
+        if (r == owner) continue;
+        Task put;
+        put.op = Op::kPut;
+        put.rank = r;
+        put.src = {r, Space::kInput, base + c};
+        put.dst = {owner, Space::kScratch, uint64_t(r) * shard_bytes + c};
+        put.bytes = len;
+        int const p = g.add(std::move(put));
+        Task red;
+        red.op = Op::kReduce;
+        red.rank = owner;
+        red.src = {owner, Space::kScratch, uint64_t(r) * shard_bytes + c};
+        red.dst = {owner, Space::kOutput, c};
+        red.bytes = len;
+        red.deps = {p, last};
+        last = g.add(std::move(red));
+      }
+    }
+  }
+  return g;
+}
+
 // ---------------------------------------------------------------------------
 // Cost model
 // ---------------------------------------------------------------------------

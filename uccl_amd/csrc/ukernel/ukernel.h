@@ -123,6 +123,16 @@ ChunkGraph plan_allreduce_oneshot(Topology const& topo, uint64_t nbytes,
 ChunkGraph plan_broadcast(Topology const& topo, int root, uint64_t nbytes,
                           uint64_t chunk_bytes);
 
+// All-gather: rank r's nbytes input becomes slot r of every rank's
+// world*nbytes output (fullmesh push, chunked).
+ChunkGraph plan_allgather(Topology const& topo, uint64_t nbytes,
+                          uint64_t chunk_bytes);
+
+// Reduce-scatter: rank r's output is the sum over ranks of shard r of
+// their world*shard_bytes inputs (fullmesh push + owner reduce).
+ChunkGraph plan_reducescatter(Topology const& topo, uint64_t shard_bytes,
+                              uint64_t elem_bytes, uint64_t chunk_bytes);
+
 // ---------------------------------------------------------------------------
 // Cost model: deterministic list-scheduling estimate of a plan's
 // completion time. Resources: every directed link serializes its kPuts at

@@ -53,6 +53,16 @@ def plan_broadcast(topo, root: int, nbytes: int,
     return _C.uk_plan_broadcast(topo, root, nbytes, chunk_bytes)
 
 
+def plan_allgather(topo, nbytes: int, chunk_bytes: int = 1 << 20) -> Graph:
+    return _C.uk_plan_allgather(topo, nbytes, chunk_bytes)
+
+
+def plan_reducescatter(topo, shard_bytes: int, elem_bytes: int = 4,
+                       chunk_bytes: int = 1 << 20) -> Graph:
+    return _C.uk_plan_reducescatter(topo, shard_bytes, elem_bytes,
+                                    chunk_bytes)
+
+
 def lower(graph: Graph) -> Graph:
     return _C.uk_lower(graph)
 
