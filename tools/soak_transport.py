@@ -13,6 +13,10 @@ def one_round(rnd: random.Random, idx: int):
     os.environ["UCCL_TP_LOSS_PCT"] = str(rnd.choice([0, 0, 2, 5, 10, 20]))
     os.environ["UCCL_TP_ACK_LOSS_PCT"] = str(rnd.choice([0, 0, 10]))
     os.environ["UCCL_TP_CC"] = rnd.choice(["timely", "swift", "eqds"])
+    # paced pull quanta occasionally (high rate so soak stays fast)
+    os.environ["UCCL_TP_EQDS_MBPS"] = str(
+        rnd.choice([0, 0, 0, 400]) if os.environ["UCCL_TP_CC"] == "eqds"
+        else 0)
     os.environ["UCCL_TP_CWND_MAX"] = str(rnd.choice([32, 256, 1024]))
     from uccl_amd import _load_native
 
