@@ -84,7 +84,7 @@ def build_plugin(verbose: bool = False) -> Path:
     reliable = CSRC / "transport" / "reliable.cpp"
     tracecc = CSRC / "core" / "trace.cpp"
     for cmd in (
-        ["g++", "-O2", "-std=c++17", "-fPIC", "-shared", str(src),
+        ["g++", "-O2", "-g", "-std=c++17", "-fPIC", "-shared", str(src),
          str(reliable), str(tracecc), "-o", str(target), "-pthread"],
         ["g++", "-O2", "-std=c++17", str(harness_src), "-o", str(harness),
          "-ldl", "-pthread"],

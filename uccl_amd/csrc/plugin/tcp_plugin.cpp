@@ -15,9 +15,12 @@
 // harness drives the vtable over both data planes).
 
 #include <fcntl.h>
+#include <unistd.h>
+#include <cstdio>
 #include <poll.h>
 
 #include <atomic>
+#include <chrono>
 #include <map>
 #include <condition_variable>
 #include <cstring>
@@ -84,6 +87,16 @@ bool use_multipath() {
   return v;
 }
 
+static bool net_debug() {
+  static bool v = uccl::env_bool("UCCL_NET_DEBUG", false);
+  return v;
+}
+#define NET_DBG(fmt, ...) \
+  do { \
+    if (net_debug()) \
+      fprintf(stderr, "[uccl-net %d] " fmt "\n", getpid(), ##__VA_ARGS__); \
+  } while (0)
+
 // --- multipath fabric singleton -------------------------------------------
 // One TransportEndpoint per process; RCCL's many listenComms are
 // distinguished by a nonce carried in the connect tag, so a global
@@ -106,6 +119,8 @@ struct MpFabric {
         while (true) {
           uint64_t tag = 0;
           uint64_t flow = tp.accept(&tag);
+          NET_DBG("fabric accept flow=%llx nonce=%llu",
+                  (unsigned long long)flow, (unsigned long long)tag);
           std::lock_guard<std::mutex> g(mu);
           queues[tag].push_back(flow);
           cv.notify_all();
@@ -176,6 +191,7 @@ Comm::~Comm() {
 // path sends as two discrete reliable messages)
 void chan_send(Comm* c, void const* buf, size_t n) {
   if (c->flow) {
+    NET_DBG("send flow=%llx n=%zu", (unsigned long long)c->flow, n);
     MpFabric::get().tp.send_msg(c->flow, buf, n);
   } else {
     send_all(c->fd, buf, n);
@@ -184,8 +200,10 @@ void chan_send(Comm* c, void const* buf, size_t n) {
 
 bool chan_recv(Comm* c, void* buf, size_t n) {
   if (c->flow) {
+    NET_DBG("recv post flow=%llx n=%zu", (unsigned long long)c->flow, n);
     try {
       MpFabric::get().tp.recv_msg(c->flow, buf, n);
+      NET_DBG("recv done flow=%llx n=%zu", (unsigned long long)c->flow, n);
       return true;
     } catch (std::exception const&) {
       return false;
@@ -199,7 +217,13 @@ void tx_loop(Comm* c) {
     SendOp op;
     {
       std::unique_lock<std::mutex> lk(c->mu);
-      c->cv.wait(lk, [&] { return !c->sendq.empty() || !c->alive; });
+      // bounded wait: re-check the predicate on a 50ms tick. Under heavy
+      // CPU oversubscription we observed rare stalls with an op queued
+      // despite the notify; the periodic re-check bounds any missed wake
+      // at 50ms (the reference's engines busy-poll with adaptive sleep
+      // for the same robustness, p2p/util/adaptive_sleeper.h).
+      while (c->sendq.empty() && c->alive)
+        c->cv.wait_for(lk, std::chrono::milliseconds(50));
       if (!c->alive) return;
       op = c->sendq.front();
       c->sendq.pop_front();
@@ -324,6 +348,8 @@ ncclResult_t p_connect(int dev, void* opaque, void** sendComm) {
     try {
       uint64_t flow = MpFabric::get().tp.connect(
           std::string(h.md, h.md_len), h.nonce);
+      NET_DBG("connect flow=%llx nonce=%llu", (unsigned long long)flow,
+              (unsigned long long)h.nonce);
       auto* c = new Comm();
       c->flow = flow;
       c->sender = true;
