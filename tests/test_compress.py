@@ -140,3 +140,34 @@ def test_object_transfer():
     assert torch.equal(o["weights"]["ids"], obj["weights"]["ids"])
     assert o["shapes"][0] == (1, 2)
     assert torch.equal(o["shapes"][1][1], torch.zeros(5))
+
+
+def test_endpoint_compressed_transfer_multipath():
+    # codec over the multipath reliable plane (env must be set before the
+    # endpoints exist; restore afterwards to keep test isolation)
+    import os
+
+    old = os.environ.get("UCCL_P2P_TRANSPORT")
+    os.environ["UCCL_P2P_TRANSPORT"] = "multipath"
+    try:
+        a = p2p.Endpoint(gpu=0, num_workers=1)
+        b = p2p.Endpoint(gpu=0, num_workers=1)
+        ids = {}
+        th = threading.Thread(target=lambda: ids.setdefault("b", b.accept()))
+        th.start()
+        cid_a = a.connect(b.metadata())
+        th.join()
+        src = torch.randn(1 << 20).bfloat16()
+        got = {}
+        rx = threading.Thread(
+            target=lambda: got.setdefault("t",
+                                          p2p.recv_compressed(b, ids["b"])))
+        rx.start()
+        p2p.send_compressed(a, cid_a, src)
+        rx.join(timeout=120)
+        assert torch.equal(got["t"], src)
+    finally:
+        if old is None:
+            os.environ.pop("UCCL_P2P_TRANSPORT", None)
+        else:
+            os.environ["UCCL_P2P_TRANSPORT"] = old
