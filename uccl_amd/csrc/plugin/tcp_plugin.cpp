@@ -269,8 +269,32 @@ void rx_loop(Comm* c) {
       f.tag = h.tag;
       f.data.resize(h.bytes);
       if (h.bytes) chan_recv(c, f.data.data(), h.bytes);
+      // re-check posted under the lock: between the first posted scan
+      // (above, which missed) and this push, p_irecv may have run —
+      // it saw an empty `unmatched` and parked its recv in `posted`.
+      // Without this second scan the frame and the posted recv would
+      // each sit in their queue forever (observed as a rare multi-pair
+      // hang under load).
       std::lock_guard<std::mutex> g(c->mu);
-      c->unmatched.push_back(std::move(f));
+      bool late_match = false;
+      for (auto it = c->posted.begin(); it != c->posted.end(); ++it) {
+        if (it->tag == f.tag) {
+          PostedRecv pr2 = *it;
+          c->posted.erase(it);
+          if (pr2.cap >= static_cast<int>(f.bytes)) {
+            if (f.bytes) memcpy(pr2.data, f.data.data(), f.bytes);
+            pr2.req->size = static_cast<int>(f.bytes);
+            pr2.req->done.store(1, std::memory_order_release);
+            late_match = true;
+          } else {
+            PLOG(NCCL_LOG_WARN,
+                 "uccl-net: recv overflow tag=%d %lu > %d (req dropped)",
+                 f.tag, (unsigned long)f.bytes, pr2.cap);
+          }
+          break;
+        }
+      }
+      if (!late_match) c->unmatched.push_back(std::move(f));
     }
   }
 }
