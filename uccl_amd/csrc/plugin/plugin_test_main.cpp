@@ -153,6 +153,51 @@ int main(int argc, char** argv) {
     printf("multi-pair OK\n");
   }
 
+  // --- rapid-fire unmatched/posted interleave stress ----------------------
+  // regression for the rx cross-match race: when a frame arrives in the
+  // window between irecv's unmatched scan and its posted push, the rx
+  // loop must deliver the late match. Alternate posting order and sizes
+  // across many quick iterations to drive both paths.
+  {
+    char hdl[NCCL_NET_HANDLE_MAXSIZE] = {};
+    void *plc = nullptr, *psc = nullptr, *prc = nullptr;
+    assert(net->listen(0, hdl, &plc) == ncclSuccess);
+    std::thread acc([&] {
+      while (!prc) assert(net->accept(plc, &prc) == ncclSuccess);
+    });
+    while (!psc) assert(net->connect(0, hdl, &psc) == ncclSuccess);
+    acc.join();
+    for (int it = 0; it < 300; ++it) {
+      int const n = 64 + (it * 97) % 4000;
+      std::vector<char> sb(n, static_cast<char>(it));
+      std::vector<char> rb(n, 0);
+      void *sr = nullptr, *rr = nullptr;
+      void* d[1] = {rb.data()};
+      int z[1] = {n};
+      int t[1] = {it};
+      void* mh[1] = {nullptr};
+      if (it & 1) {
+        // send first: the frame usually lands before the recv is posted
+        assert(net->isend(psc, sb.data(), n, it, nullptr, &sr) ==
+               ncclSuccess);
+        wait_req(net, sr, nullptr);
+        assert(net->irecv(prc, 1, d, z, t, mh, &rr) == ncclSuccess);
+      } else {
+        assert(net->irecv(prc, 1, d, z, t, mh, &rr) == ncclSuccess);
+        assert(net->isend(psc, sb.data(), n, it, nullptr, &sr) ==
+               ncclSuccess);
+        wait_req(net, sr, nullptr);
+      }
+      int got = 0;
+      wait_req(net, rr, &got);
+      assert(got == n && memcmp(sb.data(), rb.data(), n) == 0);
+    }
+    assert(net->closeSend(psc) == ncclSuccess);
+    assert(net->closeRecv(prc) == ncclSuccess);
+    assert(net->closeListen(plc) == ncclSuccess);
+    printf("interleave stress OK\n");
+  }
+
   printf("PLUGIN HARNESS OK\n");
   return 0;
 }
