@@ -68,3 +68,41 @@ def test_p2p_gpu_tcp_staging():
     if not torch.cuda.is_available():
         pytest.skip("requires GPU")
     _run_pair({"UCCL_P2P_TEST_GPU": "1", "UCCL_P2P_ENABLE_IPC": "0"})
+
+
+def test_p2p_async_fifo_cpu():
+    """Concurrent async sends/recvs on one connection must match in
+    submission order even with a multi-worker pool (per-direction ticket
+    sequencing in the engine)."""
+    import threading
+    import time
+
+    import torch
+
+    from uccl_amd import p2p
+
+    a = p2p.Endpoint(gpu=0, num_workers=3)
+    b = p2p.Endpoint(gpu=0, num_workers=3)
+    ids = {}
+    th = threading.Thread(target=lambda: ids.setdefault("b", b.accept()))
+    th.start()
+    cid_a = a.connect(b.metadata())
+    th.join(timeout=30)
+    for trial in range(5):
+        k = 6
+        srcs = [torch.full((4096,), i + trial * 10, dtype=torch.int32)
+                for i in range(k)]
+        dsts = [torch.zeros(4096, dtype=torch.int32) for _ in range(k)]
+        rids = [b.recv_async(ids["b"], d) for d in dsts]
+        sids = [a.send_async(cid_a, s) for s in srcs]
+        deadline = time.time() + 60
+        for i in sids:
+            while not a.poll_async(i):
+                assert time.time() < deadline
+                time.sleep(0.001)
+        for i in rids:
+            while not b.poll_async(i):
+                assert time.time() < deadline
+                time.sleep(0.001)
+        for s, d in zip(srcs, dsts):
+            assert torch.equal(s, d)

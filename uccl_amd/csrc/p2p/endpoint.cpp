@@ -93,6 +93,14 @@ struct Endpoint::Conn {
   std::mutex rx_mu;
   std::condition_variable rx_cv;
 
+  // per-direction FIFO tickets for ASYNC two-sided ops: the worker pool
+  // may start tasks out of submission order, but sends must hit the wire
+  // (and recvs must pop rxq) in the order the caller issued them
+  std::mutex ord_mu;
+  std::condition_variable ord_cv;
+  uint64_t tx_ticket = 0, tx_serving = 0;
+  uint64_t rx_ticket = 0, rx_serving = 0;
+
   // token -> completion latch (acks, read responses)
   std::mutex tok_mu;
   std::condition_variable tok_cv;
@@ -254,12 +262,14 @@ Endpoint::~Endpoint() {
       {
         std::lock_guard<std::mutex> g1(c->rx_mu);
         std::lock_guard<std::mutex> g2(c->tok_mu);
+        std::lock_guard<std::mutex> g3(c->ord_mu);
         c->alive = false;
       }
       ::shutdown(c->fd, SHUT_RDWR);
       if (c->flow && tp_) tp_->close_flow(c->flow);
       c->rx_cv.notify_all();
       c->tok_cv.notify_all();
+      c->ord_cv.notify_all();
     }
   }
   if (listener_.joinable()) listener_.join();
@@ -485,7 +495,11 @@ void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
       }
       default:
         UCCL_LOG_ERROR << "unknown p2p op " << h.op;
-        c->alive = false;
+        {
+          std::lock_guard<std::mutex> g(c->ord_mu);
+          c->alive = false;
+        }
+        c->ord_cv.notify_all();
         return;
     }
   }
@@ -737,11 +751,43 @@ void Endpoint::worker_loop() {
 uint64_t Endpoint::send_async(uint64_t cid, void const* p, size_t n,
                               int dev) {
   auto c = conn(cid);
-  return submit([this, c, p, n, dev] { do_send(*c, p, n, dev); });
+  uint64_t tk;
+  {
+    std::lock_guard<std::mutex> g(c->ord_mu);
+    tk = c->tx_ticket++;
+  }
+  return submit([this, c, p, n, dev, tk] {
+    {
+      std::unique_lock<std::mutex> lk(c->ord_mu);
+      c->ord_cv.wait(lk, [&] { return c->tx_serving == tk || !c->alive; });
+    }
+    do_send(*c, p, n, dev);
+    {
+      std::lock_guard<std::mutex> g(c->ord_mu);
+      ++c->tx_serving;
+    }
+    c->ord_cv.notify_all();
+  });
 }
 uint64_t Endpoint::recv_async(uint64_t cid, void* p, size_t n, int dev) {
   auto c = conn(cid);
-  return submit([this, c, p, n, dev] { do_recv(*c, p, n, dev); });
+  uint64_t tk;
+  {
+    std::lock_guard<std::mutex> g(c->ord_mu);
+    tk = c->rx_ticket++;
+  }
+  return submit([this, c, p, n, dev, tk] {
+    {
+      std::unique_lock<std::mutex> lk(c->ord_mu);
+      c->ord_cv.wait(lk, [&] { return c->rx_serving == tk || !c->alive; });
+    }
+    do_recv(*c, p, n, dev);
+    {
+      std::lock_guard<std::mutex> g(c->ord_mu);
+      ++c->rx_serving;
+    }
+    c->ord_cv.notify_all();
+  });
 }
 uint64_t Endpoint::write_async(uint64_t cid, void const* p, size_t n, int dev,
                                const std::string& ad) {
@@ -779,12 +825,14 @@ void Endpoint::close_conn(uint64_t conn_id) {
   {
     std::lock_guard<std::mutex> g1(c->rx_mu);
     std::lock_guard<std::mutex> g2(c->tok_mu);
+    std::lock_guard<std::mutex> g3(c->ord_mu);
     c->alive = false;
   }
   ::shutdown(c->fd, SHUT_RDWR);
   if (c->flow && tp_) tp_->close_flow(c->flow);
   c->rx_cv.notify_all();
   c->tok_cv.notify_all();
+  c->ord_cv.notify_all();
   if (c->rx.joinable()) c->rx.join();
   ::close(c->fd);
   std::lock_guard<std::mutex> g2(c->ipc_mu);
