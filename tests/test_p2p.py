@@ -106,3 +106,30 @@ def test_p2p_async_fifo_cpu():
                 time.sleep(0.001)
         for s, d in zip(srcs, dsts):
             assert torch.equal(s, d)
+
+
+def test_p2p_stats_cpu():
+    import threading
+
+    import torch
+
+    from uccl_amd import p2p
+
+    a = p2p.Endpoint(gpu=0, num_workers=1)
+    b = p2p.Endpoint(gpu=0, num_workers=1)
+    ids = {}
+    th = threading.Thread(target=lambda: ids.setdefault("b", b.accept()))
+    th.start()
+    cid = a.connect(b.metadata())
+    th.join(timeout=30)
+    src = torch.arange(10000, dtype=torch.float32)
+    dst = torch.zeros_like(src)
+    t = threading.Thread(target=lambda: b.recv(ids["b"], dst))
+    t.start()
+    a.send(cid, src)
+    t.join(timeout=30)
+    st = a.stats()
+    assert st["send"]["calls"] == 1
+    assert st["send"]["bytes"] == 40000
+    assert st["send"]["p50_us"] > 0
+    assert b.stats()["recv"]["calls"] == 1

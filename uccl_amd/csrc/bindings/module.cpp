@@ -229,6 +229,19 @@ PYBIND11_MODULE(_C, m) {
              return py::bytes(e.advertise(mr, off, bytes));
            },
            py::arg("mr_id"), py::arg("offset") = 0, py::arg("bytes") = 0)
+      .def("stats",
+           [](Endpoint& e) {
+             py::dict d;
+             for (auto& [name, st] : e.stats()) {
+               py::dict o;
+               o["calls"] = st.calls;
+               o["bytes"] = st.bytes;
+               o["p50_us"] = st.p50_us;
+               o["p99_us"] = st.p99_us;
+               d[py::str(name)] = o;
+             }
+             return d;
+           })
       .def("send",
            [dev_of](Endpoint& e, uint64_t cid, at::Tensor t) {
              TORCH_CHECK(t.is_contiguous());

@@ -531,6 +531,7 @@ static bool ipc_enabled() {
 
 void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
   trace::Span span("p2p", "send");
+  OpTimer ot__(st_send_, bytes);
   if (is_gpu(device) && c.same_host && ipc_enabled()) {
     // one-copy IPC path: ship {handle, offset}; receiver DtoD-copies
     IpcBlob blob{};
@@ -601,6 +602,7 @@ void Endpoint::copy_to_user(RxItem& item, void* dst, size_t bytes,
 
 void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
   trace::Span span("p2p", "recv");
+  OpTimer ot__(st_recv_, bytes);
   std::shared_ptr<RxItem> item;
   {
     std::unique_lock<std::mutex> lk(c.rx_mu);
@@ -627,6 +629,7 @@ void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
 void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
                         Advert ad) {
   trace::Span span("p2p", "write");
+  OpTimer ot__(st_write_, bytes);
   UCCL_CHECK(bytes <= ad.bytes) << "write larger than advertised window";
   uint64_t token = c.next_token++;
   if (is_gpu(device) && c.same_host && ipc_enabled()) {
@@ -662,6 +665,7 @@ void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
 void Endpoint::do_read(Conn& c, void* ptr, size_t bytes, int device,
                        Advert ad) {
   trace::Span span("p2p", "read");
+  OpTimer ot__(st_read_, bytes);
   UCCL_CHECK(bytes <= ad.bytes) << "read larger than advertised window";
   uint64_t token = c.next_token++;
   c.send_msg(MsgHdr{kReadReq, ad.mr_id, ad.offset, bytes, token});
@@ -811,6 +815,23 @@ bool Endpoint::poll_async(uint64_t xfer_id) {
     return true;
   }
   return false;
+}
+
+std::map<std::string, Endpoint::OpStat> Endpoint::stats() {
+  std::map<std::string, OpStat> out;
+  auto fill = [&](char const* name, OpRec& r) {
+    OpStat st;
+    st.calls = r.calls.load(std::memory_order_relaxed);
+    st.bytes = r.bytes.load(std::memory_order_relaxed);
+    st.p50_us = r.lat.percentile_us(50);
+    st.p99_us = r.lat.percentile_us(99);
+    out[name] = st;
+  };
+  fill("send", st_send_);
+  fill("recv", st_recv_);
+  fill("write", st_write_);
+  fill("read", st_read_);
+  return out;
 }
 
 void Endpoint::close_conn(uint64_t conn_id) {

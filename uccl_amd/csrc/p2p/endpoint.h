@@ -16,7 +16,10 @@
 
 #include <hip/hip_runtime.h>
 
+#include "../core/latency.h"
+
 #include <atomic>
+#include <chrono>
 #include <map>
 #include <condition_variable>
 #include <cstdint>
@@ -85,6 +88,15 @@ class Endpoint {
 
   bool poll_async(uint64_t xfer_id);  // true once complete (then forgets it)
 
+  // per-op-family counters + latency percentiles (reference parity:
+  // engine stats threads / proxy timing getters)
+  struct OpStat {
+    uint64_t calls = 0;
+    uint64_t bytes = 0;
+    double p50_us = 0, p99_us = 0;
+  };
+  std::map<std::string, OpStat> stats();
+
   int num_conns();
   // tear down one connection (reference parity: p2p remove_remote_endpoint,
   // p2p/engine.cc:2208); pending ops on it fail
@@ -131,6 +143,31 @@ class Endpoint {
   std::mutex xfer_mu_;
   std::unordered_map<uint64_t, std::shared_ptr<std::atomic<int>>> xfers_;
   std::atomic<uint64_t> next_xfer_{1};
+
+  // per-op observability
+  struct OpRec {
+    std::atomic<uint64_t> calls{0};
+    std::atomic<uint64_t> bytes{0};
+    LatencyHist lat;
+    void add(uint64_t n, double us) {
+      calls.fetch_add(1, std::memory_order_relaxed);
+      bytes.fetch_add(n, std::memory_order_relaxed);
+      lat.record_us(us);
+    }
+  };
+  OpRec st_send_, st_recv_, st_write_, st_read_;
+  struct OpTimer {
+    OpRec& r;
+    size_t n;
+    std::chrono::steady_clock::time_point t0;
+    OpTimer(OpRec& rr, size_t nn)
+        : r(rr), n(nn), t0(std::chrono::steady_clock::now()) {}
+    ~OpTimer() {
+      r.add(n, std::chrono::duration<double, std::micro>(
+                   std::chrono::steady_clock::now() - t0)
+                   .count());
+    }
+  };
 
   // pinned staging for GPU<->TCP
   void* staging_ = nullptr;
