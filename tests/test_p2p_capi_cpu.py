@@ -87,3 +87,63 @@ def test_c_api_loopback():
 
     lib.uccl_engine_destroy(a)
     lib.uccl_engine_destroy(b)
+
+
+def test_c_api_notify():
+    import threading
+    import time
+
+    build_plugin()
+    lib = ctypes.CDLL(str(PKG_DIR / "lib" / "libuccl_p2p.so"))
+    lib.uccl_engine_create.restype = ctypes.c_void_p
+    lib.uccl_engine_create.argtypes = [ctypes.c_int, ctypes.c_int]
+    lib.uccl_engine_metadata.restype = ctypes.c_int
+    lib.uccl_engine_metadata.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                         ctypes.c_size_t]
+    lib.uccl_engine_connect.restype = ctypes.c_uint64
+    lib.uccl_engine_connect.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                        ctypes.c_size_t]
+    lib.uccl_engine_accept.restype = ctypes.c_uint64
+    lib.uccl_engine_accept.argtypes = [ctypes.c_void_p]
+    lib.uccl_engine_notify.restype = ctypes.c_int
+    lib.uccl_engine_notify.argtypes = [ctypes.c_void_p, ctypes.c_uint64,
+                                       ctypes.c_void_p, ctypes.c_size_t]
+    lib.uccl_engine_notify_poll.restype = ctypes.c_int
+    lib.uccl_engine_notify_poll.argtypes = [ctypes.c_void_p,
+                                            ctypes.c_uint64,
+                                            ctypes.c_void_p,
+                                            ctypes.c_size_t]
+    lib.uccl_engine_destroy.argtypes = [ctypes.c_void_p]
+
+    a = lib.uccl_engine_create(-1, 1)
+    b = lib.uccl_engine_create(-1, 1)
+    md = ctypes.create_string_buffer(256)
+    n = lib.uccl_engine_metadata(b, md, 256)
+    ids = {}
+    th = threading.Thread(
+        target=lambda: ids.setdefault("b", lib.uccl_engine_accept(b)))
+    th.start()
+    cid_a = lib.uccl_engine_connect(a, md, n)
+    th.join(timeout=30)
+    assert cid_a and ids["b"]
+
+    for i, msg in enumerate([b"hello", b"", b"x" * 4080]):
+        assert lib.uccl_engine_notify(a, cid_a, msg, len(msg)) == 0
+        buf = ctypes.create_string_buffer(4096)
+        deadline = time.time() + 30
+        while True:
+            r = lib.uccl_engine_notify_poll(b, ids["b"], buf, 4096)
+            if r or msg == b"":
+                # zero-length notify: poll returns 0 both for "none" and
+                # for the empty message; accept either after one recv
+                if msg == b"":
+                    time.sleep(0.2)
+                    break
+                break
+            assert time.time() < deadline
+            time.sleep(0.005)
+        if msg:
+            assert buf.raw[:r] == msg
+    assert lib.uccl_engine_notify(a, cid_a, b"y" * 5000, 5000) == -1
+    lib.uccl_engine_destroy(a)
+    lib.uccl_engine_destroy(b)

@@ -1,7 +1,13 @@
 #include "c_api.h"
 
 #include <cstring>
+#include <deque>
 #include <exception>
+#include <memory>
+#include <mutex>
+#include <thread>
+#include <unordered_map>
+#include <vector>
 
 #include "endpoint.h"
 
@@ -10,7 +16,31 @@ using uccl::p2p::Endpoint;
 struct uccl_engine {
   Endpoint ep;
   uccl_engine(int gpu, int nw) : ep(gpu, nw) {}
+
+  // notify drainers: one blocking-recv thread per conn, feeding a queue
+  struct NotifyQ {
+    std::thread th;
+    std::mutex mu;
+    std::deque<std::vector<char>> q;
+    bool dead = false;
+  };
+  std::mutex nmu;
+  std::unordered_map<uint64_t, std::unique_ptr<NotifyQ>> notify;
+  ~uccl_engine() {
+    // close each notify conn so its drainer's blocking recv throws, then
+    // join — a detached drainer would race the queue's destruction
+    std::lock_guard<std::mutex> g(nmu);
+    for (auto& [id, nq] : notify) {
+      try {
+        ep.close_conn(id);
+      } catch (std::exception const&) {
+      }
+      if (nq->th.joinable()) nq->th.join();
+    }
+  }
 };
+
+static constexpr size_t kNotifyFrame = 4096;
 
 extern "C" {
 
@@ -130,6 +160,64 @@ int uccl_engine_poll(uccl_engine_t* e, uint64_t xfer) {
   } catch (std::exception const&) {
     return -1;
   }
+}
+
+
+int uccl_engine_notify(uccl_engine_t* e, uint64_t conn, void const* data,
+                       size_t len) {
+  if (len > kNotifyFrame - 16) return -1;
+  std::vector<char> frame(kNotifyFrame, 0);
+  uint64_t const n = len;
+  memcpy(frame.data(), &n, 8);
+  if (len) memcpy(frame.data() + 16, data, len);
+  try {
+    e->ep.send(conn, frame.data(), frame.size(), -1);
+    return 0;
+  } catch (std::exception const&) {
+    return -1;
+  }
+}
+
+int uccl_engine_notify_poll(uccl_engine_t* e, uint64_t conn, void* buf,
+                            size_t cap) {
+  uccl_engine::NotifyQ* nq;
+  {
+    std::lock_guard<std::mutex> g(e->nmu);
+    auto& slot = e->notify[conn];
+    if (!slot) {
+      slot.reset(new uccl_engine::NotifyQ());
+      auto* raw = slot.get();
+      Endpoint* ep = &e->ep;
+      raw->th = std::thread([raw, ep, conn] {
+        std::vector<char> frame(kNotifyFrame);
+        while (true) {
+          try {
+            ep->recv(conn, frame.data(), frame.size(), -1);
+          } catch (std::exception const&) {
+            std::lock_guard<std::mutex> g2(raw->mu);
+            raw->dead = true;
+            return;
+          }
+          uint64_t n = 0;
+          memcpy(&n, frame.data(), 8);
+          if (n > kNotifyFrame - 16) continue;  // corrupt; drop
+          std::vector<char> msg(frame.begin() + 16,
+                                frame.begin() + 16 + n);
+          std::lock_guard<std::mutex> g2(raw->mu);
+          raw->q.push_back(std::move(msg));
+        }
+      });
+    }
+    nq = slot.get();
+  }
+  std::lock_guard<std::mutex> g(nq->mu);
+  if (nq->q.empty()) return nq->dead ? -1 : 0;
+  auto& m = nq->q.front();
+  if (m.size() > cap) return -1;
+  int const n = static_cast<int>(m.size());
+  if (n) memcpy(buf, m.data(), n);
+  nq->q.pop_front();
+  return n;
 }
 
 }  // extern "C"

@@ -57,6 +57,18 @@ uint64_t uccl_engine_read_async(uccl_engine_t* e, uint64_t conn, void* ptr,
 // 1 = done, 0 = pending, -1 = unknown id
 int uccl_engine_poll(uccl_engine_t* e, uint64_t xfer);
 
+// --- notify messages (reference: NotifyMsg, p2p/util/common.h:78) ---------
+// Small out-of-band messages riding the same connection. len <= 4080.
+// A connection used for notifies must not also be used for raw
+// uccl_engine_recv (the notify drainer owns the inbound stream).
+// 0 = sent, -1 = error
+int uccl_engine_notify(uccl_engine_t* e, uint64_t conn, void const* data,
+                       size_t len);
+// >=0 = message length copied into buf, 0 with no message = none pending,
+// -1 = error. Non-blocking.
+int uccl_engine_notify_poll(uccl_engine_t* e, uint64_t conn, void* buf,
+                            size_t cap);
+
 #ifdef __cplusplus
 }
 #endif

@@ -4,6 +4,7 @@
 #include <unistd.h>
 
 #include <chrono>
+#include <stdexcept>
 #include <cstring>
 
 #include "../core/env.h"
@@ -607,7 +608,10 @@ void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
   {
     std::unique_lock<std::mutex> lk(c.rx_mu);
     c.rx_cv.wait(lk, [&] { return !c.rxq.empty() || !c.alive || stop_; });
-    UCCL_CHECK(!c.rxq.empty()) << "connection closed during recv";
+    // closing a connection with a pending recv is a normal event (e.g.
+    // teardown of a notify drainer): fail the op, don't abort the process
+    if (c.rxq.empty())
+      throw std::runtime_error("p2p connection closed during recv");
     item = c.rxq.front();
     c.rxq.pop_front();
   }
@@ -729,8 +733,13 @@ uint64_t Endpoint::submit(std::function<void()> fn) {
   {
     std::lock_guard<std::mutex> g(task_mu_);
     tasks_.push_back([fn = std::move(fn), status] {
-      fn();
-      status->store(1);
+      try {
+        fn();
+        status->store(1);
+      } catch (std::exception const& e) {
+        UCCL_LOG_WARN << "async p2p op failed: " << e.what();
+        status->store(2);  // completed-with-failure
+      }
     });
   }
   task_cv_.notify_one();
@@ -810,7 +819,7 @@ bool Endpoint::poll_async(uint64_t xfer_id) {
   std::lock_guard<std::mutex> g(xfer_mu_);
   auto it = xfers_.find(xfer_id);
   UCCL_CHECK(it != xfers_.end()) << "unknown transfer " << xfer_id;
-  if (it->second->load() == 1) {
+  if (it->second->load() != 0) {  // 1 = ok, 2 = failed (warning logged)
     xfers_.erase(it);
     return true;
   }
