@@ -7,6 +7,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <sstream>
+#include <stdexcept>
 #include <string>
 
 namespace uccl {
@@ -26,6 +27,11 @@ inline LogLevel log_level() {
   return lvl;
 }
 
+class CheckError : public std::runtime_error {
+ public:
+  explicit CheckError(std::string const& m) : std::runtime_error(m) {}
+};
+
 class LogMessage {
  public:
   LogMessage(LogLevel lvl, const char* file, int line, bool fatal = false)
@@ -38,7 +44,10 @@ class LogMessage {
     if (fatal_) {
       fprintf(stderr, "%s\n", ss_.str().c_str());
       fflush(stderr);
-      abort();
+      // throw, don't abort: API-reachable failures surface as Python /
+      // C++ exceptions; an unhandled throw in a worker thread still
+      // terminates, matching the old abort for truly-fatal contexts
+      throw CheckError(ss_.str());
     }
     if (lvl_ >= log_level()) {
       fprintf(stderr, "%s\n", ss_.str().c_str());
