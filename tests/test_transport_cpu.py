@@ -301,3 +301,19 @@ def test_transport_eqds_incast_sharing():
                 os.environ.pop(k, None)
             else:
                 os.environ[k] = v
+
+
+def test_transport_peer_death_aborts_cleanly():
+    """Failure detection: when the peer vanishes, the RTO
+    abort threshold marks the flow failed and pending sends raise
+    instead of hanging."""
+    import pytest
+
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0, UCCL_TP_RTO_US=2000,
+                                UCCL_TP_RTO_ABORT=8)
+    # handshake sanity first
+    xfer(a, b, fa, fb, 4096, 81)
+    del b  # peer endpoint (and its sockets) die
+    big = torch.zeros(4 << 20, dtype=torch.uint8)
+    with pytest.raises(RuntimeError):
+        a.send(fa, big)
