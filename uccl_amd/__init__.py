@@ -7,7 +7,13 @@ CDNA4 + xGMI):
   - uccl_amd.p2p        : NIXL-style point-to-point transfer engine
   - uccl_amd.ep         : DeepEP-compatible expert-parallel communication
   - uccl_amd.transport  : software multipath reliable transport (chunking,
-    path spraying, SACK selective repeat, Timely/Swift CC)
+    path spraying, SACK selective repeat, Timely/Swift/paced-EQDS CC)
+  - uccl_amd.ukernel    : chunk-graph planner / lowering / spray executor
+    (relay spraying over xGMI paths, cost-model algorithm choice)
+  - uccl_amd.p2p        : also lossless float compression
+    (send_compressed / send_object) and MR/XferDesc helpers
+  - observability       : chrome-trace recorder (UCCL_TRACE=1), latency
+    percentiles in transport/p2p stats
 """
 
 from __future__ import annotations
