@@ -774,12 +774,19 @@ uint64_t Endpoint::send_async(uint64_t cid, void const* p, size_t n,
       std::unique_lock<std::mutex> lk(c->ord_mu);
       c->ord_cv.wait(lk, [&] { return c->tx_serving == tk || !c->alive; });
     }
+    // serving must advance even if the op throws, or every later ticket
+    // on this connection would wait forever
+    struct Adv {
+      Conn* c;
+      ~Adv() {
+        {
+          std::lock_guard<std::mutex> g(c->ord_mu);
+          ++c->tx_serving;
+        }
+        c->ord_cv.notify_all();
+      }
+    } adv{c.get()};
     do_send(*c, p, n, dev);
-    {
-      std::lock_guard<std::mutex> g(c->ord_mu);
-      ++c->tx_serving;
-    }
-    c->ord_cv.notify_all();
   });
 }
 uint64_t Endpoint::recv_async(uint64_t cid, void* p, size_t n, int dev) {
@@ -794,12 +801,17 @@ uint64_t Endpoint::recv_async(uint64_t cid, void* p, size_t n, int dev) {
       std::unique_lock<std::mutex> lk(c->ord_mu);
       c->ord_cv.wait(lk, [&] { return c->rx_serving == tk || !c->alive; });
     }
+    struct Adv {
+      Conn* c;
+      ~Adv() {
+        {
+          std::lock_guard<std::mutex> g(c->ord_mu);
+          ++c->rx_serving;
+        }
+        c->ord_cv.notify_all();
+      }
+    } adv{c.get()};
     do_recv(*c, p, n, dev);
-    {
-      std::lock_guard<std::mutex> g(c->ord_mu);
-      ++c->rx_serving;
-    }
-    c->ord_cv.notify_all();
   });
 }
 uint64_t Endpoint::write_async(uint64_t cid, void const* p, size_t n, int dev,
