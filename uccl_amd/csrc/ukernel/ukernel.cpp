@@ -606,12 +606,14 @@ float* HostBackend::output(int rank) { return out_[rank].data(); }
 
 float* HostBackend::resolve(BufRef const& b) {
   std::vector<float>* v = nullptr;
+  uint64_t cap = 0;
   switch (b.space) {
-    case Space::kInput: v = &in_[b.rank]; break;
-    case Space::kOutput: v = &out_[b.rank]; break;
-    case Space::kScratch: v = &scratch_[b.rank]; break;
+    case Space::kInput: v = &in_[b.rank]; cap = in_bytes_; break;
+    case Space::kOutput: v = &out_[b.rank]; cap = out_bytes_; break;
+    case Space::kScratch: v = &scratch_[b.rank]; cap = scratch_bytes_; break;
   }
-  assert(b.offset % 4 == 0);
+  if (b.offset % 4 != 0 || b.offset > cap)
+    throw std::out_of_range("ukernel: buffer ref outside its space");
   return v->data() + b.offset / 4;
 }
 
