@@ -65,3 +65,62 @@ def test_xferdesc_roundtrip_and_split():
     assert [p.bytes for p in parts] == [4096, 4096, 1808]
     assert parts[2].offset == 4096 + 8192
     assert all(p.tag == 9 for p in parts)
+
+
+def test_prometheus_metrics_export():
+    import threading
+
+    import torch
+
+    from uccl_amd import _load_native
+    from uccl_amd.utils import metrics
+
+    C = _load_native(required=False)
+    a = C.TransportEndpoint(num_paths=2, chunk_bytes=4096)
+    b = C.TransportEndpoint(num_paths=2, chunk_bytes=4096)
+    out = {}
+    th = threading.Thread(target=lambda: out.setdefault("fb", b.accept()))
+    th.start()
+    fa = a.connect(b.metadata())
+    th.join(timeout=30)
+    src = torch.zeros(65536, dtype=torch.uint8)
+    dst = torch.zeros(65536, dtype=torch.uint8)
+    t = threading.Thread(target=lambda: b.recv(out["fb"], dst))
+    t.start()
+    a.send(fa, src)
+    t.join(timeout=30)
+
+    metrics.track_transport("tp_a", a)
+    text = metrics.render().decode()
+    assert 'uccl_transport_msgs_sent{name="tp_a"} 1.0' in text
+    assert "uccl_transport_rtt_p50_us" in text
+    metrics.untrack("transport", "tp_a")
+    assert "tp_a" not in metrics.render().decode()
+
+
+def test_prometheus_p2p_metrics():
+    import threading
+
+    import torch
+
+    from uccl_amd import p2p
+    from uccl_amd.utils import metrics
+
+    a = p2p.Endpoint(gpu=0, num_workers=1)
+    b = p2p.Endpoint(gpu=0, num_workers=1)
+    ids = {}
+    th = threading.Thread(target=lambda: ids.setdefault("b", b.accept()))
+    th.start()
+    cid = a.connect(b.metadata())
+    th.join(timeout=30)
+    src = torch.zeros(1000, dtype=torch.uint8)
+    dst = torch.zeros(1000, dtype=torch.uint8)
+    t = threading.Thread(target=lambda: b.recv(ids["b"], dst))
+    t.start()
+    a.send(cid, src)
+    t.join(timeout=30)
+    metrics.track_p2p("ep_a", a)
+    text = metrics.render().decode()
+    assert 'uccl_p2p_send_calls{name="ep_a"} 1.0' in text
+    assert 'uccl_p2p_send_bytes{name="ep_a"} 1000.0' in text
+    metrics.untrack("p2p", "ep_a")
