@@ -171,3 +171,31 @@ def test_endpoint_compressed_transfer_multipath():
             os.environ.pop("UCCL_P2P_TRANSPORT", None)
         else:
             os.environ["UCCL_P2P_TRANSPORT"] = old
+
+
+def test_decompress_fuzz_never_crashes():
+    # random and mutated frames must raise cleanly, never corrupt memory
+    import random
+
+    rnd = random.Random(7)
+    # pure garbage
+    for n in (0, 1, 8, 24, 25, 200):
+        frame = torch.randint(0, 256, (n,), dtype=torch.uint8)
+        with pytest.raises(Exception):
+            p2p.decompress(frame)
+    # valid frame with random single-byte mutations: either raises or
+    # returns (possibly wrong bytes if a payload byte flipped) — what it
+    # must never do is crash or over-read
+    base = p2p.compress(torch.randn(4096).bfloat16())
+    for _ in range(200):
+        bad = base.clone()
+        i = rnd.randrange(bad.numel())
+        bad[i] = rnd.randrange(256)
+        try:
+            p2p.decompress(bad)
+        except Exception:
+            pass
+    # truncations at every header-ish boundary
+    for cut in (1, 4, 7, 8, 16, 24, 25, 33, base.numel() - 1):
+        with pytest.raises(Exception):
+            p2p.decompress(base[:cut].clone())

@@ -96,3 +96,24 @@ def link_matrix(stats: dict, world: int):
     """stats['link_bytes'] as a world x world nested list."""
     lb = stats["link_bytes"]
     return [[lb[s * world + d] for d in range(world)] for s in range(world)]
+
+
+def to_dot(graph: Graph) -> str:
+    """Graphviz text for a (raw or lowered) plan — one node per task,
+    colored by op, edges = dependencies. Render with `dot -Tsvg`."""
+    colors = {"copy": "lightblue", "reduce": "orange", "put": "palegreen",
+              "signal": "gray80", "wait": "gray60"}
+    lines = ["digraph plan {", "  rankdir=LR;", "  node [shape=box];"]
+    for ln in graph.dump().splitlines():
+        idx, rest = ln.split(":", 1)
+        op = rest.strip().split()[0]
+        lines.append(
+            f'  t{idx} [label="{idx}: {rest.strip()[:48]}" '
+            f'style=filled fillcolor={colors.get(op, "white")}];')
+        if "deps[" in rest:
+            deps = rest.rsplit("deps[", 1)[1].rstrip("]").split(",")
+            for d in deps:
+                if d:
+                    lines.append(f"  t{d.strip()} -> t{idx};")
+    lines.append("}")
+    return "\n".join(lines)
