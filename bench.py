@@ -104,13 +104,22 @@ def main():
                                    torch.full_like(tensor, expect)))
 
     chk = torch.ones(2 << 20, dtype=torch.bfloat16, device="cuda")
-    if not check_ones(chk):
+    try:
+        staged_ok = check_ones(chk)
+    except Exception as e:  # engine raised: report, don't traceback
+        print(json.dumps({"error": f"allreduce gate raised: {e}"}))
+        sys.exit(2)
+    if not staged_ok:
         print(json.dumps({"error": "allreduce correctness check failed",
                           "got": float(chk[0])}))
         sys.exit(2)
     if args.symmetric and world > 1:
-        schk = comm.symmetric_tensor([1 << 20], torch.bfloat16)
-        if not check_ones(schk):
+        try:
+            schk = comm.symmetric_tensor([1 << 20], torch.bfloat16)
+            sym_ok = check_ones(schk)
+        except Exception:
+            sym_ok = False  # engine refused the symmetric path
+        if not sym_ok:
             args.symmetric = False
             t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
 
