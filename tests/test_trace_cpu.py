@@ -158,3 +158,27 @@ def test_ring_capacity_validation():
 
     with pytest.raises(Exception):
         C.MpmcRingU64(100)  # not a power of two
+
+
+def test_trace_ukernel_execution():
+    import json
+
+    import torch
+
+    from uccl_amd import ukernel as uk
+
+    C.trace_clear()
+    C.trace_set_enabled(True)
+    try:
+        topo = uk.Topology(2)
+        g = uk.lower(uk.plan_allreduce_oneshot(topo, 1024))
+        ins = [torch.ones(256), torch.ones(256) * 2]
+        outs, _ = uk.execute_host(g, ins)
+        assert torch.equal(outs[0], torch.full((256,), 3.0))
+    finally:
+        C.trace_set_enabled(False)
+    doc = json.loads(C.trace_dump_json())
+    names = {e["name"] for e in doc["traceEvents"]
+             if e["cat"] == "ukernel"}
+    assert {"put", "reduce", "copy", "signal"} <= names
+    C.trace_clear()

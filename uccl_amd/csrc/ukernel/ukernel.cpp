@@ -2,6 +2,8 @@
 
 #include "ukernel.h"
 
+#include "../core/trace.h"
+
 #include <algorithm>
 #include <cassert>
 #include <chrono>
@@ -545,16 +547,19 @@ ExecStats execute(ChunkGraph const& g, Backend& backend) {
         continue;
       }
       backoff = 0;
-      switch (t.op) {
-        case Op::kCopy: backend.copy(t); break;
-        case Op::kReduce: backend.reduce(t); break;
-        case Op::kPut:
-          backend.put(t);
-          link_bytes[size_t(t.src.rank) * world + t.dst.rank].fetch_add(
-              t.bytes, std::memory_order_relaxed);
-          break;
-        case Op::kSignal: backend.signal(t.flag); break;
-        case Op::kWait: break;  // poll already succeeded
+      {
+        trace::Span span("ukernel", op_name(t.op));
+        switch (t.op) {
+          case Op::kCopy: backend.copy(t); break;
+          case Op::kReduce: backend.reduce(t); break;
+          case Op::kPut:
+            backend.put(t);
+            link_bytes[size_t(t.src.rank) * world + t.dst.rank].fetch_add(
+                t.bytes, std::memory_order_relaxed);
+            break;
+          case Op::kSignal: backend.signal(t.flag); break;
+          case Op::kWait: break;  // poll already succeeded
+        }
       }
       tasks_run.fetch_add(1, std::memory_order_relaxed);
       for (int s : out[ti])
