@@ -75,9 +75,9 @@ class UcclWork : public c10d::Work {
   }
   bool wait(std::chrono::milliseconds) override {
     synchronize();
-    static bool const blocking =
-        uccl::env_bool("UCCL_BLOCKING_WAIT", false);
-    if (blocking) (void)hipEventSynchronize(event_->ev);
+    // read per call, not once per process (see STATUS.md lesson 8)
+    if (uccl::env_bool("UCCL_BLOCKING_WAIT", false))
+      (void)hipEventSynchronize(event_->ev);
     return true;
   }
   c10::intrusive_ptr<c10::ivalue::Future> getFuture() override {
