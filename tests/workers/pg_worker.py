@@ -19,6 +19,11 @@ def main():
     want = sum(r + 1 for r in range(world))
     assert torch.allclose(t, torch.full_like(t, want)), t[:4]
 
+    ta = torch.full((4096,), float(rank + 1), device="cuda")
+    dist.all_reduce(ta, op=dist.ReduceOp.AVG)
+    torch.cuda.synchronize()
+    assert torch.allclose(ta, torch.full_like(ta, want / world)), ta[:4]
+
     b = torch.full((64,), float(rank), device="cuda")
     dist.broadcast(b, src=0)
     torch.cuda.synchronize()

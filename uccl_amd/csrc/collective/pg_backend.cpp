@@ -132,12 +132,15 @@ class UcclBackend : public c10d::Backend {
       std::vector<at::Tensor>& tensors,
       const c10d::AllreduceOptions& opts) override {
     TORCH_CHECK(tensors.size() == 1, "uccl: one tensor per op");
-    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM,
-                "uccl: only SUM allreduce");
+    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM ||
+                    opts.reduceOp == c10d::ReduceOp::AVG,
+                "uccl: SUM/AVG allreduce only");
     auto& t = tensors[0];
     check(t);
     hipStream_t s = cur_stream(comm_->device());
     comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s);
+    if (opts.reduceOp == c10d::ReduceOp::AVG)
+      t.mul_(1.0 / getSize());  // stream-ordered on the current stream
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::ALLREDUCE, s,
                                          comm_->device(), tensors);
   }
@@ -192,14 +195,17 @@ class UcclBackend : public c10d::Backend {
   c10::intrusive_ptr<c10d::Work> _reduce_scatter_base(
       at::Tensor& output, at::Tensor& input,
       const c10d::ReduceScatterOptions& opts) override {
-    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM,
-                "uccl: only SUM reduce_scatter");
+    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM ||
+                    opts.reduceOp == c10d::ReduceOp::AVG,
+                "uccl: SUM/AVG reduce_scatter only");
     check(input);
     check(output);
     TORCH_CHECK(input.numel() == output.numel() * getSize());
     hipStream_t s = cur_stream(comm_->device());
     comm_->reduce_scatter(output.data_ptr(), input.data_ptr(),
                           output.numel(), to_dtype(input), s);
+    if (opts.reduceOp == c10d::ReduceOp::AVG)
+      output.mul_(1.0 / getSize());
     return c10::make_intrusive<UcclWork>(
         getRank(), c10d::OpType::_REDUCE_SCATTER_BASE, s, comm_->device(),
         std::vector<at::Tensor>{output});
