@@ -317,3 +317,57 @@ def test_transport_peer_death_aborts_cleanly():
     big = torch.zeros(4 << 20, dtype=torch.uint8)
     with pytest.raises(RuntimeError):
         a.send(fa, big)
+
+
+def test_transport_garbage_udp_ignored():
+    """Stray/garbage UDP datagrams at the data ports (wrong magic, bogus
+    flow ids, truncated or oversized frames) must be ignored without
+    crashing the endpoint or corrupting live transfers."""
+    import random
+    import socket
+    import struct
+
+    import os
+    import re
+
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0)
+    # find this process's bound UDP ports (the endpoints' path sockets)
+    inodes = set()
+    for fd in os.listdir("/proc/self/fd"):
+        try:
+            tgt = os.readlink(f"/proc/self/fd/{fd}")
+        except OSError:
+            continue
+        m = re.match(r"socket:\[(\d+)\]", tgt)
+        if m:
+            inodes.add(m.group(1))
+    ports = []
+    with open("/proc/net/udp") as f:
+        next(f)
+        for line in f:
+            parts = line.split()
+            if parts[9] in inodes:
+                ports.append(int(parts[1].split(":")[1], 16))
+    assert ports, "no UDP path sockets found"
+
+    s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    rnd = random.Random(3)
+    for _ in range(300):
+        port = rnd.choice(ports)
+        kind = rnd.randrange(4)
+        if kind == 0:
+            pkt = bytes(rnd.randrange(256) for _ in range(rnd.randrange(90)))
+        elif kind == 1:  # valid-looking data hdr, unknown flow
+            pkt = struct.pack("<IIQQQIIQ", 0x7563636c, 1, rnd.getrandbits(63),
+                              0, 0, rnd.randrange(1 << 20), 4096,
+                              rnd.getrandbits(60)) + b"x" * 100
+        elif kind == 2:  # ack-ish
+            pkt = struct.pack("<IIQI", 0x7563636c, 2, rnd.getrandbits(63),
+                              rnd.randrange(1 << 16)) + b"\x00" * 48
+        else:
+            pkt = b"\xff" * rnd.choice([1, 7, 65, 1400])
+        s.sendto(pkt, ("127.0.0.1", port))
+    # live transfer still works afterwards
+    xfer(a, b, fa, fb, 1 << 20, 91)
+    st = a.stats()
+    assert st.msgs_sent >= 1
