@@ -371,3 +371,16 @@ def test_transport_garbage_udp_ignored():
     xfer(a, b, fa, fb, 1 << 20, 91)
     st = a.stats()
     assert st.msgs_sent >= 1
+
+
+def test_transport_eqds_paced_under_loss():
+    """Pull pacing composed with 5% data loss and 5% ack loss: grants,
+    SACK recovery, and the credit-refresh path must cooperate."""
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=5, UCCL_TP_ACK_LOSS_PCT=5,
+                                UCCL_TP_CC="eqds", UCCL_TP_EQDS_MBPS=200,
+                                UCCL_TP_RWND_KB=256)
+    for n, sd in [(1 << 20, 95), (3 << 20, 96), (0, 97), (777, 98)]:
+        xfer(a, b, fa, fb, n, sd)
+    st = a.stats()
+    assert st.msgs_sent == 4
+    assert st.retransmits + st.rto_retransmits > 0  # loss actually hit
