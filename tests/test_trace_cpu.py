@@ -182,3 +182,26 @@ def test_trace_ukernel_execution():
              if e["cat"] == "ukernel"}
     assert {"put", "reduce", "copy", "signal"} <= names
     C.trace_clear()
+
+
+def test_spsc_ring_two_threads():
+    r = C.SpscRingU64(1 << 10)
+    n = 200000
+    got = []
+
+    def consumer():
+        expect = 0
+        while expect < n:
+            v = r.pop()
+            if v is not None:
+                assert v == expect  # strict FIFO
+                expect += 1
+        got.append(expect)
+
+    t = threading.Thread(target=consumer)
+    t.start()
+    for i in range(n):
+        while not r.push(i):
+            pass
+    t.join(timeout=60)
+    assert got == [n]
