@@ -51,3 +51,41 @@ def test_mrmap_invariants(regions):
     for ab, al, aid in accepted:
         hit = m.find(ab + al)
         assert hit is None or hit[0] != aid
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.integers(0, 1 << 40), st.integers(1, 100000),
+       st.integers(1, 1 << 22))
+def test_xferdesc_split_covers_exactly(offset, nbytes, chunk):
+    from uccl_amd.p2p.utils import XferDesc
+
+    d = XferDesc(mr_id=1, offset=offset, bytes=nbytes, tag=5)
+    parts = d.split(chunk)
+    assert sum(p.bytes for p in parts) == nbytes
+    pos = offset
+    for p in parts:
+        assert p.offset == pos and 0 < p.bytes <= chunk
+        pos += p.bytes
+    rt = XferDesc.deserialize(d.serialize())
+    assert rt == d
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.floats(min_value=0.001, max_value=1e7,
+                          allow_nan=False), min_size=1, max_size=300))
+def test_latency_hist_percentile_monotone(samples):
+    from uccl_amd import _load_native
+
+    C = _load_native(required=False)
+    h = C.LatencyHist()
+    for us in samples:
+        h.record_us(us)
+    assert h.count() == len(samples)
+    last = 0.0
+    for p in (0, 25, 50, 75, 90, 99, 100):
+        v = h.percentile_us(p)
+        assert v >= last
+        last = v
+    # upper bound respects bucket resolution (~41% + rounding)
+    assert h.percentile_us(100) <= max(samples) * 2 + 1e-3
+    assert h.percentile_us(0) >= min(samples) / 2 - 1e-3
