@@ -39,6 +39,10 @@ def main():
         fn.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t,
                        ctypes.c_int] + extra + [ctypes.c_void_p,
                                                 ctypes.c_void_p]
+    lib.ncclAllToAll.restype = ctypes.c_int
+    lib.ncclAllToAll.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                 ctypes.c_size_t, ctypes.c_int,
+                                 ctypes.c_void_p, ctypes.c_void_p]
     lib.ncclAllGather.restype = ctypes.c_int
     lib.ncclAllGather.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                   ctypes.c_size_t, ctypes.c_int,
@@ -112,6 +116,20 @@ def main():
                                ncclFloat32, ncclSum, comm, stream)
     torch.cuda.synchronize()
     assert rc == 0 and torch.allclose(rs_out, torch.full_like(rs_out, want))
+
+    # all_to_all (RCCL extension): segment j of rank i's input lands as
+    # segment i of rank j's output
+    a2a_in = torch.cat([torch.full((128,), float(rank * 10 + j),
+                                   device="cuda") for j in range(world)])
+    a2a_out = torch.empty_like(a2a_in)
+    rc = lib.ncclAllToAll(a2a_in.data_ptr(), a2a_out.data_ptr(), 128,
+                          ncclFloat32, comm, stream)
+    torch.cuda.synchronize()
+    assert rc == 0
+    for src in range(world):
+        seg = a2a_out[src * 128:(src + 1) * 128]
+        assert torch.allclose(
+            seg, torch.full_like(seg, float(src * 10 + rank))), (src, seg[0])
 
     lib.ncclCommDestroy(comm)
     print(f"[rank {rank}] NCCL SHIM OK", flush=True)
