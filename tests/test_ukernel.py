@@ -233,3 +233,21 @@ def test_reducescatter_plan():
     total = torch.stack(ins).sum(0)
     for r, o in enumerate(outs):
         assert torch.equal(o, total[r * shard:(r + 1) * shard]), r
+
+
+def test_alltoall_plan():
+    world, seg = 4, 500  # elems per segment
+    topo = uk.Topology(world)
+    ins = _inputs(world, world * seg, seed=31)
+    g = uk.lower(uk.plan_alltoall(topo, seg * 4, chunk_bytes=1024))
+    outs, stats = uk.execute_host(g, ins, out_bytes=world * seg * 4)
+    for dst in range(world):
+        for src in range(world):
+            got = outs[dst][src * seg:(src + 1) * seg]
+            want = ins[src][dst * seg:(dst + 1) * seg]
+            assert torch.equal(got, want), (src, dst)
+    # every directed link carried exactly seg bytes
+    lm = uk.link_matrix(stats, world)
+    for s_ in range(world):
+        for d_ in range(world):
+            assert lm[s_][d_] == (0 if s_ == d_ else seg * 4)

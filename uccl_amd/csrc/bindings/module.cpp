@@ -567,6 +567,8 @@ PYBIND11_MODULE(_C, m) {
   m.def("uk_plan_reducescatter", &uk::plan_reducescatter, py::arg("topo"),
         py::arg("shard_bytes"), py::arg("elem_bytes") = 4,
         py::arg("chunk_bytes") = 1 << 20);
+  m.def("uk_plan_alltoall", &uk::plan_alltoall, py::arg("topo"),
+        py::arg("seg_bytes"), py::arg("chunk_bytes") = 1 << 20);
   m.def("uk_lower", &uk::lower);
   m.def("uk_estimate_us", &uk::estimate_us, py::arg("graph"),
         py::arg("topo"), py::arg("link_gbps") = 150.0,

@@ -339,6 +339,28 @@ ChunkGraph plan_reducescatter(Topology const& topo, uint64_t shard_bytes,
   return g;
 }
 
+ChunkGraph plan_alltoall(Topology const& topo, uint64_t seg_bytes,
+                         uint64_t chunk_bytes) {
+  int const world = topo.world;
+  ChunkGraph g;
+  g.world = world;
+  for (int src = 0; src < world; ++src) {
+    for (int dst = 0; dst < world; ++dst) {
+      for (uint64_t c = 0; c < seg_bytes; c += chunk_bytes) {
+        uint64_t const len = std::min(chunk_bytes, seg_bytes - c);
+        Task t;
+        t.op = src == dst ? Op::kCopy : Op::kPut;
+        t.rank = src;
+        t.src = {src, Space::kInput, uint64_t(dst) * seg_bytes + c};
+        t.dst = {dst, Space::kOutput, uint64_t(src) * seg_bytes + c};
+        t.bytes = len;
+        g.add(std::move(t));
+      }
+    }
+  }
+  return g;
+}
+
 // ---------------------------------------------------------------------------
 // Cost model
 // ---------------------------------------------------------------------------

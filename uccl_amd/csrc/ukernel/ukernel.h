@@ -133,6 +133,11 @@ ChunkGraph plan_allgather(Topology const& topo, uint64_t nbytes,
 ChunkGraph plan_reducescatter(Topology const& topo, uint64_t shard_bytes,
                               uint64_t elem_bytes, uint64_t chunk_bytes);
 
+// All-to-all: segment j of rank i's input becomes segment i of rank j's
+// output (all-pairs push, chunked).
+ChunkGraph plan_alltoall(Topology const& topo, uint64_t seg_bytes,
+                         uint64_t chunk_bytes);
+
 // ---------------------------------------------------------------------------
 // Cost model: deterministic list-scheduling estimate of a plan's
 // completion time. Resources: every directed link serializes its kPuts at

@@ -63,6 +63,10 @@ def plan_reducescatter(topo, shard_bytes: int, elem_bytes: int = 4,
                                     chunk_bytes)
 
 
+def plan_alltoall(topo, seg_bytes: int, chunk_bytes: int = 1 << 20) -> Graph:
+    return _C.uk_plan_alltoall(topo, seg_bytes, chunk_bytes)
+
+
 def lower(graph: Graph) -> Graph:
     return _C.uk_lower(graph)
 
