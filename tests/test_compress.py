@@ -199,3 +199,20 @@ def test_decompress_fuzz_never_crashes():
     for cut in (1, 4, 7, 8, 16, 24, 25, 33, base.numel() - 1):
         with pytest.raises(Exception):
             p2p.decompress(base[:cut].clone())
+
+
+def test_decompress_overflow_sizes_rejected():
+    # craft frames whose plane/block size fields would wrap u64 bounds
+    import struct
+
+    base = p2p.compress(torch.randn(4096).bfloat16())
+    raw = bytearray(base.tolist())
+    # plane table starts at byte 25: [encoding u8][stored u64] per plane
+    evil = raw.copy()
+    struct.pack_into("<Q", evil, 26, (1 << 64) - 8)  # plane 0 stored size
+    with pytest.raises(Exception):
+        p2p.decompress(torch.frombuffer(bytes(evil), dtype=torch.uint8))
+    evil2 = raw.copy()
+    struct.pack_into("<Q", evil2, 26 + 9, (1 << 63))  # plane 1 stored size
+    with pytest.raises(Exception):
+        p2p.decompress(torch.frombuffer(bytes(evil2), dtype=torch.uint8))

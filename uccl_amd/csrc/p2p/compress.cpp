@@ -109,7 +109,9 @@ bool inflate_plane(uint8_t const* p, size_t n, uint8_t* out, size_t out_n) {
   size_t out_off = 0;
   std::vector<std::future<bool>> futs;
   for (uint32_t b = 0; b < nblocks; ++b) {
-    if (off + sizes[b] > n) return false;
+    // subtraction form: sizes[] comes off the wire; off + sizes[b] could
+    // wrap a u64 and bypass the bound
+    if (sizes[b] > n - off) return false;
     size_t const blen = b + 1 < nblocks ? kBlock : out_n - out_off;
     if (out_off + blen > out_n) return false;
     uint8_t const* src = p + off;
@@ -241,7 +243,7 @@ size_t decompress(void const* frame, size_t frame_bytes, void* out,
   size_t const tail = bytes - elems * elem_size;
   std::vector<uint8_t> plane(elems);
   for (int pl = 0; pl < nplanes; ++pl) {
-    if (frame_bytes < off + meta[pl].stored)
+    if (meta[pl].stored > frame_bytes - off)  // subtraction: no u64 wrap
       throw std::runtime_error("compress: truncated plane");
     uint8_t const* stored = p + off;
     uint8_t const* plane_data;
