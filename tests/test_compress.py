@@ -211,8 +211,8 @@ def test_decompress_overflow_sizes_rejected():
     evil = raw.copy()
     struct.pack_into("<Q", evil, 26, (1 << 64) - 8)  # plane 0 stored size
     with pytest.raises(Exception):
-        p2p.decompress(torch.frombuffer(bytes(evil), dtype=torch.uint8))
+        p2p.decompress(torch.frombuffer(bytearray(evil), dtype=torch.uint8))
     evil2 = raw.copy()
     struct.pack_into("<Q", evil2, 26 + 9, (1 << 63))  # plane 1 stored size
     with pytest.raises(Exception):
-        p2p.decompress(torch.frombuffer(bytes(evil2), dtype=torch.uint8))
+        p2p.decompress(torch.frombuffer(bytearray(evil2), dtype=torch.uint8))
