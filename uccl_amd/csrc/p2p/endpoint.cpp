@@ -377,6 +377,25 @@ void Endpoint::dereg(uint64_t mr_id) {
 
 void Endpoint::rx_loop(std::shared_ptr<Conn> c) {
   if (gpu_ >= 0) (void)hipSetDevice(gpu_);
+  try {
+    rx_loop_body(c);
+  } catch (std::exception const& e) {
+    // a malformed frame or failed check kills THIS connection, not the
+    // process: mark it dead so blocked ops fail (pending-ops contract)
+    UCCL_LOG_WARN << "p2p rx loop failed: " << e.what();
+    {
+      std::lock_guard<std::mutex> g1(c->rx_mu);
+      std::lock_guard<std::mutex> g2(c->tok_mu);
+      std::lock_guard<std::mutex> g3(c->ord_mu);
+      c->alive = false;
+    }
+    c->rx_cv.notify_all();
+    c->tok_cv.notify_all();
+    c->ord_cv.notify_all();
+  }
+}
+
+void Endpoint::rx_loop_body(std::shared_ptr<Conn> c) {
   while (c->alive && !stop_) {
     MsgHdr h{};
     if (!c->recv_hdr(&h)) break;

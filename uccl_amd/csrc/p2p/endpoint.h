@@ -107,6 +107,7 @@ class Endpoint {
   struct RxItem;
 
   void rx_loop(std::shared_ptr<Conn> c);
+  void rx_loop_body(std::shared_ptr<Conn> c);
   void worker_loop();
   uint64_t submit(std::function<void()> fn);
   void do_send(Conn& c, void const* ptr, size_t bytes, int device);

@@ -236,10 +236,17 @@ void tx_loop(Comm* c) {
   }
 }
 
+constexpr uint64_t kMaxWireBytes = 1ull << 31;  // frame sanity cap
+
 void rx_loop(Comm* c) {
   while (c->alive) {
     WireHdr h{};
     if (!chan_recv(c, &h, sizeof(h))) return;
+    if (h.bytes > kMaxWireBytes) {
+      PLOG(NCCL_LOG_WARN, "uccl-net: bogus frame size %llu; closing",
+           (unsigned long long)h.bytes);
+      return;
+    }
     // try to match a posted recv by tag
     PostedRecv pr{};
     bool matched = false;
