@@ -62,10 +62,13 @@ std::mutex g_mu;
 std::vector<std::unique_ptr<ExchangeServer>> g_servers;
 
 struct WireBlob {
+  uint64_t magic;  // rejects stray connections on recycled ports
   int rank;
   int nranks;
   uint32_t len;
 };
+constexpr uint64_t kBlobMagic = 0x7563636c2d696478ULL;
+constexpr uint32_t kMaxBlobLen = 1u << 20;  // IPC blobs are ~KB
 
 void serve(int listen_fd) {
   // collect blobs until all nranks have reported, then broadcast
@@ -77,7 +80,8 @@ void serve(int listen_fd) {
     int fd = ::accept(listen_fd, nullptr, nullptr);
     if (fd < 0) return;  // shut down
     WireBlob h{};
-    if (!uccl::net::recv_all(fd, &h, sizeof(h))) {
+    if (!uccl::net::recv_all(fd, &h, sizeof(h)) || h.magic != kBlobMagic ||
+        h.nranks < 1 || h.nranks > 8 || h.len > kMaxBlobLen) {
       ::close(fd);
       continue;
     }
@@ -175,7 +179,8 @@ ncclResult_t ncclCommInitRank(ncclComm_t* comm, int nranks,
     uc->comm = std::make_unique<Communicator>(rank, nranks, device, 0);
     std::string blob = uc->comm->handle_bytes();
     int fd = uccl::net::connect_to(p.ip, p.port);
-    WireBlob h{rank, nranks, static_cast<uint32_t>(blob.size())};
+    WireBlob h{kBlobMagic, rank, nranks,
+               static_cast<uint32_t>(blob.size())};
     uccl::net::send_all(fd, &h, sizeof(h));
     uccl::net::send_all(fd, blob.data(), blob.size());
     uint32_t n = 0;
@@ -189,6 +194,10 @@ ncclResult_t ncclCommInitRank(ncclComm_t* comm, int nranks,
     for (int r = 0; r < nranks; ++r) {
       uint32_t l = 0;
       uccl::net::recv_all(fd, &l, sizeof(l));
+      if (l > kMaxBlobLen) {
+        ::close(fd);
+        return ncclSystemError;
+      }
       handles[r].resize(l);
       uccl::net::recv_all(fd, handles[r].data(), l);
     }
