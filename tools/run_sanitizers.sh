@@ -1,0 +1,20 @@
+#!/bin/bash
+# Build and run the sanitizer harnesses (TSan + ASan/UBSan) over the
+# host-side native layers: reliable transport, codec, planner/executor.
+# These catch what the pytest tier cannot (races, OOB, UB) because the
+# torch extension cannot easily be sanitizer-instrumented.
+set -e
+cd "$(dirname "$0")/.."
+OUT=${TMPDIR:-/tmp}
+for san in thread address,undefined; do
+  tag=${san%%,*}
+  g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_transport.cpp \
+      uccl_amd/csrc/transport/reliable.cpp uccl_amd/csrc/core/trace.cpp \
+      -o "$OUT/san_tp_$tag" -pthread
+  g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_codec_ukernel.cpp \
+      uccl_amd/csrc/p2p/compress.cpp uccl_amd/csrc/ukernel/ukernel.cpp \
+      uccl_amd/csrc/core/trace.cpp -o "$OUT/san_cu_$tag" -pthread -lz
+  echo "== $tag: transport =="; "$OUT/san_tp_$tag"
+  echo "== $tag: codec+ukernel =="; "$OUT/san_cu_$tag"
+done
+echo "ALL SANITIZERS CLEAN"

@@ -517,50 +517,46 @@ ExecStats execute(ChunkGraph const& g, Backend& backend) {
     for (int d : g.tasks[i].deps) out[d].push_back(static_cast<int>(i));
   }
 
-  struct RankQ {
-    std::mutex mu;
-    std::condition_variable cv;
-    std::deque<int> q;
-  };
-  std::vector<RankQ> queues(world);
-  std::atomic<size_t> done{0};
+  // one shared mutex/cv for all rank queues: a host executor is not
+  // lock-contention-bound, and the single-lock design makes the
+  // shutdown handshake trivially race-free (validated under TSan by
+  // tools/run_sanitizers.sh)
+  std::mutex mu;
+  std::condition_variable cv;
+  std::vector<std::deque<int>> queues(world);
+  size_t done = 0;  // guarded by mu
   std::atomic<uint64_t> tasks_run{0}, wait_requeues{0};
   std::vector<std::atomic<uint64_t>> link_bytes(size_t(world) * world);
   for (auto& b : link_bytes) b.store(0, std::memory_order_relaxed);
 
   auto push = [&](int ti) {
-    auto& rq = queues[g.tasks[ti].rank];
     {
-      std::lock_guard<std::mutex> lk(rq.mu);
-      rq.q.push_back(ti);
+      std::lock_guard<std::mutex> lk(mu);
+      queues[g.tasks[ti].rank].push_back(ti);
     }
-    rq.cv.notify_one();
+    cv.notify_all();
   };
   for (size_t i = 0; i < n; ++i)
     if (g.tasks[i].deps.empty()) push(static_cast<int>(i));
 
   auto worker = [&](int rank) {
-    auto& rq = queues[rank];
     int backoff = 0;
-    while (done.load(std::memory_order_acquire) < n) {
+    for (;;) {
       int ti = -1;
       {
-        std::unique_lock<std::mutex> lk(rq.mu);
-        rq.cv.wait_for(lk, std::chrono::milliseconds(1),
-                       [&] { return !rq.q.empty(); });
-        if (!rq.q.empty()) {
-          ti = rq.q.front();
-          rq.q.pop_front();
-        }
+        std::unique_lock<std::mutex> lk(mu);
+        cv.wait(lk, [&] { return !queues[rank].empty() || done >= n; });
+        if (queues[rank].empty()) return;  // all tasks complete
+        ti = queues[rank].front();
+        queues[rank].pop_front();
       }
-      if (ti < 0) continue;
       Task const& t = g.tasks[ti];
       if (t.op == Op::kWait && !backend.poll(t.flag)) {
         // deferred re-queue: let other ready chunks on this rank run
         wait_requeues.fetch_add(1, std::memory_order_relaxed);
         {
-          std::lock_guard<std::mutex> lk(rq.mu);
-          rq.q.push_back(ti);
+          std::lock_guard<std::mutex> lk(mu);
+          queues[rank].push_back(ti);
         }
         if (++backoff > 64) {
           std::this_thread::sleep_for(std::chrono::microseconds(50));
@@ -586,8 +582,10 @@ ExecStats execute(ChunkGraph const& g, Backend& backend) {
       tasks_run.fetch_add(1, std::memory_order_relaxed);
       for (int s : out[ti])
         if (indeg[s].fetch_sub(1, std::memory_order_acq_rel) == 1) push(s);
-      if (done.fetch_add(1, std::memory_order_acq_rel) + 1 == n)
-        for (auto& q : queues) q.cv.notify_all();
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        if (++done == n) cv.notify_all();
+      }
     }
   };
 
