@@ -14,7 +14,13 @@ for san in thread address,undefined; do
   g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_codec_ukernel.cpp \
       uccl_amd/csrc/p2p/compress.cpp uccl_amd/csrc/ukernel/ukernel.cpp \
       uccl_amd/csrc/core/trace.cpp -o "$OUT/san_cu_$tag" -pthread -lz
+  g++ -O1 -g -std=c++17 -fsanitize=$san -D__HIP_PLATFORM_AMD__=1 \
+      -I/opt/rocm/include tools/san_p2p.cpp \
+      uccl_amd/csrc/p2p/endpoint.cpp uccl_amd/csrc/transport/reliable.cpp \
+      uccl_amd/csrc/core/trace.cpp -o "$OUT/san_p2p_$tag" -pthread \
+      -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,/opt/rocm/lib
   echo "== $tag: transport =="; "$OUT/san_tp_$tag"
   echo "== $tag: codec+ukernel =="; "$OUT/san_cu_$tag"
+  echo "== $tag: p2p endpoint =="; "$OUT/san_p2p_$tag"
 done
 echo "ALL SANITIZERS CLEAN"
