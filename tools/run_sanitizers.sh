@@ -19,8 +19,26 @@ for san in thread address,undefined; do
       uccl_amd/csrc/p2p/endpoint.cpp uccl_amd/csrc/transport/reliable.cpp \
       uccl_amd/csrc/core/trace.cpp -o "$OUT/san_p2p_$tag" -pthread \
       -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,/opt/rocm/lib
+  # net plugin: dlopen harness against a sanitizer-built .so. The TSan
+  # build swaps timed cv waits for untimed ones (UCCL_SAN_NO_TIMED_WAIT):
+  # this libtsan lacks pthread_cond_clockwait interception, so wait_for
+  # reports false double-lock/races (verified with a minimal repro).
+  extra=""
+  [ "$tag" = thread ] && extra=-DUCCL_SAN_NO_TIMED_WAIT
+  g++ -O1 -g -std=c++17 -fsanitize=$san $extra -fPIC -shared \
+      uccl_amd/csrc/plugin/tcp_plugin.cpp \
+      uccl_amd/csrc/transport/reliable.cpp uccl_amd/csrc/core/trace.cpp \
+      -o "$OUT/librccl-net-uccl-$tag.so" -pthread
+  g++ -O1 -g -std=c++17 -fsanitize=$san \
+      uccl_amd/csrc/plugin/plugin_test_main.cpp \
+      -o "$OUT/plugin_test_$tag" -ldl -pthread
   echo "== $tag: transport =="; "$OUT/san_tp_$tag"
   echo "== $tag: codec+ukernel =="; "$OUT/san_cu_$tag"
   echo "== $tag: p2p endpoint =="; "$OUT/san_p2p_$tag"
+  echo "== $tag: plugin (multipath) =="
+  "$OUT/plugin_test_$tag" "$OUT/librccl-net-uccl-$tag.so" | tail -1
+  echo "== $tag: plugin (tcp) =="
+  UCCL_NET_TRANSPORT=tcp "$OUT/plugin_test_$tag" \
+      "$OUT/librccl-net-uccl-$tag.so" | tail -1
 done
 echo "ALL SANITIZERS CLEAN"

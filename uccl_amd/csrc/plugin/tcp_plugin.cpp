@@ -223,7 +223,16 @@ void tx_loop(Comm* c) {
       // at 50ms (the reference's engines busy-poll with adaptive sleep
       // for the same robustness, p2p/util/adaptive_sleeper.h).
       while (c->sendq.empty() && c->alive)
+#ifdef UCCL_SAN_NO_TIMED_WAIT
+        // TSan build only: this toolchain's libtsan does not intercept
+        // pthread_cond_clockwait (libstdc++'s wait_for), so the in-wait
+        // unlock is invisible and every timed wait reports a false
+        // "double lock"/race (verified with a 20-line repro). The
+        // production bounded wait stays timed.
+        c->cv.wait(lk);
+#else
         c->cv.wait_for(lk, std::chrono::milliseconds(50));
+#endif
       if (!c->alive) return;
       op = c->sendq.front();
       c->sendq.pop_front();
