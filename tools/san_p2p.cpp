@@ -16,6 +16,8 @@ using uccl::p2p::Endpoint;
 int main() {
   setenv("UCCL_P2P_ENABLE_IPC", "0", 1);
   for (int round = 0; round < 4; ++round) {
+    // alternate data planes: plain TCP and the multipath reliable engine
+    setenv("UCCL_P2P_TRANSPORT", round % 2 ? "multipath" : "tcp", 1);
     // buffers BEFORE endpoints: destruction is reverse order, and the
     // endpoints' rx threads (joined in ~Endpoint) must die before any
     // memory they served one-sided reads from
