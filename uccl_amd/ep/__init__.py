@@ -100,6 +100,12 @@ class Buffer:
         return self._b.combine(expert_out, topk_idx,
                                topk_weights.float().contiguous())
 
+    def close(self):
+        """Release the native buffer (symmetric heap + IPC handles)
+        promptly instead of waiting for GC — used by ElasticBuffer on
+        membership changes."""
+        self._b = None
+
     # DeepEP-compatible aliases (low-latency intranode semantics)
     low_latency_dispatch = dispatch
     low_latency_combine = combine
