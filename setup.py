@@ -21,7 +21,8 @@ setup(
     description="MI355X-native GPU communication framework "
                 "(collectives, P2P, EP, multipath transport)",
     packages=["uccl_amd", "uccl_amd.collective", "uccl_amd.p2p",
-              "uccl_amd.ep", "uccl_amd.transport", "uccl_amd.utils"],
+              "uccl_amd.ep", "uccl_amd.transport", "uccl_amd.ukernel",
+              "uccl_amd.utils"],
     package_data={"uccl_amd": ["*.so", "lib/*.so", "csrc/**/*"]},
     cmdclass={"build_py": BuildNative},
     python_requires=">=3.10",
