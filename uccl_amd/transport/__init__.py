@@ -12,7 +12,8 @@ deterministic loss injection.
     a.send(flow, host_tensor); a.recv(flow, host_tensor)
     st = a.stats()   # counters + cwnd/srtt + RTT p50/p99
 
-Knobs (all env, NCCL_*/RCCL_* aliases honored): UCCL_TP_CC,
+Knobs (env; interface selection additionally honors the
+NCCL_/RCCL_SOCKET_IFNAME aliases): UCCL_TP_CC,
 UCCL_TP_EQDS_MBPS, UCCL_TP_CWND_MAX, UCCL_TP_RWND_KB, UCCL_TP_RTO_US,
 UCCL_TP_RTO_ABORT, UCCL_TP_PACE_MBPS, UCCL_TP_LOSS_PCT,
 UCCL_TP_ACK_LOSS_PCT, UCCL_TP_MAX_MSG_MB.
