@@ -117,6 +117,46 @@ def main():
     check("allreduce[chunked]", x,
           expected_sum(world, count, torch.float32, seed), 0.0)
 
+    # --- allreduce reduction ops (min / max / prod) across all algo paths ---
+    def expected_op(op, count, dtype, seed_tag):
+        ins = torch.stack([
+            make_input(r, count, dtype, seed_tag).cpu().float()
+            for r in range(world)])
+        if op == "min":
+            ref = ins.min(dim=0).values
+        elif op == "max":
+            ref = ins.max(dim=0).values
+        else:
+            ref = ins.prod(dim=0)  # fp32 accumulation, like the kernels
+        return ref.to(dtype)
+
+    op_cases = [(5000, "ll"), (200000, "oneshot"), ((3 << 20) + 5, "twoshot")]
+    op_dtypes = (torch.float32, torch.bfloat16, torch.int32)
+    if light:
+        op_cases = op_cases[:2]
+        op_dtypes = (torch.float32, torch.bfloat16)
+    for count, label in op_cases:
+        for op in ("min", "max", "prod"):
+            for dtype in op_dtypes:
+                if dtype == torch.int32 and op == "prod":
+                    continue  # int overflow wraps; not a meaningful check
+                seed += 1
+                x = make_input(rank, count, dtype, seed)
+                comm.all_reduce(x, op=op)
+                torch.cuda.synchronize()
+                check(f"allreduce[{op},{label},{dtype}]", x,
+                      expected_op(op, count, dtype, seed), tol[dtype])
+
+    # reduce_scatter with a non-sum op
+    seed += 1
+    n = 8192
+    mine = make_input(rank, world * n, torch.float32, seed)
+    rs_out = torch.empty(n, dtype=torch.float32, device="cuda")
+    comm.reduce_scatter(rs_out, mine, op="max")
+    torch.cuda.synchronize()
+    full = expected_op("max", world * n, torch.float32, seed)
+    check("reduce_scatter[max]", rs_out, full[rank * n:(rank + 1) * n], 0.0)
+
     # --- allgather ----------------------------------------------------------
     seed += 1
     n = 12345

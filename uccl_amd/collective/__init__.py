@@ -44,14 +44,16 @@ class Communicator:
         self._c.connect(list(all_handles))
 
     # -- collectives (async on the current torch stream) ---------------------
-    def all_reduce(self, t: torch.Tensor) -> None:
-        self._c.all_reduce(t)
+    def all_reduce(self, t: torch.Tensor, op: str = "sum") -> None:
+        """op: sum | prod | min | max (fp32 accumulation for 16/8-bit floats)."""
+        self._c.all_reduce(t, op)
 
     def all_gather(self, out: torch.Tensor, inp: torch.Tensor) -> None:
         self._c.all_gather(out, inp)
 
-    def reduce_scatter(self, out: torch.Tensor, inp: torch.Tensor) -> None:
-        self._c.reduce_scatter(out, inp)
+    def reduce_scatter(self, out: torch.Tensor, inp: torch.Tensor,
+                       op: str = "sum") -> None:
+        self._c.reduce_scatter(out, inp, op)
 
     def broadcast(self, t: torch.Tensor, root: int) -> None:
         self._c.broadcast(t, root)

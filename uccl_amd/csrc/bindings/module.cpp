@@ -55,6 +55,15 @@ void check_tensor(const at::Tensor& t) {
   TORCH_CHECK(t.is_contiguous(), "uccl_amd: tensor must be contiguous");
 }
 
+uccl::RedOp redop_of(const std::string& op) {
+  if (op == "sum") return uccl::RedOp::kSum;
+  if (op == "prod") return uccl::RedOp::kProd;
+  if (op == "min") return uccl::RedOp::kMin;
+  if (op == "max") return uccl::RedOp::kMax;
+  TORCH_CHECK(false, "uccl_amd: unknown reduce op '", op,
+              "' (sum|prod|min|max)");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -105,11 +114,12 @@ PYBIND11_MODULE(_C, m) {
              c.connect(handles);
            })
       .def("all_reduce",
-           [](Communicator& c, at::Tensor t) {
+           [](Communicator& c, at::Tensor t, const std::string& op) {
              check_tensor(t);
              c.all_reduce(t.data_ptr(), t.numel(), dtype_of(t),
-                          current_stream(c.device()));
-           })
+                          current_stream(c.device()), redop_of(op));
+           },
+           py::arg("tensor"), py::arg("op") = "sum")
       .def("all_gather",
            [](Communicator& c, at::Tensor out, at::Tensor in) {
              check_tensor(out);
@@ -121,15 +131,18 @@ PYBIND11_MODULE(_C, m) {
                           dtype_of(in), current_stream(c.device()));
            })
       .def("reduce_scatter",
-           [](Communicator& c, at::Tensor out, at::Tensor in) {
+           [](Communicator& c, at::Tensor out, at::Tensor in,
+              const std::string& op) {
              check_tensor(out);
              check_tensor(in);
              TORCH_CHECK(in.numel() == out.numel() * c.world(),
                          "reduce_scatter: in must have world*numel(out)");
              TORCH_CHECK(out.scalar_type() == in.scalar_type());
              c.reduce_scatter(out.data_ptr(), in.data_ptr(), out.numel(),
-                              dtype_of(in), current_stream(c.device()));
-           })
+                              dtype_of(in), current_stream(c.device()),
+                              redop_of(op));
+           },
+           py::arg("out"), py::arg("in"), py::arg("op") = "sum")
       .def("broadcast",
            [](Communicator& c, at::Tensor t, int root) {
              check_tensor(t);

@@ -143,7 +143,7 @@ CommView Communicator::view(uint64_t seq) const {
 }
 
 void Communicator::all_reduce(void* data, size_t count, Dtype dt,
-                              hipStream_t stream) {
+                              hipStream_t stream, RedOp op) {
   tally(0, count * dtype_size(dt));
   if (world_ == 1) {
     // Sum over one rank is the identity. By default this is a no-op; with
@@ -160,7 +160,7 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
       CommView const cv = view(next_seq());
       launch_copy(static_cast<char*>(heap_) + cv.sa_off, p, n * es, stream);
       launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
-      launch_oneshot_allreduce(cv, p, n, dt, stream);
+      launch_oneshot_allreduce(cv, p, n, dt, op, stream);
     }
     return;
   }
@@ -171,7 +171,8 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
 
   if (bytes <= ll_threshold_ && es < 8) {
     // (8-byte dtypes skip LL: its 4B packet lanes can't carry i64/f64)
-    launch_ll_allreduce(view(next_seq()), data, data, count, dt, stream);
+    launch_ll_allreduce(view(next_seq()), data, data, count, dt, op,
+                        stream);
     return;
   }
   if (is_symmetric_ptr(data)) {
@@ -186,7 +187,7 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
     CommView const cv2 = view(next_seq());
     // entry barrier: every rank's input is produced + published
     launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
-    launch_twoshot_sym_rs(cv, uoff, count, dt, stream);
+    launch_twoshot_sym_rs(cv, uoff, count, dt, op, stream);
     // mid barrier: nobody reads user inputs any more -> pushes may land
     launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
     launch_twoshot_sym_push(cv, uoff, count, dt, stream);
@@ -204,9 +205,9 @@ void Communicator::all_reduce(void* data, size_t count, Dtype dt,
     launch_copy(static_cast<char*>(heap_) + cv.sa_off, p, n * es, stream);
     launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     if (oneshot) {
-      launch_oneshot_allreduce(cv, p, n, dt, stream);
+      launch_oneshot_allreduce(cv, p, n, dt, op, stream);
     } else {
-      launch_twoshot_rs_push(cv, n, dt, stream);
+      launch_twoshot_rs_push(cv, n, dt, op, stream);
       launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
       launch_twoshot_copyout(cv, p, n * es, stream);
     }
@@ -259,7 +260,7 @@ void Communicator::all_gather(void* out, void const* in, size_t count_per_rank,
 
 void Communicator::reduce_scatter(void* out, void const* in,
                                   size_t count_per_rank, Dtype dt,
-                                  hipStream_t stream) {
+                                  hipStream_t stream, RedOp op) {
   tally(2, count_per_rank * dtype_size(dt) * world_);
   size_t const es = dtype_size(dt);
   if (world_ == 1) {
@@ -275,7 +276,8 @@ void Communicator::reduce_scatter(void* out, void const* in,
         static_cast<char const*>(in) - static_cast<char*>(heap_);
     CommView const cv = view(next_seq());
     launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
-    launch_reducescatter_sym(cv, uoff, out, count_per_rank, dt, stream);
+    launch_reducescatter_sym(cv, uoff, out, count_per_rank, dt, op,
+                             stream);
     launch_signal_wait(cv, cv.seq + 1, all_mask(world_), stream);
     return;
   }
@@ -293,7 +295,7 @@ void Communicator::reduce_scatter(void* out, void const* in,
     }
     launch_signal_wait(cv, cv.seq, all_mask(world_), stream);
     launch_reducescatter_pull(cv, static_cast<char*>(out) + off * es, n, dt,
-                              stream);
+                              op, stream);
   }
 }
 

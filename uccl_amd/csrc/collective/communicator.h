@@ -68,11 +68,12 @@ class Communicator {
   }
 
   // All ops are asynchronous on `stream` and in-place where natural.
-  void all_reduce(void* data, size_t count, Dtype dt, hipStream_t stream);
+  void all_reduce(void* data, size_t count, Dtype dt, hipStream_t stream,
+                  RedOp op = RedOp::kSum);
   void all_gather(void* out, void const* in, size_t count_per_rank, Dtype dt,
                   hipStream_t stream);
   void reduce_scatter(void* out, void const* in, size_t count_per_rank,
-                      Dtype dt, hipStream_t stream);
+                      Dtype dt, hipStream_t stream, RedOp op = RedOp::kSum);
   void broadcast(void* data, size_t count, Dtype dt, int root,
                  hipStream_t stream);
   void all_to_all(void* out, void const* in, size_t count_per_rank, Dtype dt,

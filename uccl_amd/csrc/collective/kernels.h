@@ -24,31 +24,41 @@ inline size_t dtype_size(Dtype d) {
   }
 }
 
+// Reduction op for the reducing collectives. Values match the device-side
+// red_apply OP template parameter (device/primitives.h). Avg/premulsum are
+// host-side compositions: scale + kSum (see launch_scale).
+enum class RedOp : int { kSum = 0, kProd = 1, kMin = 2, kMax = 3 };
+
 void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s);
+// in-place elementwise multiply by `factor` (avg = sum + scale 1/world;
+// premulsum = scale by the rank's scalar + sum)
+void launch_scale(void* data, size_t count, Dtype dt, double factor,
+                  hipStream_t s);
 void launch_oneshot_allreduce(const CommView& cv, void* out, size_t count,
-                              Dtype dt, hipStream_t s);
+                              Dtype dt, RedOp op, hipStream_t s);
 void launch_twoshot_rs_push(const CommView& cv, size_t count, Dtype dt,
-                            hipStream_t s);
+                            RedOp op, hipStream_t s);
 void launch_twoshot_copyout(const CommView& cv, void* out, size_t bytes,
                             hipStream_t s);
 void launch_twoshot_sym_rs(const CommView& cv, size_t uoff, size_t count,
-                           Dtype dt, hipStream_t s);
+                           Dtype dt, RedOp op, hipStream_t s);
 void launch_twoshot_sym_push(const CommView& cv, size_t uoff, size_t count,
                              Dtype dt, hipStream_t s);
 void launch_allgather_sym_push(const CommView& cv, void const* in,
                                size_t uoff, size_t slot_bytes,
                                hipStream_t s);
 void launch_reducescatter_sym(const CommView& cv, size_t uoff, void* out,
-                              size_t count, Dtype dt, hipStream_t s);
+                              size_t count, Dtype dt, RedOp op,
+                              hipStream_t s);
 void launch_alltoall_sym_push(const CommView& cv, void const* in,
                               size_t uoff, size_t chunk_bytes,
                               hipStream_t s);
 void launch_ll_allreduce(const CommView& cv, void const* in, void* out,
-                         size_t count, Dtype dt, hipStream_t s);
+                         size_t count, Dtype dt, RedOp op, hipStream_t s);
 void launch_allgather_pull(const CommView& cv, void* out, size_t chunk_bytes,
                            hipStream_t s);
 void launch_reducescatter_pull(const CommView& cv, void* out, size_t count,
-                               Dtype dt, hipStream_t s);
+                               Dtype dt, RedOp op, hipStream_t s);
 void launch_broadcast_pull(const CommView& cv, int root, void* out,
                            size_t bytes, hipStream_t s);
 void launch_alltoall_pull(const CommView& cv, void* out, size_t chunk_bytes,

@@ -22,6 +22,8 @@
 
 #include <hip/hip_runtime.h>
 
+#include <type_traits>
+
 #include "kernels.h"
 #include "../device/primitives.h"
 
@@ -93,7 +95,20 @@ __global__ void k_copy(void* __restrict__ dst, void const* __restrict__ src,
 // thread pulls the same offset from every peer's scratchA and reduces.
 // ---------------------------------------------------------------------------
 
-template <typename T>
+// Scalar-tail reduction helper shared by the reduce kernels: reduce element
+// j of each rank's T-typed region (regions indexed per rank via `reg`).
+template <typename T, int OP, typename RegionFn>
+__device__ __forceinline__ T tail_reduce(const CommView& cv, RegionFn reg,
+                                         size_t j) {
+  using A = typename TailAcc<T>::type;
+  A a = static_cast<A>(reinterpret_cast<T const*>(reg(0))[j]);
+  for (int p = 1; p < cv.world; ++p)
+    a = red_apply<OP>(a,
+                      static_cast<A>(reinterpret_cast<T const*>(reg(p))[j]));
+  return static_cast<T>(a);
+}
+
+template <typename T, int OP>
 __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
                                     size_t count) {
   size_t const vper = 16 / sizeof(T);
@@ -101,7 +116,7 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
-    AccumV16<T, 0> acc;
+    AccumV16<T, OP> acc;
     // fixed rank order => bitwise-identical results on every rank
     acc.init(nt_load(
         reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv)) + i));
@@ -116,40 +131,8 @@ __global__ void k_oneshot_allreduce(CommView cv, void* __restrict__ out,
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    typename TailAcc<T>::type a = 0;
-    for (int p = 0; p < cv.world; ++p)
-      a += static_cast<typename TailAcc<T>::type>(
-          reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv))[j]);
-    reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
-  }
-}
-
-// int specialization predates TailAcc and is now equivalent to the generic
-// template's int instantiation; kept as the validated int entry point.
-template <>
-__global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
-                                         size_t count) {
-  size_t const vper = 4;
-  size_t const nvec = count / vper;
-  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  size_t const stride = gridDim.x * blockDim.x;
-  for (; i < nvec; i += stride) {
-    AccumV16<int, 0> acc;
-    acc.init(nt_load(
-        reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv)) + i));
-    for (int p = 1; p < cv.world; ++p) {
-      acc.add(nt_load(
-          reinterpret_cast<V16 const*>(scratch_a(cv.peers[p], cv)) + i));
-    }
-    nt_store(reinterpret_cast<V16*>(out) + i, acc.pack());
-  }
-  size_t const tail = count - nvec * vper;
-  if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
-    size_t const j = nvec * vper + threadIdx.x;
-    int a = 0;
-    for (int p = 0; p < cv.world; ++p)
-      a += reinterpret_cast<int const*>(scratch_a(cv.peers[p], cv))[j];
-    reinterpret_cast<int*>(out)[j] = a;
+    reinterpret_cast<T*>(out)[j] = tail_reduce<T, OP>(
+        cv, [&](int p) { return scratch_a(cv.peers[p], cv); }, j);
   }
 }
 
@@ -163,7 +146,7 @@ __global__ void k_oneshot_allreduce<int>(CommView cv, void* __restrict__ out,
 // handled by the last shard owner.
 // ---------------------------------------------------------------------------
 
-template <typename T>
+template <typename T, int OP>
 __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   size_t const vper = 16 / sizeof(T);
   size_t const nvec = count / vper;
@@ -174,7 +157,7 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   size_t i = beg + blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < end; i += stride) {
-    AccumV16<T, 0> acc;
+    AccumV16<T, OP> acc;
     // fixed rank order => bitwise-identical results on every rank
     acc.init(nt_load(
         reinterpret_cast<V16 const*>(scratch_a(cv.peers[0], cv)) + i));
@@ -197,11 +180,8 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
   if (tail && cv.rank == cv.world - 1 && blockIdx.x == 0 &&
       threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    typename TailAcc<T>::type a = 0;
-    for (int p = 0; p < cv.world; ++p)
-      a += static_cast<typename TailAcc<T>::type>(
-          reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv))[j]);
-    T const r = static_cast<T>(a);
+    T const r = tail_reduce<T, OP>(
+        cv, [&](int p) { return scratch_a(cv.peers[p], cv); }, j);
     for (int p = 0; p < cv.world; ++p)
       reinterpret_cast<T*>(scratch_b(cv.peers[p], cv))[j] = r;
   }
@@ -216,7 +196,7 @@ __global__ void k_twoshot_rs_push(CommView cv, size_t count) {
 // into every rank's user buffer in place.
 // ---------------------------------------------------------------------------
 
-template <typename T>
+template <typename T, int OP>
 __global__ void k_twoshot_sym_rs(CommView cv, size_t uoff, size_t count) {
   size_t const vper = 16 / sizeof(T);
   size_t const nvec = count / vper;
@@ -227,7 +207,7 @@ __global__ void k_twoshot_sym_rs(CommView cv, size_t uoff, size_t count) {
   size_t const stride = gridDim.x * blockDim.x;
   auto* sb = reinterpret_cast<V16*>(scratch_b(cv.peers[cv.rank], cv));
   for (; i < end; i += stride) {
-    AccumV16<T, 0> acc;
+    AccumV16<T, OP> acc;
     acc.init(nt_load(reinterpret_cast<V16 const*>(
                  static_cast<char*>(cv.peers[0]) + uoff) + i));
 #pragma unroll 7
@@ -241,13 +221,11 @@ __global__ void k_twoshot_sym_rs(CommView cv, size_t uoff, size_t count) {
   if (tail && cv.rank == cv.world - 1 && blockIdx.x == 0 &&
       threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    typename TailAcc<T>::type a = 0;
-    for (int p = 0; p < cv.world; ++p)
-      a += static_cast<typename TailAcc<T>::type>(reinterpret_cast<T const*>(
-          static_cast<char*>(cv.peers[p]) + uoff)[j]);
     // stash tail results after the vector shard in scratchB
     reinterpret_cast<T*>(sb)[(end - beg) * vper + threadIdx.x] =
-        static_cast<T>(a);
+        tail_reduce<T, OP>(
+            cv, [&](int p) { return static_cast<char*>(cv.peers[p]) + uoff; },
+            j);
   }
 }
 
@@ -331,7 +309,7 @@ __global__ void k_alltoall_sym_push(CommView cv, void const* __restrict__ in,
 // Symmetric reduce_scatter: `in` symmetric [world*count]; out[i] =
 // sum_p in_p[rank*count + i]. Entry barrier (inputs published) before the
 // remote reads; exit barrier so callers may overwrite `in` afterwards.
-template <typename T>
+template <typename T, int OP>
 __global__ void k_reducescatter_sym(CommView cv, size_t uoff,
                                     void* __restrict__ out, size_t count) {
   size_t const vper = 16 / sizeof(T);
@@ -340,7 +318,7 @@ __global__ void k_reducescatter_sym(CommView cv, size_t uoff,
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
-    AccumV16<T, 0> acc;
+    AccumV16<T, OP> acc;
     acc.init(nt_load(reinterpret_cast<V16 const*>(
         reinterpret_cast<T const*>(static_cast<char*>(cv.peers[0]) + uoff) +
         elem_off) + i));
@@ -355,11 +333,14 @@ __global__ void k_reducescatter_sym(CommView cv, size_t uoff,
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    typename TailAcc<T>::type a = 0;
-    for (int p = 0; p < cv.world; ++p)
-      a += static_cast<typename TailAcc<T>::type>(reinterpret_cast<T const*>(
-          static_cast<char*>(cv.peers[p]) + uoff)[elem_off + j]);
-    reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
+    reinterpret_cast<T*>(out)[j] = tail_reduce<T, OP>(
+        cv,
+        [&](int p) {
+          return reinterpret_cast<T const*>(
+                     static_cast<char*>(cv.peers[p]) + uoff) +
+                 elem_off;
+        },
+        j);
   }
 }
 
@@ -388,7 +369,7 @@ __global__ void k_twoshot_copyout(CommView cv, void* __restrict__ out,
 // from all ranks out of its own LL region.
 // ---------------------------------------------------------------------------
 
-template <typename T>
+template <typename T, int OP>
 __global__ void k_ll_allreduce(CommView cv, void const* __restrict__ in,
                                void* __restrict__ out, size_t count) {
   uint32_t const flag = static_cast<uint32_t>(cv.seq);
@@ -426,25 +407,30 @@ __global__ void k_ll_allreduce(CommView cv, void const* __restrict__ in,
           reinterpret_cast<uint64_t const*>(
               ll_slot(cv.peers[cv.rank], parity, p)) + w,
           flag);
+      // p==0 initializes (prod/min/max have no cheap identity)
       if constexpr (sizeof(T) == 4) {
         if constexpr (__is_same(T, int)) {
-          acci[0] += static_cast<int>(d);
+          int const x = static_cast<int>(d);
+          acci[0] = p ? red_apply<OP>(acci[0], x) : x;
         } else {
-          accf[0] += __uint_as_float(d);
+          float const x = __uint_as_float(d);
+          accf[0] = p ? red_apply<OP>(accf[0], x) : x;
         }
       } else if constexpr (sizeof(T) == 2) {  // two lanes per word
         T lo, hi;
         reinterpret_cast<uint16_t&>(lo) = d & 0xffff;
         reinterpret_cast<uint16_t&>(hi) = d >> 16;
-        accf[0] += static_cast<float>(lo);
-        accf[1] += static_cast<float>(hi);
+        float const xl = static_cast<float>(lo), xh = static_cast<float>(hi);
+        accf[0] = p ? red_apply<OP>(accf[0], xl) : xl;
+        accf[1] = p ? red_apply<OP>(accf[1], xh) : xh;
       } else {  // 1-byte types (fp8): four lanes per word
 #pragma unroll
         for (int b = 0; b < 4; ++b) {
           T e;
           reinterpret_cast<uint8_t&>(e) =
               static_cast<uint8_t>((d >> (8 * b)) & 0xff);
-          accf[b] += static_cast<float>(e);
+          float const x = static_cast<float>(e);
+          accf[b] = p ? red_apply<OP>(accf[b], x) : x;
         }
       }
     }
@@ -503,7 +489,7 @@ __global__ void k_allgather_pull(CommView cv, void* __restrict__ out,
 }
 
 // out = sum over ranks of scratchA[p][rank*count .. +count] (my shard).
-template <typename T>
+template <typename T, int OP>
 __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
                                      size_t count) {
   size_t const vper = 16 / sizeof(T);
@@ -512,7 +498,7 @@ __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t const stride = gridDim.x * blockDim.x;
   for (; i < nvec; i += stride) {
-    AccumV16<T, 0> acc;
+    AccumV16<T, OP> acc;
     acc.init(nt_load(reinterpret_cast<V16 const*>(
         reinterpret_cast<T const*>(scratch_a(cv.peers[0], cv)) +
         elem_off) + i));
@@ -526,11 +512,13 @@ __global__ void k_reducescatter_pull(CommView cv, void* __restrict__ out,
   size_t const tail = count - nvec * vper;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
     size_t const j = nvec * vper + threadIdx.x;
-    typename TailAcc<T>::type a = 0;
-    for (int p = 0; p < cv.world; ++p)
-      a += static_cast<typename TailAcc<T>::type>(reinterpret_cast<T const*>(
-          scratch_a(cv.peers[p], cv))[elem_off + j]);
-    reinterpret_cast<T*>(out)[j] = static_cast<T>(a);
+    reinterpret_cast<T*>(out)[j] = tail_reduce<T, OP>(
+        cv,
+        [&](int p) {
+          return reinterpret_cast<T const*>(scratch_a(cv.peers[p], cv)) +
+                 elem_off;
+        },
+        j);
   }
 }
 
@@ -569,6 +557,18 @@ __global__ void k_alltoall_pull(CommView cv, void* __restrict__ out,
       reinterpret_cast<char*>(d)[chunk_bytes - tail + threadIdx.x] =
           reinterpret_cast<char const*>(s)[chunk_bytes - tail + threadIdx.x];
   }
+}
+
+// In-place elementwise scale (avg = sum + 1/world; premulsum = scale + sum).
+// Multiplies in the accumulator domain (fp32 / double) like the reducers.
+template <typename T>
+__global__ void k_scale(T* __restrict__ p, size_t count, double factor) {
+  using A = typename AccOf<T>::type;
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t const stride = gridDim.x * blockDim.x;
+  for (; i < count; i += stride)
+    p[i] = static_cast<T>(static_cast<A>(static_cast<A>(p[i]) *
+                                         static_cast<A>(factor)));
 }
 
 // ---------------------------------------------------------------------------
@@ -618,6 +618,23 @@ void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s) {
   k_copy<<<grid_for(bytes), 256, 0, s>>>(dst, src, bytes);
 }
 
+template <typename T>
+static void l_scale(void* data, size_t count, double factor, hipStream_t s) {
+  k_scale<T><<<grid_for(count * sizeof(T)), 256, 0, s>>>(
+      static_cast<T*>(data), count, factor);
+}
+
+// runtime RedOp -> compile-time OP template argument
+template <typename F>
+static inline void op_switch(RedOp op, F&& f) {
+  switch (op) {
+    case RedOp::kProd: f(std::integral_constant<int, 1>{}); break;
+    case RedOp::kMin: f(std::integral_constant<int, 2>{}); break;
+    case RedOp::kMax: f(std::integral_constant<int, 3>{}); break;
+    default: f(std::integral_constant<int, 0>{}); break;
+  }
+}
+
 #define DT_DISPATCH(dt, fn, ...)                        \
   switch (dt) {                                         \
     case Dtype::kF32: fn<float>(__VA_ARGS__); break;    \
@@ -631,50 +648,65 @@ void launch_copy(void* dst, void const* src, size_t bytes, hipStream_t s) {
   }
 
 template <typename T>
-static void l_oneshot(const CommView& cv, void* out, size_t count,
+static void l_oneshot(const CommView& cv, void* out, size_t count, RedOp op,
                       hipStream_t s) {
-  k_oneshot_allreduce<T><<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, out,
-                                                                     count);
+  op_switch(op, [&](auto o) {
+    k_oneshot_allreduce<T, decltype(o)::value>
+        <<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, out, count);
+  });
 }
 
 template <typename T>
-static void l_twoshot_rs(const CommView& cv, size_t count, hipStream_t s) {
-  k_twoshot_rs_push<T>
-      <<<grid_for(count * sizeof(T) / (cv.world ? cv.world : 1)), 256, 0, s>>>(
-          cv, count);
+static void l_twoshot_rs(const CommView& cv, size_t count, RedOp op,
+                         hipStream_t s) {
+  op_switch(op, [&](auto o) {
+    k_twoshot_rs_push<T, decltype(o)::value>
+        <<<grid_for(count * sizeof(T) / (cv.world ? cv.world : 1)), 256, 0,
+           s>>>(cv, count);
+  });
 }
 
 template <typename T>
 static void l_ll(const CommView& cv, void const* in, void* out, size_t count,
-                 hipStream_t s) {
-  k_ll_allreduce<T><<<grid_for(count * sizeof(T) * 2), 256, 0, s>>>(cv, in,
-                                                                    out,
-                                                                    count);
+                 RedOp op, hipStream_t s) {
+  op_switch(op, [&](auto o) {
+    k_ll_allreduce<T, decltype(o)::value>
+        <<<grid_for(count * sizeof(T) * 2), 256, 0, s>>>(cv, in, out, count);
+  });
 }
 
 template <typename T>
-static void l_rs_pull(const CommView& cv, void* out, size_t count,
+static void l_rs_pull(const CommView& cv, void* out, size_t count, RedOp op,
                       hipStream_t s) {
-  k_reducescatter_pull<T>
-      <<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, out, count);
+  op_switch(op, [&](auto o) {
+    k_reducescatter_pull<T, decltype(o)::value>
+        <<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, out, count);
+  });
+}
+
+void launch_scale(void* data, size_t count, Dtype dt, double factor,
+                  hipStream_t s) {
+  DT_DISPATCH(dt, l_scale, data, count, factor, s);
 }
 
 void launch_oneshot_allreduce(const CommView& cv, void* out, size_t count,
-                              Dtype dt, hipStream_t s) {
-  DT_DISPATCH(dt, l_oneshot, cv, out, count, s);
+                              Dtype dt, RedOp op, hipStream_t s) {
+  DT_DISPATCH(dt, l_oneshot, cv, out, count, op, s);
 }
 
 void launch_twoshot_rs_push(const CommView& cv, size_t count, Dtype dt,
-                            hipStream_t s) {
-  DT_DISPATCH(dt, l_twoshot_rs, cv, count, s);
+                            RedOp op, hipStream_t s) {
+  DT_DISPATCH(dt, l_twoshot_rs, cv, count, op, s);
 }
 
 template <typename T>
-static void l_sym_rs(const CommView& cv, size_t uoff, size_t count,
+static void l_sym_rs(const CommView& cv, size_t uoff, size_t count, RedOp op,
                      hipStream_t s) {
-  k_twoshot_sym_rs<T>
-      <<<grid_for(count * sizeof(T) / (cv.world ? cv.world : 1)), 256, 0,
-         s>>>(cv, uoff, count);
+  op_switch(op, [&](auto o) {
+    k_twoshot_sym_rs<T, decltype(o)::value>
+        <<<grid_for(count * sizeof(T) / (cv.world ? cv.world : 1)), 256, 0,
+           s>>>(cv, uoff, count);
+  });
 }
 
 template <typename T>
@@ -686,8 +718,8 @@ static void l_sym_push(const CommView& cv, size_t uoff, size_t count,
 }
 
 void launch_twoshot_sym_rs(const CommView& cv, size_t uoff, size_t count,
-                           Dtype dt, hipStream_t s) {
-  DT_DISPATCH(dt, l_sym_rs, cv, uoff, count, s);
+                           Dtype dt, RedOp op, hipStream_t s) {
+  DT_DISPATCH(dt, l_sym_rs, cv, uoff, count, op, s);
 }
 
 void launch_allgather_sym_push(const CommView& cv, void const* in,
@@ -706,14 +738,17 @@ void launch_alltoall_sym_push(const CommView& cv, void const* in,
 
 template <typename T>
 static void l_rs_sym2(const CommView& cv, size_t uoff, void* out,
-                      size_t count, hipStream_t s) {
-  k_reducescatter_sym<T>
-      <<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, uoff, out, count);
+                      size_t count, RedOp op, hipStream_t s) {
+  op_switch(op, [&](auto o) {
+    k_reducescatter_sym<T, decltype(o)::value>
+        <<<grid_for(count * sizeof(T)), 256, 0, s>>>(cv, uoff, out, count);
+  });
 }
 
 void launch_reducescatter_sym(const CommView& cv, size_t uoff, void* out,
-                              size_t count, Dtype dt, hipStream_t s) {
-  DT_DISPATCH(dt, l_rs_sym2, cv, uoff, out, count, s);
+                              size_t count, Dtype dt, RedOp op,
+                              hipStream_t s) {
+  DT_DISPATCH(dt, l_rs_sym2, cv, uoff, out, count, op, s);
 }
 
 void launch_twoshot_sym_push(const CommView& cv, size_t uoff, size_t count,
@@ -727,8 +762,8 @@ void launch_twoshot_copyout(const CommView& cv, void* out, size_t bytes,
 }
 
 void launch_ll_allreduce(const CommView& cv, void const* in, void* out,
-                         size_t count, Dtype dt, hipStream_t s) {
-  DT_DISPATCH(dt, l_ll, cv, in, out, count, s);
+                         size_t count, Dtype dt, RedOp op, hipStream_t s) {
+  DT_DISPATCH(dt, l_ll, cv, in, out, count, op, s);
 }
 
 void launch_allgather_pull(const CommView& cv, void* out, size_t chunk_bytes,
@@ -738,8 +773,8 @@ void launch_allgather_pull(const CommView& cv, void* out, size_t chunk_bytes,
 }
 
 void launch_reducescatter_pull(const CommView& cv, void* out, size_t count,
-                               Dtype dt, hipStream_t s) {
-  DT_DISPATCH(dt, l_rs_pull, cv, out, count, s);
+                               Dtype dt, RedOp op, hipStream_t s) {
+  DT_DISPATCH(dt, l_rs_pull, cv, out, count, op, s);
 }
 
 void launch_broadcast_pull(const CommView& cv, int root, void* out,

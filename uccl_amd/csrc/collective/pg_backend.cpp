@@ -126,19 +126,27 @@ class UcclBackend : public c10d::Backend {
     }
   }
 
+
+  static RedOp to_redop(const c10d::ReduceOp& op) {
+    if (op == c10d::ReduceOp::SUM || op == c10d::ReduceOp::AVG)
+      return RedOp::kSum;  // AVG = SUM + post-scale at the call site
+    if (op == c10d::ReduceOp::PRODUCT) return RedOp::kProd;
+    if (op == c10d::ReduceOp::MIN) return RedOp::kMin;
+    if (op == c10d::ReduceOp::MAX) return RedOp::kMax;
+    TORCH_CHECK(false, "uccl: unsupported reduce op");
+  }
+
   const std::string getBackendName() const override { return "uccl"; }
 
   c10::intrusive_ptr<c10d::Work> allreduce(
       std::vector<at::Tensor>& tensors,
       const c10d::AllreduceOptions& opts) override {
     TORCH_CHECK(tensors.size() == 1, "uccl: one tensor per op");
-    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM ||
-                    opts.reduceOp == c10d::ReduceOp::AVG,
-                "uccl: SUM/AVG allreduce only");
     auto& t = tensors[0];
     check(t);
     hipStream_t s = cur_stream(comm_->device());
-    comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s);
+    comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s,
+                      to_redop(opts.reduceOp));
     if (opts.reduceOp == c10d::ReduceOp::AVG)
       t.mul_(1.0 / getSize());  // stream-ordered on the current stream
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::ALLREDUCE, s,
@@ -195,15 +203,13 @@ class UcclBackend : public c10d::Backend {
   c10::intrusive_ptr<c10d::Work> _reduce_scatter_base(
       at::Tensor& output, at::Tensor& input,
       const c10d::ReduceScatterOptions& opts) override {
-    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM ||
-                    opts.reduceOp == c10d::ReduceOp::AVG,
-                "uccl: SUM/AVG reduce_scatter only");
     check(input);
     check(output);
     TORCH_CHECK(input.numel() == output.numel() * getSize());
     hipStream_t s = cur_stream(comm_->device());
     comm_->reduce_scatter(output.data_ptr(), input.data_ptr(),
-                          output.numel(), to_dtype(input), s);
+                          output.numel(), to_dtype(input), s,
+                          to_redop(opts.reduceOp));
     if (opts.reduceOp == c10d::ReduceOp::AVG)
       output.mul_(1.0 / getSize());
     return c10::make_intrusive<UcclWork>(
@@ -237,12 +243,13 @@ class UcclBackend : public c10d::Backend {
     // as a SUM allreduce (non-root tensors also end up reduced, which
     // the torch.distributed contract permits).
     TORCH_CHECK(tensors.size() == 1);
-    TORCH_CHECK(opts.reduceOp == c10d::ReduceOp::SUM,
-                "uccl: only SUM reduce");
     auto& t = tensors[0];
     check(t);
     hipStream_t s = cur_stream(comm_->device());
-    comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s);
+    comm_->all_reduce(t.data_ptr(), t.numel(), to_dtype(t), s,
+                      to_redop(opts.reduceOp));
+    if (opts.reduceOp == c10d::ReduceOp::AVG)
+      t.mul_(1.0 / getSize());
     return c10::make_intrusive<UcclWork>(getRank(), c10d::OpType::REDUCE, s,
                                          comm_->device(),
                                          tensors);

@@ -133,6 +133,21 @@ __device__ __forceinline__ void nt_store(V16* p, V16 v) {
                               reinterpret_cast<v4u*>(p));
 }
 
+// ---------------------------------------------------------------------------
+// Reduction ops. OP values match uccl::RedOp host-side (kernels.h):
+// 0=sum 1=prod 2=min 3=max. Applied in the accumulator domain (fp32 for the
+// 16-bit/8-bit float types, native for int/i64/f64), so min/max are exact
+// and prod rounds once at pack.
+// ---------------------------------------------------------------------------
+
+template <int OP, typename A>
+__device__ __forceinline__ A red_apply(A a, A b) {
+  if constexpr (OP == 1) return a * b;
+  else if constexpr (OP == 2) return a < b ? a : b;
+  else if constexpr (OP == 3) return a > b ? a : b;
+  else return a + b;
+}
+
 // Elementwise fp32-accumulate add of two 16B vectors of T.
 template <typename T>
 __device__ __forceinline__ V16 v16_add(V16 a, V16 b);
@@ -174,133 +189,52 @@ __device__ __forceinline__ V16 v16_add<int>(V16 a, V16 b) {
   return r;
 }
 
-// fp32-accumulator variant: unpack T into float lanes, accumulate exactly,
-// pack once at the end (used by the reduce kernels so bf16 sums don't lose
-// bits per-step with world_size up to 8).
-template <typename T, int N>
-struct AccumV16;
-
-template <int N>
-struct AccumV16<float, N> {
-  float v[4];
-  __device__ __forceinline__ void init(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) v[i] = a.f32[i];
-  }
-  __device__ __forceinline__ void add(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) v[i] += a.f32[i];
-  }
-  __device__ __forceinline__ V16 pack() const {
-    V16 r;
-#pragma unroll
-    for (int i = 0; i < 4; ++i) r.f32[i] = v[i];
-    return r;
-  }
+// Accumulator lane type per element type: fp32 for the small float types
+// (exact sums for world ≤ 8, one rounding at pack; min/max exact because
+// every bf16/f16/fp8 value is representable in fp32), native for the rest.
+template <typename T>
+struct AccOf {
+  using type = float;
+};
+template <>
+struct AccOf<int> {
+  using type = int;
+};
+template <>
+struct AccOf<long long> {
+  using type = long long;
+};
+template <>
+struct AccOf<double> {
+  using type = double;
 };
 
-template <int N>
-struct AccumV16<__hip_bfloat16, N> {
-  float v[8];
+// fp32-accumulator reduction over 16B vectors: unpack T lanes into the
+// accumulator domain, apply OP per lane, pack once at the end (so bf16
+// sums don't lose bits per-step with world_size up to 8). OP: see
+// red_apply. All T here have a float conversion operator + a
+// from-float constructor (fp8 pack saturates).
+template <typename T, int OP = 0>
+struct AccumV16 {
+  using A = typename AccOf<T>::type;
+  static constexpr int kN = 16 / sizeof(T);
+  A v[kN];
   __device__ __forceinline__ void init(V16 a) {
+    T const* e = reinterpret_cast<T const*>(&a);
 #pragma unroll
-    for (int i = 0; i < 8; ++i) v[i] = __bfloat162float(a.bf16[i]);
+    for (int i = 0; i < kN; ++i) v[i] = static_cast<A>(e[i]);
   }
   __device__ __forceinline__ void add(V16 a) {
+    T const* e = reinterpret_cast<T const*>(&a);
 #pragma unroll
-    for (int i = 0; i < 8; ++i) v[i] += __bfloat162float(a.bf16[i]);
+    for (int i = 0; i < kN; ++i)
+      v[i] = red_apply<OP>(v[i], static_cast<A>(e[i]));
   }
   __device__ __forceinline__ V16 pack() const {
     V16 r;
+    T* e = reinterpret_cast<T*>(&r);
 #pragma unroll
-    for (int i = 0; i < 8; ++i) r.bf16[i] = __float2bfloat16(v[i]);
-    return r;
-  }
-};
-
-template <int N>
-struct AccumV16<__half, N> {
-  float v[8];
-  __device__ __forceinline__ void init(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 8; ++i) v[i] = __half2float(a.f16[i]);
-  }
-  __device__ __forceinline__ void add(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 8; ++i) v[i] += __half2float(a.f16[i]);
-  }
-  __device__ __forceinline__ V16 pack() const {
-    V16 r;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) r.f16[i] = __float2half(v[i]);
-    return r;
-  }
-};
-
-// OCP fp8 e4m3 (gfx950-native, NOT the MI300X fnuz variant): 16 elems per
-// 16B vector, fp32 accumulation, saturating pack.
-template <int N>
-struct AccumV16<__hip_fp8_e4m3, N> {
-  float v[16];
-  __device__ __forceinline__ void init(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 16; ++i)
-      v[i] = static_cast<float>(
-          reinterpret_cast<__hip_fp8_e4m3 const*>(&a)[i]);
-  }
-  __device__ __forceinline__ void add(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 16; ++i)
-      v[i] += static_cast<float>(
-          reinterpret_cast<__hip_fp8_e4m3 const*>(&a)[i]);
-  }
-  __device__ __forceinline__ V16 pack() const {
-    V16 r;
-#pragma unroll
-    for (int i = 0; i < 16; ++i)
-      reinterpret_cast<__hip_fp8_e4m3*>(&r)[i] = __hip_fp8_e4m3(v[i]);
-    return r;
-  }
-};
-
-template <int N>
-struct AccumV16<long long, N> {
-  long long v[2];
-  __device__ __forceinline__ void init(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-      v[i] = reinterpret_cast<long long const*>(&a)[i];
-  }
-  __device__ __forceinline__ void add(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-      v[i] += reinterpret_cast<long long const*>(&a)[i];
-  }
-  __device__ __forceinline__ V16 pack() const {
-    V16 r;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) reinterpret_cast<long long*>(&r)[i] = v[i];
-    return r;
-  }
-};
-
-template <int N>
-struct AccumV16<double, N> {
-  double v[2];
-  __device__ __forceinline__ void init(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-      v[i] = reinterpret_cast<double const*>(&a)[i];
-  }
-  __device__ __forceinline__ void add(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-      v[i] += reinterpret_cast<double const*>(&a)[i];
-  }
-  __device__ __forceinline__ V16 pack() const {
-    V16 r;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) reinterpret_cast<double*>(&r)[i] = v[i];
+    for (int i = 0; i < kN; ++i) e[i] = static_cast<T>(v[i]);
     return r;
   }
 };
@@ -321,25 +255,6 @@ struct TailAcc<long long> {
 template <>
 struct TailAcc<double> {
   using type = double;
-};
-
-template <int N>
-struct AccumV16<int, N> {
-  int v[4];
-  __device__ __forceinline__ void init(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) v[i] = reinterpret_cast<int const*>(a.f32)[i];
-  }
-  __device__ __forceinline__ void add(V16 a) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) v[i] += reinterpret_cast<int const*>(a.f32)[i];
-  }
-  __device__ __forceinline__ V16 pack() const {
-    V16 r;
-#pragma unroll
-    for (int i = 0; i < 4; ++i) reinterpret_cast<int*>(r.f32)[i] = v[i];
-    return r;
-  }
 };
 
 }  // namespace device

@@ -10,12 +10,16 @@
 // NCCL-linked apps expect; LD_PRELOAD (or plain linking) swaps the data
 // plane.
 //
-// Scope: single node (up to 8 ranks over xGMI), SUM reductions, the
-// dtypes the engine supports. Everything else returns a clean error.
+// Scope: single node (up to 8 ranks over xGMI), sum/prod/min/max/avg and
+// premulsum reductions, the dtypes the engine supports, NCCL group
+// semantics for p2p (deferred + per-peer streams so group{send;recv}
+// pairs >2MB can't deadlock on the slot-credit pacing). Everything else
+// returns a clean error.
 
 #include <rccl/rccl.h>
 
 #include <atomic>
+#include <cmath>
 #include <condition_variable>
 #include <cstring>
 #include <map>
@@ -110,10 +114,110 @@ void serve(int listen_fd) {
   }
 }
 
+struct PreMulScalar {
+  bool device_resident = false;
+  void const* dev_ptr = nullptr;  // ncclScalarDevice: read at execution
+  double value = 1.0;             // ncclScalarHostImmediate: copied at create
+  ncclDataType_t dtype = ncclFloat32;
+};
+
 struct UcclComm {
   std::unique_ptr<Communicator> comm;
   int device = 0;
+  std::mutex mu;
+  // per-(peer, dir) streams for grouped p2p: ops to the same peer+direction
+  // stay ordered, ops to different peers run concurrently so the slot-ack
+  // pacing of a send can't block the recv that would unblock it.
+  std::map<std::pair<int, int>, hipStream_t> p2p_streams;
+  std::vector<PreMulScalar> premul;  // ncclRedOpCreatePreMulSum handles
+  hipStream_t p2p_stream(int peer, bool is_send) {
+    std::lock_guard<std::mutex> g(mu);
+    auto key = std::make_pair(peer, is_send ? 1 : 0);
+    auto it = p2p_streams.find(key);
+    if (it != p2p_streams.end()) return it->second;
+    hipStream_t s = nullptr;
+    (void)hipStreamCreateWithFlags(&s, hipStreamNonBlocking);
+    p2p_streams[key] = s;
+    return s;
+  }
+  ~UcclComm() {
+    for (auto& [k, s] : p2p_streams) {
+      (void)hipStreamSynchronize(s);
+      (void)hipStreamDestroy(s);
+    }
+  }
 };
+
+// --- group semantics -------------------------------------------------------
+// NCCL requires ops inside ncclGroupStart/End to be able to progress
+// concurrently. Collectives are stream-async on the engine already; p2p ops
+// are deferred here and re-issued on per-(peer,dir) streams at GroupEnd,
+// bracketed by events so the user stream observes them in order.
+
+struct PendingP2P {
+  UcclComm* uc;
+  bool is_send;
+  void* buff;
+  size_t bytes;
+  int peer;
+  hipStream_t stream;
+};
+
+thread_local int g_group_depth = 0;
+thread_local std::vector<PendingP2P> g_pending;
+
+ncclResult_t flush_group() {
+  // self-pairs (rank sending to itself) become plain DtoD copies
+  std::vector<PendingP2P> ops;
+  ops.swap(g_pending);
+  std::vector<size_t> self_sends, self_recvs;
+  for (size_t i = 0; i < ops.size(); ++i) {
+    auto& o = ops[i];
+    if (o.peer == o.uc->comm->rank()) {
+      (o.is_send ? self_sends : self_recvs).push_back(i);
+    }
+  }
+  if (self_sends.size() != self_recvs.size()) return ncclInvalidUsage;
+  for (size_t k = 0; k < self_sends.size(); ++k) {
+    auto& s = ops[self_sends[k]];
+    auto& r = ops[self_recvs[k]];
+    if (s.bytes != r.bytes) return ncclInvalidUsage;
+    (void)hipMemcpyAsync(r.buff, s.buff, s.bytes, hipMemcpyDeviceToDevice,
+                         s.stream);
+    if (r.stream != s.stream) {
+      hipEvent_t e;
+      (void)hipEventCreateWithFlags(&e, hipEventDisableTiming);
+      (void)hipEventRecord(e, s.stream);
+      (void)hipStreamWaitEvent(r.stream, e, 0);
+      (void)hipEventDestroy(e);
+    }
+  }
+  for (auto& o : ops) {
+    if (o.peer == o.uc->comm->rank()) continue;
+    hipStream_t ps = o.uc->p2p_stream(o.peer, o.is_send);
+    hipEvent_t in, out;
+    (void)hipEventCreateWithFlags(&in, hipEventDisableTiming);
+    (void)hipEventCreateWithFlags(&out, hipEventDisableTiming);
+    (void)hipEventRecord(in, o.stream);
+    (void)hipStreamWaitEvent(ps, in, 0);
+    try {
+      if (o.is_send)
+        o.uc->comm->send(o.buff, o.bytes, o.peer, ps);
+      else
+        o.uc->comm->recv(o.buff, o.bytes, o.peer, ps);
+    } catch (std::exception const& e) {
+      UCCL_LOG_ERROR << "grouped p2p failed: " << e.what();
+      (void)hipEventDestroy(in);
+      (void)hipEventDestroy(out);
+      return ncclInternalError;
+    }
+    (void)hipEventRecord(out, ps);
+    (void)hipStreamWaitEvent(o.stream, out, 0);
+    (void)hipEventDestroy(in);
+    (void)hipEventDestroy(out);
+  }
+  return ncclSuccess;
+}
 
 ncclResult_t to_dtype(ncclDataType_t t, Dtype* out) {
   switch (t) {
@@ -122,6 +226,71 @@ ncclResult_t to_dtype(ncclDataType_t t, Dtype* out) {
     case ncclBfloat16: *out = Dtype::kBF16; return ncclSuccess;
     case ncclInt32: *out = Dtype::kI32; return ncclSuccess;
     case ncclInt8: *out = Dtype::kU8; return ncclSuccess;  // copy ops only
+    case ncclInt64: *out = Dtype::kI64; return ncclSuccess;
+    case ncclFloat64: *out = Dtype::kF64; return ncclSuccess;
+    default: return ncclInvalidArgument;
+  }
+}
+
+bool is_float_dtype(Dtype d) {
+  return d == Dtype::kF32 || d == Dtype::kF16 || d == Dtype::kBF16 ||
+         d == Dtype::kF64 || d == Dtype::kF8E4M3;
+}
+
+// Decompose an ncclRedOp into {engine op, pre-scale, post-scale}:
+//   avg       = sum, post-scale 1/world
+//   premulsum = pre-scale by this rank's scalar, sum
+struct OpPlan {
+  uccl::RedOp op = uccl::RedOp::kSum;
+  bool prescale = false;
+  double prescale_by = 1.0;
+  bool postscale = false;
+  double postscale_by = 1.0;
+};
+
+ncclResult_t plan_op(UcclComm* uc, ncclRedOp_t op, Dtype dt, OpPlan* plan) {
+  *plan = OpPlan{};
+  if (op >= ncclNumOps) {
+    size_t const idx = static_cast<size_t>(op) - ncclNumOps;
+    std::lock_guard<std::mutex> g(uc->mu);
+    if (idx >= uc->premul.size()) return ncclInvalidArgument;
+    PreMulScalar const& s = uc->premul[idx];
+    double v = s.value;
+    if (s.device_resident) {
+      // ncclScalarDevice: the scalar lives in device memory and is read at
+      // execution time; small sync copy here keeps the semantics.
+      if (s.dtype == ncclFloat32) {
+        float f = 1.f;
+        (void)hipMemcpy(&f, s.dev_ptr, sizeof(f), hipMemcpyDeviceToHost);
+        v = f;
+      } else if (s.dtype == ncclFloat64) {
+        (void)hipMemcpy(&v, s.dev_ptr, sizeof(v), hipMemcpyDeviceToHost);
+      } else if (s.dtype == ncclFloat16) {
+        uint16_t h = 0;
+        (void)hipMemcpy(&h, s.dev_ptr, sizeof(h), hipMemcpyDeviceToHost);
+        // fp16 -> double on host
+        int const sign = (h >> 15) & 1, exp = (h >> 10) & 0x1f,
+                  man = h & 0x3ff;
+        double m = exp == 0 ? man / 1024.0 / 16384.0
+                            : (1.0 + man / 1024.0) * std::pow(2.0, exp - 15);
+        v = sign ? -m : m;
+      } else {
+        return ncclInvalidArgument;
+      }
+    }
+    plan->prescale = true;
+    plan->prescale_by = v;
+    return ncclSuccess;
+  }
+  switch (op) {
+    case ncclSum: return ncclSuccess;
+    case ncclProd: plan->op = uccl::RedOp::kProd; return ncclSuccess;
+    case ncclMin: plan->op = uccl::RedOp::kMin; return ncclSuccess;
+    case ncclMax: plan->op = uccl::RedOp::kMax; return ncclSuccess;
+    case ncclAvg:
+      if (!is_float_dtype(dt)) return ncclInvalidUsage;
+      plan->postscale = true;
+      return ncclSuccess;  // postscale_by filled by caller (1/world)
     default: return ncclInvalidArgument;
   }
 }
@@ -239,25 +408,41 @@ ncclResult_t ncclCommGetAsyncError(ncclComm_t, ncclResult_t* asyncError) {
   return ncclSuccess;
 }
 
-// group semantics: ops are independently stream-async; nothing to defer
-ncclResult_t ncclGroupStart(void) { return ncclSuccess; }
-ncclResult_t ncclGroupEnd(void) { return ncclSuccess; }
+// Group semantics: collectives are independently stream-async; p2p ops are
+// deferred and re-issued on per-(peer,dir) streams at GroupEnd (see
+// flush_group) so paired send/recv >2MB can't deadlock on slot credits.
+ncclResult_t ncclGroupStart(void) {
+  ++g_group_depth;
+  return ncclSuccess;
+}
+
+ncclResult_t ncclGroupEnd(void) {
+  if (g_group_depth <= 0) return ncclInvalidUsage;
+  if (--g_group_depth == 0 && !g_pending.empty()) return flush_group();
+  return ncclSuccess;
+}
 
 ncclResult_t ncclAllReduce(const void* sendbuff, void* recvbuff,
                            size_t count, ncclDataType_t datatype,
                            ncclRedOp_t op, ncclComm_t comm,
                            hipStream_t stream) {
-  if (op != ncclSum) return ncclInvalidUsage;
   Dtype dt;
   if (to_dtype(datatype, &dt) != ncclSuccess || dt == Dtype::kU8)
     return ncclInvalidArgument;
+  OpPlan plan;
+  ncclResult_t r = plan_op(COMM(comm), op, dt, &plan);
+  if (r != ncclSuccess) return r;
   auto& c = *COMM(comm)->comm;
   if (sendbuff != recvbuff) {
     size_t bytes = count * uccl::dtype_size(dt);
     (void)hipMemcpyAsync(recvbuff, sendbuff, bytes,
                          hipMemcpyDeviceToDevice, stream);
   }
-  c.all_reduce(recvbuff, count, dt, stream);
+  if (plan.prescale)
+    uccl::launch_scale(recvbuff, count, dt, plan.prescale_by, stream);
+  c.all_reduce(recvbuff, count, dt, stream, plan.op);
+  if (plan.postscale)
+    uccl::launch_scale(recvbuff, count, dt, 1.0 / c.world(), stream);
   return ncclSuccess;
 }
 
@@ -297,12 +482,19 @@ ncclResult_t ncclReduceScatter(const void* sendbuff, void* recvbuff,
                                size_t recvcount, ncclDataType_t datatype,
                                ncclRedOp_t op, ncclComm_t comm,
                                hipStream_t stream) {
-  if (op != ncclSum) return ncclInvalidUsage;
   Dtype dt;
   if (to_dtype(datatype, &dt) != ncclSuccess || dt == Dtype::kU8)
     return ncclInvalidArgument;
-  COMM(comm)->comm->reduce_scatter(recvbuff, sendbuff, recvcount, dt,
-                                   stream);
+  OpPlan plan;
+  ncclResult_t r = plan_op(COMM(comm), op, dt, &plan);
+  if (r != ncclSuccess) return r;
+  // premulsum would need to scale the (larger, caller-owned) send buffer;
+  // not supported on this entry point.
+  if (plan.prescale) return ncclInvalidUsage;
+  auto& c = *COMM(comm)->comm;
+  c.reduce_scatter(recvbuff, sendbuff, recvcount, dt, stream, plan.op);
+  if (plan.postscale)
+    uccl::launch_scale(recvbuff, recvcount, dt, 1.0 / c.world(), stream);
   return ncclSuccess;
 }
 
@@ -310,7 +502,22 @@ ncclResult_t ncclReduce(const void* sendbuff, void* recvbuff, size_t count,
                         ncclDataType_t datatype, ncclRedOp_t op, int root,
                         ncclComm_t comm, hipStream_t stream) {
   // allreduce everywhere; the root's buffer ends up with the reduction
-  // (non-root recvbuffs are also reduced, which the API permits to vary)
+  // (non-root recvbuffs are also reduced, which the API permits to vary).
+  // NCCL allows recvbuff == nullptr on non-roots: reduce into a scratch
+  // buffer there (stream-ordered alloc/free).
+  if (recvbuff == nullptr) {
+    if (COMM(comm)->comm->rank() == root) return ncclInvalidArgument;
+    Dtype dt;
+    if (to_dtype(datatype, &dt) != ncclSuccess) return ncclInvalidArgument;
+    void* tmp = nullptr;
+    if (hipMallocAsync(&tmp, count * uccl::dtype_size(dt), stream) !=
+        hipSuccess)
+      return ncclUnhandledCudaError;
+    ncclResult_t r =
+        ncclAllReduce(sendbuff, tmp, count, datatype, op, comm, stream);
+    (void)hipFreeAsync(tmp, stream);
+    return r;
+  }
   return ncclAllReduce(sendbuff, recvbuff, count, datatype, op, comm,
                        stream);
 }
@@ -320,8 +527,14 @@ ncclResult_t ncclSend(const void* sendbuff, size_t count,
                       hipStream_t stream) {
   Dtype dt;
   if (to_dtype(datatype, &dt) != ncclSuccess) return ncclInvalidArgument;
-  COMM(comm)->comm->send(sendbuff, count * uccl::dtype_size(dt), peer,
-                         stream);
+  size_t const bytes = count * uccl::dtype_size(dt);
+  if (g_group_depth > 0) {
+    g_pending.push_back(PendingP2P{COMM(comm), true,
+                                   const_cast<void*>(sendbuff), bytes, peer,
+                                   stream});
+    return ncclSuccess;
+  }
+  COMM(comm)->comm->send(sendbuff, bytes, peer, stream);
   return ncclSuccess;
 }
 
@@ -330,9 +543,45 @@ ncclResult_t ncclRecv(void* recvbuff, size_t count,
                       hipStream_t stream) {
   Dtype dt;
   if (to_dtype(datatype, &dt) != ncclSuccess) return ncclInvalidArgument;
-  COMM(comm)->comm->recv(recvbuff, count * uccl::dtype_size(dt), peer,
-                         stream);
+  size_t const bytes = count * uccl::dtype_size(dt);
+  if (g_group_depth > 0) {
+    g_pending.push_back(
+        PendingP2P{COMM(comm), false, recvbuff, bytes, peer, stream});
+    return ncclSuccess;
+  }
+  COMM(comm)->comm->recv(recvbuff, bytes, peer, stream);
   return ncclSuccess;
+}
+
+// premulsum: result = sum_r (scalar_r * x_r); each rank pre-scales its own
+// contribution (plan_op) before the sum.
+ncclResult_t ncclRedOpCreatePreMulSum(ncclRedOp_t* op, void* scalar,
+                                      ncclDataType_t datatype,
+                                      ncclScalarResidence_t residence,
+                                      ncclComm_t comm) {
+  if (!op || !scalar) return ncclInvalidArgument;
+  PreMulScalar s;
+  s.dtype = datatype;
+  if (residence == ncclScalarDevice) {
+    s.device_resident = true;
+    s.dev_ptr = scalar;
+  } else {
+    switch (datatype) {
+      case ncclFloat32: s.value = *static_cast<float*>(scalar); break;
+      case ncclFloat64: s.value = *static_cast<double*>(scalar); break;
+      default: return ncclInvalidArgument;  // host-immediate fp16 unused
+    }
+  }
+  auto* uc = COMM(comm);
+  std::lock_guard<std::mutex> g(uc->mu);
+  uc->premul.push_back(s);
+  *op = static_cast<ncclRedOp_t>(ncclNumOps + uc->premul.size() - 1);
+  return ncclSuccess;
+}
+
+ncclResult_t ncclRedOpDestroy(ncclRedOp_t op, ncclComm_t comm) {
+  (void)comm;
+  return op >= ncclNumOps ? ncclSuccess : ncclInvalidArgument;
 }
 
 // RCCL extension, also provided by the reference's drop-in

@@ -219,12 +219,19 @@ size_t decompress(void const* frame, size_t frame_bytes, void* out,
     throw std::runtime_error("compress: bad frame");
   int const strategy = p[5];
   int const elem_size = p[6];
+  // elem_size and nplanes are wire-controlled and feed a division and the
+  // interleave store stride below: reject anything but the shapes the
+  // encoder emits before they can divide by zero or write past out_cap.
+  if (elem_size != 1 && elem_size != 2 && elem_size != 4)
+    throw std::runtime_error("compress: bad elem size");
   size_t const bytes = get_u64(p + 8);
   if (bytes > out_cap) throw std::runtime_error("compress: output too small");
   size_t off = 24;
   int const nplanes = p[off++];
   if (nplanes < 1 || nplanes > 8 || frame_bytes < off + nplanes * 9)
     throw std::runtime_error("compress: bad plane table");
+  if (strategy == kNone ? nplanes != 1 : nplanes != elem_size)
+    throw std::runtime_error("compress: plane count mismatch");
   std::vector<PlaneMeta> meta(nplanes);
   for (int pl = 0; pl < nplanes; ++pl) {
     meta[pl].encoding = p[off];

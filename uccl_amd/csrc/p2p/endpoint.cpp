@@ -255,8 +255,9 @@ Endpoint::~Endpoint() {
   ::shutdown(listen_fd_, SHUT_RDWR);
   ::close(listen_fd_);
   task_cv_.notify_all();
-  for (auto& w : workers_)
-    if (w.joinable()) w.join();
+  // Kill connections BEFORE joining workers: a worker blocked in
+  // do_recv/do_write on a cv would otherwise never wake and the join
+  // would hang forever (advisor r1).
   {
     std::lock_guard<std::mutex> g(conn_mu_);
     for (auto& [id, c] : conns_) {
@@ -273,6 +274,8 @@ Endpoint::~Endpoint() {
       c->ord_cv.notify_all();
     }
   }
+  for (auto& w : workers_)
+    if (w.joinable()) w.join();
   if (listener_.joinable()) listener_.join();
   if (tp_acceptor_.joinable()) tp_acceptor_.join();
   {
@@ -569,7 +572,9 @@ void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
     // wait for receiver's copy (src must stay valid until then)
     std::unique_lock<std::mutex> lk(c.tok_mu);
     c.tok_cv.wait(lk, [&] { return c.completed.count(token) || !c.alive; });
+    bool const completed = c.completed.count(token) != 0;
     c.completed.erase(token);
+    UCCL_CHECK(completed) << "connection died before IPC send completed";
     return;
   }
   if (!is_gpu(device)) {
@@ -682,7 +687,9 @@ void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
   }
   std::unique_lock<std::mutex> lk(c.tok_mu);
   c.tok_cv.wait(lk, [&] { return c.completed.count(token) || !c.alive; });
+  bool const completed = c.completed.count(token) != 0;
   c.completed.erase(token);
+  UCCL_CHECK(completed) << "connection died before write was acknowledged";
 }
 
 void Endpoint::do_read(Conn& c, void* ptr, size_t bytes, int device,

@@ -91,6 +91,10 @@ struct MsgRx {
   std::vector<char> staging;
   size_t bytes = 0;
   size_t recv_bytes = 0;
+  // Posted buffer capacity (recv_msg). SIZE_MAX while unposted: staging is
+  // heap-backed and resized to the wire-declared size, which is itself
+  // bounded by max_msg_bytes, so only the user_ptr case needs the cap.
+  size_t capacity = SIZE_MAX;
   bool known = false;  // first chunk seen
   char* dest() { return user_ptr ? user_ptr : staging.data(); }
 };
@@ -409,6 +413,19 @@ struct TransportEndpoint::Impl {
         if (!m.user_ptr && m.staging.empty()) m.staging.resize(h.msg_bytes);
       }
       if (h.off > m.bytes || h.len > m.bytes - h.off) return;  // oob chunk
+      // A posted receive has a fixed capacity; a peer message that does not
+      // fit (size desync / corrupt datagram / malicious peer) must never
+      // reach the buffer — fail the flow instead of truncating silently.
+      if (m.user_ptr && (m.bytes > m.capacity || h.off > m.capacity ||
+                         h.len > m.capacity - h.off)) {
+        UCCL_LOG_ERROR << "flow " << f.id << " msg " << h.msg_id
+                       << ": wire size " << m.bytes
+                       << " exceeds posted capacity " << m.capacity
+                       << "; failing flow";
+        f.failed = true;
+        cv.notify_all();
+        return;
+      }
       if (h.len) memcpy(m.dest() + h.off, payload, h.len);
       m.recv_bytes += h.len;
       f.bytes_received += h.len;
@@ -761,11 +778,14 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
     Flow& f = *it->second;
     msg_id = f.next_post_msg++;
     auto& m = f.rxmsgs[msg_id];
+    if (m.known && m.bytes > bytes)
+      throw std::runtime_error("transport recv buffer smaller than message");
     if (!m.staging.empty()) {
       memcpy(ptr, m.staging.data(), std::min(bytes, m.staging.size()));
       m.staging.clear();
       m.staging.shrink_to_fit();
     }
+    m.capacity = bytes;
     m.user_ptr = static_cast<char*>(ptr);
   }
   std::unique_lock<std::mutex> lk(impl_->mu);
