@@ -84,10 +84,17 @@ def build_plugin(verbose: bool = False) -> Path:
     reliable = CSRC / "transport" / "reliable.cpp"
     tracecc = CSRC / "core" / "trace.cpp"
     for cmd in (
-        ["g++", "-O2", "-g", "-std=c++17", "-fPIC", "-shared", str(src),
-         str(reliable), str(tracecc), "-o", str(target), "-pthread"],
-        ["g++", "-O2", "-std=c++17", str(harness_src), "-o", str(harness),
-         "-ldl", "-pthread"],
+        # hipcc + UCCL_NET_HIP: NCCL_PTR_CUDA support (device MRs staged
+        # through pinned bounce buffers). Loading on a GPU-less box still
+        # works (amdhip64 resolves; hip calls only run for device MRs).
+        [HIPCC, "-O2", "-g", "-std=c++17", "-fPIC", "-shared",
+         f"--offload-arch={GPU_ARCH}", "-DUCCL_NET_HIP=1", str(src),
+         str(reliable), str(tracecc), "-o", str(target), "-pthread",
+         f"-L{ROCM}/lib", "-lamdhip64", f"-Wl,-rpath,{ROCM}/lib"],
+        [HIPCC, "-O2", "-std=c++17", "-DUCCL_NET_HIP_TEST=1",
+         f"--offload-arch={GPU_ARCH}", str(harness_src), "-o", str(harness),
+         "-ldl", "-pthread", f"-L{ROCM}/lib", "-lamdhip64",
+         f"-Wl,-rpath,{ROCM}/lib"],
     ):
         if verbose:
             print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)

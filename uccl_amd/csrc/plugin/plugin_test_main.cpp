@@ -5,6 +5,10 @@
 
 #include <dlfcn.h>
 
+#ifdef UCCL_NET_HIP_TEST
+#include <hip/hip_runtime.h>
+#endif
+
 #include <cassert>
 #include <cstdio>
 #include <cstring>
@@ -197,6 +201,99 @@ int main(int argc, char** argv) {
     assert(net->closeListen(plc) == ncclSuccess);
     printf("interleave stress OK\n");
   }
+
+  // --- grouped irecv (n=2, one request) -----------------------------------
+  {
+    char hdl[NCCL_NET_HANDLE_MAXSIZE] = {};
+    void *plc = nullptr, *psc = nullptr, *prc = nullptr;
+    assert(net->listen(0, hdl, &plc) == ncclSuccess);
+    std::thread acc([&] {
+      while (!prc) assert(net->accept(plc, &prc) == ncclSuccess);
+    });
+    while (!psc) assert(net->connect(0, hdl, &psc) == ncclSuccess);
+    acc.join();
+    std::vector<char> x1(4096, 'a'), x2(8192, 'b');
+    std::vector<char> y1(4096, 0), y2(8192, 0);
+    void *gs1 = nullptr, *gs2 = nullptr, *gr = nullptr;
+    assert(net->isend(psc, x1.data(), 4096, 21, nullptr, &gs1) ==
+           ncclSuccess);
+    assert(net->isend(psc, x2.data(), 8192, 22, nullptr, &gs2) ==
+           ncclSuccess);
+    void* d[2] = {y1.data(), y2.data()};
+    int z[2] = {4096, 8192};
+    int t[2] = {21, 22};
+    void* m2[2] = {nullptr, nullptr};
+    assert(net->irecv(prc, 2, d, z, t, m2, &gr) == ncclSuccess);
+    wait_req(net, gs1, nullptr);
+    wait_req(net, gs2, nullptr);
+    int done = 0, szs[2] = {0, 0};
+    while (!done) assert(net->test(gr, &done, szs) == ncclSuccess);
+    assert(szs[0] == 4096 && szs[1] == 8192);
+    assert(memcmp(x1.data(), y1.data(), 4096) == 0);
+    assert(memcmp(x2.data(), y2.data(), 8192) == 0);
+    assert(net->closeSend(psc) == ncclSuccess);
+    assert(net->closeRecv(prc) == ncclSuccess);
+    assert(net->closeListen(plc) == ncclSuccess);
+    printf("grouped irecv OK\n");
+  }
+
+#ifdef UCCL_NET_HIP_TEST
+  // --- device-MR staging (NCCL_PTR_CUDA) ----------------------------------
+  // The exact calls RCCL's proxy makes when ptrSupport advertises CUDA:
+  // regMr(device ptr) -> isend/irecv with the device MR handle.
+  {
+    int devcount = 0;
+    if (hipGetDeviceCount(&devcount) == hipSuccess && devcount > 0) {
+      (void)hipSetDevice(0);
+      size_t const n = 3 << 20;
+      std::vector<char> hsrc(n), hdst(n, 0);
+      for (size_t i = 0; i < n; ++i) hsrc[i] = static_cast<char>(i * 31);
+      void *gsrc = nullptr, *gdst = nullptr;
+      assert(hipMalloc(&gsrc, n) == hipSuccess);
+      assert(hipMalloc(&gdst, n) == hipSuccess);
+      assert(hipMemcpy(gsrc, hsrc.data(), n, hipMemcpyHostToDevice) ==
+             hipSuccess);
+      char hdl[NCCL_NET_HANDLE_MAXSIZE] = {};
+      void *plc = nullptr, *psc = nullptr, *prc = nullptr;
+      assert(net->listen(0, hdl, &plc) == ncclSuccess);
+      std::thread acc([&] {
+        while (!prc) assert(net->accept(plc, &prc) == ncclSuccess);
+      });
+      while (!psc) assert(net->connect(0, hdl, &psc) == ncclSuccess);
+      acc.join();
+      assert(net->getProperties(0, &props) == ncclSuccess);
+      assert(props.ptrSupport & NCCL_PTR_CUDA);
+      void *smh = nullptr, *rmh = nullptr;
+      assert(net->regMr(psc, gsrc, n, NCCL_PTR_CUDA, &smh) == ncclSuccess);
+      assert(net->regMr(prc, gdst, n, NCCL_PTR_CUDA, &rmh) == ncclSuccess);
+      void *sr = nullptr, *rr = nullptr;
+      assert(net->isend(psc, gsrc, static_cast<int>(n), 42, smh, &sr) ==
+             ncclSuccess);
+      void* d[1] = {gdst};
+      int z[1] = {static_cast<int>(n)};
+      int t[1] = {42};
+      void* mh[1] = {rmh};
+      assert(net->irecv(prc, 1, d, z, t, mh, &rr) == ncclSuccess);
+      wait_req(net, sr, nullptr);
+      int got = 0;
+      wait_req(net, rr, &got);
+      assert(got == static_cast<int>(n));
+      assert(hipMemcpy(hdst.data(), gdst, n, hipMemcpyDeviceToHost) ==
+             hipSuccess);
+      assert(memcmp(hsrc.data(), hdst.data(), n) == 0);
+      assert(net->deregMr(psc, smh) == ncclSuccess);
+      assert(net->deregMr(prc, rmh) == ncclSuccess);
+      assert(net->closeSend(psc) == ncclSuccess);
+      assert(net->closeRecv(prc) == ncclSuccess);
+      assert(net->closeListen(plc) == ncclSuccess);
+      (void)hipFree(gsrc);
+      (void)hipFree(gdst);
+      printf("CUDA-MR staging OK\n");
+    } else {
+      printf("CUDA-MR SKIP (no GPU)\n");
+    }
+  }
+#endif
 
   printf("PLUGIN HARNESS OK\n");
   return 0;

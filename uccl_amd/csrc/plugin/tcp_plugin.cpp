@@ -8,11 +8,17 @@
 // UcclRDMAEngine. Two data planes, selected by UCCL_NET_TRANSPORT:
 //   multipath (default) — TransportEndpoint flows (message-oriented)
 //   tcp               — plain stream sockets (NCCL-socket-transport-like)
-// Host pointers only (RCCL stages GPU data through its own pinned
-// buffers, like NCCL's built-in socket transport).
 //
-// No HIP dependency, so the plugin is fully CPU-testable (the dlopen
-// harness drives the vtable over both data planes).
+// Pointer support: NCCL_PTR_HOST always; NCCL_PTR_CUDA when built with
+// UCCL_NET_HIP (the shipped build) — registered device MRs are staged
+// through per-comm pinned bounce buffers with hipMemcpy, mirroring the
+// reference's GPU regMr+staging (nccl_plugin.cc:472-594). iflush stays a
+// no-op because recv completion is only signalled after the HtoD copy
+// has synchronized (visibility is implied, no GDR in this fabric).
+//
+// When built without UCCL_NET_HIP the plugin has no HIP dependency and
+// is fully CPU-testable (the dlopen harness drives the vtable over both
+// data planes with host pointers).
 
 #include <fcntl.h>
 #include <unistd.h>
@@ -29,6 +35,10 @@
 #include <mutex>
 #include <thread>
 #include <vector>
+
+#ifdef UCCL_NET_HIP
+#include <hip/hip_runtime.h>
+#endif
 
 #include "../core/env.h"
 #include "../core/log.h"
@@ -55,10 +65,22 @@ struct Handle {
 // process's connect-retry landing on a recycled ephemeral port)
 constexpr uint64_t kPluginCookie = 0x7563636c2d6e6574ULL;  // "uccl-net"
 
+// Registered memory record. type is NCCL_PTR_HOST or NCCL_PTR_CUDA; device
+// is resolved at registration so staging copies can set the right context.
+struct Mr {
+  int type = NCCL_PTR_HOST;
+  int device = 0;
+};
+
+// One request may cover n grouped recvs (p_irecv n>1): done counts down.
 struct Request {
-  std::atomic<int> done{0};
-  int size = 0;
+  std::atomic<int> pending{1};
+  std::atomic<int> error{0};
+  int n = 1;
+  int sizes[8] = {0};
   bool recv = false;
+  bool done() const { return pending.load(std::memory_order_acquire) == 0; }
+  void complete_one() { pending.fetch_sub(1, std::memory_order_release); }
 };
 
 struct Frame {
@@ -72,6 +94,8 @@ struct PostedRecv {
   int cap;
   int tag;
   Request* req;
+  int slot;       // index into req->sizes for grouped recvs
+  Mr mr;          // destination memory kind
 };
 
 struct SendOp {
@@ -79,6 +103,7 @@ struct SendOp {
   int size;
   int tag;
   Request* req;
+  Mr mr;          // source memory kind
 };
 
 bool use_multipath() {
@@ -147,6 +172,32 @@ struct ListenComm {
   uint64_t nonce = 0;  // multipath mode
 };
 
+#ifdef UCCL_NET_HIP
+// Grow-to-fit pinned bounce buffer for device-MR staging (per comm, per
+// direction; the worker thread owns it so no locking is needed).
+struct Staging {
+  char* buf = nullptr;
+  size_t cap = 0;
+  ~Staging() {
+    if (buf) (void)hipHostFree(buf);
+  }
+  char* ensure(size_t n) {
+    if (n > cap) {
+      if (buf) (void)hipHostFree(buf);
+      size_t c = 4096;
+      while (c < n) c <<= 1;
+      if (hipHostMalloc(&buf, c) != hipSuccess) {
+        buf = nullptr;
+        cap = 0;
+        return nullptr;
+      }
+      cap = c;
+    }
+    return buf;
+  }
+};
+#endif
+
 struct Comm {
   int fd = -1;        // tcp mode
   uint64_t flow = 0;  // multipath mode (0 = tcp)
@@ -160,11 +211,31 @@ struct Comm {
   std::deque<PostedRecv> posted;
   std::deque<Frame> unmatched;
 
+#ifdef UCCL_NET_HIP
+  Staging tx_staging, rx_staging;
+#endif
+
   std::mutex mu;
   std::condition_variable cv;
 
   ~Comm();
 };
+
+// copy `bytes` from a host buffer into pr.data (host or device MR)
+bool place_payload(Comm* c, PostedRecv const& pr, char const* src,
+                   uint64_t bytes) {
+  (void)c;
+  if (!bytes) return true;
+#ifdef UCCL_NET_HIP
+  if (pr.mr.type == NCCL_PTR_CUDA) {
+    (void)hipSetDevice(pr.mr.device);
+    return hipMemcpy(pr.data, src, bytes, hipMemcpyHostToDevice) ==
+           hipSuccess;
+  }
+#endif
+  memcpy(pr.data, src, bytes);
+  return true;
+}
 
 struct WireHdr {
   uint64_t bytes;
@@ -237,11 +308,26 @@ void tx_loop(Comm* c) {
       op = c->sendq.front();
       c->sendq.pop_front();
     }
+    void const* payload = op.data;
+#ifdef UCCL_NET_HIP
+    if (op.mr.type == NCCL_PTR_CUDA && op.size) {
+      (void)hipSetDevice(op.mr.device);
+      char* st = c->tx_staging.ensure(op.size);
+      if (!st || hipMemcpy(st, op.data, op.size, hipMemcpyDeviceToHost) !=
+                     hipSuccess) {
+        PLOG(NCCL_LOG_WARN, "uccl-net: DtoH staging failed (%d B)", op.size);
+        op.req->error.store(1, std::memory_order_relaxed);
+        op.req->complete_one();
+        continue;
+      }
+      payload = st;
+    }
+#endif
     WireHdr h{static_cast<uint64_t>(op.size), op.tag, 0};
     chan_send(c, &h, sizeof(h));
-    if (op.size) chan_send(c, op.data, op.size);
-    op.req->size = op.size;
-    op.req->done.store(1, std::memory_order_release);
+    if (op.size) chan_send(c, payload, op.size);
+    op.req->sizes[0] = op.size;
+    op.req->complete_one();
   }
 }
 
@@ -272,13 +358,35 @@ void rx_loop(Comm* c) {
     }
     if (matched) {
       if (pr.cap < static_cast<int>(h.bytes)) {
+        // Peer sent more than the posted capacity: drain the frame so the
+        // stream stays in sync and fail the REQUEST (RCCL aborts cleanly
+        // via p_test), instead of silently killing the connection.
         PLOG(NCCL_LOG_WARN, "uccl-net: recv overflow tag=%d %lu > %d",
              h.tag, (unsigned long)h.bytes, pr.cap);
-        return;
+        std::vector<char> sink(h.bytes);
+        if (h.bytes) chan_recv(c, sink.data(), h.bytes);
+        pr.req->error.store(1, std::memory_order_relaxed);
+        pr.req->complete_one();
+        continue;
       }
-      if (h.bytes) chan_recv(c, pr.data, h.bytes);
-      pr.req->size = static_cast<int>(h.bytes);
-      pr.req->done.store(1, std::memory_order_release);
+      bool ok = true;
+#ifdef UCCL_NET_HIP
+      if (pr.mr.type == NCCL_PTR_CUDA) {
+        char* st = c->rx_staging.ensure(h.bytes ? h.bytes : 1);
+        if (!st) {
+          ok = false;
+        } else {
+          if (h.bytes) chan_recv(c, st, h.bytes);
+          ok = place_payload(c, pr, st, h.bytes);
+        }
+      } else
+#endif
+      {
+        if (h.bytes) chan_recv(c, pr.data, h.bytes);
+      }
+      if (!ok) pr.req->error.store(1, std::memory_order_relaxed);
+      pr.req->sizes[pr.slot] = static_cast<int>(h.bytes);
+      pr.req->complete_one();
     } else {
       Frame f;
       f.bytes = h.bytes;
@@ -297,15 +405,18 @@ void rx_loop(Comm* c) {
         if (it->tag == f.tag) {
           PostedRecv pr2 = *it;
           c->posted.erase(it);
+          late_match = true;
           if (pr2.cap >= static_cast<int>(f.bytes)) {
-            if (f.bytes) memcpy(pr2.data, f.data.data(), f.bytes);
-            pr2.req->size = static_cast<int>(f.bytes);
-            pr2.req->done.store(1, std::memory_order_release);
-            late_match = true;
+            if (!place_payload(c, pr2, f.data.data(), f.bytes))
+              pr2.req->error.store(1, std::memory_order_relaxed);
+            pr2.req->sizes[pr2.slot] = static_cast<int>(f.bytes);
+            pr2.req->complete_one();
           } else {
             PLOG(NCCL_LOG_WARN,
-                 "uccl-net: recv overflow tag=%d %lu > %d (req dropped)",
+                 "uccl-net: recv overflow tag=%d %lu > %d (req failed)",
                  f.tag, (unsigned long)f.bytes, pr2.cap);
+            pr2.req->error.store(1, std::memory_order_relaxed);
+            pr2.req->complete_one();
           }
           break;
         }
@@ -326,24 +437,36 @@ ncclResult_t p_init(ncclDebugLogger_t logfn) {
   return ncclSuccess;
 }
 
+int num_vdevs() {
+  static int n = [] {
+    long v = uccl::env_int("UCCL_NET_NDEV", 1);
+    return static_cast<int>(v < 1 ? 1 : (v > 4 ? 4 : v));
+  }();
+  return n;
+}
+
 ncclResult_t p_devices(int* ndev) {
-  *ndev = 1;
+  *ndev = num_vdevs();
   return ncclSuccess;
 }
 
 ncclResult_t p_getProperties(int dev, ncclNetProperties_v6_t* props) {
-  static char name[] = "uccl0";
+  static char names[4][8] = {"uccl0", "uccl1", "uccl2", "uccl3"};
   static char pci[] = "";
+  if (dev < 0 || dev >= num_vdevs()) return ncclInvalidArgument;
   memset(props, 0, sizeof(*props));
-  props->name = name;
+  props->name = names[dev];
   props->pciPath = pci;
-  props->guid = 0x75636331;
+  props->guid = 0x75636331 + dev;
   props->ptrSupport = NCCL_PTR_HOST;
+#ifdef UCCL_NET_HIP
+  props->ptrSupport |= NCCL_PTR_CUDA;
+#endif
   props->speed = 100000;
   props->port = 0;
   props->latency = 20.0f;
   props->maxComms = 65536;
-  props->maxRecvs = 1;
+  props->maxRecvs = 4;
   return ncclSuccess;
 }
 
@@ -475,25 +598,48 @@ ncclResult_t p_accept(void* listenComm, void** recvComm) {
 
 ncclResult_t p_regMr(void* comm, void* data, int size, int type,
                      void** mhandle) {
-  if (type != NCCL_PTR_HOST) return ncclInternalError;
-  *mhandle = nullptr;
-  return ncclSuccess;
+  (void)comm;
+  (void)size;
+  if (type == NCCL_PTR_HOST) {
+    *mhandle = nullptr;  // null handle = host memory
+    return ncclSuccess;
+  }
+#ifdef UCCL_NET_HIP
+  if (type == NCCL_PTR_CUDA) {
+    auto* mr = new Mr();
+    mr->type = NCCL_PTR_CUDA;
+    hipPointerAttribute_t attr{};
+    if (hipPointerGetAttributes(&attr, data) == hipSuccess)
+      mr->device = attr.device;
+    *mhandle = mr;
+    return ncclSuccess;
+  }
+#endif
+  (void)data;
+  return ncclInternalError;
 }
 
 ncclResult_t p_regMrDmaBuf(void*, void*, size_t, int, uint64_t, int,
                            void**) {
-  return ncclInternalError;
+  return ncclInternalError;  // no DMA engine in this fabric
 }
 
-ncclResult_t p_deregMr(void*, void*) { return ncclSuccess; }
+ncclResult_t p_deregMr(void*, void* mhandle) {
+  delete static_cast<Mr*>(mhandle);  // null-safe (host MRs)
+  return ncclSuccess;
+}
 
-ncclResult_t p_isend(void* sendComm, void* data, int size, int tag, void*,
-                     void** request) {
+Mr mr_of(void* mhandle) {
+  return mhandle ? *static_cast<Mr*>(mhandle) : Mr{};
+}
+
+ncclResult_t p_isend(void* sendComm, void* data, int size, int tag,
+                     void* mhandle, void** request) {
   auto* c = static_cast<Comm*>(sendComm);
   auto* r = new Request();
   {
     std::lock_guard<std::mutex> g(c->mu);
-    c->sendq.push_back(SendOp{data, size, tag, r});
+    c->sendq.push_back(SendOp{data, size, tag, r, mr_of(mhandle)});
   }
   c->cv.notify_one();
   *request = r;
@@ -501,41 +647,60 @@ ncclResult_t p_isend(void* sendComm, void* data, int size, int tag, void*,
 }
 
 ncclResult_t p_irecv(void* recvComm, int n, void** data, int* sizes,
-                     int* tags, void**, void** request) {
-  if (n != 1) return ncclInternalError;
+                     int* tags, void** mhandles, void** request) {
+  if (n < 1 || n > 8) return ncclInternalError;
   auto* c = static_cast<Comm*>(recvComm);
   auto* r = new Request();
   r->recv = true;
+  r->n = n;
+  r->pending.store(n, std::memory_order_relaxed);
   {
     std::lock_guard<std::mutex> g(c->mu);
-    // match an already-arrived frame first
-    for (auto it = c->unmatched.begin(); it != c->unmatched.end(); ++it) {
-      if (it->tag == tags[0]) {
-        if (static_cast<int>(it->bytes) > sizes[0]) return ncclInternalError;
-        memcpy(data[0], it->data.data(), it->bytes);
-        r->size = static_cast<int>(it->bytes);
-        r->done.store(1, std::memory_order_release);
-        c->unmatched.erase(it);
-        *request = r;
-        return ncclSuccess;
+    for (int i = 0; i < n; ++i) {
+      Mr const mr = mr_of(mhandles ? mhandles[i] : nullptr);
+      // match an already-arrived frame first
+      bool hit = false;
+      for (auto it = c->unmatched.begin(); it != c->unmatched.end(); ++it) {
+        if (it->tag == tags[i]) {
+          hit = true;
+          if (static_cast<int>(it->bytes) > sizes[i]) {
+            r->error.store(1, std::memory_order_relaxed);
+          } else {
+            PostedRecv pr{data[i], sizes[i], tags[i], r, i, mr};
+            if (!place_payload(c, pr, it->data.data(), it->bytes))
+              r->error.store(1, std::memory_order_relaxed);
+            r->sizes[i] = static_cast<int>(it->bytes);
+          }
+          r->complete_one();
+          c->unmatched.erase(it);
+          break;
+        }
       }
+      if (!hit)
+        c->posted.push_back(PostedRecv{data[i], sizes[i], tags[i], r, i, mr});
     }
-    c->posted.push_back(PostedRecv{data[0], sizes[0], tags[0], r});
   }
   *request = r;
   return ncclSuccess;
 }
 
 ncclResult_t p_iflush(void*, int, void**, int*, void**, void** request) {
-  *request = nullptr;  // host memory: nothing to flush
+  // recv completion is only reported after the (synchronous) HtoD staging
+  // copy, so there is never un-flushed GPU data to wait on
+  *request = nullptr;
   return ncclSuccess;
 }
 
 ncclResult_t p_test(void* request, int* done, int* sizes) {
   auto* r = static_cast<Request*>(request);
-  if (r->done.load(std::memory_order_acquire)) {
+  if (r->done()) {
+    if (r->error.load(std::memory_order_relaxed)) {
+      delete r;
+      return ncclInternalError;
+    }
     *done = 1;
-    if (sizes) sizes[0] = r->size;
+    if (sizes)
+      for (int i = 0; i < r->n; ++i) sizes[i] = r->sizes[i];
     delete r;
   } else {
     *done = 0;
