@@ -26,6 +26,8 @@ SOURCES = [
     CSRC / "collective" / "pg_backend.cpp",
     CSRC / "p2p" / "endpoint.cpp",
     CSRC / "transport" / "reliable.cpp",
+    CSRC / "transport" / "udp_fabric.cpp",
+    CSRC / "transport" / "verbs_fabric.cpp",
     CSRC / "ep" / "ep_kernels.hip",
     CSRC / "ep" / "ep_buffer.cpp",
     CSRC / "ep" / "ep_proxy.cpp",
@@ -68,9 +70,12 @@ def build_plugin(verbose: bool = False) -> Path:
     capi = plugdir / "libuccl_p2p.so"
     capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
                  CSRC / "transport" / "reliable.cpp",
+                 CSRC / "transport" / "udp_fabric.cpp",
+                 CSRC / "transport" / "verbs_fabric.cpp",
                  CSRC / "core" / "trace.cpp",
                  CSRC / "p2p" / "c_api.h", CSRC / "p2p" / "endpoint.h",
-                 CSRC / "transport" / "reliable.h"]
+                 CSRC / "transport" / "reliable.h",
+                 CSRC / "transport" / "fabric.h"]
     shim_check = plugdir / "libuccl_nccl.so"
     shim_deps = [CSRC / "nccl_shim" / "nccl_shim.cpp",
                  CSRC / "collective" / "kernels.hip",
@@ -82,6 +87,8 @@ def build_plugin(verbose: bool = False) -> Path:
     import subprocess as sp
 
     reliable = CSRC / "transport" / "reliable.cpp"
+    udpfab = CSRC / "transport" / "udp_fabric.cpp"
+    verbsfab = CSRC / "transport" / "verbs_fabric.cpp"
     tracecc = CSRC / "core" / "trace.cpp"
     for cmd in (
         # hipcc + UCCL_NET_HIP: NCCL_PTR_CUDA support (device MRs staged
@@ -89,8 +96,14 @@ def build_plugin(verbose: bool = False) -> Path:
         # works (amdhip64 resolves; hip calls only run for device MRs).
         [HIPCC, "-O2", "-g", "-std=c++17", "-fPIC", "-shared",
          f"--offload-arch={GPU_ARCH}", "-DUCCL_NET_HIP=1", str(src),
-         str(reliable), str(tracecc), "-o", str(target), "-pthread",
+         str(reliable), str(udpfab), str(verbsfab), str(tracecc),
+         "-o", str(target), "-pthread", "-ldl",
          f"-L{ROCM}/lib", "-lamdhip64", f"-Wl,-rpath,{ROCM}/lib"],
+        # mock verbs provider: software-loopback RDMA for the CPU test
+        # tier (drives the whole verbs fabric without a NIC)
+        ["g++", "-O2", "-std=c++17", "-fPIC", "-shared",
+         str(CSRC / "transport" / "mock_verbs_provider.cpp"),
+         "-o", str(plugdir / "libuccl_verbs_mock.so"), "-pthread"],
         [HIPCC, "-O2", "-std=c++17", "-DUCCL_NET_HIP_TEST=1",
          f"--offload-arch={GPU_ARCH}", str(harness_src), "-o", str(harness),
          "-ldl", "-pthread", f"-L{ROCM}/lib", "-lamdhip64",
@@ -126,13 +139,26 @@ def build_plugin(verbose: bool = False) -> Path:
         if r.returncode != 0:
             raise RuntimeError(f"nccl shim build failed:\n{r.stdout.decode()}")
 
+    # real-ibverbs provider: only when rdma-core headers are installed
+    # (the dev/CI image has none; the mock provider covers the fabric
+    # logic there and this file stays compile-checked by inspection)
+    if Path("/usr/include/infiniband/verbs.h").exists():
+        cmd = ["gcc", "-O2", "-fPIC", "-shared",
+               str(CSRC / "transport" / "verbs_adapter.c"),
+               "-o", str(plugdir / "libuccl_verbs_ib.so"), "-libverbs"]
+        if verbose:
+            print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)
+        r = sp.run(cmd, stdout=sp.PIPE, stderr=sp.STDOUT)
+        if r.returncode != 0:
+            raise RuntimeError(f"verbs adapter build failed:\n{r.stdout.decode()}")
+
     # flat C API lib for NIXL-style integrators (reference: p2p/uccl_engine.h)
     if _stale(capi, capi_srcs + deps):
         cc_srcs = [x for x in capi_srcs if x.suffix == ".cpp"]
         cmd = [HIPCC, "-O2", "-std=c++17", "-fPIC", "-shared",
                f"--offload-arch={GPU_ARCH}"] + [str(x) for x in cc_srcs] + [
-               "-o", str(capi), "-pthread", f"-L{ROCM}/lib", "-lamdhip64",
-               f"-Wl,-rpath,{ROCM}/lib"]
+               "-o", str(capi), "-pthread", "-ldl", f"-L{ROCM}/lib",
+               "-lamdhip64", f"-Wl,-rpath,{ROCM}/lib"]
         if verbose:
             print("[uccl_amd build]", " ".join(cmd), file=sys.stderr)
         r = sp.run(cmd, stdout=sp.PIPE, stderr=sp.STDOUT)

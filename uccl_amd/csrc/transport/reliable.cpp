@@ -1,6 +1,5 @@
 #include "reliable.h"
 
-#include <poll.h>
 #include <sys/socket.h>
 
 #include <algorithm>
@@ -14,6 +13,7 @@
 #include "../core/trace.h"
 #include "../core/log.h"
 #include "../core/net.h"
+#include "fabric.h"
 
 namespace uccl {
 namespace transport {
@@ -21,19 +21,7 @@ namespace transport {
 namespace {
 
 constexpr uint32_t kMagic = 0x55434354;  // "UCCT"
-enum Kind : uint32_t { kData = 1, kAck = 2, kWake = 3 };
-
-struct DataHdr {
-  uint32_t magic;
-  uint32_t kind;
-  uint64_t flow;
-  uint64_t msg_id;
-  uint64_t msg_bytes;
-  uint64_t off;
-  uint32_t len;
-  uint32_t csn;
-  uint64_t ts_ns;
-};
+enum Kind : uint32_t { kAck = 2 };
 
 struct AckHdr {
   uint32_t magic;
@@ -101,7 +89,6 @@ struct MsgRx {
 
 struct TransportEndpoint::Flow {
   uint64_t id;
-  sockaddr_in peer_paths[64];
   int num_paths;
 
   // --- TX direction ---
@@ -152,10 +139,7 @@ struct TransportEndpoint::Impl {
   int dup_thres = 32;
   uint64_t rto_base_ns = 20000000;
   int rto_abort_thres = 50;
-  std::vector<int> socks;          // UDP path sockets
-  std::vector<uint16_t> ports;
-  int wake_fd = -1;                // self-addressed UDP for wakeups
-  uint16_t wake_port = 0;
+  std::unique_ptr<Fabric> fabric;  // the wire plane (udp | verbs)
   int ctrl_listen = -1;
   uint16_t ctrl_port = 0;
   std::thread ctrl_thread;
@@ -173,22 +157,10 @@ struct TransportEndpoint::Impl {
   LatencyHist rtt_hist;
 
   // ---- helpers ----
-  void wake() {
-    sockaddr_in a{};
-    a.sin_family = AF_INET;
-    a.sin_port = htons(wake_port);
-    inet_pton(AF_INET, "127.0.0.1", &a.sin_addr);
-    uint32_t w[2] = {kMagic, kWake};
-    (void)sendto(wake_fd, w, sizeof(w), 0, reinterpret_cast<sockaddr*>(&a),
-                 sizeof(a));
-  }
+  void wake() { fabric->wake(); }
 
   void send_chunk(Flow& f, uint32_t csn, ChunkTx& c) {
-    DataHdr h{kMagic, kData, f.id, c.msg->id, c.msg->bytes, c.off, c.len,
-              csn, now_ns()};
-    char buf[sizeof(DataHdr) + 65536];
-    memcpy(buf, &h, sizeof(h));
-    memcpy(buf + sizeof(h), c.msg->ptr + c.off, c.len);
+    ChunkDesc d{f.id, c.msg->id, c.msg->bytes, c.off, c.len, csn, now_ns()};
     int const path = csn % f.num_paths;  // spray round-robin across paths
     c.send_ts = now_ns();
     ++c.attempts;
@@ -196,10 +168,9 @@ struct TransportEndpoint::Impl {
       ++st.injected_drops;
       return;  // "sent" into the void
     }
-    (void)sendto(socks[path], buf, sizeof(h) + c.len, 0,
-                 reinterpret_cast<sockaddr*>(&f.peer_paths[path]),
-                 sizeof(sockaddr_in));
-    ++st.data_sent;
+    // a transient fabric would-block counts as a loss: RTO retransmits
+    if (fabric->post_chunk(f.id, path, d, c.msg->ptr + c.off))
+      ++st.data_sent;
   }
 
   bool eqds_mode() const { return cc_mode == "eqds"; }
@@ -212,6 +183,9 @@ struct TransportEndpoint::Impl {
       if (eqds_mode() &&
           f.bytes_chunked + chunk_bytes > f.credit_limit)
         break;
+      // rendezvous fabrics (verbs) gate chunking on the peer's window
+      // advertisement for this message (FIFO rendezvous)
+      if (!fabric->tx_ready(f.id, f.txq.front()->id)) break;
       if (pace_q32) {
         uint64_t const now = now_ns();
         if (f.next_send_ns > now) break;  // paced: retry on a later pump
@@ -248,8 +222,7 @@ struct TransportEndpoint::Impl {
     return n > 0 ? n : 1;
   }
 
-  void send_ack(Flow& f, int sock_idx, sockaddr_in const& to,
-                bool allow_drop = true) {
+  void send_ack(Flow& f, int path, bool allow_drop = true) {
     // grant: allow the sender to stay rwnd bytes ahead of what we've
     // seen; in paced-EQDS mode (UCCL_TP_EQDS_MBPS) the receiver doles
     // that window out as pull quanta at the configured aggregate rate,
@@ -289,8 +262,7 @@ struct TransportEndpoint::Impl {
       else
         break;
     }
-    (void)sendto(socks[sock_idx], &a, sizeof(a), 0,
-                 reinterpret_cast<sockaddr const*>(&to), sizeof(to));
+    fabric->post_ctrl(f.id, path, &a, sizeof(a));
     ++st.acks_sent;
   }
 
@@ -358,17 +330,36 @@ struct TransportEndpoint::Impl {
       rtt_hist.record_us(rtt_us);
       timely_update(f, rtt_us);
     }
+    // Placed-chunk fabrics (verbs) carry no timestamp in the data path
+    // (the 32-bit IMM has no room), so ts_echo is 0: measure RTT from
+    // the sender-side send_ts of a first-attempt chunk this ack covers
+    // (Karn's rule — retransmitted chunks are ambiguous).
+    double rtt_fallback = -1.0;
+    auto consider_rtt = [&](uint32_t csn) {
+      if (a.ts_echo) return;
+      auto it = f.inflight.find(csn);
+      if (it != f.inflight.end() && it->second.attempts == 1 &&
+          it->second.send_ts)
+        rtt_fallback = (now_ns() - it->second.send_ts) / 1000.0;
+    };
     // cumulative
-    while (!f.inflight.empty() && f.inflight.begin()->first < a.cum)
+    while (!f.inflight.empty() && f.inflight.begin()->first < a.cum) {
+      consider_rtt(f.inflight.begin()->first);
       ack_chunk(f, f.inflight.begin()->first);
+    }
     // SACK bits
     uint32_t highest_sacked = a.cum;
     for (int i = 0; i < 128; ++i) {
       bool const set = i < 64 ? (a.sack0 >> i) & 1 : (a.sack1 >> (i - 64)) & 1;
       if (set) {
+        consider_rtt(a.cum + i);
         ack_chunk(f, a.cum + i);
         highest_sacked = a.cum + i;
       }
+    }
+    if (rtt_fallback > 0) {
+      rtt_hist.record_us(rtt_fallback);
+      timely_update(f, rtt_fallback);
     }
     // SACK-hole fast retransmit: spraying reorders heavily (and the
     // progress loop drains path sockets in batches), so per-chunk dup-ack
@@ -393,13 +384,15 @@ struct TransportEndpoint::Impl {
     pump_tx(f);
   }
 
-  void handle_data(Flow& f, DataHdr const& h, char const* payload,
-                   size_t payload_len, int sock_idx,
-                   sockaddr_in const& from) {
+  // `payload` is the inline chunk bytes (UDP) or nullptr when the fabric
+  // already PLACED the data at its destination (verbs RDMA write — the
+  // NIC bounds-checked the write against the registered window).
+  void handle_data(Flow& f, ChunkDesc const& h, char const* payload,
+                   int path) {
     // every field below is wire-controlled: bound it (subtraction form —
     // addition could wrap) before it touches memory. A legit peer never
     // violates these; a corrupt/stray datagram gets dropped unacked.
-    if (h.len > payload_len || h.len > chunk_bytes) return;
+    if (h.len > chunk_bytes) return;
     if (h.msg_bytes > max_msg_bytes) return;
     ++st.data_recv;
     f.last_data_ts = h.ts_ns;
@@ -426,7 +419,7 @@ struct TransportEndpoint::Impl {
         cv.notify_all();
         return;
       }
-      if (h.len) memcpy(m.dest() + h.off, payload, h.len);
+      if (h.len && payload) memcpy(m.dest() + h.off, payload, h.len);
       m.recv_bytes += h.len;
       f.bytes_received += h.len;
       f.rx_ooo[h.csn] = true;
@@ -440,7 +433,7 @@ struct TransportEndpoint::Impl {
       }
     }
     f.last_rx_ns = now_ns();
-    send_ack(f, sock_idx, from);
+    send_ack(f, path);
   }
 
   void rto_scan() {
@@ -476,39 +469,21 @@ struct TransportEndpoint::Impl {
   }
 
   void progress_loop() {
-    std::vector<pollfd> pfds;
-    for (int s : socks) pfds.push_back({s, POLLIN, 0});
-    pfds.push_back({wake_fd, POLLIN, 0});
-    std::vector<char> buf(sizeof(DataHdr) + 65536 + 64);
-    while (!stop) {
-      (void)poll(pfds.data(), pfds.size(), 5);
+    auto on_event = [this](FabricEvent const& ev) {
       std::lock_guard<std::mutex> g(mu);
-      for (size_t i = 0; i < pfds.size(); ++i) {
-        while (true) {
-          sockaddr_in from{};
-          socklen_t fl = sizeof(from);
-          ssize_t n = recvfrom(pfds[i].fd, buf.data(), buf.size(),
-                               MSG_DONTWAIT,
-                               reinterpret_cast<sockaddr*>(&from), &fl);
-          if (n <= 0) break;
-          if (n < static_cast<ssize_t>(8)) continue;
-          auto kind = reinterpret_cast<uint32_t const*>(buf.data())[1];
-          if (kind == kWake) continue;
-          if (kind == kData && n >= static_cast<ssize_t>(sizeof(DataHdr))) {
-            auto const* h = reinterpret_cast<DataHdr const*>(buf.data());
-            auto it = flows.find(h->flow);
-            if (it != flows.end())
-              handle_data(*it->second, *h, buf.data() + sizeof(DataHdr),
-                          static_cast<size_t>(n) - sizeof(DataHdr),
-                          static_cast<int>(i), from);
-          } else if (kind == kAck &&
-                     n >= static_cast<ssize_t>(sizeof(AckHdr))) {
-            auto const* a = reinterpret_cast<AckHdr const*>(buf.data());
-            auto it = flows.find(a->flow);
-            if (it != flows.end()) handle_ack(*it->second, *a);
-          }
-        }
+      auto it = flows.find(ev.flow);
+      if (it == flows.end()) return;
+      if (ev.kind == FabricEvent::kChunk) {
+        handle_data(*it->second, ev.desc, ev.payload, ev.path);
+      } else if (ev.ctrl_len >= sizeof(AckHdr)) {
+        auto const* a = reinterpret_cast<AckHdr const*>(ev.ctrl);
+        if (a->magic == kMagic && a->kind == kAck)
+          handle_ack(*it->second, *a);
       }
+    };
+    while (!stop) {
+      fabric->poll(on_event, 5);
+      std::lock_guard<std::mutex> g(mu);
       rto_scan();
       // paced-EQDS credit refresh: a credit-stalled sender emits no
       // data, so acks (which carry grants) would never flow again —
@@ -522,22 +497,50 @@ struct TransportEndpoint::Impl {
           // refresh acks bypass loss injection: the drop hash is keyed
           // on data_recv, which is frozen while the sender is stalled,
           // so an injected drop here would repeat forever (livelock)
-          send_ack(*fp, 0, fp->peer_paths[0], /*allow_drop=*/false);
+          send_ack(*fp, 0, /*allow_drop=*/false);
         }
       }
     }
   }
 
   // ---- flow setup over TCP ctrl ----
-  struct CtrlMsg {
-    uint64_t magic;  // rejects stray connections on recycled ports
+  // Two round trips so rendezvous fabrics (verbs) can exchange PER-FLOW
+  // addressing (QPNs exist only after create_flow), the same multi-step
+  // QP metadata exchange as the reference's uccl_connect/uccl_accept
+  // (collective/rdma/transport.h:1077-1083):
+  //   C->A  Hello{tag}
+  //   A->C  Reply{flow, md_acceptor}      (A created its flow resources)
+  //   C->A  Reply{md_connector}           (C created + installed peer)
+  //   A->C  Done{paths}                   (A installed peer)
+  struct CtrlHello {
+    uint64_t magic;
+    uint64_t tag;
+  };
+  struct CtrlBlob {
     uint64_t flow;
-    uint64_t tag;  // connector-supplied peer identity
-    int num_paths;
-    uint16_t ports[64];
-    char ip[48];
+    uint32_t md_len;
+  };
+  struct CtrlDone {
+    int32_t paths;
   };
   static constexpr uint64_t kCtrlMagic = 0x756363746e737074ULL;
+  static constexpr uint32_t kMaxMd = 4096;
+
+  static bool send_blob(int fd, uint64_t flow, std::string const& md) {
+    CtrlBlob b{flow, static_cast<uint32_t>(md.size())};
+    net::send_all(fd, &b, sizeof(b));
+    if (!md.empty()) net::send_all(fd, md.data(), md.size());
+    return true;
+  }
+
+  static bool recv_blob(int fd, uint64_t* flow, std::string* md) {
+    CtrlBlob b{};
+    if (!net::recv_all(fd, &b, sizeof(b)) || b.md_len > kMaxMd) return false;
+    md->resize(b.md_len);
+    if (b.md_len && !net::recv_all(fd, md->data(), b.md_len)) return false;
+    *flow = b.flow;
+    return true;
+  }
 
   void ctrl_loop() {
     while (!stop) {
@@ -546,51 +549,52 @@ struct TransportEndpoint::Impl {
         if (stop) return;
         continue;
       }
-      timeval tv{2, 0};  // bounded read; strays must not stall accepts
+      timeval tv{2, 0};  // bounded reads; strays must not stall accepts
       setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
-      CtrlMsg peer{};
-      if (!net::recv_all(fd, &peer, sizeof(peer)) ||
-          peer.magic != kCtrlMagic) {
+      CtrlHello hello{};
+      if (!net::recv_all(fd, &hello, sizeof(hello)) ||
+          hello.magic != kCtrlMagic) {
         ::close(fd);
         continue;
       }
       uint64_t const flow = next_flow.fetch_add(1);
-      CtrlMsg mine = self_ctrl(flow);
-      net::send_all(fd, &mine, sizeof(mine));
+      std::string md_mine;
+      try {
+        md_mine = fabric->create_flow(flow, /*connector=*/false);
+      } catch (std::exception const& e) {
+        UCCL_LOG_ERROR << "fabric create_flow failed: " << e.what();
+        ::close(fd);
+        continue;
+      }
+      uint64_t peer_flow = 0;
+      std::string md_peer;
+      int paths = 0;
+      if (send_blob(fd, flow, md_mine) &&
+          recv_blob(fd, &peer_flow, &md_peer)) {
+        paths = fabric->install_peer(flow, md_peer);
+      }
+      CtrlDone done{paths};
+      net::send_all(fd, &done, sizeof(done));
       ::close(fd);
-      install_flow(flow, peer);
+      if (paths <= 0) {
+        fabric->remove_peer(flow);
+        continue;
+      }
+      install_flow(flow, paths);
       {
         std::lock_guard<std::mutex> g(mu);
         accepted.push_back(flow);
-        accepted_tags.push_back(peer.tag);
+        accepted_tags.push_back(hello.tag);
       }
       cv.notify_all();
     }
   }
 
-  CtrlMsg self_ctrl(uint64_t flow) {
-    CtrlMsg m{};
-    m.magic = kCtrlMagic;
-    m.flow = flow;
-    m.num_paths = num_paths;
-    for (int i = 0; i < num_paths; ++i) m.ports[i] = ports[i];
-    std::string ip = net::local_ip();
-    strncpy(m.ip, ip.c_str(), sizeof(m.ip) - 1);
-    return m;
-  }
-
-  void install_flow(uint64_t flow, CtrlMsg const& peer) {
+  void install_flow(uint64_t flow, int paths) {
     auto f = std::make_unique<Flow>();
     f->id = flow;
     f->credit_limit = rwnd_bytes();  // initial grant (pre-first-ack)
-    f->num_paths = std::min(num_paths, peer.num_paths);
-    for (int i = 0; i < f->num_paths; ++i) {
-      sockaddr_in a{};
-      a.sin_family = AF_INET;
-      a.sin_port = htons(peer.ports[i]);
-      inet_pton(AF_INET, peer.ip, &a.sin_addr);
-      f->peer_paths[i] = a;
-    }
+    f->num_paths = paths;
     std::lock_guard<std::mutex> g(mu);
     // self-connections (one endpoint dialing its own metadata, e.g. the
     // in-process plugin fabric) install the same id from both the ctrl
@@ -635,36 +639,7 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
   if (int64_t mbps = env_int("UCCL_TP_PACE_MBPS", 0); mbps > 0)
     impl_->pace_q32 =
         static_cast<uint64_t>((1e9 * 4294967296.0) / (mbps * 1e6));
-  for (int i = 0; i < num_paths; ++i) {
-    int s = ::socket(AF_INET, SOCK_DGRAM, 0);
-    UCCL_CHECK(s >= 0) << "udp socket";
-    int sz = 16 << 20;
-    // FORCE variants bypass net.core.{r,w}mem_max when running as root —
-    // without them loopback drops under bursts and masquerades as loss
-    if (setsockopt(s, SOL_SOCKET, SO_RCVBUFFORCE, &sz, sizeof(sz)) != 0)
-      setsockopt(s, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
-    if (setsockopt(s, SOL_SOCKET, SO_SNDBUFFORCE, &sz, sizeof(sz)) != 0)
-      setsockopt(s, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
-    sockaddr_in a{};
-    a.sin_family = AF_INET;
-    a.sin_addr.s_addr = htonl(INADDR_ANY);
-    UCCL_CHECK(::bind(s, reinterpret_cast<sockaddr*>(&a), sizeof(a)) == 0);
-    socklen_t al = sizeof(a);
-    getsockname(s, reinterpret_cast<sockaddr*>(&a), &al);
-    impl_->socks.push_back(s);
-    impl_->ports.push_back(ntohs(a.sin_port));
-  }
-  {
-    impl_->wake_fd = ::socket(AF_INET, SOCK_DGRAM, 0);
-    sockaddr_in a{};
-    a.sin_family = AF_INET;
-    a.sin_addr.s_addr = htonl(INADDR_ANY);
-    UCCL_CHECK(::bind(impl_->wake_fd, reinterpret_cast<sockaddr*>(&a),
-                      sizeof(a)) == 0);
-    socklen_t al = sizeof(a);
-    getsockname(impl_->wake_fd, reinterpret_cast<sockaddr*>(&a), &al);
-    impl_->wake_port = ntohs(a.sin_port);
-  }
+  impl_->fabric = make_fabric(num_paths, chunk_bytes);
   // Flow ids are assigned by the ACCEPTOR and used verbatim by both
   // sides (wire id == map key). Endpoints can hold flows accepted
   // locally AND flows assigned by remote acceptors, so ids must be
@@ -701,8 +676,6 @@ TransportEndpoint::~TransportEndpoint() {
   ::close(impl_->ctrl_listen);
   if (impl_->ctrl_thread.joinable()) impl_->ctrl_thread.join();
   if (impl_->progress.joinable()) impl_->progress.join();
-  for (int s : impl_->socks) ::close(s);
-  ::close(impl_->wake_fd);
 }
 
 std::string TransportEndpoint::metadata() const {
@@ -718,16 +691,21 @@ uint64_t TransportEndpoint::connect(const std::string& md, uint64_t tag) {
   std::string ip = md.substr(0, pos);
   uint16_t port = static_cast<uint16_t>(atoi(md.c_str() + pos + 1));
   int fd = net::connect_to(ip, port);
-  Impl::CtrlMsg mine = impl_->self_ctrl(0);
-  mine.tag = tag;
-  net::send_all(fd, &mine, sizeof(mine));
-  Impl::CtrlMsg peer{};
-  UCCL_CHECK(net::recv_all(fd, &peer, sizeof(peer)) &&
-             peer.magic == Impl::kCtrlMagic)
-      << "ctrl handshake";
+  Impl::CtrlHello hello{Impl::kCtrlMagic, tag};
+  net::send_all(fd, &hello, sizeof(hello));
+  uint64_t flow = 0;
+  std::string md_peer;
+  UCCL_CHECK(Impl::recv_blob(fd, &flow, &md_peer)) << "ctrl handshake";
+  std::string md_mine = impl_->fabric->create_flow(flow, /*connector=*/true);
+  UCCL_CHECK(Impl::send_blob(fd, flow, md_mine)) << "ctrl handshake";
+  int const my_paths = impl_->fabric->install_peer(flow, md_peer);
+  Impl::CtrlDone done{};
+  UCCL_CHECK(net::recv_all(fd, &done, sizeof(done)) && done.paths > 0 &&
+             my_paths > 0)
+      << "ctrl handshake rejected";
   ::close(fd);
-  impl_->install_flow(peer.flow, peer);
-  return peer.flow;
+  impl_->install_flow(flow, std::min(my_paths, static_cast<int>(done.paths)));
+  return flow;
 }
 
 uint64_t TransportEndpoint::accept(uint64_t* peer_tag) {
@@ -788,6 +766,8 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
     m.capacity = bytes;
     m.user_ptr = static_cast<char*>(ptr);
   }
+  // outside the lock: fabric does its own locking (lock-order safety)
+  impl_->fabric->post_recv_window(flow, msg_id, ptr, bytes);
   std::unique_lock<std::mutex> lk(impl_->mu);
   auto fit = impl_->flows.find(flow);
   if (fit == impl_->flows.end() || !fit->second)
