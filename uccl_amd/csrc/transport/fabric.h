@@ -12,10 +12,10 @@
 //   VerbsFabric — ibverbs RC QP pool (one data QP per path + one ctrl
 //                 QP), RDMA_WRITE_WITH_IMM straight into the
 //                 receiver-advertised message window (FIFO rendezvous),
-//                 IMM = {RID, CSN, LAST}; payload is PLACED by the NIC,
-//                 so chunk events carry no inline payload. Loaded via
-//                 dlopen (vendored ABI subset, verbs_abi.h) like the
-//                 reference's ibverbs_dl.cc.
+//                 IMM = {RID:8, CSN:24}; payload is PLACED by the NIC,
+//                 so chunk events carry no inline payload. Verbs calls
+//                 go through the dlopen'd provider seam
+//                 (verbs_provider.h) like the reference's ibverbs_dl.cc.
 //
 // Selection: UCCL_TP_FABRIC=udp|verbs (default udp; verbs falls back to
 // udp with a warning when no RDMA device is present).
@@ -23,6 +23,7 @@
 
 #include <cstddef>
 #include <cstdint>
+#include <ctime>
 #include <functional>
 #include <memory>
 #include <string>
@@ -99,6 +100,19 @@ class Fabric {
   // when idle. Returns the number of events delivered.
   virtual int poll(std::function<void(FabricEvent const&)> const& cb,
                    int timeout_ms) = 0;
+
+  // Sharded variant for multi-engine progress: engine `shard` of
+  // `nshards` drains its slice of the paths (UDP: sockets i where
+  // i % nshards == shard). Default: shard 0 gets everything, the rest
+  // idle-sleep (single-queue fabrics like the verbs CQ pair).
+  virtual int poll_shard(int shard, int nshards,
+                         std::function<void(FabricEvent const&)> const& cb,
+                         int timeout_ms) {
+    if (shard == 0 || nshards <= 1) return poll(cb, timeout_ms);
+    struct timespec ts {0, timeout_ms * 1000000L};
+    nanosleep(&ts, nullptr);
+    return 0;
+  }
 
   // Unblock a concurrent poll() (e.g. at shutdown or when new TX work
   // arrives and the progress thread is sleeping in poll).

@@ -3,7 +3,9 @@
 #include <sys/socket.h>
 
 #include <algorithm>
+#include <atomic>
 #include <random>
+#include <shared_mutex>
 #include <stdexcept>
 #include <cstdlib>
 #include <cstring>
@@ -91,6 +93,13 @@ struct TransportEndpoint::Flow {
   uint64_t id;
   int num_paths;
 
+  // Per-flow lock + cv: engines touching different flows never contend
+  // (the role of the reference's per-engine state partitioning,
+  // transport.cc:443-466); the endpoint-level maps_mu below only guards
+  // the flow TABLE, taken shared on the hot path.
+  std::mutex fmu;
+  std::condition_variable fcv;
+
   // --- TX direction ---
   uint32_t next_csn = 0;
   std::map<uint32_t, ChunkTx> inflight;   // csn -> chunk
@@ -139,22 +148,41 @@ struct TransportEndpoint::Impl {
   int dup_thres = 32;
   uint64_t rto_base_ns = 20000000;
   int rto_abort_thres = 50;
+  bool spin = false;  // UCCL_TP_SPIN: busy-poll engines + spin-then-wait
   std::unique_ptr<Fabric> fabric;  // the wire plane (udp | verbs)
   int ctrl_listen = -1;
   uint16_t ctrl_port = 0;
   std::thread ctrl_thread;
-  std::thread progress;
+  std::vector<std::thread> engines;  // N progress threads, sharded paths
+  int num_engines = 1;
   std::atomic<bool> stop{false};
 
-  std::mutex mu;
-  std::condition_variable cv;
+  // flow TABLE guard (shared on lookups; flows are never erased while
+  // the endpoint lives, so a Flow* stays valid after the lock drops)
+  std::shared_mutex maps_mu;
   std::unordered_map<uint64_t, std::unique_ptr<Flow>> flows;
+  // accept queue (cold path)
+  std::mutex acc_mu;
+  std::condition_variable acc_cv;
   std::deque<uint64_t> accepted;
   std::deque<uint64_t> accepted_tags;
   std::atomic<uint64_t> next_flow{1};
 
-  Stats st;
+  // counters are engine-parallel now: plain atomics, relaxed ordering
+  struct AtomStats {
+    std::atomic<uint64_t> data_sent{0}, data_recv{0}, acks_sent{0},
+        acks_recv{0}, retransmits{0}, rto_retransmits{0},
+        injected_drops{0}, msgs_sent{0}, msgs_recv{0};
+    std::atomic<double> srtt_us{0}, cwnd{0};
+  } st;
+  std::mutex hist_mu;
   LatencyHist rtt_hist;
+
+  Flow* find_flow(uint64_t id) {
+    std::shared_lock<std::shared_mutex> g(maps_mu);
+    auto it = flows.find(id);
+    return it == flows.end() ? nullptr : it->second.get();
+  }
 
   // ---- helpers ----
   void wake() { fabric->wake(); }
@@ -304,8 +332,8 @@ struct TransportEndpoint::Impl {
       f.cwnd *= 1.0 - beta * std::min(grad, 0.25);
     }
     f.cwnd = std::min(std::max(f.cwnd, 2.0), cwnd_max);
-    st.srtt_us = f.srtt_us;
-    st.cwnd = f.cwnd;
+    st.srtt_us.store(f.srtt_us, std::memory_order_relaxed);
+    st.cwnd.store(f.cwnd, std::memory_order_relaxed);
   }
 
   void ack_chunk(Flow& f, uint32_t csn) {
@@ -317,7 +345,7 @@ struct TransportEndpoint::Impl {
     if (m.acked_bytes >= m.bytes && !m.done) {
       m.done = true;
       ++st.msgs_sent;
-      cv.notify_all();
+      f.fcv.notify_all();
     }
     f.inflight.erase(it);
   }
@@ -327,7 +355,10 @@ struct TransportEndpoint::Impl {
     if (a.credit_cum > f.credit_limit) f.credit_limit = a.credit_cum;
     if (a.ts_echo) {
       double const rtt_us = (now_ns() - a.ts_echo) / 1000.0;
-      rtt_hist.record_us(rtt_us);
+      {
+        std::lock_guard<std::mutex> hg(hist_mu);
+        rtt_hist.record_us(rtt_us);
+      }
       timely_update(f, rtt_us);
     }
     // Placed-chunk fabrics (verbs) carry no timestamp in the data path
@@ -358,7 +389,10 @@ struct TransportEndpoint::Impl {
       }
     }
     if (rtt_fallback > 0) {
-      rtt_hist.record_us(rtt_fallback);
+      {
+        std::lock_guard<std::mutex> hg(hist_mu);
+        rtt_hist.record_us(rtt_fallback);
+      }
       timely_update(f, rtt_fallback);
     }
     // SACK-hole fast retransmit: spraying reorders heavily (and the
@@ -416,7 +450,7 @@ struct TransportEndpoint::Impl {
                        << " exceeds posted capacity " << m.capacity
                        << "; failing flow";
         f.failed = true;
-        cv.notify_all();
+        f.fcv.notify_all();
         return;
       }
       if (h.len && payload) memcpy(m.dest() + h.off, payload, h.len);
@@ -429,7 +463,7 @@ struct TransportEndpoint::Impl {
       }
       if (m.recv_bytes >= m.bytes) {
         ++st.msgs_recv;
-        cv.notify_all();
+        f.fcv.notify_all();
       }
     }
     f.last_rx_ns = now_ns();
@@ -444,8 +478,10 @@ struct TransportEndpoint::Impl {
     // marks the flow failed and fails its blocked senders/receivers.
     int const abort_thres = rto_abort_thres;
     uint64_t const now = now_ns();
+    std::shared_lock<std::shared_mutex> mg(maps_mu);
     for (auto& [fid, fp] : flows) {
       Flow& f = *fp;
+      std::lock_guard<std::mutex> fg(f.fmu);
       if (f.failed) continue;
       for (auto& [csn, c] : f.inflight) {
         uint64_t const rto =
@@ -456,7 +492,7 @@ struct TransportEndpoint::Impl {
             UCCL_LOG_ERROR << "flow " << fid << " csn " << csn
                            << " exceeded RTO abort threshold; marking dead";
             f.failed = true;
-            cv.notify_all();
+            f.fcv.notify_all();
             break;
           }
           ++st.rto_retransmits;
@@ -468,30 +504,57 @@ struct TransportEndpoint::Impl {
     }
   }
 
-  void progress_loop() {
+  // One of N engine threads. Each engine drains its shard of fabric
+  // paths; engine 0 additionally owns the timers (RTO scan, EQDS credit
+  // refresh). Flow state is guarded per flow, so engines only contend
+  // when chunks of the SAME flow land on different shards.
+  void engine_loop(int eng) {
     auto on_event = [this](FabricEvent const& ev) {
-      std::lock_guard<std::mutex> g(mu);
-      auto it = flows.find(ev.flow);
-      if (it == flows.end()) return;
+      Flow* f = find_flow(ev.flow);
+      if (!f) return;
+      std::lock_guard<std::mutex> fg(f->fmu);
       if (ev.kind == FabricEvent::kChunk) {
-        handle_data(*it->second, ev.desc, ev.payload, ev.path);
+        handle_data(*f, ev.desc, ev.payload, ev.path);
       } else if (ev.ctrl_len >= sizeof(AckHdr)) {
         auto const* a = reinterpret_cast<AckHdr const*>(ev.ctrl);
-        if (a->magic == kMagic && a->kind == kAck)
-          handle_ack(*it->second, *a);
+        if (a->magic == kMagic && a->kind == kAck) handle_ack(*f, *a);
       }
     };
+    // Busy-poll with adaptive sleep (the reference's engine discipline,
+    // p2p/util/adaptive_sleeper.h): zero-timeout polls while traffic
+    // flows, degrade to 1ms ticks after a quiet spell.
+    int idle_iters = 0;
+    uint64_t last_timers = 0;
     while (!stop) {
-      fabric->poll(on_event, 5);
-      std::lock_guard<std::mutex> g(mu);
+      int got;
+      if (spin) {
+        got = fabric->poll_shard(eng, num_engines, on_event, 0);
+        if (got) {
+          idle_iters = 0;
+        } else if (++idle_iters > 2000) {
+          got = fabric->poll_shard(eng, num_engines, on_event, 1);
+          if (got) idle_iters = 0;
+        }
+      } else {
+        got = fabric->poll_shard(eng, num_engines, on_event, 1);
+      }
+      (void)got;
+      if (eng != 0) continue;
+      // timers only need ~1ms granularity even when spinning
+      uint64_t const tnow = now_ns();
+      if (spin && tnow - last_timers < 1000000ull) continue;
+      last_timers = tnow;
       rto_scan();
       // paced-EQDS credit refresh: a credit-stalled sender emits no
       // data, so acks (which carry grants) would never flow again —
       // the receiver must top up pulls from the progress loop
       if (eqds_bytes_per_ns > 0) {
         uint64_t const now = now_ns();
+        std::shared_lock<std::shared_mutex> mg(maps_mu);
         for (auto& [id, fp] : flows) {
-          if (!fp || !fp->last_grant_ns) continue;
+          if (!fp) continue;
+          std::lock_guard<std::mutex> fg(fp->fmu);
+          if (!fp->last_grant_ns) continue;
           if (now - fp->last_rx_ns > 500'000'000ull) continue;  // idle
           if (fp->granted >= fp->bytes_received + rwnd_bytes()) continue;
           // refresh acks bypass loss injection: the drop hash is keyed
@@ -582,11 +645,11 @@ struct TransportEndpoint::Impl {
       }
       install_flow(flow, paths);
       {
-        std::lock_guard<std::mutex> g(mu);
+        std::lock_guard<std::mutex> g(acc_mu);
         accepted.push_back(flow);
         accepted_tags.push_back(hello.tag);
       }
-      cv.notify_all();
+      acc_cv.notify_all();
     }
   }
 
@@ -595,7 +658,7 @@ struct TransportEndpoint::Impl {
     f->id = flow;
     f->credit_limit = rwnd_bytes();  // initial grant (pre-first-ack)
     f->num_paths = paths;
-    std::lock_guard<std::mutex> g(mu);
+    std::unique_lock<std::shared_mutex> g(maps_mu);
     // self-connections (one endpoint dialing its own metadata, e.g. the
     // in-process plugin fabric) install the same id from both the ctrl
     // acceptor and the connector: the second install must NOT replace the
@@ -634,12 +697,27 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
       static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 20000)) * 1000;
   impl_->rto_abort_thres =
       static_cast<int>(env_int("UCCL_TP_RTO_ABORT", 50));
+  // busy-poll engines by default on many-core hosts (MI355X servers);
+  // on small boxes the burned cores cost more than the wakeup saves
+  impl_->spin = env_bool(
+      "UCCL_TP_SPIN", std::thread::hardware_concurrency() >= 32);
   // optional sender pacing (the reference's Carousel timing-wheel role;
   // bypassed by default there and here — BYPASS_PACING=1)
   if (int64_t mbps = env_int("UCCL_TP_PACE_MBPS", 0); mbps > 0)
     impl_->pace_q32 =
         static_cast<uint64_t>((1e9 * 4294967296.0) / (mbps * 1e6));
   impl_->fabric = make_fabric(num_paths, chunk_bytes);
+  // Engine-thread count: scale with the machine, not the path count —
+  // on small CPU counts extra engines oversubscribe and ADD latency
+  // (measured: 1 engine on 8 cores matches the round-1 single-thread
+  // rate; engines only win when cores are plentiful, as on real
+  // MI355X hosts). UCCL_TP_ENGINES overrides.
+  int const hw = static_cast<int>(std::thread::hardware_concurrency());
+  int const auto_engines = std::max(1, std::min(4, hw / 8));
+  impl_->num_engines =
+      static_cast<int>(env_int("UCCL_TP_ENGINES", auto_engines));
+  if (impl_->num_engines < 1) impl_->num_engines = 1;
+  if (impl_->num_engines > num_paths) impl_->num_engines = num_paths;
   // Flow ids are assigned by the ACCEPTOR and used verbatim by both
   // sides (wire id == map key). Endpoints can hold flows accepted
   // locally AND flows assigned by remote acceptors, so ids must be
@@ -649,33 +727,41 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
       1;
   impl_->ctrl_listen = net::listen_on(&impl_->ctrl_port);
   impl_->ctrl_thread = std::thread([this] { impl_->ctrl_loop(); });
-  impl_->progress = std::thread([this] { impl_->progress_loop(); });
+  for (int e = 0; e < impl_->num_engines; ++e)
+    impl_->engines.emplace_back([this, e] { impl_->engine_loop(e); });
 }
 
 void TransportEndpoint::close_flow(uint64_t flow) {
-  std::lock_guard<std::mutex> g(impl_->mu);
-  auto it = impl_->flows.find(flow);
-  if (it != impl_->flows.end()) {
-    it->second->failed = true;
-    impl_->cv.notify_all();
+  Flow* f = impl_->find_flow(flow);
+  if (f) {
+    std::lock_guard<std::mutex> g(f->fmu);
+    f->failed = true;
+    f->fcv.notify_all();
   }
 }
 
 void TransportEndpoint::shutdown() {
-  {
-    std::lock_guard<std::mutex> g(impl_->mu);  // lost-wakeup guard
-    if (impl_->stop.exchange(true)) return;
-  }
+  if (impl_->stop.exchange(true)) return;
   ::shutdown(impl_->ctrl_listen, SHUT_RDWR);
   impl_->wake();
-  impl_->cv.notify_all();
+  // wake every blocked send/recv (they re-check stop under their flow
+  // lock) and any accept() waiter
+  {
+    std::shared_lock<std::shared_mutex> mg(impl_->maps_mu);
+    for (auto& [id, fp] : impl_->flows) {
+      std::lock_guard<std::mutex> fg(fp->fmu);
+      fp->fcv.notify_all();
+    }
+  }
+  impl_->acc_cv.notify_all();
 }
 
 TransportEndpoint::~TransportEndpoint() {
   shutdown();
   ::close(impl_->ctrl_listen);
   if (impl_->ctrl_thread.joinable()) impl_->ctrl_thread.join();
-  if (impl_->progress.joinable()) impl_->progress.join();
+  for (auto& e : impl_->engines)
+    if (e.joinable()) e.join();
 }
 
 std::string TransportEndpoint::metadata() const {
@@ -709,8 +795,8 @@ uint64_t TransportEndpoint::connect(const std::string& md, uint64_t tag) {
 }
 
 uint64_t TransportEndpoint::accept(uint64_t* peer_tag) {
-  std::unique_lock<std::mutex> lk(impl_->mu);
-  impl_->cv.wait(lk, [this] {
+  std::unique_lock<std::mutex> lk(impl_->acc_mu);
+  impl_->acc_cv.wait(lk, [this] {
     return !impl_->accepted.empty() || impl_->stop;
   });
   if (impl_->accepted.empty())
@@ -725,12 +811,12 @@ uint64_t TransportEndpoint::accept(uint64_t* peer_tag) {
 void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
                                  size_t bytes) {
   trace::Span span("transport", "send_msg");
+  Flow* fp = impl_->find_flow(flow);
+  UCCL_CHECK(fp != nullptr) << "unknown flow " << flow;
+  Flow& f = *fp;
   std::shared_ptr<MsgTx> m;
   {
-    std::lock_guard<std::mutex> g(impl_->mu);
-    auto it = impl_->flows.find(flow);
-    UCCL_CHECK(it != impl_->flows.end()) << "unknown flow " << flow;
-    Flow& f = *it->second;
+    std::lock_guard<std::mutex> g(f.fmu);
     m = std::make_shared<MsgTx>();
     m->id = f.next_tx_msg++;
     m->ptr = static_cast<char const*>(ptr);
@@ -739,21 +825,32 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
     impl_->pump_tx(f);
   }
   impl_->wake();
-  std::unique_lock<std::mutex> lk(impl_->mu);
-  Flow& fl = *impl_->flows[flow];
-  impl_->cv.wait(lk, [&] { return m->done || impl_->stop || fl.failed; });
-  if (fl.failed) throw std::runtime_error("transport flow failed (RTO abort)");
+  std::unique_lock<std::mutex> lk(f.fmu);
+  if (impl_->spin) {
+    // ~50us spin before sleeping: saves the futex round trip on the
+    // common fast ack (engines are busy-polling in this mode)
+    for (int i = 0; i < 2000 && !m->done && !impl_->stop && !f.failed;
+         ++i) {
+      lk.unlock();
+      #if defined(__x86_64__)
+      __builtin_ia32_pause();
+      #endif
+      lk.lock();
+    }
+  }
+  f.fcv.wait(lk, [&] { return m->done || impl_->stop || f.failed; });
+  if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (!m->done) throw std::runtime_error("transport closed during send");
 }
 
 void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
   trace::Span span("transport", "recv_msg");
+  Flow* fp = impl_->find_flow(flow);
+  UCCL_CHECK(fp != nullptr) << "unknown flow " << flow;
+  Flow& f = *fp;
   uint64_t msg_id;
   {
-    std::lock_guard<std::mutex> g(impl_->mu);
-    auto it = impl_->flows.find(flow);
-    UCCL_CHECK(it != impl_->flows.end()) << "unknown flow " << flow;
-    Flow& f = *it->second;
+    std::lock_guard<std::mutex> g(f.fmu);
     msg_id = f.next_post_msg++;
     auto& m = f.rxmsgs[msg_id];
     if (m.known && m.bytes > bytes)
@@ -768,12 +865,8 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
   }
   // outside the lock: fabric does its own locking (lock-order safety)
   impl_->fabric->post_recv_window(flow, msg_id, ptr, bytes);
-  std::unique_lock<std::mutex> lk(impl_->mu);
-  auto fit = impl_->flows.find(flow);
-  if (fit == impl_->flows.end() || !fit->second)
-    throw std::runtime_error("transport flow vanished");
-  Flow& f = *fit->second;
-  impl_->cv.wait(lk, [&] {
+  std::unique_lock<std::mutex> lk(f.fmu);
+  f.fcv.wait(lk, [&] {
     auto it = f.rxmsgs.find(msg_id);
     return (it != f.rxmsgs.end() && it->second.known &&
             it->second.recv_bytes >= it->second.bytes) ||
@@ -785,8 +878,20 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
 }
 
 Stats TransportEndpoint::stats() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
-  Stats st = impl_->st;
+  Stats st;
+  auto& a = impl_->st;
+  st.data_sent = a.data_sent.load(std::memory_order_relaxed);
+  st.data_recv = a.data_recv.load(std::memory_order_relaxed);
+  st.acks_sent = a.acks_sent.load(std::memory_order_relaxed);
+  st.acks_recv = a.acks_recv.load(std::memory_order_relaxed);
+  st.retransmits = a.retransmits.load(std::memory_order_relaxed);
+  st.rto_retransmits = a.rto_retransmits.load(std::memory_order_relaxed);
+  st.injected_drops = a.injected_drops.load(std::memory_order_relaxed);
+  st.msgs_sent = a.msgs_sent.load(std::memory_order_relaxed);
+  st.msgs_recv = a.msgs_recv.load(std::memory_order_relaxed);
+  st.srtt_us = a.srtt_us.load(std::memory_order_relaxed);
+  st.cwnd = a.cwnd.load(std::memory_order_relaxed);
+  std::lock_guard<std::mutex> g(impl_->hist_mu);
   st.rtt_p50_us = impl_->rtt_hist.percentile_us(50);
   st.rtt_p99_us = impl_->rtt_hist.percentile_us(99);
   return st;

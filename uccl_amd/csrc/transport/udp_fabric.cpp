@@ -91,7 +91,6 @@ class UdpFabric final : public Fabric {
     socklen_t al = sizeof(a);
     getsockname(wake_fd_, reinterpret_cast<sockaddr*>(&a), &al);
     wake_port_ = ntohs(a.sin_port);
-    rxbuf_.resize(sizeof(DataWire) + 65536 + 64);
   }
 
   ~UdpFabric() override {
@@ -167,42 +166,80 @@ class UdpFabric final : public Fabric {
 
   int poll(std::function<void(FabricEvent const&)> const& cb,
            int timeout_ms) override {
+    return poll_shard(0, 1, cb, timeout_ms);
+  }
+
+  // Engine-sharded drain with recvmmsg batching: one syscall pulls up
+  // to kBatch datagrams per socket (the reference's CQ batch polling
+  // role, uc_handle_completion). Each engine owns sockets
+  // i % nshards == shard; the wake socket belongs to shard 0.
+  int poll_shard(int shard, int nshards,
+                 std::function<void(FabricEvent const&)> const& cb,
+                 int timeout_ms) override {
+    constexpr int kBatch = 64;
+    size_t const slot = sizeof(DataWire) + 65536 + 64;
+    thread_local std::vector<char> bufs;
+    if (bufs.size() < kBatch * slot) bufs.resize(kBatch * slot);
+
     std::vector<pollfd> pfds;
-    pfds.reserve(socks_.size() + 1);
-    for (int s : socks_) pfds.push_back({s, POLLIN, 0});
-    pfds.push_back({wake_fd_, POLLIN, 0});
+    std::vector<int> paths;  // original path index per pollfd
+    for (int i = shard; i < static_cast<int>(socks_.size()); i += nshards) {
+      pfds.push_back({socks_[i], POLLIN, 0});
+      paths.push_back(i);
+    }
+    if (shard == 0) {
+      pfds.push_back({wake_fd_, POLLIN, 0});
+      paths.push_back(-1);
+    }
+    if (pfds.empty()) {
+      struct timespec ts {0, timeout_ms * 1000000L};
+      nanosleep(&ts, nullptr);
+      return 0;
+    }
     (void)::poll(pfds.data(), pfds.size(), timeout_ms);
+
+    mmsghdr msgs[kBatch];
+    iovec iov[kBatch];
     int delivered = 0;
     for (size_t i = 0; i < pfds.size(); ++i) {
       while (true) {
-        ssize_t n = recvfrom(pfds[i].fd, rxbuf_.data(), rxbuf_.size(),
-                             MSG_DONTWAIT, nullptr, nullptr);
-        if (n <= 0) break;
-        if (n < static_cast<ssize_t>(sizeof(CtrlWire))) continue;
-        auto const* cw = reinterpret_cast<CtrlWire const*>(rxbuf_.data());
-        if (cw->magic != kFabMagic || cw->kind == kFabWake) continue;
-        FabricEvent ev{};
-        ev.path = static_cast<int>(i);
-        if (cw->kind == kFabData &&
-            n >= static_cast<ssize_t>(sizeof(DataWire))) {
-          auto const* h = reinterpret_cast<DataWire const*>(rxbuf_.data());
-          // wire len must match the datagram (truncation/corruption guard)
-          if (sizeof(DataWire) + h->len != static_cast<size_t>(n)) continue;
-          ev.kind = FabricEvent::kChunk;
-          ev.flow = h->flow;
-          ev.desc = ChunkDesc{h->flow, h->msg_id, h->msg_bytes,
-                              h->off,  h->len,    h->csn,     h->ts_ns};
-          ev.payload = rxbuf_.data() + sizeof(DataWire);
-        } else if (cw->kind == kFabCtrl) {
-          ev.kind = FabricEvent::kCtrl;
-          ev.flow = cw->flow;
-          ev.ctrl = rxbuf_.data() + sizeof(CtrlWire);
-          ev.ctrl_len = static_cast<size_t>(n) - sizeof(CtrlWire);
-        } else {
-          continue;
+        for (int k = 0; k < kBatch; ++k) {
+          iov[k] = {bufs.data() + k * slot, slot};
+          memset(&msgs[k].msg_hdr, 0, sizeof(msghdr));
+          msgs[k].msg_hdr.msg_iov = &iov[k];
+          msgs[k].msg_hdr.msg_iovlen = 1;
         }
-        cb(ev);
-        ++delivered;
+        int n = recvmmsg(pfds[i].fd, msgs, kBatch, MSG_DONTWAIT, nullptr);
+        if (n <= 0) break;
+        for (int k = 0; k < n; ++k) {
+          char const* buf = bufs.data() + k * slot;
+          size_t const len = msgs[k].msg_len;
+          if (len < sizeof(CtrlWire)) continue;
+          auto const* cw = reinterpret_cast<CtrlWire const*>(buf);
+          if (cw->magic != kFabMagic || cw->kind == kFabWake) continue;
+          FabricEvent ev{};
+          ev.path = paths[i] >= 0 ? paths[i] : 0;
+          if (cw->kind == kFabData && len >= sizeof(DataWire)) {
+            auto const* h = reinterpret_cast<DataWire const*>(buf);
+            // wire len must match the datagram (truncation guard)
+            if (sizeof(DataWire) + h->len != len) continue;
+            ev.kind = FabricEvent::kChunk;
+            ev.flow = h->flow;
+            ev.desc = ChunkDesc{h->flow, h->msg_id, h->msg_bytes,
+                                h->off,  h->len,    h->csn,     h->ts_ns};
+            ev.payload = buf + sizeof(DataWire);
+          } else if (cw->kind == kFabCtrl) {
+            ev.kind = FabricEvent::kCtrl;
+            ev.flow = cw->flow;
+            ev.ctrl = buf + sizeof(CtrlWire);
+            ev.ctrl_len = len - sizeof(CtrlWire);
+          } else {
+            continue;
+          }
+          cb(ev);
+          ++delivered;
+        }
+        if (n < kBatch) break;
       }
     }
     return delivered;
@@ -235,7 +272,6 @@ class UdpFabric final : public Fabric {
   uint16_t wake_port_ = 0;
   std::mutex pmu_;
   std::unordered_map<uint64_t, PeerPaths> peers_;
-  std::vector<char> rxbuf_;
 };
 
 }  // namespace
