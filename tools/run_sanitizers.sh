@@ -9,15 +9,20 @@ OUT=${TMPDIR:-/tmp}
 for san in thread address,undefined; do
   tag=${san%%,*}
   g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_transport.cpp \
-      uccl_amd/csrc/transport/reliable.cpp uccl_amd/csrc/core/trace.cpp \
-      -o "$OUT/san_tp_$tag" -pthread
+      uccl_amd/csrc/transport/reliable.cpp \
+      uccl_amd/csrc/transport/udp_fabric.cpp \
+      uccl_amd/csrc/transport/verbs_fabric.cpp \
+      uccl_amd/csrc/core/trace.cpp \
+      -o "$OUT/san_tp_$tag" -pthread -ldl
   g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_codec_ukernel.cpp \
       uccl_amd/csrc/p2p/compress.cpp uccl_amd/csrc/ukernel/ukernel.cpp \
       uccl_amd/csrc/core/trace.cpp -o "$OUT/san_cu_$tag" -pthread -lz
   g++ -O1 -g -std=c++17 -fsanitize=$san -D__HIP_PLATFORM_AMD__=1 \
       -I/opt/rocm/include tools/san_p2p.cpp \
       uccl_amd/csrc/p2p/endpoint.cpp uccl_amd/csrc/transport/reliable.cpp \
-      uccl_amd/csrc/core/trace.cpp -o "$OUT/san_p2p_$tag" -pthread \
+      uccl_amd/csrc/transport/udp_fabric.cpp \
+      uccl_amd/csrc/transport/verbs_fabric.cpp \
+      uccl_amd/csrc/core/trace.cpp -o "$OUT/san_p2p_$tag" -pthread -ldl \
       -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,/opt/rocm/lib
   # net plugin: dlopen harness against a sanitizer-built .so. The TSan
   # build swaps timed cv waits for untimed ones (UCCL_SAN_NO_TIMED_WAIT):
@@ -27,8 +32,11 @@ for san in thread address,undefined; do
   [ "$tag" = thread ] && extra=-DUCCL_SAN_NO_TIMED_WAIT
   g++ -O1 -g -std=c++17 -fsanitize=$san $extra -fPIC -shared \
       uccl_amd/csrc/plugin/tcp_plugin.cpp \
-      uccl_amd/csrc/transport/reliable.cpp uccl_amd/csrc/core/trace.cpp \
-      -o "$OUT/librccl-net-uccl-$tag.so" -pthread
+      uccl_amd/csrc/transport/reliable.cpp \
+      uccl_amd/csrc/transport/udp_fabric.cpp \
+      uccl_amd/csrc/transport/verbs_fabric.cpp \
+      uccl_amd/csrc/core/trace.cpp \
+      -o "$OUT/librccl-net-uccl-$tag.so" -pthread -ldl
   g++ -O1 -g -std=c++17 -fsanitize=$san \
       uccl_amd/csrc/plugin/plugin_test_main.cpp \
       -o "$OUT/plugin_test_$tag" -ldl -pthread
