@@ -1,7 +1,10 @@
 #!/usr/bin/env python3
 """EP dispatch/combine latency benchmark (BASELINE.json config:
 "DeepEP low-latency dispatch+combine, 8 experts x 4096 tokens x 7168
-hidden bf16, 8 MI355X intranode").
+hidden bf16, 8 MI355X intranode"). Default topk=8 — the reference
+DeepEP workload shape (experimental/misc/ep_results.md:18, 256 experts
+top-8 moves 4x the data of top-2; the r1 default of top-2 understated
+the work per token).
 
 Runs at any world size (1..8); for N>1 launch one rank per GPU via
 torch.distributed.run with gloo rendezvous. Prints one JSON line with p50
@@ -22,7 +25,7 @@ def main():
     p.add_argument("--tokens", type=int, default=4096)
     p.add_argument("--hidden", type=int, default=7168)
     p.add_argument("--experts", type=int, default=8)
-    p.add_argument("--topk", type=int, default=2)
+    p.add_argument("--topk", type=int, default=8)
     p.add_argument("--iters", type=int, default=30)
     p.add_argument("--warmup", type=int, default=5)
     args = p.parse_args()

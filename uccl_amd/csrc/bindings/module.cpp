@@ -382,6 +382,83 @@ PYBIND11_MODULE(_C, m) {
                      .device(x.device()));
              return py::make_tuple(recv_x, counts);
            })
+      // phase-split halves (DeepEP SEND|RECV split + cached-handle
+      // replay; see EpBuffer::dispatch_send)
+      .def("dispatch_send",
+           [](uccl::ep::EpBuffer& b, at::Tensor x, at::Tensor topk_idx,
+              bool reuse_plan) {
+             TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+             TORCH_CHECK(topk_idx.scalar_type() == at::kLong &&
+                         topk_idx.is_cuda() && topk_idx.is_contiguous());
+             auto const& v = b.view();
+             TORCH_CHECK(x.dim() == 2 && x.size(1) == v.hidden);
+             TORCH_CHECK(topk_idx.dim() == 2 && topk_idx.size(1) == v.topk);
+             b.dispatch_send(x.data_ptr(), topk_idx.data_ptr<int64_t>(),
+                             static_cast<int>(x.size(0)), reuse_plan,
+                             current_stream(b.device()));
+           },
+           py::arg("x"), py::arg("topk_idx"),
+           py::arg("reuse_plan") = false)
+      .def("dispatch_recv",
+           [](uccl::ep::EpBuffer& b, at::Tensor counts) {
+             auto const& v = b.view();
+             TORCH_CHECK(counts.is_cuda() && counts.is_contiguous() &&
+                         counts.scalar_type() == at::kInt &&
+                         counts.numel() == v.local_experts * v.world);
+             b.dispatch_recv(counts.data_ptr<int>(),
+                             current_stream(b.device()));
+             return counts;
+           })
+      .def("recv_x_view",
+           [](uccl::ep::EpBuffer& b, py::object dtype_obj) {
+             auto const& v = b.view();
+             auto dt = v.disp_fp8
+                           ? at::kFloat8_e4m3fn
+                           : (v.elem_size == 2 ? at::kBFloat16 : at::kFloat);
+             return at::from_blob(
+                 b.recv_x_ptr(),
+                 {v.local_experts,
+                  static_cast<int64_t>(v.world) * v.max_tokens, v.hidden},
+                 at::TensorOptions().dtype(dt).device(
+                     at::Device(at::kCUDA, b.device())));
+           },
+           py::arg("dtype") = py::none())
+      .def("recv_scale_view",
+           [](uccl::ep::EpBuffer& b) {
+             auto const& v = b.view();
+             TORCH_CHECK(v.disp_fp8, "scales exist only in fp8 mode");
+             return at::from_blob(
+                 b.recv_scale_ptr(),
+                 {v.local_experts,
+                  static_cast<int64_t>(v.world) * v.max_tokens,
+                  v.hidden / 128},
+                 at::TensorOptions().dtype(at::kFloat).device(
+                     at::Device(at::kCUDA, b.device())));
+           })
+      .def("combine_send",
+           [](uccl::ep::EpBuffer& b, at::Tensor expert_out) {
+             TORCH_CHECK(expert_out.is_cuda() && expert_out.is_contiguous());
+             TORCH_CHECK(expert_out.element_size() == b.view().elem_size);
+             b.combine_send(expert_out.data_ptr(),
+                            current_stream(b.device()));
+           })
+      .def("combine_recv",
+           [](uccl::ep::EpBuffer& b, at::Tensor out, at::Tensor topk_idx,
+              at::Tensor topk_w) {
+             auto const& v = b.view();
+             TORCH_CHECK(out.is_cuda() && out.is_contiguous() &&
+                         out.element_size() == v.elem_size &&
+                         out.dim() == 2 && out.size(0) == topk_idx.size(0) &&
+                         out.size(1) == v.hidden);
+             TORCH_CHECK(topk_idx.scalar_type() == at::kLong &&
+                         topk_idx.is_contiguous());
+             TORCH_CHECK(topk_w.scalar_type() == at::kFloat &&
+                         topk_w.is_contiguous());
+             b.combine_recv(out.data_ptr(), topk_idx.data_ptr<int64_t>(),
+                            topk_w.data_ptr<float>(),
+                            current_stream(b.device()));
+             return out;
+           })
       .def("combine",
            [](uccl::ep::EpBuffer& b, at::Tensor expert_out,
               at::Tensor topk_idx, at::Tensor topk_w) {

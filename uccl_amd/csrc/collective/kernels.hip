@@ -65,7 +65,8 @@ __global__ void k_signal_wait(CommView cv, uint64_t val, unsigned wait_mask) {
   if (threadIdx.x < static_cast<unsigned>(cv.world)) {
     st_release_sys(flag_ptr(cv.peers[threadIdx.x], cv.rank, cv.channel), val);
     if ((wait_mask >> threadIdx.x) & 1u) {
-      wait_flag_ge(flag_ptr(cv.peers[cv.rank], threadIdx.x, cv.channel), val);
+      wait_flag_ge(flag_ptr(cv.peers[cv.rank], threadIdx.x, cv.channel), val,
+                   cv.rank, static_cast<int>(threadIdx.x), cv.channel);
     }
   }
 }
@@ -584,7 +585,8 @@ __global__ void k_signal_peer(CommView cv, int dst, int ch, uint64_t val) {
 // Wait until our flags[src][ch] >= val
 __global__ void k_wait_peer(CommView cv, int src, int ch, uint64_t val) {
   if (threadIdx.x == 0 && blockIdx.x == 0)
-    wait_flag_ge(flag_ptr(cv.peers[cv.rank], src, ch), val);
+    wait_flag_ge(flag_ptr(cv.peers[cv.rank], src, ch), val, cv.rank, src,
+                 ch);
 }
 
 // Copy from a peer's heap (+byte offset from heap base) into local memory.

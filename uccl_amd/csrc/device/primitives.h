@@ -57,8 +57,11 @@ __device__ __forceinline__ void backoff() { __builtin_amdgcn_s_sleep(2); }
 // Bounded spin-wait: waits until *flag >= target.  Traps (aborting the
 // kernel, and with it the process) instead of hanging the GPU forever —
 // a deadlocked collective should kill the job, not wedge the node.
+// rank/peer/ch identify WHICH wait stalled in the abort log (real-xGMI
+// bring-up debuggability; printf costs nothing until the timeout path).
 __device__ __forceinline__ void wait_flag_ge(uint64_t const* flag,
-                                             uint64_t target) {
+                                             uint64_t target, int rank = -1,
+                                             int peer = -1, int ch = -1) {
   // ~2.7e8 polls with s_sleep backoff ≈ tens of seconds: far beyond any
   // sane collective wait, but short enough that a deadlock aborts the
   // kernel instead of wedging the GPU (gpurun strike avoidance).
@@ -66,6 +69,12 @@ __device__ __forceinline__ void wait_flag_ge(uint64_t const* flag,
     if (ld_acquire_sys(flag) >= target) return;
     backoff();
   }
+  if (threadIdx.x % 64 == 0)
+    printf(
+        "uccl: device wait TIMEOUT rank=%d peer=%d ch=%d want>=%llu "
+        "last=%llu\n",
+        rank, peer, ch, (unsigned long long)target,
+        (unsigned long long)ld_relaxed_sys(flag));
   __builtin_trap();
 }
 

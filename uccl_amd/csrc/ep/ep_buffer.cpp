@@ -180,24 +180,35 @@ void EpBuffer::connect(const std::vector<std::string>& handles) {
   connected_ = true;
 }
 
-void EpBuffer::dispatch(void const* x, int64_t const* topk_idx,
-                        int num_tokens, int* out_counts,
-                        hipStream_t stream) {
+void EpBuffer::dispatch_send(void const* x, int64_t const* topk_idx,
+                             int num_tokens, bool reuse_plan,
+                             hipStream_t stream) {
   UCCL_CHECK(connected_ || world_ == 1) << "connect() not called";
   UCCL_CHECK(num_tokens <= v_.max_tokens)
       << num_tokens << " tokens > max_tokens " << v_.max_tokens;
+  UCCL_CHECK(!reuse_plan || last_num_tokens_ == num_tokens)
+      << "cached dispatch requires the same token count as the plan";
   ++v_.seq;
-  launch_ep_dispatch(v_, x, topk_idx, num_tokens, out_counts, stream);
+  launch_ep_dispatch_send(v_, x, topk_idx, num_tokens, reuse_plan, stream);
+  last_num_tokens_ = num_tokens;
+}
+
+void EpBuffer::dispatch_recv(int* out_counts, hipStream_t stream) {
+  launch_ep_dispatch_recv(v_, out_counts, stream);
   // pinned host copy of counts (combine proxy shipping needs them)
   UCCL_CHECK_HIP(hipMemcpyAsync(host_counts_, out_counts,
                                 sizeof(int) * v_.local_experts * world_,
                                 hipMemcpyDeviceToHost, stream));
-  last_num_tokens_ = num_tokens;
 }
 
-void EpBuffer::combine(void const* expert_out, void* out,
-                       int64_t const* topk_idx, float const* topk_w,
-                       hipStream_t stream) {
+void EpBuffer::dispatch(void const* x, int64_t const* topk_idx,
+                        int num_tokens, int* out_counts,
+                        hipStream_t stream) {
+  dispatch_send(x, topk_idx, num_tokens, /*reuse_plan=*/false, stream);
+  dispatch_recv(out_counts, stream);
+}
+
+void EpBuffer::combine_send(void const* expert_out, hipStream_t stream) {
   UCCL_CHECK(last_num_tokens_ >= 0) << "combine without a prior dispatch";
   launch_ep_combine_send(v_, expert_out, stream);
   if (v_.proxy_mask) {
@@ -208,8 +219,19 @@ void EpBuffer::combine(void const* expert_out, void* out,
     UCCL_CHECK_HIP(hipEventRecord(ev, stream));
     proxy_->enqueue_combine(expert_out, v_.seq, ev, host_counts_);
   }
+}
+
+void EpBuffer::combine_recv(void* out, int64_t const* topk_idx,
+                            float const* topk_w, hipStream_t stream) {
   launch_ep_combine_finish(v_, out, topk_idx, topk_w, last_num_tokens_,
                            stream);
+}
+
+void EpBuffer::combine(void const* expert_out, void* out,
+                       int64_t const* topk_idx, float const* topk_w,
+                       hipStream_t stream) {
+  combine_send(expert_out, stream);
+  combine_recv(out, topk_idx, topk_w, stream);
 }
 
 }  // namespace ep

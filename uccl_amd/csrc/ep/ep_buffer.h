@@ -31,10 +31,30 @@ class EpBuffer {
   void dispatch(void const* x, int64_t const* topk_idx, int num_tokens,
                 int* out_counts, hipStream_t stream);
 
+  // Phase-split dispatch (DeepEP SEND|RECV split, internode_ll.cu:62):
+  // send launches plan/copy/publish; recv launches the count wait —
+  // callers overlap compute between the two (recv-hook support).
+  // reuse_plan=true skips the plan/prefix kernels and replays the
+  // previous dispatch's compaction lists (DeepEP cached-handle mode,
+  // intranode.cu:150) — only valid when topk_idx is unchanged.
+  void dispatch_send(void const* x, int64_t const* topk_idx,
+                     int num_tokens, bool reuse_plan, hipStream_t stream);
+  void dispatch_recv(int* out_counts, hipStream_t stream);
+
   // expert_out: [local_experts, world*max_tokens, hidden];
   // out: [num_tokens, hidden]; topk_w: [num_tokens, topk] f32.
   void combine(void const* expert_out, void* out, int64_t const* topk_idx,
                float const* topk_w, hipStream_t stream);
+
+  // Phase-split combine: send returns expert outputs to source cells and
+  // signals; recv waits for all ranks' returns then reduces.
+  void combine_send(void const* expert_out, hipStream_t stream);
+  void combine_recv(void* out, int64_t const* topk_idx,
+                    float const* topk_w, hipStream_t stream);
+
+  // true when a plan from a prior dispatch can be replayed (same
+  // topk shape; the heap's plan scratch is untouched since)
+  bool plan_cached() const { return last_num_tokens_ >= 0; }
 
   const EpView& view() const { return v_; }
   void* recv_x_ptr() const {

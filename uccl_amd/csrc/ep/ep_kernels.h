@@ -10,6 +10,12 @@ namespace ep {
 void launch_ep_dispatch(const EpView& v, void const* x,
                         int64_t const* topk_idx, int num_tokens,
                         int* out_counts, hipStream_t s);
+// phase-split halves of launch_ep_dispatch (SEND|RECV, recv-hook support)
+void launch_ep_dispatch_send(const EpView& v, void const* x,
+                             int64_t const* topk_idx, int num_tokens,
+                             bool reuse_plan, hipStream_t s);
+void launch_ep_dispatch_recv(const EpView& v, int* out_counts,
+                             hipStream_t s);
 void launch_ep_combine_send(const EpView& v, void const* expert_out,
                             hipStream_t s);
 void launch_ep_combine_finish(const EpView& v, void* out,

@@ -148,15 +148,33 @@ __global__ void k_ep_dispatch_plan(EpView v,
                           (static_cast<uint32_t>(k) << 24);
 }
 
-// single block: exclusive prefix of per-expert counts -> plan[e][1]
-// (egress row offsets for the proxy path)
+// Exclusive prefix of per-expert counts -> plan[e][1] (egress row
+// offsets for the proxy path). 256-thread Hillis-Steele scan in LDS per
+// tile of experts — the round-1 single-thread walk serialized at 256
+// experts.
 __global__ void k_ep_plan_prefix(EpView v) {
-  if (threadIdx.x != 0) return;
-  uint32_t acc = 0;
-  for (int e = 0; e < v.num_experts; ++e) {
-    uint32_t* plan = plan_ptr(v.peers[v.rank], v, e);
-    plan[1] = acc;
-    acc += plan[0];
+  __shared__ uint32_t tile[256];
+  uint32_t carry = 0;
+  for (int base = 0; base < v.num_experts; base += 256) {
+    int const e = base + static_cast<int>(threadIdx.x);
+    uint32_t const cnt =
+        e < v.num_experts ? plan_ptr(v.peers[v.rank], v, e)[0] : 0;
+    tile[threadIdx.x] = cnt;
+    __syncthreads();
+#pragma unroll
+    for (int off = 1; off < 256; off <<= 1) {
+      uint32_t const add = threadIdx.x >= static_cast<unsigned>(off)
+                               ? tile[threadIdx.x - off]
+                               : 0;
+      __syncthreads();
+      tile[threadIdx.x] += add;
+      __syncthreads();
+    }
+    if (e < v.num_experts)
+      plan_ptr(v.peers[v.rank], v, e)[1] = carry + tile[threadIdx.x] - cnt;
+    uint32_t const tot = tile[255];
+    __syncthreads();
+    carry += tot;
   }
 }
 
@@ -227,12 +245,15 @@ __global__ void k_ep_dispatch_publish(EpView v) {
   }
 }
 
-// single-block wait: spin until every (local_expert, src) count carries this
-// seq tag, then write the plain counts into out_counts [local_experts][world]
+// sharded wait: spin until every (local_expert, src) count carries this
+// seq tag, then write the plain counts into out_counts
+// [local_experts][world]. Grid-strided so 256-expert shapes spread the
+// spinning across blocks instead of serializing in one.
 __global__ void k_ep_dispatch_wait(EpView v,
                                    int* __restrict__ out_counts) {
   int const n = v.local_experts * v.world;
-  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
     int const le = i / v.world;
     int const src = i % v.world;
     uint64_t const* p = disp_count_ptr(v.peers[v.rank], v, le, src);
@@ -240,7 +261,13 @@ __global__ void k_ep_dispatch_wait(EpView v,
     for (uint64_t it = 0;; ++it) {
       got = ld_acquire_sys(p);
       if ((got >> 32) == v.seq) break;
-      if (it > (1ull << 28)) __builtin_trap();
+      if (it > (1ull << 28)) {
+        printf("uccl_ep: dispatch wait TIMEOUT rank=%d le=%d src=%d "
+               "seq=%llu got_tag=%llu\n",
+               v.rank, le, src, (unsigned long long)v.seq,
+               (unsigned long long)(got >> 32));
+        __builtin_trap();
+      }
       backoff();
     }
     out_counts[i] = static_cast<int>(got & 0xffffffffu);
@@ -297,7 +324,12 @@ __global__ void k_ep_combine_wait(EpView v) {
     uint64_t const* p = comb_flag_ptr(v.peers[v.rank], v, threadIdx.x);
     for (uint64_t it = 0;; ++it) {
       if (ld_acquire_sys(p) >= v.seq) break;
-      if (it > (1ull << 28)) __builtin_trap();
+      if (it > (1ull << 28)) {
+        printf("uccl_ep: combine wait TIMEOUT rank=%d src=%d seq=%llu\n",
+               v.rank, static_cast<int>(threadIdx.x),
+               (unsigned long long)v.seq);
+        __builtin_trap();
+      }
       backoff();
     }
   }
@@ -369,17 +401,32 @@ __global__ void k_ep_combine_reduce(EpView v, void* __restrict__ out,
 // launchers
 // ---------------------------------------------------------------------------
 
-void launch_ep_dispatch(const EpView& v, void const* x,
-                        int64_t const* topk_idx, int num_tokens,
-                        int* out_counts, hipStream_t s) {
-  size_t const smem = 257 * sizeof(uint32_t);
-  k_ep_dispatch_plan<<<v.num_experts, 256, smem, s>>>(v, topk_idx,
-                                                      num_tokens);
-  k_ep_plan_prefix<<<1, 64, 0, s>>>(v);
+void launch_ep_dispatch_send(const EpView& v, void const* x,
+                             int64_t const* topk_idx, int num_tokens,
+                             bool reuse_plan, hipStream_t s) {
+  if (!reuse_plan) {
+    size_t const smem = 257 * sizeof(uint32_t);
+    k_ep_dispatch_plan<<<v.num_experts, 256, smem, s>>>(v, topk_idx,
+                                                        num_tokens);
+    k_ep_plan_prefix<<<1, 256, 0, s>>>(v);
+  }
   k_ep_dispatch_copy<<<v.num_experts * fanout_for(v.num_experts), 256, 0,
                        s>>>(v, x);
   k_ep_dispatch_publish<<<1, 256, 0, s>>>(v);
-  k_ep_dispatch_wait<<<1, 256, 0, s>>>(v, out_counts);
+}
+
+void launch_ep_dispatch_recv(const EpView& v, int* out_counts,
+                             hipStream_t s) {
+  int const n = v.local_experts * v.world;
+  int const grid = std::min(64, (n + 255) / 256 + 1);
+  k_ep_dispatch_wait<<<grid, 256, 0, s>>>(v, out_counts);
+}
+
+void launch_ep_dispatch(const EpView& v, void const* x,
+                        int64_t const* topk_idx, int num_tokens,
+                        int* out_counts, hipStream_t s) {
+  launch_ep_dispatch_send(v, x, topk_idx, num_tokens, false, s);
+  launch_ep_dispatch_recv(v, out_counts, s);
 }
 
 void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,

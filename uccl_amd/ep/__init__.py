@@ -95,10 +95,52 @@ class Buffer:
         source rank r live at [e, r*max_tokens : + recv_count[e, r]]."""
         return self._b.dispatch(x, topk_idx)
 
+    def dispatch_send(self, x: torch.Tensor, topk_idx: torch.Tensor,
+                      reuse_plan: bool = False) -> None:
+        """SEND phase only (DeepEP phase split, internode_ll.cu:62):
+        plans (unless reuse_plan replays the cached compaction lists),
+        copies tokens into destination slot arrays over xGMI and
+        publishes counts. Pair with dispatch_recv(); compute launched in
+        between overlaps the communication."""
+        self._b.dispatch_send(x, topk_idx, reuse_plan)
+
+    def dispatch_recv(self, counts: Optional[torch.Tensor] = None
+                      ) -> torch.Tensor:
+        """RECV phase: waits for every (expert, src) count of this seq;
+        fills (or allocates) recv_count [local_E, world] int32."""
+        if counts is None:
+            counts = torch.empty(self.local_experts, self.world,
+                                 dtype=torch.int32, device="cuda")
+        return self._b.dispatch_recv(counts)
+
+    def recv_x_view(self) -> torch.Tensor:
+        """Zero-copy view of the dispatch slot arrays
+        [local_E, world*max_tokens, hidden] (valid after dispatch_recv)."""
+        return self._b.recv_x_view()
+
+    def recv_scale_view(self) -> torch.Tensor:
+        return self._b.recv_scale_view()
+
     def combine(self, expert_out: torch.Tensor, topk_idx: torch.Tensor,
                 topk_weights: torch.Tensor) -> torch.Tensor:
         return self._b.combine(expert_out, topk_idx,
                                topk_weights.float().contiguous())
+
+    def combine_send(self, expert_out: torch.Tensor) -> None:
+        """SEND phase of combine: return expert outputs to their source
+        (token, k) cells and signal completion to every rank."""
+        self._b.combine_send(expert_out)
+
+    def combine_recv(self, topk_idx: torch.Tensor,
+                     topk_weights: torch.Tensor,
+                     out: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """RECV phase of combine: wait for all ranks' returns, then the
+        fp32 top-k weighted reduction (into `out` when given)."""
+        if out is None:
+            out = torch.empty(topk_idx.shape[0], self.hidden,
+                              dtype=self.dtype, device="cuda")
+        return self._b.combine_recv(out, topk_idx,
+                                    topk_weights.float().contiguous())
 
     def close(self):
         """Release the native buffer (symmetric heap + IPC handles)

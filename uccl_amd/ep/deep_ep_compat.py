@@ -12,10 +12,14 @@ existing DeepEP callers can switch imports).
 
 Differences from NVIDIA DeepEP (documented, not hidden):
   - events are torch.cuda.Event objects recorded on the current stream
-    (ops are stream-ordered; there is no proxy hook to defer, so `hook`
-    is a no-op callable)
-  - fp8 dispatch returns (payload, scales) like DeepEP's
-    use_fp8=True path
+  - with return_recv_hook=True the RECV phase (count/flag waits) is NOT
+    launched until hook() runs, matching DeepEP's comm-compute overlap
+    contract; outputs are pre-allocated tensors whose contents are valid
+    after hook() + stream order
+  - fp8 dispatch returns (payload, scales) like DeepEP's use_fp8=True
+  - cached_handle=<handle from a prior dispatch> replays the cached
+    token compaction plan (skips the plan/prefix kernels) when the
+    routing is unchanged
 """
 
 from __future__ import annotations
@@ -39,9 +43,15 @@ class Config:
 
 
 class _Handle:
-    def __init__(self, topk_idx, num_tokens):
+    """Dispatch handle: carries the routing (topk_idx) plus the plan
+    generation so a later dispatch with the same handle can replay the
+    cached compaction lists (DeepEP cached-handle mode,
+    reference ep/src/intranode.cu:150 cached_notify_dispatch)."""
+
+    def __init__(self, topk_idx, num_tokens, plan_gen):
         self.topk_idx = topk_idx
         self.num_tokens = num_tokens
+        self.plan_gen = plan_gen
 
 
 def _event():
@@ -52,10 +62,6 @@ def _event():
     return ev
 
 
-def _noop_hook():
-    return None
-
-
 class Buffer:
     def __init__(self, group=None, num_nvl_bytes: int = 0,
                  num_rdma_bytes: int = 0, low_latency_mode: bool = True,
@@ -64,6 +70,7 @@ class Buffer:
         self._group = group
         self._native: Optional[_NativeBuffer] = None
         self._cfg = None
+        self._plan_gen = 0  # bumps whenever the native plan scratch changes
 
     # -- DeepEP static helpers ----------------------------------------------
     @staticmethod
@@ -94,28 +101,73 @@ class Buffer:
                              num_max_dispatch_tokens_per_rank: int,
                              num_experts: int, use_fp8: bool = False,
                              async_finish: bool = False,
-                             return_recv_hook: bool = False):
+                             return_recv_hook: bool = False,
+                             cached_handle: Optional[_Handle] = None,
+                             num_worst_tokens: int = 0):
+        if num_worst_tokens:
+            # DeepEP capacity hint: validate instead of silently
+            # overflowing the slot arrays
+            assert num_worst_tokens <= \
+                num_max_dispatch_tokens_per_rank * max(self.group_size, 1), \
+                "num_worst_tokens exceeds buffer capacity"
         nb = self._ensure(x.shape[1], num_max_dispatch_tokens_per_rank,
                           num_experts, topk_idx.shape[1], x.dtype, use_fp8)
-        out = nb.dispatch(x, topk_idx)
-        handle = _Handle(topk_idx, x.shape[0])
+        # cached-handle replay (DeepEP cached dispatch,
+        # reference ep/src/intranode.cu:150): same routing as the
+        # handle's dispatch -> skip the plan/prefix kernels
+        reuse = (cached_handle is not None and
+                 cached_handle.plan_gen == self._plan_gen and
+                 cached_handle.num_tokens == x.shape[0])
+        nb.dispatch_send(x, topk_idx, reuse_plan=reuse)
+        if not reuse:
+            self._plan_gen += 1
+        handle = _Handle(topk_idx, x.shape[0], self._plan_gen)
+
+        counts = torch.empty(nb.local_experts, nb.world, dtype=torch.int32,
+                             device=x.device)
         if use_fp8:
-            recv_x, counts, scales = out
-            packed = (recv_x, scales)
+            packed = (nb.recv_x_view(), nb.recv_scale_view())
         else:
-            recv_x, counts = out
-            packed = recv_x
-        # DeepEP's packed_recv_count is per-local-expert
-        recv_count = counts.sum(dim=1)
-        return packed, recv_count, handle, _event(), _noop_hook
+            packed = nb.recv_x_view()
+        recv_count = torch.empty(nb.local_experts, dtype=torch.int32,
+                                 device=x.device)
+
+        def finish():
+            nb.dispatch_recv(counts)
+            # DeepEP's packed_recv_count is per-local-expert
+            torch.sum(counts, dim=1, out=recv_count)
+
+        if return_recv_hook:
+            # REAL hook: the RECV phase is NOT yet launched. hook()
+            # enqueues the count wait on the caller's current stream, so
+            # compute issued before hook() overlaps the xGMI copies
+            # (DeepEP return_recv_hook contract, internode_ll.cu:62
+            # SEND|RECV phase split).
+            def hook():
+                finish()
+                return None
+
+            return packed, recv_count, handle, _event(), hook
+        finish()
+        return packed, recv_count, handle, _event(), None
 
     def low_latency_combine(self, x: torch.Tensor, topk_idx: torch.Tensor,
                             topk_weights: torch.Tensor, handle: _Handle,
                             async_finish: bool = False,
                             return_recv_hook: bool = False):
         assert self._native is not None, "combine before dispatch"
-        combined = self._native.combine(x, topk_idx, topk_weights)
-        return combined, _event(), _noop_hook
+        nb = self._native
+        combined = torch.empty(handle.num_tokens, nb.hidden,
+                               dtype=nb.dtype, device=x.device)
+        nb.combine_send(x)
+        if return_recv_hook:
+            def hook():
+                nb.combine_recv(topk_idx, topk_weights, out=combined)
+                return None
+
+            return combined, _event(), hook
+        nb.combine_recv(topk_idx, topk_weights, out=combined)
+        return combined, _event(), None
 
     # -- maintenance / config surface ---------------------------------------
     def clean_low_latency_buffer(self, num_max_dispatch_tokens_per_rank=None,
