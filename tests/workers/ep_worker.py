@@ -252,6 +252,24 @@ def main():
         buf2.close()
         print(f"[rank {rank}] 256-expert top-8 OK", flush=True)
 
+    # ---- proxy sync commands (ATOMIC / BARRIER / QUIET) ---------------------
+    if os.environ.get("UCCL_EP_FORCE_PROXY", "0") == "1" and world > 1:
+        b = buf._b
+        b.quiet()
+        b.barrier()
+        # each rank adds (rank+1) into every peer's atomic scratch word;
+        # per-flow ordering + the barrier round make all adds visible
+        # before the barrier completes
+        for dst in range(world):
+            if dst != rank:
+                b.atomic_add(dst, rank + 1)
+        b.barrier()
+        torch.cuda.synchronize()
+        got = b.read_sync_word(2)
+        want = sum(r + 1 for r in range(world) if r != rank)
+        assert got == want, (got, want)
+        print(f"[rank {rank}] proxy sync cmds OK", flush=True)
+
     if os.environ.get("UCCL_EP_FORCE_PROXY", "0") != "1":
         fp8_check()
     print(f"[rank {rank}] EP ALL OK", flush=True)

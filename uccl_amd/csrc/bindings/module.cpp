@@ -459,6 +459,23 @@ PYBIND11_MODULE(_C, m) {
                             current_stream(b.device()));
              return out;
            })
+      // proxy sync commands (reference proxy ATOMIC/BARRIER/QUIET parity)
+      .def("barrier",
+           [](uccl::ep::EpBuffer& b) {
+             b.barrier(current_stream(b.device()));
+           })
+      .def("quiet",
+           [](uccl::ep::EpBuffer& b) {
+             b.quiet(current_stream(b.device()));
+           })
+      .def("atomic_add",
+           [](uccl::ep::EpBuffer& b, int dst, uint64_t value) {
+             b.atomic_add(dst, value, current_stream(b.device()));
+           })
+      .def("read_sync_word",
+           [](uccl::ep::EpBuffer& b, int idx) {
+             return b.read_sync_word(idx);
+           })
       .def("combine",
            [](uccl::ep::EpBuffer& b, at::Tensor expert_out,
               at::Tensor topk_idx, at::Tensor topk_w) {

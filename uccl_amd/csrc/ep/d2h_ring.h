@@ -18,6 +18,11 @@ namespace ep {
 enum class CmdOp : uint32_t {
   kNone = 0,
   kDispatchWrite = 1,  // a = global expert, b = egress row offset, c = count
+  // parity with the reference proxy's command set
+  // (ep/src/proxy.cpp:828 post_gpu_command op switch):
+  kAtomicAdd = 2,  // a = dst rank, b = heap byte offset, c = add value
+  kBarrier = 3,    // a = seq: CPU-side barrier across proxy peers
+  kQuiet = 4,      // a = seq: all prior ring cmds fully shipped
 };
 
 struct TransferCmd {

@@ -52,6 +52,10 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   off = align256(off + sizeof(uint64_t) * v_.local_experts * world);
   v_.off_comb_flag = off;
   off = align256(off + sizeof(uint64_t) * world);
+  v_.off_sync = off;
+  off = align256(off + sizeof(uint64_t) * 4);
+  v_.off_consumed = off;
+  off = align256(off + sizeof(uint64_t) * world);
   v_.off_disp_meta = off;
   off = align256(off + sizeof(uint32_t) * v_.local_experts * world *
                            static_cast<size_t>(max_tokens));
@@ -206,6 +210,31 @@ void EpBuffer::dispatch(void const* x, int64_t const* topk_idx,
                         hipStream_t stream) {
   dispatch_send(x, topk_idx, num_tokens, /*reuse_plan=*/false, stream);
   dispatch_recv(out_counts, stream);
+}
+
+void EpBuffer::barrier(hipStream_t stream) {
+  UCCL_CHECK(v_.ring) << "barrier requires the proxy path";
+  ++sync_seq_;
+  launch_ep_barrier(v_, sync_seq_, stream);
+}
+
+void EpBuffer::quiet(hipStream_t stream) {
+  UCCL_CHECK(v_.ring) << "quiet requires the proxy path";
+  ++sync_seq_;
+  launch_ep_quiet(v_, sync_seq_, stream);
+}
+
+void EpBuffer::atomic_add(int dst, uint64_t value, hipStream_t stream) {
+  UCCL_CHECK(v_.ring) << "atomic_add requires the proxy path";
+  launch_ep_atomic_add(v_, dst, v_.off_sync + 2 * sizeof(uint64_t), value,
+                       stream);
+}
+
+uint64_t EpBuffer::read_sync_word(int idx) {
+  uint64_t v = 0;
+  UCCL_CHECK_HIP(hipMemcpy(&v, sync_ptr(heap_, v_, idx), sizeof(v),
+                           hipMemcpyDeviceToHost));
+  return v;
 }
 
 void EpBuffer::combine_send(void const* expert_out, hipStream_t stream) {

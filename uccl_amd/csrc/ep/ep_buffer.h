@@ -56,6 +56,15 @@ class EpBuffer {
   // topk shape; the heap's plan scratch is untouched since)
   bool plan_cached() const { return last_num_tokens_ >= 0; }
 
+  // Proxy-path synchronization commands (parity: the reference proxy's
+  // ATOMIC/BARRIER/QUIET cmd types, ep/src/proxy.cpp:1629-1718 + D2H
+  // cmd switch). All are GPU-initiated via the D2H ring and complete by
+  // a device-visible flag the paired wait kernel spins on.
+  void barrier(hipStream_t stream);       // across all proxy peers
+  void quiet(hipStream_t stream);         // prior ring cmds fully shipped
+  void atomic_add(int dst, uint64_t value, hipStream_t stream);
+  uint64_t read_sync_word(int idx);       // test/diagnostic accessor
+
   const EpView& view() const { return v_; }
   void* recv_x_ptr() const {
     return static_cast<char*>(heap_) + v_.off_disp_x;
@@ -79,6 +88,7 @@ class EpBuffer {
   std::array<bool, kMaxRanks> ipc_opened_{};
   bool connected_ = false;
   int last_num_tokens_ = -1;
+  uint64_t sync_seq_ = 0;
 
   // proxy path (internode / forced): D2H command ring + CPU proxy over
   // the reliable transport

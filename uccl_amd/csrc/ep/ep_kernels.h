@@ -23,6 +23,11 @@ void launch_ep_combine_finish(const EpView& v, void* out,
                               int num_tokens, hipStream_t s);
 void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
                             hipStream_t s);
+// proxy sync commands (D2H ring push + device flag wait)
+void launch_ep_barrier(const EpView& v, uint64_t seq, hipStream_t s);
+void launch_ep_quiet(const EpView& v, uint64_t seq, hipStream_t s);
+void launch_ep_atomic_add(const EpView& v, int dst, uint64_t off,
+                          uint64_t value, hipStream_t s);
 
 }  // namespace ep
 }  // namespace uccl

@@ -117,6 +117,36 @@ __host__ __device__ inline int fanout_for(int pairs) {
   return f < 1 ? 1 : (f > 256 ? 256 : f);
 }
 
+// Back-to-back dispatch safety: a new dispatch generation must not
+// overwrite peers' count/meta/slot words while they are still reading
+// the previous one (the combine round-trip provides this implicitly in
+// the dispatch;combine;dispatch pattern, but DeepEP's phase-split and
+// cached modes allow consecutive dispatches). dispatch_recv SIGNALS
+// "generation consumed" into every peer; the next dispatch_send GATES
+// on all peers having consumed the prior generation.
+__global__ void k_ep_consume_signal(EpView v, uint64_t seq) {
+  if (threadIdx.x < static_cast<unsigned>(v.world) &&
+      !((v.proxy_mask >> threadIdx.x) & 1u))
+    st_release_sys(consumed_ptr(v.peers[threadIdx.x], v, v.rank), seq);
+}
+
+__global__ void k_ep_consume_gate(EpView v, uint64_t prev_seq) {
+  if (threadIdx.x < static_cast<unsigned>(v.world) &&
+      !((v.proxy_mask >> threadIdx.x) & 1u)) {
+    uint64_t const* p = consumed_ptr(v.peers[v.rank], v, threadIdx.x);
+    for (uint64_t it = 0;; ++it) {
+      if (ld_acquire_sys(p) >= prev_seq) return;
+      if (it > (1ull << 28)) {
+        printf("uccl_ep: consume gate TIMEOUT rank=%d peer=%d prev=%llu\n",
+               v.rank, static_cast<int>(threadIdx.x),
+               (unsigned long long)prev_seq);
+        __builtin_trap();
+      }
+      backoff();
+    }
+  }
+}
+
 __global__ void k_ep_dispatch_plan(EpView v,
                                    int64_t const* __restrict__ topk_idx,
                                    int num_tokens) {
@@ -398,12 +428,41 @@ __global__ void k_ep_combine_reduce(EpView v, void* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
+// proxy sync commands (ATOMIC/BARRIER/QUIET): one-thread push into the
+// D2H ring + a paired single-block wait on the device-visible flag the
+// proxy writes back (same trap-not-wedge discipline as the other waits).
+// ---------------------------------------------------------------------------
+
+__global__ void k_ep_sync_push(EpView v, uint32_t op, uint64_t a,
+                               uint64_t b, uint64_t c) {
+  if (threadIdx.x == 0 && blockIdx.x == 0)
+    ring_push(v.ring, TransferCmd{op, static_cast<uint32_t>(v.seq), a, b, c});
+}
+
+__global__ void k_ep_sync_wait(EpView v, int idx, uint64_t seq) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    uint64_t const* p = sync_ptr(v.peers[v.rank], v, idx);
+    for (uint64_t it = 0;; ++it) {
+      if (ld_acquire_sys(p) >= seq) return;
+      if (it > (1ull << 28)) {
+        printf("uccl_ep: sync wait TIMEOUT rank=%d idx=%d seq=%llu\n",
+               v.rank, idx, (unsigned long long)seq);
+        __builtin_trap();
+      }
+      backoff();
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
 
 void launch_ep_dispatch_send(const EpView& v, void const* x,
                              int64_t const* topk_idx, int num_tokens,
                              bool reuse_plan, hipStream_t s) {
+  if (v.seq > 1 && v.world > 1)
+    k_ep_consume_gate<<<1, 64, 0, s>>>(v, v.seq - 1);
   if (!reuse_plan) {
     size_t const smem = 257 * sizeof(uint32_t);
     k_ep_dispatch_plan<<<v.num_experts, 256, smem, s>>>(v, topk_idx,
@@ -420,6 +479,7 @@ void launch_ep_dispatch_recv(const EpView& v, int* out_counts,
   int const n = v.local_experts * v.world;
   int const grid = std::min(64, (n + 255) / 256 + 1);
   k_ep_dispatch_wait<<<grid, 256, 0, s>>>(v, out_counts);
+  if (v.world > 1) k_ep_consume_signal<<<1, 64, 0, s>>>(v, v.seq);
 }
 
 void launch_ep_dispatch(const EpView& v, void const* x,
@@ -427,6 +487,25 @@ void launch_ep_dispatch(const EpView& v, void const* x,
                         int* out_counts, hipStream_t s) {
   launch_ep_dispatch_send(v, x, topk_idx, num_tokens, false, s);
   launch_ep_dispatch_recv(v, out_counts, s);
+}
+
+void launch_ep_barrier(const EpView& v, uint64_t seq, hipStream_t s) {
+  k_ep_sync_push<<<1, 64, 0, s>>>(
+      v, static_cast<uint32_t>(CmdOp::kBarrier), seq, 0, 0);
+  k_ep_sync_wait<<<1, 64, 0, s>>>(v, 0, seq);
+}
+
+void launch_ep_quiet(const EpView& v, uint64_t seq, hipStream_t s) {
+  k_ep_sync_push<<<1, 64, 0, s>>>(
+      v, static_cast<uint32_t>(CmdOp::kQuiet), seq, 0, 0);
+  k_ep_sync_wait<<<1, 64, 0, s>>>(v, 1, seq);
+}
+
+void launch_ep_atomic_add(const EpView& v, int dst, uint64_t off,
+                          uint64_t value, hipStream_t s) {
+  k_ep_sync_push<<<1, 64, 0, s>>>(
+      v, static_cast<uint32_t>(CmdOp::kAtomicAdd),
+      static_cast<uint64_t>(dst), off, value);
 }
 
 void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,

@@ -46,6 +46,11 @@ struct EpView {
   size_t off_disp_meta;   // u32 [local_experts][world*max_tokens]
                           //   meta = src_token_idx | (k << 24)
   size_t off_comb_flag;   // u64 [world]                   (seq)
+  size_t off_sync;        // u64 [4]: [0] barrier seq, [1] quiet seq,
+                          //          [2] atomic scratch, [3] reserved
+  size_t off_consumed;    // u64 [world]: consumed[r] on MY heap = the
+                          //   dispatch generation rank r has finished
+                          //   reading from me (next-dispatch gate)
   size_t off_comb_x;      // [max_tokens][topk][hidden] elems
   size_t off_plan;        // u32 [num_experts][2 + max_tokens]  (private
                           //   per-rank scratch: count, egress prefix, list)
@@ -130,6 +135,19 @@ __host__ __device__ inline uint64_t* comb_flag_ptr(void* base,
   return reinterpret_cast<uint64_t*>(static_cast<char*>(base) +
                                      v.off_comb_flag) +
          src;
+}
+
+__host__ __device__ inline uint64_t* sync_ptr(void* base, const EpView& v,
+                                              int idx) {
+  return reinterpret_cast<uint64_t*>(static_cast<char*>(base) + v.off_sync) +
+         idx;
+}
+
+__host__ __device__ inline uint64_t* consumed_ptr(void* base,
+                                                  const EpView& v, int r) {
+  return reinterpret_cast<uint64_t*>(static_cast<char*>(base) +
+                                     v.off_consumed) +
+         r;
 }
 
 __host__ __device__ inline char* comb_x_ptr(void* base, const EpView& v,

@@ -1,8 +1,11 @@
 #include "ep_proxy.h"
 
+#include <pthread.h>
+#include <sched.h>
 #include <unistd.h>
 
 #include <cstring>
+#include <thread>
 
 #include "../core/env.h"
 #include "../core/log.h"
@@ -11,8 +14,39 @@ namespace uccl {
 namespace ep {
 
 namespace {
-constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3;
+constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3, kBar = 4, kAtom = 5;
 constexpr size_t kStageBytes = 16ull << 20;  // pinned staging chunk
+
+// CPU pinning for proxy threads (parity: the reference pins proxies per
+// NUMA node / NIC, ep/src/proxy.cpp:168-214). Without libnuma in this
+// image, pin to explicit cores: UCCL_EP_PROXY_CORES="c0,c1,.." or
+// default to the HIGHEST cores (kernels/apps start filling from 0).
+void pin_proxy_thread(int slot) {
+  std::string const spec = env_str("UCCL_EP_PROXY_CORES", "");
+  int core = -1;
+  if (!spec.empty()) {
+    int idx = 0;
+    size_t pos = 0;
+    while (pos < spec.size()) {
+      size_t nxt = spec.find(',', pos);
+      if (nxt == std::string::npos) nxt = spec.size();
+      if (idx == slot) {
+        core = atoi(spec.substr(pos, nxt - pos).c_str());
+        break;
+      }
+      pos = nxt + 1;
+      ++idx;
+    }
+  } else {
+    int const hw = static_cast<int>(std::thread::hardware_concurrency());
+    if (hw >= 16) core = hw - 1 - (slot % (hw / 4));
+  }
+  if (core < 0) return;
+  cpu_set_t set;
+  CPU_ZERO(&set);
+  CPU_SET(core, &set);
+  (void)pthread_setaffinity_np(pthread_self(), sizeof(set), &set);
+}
 }  // namespace
 
 int EpProxy::flow_peer(uint64_t flow) const {
@@ -25,10 +59,12 @@ EpProxy::Lane::Lane(int device, size_t bytes) {
   UCCL_CHECK_HIP(hipSetDevice(device));
   UCCL_CHECK_HIP(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
   UCCL_CHECK_HIP(hipHostMalloc(&buf, bytes));
+  UCCL_CHECK_HIP(hipHostMalloc(&buf2, bytes));
 }
 
 EpProxy::Lane::~Lane() {
   if (buf) (void)hipHostFree(buf);
+  if (buf2) (void)hipHostFree(buf2);
   if (stream) (void)hipStreamDestroy(stream);
 }
 
@@ -98,13 +134,32 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
   size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
   size_t const total = h.count * row_bytes;
   tp_->send_msg(flow, &h, sizeof(h));
-  for (size_t off = 0; off < total; off += stage_bytes_) {
-    size_t const n = std::min(stage_bytes_, total - off);
-    UCCL_CHECK_HIP(hipMemcpyAsync(lane.buf,
-                                  static_cast<char const*>(dev_rows) + off,
-                                  n, hipMemcpyDeviceToHost, lane.stream));
+  // double-buffered pipeline: the D2H copy of chunk i+1 runs while the
+  // transport ships chunk i (the reference's batched-posting role,
+  // proxy.cpp:1203 post_gpu_commands_mixed)
+  void* bufs[2] = {lane.buf, lane.buf2};
+  size_t off = 0;
+  int cur = 0;
+  if (total) {
+    size_t const n0 = std::min(stage_bytes_, total);
+    UCCL_CHECK_HIP(hipMemcpyAsync(bufs[cur], dev_rows, n0,
+                                  hipMemcpyDeviceToHost, lane.stream));
     UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
-    tp_->send_msg(flow, lane.buf, n);
+  }
+  while (off < total) {
+    size_t const n = std::min(stage_bytes_, total - off);
+    size_t const next_off = off + n;
+    if (next_off < total) {
+      size_t const n1 = std::min(stage_bytes_, total - next_off);
+      UCCL_CHECK_HIP(hipMemcpyAsync(
+          bufs[cur ^ 1], static_cast<char const*>(dev_rows) + next_off, n1,
+          hipMemcpyDeviceToHost, lane.stream));
+    }
+    tp_->send_msg(flow, bufs[cur], n);
+    if (next_off < total)
+      UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+    off = next_off;
+    cur ^= 1;
   }
   std::vector<uint32_t> metas;
   if (host_metas) {
@@ -119,8 +174,37 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
   if (h.count) tp_->send_msg(flow, metas.data(), h.count * sizeof(uint32_t));
 }
 
+int EpProxy::num_proxy_peers() const {
+  int n = 0;
+  for (int r = 0; r < v_.world; ++r)
+    if ((proxy_mask_ >> r) & 1u) ++n;
+  return n;
+}
+
+void EpProxy::write_sync_flag(Lane& lane, int idx, uint64_t seq) {
+  UCCL_CHECK_HIP(hipMemcpyAsync(sync_ptr(heap_, v_, idx), &seq, sizeof(seq),
+                                hipMemcpyHostToDevice, lane.stream));
+  UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+}
+
+// one arrival (self or a peer's kBar message) for barrier `seq`; when
+// self + every proxy peer have arrived, publish the device flag
+void EpProxy::handle_barrier_arrival(Lane& lane, uint64_t seq) {
+  bool complete = false;
+  {
+    std::lock_guard<std::mutex> g(bar_mu_);
+    int const need = 1 + num_proxy_peers();
+    if (++bar_seen_[seq] >= need) {
+      bar_seen_.erase(seq);
+      complete = true;
+    }
+  }
+  if (complete) write_sync_flag(lane, 0, seq);
+}
+
 void EpProxy::ring_loop() {
   (void)hipSetDevice(device_);
+  pin_proxy_thread(0);
   Lane lane(device_, stage_bytes_);
   uint64_t head = 0;
   while (!stop_) {
@@ -138,14 +222,51 @@ void EpProxy::ring_loop() {
       ++head;
       __atomic_store_n(const_cast<uint64_t*>(&ring_->head), head,
                        __ATOMIC_RELEASE);
-      if (c.op != static_cast<uint32_t>(CmdOp::kDispatchWrite)) continue;
-      int const e = static_cast<int>(c.a);
-      int const dst = e / v_.local_experts;
-      WireHdr h{kDisp, c.seq32, static_cast<uint32_t>(e % v_.local_experts),
-                static_cast<uint32_t>(v_.rank), c.c};
       try {
-        ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
-                  egress_meta(heap_, v_, c.b), nullptr);
+        switch (static_cast<CmdOp>(c.op)) {
+          case CmdOp::kDispatchWrite: {
+            int const e = static_cast<int>(c.a);
+            int const dst = e / v_.local_experts;
+            WireHdr h{kDisp, c.seq32,
+                      static_cast<uint32_t>(e % v_.local_experts),
+                      static_cast<uint32_t>(v_.rank), c.c, 0};
+            ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
+                      egress_meta(heap_, v_, c.b), nullptr);
+            break;
+          }
+          case CmdOp::kAtomicAdd: {
+            // a = dst rank, b = heap byte offset, c = value
+            int const dst = static_cast<int>(c.a);
+            WireHdr h{kAtom, c.seq32, 0, static_cast<uint32_t>(v_.rank),
+                      c.c, c.b};
+            std::lock_guard<std::mutex> guard(*flow_mu_[dst]);
+            tp_->send_msg(flows_[dst], &h, sizeof(h));
+            break;
+          }
+          case CmdOp::kBarrier: {
+            // announce to every proxy peer, count self
+            uint64_t const seq = c.a;
+            WireHdr h{kBar, static_cast<uint32_t>(seq), 0,
+                      static_cast<uint32_t>(v_.rank), seq, 0};
+            for (int r = 0; r < v_.world; ++r) {
+              if (!((proxy_mask_ >> r) & 1u)) continue;
+              std::lock_guard<std::mutex> guard(*flow_mu_[r]);
+              tp_->send_msg(flows_[r], &h, sizeof(h));
+            }
+            handle_barrier_arrival(lane, seq);
+            break;
+          }
+          case CmdOp::kQuiet: {
+            // ring cmds execute in FIFO order and ship_rows is
+            // synchronous, so reaching this cmd means every prior
+            // transfer has been fully handed to the reliable transport
+            // (the reference's quiet_cq role, proxy.cpp:1341)
+            write_sync_flag(lane, 1, c.a);
+            break;
+          }
+          default:
+            break;
+        }
       } catch (std::exception const&) {
         return;  // transport closed
       }
@@ -155,6 +276,7 @@ void EpProxy::ring_loop() {
 
 void EpProxy::comb_tx_loop() {
   (void)hipSetDevice(device_);
+  pin_proxy_thread(1);
   Lane lane(device_, stage_bytes_);
   while (true) {
     CombTask task;
@@ -220,6 +342,7 @@ void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
 
 void EpProxy::rx_loop(int peer) {
   (void)hipSetDevice(device_);
+  pin_proxy_thread(2 + peer);
   Lane lane(device_, stage_bytes_);
   uint64_t const flow = flows_[peer];
   size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
@@ -228,6 +351,25 @@ void EpProxy::rx_loop(int peer) {
     while (!stop_) {
       WireHdr h{};
       tp_->recv_msg(flow, &h, sizeof(h));
+      if (h.kind == kBar) {
+        handle_barrier_arrival(lane, h.count);
+        continue;
+      }
+      if (h.kind == kAtom) {
+        // serialize RMW on the heap word (rx threads are the only
+        // writers of these offsets; the mutex covers multi-peer adds)
+        std::lock_guard<std::mutex> g(atomic_mu_);
+        uint64_t cur = 0;
+        char* addr = static_cast<char*>(heap_) + h.aux;
+        UCCL_CHECK_HIP(hipMemcpyAsync(&cur, addr, sizeof(cur),
+                                      hipMemcpyDeviceToHost, lane.stream));
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+        cur += h.count;
+        UCCL_CHECK_HIP(hipMemcpyAsync(addr, &cur, sizeof(cur),
+                                      hipMemcpyHostToDevice, lane.stream));
+        UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+        continue;
+      }
       if (h.kind == kCombDone) {
         // all combine payloads from `peer` for this seq have been
         // scattered (lane.stream is in-order); publish the flag

@@ -15,6 +15,7 @@
 #include <atomic>
 #include <condition_variable>
 #include <deque>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <thread>
@@ -49,11 +50,13 @@ class EpProxy {
 
  private:
   struct WireHdr {
-    uint32_t kind;   // 1=dispatch, 2=combine, 3=combine-done
+    uint32_t kind;   // 1=dispatch, 2=combine, 3=combine-done,
+                     // 4=barrier, 5=atomic-add
     uint32_t seq32;
     uint32_t le;
     uint32_t src;
-    uint64_t count;
+    uint64_t count;  // barrier: unused; atomic: the add value
+    uint64_t aux;    // atomic: target heap byte offset
   };
 
   // Per-thread pinned staging + non-blocking stream: the ring thread,
@@ -61,6 +64,8 @@ class EpProxy {
   // and must never share host staging (corruption observed at world=4).
   struct Lane {
     void* buf = nullptr;
+    void* buf2 = nullptr;  // double buffer: D2H of chunk i+1 overlaps the
+                           // transport send of chunk i (batched posting)
     hipStream_t stream = nullptr;
     Lane(int device, size_t bytes);
     ~Lane();
@@ -69,6 +74,9 @@ class EpProxy {
   void ring_loop();
   void rx_loop(int peer);
   void comb_tx_loop();
+  void handle_barrier_arrival(Lane& lane, uint64_t seq);
+  void write_sync_flag(Lane& lane, int idx, uint64_t seq);
+  int num_proxy_peers() const;
   int flow_peer(uint64_t flow) const;
   void ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                  void const* dev_rows, uint32_t const* dev_metas_or_null,
@@ -96,6 +104,12 @@ class EpProxy {
                         // has completed (the D2H precedes it in-stream)
   };
   std::deque<CombTask> comb_q_;
+  // barrier bookkeeping: arrivals per seq (self + every proxy peer)
+  std::mutex bar_mu_;
+  std::map<uint64_t, int> bar_seen_;
+  std::map<uint64_t, Lane*> bar_lane_;
+  // serializes atomic-add read-modify-writes into the heap
+  std::mutex atomic_mu_;
   std::mutex mu_;
   std::condition_variable cv_;
   std::atomic<bool> stop_{false};
