@@ -32,7 +32,9 @@ SOURCES = [
     CSRC / "ep" / "ep_buffer.cpp",
     CSRC / "ep" / "ep_proxy.cpp",
     CSRC / "ukernel" / "ukernel.cpp",
+    CSRC / "ukernel" / "uk_device.hip",
     CSRC / "p2p" / "compress.cpp",
+    CSRC / "p2p" / "gpu_codec.hip",
     CSRC / "core" / "trace.cpp",
     CSRC / "bindings" / "module.cpp",
 ]

@@ -105,9 +105,10 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
     UCCL_CHECK_HIP(hipHostGetDevicePointer(&dev, ring_host_, 0));
     v_.ring = static_cast<D2HRing*>(dev);
   }
-  UCCL_CHECK_HIP(hipHostMalloc(
-      reinterpret_cast<void**>(&host_counts_),
-      sizeof(int) * v_.local_experts * world_));
+  for (int i = 0; i < kCountSlots; ++i)
+    UCCL_CHECK_HIP(hipHostMalloc(
+        reinterpret_cast<void**>(&host_counts_[i]),
+        sizeof(int) * v_.local_experts * world_));
   UCCL_LOG_INFO << "EpBuffer rank " << rank << "/" << world << " experts="
                 << num_experts << " hidden=" << hidden << " max_tokens="
                 << max_tokens << " heap=" << (v_.heap_bytes >> 20) << "MB";
@@ -119,7 +120,8 @@ EpBuffer::~EpBuffer() {
     if (ipc_opened_[r] && v_.peers[r]) (void)hipIpcCloseMemHandle(v_.peers[r]);
   if (heap_) (void)hipFree(heap_);
   if (ring_host_) (void)hipHostFree(ring_host_);
-  if (host_counts_) (void)hipHostFree(host_counts_);
+  for (int i = 0; i < kCountSlots; ++i)
+    if (host_counts_[i]) (void)hipHostFree(host_counts_[i]);
 }
 
 std::string EpBuffer::handle_bytes() const {
@@ -199,8 +201,10 @@ void EpBuffer::dispatch_send(void const* x, int64_t const* topk_idx,
 
 void EpBuffer::dispatch_recv(int* out_counts, hipStream_t stream) {
   launch_ep_dispatch_recv(v_, out_counts, stream);
-  // pinned host copy of counts (combine proxy shipping needs them)
-  UCCL_CHECK_HIP(hipMemcpyAsync(host_counts_, out_counts,
+  // pinned host copy of counts (combine proxy shipping needs them);
+  // rotate slots so an in-flight combine's pointer stays valid
+  UCCL_CHECK_HIP(hipMemcpyAsync(host_counts_[v_.seq % kCountSlots],
+                                out_counts,
                                 sizeof(int) * v_.local_experts * world_,
                                 hipMemcpyDeviceToHost, stream));
 }
@@ -246,7 +250,8 @@ void EpBuffer::combine_send(void const* expert_out, hipStream_t stream) {
     hipEvent_t ev = nullptr;
     UCCL_CHECK_HIP(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
     UCCL_CHECK_HIP(hipEventRecord(ev, stream));
-    proxy_->enqueue_combine(expert_out, v_.seq, ev, host_counts_);
+    proxy_->enqueue_combine(expert_out, v_.seq, ev,
+                            host_counts_[v_.seq % kCountSlots]);
   }
 }
 

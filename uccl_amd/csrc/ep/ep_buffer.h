@@ -94,7 +94,11 @@ class EpBuffer {
   // the reliable transport
   std::unique_ptr<EpProxy> proxy_;
   D2HRing* ring_host_ = nullptr;
-  int* host_counts_ = nullptr;  // pinned copy of dispatch counts
+  // pinned copies of dispatch counts, rotated per generation so a
+  // combine in flight never reads a buffer the next dispatch_recv is
+  // overwriting
+  static constexpr int kCountSlots = 4;
+  int* host_counts_[kCountSlots] = {};
 };
 
 }  // namespace ep

@@ -147,6 +147,28 @@ __global__ void k_ep_consume_gate(EpView v, uint64_t prev_seq) {
   }
 }
 
+// proxy egress reuse gate: spin until the CPU proxy has shipped every
+// queued ring command (head catches tail) so the egress staging rows
+// can be overwritten by the next generation
+__global__ void k_ep_ring_wait_empty(EpView v) {
+  if (threadIdx.x != 0 || blockIdx.x != 0 || !v.ring) return;
+  for (uint64_t it = 0;; ++it) {
+    uint64_t const t = __hip_atomic_load(
+        const_cast<uint64_t*>(&v.ring->tail), __ATOMIC_RELAXED,
+        __HIP_MEMORY_SCOPE_SYSTEM);
+    uint64_t const h = __hip_atomic_load(
+        const_cast<uint64_t*>(&v.ring->head), __ATOMIC_ACQUIRE,
+        __HIP_MEMORY_SCOPE_SYSTEM);
+    if (h == t) return;
+    if (it > (1ull << 28)) {
+      printf("uccl_ep: ring drain TIMEOUT rank=%d head=%llu tail=%llu\n",
+             v.rank, (unsigned long long)h, (unsigned long long)t);
+      __builtin_trap();
+    }
+    backoff();
+  }
+}
+
 __global__ void k_ep_dispatch_plan(EpView v,
                                    int64_t const* __restrict__ topk_idx,
                                    int num_tokens) {
@@ -463,6 +485,7 @@ void launch_ep_dispatch_send(const EpView& v, void const* x,
                              bool reuse_plan, hipStream_t s) {
   if (v.seq > 1 && v.world > 1)
     k_ep_consume_gate<<<1, 64, 0, s>>>(v, v.seq - 1);
+  if (v.ring && v.proxy_mask) k_ep_ring_wait_empty<<<1, 64, 0, s>>>(v);
   if (!reuse_plan) {
     size_t const smem = 257 * sizeof(uint32_t);
     k_ep_dispatch_plan<<<v.num_experts, 256, smem, s>>>(v, topk_idx,

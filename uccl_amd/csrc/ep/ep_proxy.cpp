@@ -219,9 +219,6 @@ void EpProxy::ring_loop() {
       TransferCmd c;
       memcpy(&c, const_cast<TransferCmd*>(&ring_->cmds[head % kRingSlots]),
              sizeof(c));
-      ++head;
-      __atomic_store_n(const_cast<uint64_t*>(&ring_->head), head,
-                       __ATOMIC_RELEASE);
       try {
         switch (static_cast<CmdOp>(c.op)) {
           case CmdOp::kDispatchWrite: {
@@ -267,9 +264,19 @@ void EpProxy::ring_loop() {
           default:
             break;
         }
-      } catch (std::exception const&) {
-        return;  // transport closed
+      } catch (std::exception const& e) {
+        if (!stop_)
+          UCCL_LOG_ERROR << "ep proxy ring_loop died: " << e.what();
+        return;
       }
+      // publish head ONLY after the command is fully shipped: the GPU
+      // gates the next dispatch's egress reuse on head==tail
+      // (k_ep_ring_wait_empty), so early publication would let the copy
+      // kernel clobber rows mid-ship (observed as corrupt meta ->
+      // out-of-bounds combine writes)
+      ++head;
+      __atomic_store_n(const_cast<uint64_t*>(&ring_->head), head,
+                       __ATOMIC_RELEASE);
     }
   }
 }
@@ -315,7 +322,9 @@ void EpProxy::comb_tx_loop() {
           tp_->send_msg(flows_[src], &done, sizeof(done));
         }
       }
-    } catch (std::exception const&) {
+    } catch (std::exception const& e) {
+      if (!stop_)
+        UCCL_LOG_ERROR << "ep proxy comb_tx_loop died: " << e.what();
       return;
     }
     (void)hipEventDestroy(task.ready);
@@ -426,8 +435,10 @@ void EpProxy::rx_loop(int peer) {
         }
       }
     }
-  } catch (std::exception const&) {
-    // transport closed
+  } catch (std::exception const& e) {
+    if (!stop_)
+      UCCL_LOG_ERROR << "ep proxy rx_loop(" << peer
+                     << ") died: " << e.what();
   }
 }
 
