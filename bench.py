@@ -117,9 +117,16 @@ def main():
         try:
             schk = comm.symmetric_tensor([1 << 20], torch.bfloat16)
             sym_ok = check_ones(schk)
-        except Exception:
+        except Exception as e:
             sym_ok = False  # engine refused the symmetric path
+            print(f"[bench rank {rank}] symmetric gate RAISED: {e}",
+                  file=sys.stderr, flush=True)
         if not sym_ok:
+            # per-rank diagnostic so a multi-GPU bring-up shows WHICH rank
+            # failed the zero-copy gate and the run is auditable
+            print(f"[bench rank {rank}] symmetric path gate failed; "
+                  "falling back to the staged engine", file=sys.stderr,
+                  flush=True)
             args.symmetric = False
             t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
 
