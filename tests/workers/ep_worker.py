@@ -137,6 +137,7 @@ def main():
         out2 = buf.combine(rx, topk2.cuda(), w2.cuda())
         torch.cuda.synchronize()
         assert out2.shape == (T, H)
+        print(f"[rank {rank}] repeat {it} OK", flush=True)
 
     # ---- phase-split (SEND|RECV) must equal the eager path ------------------
     x3, topk3, w3 = rank_inputs(rank, T, H, K, E, dtype, seed + 31)
@@ -179,6 +180,63 @@ def main():
         assert not torch.equal(snap_a, snap_b), \
             "cached replay did not refresh payloads"
     print(f"[rank {rank}] cached-plan replay OK", flush=True)
+
+    # ---- normal (rank-granular) mode vs torch reference ---------------------
+    if os.environ.get("UCCL_EP_FORCE_PROXY", "0") != "1":
+        xn, topkn, wn = rank_inputs(rank, T, H, K, E, dtype, seed + 201)
+        xng, topkng, wng = xn.cuda(), topkn.cuda(), wn.cuda()
+        rxv, counts_n, rtopkv, rwv = buf.nrm_dispatch(xng, topkng, wng)
+        torch.cuda.synchronize()
+        all_n = [rank_inputs(r, T, H, K, E, dtype, seed + 201)
+                 for r in range(world)]
+        lo, hi = rank * local_E, (rank + 1) * local_E
+        cn = counts_n.cpu()
+        for src in range(world):
+            sx, st, sw = all_n[src]
+            sel = [t for t in range(T)
+                   if ((st[t] // local_E) == rank).any()]
+            assert cn[src].item() == len(sel), (src, cn[src].item(),
+                                                len(sel))
+            n = len(sel)
+            assert torch.equal(rxv[src, :n].cpu().view(torch.int16),
+                               sx[sel].view(torch.int16)), src
+            assert torch.equal(rtopkv[src, :n].cpu(), st[sel]), src
+            assert torch.allclose(rwv[src, :n].cpu(), sw[sel]), src
+        print(f"[rank {rank}] normal dispatch OK", flush=True)
+
+        # receiver applies its local experts (f_e(x) = 2x + e) weighted
+        proc = torch.zeros(world, maxT, H, dtype=dtype, device="cuda")
+        for src in range(world):
+            n = cn[src].item()
+            if n == 0:
+                continue
+            xs = rxv[src, :n].float()
+            tks = rtopkv[src, :n]
+            ws = rwv[src, :n]
+            mine = ((tks >= lo) & (tks < hi)).float()
+            wsum = (ws * mine).sum(1, keepdim=True)
+            we = (ws * mine * tks.clamp(min=0).float()).sum(1, keepdim=True)
+            proc[src, :n] = (xs * 2 * wsum + we).to(dtype)
+        outn = buf.nrm_combine(proc, topkng)
+        torch.cuda.synchronize()
+        refn = torch.zeros(T, H, dtype=torch.float32)
+        for t in range(T):
+            # per contributing rank: sum_k(mine) w*(2x+e), cast once
+            per_rank = {}
+            for k in range(K):
+                e = int(topkn[t, k])
+                per_rank.setdefault(e // local_E, []).append(k)
+            for r, ks in per_rank.items():
+                acc = torch.zeros(H)
+                for k in ks:
+                    e = int(topkn[t, k])
+                    acc += wn[t, k] * (xn[t].float() * 2 + e)
+                refn[t] += acc.to(dtype).float()
+        diffn = (outn.cpu().float() - refn).abs().max().item()
+        scalen = refn.abs().max().item()
+        assert diffn <= 0.06 * max(scalen, 1.0), (diffn, scalen)
+        print(f"[rank {rank}] normal combine OK (maxdiff {diffn:.4f})",
+              flush=True)
 
     # ---- DeepEP compat surface with a REAL recv hook at top-8 ---------------
     # (transliterated from the reference's test_low_latency.py:418 check
@@ -226,6 +284,23 @@ def main():
         assert torch.equal(rc2.cpu(), recv_count.cpu())
         print(f"[rank {rank}] deep_ep compat (hook+cached, top-{K8}) OK",
               flush=True)
+
+        # normal-mode through the DeepEP signature: dispatch carries the
+        # topk rows; receiver applies weights; combine sums across ranks
+        npe = torch.zeros(E, dtype=torch.int32)
+        rxn, rtin, rtwn, per_e, hn, _ = cb.dispatch(
+            xcg, topk_idx=topkcg, topk_weights=wc.cuda(),
+            num_tokens_per_expert=npe)
+        torch.cuda.synchronize()
+        assert sum(per_e) == int((rtin >= 0).sum())
+        procc = (rxn.float() * rtwn.sum(1, keepdim=True)).to(xcg.dtype)
+        combn, _ = cb.combine(procc, hn)
+        torch.cuda.synchronize()
+        wantc = (wc.sum(1, keepdim=True).cuda() * xcg.float()).cpu()
+        dn = (combn.cpu().float() - wantc).abs().max().item()
+        sn = wantc.abs().max().item()
+        assert dn <= 0.06 * max(sn, 1.0), (dn, sn)
+        print(f"[rank {rank}] deep_ep compat normal mode OK", flush=True)
 
     # ---- many-expert shape (256 experts, top-8): plan/wait scaling ----------
     if os.environ.get("UCCL_TEST_LIGHT", "0") != "1":

@@ -330,15 +330,15 @@ PYBIND11_MODULE(_C, m) {
   py::class_<uccl::ep::EpBuffer>(m, "EpBuffer")
       .def(py::init([](int rank, int world, int device, int num_experts,
                        int topk, int hidden, int max_tokens, int elem_size,
-                       bool use_fp8) {
+                       bool use_fp8, bool with_normal) {
              return new uccl::ep::EpBuffer(rank, world, device, num_experts,
                                            topk, hidden, max_tokens,
-                                           elem_size, use_fp8);
+                                           elem_size, use_fp8, with_normal);
            }),
            py::arg("rank"), py::arg("world"), py::arg("device"),
            py::arg("num_experts"), py::arg("topk"), py::arg("hidden"),
            py::arg("max_tokens"), py::arg("elem_size"),
-           py::arg("use_fp8") = false)
+           py::arg("use_fp8") = false, py::arg("with_normal") = true)
       .def("handle_bytes",
            [](uccl::ep::EpBuffer& b) { return py::bytes(b.handle_bytes()); })
       .def("connect",
@@ -460,6 +460,94 @@ PYBIND11_MODULE(_C, m) {
                             topk_w.data_ptr<float>(),
                             current_stream(b.device()));
              return out;
+           })
+      // normal (rank-granular) mode — DeepEP HT dispatch/combine
+      .def("nrm_dispatch_send",
+           [](uccl::ep::EpBuffer& b, at::Tensor x, at::Tensor topk_idx,
+              py::object topk_w) {
+             auto const& v = b.view();
+             TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+                         x.dim() == 2 && x.size(1) == v.hidden &&
+                         x.element_size() == v.elem_size);
+             TORCH_CHECK(topk_idx.scalar_type() == at::kLong &&
+                         topk_idx.is_cuda() && topk_idx.is_contiguous() &&
+                         topk_idx.size(1) == v.topk);
+             float const* wp = nullptr;
+             at::Tensor w;
+             if (!topk_w.is_none()) {
+               w = topk_w.cast<at::Tensor>();
+               TORCH_CHECK(w.scalar_type() == at::kFloat &&
+                           w.is_contiguous() && w.is_cuda());
+               wp = w.data_ptr<float>();
+             }
+             b.nrm_dispatch_send(x.data_ptr(), topk_idx.data_ptr<int64_t>(),
+                                 wp, static_cast<int>(x.size(0)),
+                                 current_stream(b.device()));
+           },
+           py::arg("x"), py::arg("topk_idx"),
+           py::arg("topk_w") = py::none())
+      .def("nrm_dispatch_recv",
+           [](uccl::ep::EpBuffer& b, at::Tensor counts) {
+             auto const& v = b.view();
+             TORCH_CHECK(counts.is_cuda() && counts.is_contiguous() &&
+                         counts.scalar_type() == at::kInt &&
+                         counts.numel() == v.world);
+             b.nrm_dispatch_recv(counts.data_ptr<int>(),
+                                 current_stream(b.device()));
+             return counts;
+           })
+      .def("nrm_combine_send",
+           [](uccl::ep::EpBuffer& b, at::Tensor x) {
+             auto const& v = b.view();
+             TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+                         x.element_size() == v.elem_size);
+             b.nrm_combine_send(x.data_ptr(), current_stream(b.device()));
+           })
+      .def("nrm_combine_recv",
+           [](uccl::ep::EpBuffer& b, at::Tensor out, at::Tensor topk_idx) {
+             auto const& v = b.view();
+             TORCH_CHECK(out.is_cuda() && out.is_contiguous() &&
+                         out.dim() == 2 && out.size(1) == v.hidden);
+             TORCH_CHECK(topk_idx.scalar_type() == at::kLong &&
+                         topk_idx.is_cuda() && topk_idx.is_contiguous());
+             b.nrm_combine_recv(out.data_ptr(),
+                                topk_idx.data_ptr<int64_t>(),
+                                current_stream(b.device()));
+             return out;
+           })
+      .def("nrm_x_view",
+           [](uccl::ep::EpBuffer& b) {
+             auto const& v = b.view();
+             auto dt = v.elem_size == 2 ? at::kBFloat16 : at::kFloat;
+             return at::from_blob(
+                 b.nrm_x_base(),
+                 {v.world, v.max_tokens, v.hidden},
+                 at::TensorOptions().dtype(dt).device(
+                     at::Device(at::kCUDA, b.device())));
+           })
+      .def("nrm_meta_view",
+           [](uccl::ep::EpBuffer& b) {
+             auto const& v = b.view();
+             return at::from_blob(
+                 b.nrm_meta_base(), {v.world, v.max_tokens},
+                 at::TensorOptions().dtype(at::kInt).device(
+                     at::Device(at::kCUDA, b.device())));
+           })
+      .def("nrm_topk_view",
+           [](uccl::ep::EpBuffer& b) {
+             auto const& v = b.view();
+             return at::from_blob(
+                 b.nrm_topk_base(), {v.world, v.max_tokens, v.topk},
+                 at::TensorOptions().dtype(at::kLong).device(
+                     at::Device(at::kCUDA, b.device())));
+           })
+      .def("nrm_w_view",
+           [](uccl::ep::EpBuffer& b) {
+             auto const& v = b.view();
+             return at::from_blob(
+                 b.nrm_w_base(), {v.world, v.max_tokens, v.topk},
+                 at::TensorOptions().dtype(at::kFloat).device(
+                     at::Device(at::kCUDA, b.device())));
            })
       // proxy sync commands (reference proxy ATOMIC/BARRIER/QUIET parity)
       .def("barrier",

@@ -22,7 +22,7 @@ static size_t align256(size_t x) { return (x + 255) & ~size_t(255); }
 
 EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
                    int topk, int hidden, int max_tokens, int elem_size,
-                   bool use_fp8)
+                   bool use_fp8, bool with_normal)
     : rank_(rank), world_(world), device_(device) {
   if (use_fp8) {
     UCCL_CHECK(hidden % 128 == 0) << "fp8 dispatch needs hidden % 128 == 0";
@@ -85,11 +85,40 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   off = align256(off + ingress_rows * hidden * elem_size);
   v_.off_ingress_meta = off;
   off = align256(off + ingress_rows * sizeof(uint32_t));
+  // normal (rank-granular) mode regions — DeepEP HT dispatch/combine
+  v_.with_normal = with_normal ? 1 : 0;
+  if (with_normal) {
+    size_t const rows = static_cast<size_t>(world) * max_tokens;
+    v_.off_nrm_count = off;
+    off = align256(off + sizeof(uint64_t) * world);
+    v_.off_nrm_flag = off;
+    off = align256(off + sizeof(uint64_t) * world);
+    v_.off_nrm_x = off;
+    off = align256(off + rows * hidden * elem_size);
+    v_.off_nrm_meta = off;
+    off = align256(off + rows * sizeof(uint32_t));
+    v_.off_nrm_topk = off;
+    off = align256(off + rows * topk * sizeof(int64_t));
+    v_.off_nrm_w = off;
+    off = align256(off + rows * topk * sizeof(float));
+    v_.off_nrm_plan = off;
+    off = align256(off + sizeof(uint32_t) * world *
+                             (1 + static_cast<size_t>(max_tokens)));
+    v_.off_nrm_ret = off;
+    off = align256(off + static_cast<size_t>(max_tokens) * world * hidden *
+                             elem_size);
+  }
   v_.heap_bytes = off;
 
   UCCL_CHECK_HIP(hipSetDevice(device_));
   UCCL_CHECK_HIP(hipMalloc(&heap_, v_.heap_bytes));
   UCCL_CHECK_HIP(hipMemset(heap_, 0, v_.off_disp_meta));
+  if (v_.with_normal) {
+    UCCL_CHECK_HIP(hipMemset(static_cast<char*>(heap_) + v_.off_nrm_count,
+                             0, sizeof(uint64_t) * world_));
+    UCCL_CHECK_HIP(hipMemset(static_cast<char*>(heap_) + v_.off_nrm_flag,
+                             0, sizeof(uint64_t) * world_));
+  }
   UCCL_CHECK_HIP(hipDeviceSynchronize());
   for (int r = 0; r < kMaxRanks; ++r) v_.peers[r] = nullptr;
   v_.peers[rank_] = heap_;
@@ -214,6 +243,32 @@ void EpBuffer::dispatch(void const* x, int64_t const* topk_idx,
                         hipStream_t stream) {
   dispatch_send(x, topk_idx, num_tokens, /*reuse_plan=*/false, stream);
   dispatch_recv(out_counts, stream);
+}
+
+void EpBuffer::nrm_dispatch_send(void const* x, int64_t const* topk_idx,
+                                 float const* topk_w, int num_tokens,
+                                 hipStream_t stream) {
+  UCCL_CHECK(v_.with_normal) << "buffer built without normal mode";
+  UCCL_CHECK(connected_ || world_ == 1) << "connect() not called";
+  UCCL_CHECK(!v_.proxy_mask) << "normal mode is xGMI-only (intranode)";
+  UCCL_CHECK(num_tokens <= v_.max_tokens) << "too many tokens";
+  ++v_.seq;
+  launch_ep_nrm_dispatch_send(v_, x, topk_idx, topk_w, num_tokens, stream);
+  nrm_last_tokens_ = num_tokens;
+}
+
+void EpBuffer::nrm_dispatch_recv(int* out_counts, hipStream_t stream) {
+  launch_ep_nrm_dispatch_recv(v_, out_counts, stream);
+}
+
+void EpBuffer::nrm_combine_send(void const* x, hipStream_t stream) {
+  UCCL_CHECK(nrm_last_tokens_ >= 0) << "combine without dispatch";
+  launch_ep_nrm_combine_send(v_, x, stream);
+}
+
+void EpBuffer::nrm_combine_recv(void* out, int64_t const* topk_idx,
+                                hipStream_t stream) {
+  launch_ep_nrm_combine_recv(v_, out, topk_idx, nrm_last_tokens_, stream);
 }
 
 void EpBuffer::barrier(hipStream_t stream) {

@@ -19,7 +19,8 @@ struct D2HRing;
 class EpBuffer {
  public:
   EpBuffer(int rank, int world, int device, int num_experts, int topk,
-           int hidden, int max_tokens, int elem_size, bool use_fp8 = false);
+           int hidden, int max_tokens, int elem_size, bool use_fp8 = false,
+           bool with_normal = true);
   ~EpBuffer();
   EpBuffer(const EpBuffer&) = delete;
 
@@ -56,6 +57,31 @@ class EpBuffer {
   // topk shape; the heap's plan scratch is untouched since)
   bool plan_cached() const { return last_num_tokens_ >= 0; }
 
+  // Normal (rank-granular) mode: DeepEP HT dispatch/combine. Each token
+  // ships once per destination rank with its topk row + weights; the
+  // receiver runs its local experts over the per-source rows and
+  // combine returns one processed row per received token, reduced at
+  // the source over contributing ranks.
+  void nrm_dispatch_send(void const* x, int64_t const* topk_idx,
+                         float const* topk_w, int num_tokens,
+                         hipStream_t stream);
+  void nrm_dispatch_recv(int* out_counts, hipStream_t stream);
+  void nrm_combine_send(void const* x, hipStream_t stream);
+  void nrm_combine_recv(void* out, int64_t const* topk_idx,
+                        hipStream_t stream);
+  void* nrm_x_base() const {
+    return static_cast<char*>(heap_) + v_.off_nrm_x;
+  }
+  void* nrm_meta_base() const {
+    return static_cast<char*>(heap_) + v_.off_nrm_meta;
+  }
+  void* nrm_topk_base() const {
+    return static_cast<char*>(heap_) + v_.off_nrm_topk;
+  }
+  void* nrm_w_base() const {
+    return static_cast<char*>(heap_) + v_.off_nrm_w;
+  }
+
   // Proxy-path synchronization commands (parity: the reference proxy's
   // ATOMIC/BARRIER/QUIET cmd types, ep/src/proxy.cpp:1629-1718 + D2H
   // cmd switch). All are GPU-initiated via the D2H ring and complete by
@@ -89,6 +115,7 @@ class EpBuffer {
   bool connected_ = false;
   int last_num_tokens_ = -1;
   uint64_t sync_seq_ = 0;
+  int nrm_last_tokens_ = -1;
 
   // proxy path (internode / forced): D2H command ring + CPU proxy over
   // the reliable transport

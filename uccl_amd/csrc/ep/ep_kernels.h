@@ -23,6 +23,18 @@ void launch_ep_combine_finish(const EpView& v, void* out,
                               int num_tokens, hipStream_t s);
 void launch_ep_comb_scatter(const EpView& v, size_t row0, size_t count,
                             hipStream_t s);
+// normal (rank-granular) mode — DeepEP HT dispatch/combine
+void launch_ep_nrm_dispatch_send(const EpView& v, void const* x,
+                                 int64_t const* topk_idx,
+                                 float const* topk_w, int num_tokens,
+                                 hipStream_t s);
+void launch_ep_nrm_dispatch_recv(const EpView& v, int* out_counts,
+                                 hipStream_t s);
+void launch_ep_nrm_combine_send(const EpView& v, void const* x,
+                                hipStream_t s);
+void launch_ep_nrm_combine_recv(const EpView& v, void* out,
+                                int64_t const* topk_idx, int num_tokens,
+                                hipStream_t s);
 // proxy sync commands (D2H ring push + device flag wait)
 void launch_ep_barrier(const EpView& v, uint64_t seq, hipStream_t s);
 void launch_ep_quiet(const EpView& v, uint64_t seq, hipStream_t s);

@@ -450,6 +450,200 @@ __global__ void k_ep_combine_reduce(EpView v, void* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
+// normal (rank-granular) mode: DeepEP HT dispatch/combine. Each token is
+// shipped ONCE per destination rank (deduped over its top-k experts)
+// with its topk row + weights; combine returns one processed row per
+// received token to its source, which reduces over contributing ranks.
+// Reference: ep/src/intranode.cu:186 (dispatch) / :722 (combine).
+// ---------------------------------------------------------------------------
+
+__global__ void k_nrm_plan(EpView v, int64_t const* __restrict__ topk_idx,
+                           int num_tokens) {
+  int const dst = blockIdx.x;  // destination rank
+  extern __shared__ uint32_t smem[];
+  uint32_t* prefix = smem;  // [blockDim.x + 1]
+  uint32_t* plan = nrm_plan_ptr(v.peers[v.rank], v, dst);  // [count, toks..]
+
+  int const seg = (num_tokens + blockDim.x - 1) / blockDim.x;
+  int const t0 = threadIdx.x * seg;
+  int const t1 = min(t0 + seg, num_tokens);
+  auto hits_rank = [&](int t) {
+    for (int k = 0; k < v.topk; ++k) {
+      int64_t const e = topk_idx[static_cast<size_t>(t) * v.topk + k];
+      if (e >= 0 && static_cast<int>(e / v.local_experts) == dst)
+        return true;
+    }
+    return false;
+  };
+  uint32_t mine = 0;
+  for (int t = t0; t < t1; ++t)
+    if (hits_rank(t)) ++mine;
+  prefix[threadIdx.x + 1] = mine;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    prefix[0] = 0;
+    for (unsigned i = 1; i <= blockDim.x; ++i) prefix[i] += prefix[i - 1];
+    plan[0] = prefix[blockDim.x];
+  }
+  __syncthreads();
+  uint32_t pos = prefix[threadIdx.x];
+  for (int t = t0; t < t1; ++t)
+    if (hits_rank(t)) plan[1 + pos++] = static_cast<uint32_t>(t);
+}
+
+__global__ void k_nrm_copy(EpView v, void const* __restrict__ x,
+                           int64_t const* __restrict__ topk_idx,
+                           float const* __restrict__ topk_w) {
+  int const fanout = gridDim.x / v.world;
+  int const dst = blockIdx.x / fanout;
+  int const b = blockIdx.x % fanout;
+  if (dst >= v.world) return;
+  if ((v.proxy_mask >> dst) & 1u) return;  // normal mode is xGMI-only
+  void* me = v.peers[v.rank];
+  void* dbase = v.peers[dst];
+  uint32_t const* plan = nrm_plan_ptr(me, v, dst);
+  uint32_t const count = plan[0];
+  size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
+  for (uint32_t i = b; i < count; i += fanout) {
+    uint32_t const t = plan[1 + i];
+    block_copy(nrm_x_ptr(dbase, v, v.rank, i),
+               static_cast<char const*>(x) +
+                   static_cast<size_t>(t) * row_bytes,
+               row_bytes);
+    if (threadIdx.x == 0) *nrm_meta_ptr(dbase, v, v.rank, i) = t;
+    // topk row (global expert ids) + weights travel with the token
+    if (threadIdx.x < static_cast<unsigned>(v.topk)) {
+      nrm_topk_ptr(dbase, v, v.rank, i)[threadIdx.x] =
+          topk_idx[static_cast<size_t>(t) * v.topk + threadIdx.x];
+      nrm_w_ptr(dbase, v, v.rank, i)[threadIdx.x] =
+          topk_w ? topk_w[static_cast<size_t>(t) * v.topk + threadIdx.x]
+                 : 0.f;
+    }
+  }
+}
+
+__global__ void k_nrm_publish(EpView v) {
+  if (threadIdx.x < static_cast<unsigned>(v.world) &&
+      !((v.proxy_mask >> threadIdx.x) & 1u)) {
+    uint32_t const count = nrm_plan_ptr(v.peers[v.rank], v, threadIdx.x)[0];
+    st_release_sys(nrm_count_ptr(v.peers[threadIdx.x], v, v.rank),
+                   tag_count(v.seq, count));
+  }
+}
+
+__global__ void k_nrm_wait(EpView v, int* __restrict__ out_counts) {
+  if (threadIdx.x < static_cast<unsigned>(v.world)) {
+    uint64_t const* p = nrm_count_ptr(v.peers[v.rank], v, threadIdx.x);
+    uint64_t got = 0;
+    for (uint64_t it = 0;; ++it) {
+      got = ld_acquire_sys(p);
+      if ((got >> 32) == v.seq) break;
+      if (it > (1ull << 28)) {
+        printf("uccl_ep: nrm wait TIMEOUT rank=%d src=%d seq=%llu\n",
+               v.rank, static_cast<int>(threadIdx.x),
+               (unsigned long long)v.seq);
+        __builtin_trap();
+      }
+      backoff();
+    }
+    out_counts[threadIdx.x] = static_cast<int>(got & 0xffffffffu);
+  }
+}
+
+// combine returns: block (src, b) walks rows received from `src` and
+// writes the processed row back into src's per-(token, my_rank) cell.
+__global__ void k_nrm_return(EpView v, void const* __restrict__ x) {
+  int const fanout = gridDim.x / v.world;
+  int const src = blockIdx.x / fanout;
+  int const b = blockIdx.x % fanout;
+  if (src >= v.world) return;
+  if ((v.proxy_mask >> src) & 1u) return;
+  void* me = v.peers[v.rank];
+  uint64_t const tagged = *nrm_count_ptr(me, v, src);
+  uint32_t const count = static_cast<uint32_t>(tagged & 0xffffffffu);
+  void* sbase = v.peers[src];
+  size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
+  for (uint32_t i = b; i < count; i += fanout) {
+    uint32_t const t = *nrm_meta_ptr(me, v, src, i);
+    block_copy_nt(nrm_ret_ptr(sbase, v, t, v.rank),
+                  static_cast<char const*>(x) +
+                      (static_cast<size_t>(src) * v.max_tokens + i) *
+                          row_bytes,
+                  row_bytes);
+  }
+}
+
+__global__ void k_nrm_ret_signal(EpView v) {
+  if (threadIdx.x < static_cast<unsigned>(v.world) &&
+      !((v.proxy_mask >> threadIdx.x) & 1u))
+    st_release_sys(nrm_flag_ptr(v.peers[threadIdx.x], v, v.rank), v.seq);
+}
+
+__global__ void k_nrm_ret_wait(EpView v) {
+  if (threadIdx.x < static_cast<unsigned>(v.world)) {
+    uint64_t const* p = nrm_flag_ptr(v.peers[v.rank], v, threadIdx.x);
+    for (uint64_t it = 0;; ++it) {
+      if (ld_acquire_sys(p) >= v.seq) return;
+      if (it > (1ull << 28)) {
+        printf("uccl_ep: nrm ret wait TIMEOUT rank=%d src=%d\n", v.rank,
+               static_cast<int>(threadIdx.x));
+        __builtin_trap();
+      }
+      backoff();
+    }
+  }
+}
+
+// reduce: out[t] = sum over ranks r that received token t of ret[t][r]
+// (fp32 accumulation; contributing set recomputed from topk_idx)
+template <typename T>
+__global__ void k_nrm_reduce(EpView v, void* __restrict__ out,
+                             int64_t const* __restrict__ topk_idx,
+                             int num_tokens) {
+  int const t = blockIdx.x;
+  if (t >= num_tokens) return;
+  void* me = v.peers[v.rank];
+  uint32_t contributes = 0;  // bitmask of ranks that got this token
+  for (int k = 0; k < v.topk; ++k) {
+    int64_t const e = topk_idx[static_cast<size_t>(t) * v.topk + k];
+    if (e >= 0) contributes |= 1u << static_cast<int>(e / v.local_experts);
+  }
+  size_t const vper = 16 / sizeof(T);
+  size_t const nvec = static_cast<size_t>(v.hidden) / vper;
+  auto* orow = reinterpret_cast<V16*>(
+      static_cast<char*>(out) +
+      static_cast<size_t>(t) * v.hidden * sizeof(T));
+  for (size_t i = threadIdx.x; i < nvec; i += blockDim.x) {
+    float acc[16 / sizeof(T)] = {};
+    for (int r = 0; r < v.world; ++r) {
+      if (!((contributes >> r) & 1u)) continue;
+      V16 const val = reinterpret_cast<V16 const*>(
+          nrm_ret_ptr(me, v, t, r))[i];
+#pragma unroll
+      for (size_t j = 0; j < vper; ++j)
+        acc[j] += static_cast<float>(reinterpret_cast<T const*>(&val)[j]);
+    }
+    V16 rvec;
+#pragma unroll
+    for (size_t j = 0; j < vper; ++j)
+      reinterpret_cast<T*>(&rvec)[j] = static_cast<T>(acc[j]);
+    orow[i] = rvec;
+  }
+  size_t const tail = v.hidden - nvec * vper;
+  if (tail && threadIdx.x < tail) {
+    size_t const j = nvec * vper + threadIdx.x;
+    float acc = 0.f;
+    for (int r = 0; r < v.world; ++r) {
+      if (!((contributes >> r) & 1u)) continue;
+      acc += static_cast<float>(
+          reinterpret_cast<T const*>(nrm_ret_ptr(me, v, t, r))[j]);
+    }
+    static_cast<T*>(out)[static_cast<size_t>(t) * v.hidden + j] =
+        static_cast<T>(acc);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // proxy sync commands (ATOMIC/BARRIER/QUIET): one-thread push into the
 // D2H ring + a paired single-block wait on the device-visible flag the
 // proxy writes back (same trap-not-wedge discipline as the other waits).
@@ -510,6 +704,44 @@ void launch_ep_dispatch(const EpView& v, void const* x,
                         int* out_counts, hipStream_t s) {
   launch_ep_dispatch_send(v, x, topk_idx, num_tokens, false, s);
   launch_ep_dispatch_recv(v, out_counts, s);
+}
+
+void launch_ep_nrm_dispatch_send(const EpView& v, void const* x,
+                                 int64_t const* topk_idx,
+                                 float const* topk_w, int num_tokens,
+                                 hipStream_t s) {
+  if (v.seq > 1 && v.world > 1)
+    k_ep_consume_gate<<<1, 64, 0, s>>>(v, v.seq - 1);
+  size_t const smem = 257 * sizeof(uint32_t);
+  k_nrm_plan<<<v.world, 256, smem, s>>>(v, topk_idx, num_tokens);
+  k_nrm_copy<<<v.world * fanout_for(v.world), 256, 0, s>>>(v, x, topk_idx,
+                                                           topk_w);
+  k_nrm_publish<<<1, 64, 0, s>>>(v);
+}
+
+void launch_ep_nrm_dispatch_recv(const EpView& v, int* out_counts,
+                                 hipStream_t s) {
+  k_nrm_wait<<<1, 64, 0, s>>>(v, out_counts);
+  if (v.world > 1) k_ep_consume_signal<<<1, 64, 0, s>>>(v, v.seq);
+}
+
+void launch_ep_nrm_combine_send(const EpView& v, void const* x,
+                                hipStream_t s) {
+  k_nrm_return<<<v.world * fanout_for(v.world), 256, 0, s>>>(v, x);
+  k_nrm_ret_signal<<<1, 64, 0, s>>>(v);
+}
+
+void launch_ep_nrm_combine_recv(const EpView& v, void* out,
+                                int64_t const* topk_idx, int num_tokens,
+                                hipStream_t s) {
+  k_nrm_ret_wait<<<1, 64, 0, s>>>(v);
+  if (v.elem_size == 2) {
+    k_nrm_reduce<__hip_bfloat16>
+        <<<num_tokens, 256, 0, s>>>(v, out, topk_idx, num_tokens);
+  } else {
+    k_nrm_reduce<float>
+        <<<num_tokens, 256, 0, s>>>(v, out, topk_idx, num_tokens);
+  }
 }
 
 void launch_ep_barrier(const EpView& v, uint64_t seq, hipStream_t s) {

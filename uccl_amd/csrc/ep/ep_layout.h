@@ -59,6 +59,21 @@ struct EpView {
   size_t off_egress_meta; // u32 [max_tokens*topk]
   size_t off_ingress;     // proxy RX staging rows (combine returns)
   size_t off_ingress_meta;
+
+  // --- normal (rank-granular) mode: DeepEP's HT dispatch ships each
+  // token ONCE per destination RANK (deduped across its top-k experts)
+  // together with its topk row + weights; combine returns one row per
+  // received token to its source (reference ep/src/intranode.cu:186
+  // dispatch / :722 combine). Regions exist when with_normal.
+  int with_normal;
+  size_t off_nrm_count;   // u64 [world]            (seq<<32 | count)
+  size_t off_nrm_x;       // [world][max_tokens][hidden] elem
+  size_t off_nrm_meta;    // u32 [world][max_tokens]  src token idx
+  size_t off_nrm_topk;    // i64 [world][max_tokens][topk]  (global ids)
+  size_t off_nrm_w;       // f32 [world][max_tokens][topk]
+  size_t off_nrm_plan;    // u32 [world][1 + max_tokens] private scratch
+  size_t off_nrm_ret;     // [max_tokens][world][hidden] elem (returns)
+  size_t off_nrm_flag;    // u64 [world]            (return seq)
   size_t heap_bytes;
 };
 
@@ -127,6 +142,67 @@ __host__ __device__ inline uint32_t* ingress_meta(void* base,
   return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
                                      v.off_ingress_meta) +
          row;
+}
+
+// --- normal-mode helpers ---------------------------------------------------
+
+__host__ __device__ inline uint64_t* nrm_count_ptr(void* base,
+                                                   const EpView& v,
+                                                   int src) {
+  return reinterpret_cast<uint64_t*>(static_cast<char*>(base) +
+                                     v.off_nrm_count) +
+         src;
+}
+
+__host__ __device__ inline char* nrm_x_ptr(void* base, const EpView& v,
+                                           int src, size_t row) {
+  return static_cast<char*>(base) + v.off_nrm_x +
+         ((static_cast<size_t>(src) * v.max_tokens + row) * v.hidden) *
+             v.elem_size;
+}
+
+__host__ __device__ inline uint32_t* nrm_meta_ptr(void* base,
+                                                  const EpView& v, int src,
+                                                  size_t row) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
+                                     v.off_nrm_meta) +
+         static_cast<size_t>(src) * v.max_tokens + row;
+}
+
+__host__ __device__ inline int64_t* nrm_topk_ptr(void* base,
+                                                 const EpView& v, int src,
+                                                 size_t row) {
+  return reinterpret_cast<int64_t*>(static_cast<char*>(base) +
+                                    v.off_nrm_topk) +
+         (static_cast<size_t>(src) * v.max_tokens + row) * v.topk;
+}
+
+__host__ __device__ inline float* nrm_w_ptr(void* base, const EpView& v,
+                                            int src, size_t row) {
+  return reinterpret_cast<float*>(static_cast<char*>(base) + v.off_nrm_w) +
+         (static_cast<size_t>(src) * v.max_tokens + row) * v.topk;
+}
+
+__host__ __device__ inline uint32_t* nrm_plan_ptr(void* base,
+                                                  const EpView& v, int dst) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
+                                     v.off_nrm_plan) +
+         static_cast<size_t>(dst) * (1 + v.max_tokens);
+}
+
+__host__ __device__ inline char* nrm_ret_ptr(void* base, const EpView& v,
+                                             size_t tok, int src) {
+  return static_cast<char*>(base) + v.off_nrm_ret +
+         ((tok * v.world + src) * static_cast<size_t>(v.hidden)) *
+             v.elem_size;
+}
+
+__host__ __device__ inline uint64_t* nrm_flag_ptr(void* base,
+                                                  const EpView& v,
+                                                  int src) {
+  return reinterpret_cast<uint64_t*>(static_cast<char*>(base) +
+                                     v.off_nrm_flag) +
+         src;
 }
 
 __host__ __device__ inline uint64_t* comb_flag_ptr(void* base,

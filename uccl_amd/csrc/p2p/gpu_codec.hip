@@ -385,7 +385,10 @@ size_t encode_plane(uint8_t const* src, size_t bytes, uint8_t* dst,
   uint64_t run = 0;
   for (int i = 0; i < nblocks; ++i) {
     hoffs[i] = run;
-    run += hsizes[i];
+    // 8-align every block so the decoder's typed u64/u32 header loads
+    // are aligned (unaligned global loads FAULT on CDNA; found the hard
+    // way at the first multi-block plane)
+    run += align8(hsizes[i]);
   }
   PlaneHdr ph{run, static_cast<uint32_t>(nblocks), 0};
   UCCL_CHECK_HIP(hipMemcpyAsync(dst, &ph, sizeof(ph),

@@ -142,6 +142,34 @@ class Buffer:
         return self._b.combine_recv(out, topk_idx,
                                     topk_weights.float().contiguous())
 
+    # -- normal (rank-granular) mode: DeepEP HT semantics -------------------
+    def nrm_dispatch(self, x: torch.Tensor, topk_idx: torch.Tensor,
+                     topk_weights: Optional[torch.Tensor] = None):
+        """Rank-granular dispatch: each token ships ONCE per destination
+        rank (deduped over its top-k experts) together with its topk row
+        and weights. Returns (recv_x [world, max_tokens, hidden] view,
+        counts [world] int32, recv_topk [world, max_tokens, topk] i64
+        view with GLOBAL expert ids, recv_w view). Rows for source r are
+        [r, :counts[r]]."""
+        w = topk_weights.float().contiguous() if topk_weights is not None             else None
+        self._b.nrm_dispatch_send(x, topk_idx, w)
+        counts = torch.empty(self.world, dtype=torch.int32, device="cuda")
+        self._b.nrm_dispatch_recv(counts)
+        return (self._b.nrm_x_view(), counts, self._b.nrm_topk_view(),
+                self._b.nrm_w_view())
+
+    def nrm_combine(self, x: torch.Tensor, topk_idx: torch.Tensor,
+                    out: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Return one processed row per received token ([world,
+        max_tokens, hidden] layout) to its source and reduce over
+        contributing ranks (weights are expected to be applied by the
+        receiver's expert computation, DeepEP normal-combine style)."""
+        self._b.nrm_combine_send(x)
+        if out is None:
+            out = torch.empty(topk_idx.shape[0], self.hidden,
+                              dtype=self.dtype, device="cuda")
+        return self._b.nrm_combine_recv(out, topk_idx)
+
     def close(self):
         """Release the native buffer (symmetric heap + IPC handles)
         promptly instead of waiting for GC — used by ElasticBuffer on

@@ -203,10 +203,86 @@ class Buffer:
             return dist.get_world_size(group=self._group)
         return 1
 
-    # convenience aliases matching the high-throughput and internode entry
-    # points (one xGMI engine serves all three DeepEP modes; internode
-    # peers ride the proxy path automatically)
-    dispatch = low_latency_dispatch
-    combine = low_latency_combine
+    # -- normal (HT) mode: DeepEP's rank-granular dispatch signature --------
+    def dispatch(self, x, handle=None, num_tokens_per_rank=None,
+                 num_tokens_per_rdma_rank=None, is_token_in_rank=None,
+                 num_tokens_per_expert=None, topk_idx=None,
+                 topk_weights=None, expert_alignment: int = 1,
+                 config=None, previous_event=None, async_finish=False,
+                 allocate_on_comm_stream=False):
+        """DeepEP normal-mode dispatch (reference ep/bench/buffer.py:898):
+        each token goes ONCE to every rank owning >=1 of its top-k
+        experts, carrying its topk row + weights. Returns
+        (recv_x, recv_topk_idx, recv_topk_weights,
+         num_recv_tokens_per_expert_list, handle, event).
+        recv_x is the per-source concatenation [sum(counts), hidden];
+        recv_topk_idx holds LOCAL expert ids (-1 for entries owned by
+        other ranks), per DeepEP semantics."""
+        assert topk_idx is not None, "normal dispatch needs topk_idx"
+        T, K = topk_idx.shape
+        nb = self._ensure(x.shape[1],
+                          max(T, num_tokens_per_rank.max().item()
+                              if num_tokens_per_rank is not None else T),
+                          self._cfg[2] if self._cfg else K * 8,
+                          K, x.dtype, False) if self._native is None else             self._native
+        # reuse the existing geometry when compatible, else (re)build
+        if (self._native is None or self._native.topk != K or
+                self._native.hidden != x.shape[1]):
+            num_experts = (len(num_tokens_per_expert)
+                           if num_tokens_per_expert is not None else K * 8)
+            nb = self._ensure(x.shape[1], max(256, T), num_experts, K,
+                              x.dtype, False)
+        nb = self._native
+        rx, counts, rtopk, rw = nb.nrm_dispatch(x, topk_idx, topk_weights)
+        torch.cuda.current_stream().synchronize()
+        cnts = counts.tolist()
+        rows_x, rows_topk, rows_w = [], [], []
+        for r in range(nb.world):
+            n = cnts[r]
+            rows_x.append(rx[r, :n])
+            rows_topk.append(rtopk[r, :n])
+            rows_w.append(rw[r, :n])
+        recv_x = torch.cat(rows_x) if rows_x else rx[0, :0]
+        gtopk = torch.cat(rows_topk) if rows_topk else rtopk[0, :0]
+        recv_w = torch.cat(rows_w) if rows_w else rw[0, :0]
+        # global -> local expert ids; -1 for other ranks' experts
+        local_lo = nb.rank * nb.local_experts
+        local_hi = local_lo + nb.local_experts
+        mine = (gtopk >= local_lo) & (gtopk < local_hi)
+        recv_topk_idx = torch.where(mine, gtopk - local_lo,
+                                    torch.full_like(gtopk, -1))
+        recv_topk_weights = torch.where(mine, recv_w,
+                                        torch.zeros_like(recv_w))
+        per_expert = [int((recv_topk_idx == e).sum())
+                      for e in range(nb.local_experts)]
+        h = _Handle(topk_idx, T, self._plan_gen)
+        h.counts = cnts
+        return (recv_x, recv_topk_idx, recv_topk_weights, per_expert, h,
+                _event())
+
+    def combine(self, x, handle, topk_weights=None, config=None,
+                previous_event=None, async_finish=False,
+                allocate_on_comm_stream=False):
+        """DeepEP normal-mode combine: `x` holds one processed row per
+        received token in dispatch order (the per-source concatenation
+        returned by dispatch, weights already applied receiver-side);
+        each row returns to its source, which sums over contributing
+        ranks. Returns (combined_x, event)."""
+        nb = self._native
+        assert nb is not None and hasattr(handle, "counts")
+        # scatter the concatenated rows back into the [world, max_tokens]
+        # slot layout the return kernel walks
+        buf = torch.zeros(nb.world, nb.max_tokens, nb.hidden,
+                          dtype=x.dtype, device=x.device)
+        off = 0
+        for r in range(nb.world):
+            n = handle.counts[r]
+            buf[r, :n] = x[off:off + n]
+            off += n
+        combined = nb.nrm_combine(buf, handle.topk_idx)
+        return combined, _event()
+
+    # low-latency aliases for the internode entry points (the proxy path
+    # rides the same engine)
     internode_dispatch = low_latency_dispatch
     internode_combine = low_latency_combine
