@@ -133,3 +133,39 @@ def test_p2p_stats_cpu():
     assert st["send"]["bytes"] == 40000
     assert st["send"]["p50_us"] > 0
     assert b.stats()["recv"]["calls"] == 1
+
+
+def test_rccl_plane_graceful_fallback(monkeypatch):
+    """UCCL_P2P_TRANSPORT=rccl on a GPU-less box: both sides negotiate,
+    detect the plane is unusable, and fall back to the TCP plane — the
+    transfer still completes (reference parity: the p2p NCCL backend is
+    the portability fallback, p2p/nccl/nccl_endpoint.h:58)."""
+    import torch
+
+    monkeypatch.setenv("UCCL_P2P_TRANSPORT", "rccl")
+    from uccl_amd import p2p
+
+    a = p2p.Endpoint(gpu=0, num_workers=1)
+    b = p2p.Endpoint(gpu=0, num_workers=1)
+    import threading
+
+    ids = {}
+    t = threading.Thread(target=lambda: ids.__setitem__("b", b.accept()))
+    t.start()
+    ids["a"] = a.connect(b.metadata())
+    t.join(timeout=20)
+    src = torch.arange(4096, dtype=torch.uint8)
+    dst = torch.zeros(4096, dtype=torch.uint8)
+    exc = []
+
+    def rx():
+        try:
+            b.recv(ids["b"], dst)
+        except Exception as e:
+            exc.append(e)
+
+    t2 = threading.Thread(target=rx)
+    t2.start()
+    a.send(ids["a"], src)
+    t2.join(timeout=30)
+    assert not exc and torch.equal(src, dst)

@@ -25,6 +25,7 @@ SOURCES = [
     CSRC / "collective" / "communicator.cpp",
     CSRC / "collective" / "pg_backend.cpp",
     CSRC / "p2p" / "endpoint.cpp",
+    CSRC / "p2p" / "rccl_plane.cpp",
     CSRC / "transport" / "reliable.cpp",
     CSRC / "transport" / "udp_fabric.cpp",
     CSRC / "transport" / "verbs_fabric.cpp",
@@ -71,6 +72,7 @@ def build_plugin(verbose: bool = False) -> Path:
         CSRC / "transport" / "reliable.cpp"]
     capi = plugdir / "libuccl_p2p.so"
     capi_srcs = [CSRC / "p2p" / "c_api.cpp", CSRC / "p2p" / "endpoint.cpp",
+                 CSRC / "p2p" / "rccl_plane.cpp",
                  CSRC / "transport" / "reliable.cpp",
                  CSRC / "transport" / "udp_fabric.cpp",
                  CSRC / "transport" / "verbs_fabric.cpp",

@@ -4,6 +4,8 @@
 #include <sched.h>
 #include <unistd.h>
 
+#include <chrono>
+
 #include <cstring>
 #include <thread>
 
@@ -81,6 +83,27 @@ EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
 }
 
 EpProxy::~EpProxy() {
+  // DRAIN before shutdown: a rank may destroy its buffer as soon as ITS
+  // combine completed, while peers still await the returns this proxy is
+  // shipping asynchronously (comb_tx_loop) or dispatch rows still queued
+  // in the D2H ring. Killing the transport mid-ship strands those peers
+  // in their device waits (observed at 256-expert forced-proxy: the
+  // faster rank's close() trapped the slower rank).
+  {
+    std::unique_lock<std::mutex> lk(mu_);
+    cv_.wait_for(lk, std::chrono::seconds(20),
+                 [this] { return comb_q_.empty() && !comb_busy_; });
+  }
+  if (ring_) {
+    for (int i = 0; i < 20000; ++i) {  // up to ~20s
+      uint64_t const h = __atomic_load_n(
+          const_cast<uint64_t*>(&ring_->head), __ATOMIC_ACQUIRE);
+      uint64_t const t = __atomic_load_n(
+          const_cast<uint64_t*>(&ring_->tail), __ATOMIC_ACQUIRE);
+      if (h == t) break;
+      usleep(1000);
+    }
+  }
   {
     std::lock_guard<std::mutex> g(mu_);  // lost-wakeup guard
     stop_ = true;
@@ -293,6 +316,7 @@ void EpProxy::comb_tx_loop() {
       if (stop_ && comb_q_.empty()) return;
       task = std::move(comb_q_.front());
       comb_q_.pop_front();
+      comb_busy_ = true;
     }
     (void)hipEventSynchronize(task.ready);
     size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
@@ -325,9 +349,19 @@ void EpProxy::comb_tx_loop() {
     } catch (std::exception const& e) {
       if (!stop_)
         UCCL_LOG_ERROR << "ep proxy comb_tx_loop died: " << e.what();
+      {
+        std::lock_guard<std::mutex> g(mu_);
+        comb_busy_ = false;
+      }
+      cv_.notify_all();
       return;
     }
     (void)hipEventDestroy(task.ready);
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      comb_busy_ = false;
+    }
+    cv_.notify_all();
   }
 }
 

@@ -104,6 +104,7 @@ class EpProxy {
                         // has completed (the D2H precedes it in-stream)
   };
   std::deque<CombTask> comb_q_;
+  bool comb_busy_ = false;  // a task is being shipped right now
   // barrier bookkeeping: arrivals per seq (self + every proxy peer)
   std::mutex bar_mu_;
   std::map<uint64_t, int> bar_seen_;

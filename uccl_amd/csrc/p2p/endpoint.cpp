@@ -1,4 +1,5 @@
 #include "endpoint.h"
+#include "rccl_plane.h"
 
 #include <hip/hip_runtime.h>
 #include <unistd.h>
@@ -31,6 +32,12 @@ enum Op : uint64_t {
   kReadReq = 7,      // a=mr, b=off, c=bytes, d=token
   kReadResp = 8,     // a=bytes, d=token                      + payload
   kIpcDone = 9,      // d=token
+  // RCCL-as-transport plane (UCCL_P2P_TRANSPORT=rccl): headers ride the
+  // TCP channel for matching/ordering; payloads move device-to-device
+  // as ncclSend/ncclRecv on the connection's 2-rank comm
+  kSendRccl = 10,    // a=bytes
+  kWriteRccl = 11,   // a=mr, b=off, c=bytes, d=token
+  kReadReqRccl = 12, // a=mr, b=off, c=bytes, d=token
 };
 
 struct MsgHdr {
@@ -57,6 +64,12 @@ struct Meta {
 
 bool is_gpu(int device) { return device >= 0; }
 
+bool p2p_rccl() {
+  static bool v = uccl::env_str("UCCL_P2P_TRANSPORT", "tcp") ==
+                  std::string("rccl");
+  return v;
+}
+
 bool p2p_multipath() {
   static bool v = uccl::env_str("UCCL_P2P_TRANSPORT", "tcp") ==
                   std::string("multipath");
@@ -68,9 +81,11 @@ constexpr size_t kMsgChunk = 8ull << 20;  // payload message granule
 }  // namespace
 
 struct Endpoint::RxItem {
-  // Either inline host data, or an IPC descriptor to copy from.
+  // Inline host data, an IPC descriptor to copy from, or an RCCL-plane
+  // payload to be received directly into the consumer's buffer.
   std::vector<char> data;
   bool ipc = false;
+  bool rccl = false;
   IpcBlob blob{};
   uint64_t token = 0;
   size_t bytes = 0;
@@ -111,6 +126,12 @@ struct Endpoint::Conn {
   // IPC handle cache: src (pid, base-handle bytes) -> mapped ptr
   std::unordered_map<std::string, void*> ipc_cache;
   std::mutex ipc_mu;
+
+  // RCCL-as-transport plane state (UCCL_P2P_TRANSPORT=rccl)
+  void* rccl_comm = nullptr;
+  int rccl_rank = -1;  // 0 = acceptor, 1 = connector; peer = 1 - rank
+  hipStream_t rccl_tx = nullptr;  // sends and recvs ride separate
+  hipStream_t rccl_rx = nullptr;  // streams: no bidirectional deadlock
 
   // one logical message = header + chunked payload; over the multipath
   // plane each piece is a discrete reliable message in the same order
@@ -204,6 +225,7 @@ Endpoint::Endpoint(int gpu, int num_workers) : gpu_(gpu) {
         c->flow = tp_flows_[h.d];
         tp_flows_.erase(h.d);
       }
+      if (p2p_rccl()) setup_rccl(*c, /*acceptor=*/true);
       {
         std::lock_guard<std::mutex> g(conn_mu_);
         conns_[c->id] = c;
@@ -282,6 +304,12 @@ Endpoint::~Endpoint() {
     std::lock_guard<std::mutex> g(conn_mu_);
     for (auto& [id, c] : conns_) {
       if (c->rx.joinable()) c->rx.join();
+      if (c->rccl_comm) {
+        RcclPlane::get().comm_destroy(c->rccl_comm);
+        c->rccl_comm = nullptr;
+      }
+      if (c->rccl_tx) (void)hipStreamDestroy(c->rccl_tx);
+      if (c->rccl_rx) (void)hipStreamDestroy(c->rccl_rx);
       ::close(c->fd);
       for (auto& [k, p] : c->ipc_cache) (void)hipIpcCloseMemHandle(p);
     }
@@ -306,6 +334,41 @@ std::string Endpoint::metadata() const {
     memcpy(m.tp_md, md.data(), std::min(md.size(), sizeof(m.tp_md)));
   }
   return std::string(reinterpret_cast<char*>(&m), sizeof(m));
+}
+
+// RCCL-as-transport bootstrap over the fresh TCP connection: exchange
+// availability flags, ship the uniqueId acceptor->connector, then the
+// collective 2-rank comm init (reference parity: p2p/nccl backend,
+// nccl_endpoint.h:58). Any failure leaves rccl_comm null and the
+// connection on the default plane.
+void Endpoint::setup_rccl(Conn& c, bool acceptor) {
+  uint8_t mine = (gpu_ >= 0 && RcclPlane::available()) ? 1 : 0;
+  uint8_t theirs = 0;
+  net::send_all(c.fd, &mine, 1);
+  if (!net::recv_all(c.fd, &theirs, 1)) return;
+  if (!mine || !theirs) {
+    UCCL_LOG_WARN << "p2p rccl plane unavailable; using default plane";
+    return;
+  }
+  RcclPlane::UniqueId id{};
+  auto& plane = RcclPlane::get();
+  if (acceptor) {
+    if (!plane.create_unique_id(&id)) return;
+    net::send_all(c.fd, id.data, sizeof(id.data));
+  } else {
+    if (!net::recv_all(c.fd, id.data, sizeof(id.data))) return;
+  }
+  c.rccl_rank = acceptor ? 0 : 1;
+  c.rccl_comm = plane.comm_init(c.rccl_rank, id, gpu_);
+  if (!c.rccl_comm) {
+    c.rccl_rank = -1;
+    return;
+  }
+  UCCL_CHECK_HIP(hipStreamCreateWithFlags(&c.rccl_tx,
+                                          hipStreamNonBlocking));
+  UCCL_CHECK_HIP(hipStreamCreateWithFlags(&c.rccl_rx,
+                                          hipStreamNonBlocking));
+  UCCL_LOG_INFO << "p2p conn " << c.id << " using the RCCL data plane";
 }
 
 uint64_t Endpoint::connect(const std::string& remote_metadata) {
@@ -338,6 +401,7 @@ uint64_t Endpoint::connect(const std::string& remote_metadata) {
     c->tp = tp_.get();
     c->flow = tp_->connect(std::string(m.tp_md, m.tp_len), nonce);
   }
+  if (p2p_rccl()) setup_rccl(*c, /*acceptor=*/false);
   {
     std::lock_guard<std::mutex> g(conn_mu_);
     conns_[c->id] = c;
@@ -507,6 +571,74 @@ void Endpoint::rx_loop_body(std::shared_ptr<Conn> c) {
         c->tok_cv.notify_all();
         break;
       }
+      case kSendRccl: {
+        auto item = std::make_shared<RxItem>();
+        item->bytes = h.a;
+        item->rccl = true;
+        {
+          std::lock_guard<std::mutex> g(c->rx_mu);
+          c->rxq.push_back(item);
+        }
+        c->rx_cv.notify_all();
+        break;
+      }
+      case kWriteRccl: {
+        MR mr;
+        {
+          std::lock_guard<std::mutex> g(mr_mu_);
+          auto it = mrs_.find(h.a);
+          UCCL_CHECK(it != mrs_.end()) << "rccl write to unknown mr";
+          mr = it->second;
+        }
+        UCCL_CHECK(h.b + h.c <= mr.bytes) << "rccl write overflows mr";
+        UCCL_CHECK(c->rccl_comm) << "rccl frame on non-rccl conn";
+        char* dst = static_cast<char*>(mr.ptr) + h.b;
+        if (is_gpu(mr.device)) {
+          UCCL_CHECK(RcclPlane::get().recv(c->rccl_comm, dst, h.c,
+                                           1 - c->rccl_rank, c->rccl_rx))
+              << "rccl recv failed";
+        } else {
+          // host MR: land in a temp device buffer, then DtoH
+          void* tmp = nullptr;
+          UCCL_CHECK_HIP(hipMalloc(&tmp, h.c));
+          bool ok = RcclPlane::get().recv(c->rccl_comm, tmp, h.c,
+                                          1 - c->rccl_rank, c->rccl_rx);
+          if (ok)
+            UCCL_CHECK_HIP(hipMemcpy(dst, tmp, h.c,
+                                     hipMemcpyDeviceToHost));
+          (void)hipFree(tmp);
+          UCCL_CHECK(ok) << "rccl recv failed";
+        }
+        if (h.d) c->send_msg(MsgHdr{kWriteAck, 0, 0, 0, h.d});
+        break;
+      }
+      case kReadReqRccl: {
+        MR mr;
+        {
+          std::lock_guard<std::mutex> g(mr_mu_);
+          auto it = mrs_.find(h.a);
+          UCCL_CHECK(it != mrs_.end()) << "rccl read of unknown mr";
+          mr = it->second;
+        }
+        UCCL_CHECK(h.b + h.c <= mr.bytes) << "rccl read overflows mr";
+        UCCL_CHECK(c->rccl_comm) << "rccl frame on non-rccl conn";
+        char* src = static_cast<char*>(mr.ptr) + h.b;
+        if (is_gpu(mr.device)) {
+          UCCL_CHECK(RcclPlane::get().send(c->rccl_comm, src, h.c,
+                                           1 - c->rccl_rank, c->rccl_tx))
+              << "rccl send failed";
+        } else {
+          void* tmp = nullptr;
+          UCCL_CHECK_HIP(hipMalloc(&tmp, h.c));
+          UCCL_CHECK_HIP(hipMemcpy(tmp, src, h.c,
+                                   hipMemcpyHostToDevice));
+          bool ok = RcclPlane::get().send(c->rccl_comm, tmp, h.c,
+                                          1 - c->rccl_rank, c->rccl_tx);
+          (void)hipFree(tmp);
+          UCCL_CHECK(ok) << "rccl send failed";
+        }
+        break;
+      }
       case kWriteAck:
       case kIpcDone: {
         {
@@ -555,6 +687,15 @@ static bool ipc_enabled() {
 void Endpoint::do_send(Conn& c, void const* ptr, size_t bytes, int device) {
   trace::Span span("p2p", "send");
   OpTimer ot__(st_send_, bytes);
+  if (c.rccl_comm && is_gpu(device)) {
+    // RCCL plane: header over TCP for matching order; payload moves
+    // device-to-device as ncclSend (blocks until the peer's recv posts)
+    c.send_msg(MsgHdr{kSendRccl, bytes, 0, 0, 0});
+    UCCL_CHECK(RcclPlane::get().send(c.rccl_comm, ptr, bytes,
+                                     1 - c.rccl_rank, c.rccl_tx))
+        << "rccl plane send failed";
+    return;
+  }
   if (is_gpu(device) && c.same_host && ipc_enabled()) {
     // one-copy IPC path: ship {handle, offset}; receiver DtoD-copies
     IpcBlob blob{};
@@ -639,6 +780,26 @@ void Endpoint::do_recv(Conn& c, void* ptr, size_t bytes, int device) {
     item = c.rxq.front();
     c.rxq.pop_front();
   }
+  if (item->rccl) {
+    UCCL_CHECK(item->bytes <= bytes) << "recv buffer too small";
+    UCCL_CHECK(c.rccl_comm) << "rccl frame on non-rccl conn";
+    if (is_gpu(device)) {
+      UCCL_CHECK(RcclPlane::get().recv(c.rccl_comm, ptr, item->bytes,
+                                       1 - c.rccl_rank, c.rccl_rx))
+          << "rccl plane recv failed";
+    } else {
+      void* tmp = nullptr;
+      UCCL_CHECK_HIP(hipMalloc(&tmp, item->bytes));
+      bool ok = RcclPlane::get().recv(c.rccl_comm, tmp, item->bytes,
+                                      1 - c.rccl_rank, c.rccl_rx);
+      if (ok)
+        UCCL_CHECK_HIP(hipMemcpy(ptr, tmp, item->bytes,
+                                 hipMemcpyDeviceToHost));
+      (void)hipFree(tmp);
+      UCCL_CHECK(ok) << "rccl plane recv failed";
+    }
+    return;
+  }
   if (item->ipc) {
     UCCL_CHECK(is_gpu(device)) << "IPC send into host recv buffer";
     UCCL_CHECK(item->bytes <= bytes) << "recv buffer too small";
@@ -660,7 +821,12 @@ void Endpoint::do_write(Conn& c, void const* ptr, size_t bytes, int device,
   OpTimer ot__(st_write_, bytes);
   UCCL_CHECK(bytes <= ad.bytes) << "write larger than advertised window";
   uint64_t token = c.next_token++;
-  if (is_gpu(device) && c.same_host && ipc_enabled()) {
+  if (c.rccl_comm && is_gpu(device)) {
+    c.send_msg(MsgHdr{kWriteRccl, ad.mr_id, ad.offset, bytes, token});
+    UCCL_CHECK(RcclPlane::get().send(c.rccl_comm, ptr, bytes,
+                                     1 - c.rccl_rank, c.rccl_tx))
+        << "rccl plane send failed";
+  } else if (is_gpu(device) && c.same_host && ipc_enabled()) {
     IpcBlob blob{};
     void* base = nullptr;
     size_t base_sz = 0;
@@ -698,6 +864,15 @@ void Endpoint::do_read(Conn& c, void* ptr, size_t bytes, int device,
   OpTimer ot__(st_read_, bytes);
   UCCL_CHECK(bytes <= ad.bytes) << "read larger than advertised window";
   uint64_t token = c.next_token++;
+  if (c.rccl_comm && is_gpu(device)) {
+    // rccl plane: the owner's rx loop ncclSends from the MR; our recv
+    // completing IS the data arrival
+    c.send_msg(MsgHdr{kReadReqRccl, ad.mr_id, ad.offset, bytes, token});
+    UCCL_CHECK(RcclPlane::get().recv(c.rccl_comm, ptr, bytes,
+                                     1 - c.rccl_rank, c.rccl_rx))
+        << "rccl plane recv failed";
+    return;
+  }
   c.send_msg(MsgHdr{kReadReq, ad.mr_id, ad.offset, bytes, token});
   std::shared_ptr<RxItem> item;
   {

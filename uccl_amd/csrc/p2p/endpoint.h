@@ -117,6 +117,7 @@ class Endpoint {
   void do_read(Conn& c, void* ptr, size_t bytes, int device, Advert ad);
   void copy_to_user(RxItem& item, void* dst, size_t bytes, int device);
   void* open_ipc(Conn& c, const void* handle_bytes, int src_device);
+  void setup_rccl(Conn& c, bool acceptor);
   std::shared_ptr<Conn> conn(uint64_t id);
 
   int gpu_;
