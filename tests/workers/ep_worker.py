@@ -87,6 +87,7 @@ def main():
 
     recv_x, counts = buf.dispatch(xg, topkg)
     torch.cuda.synchronize()
+    buf._b.check_error()
 
     # ---- reference: recompute all ranks' routing on CPU --------------------
     all_inputs = [rank_inputs(r, T, H, K, E, dtype, seed) for r in range(world)]
@@ -136,6 +137,7 @@ def main():
         rx, cnts = buf.dispatch(x2.cuda(), topk2.cuda())
         out2 = buf.combine(rx, topk2.cuda(), w2.cuda())
         torch.cuda.synchronize()
+        buf._b.check_error()
         assert out2.shape == (T, H)
         print(f"[rank {rank}] repeat {it} OK", flush=True)
 
@@ -151,6 +153,7 @@ def main():
     torch.cuda.synchronize()
     eager_rx, eager_counts = buf.dispatch(x3g, topk3g)
     torch.cuda.synchronize()
+    buf._b.check_error()
     assert torch.equal(counts3.cpu(), eager_counts.cpu())
     del dummy
     print(f"[rank {rank}] phase-split dispatch OK", flush=True)
@@ -161,6 +164,7 @@ def main():
     torch.cuda.synchronize()
     out3b = buf.combine(eo3, topk3g, w3.cuda())
     torch.cuda.synchronize()
+    buf._b.check_error()
     assert torch.equal(out3a.cpu(), out3b.cpu())
     print(f"[rank {rank}] phase-split combine OK", flush=True)
 
@@ -175,6 +179,7 @@ def main():
     c4b = buf.dispatch_recv().cpu()
     snap_b = buf.recv_x_view()[0, : 4].clone()
     torch.cuda.synchronize()
+    buf._b.check_error()
     assert torch.equal(c4a, c4b), "cached replay changed counts"
     if int(c4a[0, rank]) > 0:
         assert not torch.equal(snap_a, snap_b), \
@@ -323,6 +328,7 @@ def main():
                 assert c5c[le, src].item() == sel, (e, src)
         out5 = buf2.combine(rx5.clone(), topk5.cuda(), w5.cuda())
         torch.cuda.synchronize()
+        buf2._b.check_error()
         assert out5.shape == (T2, 256)
         buf2.close()
         print(f"[rank {rank}] 256-expert top-8 OK", flush=True)
@@ -340,6 +346,7 @@ def main():
                 b.atomic_add(dst, rank + 1)
         b.barrier()
         torch.cuda.synchronize()
+        b.check_error()
         got = b.read_sync_word(2)
         want = sum(r + 1 for r in range(world) if r != rank)
         assert got == want, (got, want)

@@ -566,6 +566,8 @@ PYBIND11_MODULE(_C, m) {
            [](uccl::ep::EpBuffer& b, int idx) {
              return b.read_sync_word(idx);
            })
+      .def("check_error",
+           [](uccl::ep::EpBuffer& b) { b.check_error(); })
       .def("combine",
            [](uccl::ep::EpBuffer& b, at::Tensor expert_out,
               at::Tensor topk_idx, at::Tensor topk_w) {

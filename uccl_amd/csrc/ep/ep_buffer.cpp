@@ -289,6 +289,20 @@ void EpBuffer::atomic_add(int dst, uint64_t value, hipStream_t stream) {
                        stream);
 }
 
+void EpBuffer::check_error() {
+  uint64_t const e = read_sync_word(3);
+  if (!e) return;
+  static char const* names[] = {"?",            "dispatch_wait",
+                                "combine_wait", "consume_gate",
+                                "sync_wait",    "nrm_wait",
+                                "nrm_ret_wait", "ring_drain"};
+  int const code = static_cast<int>(e >> 56);
+  UCCL_CHECK(false) << "EP device wait TIMED OUT: "
+                    << names[code >= 0 && code <= 7 ? code : 0]
+                    << " rank=" << ((e >> 48) & 0xff) << " aux="
+                    << ((e >> 32) & 0xffff) << " seq=" << (e & 0xffffffffull);
+}
+
 uint64_t EpBuffer::read_sync_word(int idx) {
   uint64_t v = 0;
   UCCL_CHECK_HIP(hipMemcpy(&v, sync_ptr(heap_, v_, idx), sizeof(v),

@@ -90,6 +90,9 @@ class EpBuffer {
   void quiet(hipStream_t stream);         // prior ring cmds fully shipped
   void atomic_add(int dst, uint64_t value, hipStream_t stream);
   uint64_t read_sync_word(int idx);       // test/diagnostic accessor
+  // raises (with the decoded wait id) if any device wait timed out and
+  // recorded a diagnostic instead of wedging/trapping the queue
+  void check_error();
 
   const EpView& view() const { return v_; }
   void* recv_x_ptr() const {

@@ -32,6 +32,30 @@ __device__ inline uint64_t tag_count(uint64_t seq, uint32_t count) {
   return (seq << 32) | count;
 }
 
+// Wait-timeout diagnostics: instead of trapping (which loses the device
+// printf and kills the queue with an opaque 0x1016), record WHICH wait
+// stalled into the heap's sync[3] word and return; the host checks the
+// word after its next sync and raises with the decoded context
+// (EpBuffer::check_error). err = code:8 | rank:8 | aux:16 | seq:32.
+enum EpWaitCode : uint64_t {
+  kErrDispatchWait = 1,
+  kErrCombineWait = 2,
+  kErrConsumeGate = 3,
+  kErrSyncWait = 4,
+  kErrNrmWait = 5,
+  kErrNrmRetWait = 6,
+  kErrRingDrain = 7,
+};
+
+__device__ inline void record_wait_err(const EpView& v, uint64_t code,
+                                       uint64_t aux) {
+  uint64_t const e = (code << 56) |
+                     (static_cast<uint64_t>(v.rank & 0xff) << 48) |
+                     ((aux & 0xffff) << 32) |
+                     (v.seq & 0xffffffffull);
+  st_release_sys(sync_ptr(v.peers[v.rank], v, 3), e);
+}
+
 // Quantize one hidden row (bf16) to fp8-e4m3 with per-128-element scales,
 // cooperatively with a 256-thread block (two 128-element groups in
 // flight; cross-wave amax via LDS). DeepEP LL fp8 semantics
@@ -137,10 +161,8 @@ __global__ void k_ep_consume_gate(EpView v, uint64_t prev_seq) {
     for (uint64_t it = 0;; ++it) {
       if (ld_acquire_sys(p) >= prev_seq) return;
       if (it > (1ull << 28)) {
-        printf("uccl_ep: consume gate TIMEOUT rank=%d peer=%d prev=%llu\n",
-               v.rank, static_cast<int>(threadIdx.x),
-               (unsigned long long)prev_seq);
-        __builtin_trap();
+        record_wait_err(v, kErrConsumeGate, threadIdx.x);
+        return;
       }
       backoff();
     }
@@ -161,9 +183,8 @@ __global__ void k_ep_ring_wait_empty(EpView v) {
         __HIP_MEMORY_SCOPE_SYSTEM);
     if (h == t) return;
     if (it > (1ull << 28)) {
-      printf("uccl_ep: ring drain TIMEOUT rank=%d head=%llu tail=%llu\n",
-             v.rank, (unsigned long long)h, (unsigned long long)t);
-      __builtin_trap();
+      record_wait_err(v, kErrRingDrain, static_cast<uint64_t>(t - h));
+      return;
     }
     backoff();
   }
@@ -314,11 +335,9 @@ __global__ void k_ep_dispatch_wait(EpView v,
       got = ld_acquire_sys(p);
       if ((got >> 32) == v.seq) break;
       if (it > (1ull << 28)) {
-        printf("uccl_ep: dispatch wait TIMEOUT rank=%d le=%d src=%d "
-               "seq=%llu got_tag=%llu\n",
-               v.rank, le, src, (unsigned long long)v.seq,
-               (unsigned long long)(got >> 32));
-        __builtin_trap();
+        record_wait_err(v, kErrDispatchWait,
+                        static_cast<uint64_t>(le) * v.world + src);
+        break;
       }
       backoff();
     }
@@ -353,6 +372,11 @@ __global__ void k_ep_combine_send(EpView v,
     uint32_t const meta = *disp_meta_ptr(me, v, le, slot);
     uint32_t const t = meta & kMetaTokMask;
     uint32_t const k = meta >> 24;
+    // defensive: corrupt meta (e.g. after an error-path early return)
+    // must never index outside the return region
+    if (t >= static_cast<uint32_t>(v.max_tokens) ||
+        k >= static_cast<uint32_t>(v.topk))
+      continue;
     char const* srcrow =
         static_cast<char const*>(expert_out) +
         ((static_cast<size_t>(le) * v.world * v.max_tokens + slot) *
@@ -377,10 +401,8 @@ __global__ void k_ep_combine_wait(EpView v) {
     for (uint64_t it = 0;; ++it) {
       if (ld_acquire_sys(p) >= v.seq) break;
       if (it > (1ull << 28)) {
-        printf("uccl_ep: combine wait TIMEOUT rank=%d src=%d seq=%llu\n",
-               v.rank, static_cast<int>(threadIdx.x),
-               (unsigned long long)v.seq);
-        __builtin_trap();
+        record_wait_err(v, kErrCombineWait, threadIdx.x);
+        break;
       }
       backoff();
     }
@@ -396,6 +418,9 @@ __global__ void k_ep_comb_scatter(EpView v, size_t row0, size_t count) {
     uint32_t const meta = *ingress_meta(me, v, row0 + i);
     uint32_t const t = meta & kMetaTokMask;
     uint32_t const k = meta >> 24;
+    if (t >= static_cast<uint32_t>(v.max_tokens) ||
+        k >= static_cast<uint32_t>(v.topk))
+      continue;
     block_copy(comb_x_ptr(me, v, t, k), ingress_row(me, v, row0 + i),
                row_bytes);
   }
@@ -539,10 +564,8 @@ __global__ void k_nrm_wait(EpView v, int* __restrict__ out_counts) {
       got = ld_acquire_sys(p);
       if ((got >> 32) == v.seq) break;
       if (it > (1ull << 28)) {
-        printf("uccl_ep: nrm wait TIMEOUT rank=%d src=%d seq=%llu\n",
-               v.rank, static_cast<int>(threadIdx.x),
-               (unsigned long long)v.seq);
-        __builtin_trap();
+        record_wait_err(v, kErrNrmWait, threadIdx.x);
+        break;
       }
       backoff();
     }
@@ -565,6 +588,7 @@ __global__ void k_nrm_return(EpView v, void const* __restrict__ x) {
   size_t const row_bytes = static_cast<size_t>(v.hidden) * v.elem_size;
   for (uint32_t i = b; i < count; i += fanout) {
     uint32_t const t = *nrm_meta_ptr(me, v, src, i);
+    if (t >= static_cast<uint32_t>(v.max_tokens)) continue;
     block_copy_nt(nrm_ret_ptr(sbase, v, t, v.rank),
                   static_cast<char const*>(x) +
                       (static_cast<size_t>(src) * v.max_tokens + i) *
@@ -585,9 +609,8 @@ __global__ void k_nrm_ret_wait(EpView v) {
     for (uint64_t it = 0;; ++it) {
       if (ld_acquire_sys(p) >= v.seq) return;
       if (it > (1ull << 28)) {
-        printf("uccl_ep: nrm ret wait TIMEOUT rank=%d src=%d\n", v.rank,
-               static_cast<int>(threadIdx.x));
-        __builtin_trap();
+        record_wait_err(v, kErrNrmRetWait, threadIdx.x);
+        return;
       }
       backoff();
     }
@@ -661,9 +684,8 @@ __global__ void k_ep_sync_wait(EpView v, int idx, uint64_t seq) {
     for (uint64_t it = 0;; ++it) {
       if (ld_acquire_sys(p) >= seq) return;
       if (it > (1ull << 28)) {
-        printf("uccl_ep: sync wait TIMEOUT rank=%d idx=%d seq=%llu\n",
-               v.rank, idx, (unsigned long long)seq);
-        __builtin_trap();
+        record_wait_err(v, kErrSyncWait, idx);
+        return;
       }
       backoff();
     }
