@@ -131,6 +131,56 @@ def main():
         assert torch.allclose(
             seg, torch.full_like(seg, float(src * 10 + rank))), (src, seg[0])
 
+    # --- op matrix: prod / min / max / avg (rccl-tests patterns) ------------
+    ncclProd, ncclMax, ncclMin, ncclAvg = 1, 2, 3, 4
+    prod_want = 1.0
+    for r in range(world):
+        prod_want *= (r + 1.0)
+    for op, wantv in ((ncclProd, prod_want), (ncclMin, 1.0),
+                      (ncclMax, float(world)),
+                      (ncclAvg, want / world)):
+        t = torch.full((8192,), float(rank + 1), device="cuda")
+        rc = lib.ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                               ncclFloat32, op, comm, stream)
+        torch.cuda.synchronize()
+        assert rc == 0 and torch.allclose(
+            t, torch.full_like(t, wantv)), (op, float(t[0]), wantv)
+        # reduce_scatter with the same op
+        rs_in = torch.full((256 * world,), float(rank + 1), device="cuda")
+        rs_out = torch.empty(256, device="cuda")
+        rc = lib.ncclReduceScatter(rs_in.data_ptr(), rs_out.data_ptr(),
+                                   256, ncclFloat32, op, comm, stream)
+        torch.cuda.synchronize()
+        assert rc == 0 and torch.allclose(
+            rs_out, torch.full_like(rs_out, wantv)), (op, float(rs_out[0]))
+
+    # --- grouped send/recv (the advisor-r1 deadlock pattern): paired
+    # >2MB sendrecv inside ncclGroupStart/End must complete ---------------
+    if world >= 2:
+        lib.ncclGroupStart.restype = ctypes.c_int
+        lib.ncclGroupEnd.restype = ctypes.c_int
+        lib.ncclSend.restype = ctypes.c_int
+        lib.ncclSend.argtypes = [ctypes.c_void_p, ctypes.c_size_t,
+                                 ctypes.c_int, ctypes.c_int,
+                                 ctypes.c_void_p, ctypes.c_void_p]
+        lib.ncclRecv.restype = ctypes.c_int
+        lib.ncclRecv.argtypes = [ctypes.c_void_p, ctypes.c_size_t,
+                                 ctypes.c_int, ctypes.c_int,
+                                 ctypes.c_void_p, ctypes.c_void_p]
+        peer = rank ^ 1
+        n = 3 << 20  # 12 MB of fp32: multiple 2MB slot credits deep
+        sbuf = torch.full((n,), float(rank + 3), device="cuda")
+        rbuf = torch.zeros(n, device="cuda")
+        assert lib.ncclGroupStart() == 0
+        assert lib.ncclSend(sbuf.data_ptr(), n, ncclFloat32, peer, comm,
+                            stream) == 0
+        assert lib.ncclRecv(rbuf.data_ptr(), n, ncclFloat32, peer, comm,
+                            stream) == 0
+        assert lib.ncclGroupEnd() == 0
+        torch.cuda.synchronize()
+        assert torch.allclose(rbuf, torch.full_like(rbuf, float(peer + 3)))
+        print(f"[rank {rank}] grouped sendrecv OK", flush=True)
+
     lib.ncclCommDestroy(comm)
     print(f"[rank {rank}] NCCL SHIM OK", flush=True)
 
