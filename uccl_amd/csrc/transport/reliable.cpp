@@ -426,7 +426,11 @@ struct TransportEndpoint::Impl {
     // every field below is wire-controlled: bound it (subtraction form —
     // addition could wrap) before it touches memory. A legit peer never
     // violates these; a corrupt/stray datagram gets dropped unacked.
-    if (h.len > chunk_bytes) return;
+    // NB: bound against the PROTOCOL max, not our local chunk_bytes —
+    // the peer picks its own chunk size and endpoints may differ
+    // (asymmetric pairs deadlocked on this: dropped-unacked chunks ->
+    // sender RTO abort; the fabric already verified the payload length)
+    if (h.len > 65536) return;
     if (h.msg_bytes > max_msg_bytes) return;
     ++st.data_recv;
     f.last_data_ts = h.ts_ns;
