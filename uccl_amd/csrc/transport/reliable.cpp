@@ -493,8 +493,17 @@ struct TransportEndpoint::Impl {
                                            static_cast<uint64_t>(f.srtt_us));
         if (c.send_ts && now - c.send_ts > rto) {
           if (++c.rto_count >= abort_thres) {
-            UCCL_LOG_ERROR << "flow " << fid << " csn " << csn
-                           << " exceeded RTO abort threshold; marking dead";
+            UCCL_LOG_ERROR
+                << "flow " << fid << " csn " << csn
+                << " exceeded RTO abort threshold; marking dead"
+                << " [attempts=" << c.attempts << " len=" << c.len
+                << " inflight=" << f.inflight.size() << " cwnd=" << f.cwnd
+                << " ds=" << st.data_sent.load()
+                << " dr=" << st.data_recv.load()
+                << " as=" << st.acks_sent.load()
+                << " ar=" << st.acks_recv.load()
+                << " rtx=" << st.retransmits.load() << "+"
+                << st.rto_retransmits.load() << "]";
             f.failed = true;
             f.fcv.notify_all();
             break;
