@@ -29,7 +29,8 @@ def _inputs(world, elems, seed=0):
 @pytest.mark.parametrize("world", [2, 4])
 def test_sendrecv_spray_device(world):
     elems = 1 << 16
-    g = uk.uk_plan_sendrecv_spray(world, 0, 1, elems * 4, 16384)
+    topo = uk.UkTopology(world)
+    g = uk.uk_plan_sendrecv(topo, 0, 1, elems * 4, 16384)
     g = uk.uk_lower(g)
     ins = _inputs(world, elems, seed=11)
     host_outs, _ = uk.uk_execute_host(g, ins, elems * 4)
@@ -42,7 +43,7 @@ def test_sendrecv_spray_device(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_allreduce_auto_device(world):
     elems = 1 << 15
-    g = uk.uk_plan_allreduce_auto(world, elems * 4, 4, 16384)
+    g = uk.uk_plan_allreduce_auto(uk.UkTopology(world), elems * 4, 4, 16384)
     ins = _inputs(world, elems, seed=13)
     host_outs, _ = uk.uk_execute_host(g, ins, elems * 4)
     dev_outs, _ = uk.uk_execute_device(g, ins, elems * 4)
@@ -52,7 +53,7 @@ def test_allreduce_auto_device(world):
 
 def test_allgather_device():
     world, elems = 4, 4096
-    g = uk.uk_plan_allgather(world, elems * 4, 8192)
+    g = uk.uk_plan_allgather(uk.UkTopology(world), elems * 4, 8192)
     g = uk.uk_lower(g)
     ins = _inputs(world, elems, seed=15)
     host_outs, _ = uk.uk_execute_host(g, ins, world * elems * 4)
