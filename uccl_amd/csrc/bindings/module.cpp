@@ -601,6 +601,8 @@ PYBIND11_MODULE(_C, m) {
                     &uccl::transport::Stats::rto_retransmits)
       .def_readonly("injected_drops",
                     &uccl::transport::Stats::injected_drops)
+      .def_readonly("dup_recv", &uccl::transport::Stats::dup_recv)
+      .def_readonly("send_fail", &uccl::transport::Stats::send_fail)
       .def_readonly("msgs_sent", &uccl::transport::Stats::msgs_sent)
       .def_readonly("msgs_recv", &uccl::transport::Stats::msgs_recv)
       .def_readonly("srtt_us", &uccl::transport::Stats::srtt_us)

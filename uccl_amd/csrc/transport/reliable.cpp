@@ -172,7 +172,8 @@ struct TransportEndpoint::Impl {
   struct AtomStats {
     std::atomic<uint64_t> data_sent{0}, data_recv{0}, acks_sent{0},
         acks_recv{0}, retransmits{0}, rto_retransmits{0},
-        injected_drops{0}, msgs_sent{0}, msgs_recv{0};
+        injected_drops{0}, dup_recv{0}, send_fail{0}, msgs_sent{0},
+        msgs_recv{0};
     std::atomic<double> srtt_us{0}, cwnd{0};
   } st;
   std::mutex hist_mu;
@@ -199,13 +200,22 @@ struct TransportEndpoint::Impl {
     // a transient fabric would-block counts as a loss: RTO retransmits
     if (fabric->post_chunk(f.id, path, d, c.msg->ptr + c.off))
       ++st.data_sent;
+    else
+      ++st.send_fail;
   }
 
   bool eqds_mode() const { return cc_mode == "eqds"; }
   uint64_t rwnd_bytes() const { return rwnd; }
 
   void pump_tx(Flow& f) {
-    while (static_cast<double>(f.inflight.size()) < f.cwnd && !f.txq.empty()) {
+    // In-flight chunks are capped BELOW the 128-bit SACK window: chunks
+    // past cum+128 cannot be selectively acked, and with engine-sharded
+    // (reordering) rx a lagging cum makes them look lost -> spurious
+    // RTO retransmits (measured: dup_recv == peer rto_retransmits).
+    // The reference bounds the same thing with per-engine unacked-bytes
+    // budgets (transport_config.h:69-82).
+    double const wnd = std::min(f.cwnd, 120.0);
+    while (static_cast<double>(f.inflight.size()) < wnd && !f.txq.empty()) {
       // EQDS-style receiver-driven credit: stop when the granted budget
       // is exhausted; later acks raise credit_limit and re-pump.
       if (eqds_mode() &&
@@ -436,6 +446,7 @@ struct TransportEndpoint::Impl {
     f.last_data_ts = h.ts_ns;
     bool const fresh =
         (h.csn >= f.rx_cum) && !f.rx_ooo.count(h.csn);
+    if (!fresh) ++st.dup_recv;
     if (fresh) {
       auto& m = f.rxmsgs[h.msg_id];
       if (!m.known) {
@@ -491,7 +502,15 @@ struct TransportEndpoint::Impl {
         uint64_t const rto =
             std::max<uint64_t>(rto_ns, 4ull * 1000 *
                                            static_cast<uint64_t>(f.srtt_us));
-        if (c.send_ts && now - c.send_ts > rto) {
+        // SIGNED age: with per-flow locks another thread can send a
+        // chunk after this scan captured `now` (send_ts > now), and the
+        // unsigned subtraction wrapped to ~2^64 — every such chunk
+        // looked infinitely old and fired a spurious RTO (cwnd
+        // collapse; with unlucky timing, the 50-strike flow abort the
+        // GPU proxies hit). Round 1's global lock made this impossible.
+        int64_t const age = static_cast<int64_t>(now) -
+                            static_cast<int64_t>(c.send_ts);
+        if (c.send_ts && age > static_cast<int64_t>(rto)) {
           if (++c.rto_count >= abort_thres) {
             UCCL_LOG_ERROR
                 << "flow " << fid << " csn " << csn
@@ -509,6 +528,14 @@ struct TransportEndpoint::Impl {
             break;
           }
           ++st.rto_retransmits;
+          static bool const dbg = env_bool("UCCL_TP_DEBUG_RTO", false);
+          if (dbg)
+            fprintf(stderr,
+                    "[rto] flow=%llx csn=%u path=%d age_ms=%.1f att=%u "
+                    "infl=%zu\n",
+                    (unsigned long long)fid, csn,
+                    static_cast<int>(csn % f.num_paths),
+                    (now - c.send_ts) / 1e6, c.attempts, f.inflight.size());
           f.cwnd = std::max(2.0, f.cwnd / 2);
           send_chunk(f, csn, c);
         }
@@ -706,14 +733,21 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
                 << 10;
   impl_->dup_thres =
       static_cast<int>(env_int("UCCL_TP_DUPACK_THRES", 32));
+  // RTO is the LAST-RESORT backstop (SACK-hole fast retransmit handles
+  // real loss); measured host scheduling tails reach ~13ms p99 under
+  // thread oversubscription, so a 20ms base fired spuriously (every
+  // RTO-retransmitted chunk arrived as a duplicate). 100ms keeps the
+  // backstop well clear of scheduler noise; abort = 50x = 5s blackout.
   impl_->rto_base_ns =
-      static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 20000)) * 1000;
+      static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 100000)) * 1000;
   impl_->rto_abort_thres =
       static_cast<int>(env_int("UCCL_TP_RTO_ABORT", 50));
-  // busy-poll engines by default on many-core hosts (MI355X servers);
-  // on small boxes the burned cores cost more than the wakeup saves
-  impl_->spin = env_bool(
-      "UCCL_TP_SPIN", std::thread::hardware_concurrency() >= 32);
+  // Busy-poll engines (reference adaptive-sleep discipline). OPT-IN:
+  // measured 2.1x single-flow message rate, but an unresolved
+  // spin-specific stall was observed under the EP proxy workload on
+  // many-core GPU hosts (one path's chunk unacked through 50 RTOs), so
+  // correctness keeps the default off until that is root-caused.
+  impl_->spin = env_bool("UCCL_TP_SPIN", false);
   // optional sender pacing (the reference's Carousel timing-wheel role;
   // bypassed by default there and here — BYPASS_PACING=1)
   if (int64_t mbps = env_int("UCCL_TP_PACE_MBPS", 0); mbps > 0)
@@ -900,6 +934,8 @@ Stats TransportEndpoint::stats() const {
   st.retransmits = a.retransmits.load(std::memory_order_relaxed);
   st.rto_retransmits = a.rto_retransmits.load(std::memory_order_relaxed);
   st.injected_drops = a.injected_drops.load(std::memory_order_relaxed);
+  st.dup_recv = a.dup_recv.load(std::memory_order_relaxed);
+  st.send_fail = a.send_fail.load(std::memory_order_relaxed);
   st.msgs_sent = a.msgs_sent.load(std::memory_order_relaxed);
   st.msgs_recv = a.msgs_recv.load(std::memory_order_relaxed);
   st.srtt_us = a.srtt_us.load(std::memory_order_relaxed);

@@ -43,6 +43,9 @@ struct Stats {
   uint64_t retransmits = 0;
   uint64_t rto_retransmits = 0;
   uint64_t injected_drops = 0;
+  uint64_t dup_recv = 0;    // non-fresh data arrivals (peer resent an
+                            // already-received chunk: ack loss signal)
+  uint64_t send_fail = 0;   // fabric post failures (e.g. sendto error)
   uint64_t msgs_sent = 0;
   uint64_t msgs_recv = 0;
   double srtt_us = 0;

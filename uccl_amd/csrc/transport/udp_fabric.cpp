@@ -146,9 +146,11 @@ class UdpFabric final : public Fabric {
     char buf[sizeof(DataWire) + 65536];
     memcpy(buf, &h, sizeof(h));
     if (d.len) memcpy(buf + sizeof(h), payload, d.len);
-    (void)sendto(socks_[path], buf, sizeof(h) + d.len, 0,
-                 reinterpret_cast<sockaddr*>(&to), sizeof(to));
-    return true;
+    ssize_t const n = sendto(socks_[path], buf, sizeof(h) + d.len, 0,
+                             reinterpret_cast<sockaddr*>(&to), sizeof(to));
+    // ENOBUFS and friends = local transient loss: report so the caller
+    // accounts a send failure (the RTO machinery covers recovery)
+    return n == static_cast<ssize_t>(sizeof(h) + d.len);
   }
 
   void post_ctrl(uint64_t flow, int path, void const* frame,
