@@ -19,7 +19,8 @@ for san in thread address,undefined; do
       uccl_amd/csrc/core/trace.cpp -o "$OUT/san_cu_$tag" -pthread -lz
   g++ -O1 -g -std=c++17 -fsanitize=$san -D__HIP_PLATFORM_AMD__=1 \
       -I/opt/rocm/include tools/san_p2p.cpp \
-      uccl_amd/csrc/p2p/endpoint.cpp uccl_amd/csrc/transport/reliable.cpp \
+      uccl_amd/csrc/p2p/endpoint.cpp uccl_amd/csrc/p2p/rccl_plane.cpp \
+      uccl_amd/csrc/transport/reliable.cpp \
       uccl_amd/csrc/transport/udp_fabric.cpp \
       uccl_amd/csrc/transport/verbs_fabric.cpp \
       uccl_amd/csrc/core/trace.cpp -o "$OUT/san_p2p_$tag" -pthread -ldl \
