@@ -72,6 +72,7 @@ def main():
     comm = ctypes.c_void_p()
     rc = lib.ncclCommInitRank(ctypes.byref(comm), world, uid, rank)
     assert rc == 0, f"CommInitRank rc={rc}"
+    print(f"[rank {rank}] init OK", flush=True)
 
     stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
     want = sum(r + 1.0 for r in range(world))
@@ -82,6 +83,7 @@ def main():
                            ncclFloat32, ncclSum, comm, stream)
     torch.cuda.synchronize()
     assert rc == 0 and torch.allclose(t, torch.full_like(t, want)), t[:3]
+    print(f"[rank {rank}] allreduce OK", flush=True)
 
     # out-of-place allreduce
     src = torch.full((4096,), float(rank + 1), device="cuda")
@@ -91,12 +93,16 @@ def main():
     torch.cuda.synchronize()
     assert rc == 0 and torch.allclose(dst, torch.full_like(dst, want))
 
+    print(f"[rank {rank}] oop allreduce OK", flush=True)
+
     # broadcast
     b = torch.full((1024,), float(rank * 5), device="cuda")
     rc = lib.ncclBroadcast(b.data_ptr(), b.data_ptr(), 1024, ncclFloat32,
                            0, comm, stream)
     torch.cuda.synchronize()
     assert rc == 0 and torch.allclose(b, torch.zeros_like(b))
+
+    print(f"[rank {rank}] broadcast OK", flush=True)
 
     # allgather
     ag_in = torch.full((256,), float(rank), device="cuda")
@@ -109,6 +115,8 @@ def main():
         assert torch.allclose(ag_out[r * 256:(r + 1) * 256],
                               torch.full((256,), float(r), device="cuda"))
 
+    print(f"[rank {rank}] allgather OK", flush=True)
+
     # reduce_scatter
     rs_in = torch.full((512 * world,), float(rank + 1), device="cuda")
     rs_out = torch.empty(512, device="cuda")
@@ -116,6 +124,8 @@ def main():
                                ncclFloat32, ncclSum, comm, stream)
     torch.cuda.synchronize()
     assert rc == 0 and torch.allclose(rs_out, torch.full_like(rs_out, want))
+
+    print(f"[rank {rank}] reduce_scatter OK", flush=True)
 
     # all_to_all (RCCL extension): segment j of rank i's input lands as
     # segment i of rank j's output
@@ -130,6 +140,8 @@ def main():
         seg = a2a_out[src * 128:(src + 1) * 128]
         assert torch.allclose(
             seg, torch.full_like(seg, float(src * 10 + rank))), (src, seg[0])
+
+    print(f"[rank {rank}] alltoall OK", flush=True)
 
     # --- op matrix: prod / min / max / avg (rccl-tests patterns) ------------
     ncclProd, ncclMax, ncclMin, ncclAvg = 1, 2, 3, 4
@@ -153,6 +165,8 @@ def main():
         torch.cuda.synchronize()
         assert rc == 0 and torch.allclose(
             rs_out, torch.full_like(rs_out, wantv)), (op, float(rs_out[0]))
+
+    print(f"[rank {rank}] op matrix OK", flush=True)
 
     # --- grouped send/recv (the advisor-r1 deadlock pattern): paired
     # >2MB sendrecv inside ncclGroupStart/End must complete ---------------

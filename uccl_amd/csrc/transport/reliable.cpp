@@ -190,9 +190,17 @@ struct TransportEndpoint::Impl {
 
   void send_chunk(Flow& f, uint32_t csn, ChunkTx& c) {
     ChunkDesc d{f.id, c.msg->id, c.msg->bytes, c.off, c.len, csn, now_ns()};
-    int const path = csn % f.num_paths;  // spray round-robin across paths
     c.send_ts = now_ns();
     ++c.attempts;
+    // First attempt sprays round-robin (csn % paths); every RETRANSMIT
+    // rotates to the next path. A single black-holed path (socket not
+    // drained, one-direction loopback drop, broken QP) must never strand
+    // a chunk: pinned retransmits turned exactly that into a 50-strike
+    // flow abort on the GPU box (csn stuck for 5s while every other csn
+    // delivered). The reference migrates timed-out chunks off their path
+    // the same way (path selection on retransmission).
+    int const path =
+        static_cast<int>((csn + (c.attempts - 1)) % f.num_paths);
     if (inject_drop(csn, c.attempts, loss_pct)) {
       ++st.injected_drops;
       return;  // "sent" into the void
@@ -511,6 +519,16 @@ struct TransportEndpoint::Impl {
         int64_t const age = static_cast<int64_t>(now) -
                             static_cast<int64_t>(c.send_ts);
         if (c.send_ts && age > static_cast<int64_t>(rto)) {
+          if (c.rto_count == abort_thres / 2)
+            UCCL_LOG_WARN << "flow " << fid << " csn " << csn
+                          << " still unacked after " << c.rto_count
+                          << " RTOs [len=" << c.len
+                          << " attempts=" << c.attempts << " next path="
+                          << (csn + c.attempts) % f.num_paths << "/"
+                          << f.num_paths << " cum_dup_recv="
+                          << st.dup_recv.load() << " send_fail="
+                          << st.send_fail.load() << " inj="
+                          << st.injected_drops.load() << "]";
           if (++c.rto_count >= abort_thres) {
             UCCL_LOG_ERROR
                 << "flow " << fid << " csn " << csn
@@ -522,7 +540,9 @@ struct TransportEndpoint::Impl {
                 << " as=" << st.acks_sent.load()
                 << " ar=" << st.acks_recv.load()
                 << " rtx=" << st.retransmits.load() << "+"
-                << st.rto_retransmits.load() << "]";
+                << st.rto_retransmits.load() << " sf="
+                << st.send_fail.load() << " inj="
+                << st.injected_drops.load() << "]";
             f.failed = true;
             f.fcv.notify_all();
             break;
