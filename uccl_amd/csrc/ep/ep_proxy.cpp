@@ -242,6 +242,7 @@ void EpProxy::ring_loop() {
       TransferCmd c;
       memcpy(&c, const_cast<TransferCmd*>(&ring_->cmds[head % kRingSlots]),
              sizeof(c));
+      auto const t0 = std::chrono::steady_clock::now();
       try {
         switch (static_cast<CmdOp>(c.op)) {
           case CmdOp::kDispatchWrite: {
@@ -291,6 +292,16 @@ void EpProxy::ring_loop() {
         if (!stop_)
           UCCL_LOG_ERROR << "ep proxy ring_loop died: " << e.what();
         return;
+      }
+      {
+        // slow-command diagnostic: ship_rows should complete in ms; a
+        // multi-second command pins down WHICH ring entry wedged when a
+        // peer later times out waiting for the generation's tail
+        auto const t1 = std::chrono::steady_clock::now();
+        double const s = std::chrono::duration<double>(t1 - t0).count();
+        if (s > 3.0)
+          UCCL_LOG_WARN << "ep proxy ring cmd slow: op " << c.op << " seq "
+                        << c.seq32 << " a " << c.a << " took " << s << "s";
       }
       // publish head ONLY after the command is fully shipped: the GPU
       // gates the next dispatch's egress reuse on head==tail

@@ -18,6 +18,7 @@
 
 #include <rccl/rccl.h>
 
+#include <algorithm>
 #include <atomic>
 #include <cmath>
 #include <condition_variable>
@@ -125,27 +126,7 @@ struct UcclComm {
   std::unique_ptr<Communicator> comm;
   int device = 0;
   std::mutex mu;
-  // per-(peer, dir) streams for grouped p2p: ops to the same peer+direction
-  // stay ordered, ops to different peers run concurrently so the slot-ack
-  // pacing of a send can't block the recv that would unblock it.
-  std::map<std::pair<int, int>, hipStream_t> p2p_streams;
   std::vector<PreMulScalar> premul;  // ncclRedOpCreatePreMulSum handles
-  hipStream_t p2p_stream(int peer, bool is_send) {
-    std::lock_guard<std::mutex> g(mu);
-    auto key = std::make_pair(peer, is_send ? 1 : 0);
-    auto it = p2p_streams.find(key);
-    if (it != p2p_streams.end()) return it->second;
-    hipStream_t s = nullptr;
-    (void)hipStreamCreateWithFlags(&s, hipStreamNonBlocking);
-    p2p_streams[key] = s;
-    return s;
-  }
-  ~UcclComm() {
-    for (auto& [k, s] : p2p_streams) {
-      (void)hipStreamSynchronize(s);
-      (void)hipStreamDestroy(s);
-    }
-  }
 };
 
 // --- group semantics -------------------------------------------------------
@@ -192,29 +173,36 @@ ncclResult_t flush_group() {
       (void)hipEventDestroy(e);
     }
   }
-  for (auto& o : ops) {
-    if (o.peer == o.uc->comm->rank()) continue;
-    hipStream_t ps = o.uc->p2p_stream(o.peer, o.is_send);
-    hipEvent_t in, out;
-    (void)hipEventCreateWithFlags(&in, hipEventDisableTiming);
-    (void)hipEventCreateWithFlags(&out, hipEventDisableTiming);
-    (void)hipEventRecord(in, o.stream);
-    (void)hipStreamWaitEvent(ps, in, 0);
+  // Peer ops are re-issued on the USER stream in a canonical order both
+  // sides agree on: sort by peer rank, and within a pairing the lower
+  // rank enqueues its sends before its recvs while the higher rank
+  // enqueues recvs first. Each matched send/recv chain then completes
+  // pair-by-pair on a single stream — no credit-window deadlock, and no
+  // extra streams (per-(peer,dir) streams deadlocked on ROCm when many
+  // spin-wait kernels shared the ~4 HW queues of one device).
+  std::vector<PendingP2P> peer_ops;
+  for (auto& o : ops)
+    if (o.peer != o.uc->comm->rank()) peer_ops.push_back(o);
+  std::stable_sort(peer_ops.begin(), peer_ops.end(),
+                   [](PendingP2P const& a, PendingP2P const& b) {
+                     if (a.peer != b.peer) return a.peer < b.peer;
+                     int const ra = a.uc->comm->rank();
+                     int const pa =
+                         (a.is_send == (ra < a.peer)) ? 0 : 1;
+                     int const pb =
+                         (b.is_send == (ra < b.peer)) ? 0 : 1;
+                     return pa < pb;
+                   });
+  for (auto& o : peer_ops) {
     try {
       if (o.is_send)
-        o.uc->comm->send(o.buff, o.bytes, o.peer, ps);
+        o.uc->comm->send(o.buff, o.bytes, o.peer, o.stream);
       else
-        o.uc->comm->recv(o.buff, o.bytes, o.peer, ps);
+        o.uc->comm->recv(o.buff, o.bytes, o.peer, o.stream);
     } catch (std::exception const& e) {
       UCCL_LOG_ERROR << "grouped p2p failed: " << e.what();
-      (void)hipEventDestroy(in);
-      (void)hipEventDestroy(out);
       return ncclInternalError;
     }
-    (void)hipEventRecord(out, ps);
-    (void)hipStreamWaitEvent(o.stream, out, 0);
-    (void)hipEventDestroy(in);
-    (void)hipEventDestroy(out);
   }
   return ncclSuccess;
 }

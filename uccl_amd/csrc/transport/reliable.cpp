@@ -4,6 +4,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <chrono>
 #include <random>
 #include <shared_mutex>
 #include <stdexcept>
@@ -905,7 +906,18 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
       lk.lock();
     }
   }
-  f.fcv.wait(lk, [&] { return m->done || impl_->stop || f.failed; });
+  while (!f.fcv.wait_for(lk, std::chrono::seconds(5), [&] {
+    return m->done || impl_->stop || f.failed;
+  })) {
+    // stall diagnostics: a healthy send completes in ms; log enough flow
+    // state to tell a transport stall (unacked inflight) from a lost
+    // wakeup (acked_bytes==bytes but done never observed)
+    UCCL_LOG_WARN << "send_msg stalled 5s: flow " << flow << " msg "
+                  << m->id << " bytes " << m->bytes << " acked "
+                  << m->acked_bytes << " done " << m->done << " inflight "
+                  << f.inflight.size() << " txq " << f.txq.size()
+                  << " cwnd " << f.cwnd << " next_csn " << f.next_csn;
+  }
   if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (!m->done) throw std::runtime_error("transport closed during send");
 }
@@ -933,12 +945,22 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
   // outside the lock: fabric does its own locking (lock-order safety)
   impl_->fabric->post_recv_window(flow, msg_id, ptr, bytes);
   std::unique_lock<std::mutex> lk(f.fmu);
-  f.fcv.wait(lk, [&] {
+  auto done_pred = [&] {
     auto it = f.rxmsgs.find(msg_id);
     return (it != f.rxmsgs.end() && it->second.known &&
             it->second.recv_bytes >= it->second.bytes) ||
            impl_->stop || f.failed;
-  });
+  };
+  while (!f.fcv.wait_for(lk, std::chrono::seconds(5), done_pred)) {
+    auto it = f.rxmsgs.find(msg_id);
+    UCCL_LOG_WARN << "recv_msg stalled 5s: flow " << flow << " msg "
+                  << msg_id << " cap " << bytes << " known "
+                  << (it != f.rxmsgs.end() && it->second.known) << " got "
+                  << (it != f.rxmsgs.end() ? it->second.recv_bytes : 0)
+                  << "/"
+                  << (it != f.rxmsgs.end() ? it->second.bytes : 0)
+                  << " rx_cum " << f.rx_cum << " ooo " << f.rx_ooo.size();
+  }
   if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (impl_->stop) throw std::runtime_error("transport closed during recv");
   f.rxmsgs.erase(msg_id);
