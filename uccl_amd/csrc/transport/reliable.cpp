@@ -606,6 +606,19 @@ struct TransportEndpoint::Impl {
       if (spin && tnow - last_timers < 1000000ull) continue;
       last_timers = tnow;
       rto_scan();
+      // safety re-pump: pump_tx normally runs on enqueue and on acks, but
+      // a flow with an empty inflight set receives no acks — if its
+      // enqueue-time pump sent nothing (pacing gate, rendezvous window
+      // not yet advertised), nothing would ever retry. The 1ms timer
+      // closes that hole.
+      {
+        std::shared_lock<std::shared_mutex> mg(maps_mu);
+        for (auto& [id, fp] : flows) {
+          if (!fp) continue;
+          std::lock_guard<std::mutex> fg(fp->fmu);
+          if (!fp->failed && !fp->txq.empty()) pump_tx(*fp);
+        }
+      }
       // paced-EQDS credit refresh: a credit-stalled sender emits no
       // data, so acks (which carry grants) would never flow again —
       // the receiver must top up pulls from the progress loop
