@@ -57,6 +57,42 @@ int EpProxy::flow_peer(uint64_t flow) const {
   return 0;
 }
 
+// A flow mutex held for seconds means another lane is wedged mid-ship;
+// log instead of blocking invisibly (a silent block here produced
+// "last expert never shipped" stalls that were unattributable).
+// hipStreamSynchronize that logs instead of blocking invisibly: a lane
+// sync stuck for seconds means the copy engine / device is wedged.
+static void lane_sync(hipStream_t stream, char const* who) {
+  auto const t0 = std::chrono::steady_clock::now();
+  for (;;) {
+    hipError_t const e = hipStreamQuery(stream);
+    if (e == hipSuccess) return;
+    if (e != hipErrorNotReady) {
+      UCCL_CHECK_HIP(e);
+      return;
+    }
+    auto const el = std::chrono::duration<double>(
+                        std::chrono::steady_clock::now() - t0)
+                        .count();
+    if (el > 5.0) {
+      UCCL_LOG_WARN << "ep proxy lane sync stalled " << el << "s (" << who
+                    << ")";
+      (void)hipStreamSynchronize(stream);
+      return;
+    }
+    usleep(50);
+  }
+}
+
+struct TimedFlowLock {
+  std::timed_mutex& m;
+  TimedFlowLock(std::timed_mutex& mu, char const* who) : m(mu) {
+    while (!m.try_lock_for(std::chrono::seconds(5)))
+      UCCL_LOG_WARN << "ep proxy flow mutex contended 5s (" << who << ")";
+  }
+  ~TimedFlowLock() { m.unlock(); }
+};
+
 EpProxy::Lane::Lane(int device, size_t bytes) {
   UCCL_CHECK_HIP(hipSetDevice(device));
   UCCL_CHECK_HIP(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
@@ -78,7 +114,7 @@ EpProxy::EpProxy(const EpView& view, void* heap, D2HRing* ring_host,
       static_cast<size_t>(env_int("UCCL_EP_PROXY_CHUNK", 16384)));
   flows_.resize(v_.world, 0);
   for (int r = 0; r < v_.world; ++r)
-    flow_mu_.emplace_back(new std::mutex());
+    flow_mu_.emplace_back(new std::timed_mutex());
   stage_bytes_ = kStageBytes;
 }
 
@@ -153,7 +189,7 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                         void const* dev_rows,
                         uint32_t const* dev_metas_or_null,
                         std::vector<uint32_t> const* host_metas) {
-  std::lock_guard<std::mutex> guard(*flow_mu_[flow_peer(flow)]);
+  TimedFlowLock guard(*flow_mu_[flow_peer(flow)], "ship_rows");
   size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
   size_t const total = h.count * row_bytes;
   tp_->send_msg(flow, &h, sizeof(h));
@@ -167,7 +203,7 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
     size_t const n0 = std::min(stage_bytes_, total);
     UCCL_CHECK_HIP(hipMemcpyAsync(bufs[cur], dev_rows, n0,
                                   hipMemcpyDeviceToHost, lane.stream));
-    UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+    lane_sync(lane.stream, "ship first d2h");
   }
   while (off < total) {
     size_t const n = std::min(stage_bytes_, total - off);
@@ -179,8 +215,7 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
           hipMemcpyDeviceToHost, lane.stream));
     }
     tp_->send_msg(flow, bufs[cur], n);
-    if (next_off < total)
-      UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+    if (next_off < total) lane_sync(lane.stream, "ship next d2h");
     off = next_off;
     cur ^= 1;
   }
@@ -192,7 +227,7 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
     UCCL_CHECK_HIP(hipMemcpyAsync(metas.data(), dev_metas_or_null,
                                   h.count * sizeof(uint32_t),
                                   hipMemcpyDeviceToHost, lane.stream));
-    UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+    lane_sync(lane.stream, "ship metas d2h");
   }
   if (h.count) tp_->send_msg(flow, metas.data(), h.count * sizeof(uint32_t));
 }
@@ -230,14 +265,26 @@ void EpProxy::ring_loop() {
   pin_proxy_thread(0);
   Lane lane(device_, stage_bytes_);
   uint64_t head = 0;
+  uint64_t idle_us = 0;
   while (!stop_) {
     uint64_t tail =
         __atomic_load_n(const_cast<uint64_t*>(&ring_->tail),
                         __ATOMIC_ACQUIRE);
     if (head == tail) {
       usleep(20);
+      // idle diagnostic: if a peer later reports missing tail entries,
+      // "ring idle at H" here proves the GPU publish never pushed them
+      // (vs the loop being wedged processing one)
+      idle_us += 20;
+      if (idle_us >= 5'000'000) {
+        idle_us = 0;
+        static bool const dbg = env_bool("UCCL_TP_DEBUG_RTO", false);
+        if (dbg)
+          UCCL_LOG_WARN << "ep proxy ring idle: head=tail=" << head;
+      }
       continue;
     }
+    idle_us = 0;
     while (head != tail) {
       TransferCmd c;
       memcpy(&c, const_cast<TransferCmd*>(&ring_->cmds[head % kRingSlots]),
@@ -260,7 +307,7 @@ void EpProxy::ring_loop() {
             int const dst = static_cast<int>(c.a);
             WireHdr h{kAtom, c.seq32, 0, static_cast<uint32_t>(v_.rank),
                       c.c, c.b};
-            std::lock_guard<std::mutex> guard(*flow_mu_[dst]);
+            TimedFlowLock guard(*flow_mu_[dst], "atomic");
             tp_->send_msg(flows_[dst], &h, sizeof(h));
             break;
           }
@@ -271,7 +318,7 @@ void EpProxy::ring_loop() {
                       static_cast<uint32_t>(v_.rank), seq, 0};
             for (int r = 0; r < v_.world; ++r) {
               if (!((proxy_mask_ >> r) & 1u)) continue;
-              std::lock_guard<std::mutex> guard(*flow_mu_[r]);
+              TimedFlowLock guard(*flow_mu_[r], "barrier");
               tp_->send_msg(flows_[r], &h, sizeof(h));
             }
             handle_barrier_arrival(lane, seq);
@@ -353,7 +400,7 @@ void EpProxy::comb_tx_loop() {
         WireHdr done{kCombDone, static_cast<uint32_t>(task.seq), 0,
                      static_cast<uint32_t>(v_.rank), 0};
         {
-          std::lock_guard<std::mutex> guard(*flow_mu_[src]);
+          TimedFlowLock guard(*flow_mu_[src], "comb_done");
           tp_->send_msg(flows_[src], &done, sizeof(done));
         }
       }

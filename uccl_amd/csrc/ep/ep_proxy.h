@@ -92,7 +92,7 @@ class EpProxy {
   std::vector<uint64_t> flows_;
   // serializes multi-message sequences (hdr, payload..., metas) per flow:
   // the ring thread and the combine-tx thread share peer flows
-  std::vector<std::unique_ptr<std::mutex>> flow_mu_;
+  std::vector<std::unique_ptr<std::timed_mutex>> flow_mu_;
 
   size_t stage_bytes_ = 0;
 
