@@ -23,6 +23,8 @@ enum class CmdOp : uint32_t {
   kAtomicAdd = 2,  // a = dst rank, b = heap byte offset, c = add value
   kBarrier = 3,    // a = seq: CPU-side barrier across proxy peers
   kQuiet = 4,      // a = seq: all prior ring cmds fully shipped
+  kConsume = 5,    // a = seq: tell proxied peers this generation's counts
+                   // were consumed (their next dispatch may overwrite)
 };
 
 struct TransferCmd {

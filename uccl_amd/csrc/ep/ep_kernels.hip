@@ -152,11 +152,22 @@ __global__ void k_ep_consume_signal(EpView v, uint64_t seq) {
   if (threadIdx.x < static_cast<unsigned>(v.world) &&
       !((v.proxy_mask >> threadIdx.x) & 1u))
     st_release_sys(consumed_ptr(v.peers[threadIdx.x], v, v.rank), seq);
+  // proxied peers have no IPC mapping: route the signal through the CPU
+  // proxy (ring cmd -> tiny wire message -> peer rx writes our consumed
+  // word). Without this, a fast peer's generation N+1 counts overwrote
+  // generation N tags before this rank's wait kernel sampled them —
+  // observed as "last-shipped expert slot never matches its seq tag"
+  // under UCCL_EP_FORCE_PROXY with back-to-back dispatches.
+  __syncthreads();
+  if (threadIdx.x == 0 && v.proxy_mask && v.ring)
+    ring_push(v.ring, TransferCmd{static_cast<uint32_t>(CmdOp::kConsume),
+                                  static_cast<uint32_t>(seq), seq, 0, 0});
 }
 
 __global__ void k_ep_consume_gate(EpView v, uint64_t prev_seq) {
-  if (threadIdx.x < static_cast<unsigned>(v.world) &&
-      !((v.proxy_mask >> threadIdx.x) & 1u)) {
+  // gate on EVERY peer (IPC-signalled or proxy-relayed): the consumed
+  // word for proxied peers is written host-side by the proxy rx loop
+  if (threadIdx.x < static_cast<unsigned>(v.world)) {
     uint64_t const* p = consumed_ptr(v.peers[v.rank], v, threadIdx.x);
     for (uint64_t it = 0;; ++it) {
       if (ld_acquire_sys(p) >= prev_seq) return;

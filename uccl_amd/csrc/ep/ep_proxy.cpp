@@ -16,7 +16,8 @@ namespace uccl {
 namespace ep {
 
 namespace {
-constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3, kBar = 4, kAtom = 5;
+constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3, kBar = 4, kAtom = 5,
+                   kCons = 6;
 constexpr size_t kStageBytes = 16ull << 20;  // pinned staging chunk
 
 // CPU pinning for proxy threads (parity: the reference pins proxies per
@@ -324,6 +325,18 @@ void EpProxy::ring_loop() {
             handle_barrier_arrival(lane, seq);
             break;
           }
+          case CmdOp::kConsume: {
+            // relay "generation a consumed" to every proxied peer so
+            // their next dispatch_send's consume gate can open
+            WireHdr h{kCons, c.seq32, 0, static_cast<uint32_t>(v_.rank),
+                      c.a, 0};
+            for (int r = 0; r < v_.world; ++r) {
+              if (!((proxy_mask_ >> r) & 1u)) continue;
+              TimedFlowLock guard(*flow_mu_[r], "consume");
+              tp_->send_msg(flows_[r], &h, sizeof(h));
+            }
+            break;
+          }
           case CmdOp::kQuiet: {
             // ring cmds execute in FIFO order and ship_rows is
             // synchronous, so reaching this cmd means every prior
@@ -454,6 +467,16 @@ void EpProxy::rx_loop(int peer) {
       tp_->recv_msg(flow, &h, sizeof(h));
       if (h.kind == kBar) {
         handle_barrier_arrival(lane, h.count);
+        continue;
+      }
+      if (h.kind == kCons) {
+        // peer consumed generation h.count of OUR egress: open our next
+        // dispatch_send's consume gate for that peer
+        uint64_t const seq = h.count;
+        UCCL_CHECK_HIP(hipMemcpyAsync(consumed_ptr(heap_, v_, peer), &seq,
+                                      sizeof(seq), hipMemcpyHostToDevice,
+                                      lane.stream));
+        lane_sync(lane.stream, "consume h2d");
         continue;
       }
       if (h.kind == kAtom) {
