@@ -964,7 +964,13 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
             it->second.recv_bytes >= it->second.bytes) ||
            impl_->stop || f.failed;
   };
+  // NB: a posted recv legitimately blocks for as long as the peer has
+  // nothing to send (proxy rx loops idle between generations), so the
+  // stall diagnostic is opt-in — unlike send_msg, where 5s always means
+  // something is wrong.
+  static bool const rx_dbg = env_bool("UCCL_TP_DEBUG_RTO", false);
   while (!f.fcv.wait_for(lk, std::chrono::seconds(5), done_pred)) {
+    if (!rx_dbg) continue;
     auto it = f.rxmsgs.find(msg_id);
     UCCL_LOG_WARN << "recv_msg stalled 5s: flow " << flow << " msg "
                   << msg_id << " cap " << bytes << " known "
