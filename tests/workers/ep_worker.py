@@ -307,32 +307,6 @@ def main():
         assert dn <= 0.06 * max(sn, 1.0), (dn, sn)
         print(f"[rank {rank}] deep_ep compat normal mode OK", flush=True)
 
-    # ---- many-expert shape (256 experts, top-8): plan/wait scaling ----------
-    if os.environ.get("UCCL_TEST_LIGHT", "0") != "1":
-        E2 = 256
-        local_E2 = E2 // world
-        T2 = 64
-        buf2 = uep.Buffer(num_experts=E2, topk=8, hidden=256,
-                          max_tokens=128, dtype=dtype)
-        x5, topk5, w5 = rank_inputs(rank, T2, 256, 8, E2, dtype, seed + 13)
-        rx5, c5 = buf2.dispatch(x5.cuda(), topk5.cuda())
-        torch.cuda.synchronize()
-        all5 = [rank_inputs(r, T2, 256, 8, E2, dtype, seed + 13)
-                for r in range(world)]
-        c5c = c5.cpu()
-        for le in range(0, local_E2, max(1, local_E2 // 8)):  # spot-check
-            e = rank * local_E2 + le
-            for src in range(world):
-                _, st, _ = all5[src]
-                sel = sum(1 for t in range(T2) if (st[t] == e).any())
-                assert c5c[le, src].item() == sel, (e, src)
-        out5 = buf2.combine(rx5.clone(), topk5.cuda(), w5.cuda())
-        torch.cuda.synchronize()
-        buf2._b.check_error()
-        assert out5.shape == (T2, 256)
-        buf2.close()
-        print(f"[rank {rank}] 256-expert top-8 OK", flush=True)
-
     # ---- proxy sync commands (ATOMIC / BARRIER / QUIET) ---------------------
     if os.environ.get("UCCL_EP_FORCE_PROXY", "0") == "1" and world > 1:
         b = buf._b
@@ -351,6 +325,43 @@ def main():
         want = sum(r + 1 for r in range(world) if r != rank)
         assert got == want, (got, want)
         print(f"[rank {rank}] proxy sync cmds OK", flush=True)
+
+    # free the main buffer's proxy threads before the 256-expert
+    # shape: two live proxies x world ranks oversubscribe the small
+    # CPU of a 1-GPU test box enough to trip device wait timeouts.
+    # Barrier first: a rank must not free a heap IPC peers still read.
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+    buf.close()
+
+    # ---- many-expert shape (256 experts, top-8): plan/wait scaling ----------
+    if os.environ.get("UCCL_TEST_LIGHT", "0") != "1":
+        E2 = 256
+        local_E2 = E2 // world
+        T2 = 64
+        buf2 = uep.Buffer(num_experts=E2, topk=8, hidden=256,
+                          max_tokens=128, dtype=dtype)
+        x5, topk5, w5 = rank_inputs(rank, T2, 256, 8, E2, dtype, seed + 13)
+        rx5, c5 = buf2.dispatch(x5.cuda(), topk5.cuda())
+        torch.cuda.synchronize()
+        buf2._b.check_error()  # surface device wait timeouts before asserts
+        all5 = [rank_inputs(r, T2, 256, 8, E2, dtype, seed + 13)
+                for r in range(world)]
+        c5c = c5.cpu()
+        for le in range(0, local_E2, max(1, local_E2 // 8)):  # spot-check
+            e = rank * local_E2 + le
+            for src in range(world):
+                _, st, _ = all5[src]
+                sel = sum(1 for t in range(T2) if (st[t] == e).any())
+                assert c5c[le, src].item() == sel, (e, src)
+        out5 = buf2.combine(rx5.clone(), topk5.cuda(), w5.cuda())
+        torch.cuda.synchronize()
+        buf2._b.check_error()
+        assert out5.shape == (T2, 256)
+        buf2.close()
+        print(f"[rank {rank}] 256-expert top-8 OK", flush=True)
 
     if os.environ.get("UCCL_EP_FORCE_PROXY", "0") != "1":
         fp8_check()
