@@ -32,6 +32,9 @@ def main():
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29572")
+    # proxy lane streams must not share HW queues with spin-wait kernels
+    # (see tests/workers/ep_worker.py header note)
+    os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
 
     import torch
 

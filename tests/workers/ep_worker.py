@@ -13,6 +13,13 @@ import signal
 
 signal.alarm(int(os.environ.get("UCCL_TEST_ALARM", "240")))
 
+# More HW queues than the ROCm default (4): the proxy runs ~6 streams
+# (lane copies) beside the user stream's spin-wait kernels; when streams
+# share a HW queue the copy packets serialize BEHIND a spinning kernel
+# (observed: 5 s D2H stall -> flow-mutex cascade at 4 procs/GPU). Must be
+# set before the HSA runtime initializes, i.e. before importing torch.
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 import torch
 
 
