@@ -8,7 +8,12 @@ cd "$(dirname "$0")/.."
 OUT=${TMPDIR:-/tmp}
 for san in thread address,undefined; do
   tag=${san%%,*}
-  g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_transport.cpp \
+  # TSan builds define UCCL_SAN_NO_TIMED_WAIT everywhere: this libtsan
+  # lacks pthread_cond_clockwait interception, so any cv wait_for reports
+  # phantom double-locks/races (minimal repro verified)
+  tw=""
+  [ "$tag" = thread ] && tw=-DUCCL_SAN_NO_TIMED_WAIT
+  g++ -O1 -g -std=c++17 -fsanitize=$san $tw tools/san_transport.cpp \
       uccl_amd/csrc/transport/reliable.cpp \
       uccl_amd/csrc/transport/udp_fabric.cpp \
       uccl_amd/csrc/transport/verbs_fabric.cpp \
@@ -17,7 +22,7 @@ for san in thread address,undefined; do
   g++ -O1 -g -std=c++17 -fsanitize=$san tools/san_codec_ukernel.cpp \
       uccl_amd/csrc/p2p/compress.cpp uccl_amd/csrc/ukernel/ukernel.cpp \
       uccl_amd/csrc/core/trace.cpp -o "$OUT/san_cu_$tag" -pthread -lz
-  g++ -O1 -g -std=c++17 -fsanitize=$san -D__HIP_PLATFORM_AMD__=1 \
+  g++ -O1 -g -std=c++17 -fsanitize=$san $tw -D__HIP_PLATFORM_AMD__=1 \
       -I/opt/rocm/include tools/san_p2p.cpp \
       uccl_amd/csrc/p2p/endpoint.cpp uccl_amd/csrc/p2p/rccl_plane.cpp \
       uccl_amd/csrc/transport/reliable.cpp \
@@ -29,8 +34,7 @@ for san in thread address,undefined; do
   # build swaps timed cv waits for untimed ones (UCCL_SAN_NO_TIMED_WAIT):
   # this libtsan lacks pthread_cond_clockwait interception, so wait_for
   # reports false double-lock/races (verified with a minimal repro).
-  extra=""
-  [ "$tag" = thread ] && extra=-DUCCL_SAN_NO_TIMED_WAIT
+  extra="$tw"
   g++ -O1 -g -std=c++17 -fsanitize=$san $extra -fPIC -shared \
       uccl_amd/csrc/plugin/tcp_plugin.cpp \
       uccl_amd/csrc/transport/reliable.cpp \

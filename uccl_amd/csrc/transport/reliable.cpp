@@ -919,6 +919,12 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
       lk.lock();
     }
   }
+#ifdef UCCL_SAN_NO_TIMED_WAIT
+  // this libtsan lacks pthread_cond_clockwait interception: wait_for
+  // reports phantom double-locks/races, so TSan builds use the plain
+  // wait (losing only the stall diagnostics)
+  f.fcv.wait(lk, [&] { return m->done || impl_->stop || f.failed; });
+#else
   while (!f.fcv.wait_for(lk, std::chrono::seconds(5), [&] {
     return m->done || impl_->stop || f.failed;
   })) {
@@ -931,6 +937,7 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
                   << f.inflight.size() << " txq " << f.txq.size()
                   << " cwnd " << f.cwnd << " next_csn " << f.next_csn;
   }
+#endif
   if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (!m->done) throw std::runtime_error("transport closed during send");
 }
@@ -968,6 +975,9 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
   // nothing to send (proxy rx loops idle between generations), so the
   // stall diagnostic is opt-in — unlike send_msg, where 5s always means
   // something is wrong.
+#ifdef UCCL_SAN_NO_TIMED_WAIT
+  f.fcv.wait(lk, done_pred);  // see send_msg: libtsan clockwait gap
+#else
   static bool const rx_dbg = env_bool("UCCL_TP_DEBUG_RTO", false);
   while (!f.fcv.wait_for(lk, std::chrono::seconds(5), done_pred)) {
     if (!rx_dbg) continue;
@@ -980,6 +990,7 @@ void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {
                   << (it != f.rxmsgs.end() ? it->second.bytes : 0)
                   << " rx_cum " << f.rx_cum << " ooo " << f.rx_ooo.size();
   }
+#endif
   if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (impl_->stop) throw std::runtime_error("transport closed during recv");
   f.rxmsgs.erase(msg_id);
