@@ -72,6 +72,8 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   v_.off_plan = off;
   off = align256(off + sizeof(uint32_t) * num_experts *
                            (2 + static_cast<size_t>(max_tokens)));
+  v_.off_plan_chunks = off;
+  off = align256(off + sizeof(uint32_t) * num_experts * kPlanChunks);
   size_t const egress_rows = static_cast<size_t>(max_tokens) * topk;
   v_.off_egress = off;
   off = align256(off + egress_rows * hidden * elem_size);

@@ -54,6 +54,11 @@ struct EpView {
   size_t off_comb_x;      // [max_tokens][topk][hidden] elems
   size_t off_plan;        // u32 [num_experts][2 + max_tokens]  (private
                           //   per-rank scratch: count, egress prefix, list)
+  size_t off_plan_chunks; // u32 [num_experts][kPlanChunks] per-(expert,
+                          //   token-chunk) counts, then in-chunk bases
+                          //   (the dispatch plan parallelizes over E*C
+                          //   blocks; one block per expert left 97% of
+                          //   the chip idle at small expert counts)
   size_t off_egress;      // packed egress rows [max_tokens*topk][hidden]
                           //   (proxy mode: rows destined to remote ranks)
   size_t off_egress_meta; // u32 [max_tokens*topk]
@@ -78,6 +83,15 @@ struct EpView {
 };
 
 constexpr uint32_t kMetaTokMask = 0x00ffffffu;
+constexpr int kPlanChunks = 16;  // token chunks per expert in plan build
+
+__host__ __device__ inline uint32_t* plan_chunks_ptr(void* base,
+                                                     const EpView& v,
+                                                     int e) {
+  return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +
+                                     v.off_plan_chunks) +
+         static_cast<size_t>(e) * kPlanChunks;
+}
 
 __host__ __device__ inline uint64_t* disp_count_ptr(void* base,
                                                     const EpView& v, int le,
