@@ -17,7 +17,7 @@ namespace ep {
 
 namespace {
 constexpr uint32_t kDisp = 1, kComb = 2, kCombDone = 3, kBar = 4, kAtom = 5,
-                   kCons = 6;
+                   kCons = 6, kDispB = 7;
 constexpr size_t kStageBytes = 16ull << 20;  // pinned staging chunk
 
 // CPU pinning for proxy threads (parity: the reference pins proxies per
@@ -233,6 +233,62 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
   if (h.count) tp_->send_msg(flow, metas.data(), h.count * sizeof(uint32_t));
 }
 
+// Batched dispatch shipping: one wire transaction for a contiguous run
+// of per-expert egress spans headed to the same peer (hdr, (le,count)
+// pairs, one chunked row stream, one meta block) instead of
+// 3*n_experts blocking sends. This is the latency lever for many-expert
+// shapes: per-message round trips dominated the 256-expert generation
+// time before batching.
+void EpProxy::ship_batch(Lane& lane, uint64_t flow, uint32_t seq,
+                         uint64_t row0, std::vector<uint32_t> const& les,
+                         std::vector<uint32_t> const& cnts,
+                         uint64_t total_rows) {
+  TimedFlowLock guard(*flow_mu_[flow_peer(flow)], "ship_batch");
+  size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
+  size_t const total = total_rows * row_bytes;
+  WireHdr h{kDispB, seq, les[0], static_cast<uint32_t>(v_.rank),
+            total_rows, les.size()};
+  tp_->send_msg(flow, &h, sizeof(h));
+  std::vector<uint32_t> pairs(2 * les.size());
+  for (size_t i = 0; i < les.size(); ++i) {
+    pairs[2 * i] = les[i];
+    pairs[2 * i + 1] = cnts[i];
+  }
+  tp_->send_msg(flow, pairs.data(), pairs.size() * sizeof(uint32_t));
+  char const* dev_rows = egress_row(heap_, v_, row0);
+  void* bufs[2] = {lane.buf, lane.buf2};
+  size_t off = 0;
+  int cur = 0;
+  if (total) {
+    size_t const n0 = std::min(stage_bytes_, total);
+    UCCL_CHECK_HIP(hipMemcpyAsync(bufs[cur], dev_rows, n0,
+                                  hipMemcpyDeviceToHost, lane.stream));
+    lane_sync(lane.stream, "batch first d2h");
+  }
+  while (off < total) {
+    size_t const n = std::min(stage_bytes_, total - off);
+    size_t const next_off = off + n;
+    if (next_off < total) {
+      size_t const n1 = std::min(stage_bytes_, total - next_off);
+      UCCL_CHECK_HIP(hipMemcpyAsync(bufs[cur ^ 1], dev_rows + next_off, n1,
+                                    hipMemcpyDeviceToHost, lane.stream));
+    }
+    tp_->send_msg(flow, bufs[cur], n);
+    if (next_off < total) lane_sync(lane.stream, "batch next d2h");
+    off = next_off;
+    cur ^= 1;
+  }
+  if (total_rows) {
+    std::vector<uint32_t> metas(total_rows);
+    UCCL_CHECK_HIP(hipMemcpyAsync(metas.data(),
+                                  egress_meta(heap_, v_, row0),
+                                  total_rows * sizeof(uint32_t),
+                                  hipMemcpyDeviceToHost, lane.stream));
+    lane_sync(lane.stream, "batch metas d2h");
+    tp_->send_msg(flow, metas.data(), total_rows * sizeof(uint32_t));
+  }
+}
+
 int EpProxy::num_proxy_peers() const {
   int n = 0;
   for (int r = 0; r < v_.world; ++r)
@@ -296,11 +352,46 @@ void EpProxy::ring_loop() {
           case CmdOp::kDispatchWrite: {
             int const e = static_cast<int>(c.a);
             int const dst = e / v_.local_experts;
-            WireHdr h{kDisp, c.seq32,
-                      static_cast<uint32_t>(e % v_.local_experts),
-                      static_cast<uint32_t>(v_.rank), c.c, 0};
-            ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
-                      egress_meta(heap_, v_, c.b), nullptr);
+            // BATCH: fold the maximal run of queued dispatch cmds for
+            // the same (dst, seq) with contiguous egress spans into one
+            // wire transaction — the reference proxy's batched posting
+            // (proxy.cpp:1203 post_gpu_commands_mixed). The publish
+            // kernel emits experts in pfx order, so a generation's
+            // cmds per dst are contiguous and this collapses
+            // 3*local_experts blocking sends into ~3.
+            std::vector<uint32_t> les{static_cast<uint32_t>(
+                e % v_.local_experts)};
+            std::vector<uint32_t> cnts{static_cast<uint32_t>(c.c)};
+            uint64_t row0 = c.b;
+            uint64_t rows = c.c;
+            uint64_t scan = head + 1;
+            while (scan != tail) {
+              TransferCmd n;
+              memcpy(&n,
+                     const_cast<TransferCmd*>(
+                         &ring_->cmds[scan % kRingSlots]),
+                     sizeof(n));
+              if (static_cast<CmdOp>(n.op) != CmdOp::kDispatchWrite ||
+                  n.seq32 != c.seq32 ||
+                  static_cast<int>(n.a) / v_.local_experts != dst ||
+                  n.b != row0 + rows)
+                break;
+              les.push_back(
+                  static_cast<uint32_t>(n.a % v_.local_experts));
+              cnts.push_back(static_cast<uint32_t>(n.c));
+              rows += n.c;
+              ++scan;
+            }
+            if (les.size() > 1) {
+              ship_batch(lane, flows_[dst], c.seq32, row0, les, cnts,
+                         rows);
+              head = scan - 1;  // ++head below completes the batch
+            } else {
+              WireHdr h{kDisp, c.seq32, les[0],
+                        static_cast<uint32_t>(v_.rank), c.c, 0};
+              ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
+                        egress_meta(heap_, v_, c.b), nullptr);
+            }
             break;
           }
           case CmdOp::kAtomicAdd: {
@@ -506,6 +597,63 @@ void EpProxy::rx_loop(int peer) {
                                       sizeof(seq), hipMemcpyHostToDevice,
                                       lane.stream));
         UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+        continue;
+      }
+      if (h.kind == kDispB) {
+        // batched dispatch: (le,count) pairs, then one contiguous row
+        // stream split across the per-expert slot regions, then metas
+        size_t const nles = h.aux;
+        std::vector<uint32_t> pairs(2 * nles);
+        tp_->recv_msg(flow, pairs.data(), pairs.size() * sizeof(uint32_t));
+        // per-le destination spans and cumulative row boundaries
+        std::vector<uint64_t> bound(nles + 1, 0);
+        for (size_t i = 0; i < nles; ++i)
+          bound[i + 1] = bound[i] + pairs[2 * i + 1];
+        size_t const total = h.count * row_bytes;
+        size_t off = 0;
+        size_t li = 0;
+        while (off < total) {
+          size_t const n = std::min(stage_bytes_, total - off);
+          tp_->recv_msg(flow, lane.buf, n);
+          size_t done = 0;
+          while (done < n) {
+            size_t const gpos = off + done;  // global byte pos in stream
+            while (li + 1 < nles && gpos >= bound[li + 1] * row_bytes)
+              ++li;
+            size_t const le_end = bound[li + 1] * row_bytes;
+            size_t const span = std::min(n - done, le_end - gpos);
+            char* dst = disp_x_ptr(heap_, v_, pairs[2 * li],
+                                   static_cast<size_t>(h.src) *
+                                       v_.max_tokens) +
+                        (gpos - bound[li] * row_bytes);
+            UCCL_CHECK_HIP(
+                hipMemcpyAsync(dst, static_cast<char*>(lane.buf) + done,
+                               span, hipMemcpyHostToDevice, lane.stream));
+            done += span;
+          }
+          lane_sync(lane.stream, "batch rows h2d");
+          off += n;
+        }
+        metas.resize(h.count);
+        if (h.count)
+          tp_->recv_msg(flow, metas.data(), h.count * sizeof(uint32_t));
+        for (size_t i = 0; i < nles; ++i) {
+          uint32_t const le = pairs[2 * i];
+          uint64_t const cnt = pairs[2 * i + 1];
+          if (cnt) {
+            UCCL_CHECK_HIP(hipMemcpyAsync(
+                disp_meta_ptr(heap_, v_, le,
+                              static_cast<size_t>(h.src) * v_.max_tokens),
+                metas.data() + bound[i], cnt * sizeof(uint32_t),
+                hipMemcpyHostToDevice, lane.stream));
+          }
+          uint64_t const tagged =
+              (static_cast<uint64_t>(h.seq32) << 32) | cnt;
+          UCCL_CHECK_HIP(hipMemcpyAsync(
+              disp_count_ptr(heap_, v_, le, h.src), &tagged,
+              sizeof(tagged), hipMemcpyHostToDevice, lane.stream));
+          lane_sync(lane.stream, "batch tag h2d");
+        }
         continue;
       }
       size_t const total = h.count * row_bytes;

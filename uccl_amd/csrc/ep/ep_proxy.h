@@ -81,6 +81,9 @@ class EpProxy {
   void ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                  void const* dev_rows, uint32_t const* dev_metas_or_null,
                  std::vector<uint32_t> const* host_metas);
+  void ship_batch(Lane& lane, uint64_t flow, uint32_t seq, uint64_t row0,
+                  std::vector<uint32_t> const& les,
+                  std::vector<uint32_t> const& cnts, uint64_t total_rows);
 
   EpView v_;
   void* heap_;
