@@ -8,6 +8,7 @@
 #include <random>
 #include <shared_mutex>
 #include <stdexcept>
+#include <thread>
 #include <cstdlib>
 #include <cstring>
 
@@ -776,12 +777,14 @@ TransportEndpoint::TransportEndpoint(int num_paths, size_t chunk_bytes)
       static_cast<uint64_t>(env_int("UCCL_TP_RTO_US", 100000)) * 1000;
   impl_->rto_abort_thres =
       static_cast<int>(env_int("UCCL_TP_RTO_ABORT", 50));
-  // Busy-poll engines (reference adaptive-sleep discipline). OPT-IN:
-  // measured 2.1x single-flow message rate, but an unresolved
-  // spin-specific stall was observed under the EP proxy workload on
-  // many-core GPU hosts (one path's chunk unacked through 50 RTOs), so
-  // correctness keeps the default off until that is root-caused.
-  impl_->spin = env_bool("UCCL_TP_SPIN", false);
+  // Busy-poll engines (reference adaptive-sleep discipline): 2.1x
+  // single-flow message rate measured. Default ON only where cores are
+  // plentiful — burning a core per engine on a small host starves the
+  // proxies/workers that share it. (The stall once blamed on spin was
+  // root-caused to the proxied consume-gate race + RTO path pinning,
+  // both fixed; spin itself was exonerated by the same logs.)
+  impl_->spin = env_bool(
+      "UCCL_TP_SPIN", std::thread::hardware_concurrency() >= 32);
   // optional sender pacing (the reference's Carousel timing-wheel role;
   // bypassed by default there and here — BYPASS_PACING=1)
   if (int64_t mbps = env_int("UCCL_TP_PACE_MBPS", 0); mbps > 0)
