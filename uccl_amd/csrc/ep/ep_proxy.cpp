@@ -656,14 +656,28 @@ void EpProxy::rx_loop(int peer) {
       }
       if (h.kind == kDispB) {
         // batched dispatch: (le,count) pairs, then one contiguous row
-        // stream split across the per-expert slot regions, then metas
+        // stream split across the per-expert slot regions, then metas.
+        // Every wire-controlled value is bounded before it indexes the
+        // heap (same discipline as the transport's handle_data).
         size_t const nles = h.aux;
+        UCCL_CHECK(nles >= 1 &&
+                   nles <= static_cast<size_t>(v_.local_experts))
+            << "dispatch batch nles " << nles;
         std::vector<uint32_t> pairs(2 * nles);
         tp_->recv_msg(flow, pairs.data(), pairs.size() * sizeof(uint32_t));
         // per-le destination spans and cumulative row boundaries
         std::vector<uint64_t> bound(nles + 1, 0);
-        for (size_t i = 0; i < nles; ++i)
+        for (size_t i = 0; i < nles; ++i) {
+          UCCL_CHECK(pairs[2 * i] <
+                         static_cast<uint32_t>(v_.local_experts) &&
+                     pairs[2 * i + 1] <=
+                         static_cast<uint32_t>(v_.max_tokens))
+              << "dispatch batch pair (" << pairs[2 * i] << ","
+              << pairs[2 * i + 1] << ")";
           bound[i + 1] = bound[i] + pairs[2 * i + 1];
+        }
+        UCCL_CHECK(bound[nles] == h.count)
+            << "dispatch batch rows " << h.count << " != " << bound[nles];
         size_t const total = h.count * row_bytes;
         size_t off = 0;
         size_t li = 0;
