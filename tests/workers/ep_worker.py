@@ -370,8 +370,7 @@ def main():
         buf2.close()
         print(f"[rank {rank}] 256-expert top-8 OK", flush=True)
 
-    if os.environ.get("UCCL_EP_FORCE_PROXY", "0") != "1":
-        fp8_check()
+    fp8_check()  # runs under the proxy too: quantized egress + scales
     print(f"[rank {rank}] EP ALL OK", flush=True)
     if world > 1:
         import torch.distributed as dist

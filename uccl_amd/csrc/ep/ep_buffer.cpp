@@ -126,8 +126,9 @@ EpBuffer::EpBuffer(int rank, int world, int device, int num_experts,
   v_.peers[rank_] = heap_;
   v_.proxy_mask = 0;
   v_.ring = nullptr;
-  UCCL_CHECK(!(use_fp8 && env_bool("UCCL_EP_FORCE_PROXY", false)))
-      << "fp8 dispatch not yet supported on the proxy path";
+  // fp8 + proxy: the copy kernel quantizes into split egress streams
+  // (rows + scales) and the proxy ships both — half the wire bytes of
+  // raw bf16 (see egress_x_fp8/egress_scale_fp8).
   if (env_bool("UCCL_EP_FORCE_PROXY", false) && world_ > 1) {
     UCCL_CHECK_HIP(hipHostMalloc(&ring_host_, sizeof(D2HRing),
                                  hipHostMallocMapped));

@@ -337,8 +337,18 @@ __global__ void k_ep_dispatch_copy(EpView v, void const* __restrict__ x) {
     char const* src_row =
         static_cast<char const*>(x) + static_cast<size_t>(t) * row_bytes;
     if (via_proxy) {
-      // stage into packed local egress; the CPU proxy ships it
-      block_copy(egress_row(me, v, pfx + i), src_row, row_bytes);
+      // stage into packed local egress; the CPU proxy ships it. fp8
+      // mode quantizes HERE so the wire carries 1B/elem + scales
+      // (half the D2H + transport bytes of raw bf16).
+      if (v.disp_fp8) {
+        __shared__ float redp[256];
+        block_quant_row_fp8(
+            egress_x_fp8(me, v, pfx + i), egress_scale_fp8(me, v, pfx + i),
+            reinterpret_cast<__hip_bfloat16 const*>(src_row), v.hidden,
+            redp);
+      } else {
+        block_copy(egress_row(me, v, pfx + i), src_row, row_bytes);
+      }
       if (threadIdx.x == 0) *egress_meta(me, v, pfx + i) = tk;
     } else {
       size_t const slot = static_cast<size_t>(v.rank) * v.max_tokens + i;

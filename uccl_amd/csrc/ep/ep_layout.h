@@ -137,6 +137,26 @@ __host__ __device__ inline char* egress_row(void* base, const EpView& v,
          row * static_cast<size_t>(v.hidden) * v.elem_size;
 }
 
+// fp8 proxy egress: quantized rows and their per-128 scales live as two
+// SEPARATE contiguous streams inside the (bf16-sized, so 2x larger than
+// needed) egress region — the receiver then lands each with one
+// contiguous H2D per expert span instead of per-row unpacking.
+__host__ __device__ inline char* egress_x_fp8(void* base, const EpView& v,
+                                              size_t row) {
+  return static_cast<char*>(base) + v.off_egress +
+         row * static_cast<size_t>(v.hidden);
+}
+
+__host__ __device__ inline float* egress_scale_fp8(void* base,
+                                                   const EpView& v,
+                                                   size_t row) {
+  size_t const nrows =
+      static_cast<size_t>(v.max_tokens) * v.topk;  // region capacity
+  return reinterpret_cast<float*>(static_cast<char*>(base) + v.off_egress +
+                                  nrows * static_cast<size_t>(v.hidden)) +
+         row * (v.hidden / 128);
+}
+
 __host__ __device__ inline uint32_t* egress_meta(void* base, const EpView& v,
                                                  size_t row) {
   return reinterpret_cast<uint32_t*>(static_cast<char*>(base) +

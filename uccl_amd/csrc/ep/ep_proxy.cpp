@@ -189,9 +189,11 @@ void EpProxy::start() {
 void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                         void const* dev_rows,
                         uint32_t const* dev_metas_or_null,
-                        std::vector<uint32_t> const* host_metas) {
+                        std::vector<uint32_t> const* host_metas,
+                        size_t row_bytes, void const* dev_scales,
+                        size_t scale_row_bytes) {
   TimedFlowLock guard(*flow_mu_[flow_peer(flow)], "ship_rows");
-  size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
+  if (!row_bytes) row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
   size_t const total = h.count * row_bytes;
   tp_->send_msg(flow, &h, sizeof(h));
   // double-buffered pipeline: the D2H copy of chunk i+1 runs while the
@@ -220,6 +222,20 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
     off = next_off;
     cur ^= 1;
   }
+  if (dev_scales && h.count) {
+    // fp8 egress: second contiguous stream of per-row scales
+    size_t const stotal = h.count * scale_row_bytes;
+    size_t soff = 0;
+    while (soff < stotal) {
+      size_t const n = std::min(stage_bytes_, stotal - soff);
+      UCCL_CHECK_HIP(hipMemcpyAsync(
+          lane.buf, static_cast<char const*>(dev_scales) + soff, n,
+          hipMemcpyDeviceToHost, lane.stream));
+      lane_sync(lane.stream, "ship scales d2h");
+      tp_->send_msg(flow, lane.buf, n);
+      soff += n;
+    }
+  }
   std::vector<uint32_t> metas;
   if (host_metas) {
     metas = *host_metas;
@@ -244,7 +260,9 @@ void EpProxy::ship_batch(Lane& lane, uint64_t flow, uint32_t seq,
                          std::vector<uint32_t> const& cnts,
                          uint64_t total_rows) {
   TimedFlowLock guard(*flow_mu_[flow_peer(flow)], "ship_batch");
-  size_t const row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
+  size_t const row_bytes =
+      v_.disp_fp8 ? static_cast<size_t>(v_.hidden)
+                  : static_cast<size_t>(v_.hidden) * v_.elem_size;
   size_t const total = total_rows * row_bytes;
   WireHdr h{kDispB, seq, les[0], static_cast<uint32_t>(v_.rank),
             total_rows, les.size()};
@@ -255,7 +273,9 @@ void EpProxy::ship_batch(Lane& lane, uint64_t flow, uint32_t seq,
     pairs[2 * i + 1] = cnts[i];
   }
   tp_->send_msg(flow, pairs.data(), pairs.size() * sizeof(uint32_t));
-  char const* dev_rows = egress_row(heap_, v_, row0);
+  char const* dev_rows = v_.disp_fp8
+                             ? egress_x_fp8(heap_, v_, row0)
+                             : egress_row(heap_, v_, row0);
   void* bufs[2] = {lane.buf, lane.buf2};
   size_t off = 0;
   int cur = 0;
@@ -277,6 +297,21 @@ void EpProxy::ship_batch(Lane& lane, uint64_t flow, uint32_t seq,
     if (next_off < total) lane_sync(lane.stream, "batch next d2h");
     off = next_off;
     cur ^= 1;
+  }
+  if (v_.disp_fp8 && total_rows) {
+    size_t const srow = (static_cast<size_t>(v_.hidden) / 128) * 4;
+    size_t const stotal = total_rows * srow;
+    char const* dev_scales = reinterpret_cast<char const*>(
+        egress_scale_fp8(heap_, v_, row0));
+    size_t soff = 0;
+    while (soff < stotal) {
+      size_t const n = std::min(stage_bytes_, stotal - soff);
+      UCCL_CHECK_HIP(hipMemcpyAsync(lane.buf, dev_scales + soff, n,
+                                    hipMemcpyDeviceToHost, lane.stream));
+      lane_sync(lane.stream, "batch scales d2h");
+      tp_->send_msg(flow, lane.buf, n);
+      soff += n;
+    }
   }
   if (total_rows) {
     std::vector<uint32_t> metas(total_rows);
@@ -389,8 +424,16 @@ void EpProxy::ring_loop() {
             } else {
               WireHdr h{kDisp, c.seq32, les[0],
                         static_cast<uint32_t>(v_.rank), c.c, 0};
-              ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
-                        egress_meta(heap_, v_, c.b), nullptr);
+              if (v_.disp_fp8)
+                ship_rows(lane, flows_[dst], h,
+                          egress_x_fp8(heap_, v_, c.b),
+                          egress_meta(heap_, v_, c.b), nullptr,
+                          static_cast<size_t>(v_.hidden),
+                          egress_scale_fp8(heap_, v_, c.b),
+                          (static_cast<size_t>(v_.hidden) / 128) * 4);
+              else
+                ship_rows(lane, flows_[dst], h, egress_row(heap_, v_, c.b),
+                          egress_meta(heap_, v_, c.b), nullptr);
             }
             break;
           }
@@ -678,31 +721,44 @@ void EpProxy::rx_loop(int peer) {
         }
         UCCL_CHECK(bound[nles] == h.count)
             << "dispatch batch rows " << h.count << " != " << bound[nles];
-        size_t const total = h.count * row_bytes;
-        size_t off = 0;
-        size_t li = 0;
-        while (off < total) {
-          size_t const n = std::min(stage_bytes_, total - off);
-          tp_->recv_msg(flow, lane.buf, n);
-          size_t done = 0;
-          while (done < n) {
-            size_t const gpos = off + done;  // global byte pos in stream
-            while (li + 1 < nles && gpos >= bound[li + 1] * row_bytes)
-              ++li;
-            size_t const le_end = bound[li + 1] * row_bytes;
-            size_t const span = std::min(n - done, le_end - gpos);
-            char* dst = disp_x_ptr(heap_, v_, pairs[2 * li],
-                                   static_cast<size_t>(h.src) *
-                                       v_.max_tokens) +
-                        (gpos - bound[li] * row_bytes);
-            UCCL_CHECK_HIP(
-                hipMemcpyAsync(dst, static_cast<char*>(lane.buf) + done,
-                               span, hipMemcpyHostToDevice, lane.stream));
-            done += span;
+        // receive a contiguous stream of per-row records of size `rsz`
+        // and split it across the per-expert destination spans
+        auto recv_split = [&](size_t rsz, auto dst_of_le) {
+          size_t const total = h.count * rsz;
+          size_t off = 0;
+          size_t li = 0;
+          while (off < total) {
+            size_t const n = std::min(stage_bytes_, total - off);
+            tp_->recv_msg(flow, lane.buf, n);
+            size_t done = 0;
+            while (done < n) {
+              size_t const gpos = off + done;
+              while (li + 1 < nles && gpos >= bound[li + 1] * rsz) ++li;
+              size_t const le_end = bound[li + 1] * rsz;
+              size_t const span = std::min(n - done, le_end - gpos);
+              char* dst = dst_of_le(pairs[2 * li]) +
+                          (gpos - bound[li] * rsz);
+              UCCL_CHECK_HIP(hipMemcpyAsync(
+                  dst, static_cast<char*>(lane.buf) + done, span,
+                  hipMemcpyHostToDevice, lane.stream));
+              done += span;
+            }
+            lane_sync(lane.stream, "batch rows h2d");
+            off += n;
           }
-          lane_sync(lane.stream, "batch rows h2d");
-          off += n;
-        }
+        };
+        size_t const slot0b = static_cast<size_t>(h.src) * v_.max_tokens;
+        size_t const rb =
+            v_.disp_fp8 ? static_cast<size_t>(v_.hidden) : row_bytes;
+        recv_split(rb, [&](uint32_t le) {
+          return disp_x_ptr(heap_, v_, le, slot0b);
+        });
+        if (v_.disp_fp8)
+          recv_split((static_cast<size_t>(v_.hidden) / 128) * 4,
+                     [&](uint32_t le) {
+                       return reinterpret_cast<char*>(
+                           disp_scale_ptr(heap_, v_, le, slot0b));
+                     });
         metas.resize(h.count);
         if (h.count)
           tp_->recv_msg(flow, metas.data(), h.count * sizeof(uint32_t));
@@ -725,7 +781,10 @@ void EpProxy::rx_loop(int peer) {
         }
         continue;
       }
-      size_t const total = h.count * row_bytes;
+      bool const fp8_disp = v_.disp_fp8 && h.kind == kDisp;
+      size_t const rb = fp8_disp ? static_cast<size_t>(v_.hidden)
+                                 : row_bytes;
+      size_t const total = h.count * rb;
       // destination in device memory
       size_t const ing0 = static_cast<size_t>(peer) * v_.max_tokens;
       char* dev_dst = (h.kind == kDisp)
@@ -739,6 +798,21 @@ void EpProxy::rx_loop(int peer) {
         UCCL_CHECK_HIP(hipMemcpyAsync(dev_dst + off, lane.buf, n,
                                       hipMemcpyHostToDevice, lane.stream));
         UCCL_CHECK_HIP(hipStreamSynchronize(lane.stream));
+      }
+      if (fp8_disp && h.count) {
+        // second stream: per-row scales into the contiguous scale span
+        size_t const srow = (static_cast<size_t>(v_.hidden) / 128) * 4;
+        size_t const stotal = h.count * srow;
+        char* sdst = reinterpret_cast<char*>(disp_scale_ptr(
+            heap_, v_, h.le, static_cast<size_t>(h.src) * v_.max_tokens));
+        for (size_t off = 0; off < stotal; off += stage_bytes_) {
+          size_t const n = std::min(stage_bytes_, stotal - off);
+          tp_->recv_msg(flow, lane.buf, n);
+          UCCL_CHECK_HIP(hipMemcpyAsync(sdst + off, lane.buf, n,
+                                        hipMemcpyHostToDevice,
+                                        lane.stream));
+          lane_sync(lane.stream, "disp scales h2d");
+        }
       }
       metas.resize(h.count);
       if (h.count)

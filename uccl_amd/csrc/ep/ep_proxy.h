@@ -80,7 +80,9 @@ class EpProxy {
   int flow_peer(uint64_t flow) const;
   void ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                  void const* dev_rows, uint32_t const* dev_metas_or_null,
-                 std::vector<uint32_t> const* host_metas);
+                 std::vector<uint32_t> const* host_metas,
+                 size_t row_bytes = 0, void const* dev_scales = nullptr,
+                 size_t scale_row_bytes = 0);
   void ship_batch(Lane& lane, uint64_t flow, uint32_t seq, uint64_t row0,
                   std::vector<uint32_t> const& les,
                   std::vector<uint32_t> const& cnts, uint64_t total_rows);
