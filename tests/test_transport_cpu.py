@@ -384,3 +384,52 @@ def test_transport_eqds_paced_under_loss():
     st = a.stats()
     assert st.msgs_sent == 4
     assert st.retransmits + st.rto_retransmits > 0  # loss actually hit
+
+
+def test_transport_async_post_flush():
+    """post_send enqueues without waiting; flush blocks until everything
+    posted on the flow is acked; receivers see posting order."""
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0)
+    msgs = [torch.randint(0, 256, (n,), dtype=torch.uint8)
+            for n in (64, 4096, 1 << 20, 32, 0, 123456)]
+    outs = [torch.zeros_like(m) for m in msgs]
+    done = {}
+
+    def rx():
+        for o in outs:
+            b.recv(fb, o)
+        done["rx"] = True
+
+    t = threading.Thread(target=rx, daemon=True)
+    t.start()
+    for m in msgs:
+        a.post_send(fa, m)  # returns without waiting for acks
+    a.flush(fa)  # all acked from here
+    t.join(timeout=60)
+    assert done.get("rx")
+    for m, o in zip(msgs, outs):
+        assert torch.equal(m, o)
+    assert a.stats().msgs_sent == len(msgs)
+
+
+def test_transport_async_flush_survives_loss():
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=15)
+    msgs = [torch.randint(0, 256, (8192,), dtype=torch.uint8)
+            for _ in range(8)]
+    outs = [torch.zeros_like(m) for m in msgs]
+    done = {}
+
+    def rx():
+        for o in outs:
+            b.recv(fb, o)
+        done["rx"] = True
+
+    t = threading.Thread(target=rx, daemon=True)
+    t.start()
+    for m in msgs:
+        a.post_send(fa, m)
+    a.flush(fa)
+    t.join(timeout=120)
+    assert done.get("rx")
+    for m, o in zip(msgs, outs):
+        assert torch.equal(m, o)

@@ -634,6 +634,21 @@ PYBIND11_MODULE(_C, m) {
              py::gil_scoped_release rel;
              e.send_msg(flow, p, n);
            })
+      .def("post_send",
+           [](TransportEndpoint& e, uint64_t flow, at::Tensor t) {
+             TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),
+                         "transport post_send: host tensors");
+             // caller keeps the tensor alive until flush()
+             void* p = t.data_ptr();
+             size_t n = t.numel() * t.element_size();
+             py::gil_scoped_release rel;
+             e.send_msg_async(flow, p, n);
+           })
+      .def("flush",
+           [](TransportEndpoint& e, uint64_t flow) {
+             py::gil_scoped_release rel;
+             e.flush_sends(flow);
+           })
       .def("recv",
            [](TransportEndpoint& e, uint64_t flow, at::Tensor t) {
              TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),

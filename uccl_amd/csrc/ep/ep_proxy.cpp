@@ -454,21 +454,27 @@ void EpProxy::ring_loop() {
             for (int r = 0; r < v_.world; ++r) {
               if (!((proxy_mask_ >> r) & 1u)) continue;
               TimedFlowLock guard(*flow_mu_[r], "barrier");
-              tp_->send_msg(flows_[r], &h, sizeof(h));
+              tp_->send_msg_async(flows_[r], &h, sizeof(h));
             }
+            for (int r = 0; r < v_.world; ++r)
+              if ((proxy_mask_ >> r) & 1u) tp_->flush_sends(flows_[r]);
             handle_barrier_arrival(lane, seq);
             break;
           }
           case CmdOp::kConsume: {
             // relay "generation a consumed" to every proxied peer so
-            // their next dispatch_send's consume gate can open
+            // their next dispatch_send's consume gate can open. Posted
+            // async to ALL peers, then flushed once — a serial blocking
+            // send per peer paid one transport round trip each.
             WireHdr h{kCons, c.seq32, 0, static_cast<uint32_t>(v_.rank),
                       c.a, 0};
             for (int r = 0; r < v_.world; ++r) {
               if (!((proxy_mask_ >> r) & 1u)) continue;
               TimedFlowLock guard(*flow_mu_[r], "consume");
-              tp_->send_msg(flows_[r], &h, sizeof(h));
+              tp_->send_msg_async(flows_[r], &h, sizeof(h));
             }
+            for (int r = 0; r < v_.world; ++r)
+              if ((proxy_mask_ >> r) & 1u) tp_->flush_sends(flows_[r]);
             break;
           }
           case CmdOp::kQuiet: {

@@ -114,6 +114,7 @@ struct TransportEndpoint::Flow {
   uint32_t last_cum = 0;     // cum of the previous ack (hole detection)
   int hole_dupacks = 0;      // acks with stalled cum + new SACKs above
   uint64_t bytes_chunked = 0;   // cumulative bytes handed to the wire
+  uint64_t tx_unacked_msgs = 0;  // posted (async or blocking), not done
   uint64_t credit_limit = 0;    // eqds: granted chunk-byte budget
   uint64_t bytes_received = 0;  // rx side: cumulative fresh payload
   uint64_t next_send_ns = 0;    // pacing release time (UCCL_TP_PACE_MBPS)
@@ -365,6 +366,7 @@ struct TransportEndpoint::Impl {
     if (m.acked_bytes >= m.bytes && !m.done) {
       m.done = true;
       ++st.msgs_sent;
+      if (f.tx_unacked_msgs) --f.tx_unacked_msgs;
       f.fcv.notify_all();
     }
     f.inflight.erase(it);
@@ -906,6 +908,7 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
     m->ptr = static_cast<char const*>(ptr);
     m->bytes = bytes;
     f.txq.push_back(m);
+    ++f.tx_unacked_msgs;
     impl_->pump_tx(f);
   }
   impl_->wake();
@@ -943,6 +946,46 @@ void TransportEndpoint::send_msg(uint64_t flow, void const* ptr,
 #endif
   if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
   if (!m->done) throw std::runtime_error("transport closed during send");
+}
+
+void TransportEndpoint::send_msg_async(uint64_t flow, void const* ptr,
+                                       size_t bytes) {
+  Flow* fp = impl_->find_flow(flow);
+  UCCL_CHECK(fp != nullptr) << "unknown flow " << flow;
+  Flow& f = *fp;
+  {
+    std::lock_guard<std::mutex> g(f.fmu);
+    if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
+    auto m = std::make_shared<MsgTx>();
+    m->id = f.next_tx_msg++;
+    m->ptr = static_cast<char const*>(ptr);
+    m->bytes = bytes;
+    f.txq.push_back(std::move(m));
+    ++f.tx_unacked_msgs;
+    impl_->pump_tx(f);
+  }
+  impl_->wake();
+}
+
+void TransportEndpoint::flush_sends(uint64_t flow) {
+  Flow* fp = impl_->find_flow(flow);
+  UCCL_CHECK(fp != nullptr) << "unknown flow " << flow;
+  Flow& f = *fp;
+  std::unique_lock<std::mutex> lk(f.fmu);
+  auto done = [&] {
+    return f.tx_unacked_msgs == 0 || impl_->stop || f.failed;
+  };
+#ifdef UCCL_SAN_NO_TIMED_WAIT
+  f.fcv.wait(lk, done);
+#else
+  while (!f.fcv.wait_for(lk, std::chrono::seconds(5), done)) {
+    UCCL_LOG_WARN << "flush_sends stalled 5s: flow " << flow
+                  << " outstanding " << f.tx_unacked_msgs << " inflight "
+                  << f.inflight.size() << " txq " << f.txq.size();
+  }
+#endif
+  if (f.failed) throw std::runtime_error("transport flow failed (RTO abort)");
+  if (impl_->stop) throw std::runtime_error("transport closed during flush");
 }
 
 void TransportEndpoint::recv_msg(uint64_t flow, void* ptr, size_t bytes) {

@@ -70,6 +70,16 @@ class TransportEndpoint {
   void send_msg(uint64_t flow, void const* ptr, size_t bytes);
   void recv_msg(uint64_t flow, void* ptr, size_t bytes);
 
+  // Async posting (the reference's post-then-poll-CQ discipline,
+  // collective/rdma/transport.cc post_send/uc_poll_cq): enqueue without
+  // waiting for acks. `ptr` must stay valid until flush_sends() returns.
+  // Message order on the flow is the posting order, identical to
+  // blocking sends — receivers cannot tell the difference.
+  void send_msg_async(uint64_t flow, void const* ptr, size_t bytes);
+  // Wait until every message posted on `flow` (async or blocking) has
+  // been fully acked; throws if the flow failed.
+  void flush_sends(uint64_t flow);
+
   Stats stats() const;
 
   // Unblock all pending send/recv (they throw std::runtime_error) and stop
