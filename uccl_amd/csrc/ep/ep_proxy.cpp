@@ -195,7 +195,9 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
   TimedFlowLock guard(*flow_mu_[flow_peer(flow)], "ship_rows");
   if (!row_bytes) row_bytes = static_cast<size_t>(v_.hidden) * v_.elem_size;
   size_t const total = h.count * row_bytes;
-  tp_->send_msg(flow, &h, sizeof(h));
+  // control messages post async (acked while the payload streams);
+  // flush_sends() before return bounds every lifetime to this call
+  tp_->send_msg_async(flow, &h, sizeof(h));
   // double-buffered pipeline: the D2H copy of chunk i+1 runs while the
   // transport ships chunk i (the reference's batched-posting role,
   // proxy.cpp:1203 post_gpu_commands_mixed)
@@ -246,7 +248,9 @@ void EpProxy::ship_rows(Lane& lane, uint64_t flow, WireHdr const& h,
                                   hipMemcpyDeviceToHost, lane.stream));
     lane_sync(lane.stream, "ship metas d2h");
   }
-  if (h.count) tp_->send_msg(flow, metas.data(), h.count * sizeof(uint32_t));
+  if (h.count)
+    tp_->send_msg_async(flow, metas.data(), h.count * sizeof(uint32_t));
+  tp_->flush_sends(flow);
 }
 
 // Batched dispatch shipping: one wire transaction for a contiguous run
@@ -266,13 +270,13 @@ void EpProxy::ship_batch(Lane& lane, uint64_t flow, uint32_t seq,
   size_t const total = total_rows * row_bytes;
   WireHdr h{kDispB, seq, les[0], static_cast<uint32_t>(v_.rank),
             total_rows, les.size()};
-  tp_->send_msg(flow, &h, sizeof(h));
+  tp_->send_msg_async(flow, &h, sizeof(h));
   std::vector<uint32_t> pairs(2 * les.size());
   for (size_t i = 0; i < les.size(); ++i) {
     pairs[2 * i] = les[i];
     pairs[2 * i + 1] = cnts[i];
   }
-  tp_->send_msg(flow, pairs.data(), pairs.size() * sizeof(uint32_t));
+  tp_->send_msg_async(flow, pairs.data(), pairs.size() * sizeof(uint32_t));
   char const* dev_rows = v_.disp_fp8
                              ? egress_x_fp8(heap_, v_, row0)
                              : egress_row(heap_, v_, row0);
@@ -320,8 +324,9 @@ void EpProxy::ship_batch(Lane& lane, uint64_t flow, uint32_t seq,
                                   total_rows * sizeof(uint32_t),
                                   hipMemcpyDeviceToHost, lane.stream));
     lane_sync(lane.stream, "batch metas d2h");
-    tp_->send_msg(flow, metas.data(), total_rows * sizeof(uint32_t));
+    tp_->send_msg_async(flow, metas.data(), total_rows * sizeof(uint32_t));
   }
+  tp_->flush_sends(flow);
 }
 
 int EpProxy::num_proxy_peers() const {
@@ -600,10 +605,12 @@ void EpProxy::comb_tx_loop() {
                     static_cast<uint32_t>(le0),
                     static_cast<uint32_t>(v_.rank), rows};
           TimedFlowLock guard(*flow_mu_[src], "comb_batch");
-          tp_->send_msg(flows_[src], &h, sizeof(h));
-          tp_->send_msg(flows_[src], lane.buf, rows * row_bytes);
-          tp_->send_msg(flows_[src], metas.data(),
-                        rows * sizeof(uint32_t));
+          tp_->send_msg_async(flows_[src], &h, sizeof(h));
+          tp_->send_msg_async(flows_[src], lane.buf, rows * row_bytes);
+          tp_->send_msg_async(flows_[src], metas.data(),
+                              rows * sizeof(uint32_t));
+          // flush bounds h/metas/lane.buf lifetimes before group reuse
+          tp_->flush_sends(flows_[src]);
         }
         WireHdr done{kCombDone, static_cast<uint32_t>(task.seq), 0,
                      static_cast<uint32_t>(v_.rank), 0};
