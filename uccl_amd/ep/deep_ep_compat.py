@@ -275,6 +275,7 @@ class Buffer:
         h = _Handle(topk_idx, T, self._plan_gen)
         h.counts = chunk_meta[0][1] if len(chunk_meta) == 1 else None
         h.chunks = chunk_meta
+        h.topk_w = topk_weights
         return (recv_x, recv_topk_idx, recv_topk_weights, per_expert, h,
                 _event())
 
@@ -295,7 +296,21 @@ class Buffer:
         outs = []
         off = 0
         tok0 = 0
+        multi = len(handle.chunks) > 1
         for chunk_T, cnts in handle.chunks:
+            if multi:
+                # the device's slot metadata holds only ONE generation;
+                # later chunk dispatches overwrote the earlier ones.
+                # Re-run the routing round (payload ignored) so this
+                # chunk's return walk sees its own metas/counts — a
+                # collective step, symmetric on every rank.
+                tslice = handle.topk_idx[tok0:tok0 + chunk_T]
+                wslice = (handle.topk_w[tok0:tok0 + chunk_T]
+                          if handle.topk_w is not None else None)
+                dummy = torch.zeros(chunk_T, nb.hidden, dtype=x.dtype,
+                                    device=x.device)
+                nb.nrm_dispatch(dummy, tslice, wslice)
+                torch.cuda.current_stream().synchronize()
             buf = torch.zeros(nb.world, nb.max_tokens, nb.hidden,
                               dtype=x.dtype, device=x.device)
             for r in range(nb.world):
