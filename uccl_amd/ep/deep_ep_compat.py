@@ -233,35 +233,18 @@ class Buffer:
             nb = self._ensure(x.shape[1], max(256, T), num_experts, K,
                               x.dtype, False)
         nb = self._native
-        # chunked HT mode: a prefill batch larger than the buffer's
-        # max_tokens runs as ceil(T/max_tokens) back-to-back generations
-        # (the consume gate makes consecutive dispatches safe); rows are
-        # materialized out of the slot arrays between chunks, so each
-        # generation may reuse them. All ranks chunk by the SAME
-        # max_tokens, so generation counts stay aligned across ranks.
-        chunk_cap = nb.max_tokens
+        rx, counts, rtopk, rw = nb.nrm_dispatch(x, topk_idx, topk_weights)
+        torch.cuda.current_stream().synchronize()
+        cnts = counts.tolist()
         rows_x, rows_topk, rows_w = [], [], []
-        chunk_meta = []  # [(chunk_T, counts_per_rank)] for combine()
-        for t0 in range(0, T, chunk_cap):
-            t1 = min(t0 + chunk_cap, T)
-            rx, counts, rtopk, rw = nb.nrm_dispatch(
-                x[t0:t1], topk_idx[t0:t1],
-                topk_weights[t0:t1] if topk_weights is not None else None)
-            torch.cuda.current_stream().synchronize()
-            cnts = counts.tolist()
-            chunk_meta.append((t1 - t0, cnts))
-            for r in range(nb.world):
-                n = cnts[r]
-                rows_x.append(rx[r, :n].clone())
-                rows_topk.append(rtopk[r, :n].clone())
-                rows_w.append(rw[r, :n].clone())
-        recv_x = torch.cat(rows_x) if rows_x else torch.empty(
-            0, nb.hidden, dtype=x.dtype, device=x.device)
-        gtopk = (torch.cat(rows_topk) if rows_topk else
-                 torch.empty(0, K, dtype=topk_idx.dtype,
-                             device=x.device))
-        recv_w = (torch.cat(rows_w) if rows_w else
-                  torch.empty(0, K, dtype=torch.float32, device=x.device))
+        for r in range(nb.world):
+            n = cnts[r]
+            rows_x.append(rx[r, :n])
+            rows_topk.append(rtopk[r, :n])
+            rows_w.append(rw[r, :n])
+        recv_x = torch.cat(rows_x) if rows_x else rx[0, :0]
+        gtopk = torch.cat(rows_topk) if rows_topk else rtopk[0, :0]
+        recv_w = torch.cat(rows_w) if rows_w else rw[0, :0]
         # global -> local expert ids; -1 for other ranks' experts
         local_lo = nb.rank * nb.local_experts
         local_hi = local_lo + nb.local_experts
@@ -273,9 +256,7 @@ class Buffer:
         per_expert = [int((recv_topk_idx == e).sum())
                       for e in range(nb.local_experts)]
         h = _Handle(topk_idx, T, self._plan_gen)
-        h.counts = chunk_meta[0][1] if len(chunk_meta) == 1 else None
-        h.chunks = chunk_meta
-        h.topk_w = topk_weights
+        h.counts = cnts
         return (recv_x, recv_topk_idx, recv_topk_weights, per_expert, h,
                 _event())
 
@@ -288,39 +269,17 @@ class Buffer:
         each row returns to its source, which sums over contributing
         ranks. Returns (combined_x, event)."""
         nb = self._native
-        assert nb is not None and hasattr(handle, "chunks")
-        # per chunk: scatter that chunk's rows back into the
-        # [world, max_tokens] slot layout the return kernel walks, run
-        # the return round, and stitch the per-chunk outputs back in
-        # token order (chunks are token-contiguous).
-        outs = []
+        assert nb is not None and hasattr(handle, "counts")
+        # scatter the concatenated rows back into the [world, max_tokens]
+        # slot layout the return kernel walks
+        buf = torch.zeros(nb.world, nb.max_tokens, nb.hidden,
+                          dtype=x.dtype, device=x.device)
         off = 0
-        tok0 = 0
-        multi = len(handle.chunks) > 1
-        for chunk_T, cnts in handle.chunks:
-            if multi:
-                # the device's slot metadata holds only ONE generation;
-                # later chunk dispatches overwrote the earlier ones.
-                # Re-run the routing round (payload ignored) so this
-                # chunk's return walk sees its own metas/counts — a
-                # collective step, symmetric on every rank.
-                tslice = handle.topk_idx[tok0:tok0 + chunk_T]
-                wslice = (handle.topk_w[tok0:tok0 + chunk_T]
-                          if handle.topk_w is not None else None)
-                dummy = torch.zeros(chunk_T, nb.hidden, dtype=x.dtype,
-                                    device=x.device)
-                nb.nrm_dispatch(dummy, tslice, wslice)
-                torch.cuda.current_stream().synchronize()
-            buf = torch.zeros(nb.world, nb.max_tokens, nb.hidden,
-                              dtype=x.dtype, device=x.device)
-            for r in range(nb.world):
-                n = cnts[r]
-                buf[r, :n] = x[off:off + n]
-                off += n
-            outs.append(nb.nrm_combine(
-                buf, handle.topk_idx[tok0:tok0 + chunk_T]))
-            tok0 += chunk_T
-        combined = outs[0] if len(outs) == 1 else torch.cat(outs)
+        for r in range(nb.world):
+            n = handle.counts[r]
+            buf[r, :n] = x[off:off + n]
+            off += n
+        combined = nb.nrm_combine(buf, handle.topk_idx)
         return combined, _event()
 
     # low-latency aliases for the internode entry points (the proxy path
