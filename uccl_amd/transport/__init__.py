@@ -10,6 +10,8 @@ deterministic loss injection.
     a = TransportEndpoint(num_paths=8, chunk_bytes=8192)
     flow = a.connect(peer_metadata, tag=rank)   # or a.accept()
     a.send(flow, host_tensor); a.recv(flow, host_tensor)
+    a.post_send(flow, t)   # async: keep t alive until a.flush(flow)
+    a.flush(flow)          # wait until everything posted is acked
     st = a.stats()   # counters + cwnd/srtt + RTT p50/p99
 
 Knobs (env; interface selection additionally honors the
