@@ -433,3 +433,20 @@ def test_transport_async_flush_survives_loss():
     assert done.get("rx")
     for m, o in zip(msgs, outs):
         assert torch.equal(m, o)
+
+
+def test_transport_flush_after_close_raises():
+    """Posting then closing the flow must fail the flush (and any
+    blocked senders) instead of hanging."""
+    C, a, b, fa, fb = make_pair(UCCL_TP_LOSS_PCT=0)
+    t = torch.randint(0, 256, (4096,), dtype=torch.uint8)
+    o = torch.zeros_like(t)
+    a.post_send(fa, t)
+    b.recv(fb, o)  # drain so the flow is idle
+    a.flush(fa)
+    assert torch.equal(t, o)
+    a.close_flow(fa)
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        a.post_send(fa, t)
+        a.flush(fa)

@@ -649,6 +649,11 @@ PYBIND11_MODULE(_C, m) {
              py::gil_scoped_release rel;
              e.flush_sends(flow);
            })
+      .def("close_flow",
+           [](TransportEndpoint& e, uint64_t flow) {
+             py::gil_scoped_release rel;
+             e.close_flow(flow);
+           })
       .def("recv",
            [](TransportEndpoint& e, uint64_t flow, at::Tensor t) {
              TORCH_CHECK(t.is_contiguous() && !t.is_cuda(),
