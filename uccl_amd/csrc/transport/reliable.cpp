@@ -558,7 +558,7 @@ struct TransportEndpoint::Impl {
                     "[rto] flow=%llx csn=%u path=%d age_ms=%.1f att=%u "
                     "infl=%zu\n",
                     (unsigned long long)fid, csn,
-                    static_cast<int>(csn % f.num_paths),
+                    static_cast<int>((csn + c.attempts) % f.num_paths),
                     (now - c.send_ts) / 1e6, c.attempts, f.inflight.size());
           f.cwnd = std::max(2.0, f.cwnd / 2);
           send_chunk(f, csn, c);
