@@ -21,6 +21,8 @@ int main() {
     setenv("UCCL_TP_CC", round % 3 == 0 ? "timely"
                          : round % 3 == 1 ? "swift" : "eqds", 1);
     setenv("UCCL_TP_LOSS_PCT", round % 2 ? "3" : "0", 1);
+    // exercise the paced path + timing wheel under the sanitizers
+    setenv("UCCL_TP_PACE_MBPS", round % 3 == 1 ? "200" : "0", 1);
     TransportEndpoint a(2, 4096), b(2, 4096);
     uint64_t fb = 0;
     std::thread acc([&] { fb = b.accept(nullptr); });

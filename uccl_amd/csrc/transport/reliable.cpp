@@ -14,6 +14,7 @@
 
 #include "../core/env.h"
 #include "../core/latency.h"
+#include "../core/timing_wheel.h"
 #include "../core/trace.h"
 #include "../core/log.h"
 #include "../core/net.h"
@@ -140,6 +141,7 @@ struct TransportEndpoint::Impl {
   int loss_pct;
   int ack_loss_pct;
   uint64_t pace_q32 = 0;  // ns-per-byte in Q32 (0 = pacing bypassed)
+  TimingWheel wheel;      // Carousel-style release of paced flows
   // CC configuration, resolved per endpoint at construction (statics
   // would freeze the first process-wide value and break test isolation)
   std::string cc_mode = "timely";
@@ -237,7 +239,12 @@ struct TransportEndpoint::Impl {
       if (!fabric->tx_ready(f.id, f.txq.front()->id)) break;
       if (pace_q32) {
         uint64_t const now = now_ns();
-        if (f.next_send_ns > now) break;  // paced: retry on a later pump
+        if (f.next_send_ns > now) {
+          // paced: file the flow on the timing wheel; engine-0 releases
+          // it at (about) next_send_ns instead of the coarse 1ms rescan
+          wheel.schedule(f.id, f.next_send_ns);
+          break;
+        }
       }
       auto m = f.txq.front();
       uint32_t const csn = f.next_csn++;
@@ -604,6 +611,15 @@ struct TransportEndpoint::Impl {
       }
       (void)got;
       if (eng != 0) continue;
+      // paced-flow release (cheap no-op while the wheel is empty)
+      if (pace_q32 && !wheel.empty()) {
+        wheel.advance(now_ns(), [this](uint64_t fid) {
+          Flow* f = find_flow(fid);
+          if (!f) return;
+          std::lock_guard<std::mutex> fg(f->fmu);
+          if (!f->failed) pump_tx(*f);
+        });
+      }
       // timers only need ~1ms granularity even when spinning
       uint64_t const tnow = now_ns();
       if (spin && tnow - last_timers < 1000000ull) continue;
