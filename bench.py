@@ -114,19 +114,38 @@ def main():
                           "got": float(chk[0])}))
         sys.exit(2)
     if args.symmetric and world > 1:
+        # The sym-vs-staged decision MUST be collective: a single rank
+        # falling back alone would run a different flag protocol than its
+        # peers and wedge the scaling run. Consensus in two steps so the
+        # collective gate itself is only entered by ALL ranks together:
+        #   1) every rank allocates the probe tensor; gloo-MIN agreement
+        #   2) all ranks run the collective gate; gloo-MIN on the result
+        def all_agree(ok: bool) -> bool:
+            flag = torch.tensor([1 if ok else 0])
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            return bool(flag[0])
+
+        schk = None
         try:
             schk = comm.symmetric_tensor([1 << 20], torch.bfloat16)
-            sym_ok = check_ones(schk)
         except Exception as e:
-            sym_ok = False  # engine refused the symmetric path
-            print(f"[bench rank {rank}] symmetric gate RAISED: {e}",
+            print(f"[bench rank {rank}] symmetric alloc RAISED: {e}",
                   file=sys.stderr, flush=True)
+        sym_ok = all_agree(schk is not None)
+        if sym_ok:
+            try:
+                sym_ok = check_ones(schk)
+            except Exception as e:
+                sym_ok = False
+                print(f"[bench rank {rank}] symmetric gate RAISED: {e}",
+                      file=sys.stderr, flush=True)
+            sym_ok = all_agree(sym_ok)
         if not sym_ok:
             # per-rank diagnostic so a multi-GPU bring-up shows WHICH rank
             # failed the zero-copy gate and the run is auditable
-            print(f"[bench rank {rank}] symmetric path gate failed; "
-                  "falling back to the staged engine", file=sys.stderr,
-                  flush=True)
+            print(f"[bench rank {rank}] symmetric path gate failed "
+                  "somewhere in the job; all ranks falling back to the "
+                  "staged engine together", file=sys.stderr, flush=True)
             args.symmetric = False
             t = torch.randn(count, dtype=torch.bfloat16, device="cuda")
 
